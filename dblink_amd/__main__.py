@@ -1,0 +1,4 @@
+from dblink_amd.api.cli import main
+
+if __name__ == "__main__":
+    raise SystemExit(main())

@@ -1,0 +1,34 @@
+"""Command-line entry: ``python -m dblink_amd <config.conf>``.
+
+Single-process by default; multi-GPU via
+``torchrun --standalone --nproc-per-node N -m dblink_amd <config.conf>``
+(one rank per MI355X GPU over RCCL).
+"""
+
+from __future__ import annotations
+
+import argparse
+import logging
+import sys
+
+
+def main(argv=None):
+    parser = argparse.ArgumentParser(
+        prog="dblink_amd", description="MI355X-native distributed Bayesian entity resolution"
+    )
+    parser.add_argument("config", help="path to HOCON configuration file")
+    parser.add_argument("-v", "--verbose", action="store_true")
+    args = parser.parse_args(argv)
+
+    logging.basicConfig(
+        level=logging.DEBUG if args.verbose else logging.INFO,
+        format="%(asctime)s %(levelname)s %(name)s: %(message)s",
+    )
+    from .project import run_config
+
+    run_config(args.config)
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,73 @@
+"""Native-op dispatch layer.
+
+Loads the in-tree native extension ``dblink_amd._C`` (C++/OpenMP host paths +
+HIP gfx950 device kernels, built by ``setup.py build_ext --inplace`` or
+``__graft_entry__.build()``).
+
+Policy:
+- on CPU, pure-Python/numpy fallbacks are allowed (used by tests as oracles
+  and when the extension has not been built yet);
+- on GPU (CUDA/HIP tensors), the native extension is REQUIRED — ops raise
+  rather than silently falling back to an eager path.
+"""
+
+from __future__ import annotations
+
+import math
+
+import numpy as np
+
+_C = None
+_C_ERR = None
+try:  # torch must be imported first so the extension can link against it
+    import torch  # noqa: F401
+    from dblink_amd import _C as _C  # type: ignore
+except Exception as e:  # pragma: no cover - exercised only without built ext
+    _C = None
+    _C_ERR = e
+
+
+def have_native() -> bool:
+    return _C is not None
+
+
+def native():
+    if _C is None:
+        raise RuntimeError(
+            f"dblink_amd._C native extension is required but not available: {_C_ERR}\n"
+            "Build it with `python setup.py build_ext --inplace`."
+        )
+    return _C
+
+
+def sim_pairs(values, similarity_fn):
+    """Build the sparse exp-similarity index over a domain of strings.
+
+    Returns a ``SimIndexCSR``. Replaces the reference's Spark ``cartesian``
+    V x V sweep (``AttributeIndex.scala:219-231``) with a length-pruned
+    banded Levenshtein pass.
+    """
+    from ..models.attribute_index import SimIndexCSR, _python_sim_pairs
+
+    if similarity_fn.is_constant:
+        n = len(values)
+        return SimIndexCSR(np.zeros(n + 1, dtype=np.int64), np.empty(0, np.int32), np.empty(0))
+
+    if _C is not None and hasattr(_C, "sim_pairs_cpu"):
+        enc = [v.encode("utf-8", "surrogatepass") for v in values]
+        lens = np.array([len(e) for e in enc], dtype=np.int32)
+        maxlen = int(lens.max()) if len(lens) else 0
+        buf = np.zeros((len(enc), max(maxlen, 1)), dtype=np.uint8)
+        for i, e in enumerate(enc):
+            buf[i, : len(e)] = np.frombuffer(e, dtype=np.uint8)
+        import torch
+
+        row_ptr, col, expsim = _C.sim_pairs_cpu(
+            torch.from_numpy(buf),
+            torch.from_numpy(lens),
+            float(similarity_fn.threshold),
+            float(similarity_fn.max_similarity),
+        )
+        return SimIndexCSR(row_ptr.numpy().astype(np.int64), col.numpy(), expsim.numpy())
+
+    return _python_sim_pairs(values, similarity_fn)

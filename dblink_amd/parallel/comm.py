@@ -1,0 +1,132 @@
+"""Distributed communication layer: one process per GPU over RCCL.
+
+Replaces the reference's Spark shuffle / broadcast / accumulators
+(``SURVEY.md §2.3``) with torch.distributed collectives:
+
+- summary reduction  -> all_reduce on a small packed tensor (RCCL ring is fine)
+- cluster migration  -> all_to_all_v over xGMI (direct pairwise on RCCL)
+- theta              -> recomputed redundantly on every rank from the reduced
+                        counts with the same Philox stream (no broadcast needed)
+
+Backend: "nccl" (= RCCL on ROCm) when CUDA devices are available, else
+"gloo" for CPU multi-process tests. gloo has no all_to_all, so the wrapper
+falls back to batched isend/irecv there.
+"""
+
+from __future__ import annotations
+
+import datetime
+import os
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+
+def init_from_env(backend=None):
+    """Initialize the default process group from torchrun env vars.
+
+    Returns (rank, world_size, device). Safe to call in single-process mode
+    (no env) — returns (0, 1, cpu/cuda:0) without creating a group.
+    """
+    if "RANK" not in os.environ or int(os.environ.get("WORLD_SIZE", "1")) <= 1:
+        if torch.cuda.is_available():
+            return 0, 1, torch.device("cuda", 0)
+        return 0, 1, torch.device("cpu")
+    rank = int(os.environ["RANK"])
+    world = int(os.environ["WORLD_SIZE"])
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    if backend is None:
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+    if not dist.is_initialized():
+        dist.init_process_group(backend=backend, timeout=datetime.timedelta(minutes=10))
+    if torch.cuda.is_available():
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+    else:
+        device = torch.device("cpu")
+    return rank, world, device
+
+
+def rank_world():
+    if dist.is_available() and dist.is_initialized():
+        return dist.get_rank(), dist.get_world_size()
+    return 0, 1
+
+
+def is_distributed():
+    return dist.is_available() and dist.is_initialized()
+
+
+def barrier():
+    if is_distributed():
+        dist.barrier()
+
+
+def all_reduce_sum_(t: torch.Tensor):
+    if is_distributed():
+        dist.all_reduce(t, op=dist.ReduceOp.SUM)
+    return t
+
+
+def all_gather_object(obj):
+    if not is_distributed():
+        return [obj]
+    out = [None] * dist.get_world_size()
+    dist.all_gather_object(out, obj)
+    return out
+
+
+def _backend_supports_all_to_all():
+    try:
+        return dist.get_backend() == "nccl"
+    except Exception:
+        return False
+
+
+def all_to_all_v(tensor: torch.Tensor, send_counts, device=None):
+    """Exchange row-slices of ``tensor`` between all ranks.
+
+    ``send_counts``: int list/array of length world, rows destined per rank
+    (rows must already be grouped by destination rank, ascending).
+    Returns (received tensor, recv_counts list).
+    """
+    if not is_distributed():
+        return tensor, [int(send_counts[0])]
+    world = dist.get_world_size()
+    send_counts = [int(c) for c in send_counts]
+    sc = torch.tensor(send_counts, dtype=torch.int64, device=tensor.device)
+    rc = torch.empty_like(sc)
+    dist.all_to_all_single(rc, sc)
+    recv_counts = [int(x) for x in rc.cpu()]
+    out_shape = (sum(recv_counts),) + tuple(tensor.shape[1:])
+    out = torch.empty(out_shape, dtype=tensor.dtype, device=tensor.device)
+    if _backend_supports_all_to_all():
+        dist.all_to_all_single(out, tensor.contiguous(), recv_counts, send_counts)
+        return out, recv_counts
+    # gloo fallback: pairwise isend/irecv
+    rank = dist.get_rank()
+    send_offsets = np.concatenate([[0], np.cumsum(send_counts)])
+    recv_offsets = np.concatenate([[0], np.cumsum(recv_counts)])
+    reqs = []
+    tensor = tensor.contiguous()
+    for peer in range(world):
+        if peer == rank:
+            continue
+        chunk = tensor[send_offsets[peer] : send_offsets[peer + 1]]
+        if chunk.numel():
+            reqs.append(dist.isend(chunk.clone(), dst=peer))
+    for peer in range(world):
+        if peer == rank:
+            out[recv_offsets[peer] : recv_offsets[peer + 1]] = tensor[
+                send_offsets[peer] : send_offsets[peer + 1]
+            ]
+            continue
+        dst = out[recv_offsets[peer] : recv_offsets[peer + 1]]
+        if dst.numel():
+            buf = dst.contiguous()
+            dist.recv(buf, src=peer)
+            dst.copy_(buf)
+    for r in reqs:
+        r.wait()
+    return out, recv_counts

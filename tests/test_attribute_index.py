@@ -1,0 +1,133 @@
+"""Attribute-index tests with golden values from the reference's
+AttributeIndexTest.scala:38-100 (Australian state names fixture)."""
+
+import numpy as np
+import pytest
+
+from dblink_amd.models.attribute_index import AttributeIndex, _python_sim_pairs
+from dblink_amd.models.similarity import ConstantSimilarityFn, LevenshteinSimilarityFn
+
+STATE_WEIGHTS = {
+    "Australian Capital Territory": 0.410,
+    "New South Wales": 7.86,
+    "Northern Territory": 0.246,
+    "Queensland": 4.92,
+    "South Australia": 1.72,
+    "Tasmania": 0.520,
+    "Victoria": 6.32,
+    "Western Australia": 2.58,
+}
+
+# Golden sim normalizations from AttributeIndexTest.scala:48-56
+# (LevenshteinSimilarityFn(5.0, 10.0))
+GOLDEN_NORMS = {
+    "Australian Capital Territory": 0.0027140755302269004,
+    "New South Wales": 1.4193905286944585e-4,
+    "Northern Territory": 0.00451528932619675,
+    "Queensland": 2.2673706056780077e-4,
+    "South Australia": 6.465919296781136e-4,
+    "Tasmania": 0.00214117348291189,
+    "Victoria": 1.7651936247903708e-4,
+    "Western Australia": 4.317863538883541e-4,
+}
+
+
+@pytest.fixture(scope="module")
+def const_index():
+    return AttributeIndex(STATE_WEIGHTS, ConstantSimilarityFn(), pair_sweep=_python_sim_pairs)
+
+
+@pytest.fixture(scope="module")
+def lev_index():
+    return AttributeIndex(
+        STATE_WEIGHTS, LevenshteinSimilarityFn(5.0, 10.0), pair_sweep=_python_sim_pairs
+    )
+
+
+def test_value_ids_sorted(const_index):
+    # ids assigned by lexicographic sort of distinct values
+    assert const_index.value_id_of("Australian Capital Territory") == 0
+    assert const_index.value_id_of("Western Australia") == 7
+    assert const_index.value_id_of("not a state") == -1
+
+
+def test_probabilities(const_index):
+    total = sum(STATE_WEIGHTS.values())
+    for name, w in STATE_WEIGHTS.items():
+        vid = const_index.value_id_of(name)
+        assert const_index.probability_of(vid) == pytest.approx(w / total)
+    assert sum(const_index.probs) == pytest.approx(1.0)
+
+
+def test_probability_out_of_range(const_index):
+    with pytest.raises(IndexError):
+        const_index.probability_of(-1)
+    with pytest.raises(IndexError):
+        const_index.probability_of(const_index.num_values)
+
+
+def test_constant_norms_and_sims(const_index):
+    for v in range(const_index.num_values):
+        assert const_index.sim_normalization_of(v) == 1.0
+        assert const_index.sim_values_of(v) == {}
+        for w in range(const_index.num_values):
+            assert const_index.exp_sim_of(v, w) == 1.0
+
+
+def test_golden_sim_normalizations(lev_index):
+    for name, golden in GOLDEN_NORMS.items():
+        vid = lev_index.value_id_of(name)
+        assert lev_index.sim_normalization_of(vid) == pytest.approx(golden, abs=1e-4)
+
+
+def test_golden_sim_values_sa(lev_index):
+    # AttributeIndexTest.scala:58: simValuesOf("South Australia") has exactly
+    # {7 -> 39.813678..., 4 -> 22026.4657...} (ids 7=WA, 4=SA itself)
+    vid = lev_index.value_id_of("South Australia")
+    sims = lev_index.sim_values_of(vid)
+    assert set(sims.keys()) == {4, 7}
+    assert sims[7] == pytest.approx(39.813678188084864, rel=1e-6)
+    assert sims[4] == pytest.approx(22026.465794806718, rel=1e-6)
+
+
+def test_golden_exp_sims(lev_index):
+    sa = lev_index.value_id_of("South Australia")
+    wa = lev_index.value_id_of("Western Australia")
+    assert lev_index.exp_sim_of(sa, wa) == pytest.approx(39.813678188084864, rel=1e-6)
+    vic = lev_index.value_id_of("Victoria")
+    tas = lev_index.value_id_of("Tasmania")
+    assert lev_index.exp_sim_of(vic, tas) == 1.0
+
+
+def test_power_distribution(lev_index):
+    d1 = lev_index.sim_norm_dist(1)
+    # p(v) proportional to phi(v) * norm(v)
+    expect = lev_index.probs * lev_index.sim_norms
+    expect /= expect.sum()
+    np.testing.assert_allclose(d1.probs, expect, rtol=1e-12)
+    # sim_norm_prob matches
+    for v in range(lev_index.num_values):
+        assert lev_index.sim_norm_prob(v, 1) == pytest.approx(expect[v])
+    with pytest.raises(ValueError):
+        lev_index.sim_norm_dist(0)
+
+
+def test_constant_power_dist_is_phi(const_index):
+    d = const_index.sim_norm_dist(3)
+    np.testing.assert_allclose(d.probs, const_index.probs)
+
+
+def test_native_pair_sweep_matches_python():
+    """If the native extension is built, its sim_pairs must agree with the
+    pure-Python oracle on this fixture."""
+    from dblink_amd import ops
+
+    if not ops.have_native():
+        pytest.skip("native extension not built")
+    fn = LevenshteinSimilarityFn(5.0, 10.0)
+    values = sorted(STATE_WEIGHTS)
+    native = ops.sim_pairs(values, fn)
+    ref = _python_sim_pairs(values, fn)
+    np.testing.assert_array_equal(native.row_ptr, ref.row_ptr)
+    np.testing.assert_array_equal(native.col, ref.col)
+    np.testing.assert_allclose(native.expsim, ref.expsim, rtol=1e-6)

@@ -1,0 +1,107 @@
+"""HOCON parser tests, including the exact shape of the reference's example
+configs (examples/RLdata500.conf structure)."""
+
+import textwrap
+
+import pytest
+
+from dblink_amd.utils import hocon
+
+EXAMPLE = textwrap.dedent(
+    """
+    dblink : {
+        // comment
+        lowDistortion : {alpha : 0.5, beta : 50.0}
+        constSimFn : {
+            name : "ConstantSimilarityFn",
+        }
+        levSimFn : {
+            name : "LevenshteinSimilarityFn",
+            parameters : {
+                threshold : 7.0
+                maxSimilarity : 10.0
+            }
+        }
+        data : {
+            path : "./examples/data.csv"
+            recordIdentifier : "rec_id",
+            entityIdentifier : "ent_id" // optional
+            nullValue : "NA"
+            matchingAttributes : [
+                {name : "by", similarityFunction : ${dblink.constSimFn}, distortionPrior : ${dblink.lowDistortion}},
+                {name : "fname_c1", similarityFunction : ${dblink.levSimFn}, distortionPrior : ${dblink.lowDistortion}}
+            ]
+        }
+        randomSeed : 319158
+        expectedMaxClusterSize : 10
+        partitioner : {
+            name : "KDTreePartitioner",
+            parameters : {
+                numLevels : 0,
+                matchingAttributes : []
+            }
+        }
+        outputPath : "./out/"
+        checkpointPath : "/tmp/ckpt/"
+        steps : [
+            {name : "sample", parameters : {sampleSize : 100, burninInterval : 0,
+                thinningInterval : 10, resume : false, sampler : "PCG-I"}},
+            {name : "evaluate", parameters : {lowerIterationCutoff : 100,
+                metrics : ["pairwise", "cluster"], useExistingSMPC : false}}
+        ]
+    }
+    """
+)
+
+
+def test_parse_example():
+    cfg = hocon.parse_string(EXAMPLE)
+    assert cfg.get_string("dblink.data.path") == "./examples/data.csv"
+    assert cfg.get_long("dblink.randomSeed") == 319158
+    assert cfg.get_int("dblink.expectedMaxClusterSize") == 10
+    attrs = cfg.get_config_list("dblink.data.matchingAttributes")
+    assert len(attrs) == 2
+    # substitution resolution
+    assert attrs[0].get_string("similarityFunction.name") == "ConstantSimilarityFn"
+    assert attrs[1].get_double("similarityFunction.parameters.threshold") == 7.0
+    assert attrs[0].get_double("distortionPrior.alpha") == 0.5
+    steps = cfg.get_config_list("dblink.steps")
+    assert steps[0].get_string("name") == "sample"
+    assert steps[0].get_int("parameters.sampleSize") == 100
+    assert steps[0].get_or("parameters.resume", True) is False
+    assert steps[1].get_string_list("parameters.metrics") == ["pairwise", "cluster"]
+
+
+def test_missing_and_defaults():
+    cfg = hocon.parse_string("a { b : 1 }")
+    assert cfg.get_int("a.b") == 1
+    assert not cfg.has_path("a.c")
+    assert cfg.get_or("a.c", 42) == 42
+    with pytest.raises(hocon.ConfigMissingError):
+        cfg.get("a.c")
+
+
+def test_newline_separated_and_equals():
+    cfg = hocon.parse_string("a = 1\nb : two\nc { d = true }\n")
+    assert cfg.get_int("a") == 1
+    assert cfg.get_string("b") == "two"
+    assert cfg.get_bool("c.d") is True
+
+
+def test_duplicate_key_merge():
+    cfg = hocon.parse_string("a { x : 1 }\na { y : 2 }\n")
+    assert cfg.get_int("a.x") == 1
+    assert cfg.get_int("a.y") == 2
+
+
+def test_dotted_keys():
+    cfg = hocon.parse_string("a.b.c : 3")
+    assert cfg.get_int("a.b.c") == 3
+
+
+def test_unquoted_strings_and_numbers():
+    cfg = hocon.parse_string("x : hello\ny : 2.5\nz : -3\nw : null")
+    assert cfg.get_string("x") == "hello"
+    assert cfg.get_double("y") == 2.5
+    assert cfg.get_int("z") == -3
+    assert cfg.get("w") is None

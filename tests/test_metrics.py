@@ -1,0 +1,82 @@
+import math
+
+import pytest
+
+from dblink_amd.analysis.baselines import exact_match_clusters, near_clusters
+from dblink_amd.analysis.metrics import (
+    ClusteringMetrics,
+    PairwiseMetrics,
+    adjusted_rand_index,
+    membership_to_clusters,
+    to_membership,
+    to_pairwise_links,
+)
+
+
+def test_pairwise_links():
+    clusters = [{"a", "b", "c"}, {"d"}, {"e", "f"}]
+    links = to_pairwise_links(clusters)
+    assert links == {("a", "b"), ("a", "c"), ("b", "c"), ("e", "f")}
+
+
+def test_pairwise_metrics_perfect():
+    clusters = [{"a", "b"}, {"c", "d", "e"}]
+    m = PairwiseMetrics.compute(clusters, clusters)
+    assert m.precision == 1.0 and m.recall == 1.0 and m.f1score == 1.0
+
+
+def test_pairwise_metrics_partial():
+    pred = [{"a", "b"}, {"c"}, {"d", "e"}]
+    true = [{"a", "b", "c"}, {"d"}, {"e"}]
+    # pred links: ab, de ; true links: ab, ac, bc
+    m = PairwiseMetrics.compute(pred, true)
+    assert m.precision == pytest.approx(1 / 2)
+    assert m.recall == pytest.approx(1 / 3)
+    assert m.f1score == pytest.approx(2 * (1 / 2) * (1 / 3) / (1 / 2 + 1 / 3))
+
+
+def test_ari_identical():
+    clusters = [{"a", "b"}, {"c"}, {"d", "e", "f"}]
+    assert adjusted_rand_index(clusters, clusters) == pytest.approx(1.0)
+
+
+def test_ari_known_value():
+    # Hand-computed: contingency nij = [[2,1],[1,2]], n=6 ->
+    # sum comb2(nij)=2, pred/true comb sums = 6, expected = 36/15 = 2.4,
+    # ARI = (2-2.4)/(6-2.4) = -1/9.
+    pred = membership_to_clusters({i: l for i, l in enumerate([0, 0, 1, 1, 0, 1])})
+    true = membership_to_clusters({i: l for i, l in enumerate([0, 0, 0, 1, 1, 1])})
+    ari = adjusted_rand_index(pred, true)
+    assert ari == pytest.approx(-1 / 9, abs=1e-9)
+
+    # cross-check against sklearn if available
+    try:
+        from sklearn.metrics import adjusted_rand_score
+    except ImportError:
+        return
+    assert ari == pytest.approx(adjusted_rand_score([0, 0, 1, 1, 0, 1], [0, 0, 0, 1, 1, 1]))
+
+
+def test_ari_mismatched_elements():
+    with pytest.raises(ValueError):
+        adjusted_rand_index([{"a"}], [{"b"}])
+
+
+def test_membership_roundtrip():
+    clusters = [{"a", "b"}, {"c"}]
+    m = to_membership(clusters)
+    back = membership_to_clusters(m)
+    assert sorted(map(sorted, back)) == sorted(map(sorted, clusters))
+
+
+def test_exact_match_clusters():
+    recs = [("r1", ["x", "y"]), ("r2", ["x", "y"]), ("r3", ["x", "z"])]
+    clusters = exact_match_clusters(recs)
+    assert sorted(map(sorted, clusters)) == [["r1", "r2"], ["r3"]]
+
+
+def test_near_clusters():
+    recs = [("r1", ["x", "y"]), ("r2", ["x", "z"]), ("r3", ["w", "w"])]
+    clusters = near_clusters(recs, 1)
+    merged = [c for c in clusters if len(c) > 1]
+    assert {"r1", "r2"} in merged
