@@ -1,0 +1,167 @@
+"""Flagship benchmark: Gibbs iterations/sec on RLdata10000-shaped data.
+
+Driver contract:
+    python bench.py --gpus N --steps K --warmup W
+(for N > 1 launched via torch.distributed.run with one rank per GPU over RCCL)
+
+Measures the BASELINE.json metric — "Gibbs iterations/sec (whole node) on
+RLdata10000-shape" — with the PCG-I sampler on synthetic RLdata-shaped
+records (same schema: 3 constant-sim + 2 Levenshtein attributes, 10%
+duplicates; there is no network access for the original RLdata files) and
+random/deterministic-init latent state. Weak scaling: each GPU holds an
+RLdata10000-shaped shard (10,000 records, 4 KD-tree partitions), so the
+whole-job iteration size grows with N while iterations/sec is reported for
+the whole job.
+
+Timing: W untimed warm-up sweeps, then K sweeps bracketed by a barrier +
+torch.cuda.synchronize() on both sides; MAX elapsed over ranks; rank 0
+prints one JSON line. Every timed sweep includes the full Markov transition
+the reference times in its diagnostics loop: theta update, link/value/
+distortion updates, partition reassignment, migration all-to-all, and the
+summary all-reduce. Sample/diagnostic file writes are excluded (the
+reference amortizes them 1/thinningInterval = 1/10; checkpointing off).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import math
+import os
+import sys
+import time
+
+import numpy as np
+
+
+def build_cache_and_records(total_records, seed):
+    from dblink_amd.models.records import (
+        Attribute,
+        BetaShapeParameters,
+        RecordsCache,
+        RecordsTable,
+    )
+    from dblink_amd.models.similarity import ConstantSimilarityFn, LevenshteinSimilarityFn
+    from dblink_amd.utils.synthdata import generate
+
+    cols, header = generate(total_records, dup_fraction=0.1, seed=seed)
+    attr_names = ["by", "bm", "bd", "fname_c1", "lname_c1"]
+    values = [
+        [cols[a][i] if cols[a][i] != "NA" else None for a in attr_names]
+        for i in range(total_records)
+    ]
+    table = RecordsTable(cols["rec_id"], ["0"] * total_records, values)
+    prior = BetaShapeParameters(10.0, 1000.0)  # RLdata10000.conf:4
+    attrs = [
+        Attribute("by", ConstantSimilarityFn(), prior),
+        Attribute("bm", ConstantSimilarityFn(), prior),
+        Attribute("bd", ConstantSimilarityFn(), prior),
+        Attribute("fname_c1", LevenshteinSimilarityFn(7.0, 10.0), prior),
+        Attribute("lname_c1", LevenshteinSimilarityFn(7.0, 10.0), prior),
+    ]
+    cache = RecordsCache.build(table, attrs, max_cluster_size=10)
+    rec_values, rec_files = cache.transform_records(table)
+    return cache, rec_values, rec_files
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=50)
+    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--records-per-gpu", type=int, default=10000)
+    ap.add_argument("--partitions-per-gpu", type=int, default=4)
+    ap.add_argument("--sampler", default="PCG-I")
+    ap.add_argument("--seed", type=int, default=319158)
+    ap.add_argument("--cpu", action="store_true", help="force the CPU engine")
+    args = ap.parse_args()
+
+    sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+    import torch
+
+    from dblink_amd.engine.cpu_engine import CpuEngine, SamplerFlags
+    from dblink_amd.engine.init import deterministic_init
+    from dblink_amd.parallel import comm
+    from dblink_amd.parallel.partitioning import KDTreePartitioner
+
+    rank, world, device = comm.init_from_env()
+    n_gpus = world if world > 1 else args.gpus
+    use_gpu = torch.cuda.is_available() and not args.cpu
+
+    total_records = args.records_per_gpu * n_gpus
+    total_partitions = args.partitions_per_gpu * n_gpus
+    num_levels = max(0, int(round(math.log2(total_partitions))))
+
+    cache, rec_values, rec_files = build_cache_and_records(total_records, args.seed)
+
+    partitioner = KDTreePartitioner(num_levels, [3, 4])  # fname_c1, lname_c1
+    bounds = np.linspace(0, total_records, world + 1).astype(np.int64)
+    lo, hi = int(bounds[rank]), int(bounds[rank + 1])
+    state = deterministic_init(
+        rec_values[lo:hi], rec_files[lo:hi], np.arange(lo, hi, dtype=np.int64),
+        cache, partitioner, args.seed, rank=rank, world_size=world,
+    )
+
+    if use_gpu:
+        from dblink_amd.engine.gpu_engine import GpuEngine
+
+        engine = GpuEngine(cache, partitioner, world_size=world, rank=rank, device=device)
+    else:
+        engine = CpuEngine(cache, partitioner, world_size=world, rank=rank)
+    engine.initial_summary(state)
+    flags = SamplerFlags.for_sampler(args.sampler)
+
+    def sync():
+        comm.barrier()
+        if use_gpu:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        engine.step(state, flags)
+    sync()
+    t0 = time.time()
+    for _ in range(args.steps):
+        engine.step(state, flags)
+    sync()
+    elapsed = time.time() - t0
+
+    # max over ranks
+    t = torch.tensor([elapsed], dtype=torch.float64,
+                     device=device if use_gpu else "cpu")
+    if comm.is_distributed():
+        import torch.distributed as dist
+
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t.cpu())
+
+    if rank == 0:
+        value = args.steps / elapsed
+        out = {
+            "metric": "Gibbs iterations/sec (whole node) on RLdata10000-shape",
+            "value": value,
+            "unit": "iterations/sec",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "dblink PCG-I partitioned Gibbs (RLdata10000 schema: "
+                         "3 constant + 2 Levenshtein attributes, 10% duplicates)",
+                "records_per_gpu": args.records_per_gpu,
+                "global_records": total_records,
+                "partitions": total_partitions,
+                "sampler": args.sampler,
+                "parallelism": f"entity-partitioned Gibbs, {n_gpus} rank(s) over RCCL",
+                "engine": "gpu" if use_gpu else "cpu",
+            },
+        }
+        print(json.dumps(out))
+
+
+if __name__ == "__main__":
+    main()

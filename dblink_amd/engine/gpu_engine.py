@@ -1,0 +1,431 @@
+"""GPU engine: the partitioned Gibbs sweep as CDNA4 HIP kernels.
+
+Design (MI355X-first, not a port of the Spark flow):
+
+- all chain state lives in HBM3E as flat SoA tensors sorted by partition id;
+  per-partition segments are addressed by offset tables, so ONE kernel launch
+  covers every partition this rank owns (256 CUs want >>256 workgroups)
+- the inverted index is rebuilt per sweep as a sorted posting array
+  (radix sort via torch.sort) + batched searchsorted for per-record ranges
+- link / value / distortion updates run as the fused kernels of
+  ``ops/csrc/kernels.hip`` with Philox counter RNG (order-independent draws)
+- summaries reduce on-device to a small packed tensor -> one RCCL all-reduce
+- cluster migration is an RCCL all-to-all-v over xGMI, then a device-side
+  re-sort by partition id
+
+The host syncs once per iteration (summary + theta update, as in the
+reference's driver loop) and pulls full state back only when a sample is
+written or the state is saved.
+"""
+
+from __future__ import annotations
+
+import logging
+
+import numpy as np
+import torch
+
+from ..models.distortion import DistortionProbs
+from ..parallel import comm
+from .cpu_engine import CpuEngine, SamplerFlags
+from .state import ChainState, SummaryVars
+
+log = logging.getLogger("dblink_amd.gpu")
+
+
+def _require_native():
+    from .. import ops
+
+    return ops.native()
+
+
+class GpuModel:
+    """Device-resident read-only model tensors (the broadcast RecordsCache)."""
+
+    def __init__(self, cache, device, max_cluster_size):
+        self.cache = cache
+        self.device = device
+        attrs = cache.indexed_attributes
+        A = len(attrs)
+        self.A = A
+        self.F = cache.num_files
+        self.Kc = int(max_cluster_size)
+
+        voff = np.zeros(A + 1, dtype=np.int64)
+        for a, ia in enumerate(attrs):
+            voff[a + 1] = voff[a] + ia.index.num_values
+        self.Vtot = int(voff[A])
+        self.Vmax = int(max(ia.index.num_values for ia in attrs))
+
+        phi = np.concatenate([ia.index.probs for ia in attrs]).astype(np.float64)
+        norm = np.concatenate([ia.index.sim_norms for ia in attrs]).astype(np.float64)
+
+        # CSR similarity index with per-attribute row offsets
+        row_ptr = np.zeros(self.Vtot + 1, dtype=np.int64)
+        cols, sims = [], []
+        nnz = 0
+        for a, ia in enumerate(attrs):
+            V = ia.index.num_values
+            if ia.is_constant:
+                row_ptr[voff[a] + 1 : voff[a] + V + 1] = nnz
+            else:
+                si = ia.index.sim_index
+                row_ptr[voff[a] + 1 : voff[a] + V + 1] = nnz + si.row_ptr[1:]
+                cols.append(si.col)
+                sims.append(np.log(si.expsim))  # store sim = log(expsim)
+                nnz += int(si.row_ptr[-1])
+        csr_col = np.concatenate(cols) if cols else np.empty(0, dtype=np.int32)
+        csr_sim = np.concatenate(sims) if sims else np.empty(0, dtype=np.float64)
+
+        # alias tables
+        phi_prob = np.concatenate([ia.index.distribution.prob for ia in attrs])
+        phi_alias = np.concatenate([ia.index.distribution.alias for ia in attrs])
+
+        pow_prob, pow_alias = [], []
+        pow_off = np.full(A, -1, dtype=np.int64)
+        log_pow_total = np.zeros((A, self.Kc + 1), dtype=np.float64)
+        off = 0
+        for a, ia in enumerate(attrs):
+            if ia.is_constant:
+                continue
+            V = ia.index.num_values
+            pow_off[a] = off
+            for k in range(1, self.Kc + 1):
+                t = ia.index.sim_norm_dist(k)
+                pow_prob.append(t.prob)
+                pow_alias.append(t.alias)
+                log_pow_total[a, k] = np.log(ia.index.sim_norm_total(k))
+            off += self.Kc * V
+
+        self_expsim = np.concatenate(
+            [
+                np.full(
+                    ia.index.num_values,
+                    1.0 if ia.is_constant else float(np.exp(ia.spec.similarity_fn.max_similarity)),
+                )
+                for ia in attrs
+            ]
+        )
+
+        def dev(arr, dtype):
+            return torch.from_numpy(np.ascontiguousarray(arr)).to(dtype).to(device)
+
+        self.voff = dev(voff, torch.int64)
+        self.phi = dev(phi, torch.float32)
+        self.log_phi = dev(np.log(phi), torch.float32)
+        self.norm_lin = dev(norm, torch.float32)
+        self.log_norm = dev(np.log(norm), torch.float32)
+        self.csr_row_ptr = dev(row_ptr, torch.int64)
+        self.csr_col = dev(csr_col, torch.int32)
+        self.csr_sim = dev(csr_sim, torch.float32)
+        self.phi_prob = dev(phi_prob, torch.float32)
+        self.phi_alias = dev(phi_alias, torch.int32)
+        self.pow_prob = dev(np.concatenate(pow_prob) if pow_prob else np.empty(0), torch.float32)
+        self.pow_alias = dev(
+            np.concatenate(pow_alias) if pow_alias else np.empty(0, np.int64), torch.int32
+        )
+        self.pow_off = dev(pow_off, torch.int64)
+        self.log_pow_total = dev(log_pow_total.reshape(-1), torch.float32)
+        self.attr_const = dev(
+            np.array([1 if ia.is_constant else 0 for ia in attrs], dtype=np.uint8), torch.uint8
+        )
+        self.self_expsim = dev(self_expsim, torch.float32)
+        self.theta = torch.zeros((A, self.F), dtype=torch.float32, device=device)
+
+
+class GpuStateTensors:
+    """Device mirror of ChainState's mutable arrays."""
+
+    FIELDS = ("ent_values", "ent_part", "rec_values", "rec_file", "rec_ent", "rec_dist", "rec_gid")
+    DTYPES = {
+        "ent_values": torch.int32,
+        "ent_part": torch.int32,
+        "rec_values": torch.int32,
+        "rec_file": torch.int32,
+        "rec_ent": torch.int64,
+        "rec_dist": torch.uint8,
+        "rec_gid": torch.int64,
+    }
+
+    def __init__(self, state: ChainState, device):
+        for f in self.FIELDS:
+            arr = getattr(state, f)
+            setattr(self, f, torch.from_numpy(np.ascontiguousarray(arr)).to(device))
+        self.rec_part = self.ent_part[self.rec_ent].contiguous()
+        self.device = device
+
+    def to_host(self, state: ChainState):
+        for f in self.FIELDS:
+            setattr(state, f, getattr(self, f).cpu().numpy())
+
+    @property
+    def E(self):
+        return int(self.ent_values.shape[0])
+
+    @property
+    def R(self):
+        return int(self.rec_values.shape[0])
+
+
+class GpuEngine(CpuEngine):
+    """Drop-in engine with the sweep on a ROCm device.
+
+    Inherits host-side pieces (theta update, summary packing/reduction,
+    linkage structure) from CpuEngine; the Markov transition itself runs as
+    HIP kernels.
+    """
+
+    def __init__(self, cache, partitioner, world_size=1, rank=0, device=None):
+        super().__init__(cache, partitioner, world_size=world_size, rank=rank)
+        if device is None:
+            device = torch.device("cuda", 0)
+        self.device = device
+        self.C = _require_native()
+        self.model = GpuModel(cache, device, cache_kc(cache, partitioner))
+        self.flat_tree = None
+        self._gs = None
+        self._ent_id_base = rank << 40
+        self._err = torch.zeros(1, dtype=torch.int32, device=device)
+        self._loglik_buf = torch.zeros(1, dtype=torch.float64, device=device)
+
+    # ---- state residency -----------------------------------------------------
+
+    def _gpu_state(self, state: ChainState) -> GpuStateTensors:
+        if self._gs is None:
+            self._gs = GpuStateTensors(state, self.device)
+            self._upload_tree()
+        return self._gs
+
+    def _upload_tree(self):
+        flat = self.partitioner.as_flat()
+        self.flat_tree = {
+            k: torch.from_numpy(np.ascontiguousarray(v.astype(np.int32))).to(self.device)
+            for k, v in flat.items()
+        }
+        if self.flat_tree["rset"].numel() == 0:
+            self.flat_tree["rset"] = torch.zeros(1, dtype=torch.int32, device=self.device)
+
+    def sync_state(self, state: ChainState):
+        if self._gs is not None:
+            self._gs.to_host(state)
+
+    # ---- the transition ------------------------------------------------------
+
+    def step(self, state: ChainState, flags: SamplerFlags):
+        gs = self._gpu_state(state)
+        m = self.model
+
+        # theta from previous summary (host, Philox stream keyed by iteration)
+        self._update_dist_probs(state)
+        m.theta.copy_(torch.from_numpy(state.dist_probs.probs).float())
+
+        seed = state.current_seed
+        it = state.iteration + 1
+        A, E, R = m.A, gs.E, gs.R
+
+        # --- inverted index (sorted postings) --------------------------------
+        if not flags.sequential and not flags.collapsed_entity_ids:
+            arangeA = torch.arange(A, device=self.device, dtype=torch.int64).view(A, 1)
+            keys = (
+                (gs.ent_part.to(torch.int64).view(1, E) * A + arangeA) * m.Vmax
+                + gs.ent_values.t().to(torch.int64)
+            ).reshape(-1)
+            sorted_keys, perm = torch.sort(keys, stable=True)
+            postings = (perm % E).to(torch.int32)
+            qkeys = (
+                (gs.rec_part.to(torch.int64).view(R, 1) * A
+                 + torch.arange(A, device=self.device, dtype=torch.int64).view(1, A))
+                * m.Vmax
+                + gs.rec_values.clamp_min(0).to(torch.int64)
+            ).reshape(-1)
+            cand_lo = torch.searchsorted(sorted_keys, qkeys, right=False).view(R, A)
+            cand_hi = torch.searchsorted(sorted_keys, qkeys, right=True).view(R, A)
+
+        ent_ptr = torch.searchsorted(
+            gs.ent_part.to(torch.int64).contiguous(),
+            torch.arange(self.num_partitions + 1, device=self.device, dtype=torch.int64),
+        )
+
+        # --- phase 1: link update --------------------------------------------
+        rec_ent_new = torch.empty_like(gs.rec_ent)
+        if flags.sequential or flags.collapsed_entity_ids:
+            self.C.link_update_dense(
+                gs.rec_values, gs.rec_dist, gs.rec_gid, gs.rec_part, gs.rec_file,
+                gs.ent_values, ent_ptr, m.theta, m.phi, m.norm_lin, m.voff,
+                m.csr_row_ptr, m.csr_col, m.csr_sim, m.attr_const,
+                1 if flags.collapsed_entity_ids else 0, seed, it, rec_ent_new,
+            )
+        else:
+            self.C.link_update(
+                gs.rec_values, gs.rec_dist, gs.rec_gid, gs.rec_part,
+                cand_lo.contiguous(), cand_hi.contiguous(), postings, gs.ent_values,
+                ent_ptr, m.log_norm, m.voff, m.csr_row_ptr, m.csr_col, m.csr_sim,
+                m.attr_const, seed, it, rec_ent_new, gs.rec_ent, self._err,
+            )
+        gs.rec_ent = rec_ent_new
+
+        # --- entity -> records CSR -------------------------------------------
+        sorted_re, order = torch.sort(gs.rec_ent, stable=True)
+        ent_rec_ptr = torch.searchsorted(
+            sorted_re, torch.arange(E + 1, device=self.device, dtype=torch.int64)
+        )
+        ent_rec_idx = order
+
+        # --- phase 2: value update (in place) --------------------------------
+        self.C.value_update(
+            gs.rec_values, gs.rec_dist, gs.rec_file, ent_rec_ptr, ent_rec_idx,
+            gs.ent_values, m.theta, m.phi, m.log_phi, m.norm_lin, m.log_norm,
+            m.voff, m.csr_row_ptr, m.csr_col, m.csr_sim, m.phi_prob, m.phi_alias,
+            m.pow_prob, m.pow_alias, m.pow_off, m.log_pow_total, m.attr_const,
+            m.Kc, 1 if flags.collapsed_entity_values else 0,
+            1 if flags.sequential else 0, seed, it, self._ent_id_base, self._err,
+        )
+
+        # --- phase 3: distortion update --------------------------------------
+        self.C.distortion_update(
+            gs.rec_values, gs.rec_dist, gs.rec_file, gs.rec_gid, gs.rec_ent,
+            gs.ent_values, m.theta, m.phi, m.norm_lin, m.self_expsim, m.voff,
+            m.attr_const, seed, it,
+        )
+
+        # --- partition reassignment ------------------------------------------
+        ent_part_new = torch.empty_like(gs.ent_part)
+        self.C.kd_descent(
+            gs.ent_values, self.flat_tree["kind"], self.flat_tree["attr"],
+            self.flat_tree["a"], self.flat_tree["b"], self.flat_tree["rset"],
+            ent_part_new,
+        )
+        gs.ent_part = ent_part_new
+
+        state.current_seed += self.num_partitions
+        state.iteration += 1
+
+        # --- migration + local re-sort ---------------------------------------
+        self._migrate_and_sort(gs)
+
+        # --- summary ----------------------------------------------------------
+        summary = self._summary_device(gs, state)
+        state.summary = summary
+        return state
+
+    # ---- helpers -------------------------------------------------------------
+
+    def _migrate_and_sort(self, gs: GpuStateTensors):
+        world = self.world_size
+        if world > 1 and comm.is_distributed():
+            dest_e = gs.ent_part.to(torch.int64) % world
+            order_e = torch.argsort(dest_e, stable=True)
+            inv_e = torch.empty_like(order_e)
+            inv_e[order_e] = torch.arange(order_e.numel(), device=self.device)
+            send_e = torch.bincount(dest_e, minlength=world)
+            dest_r = dest_e[gs.rec_ent]
+            order_r = torch.argsort(dest_r, stable=True)
+            send_r = torch.bincount(dest_r, minlength=world)
+            base = torch.cumsum(
+                torch.cat([torch.zeros(1, dtype=torch.int64, device=self.device), send_e[:-1]]), 0
+            )
+            pos_in_dest = inv_e - base[dest_e]
+
+            send_e_l = [int(x) for x in send_e.cpu()]
+            send_r_l = [int(x) for x in send_r.cpu()]
+
+            new_ev, recv_e = comm.all_to_all_v(gs.ent_values[order_e].contiguous(), send_e_l)
+            new_ep, _ = comm.all_to_all_v(gs.ent_part[order_e].contiguous(), send_e_l)
+            rel = pos_in_dest[gs.rec_ent][order_r].contiguous()
+            new_rel, recv_r = comm.all_to_all_v(rel, send_r_l)
+            new_rv, _ = comm.all_to_all_v(gs.rec_values[order_r].contiguous(), send_r_l)
+            new_rf, _ = comm.all_to_all_v(gs.rec_file[order_r].contiguous(), send_r_l)
+            new_rd, _ = comm.all_to_all_v(gs.rec_dist[order_r].contiguous(), send_r_l)
+            new_rg, _ = comm.all_to_all_v(gs.rec_gid[order_r].contiguous(), send_r_l)
+
+            eoff = np.concatenate([[0], np.cumsum(recv_e)])
+            roff = np.concatenate([[0], np.cumsum(recv_r)])
+            new_re = torch.empty(int(roff[-1]), dtype=torch.int64, device=self.device)
+            for s in range(world):
+                new_re[int(roff[s]) : int(roff[s + 1])] = (
+                    new_rel[int(roff[s]) : int(roff[s + 1])] + int(eoff[s])
+                )
+            gs.ent_values, gs.ent_part = new_ev, new_ep
+            gs.rec_values, gs.rec_file, gs.rec_dist, gs.rec_gid = new_rv, new_rf, new_rd, new_rg
+            gs.rec_ent = new_re
+
+        # local re-sort by partition id (stable)
+        order = torch.argsort(gs.ent_part.to(torch.int64), stable=True)
+        inv = torch.empty_like(order)
+        inv[order] = torch.arange(order.numel(), device=self.device)
+        gs.ent_values = gs.ent_values[order].contiguous()
+        gs.ent_part = gs.ent_part[order].contiguous()
+        new_rec_ent = inv[gs.rec_ent]
+        rorder = torch.argsort(new_rec_ent, stable=True)
+        gs.rec_ent = new_rec_ent[rorder].contiguous()
+        gs.rec_values = gs.rec_values[rorder].contiguous()
+        gs.rec_file = gs.rec_file[rorder].contiguous()
+        gs.rec_dist = gs.rec_dist[rorder].contiguous()
+        gs.rec_gid = gs.rec_gid[rorder].contiguous()
+        gs.rec_part = gs.ent_part[gs.rec_ent].contiguous()
+
+    def _summary_device(self, gs: GpuStateTensors, state: ChainState) -> SummaryVars:
+        m = self.model
+        A, F = m.A, m.F
+        E, R = gs.E, gs.R
+        self._loglik_buf.zero_()
+        self.C.summary_loglik(
+            gs.ent_values, gs.rec_values, gs.rec_dist, gs.rec_ent, m.log_phi,
+            m.log_norm, m.voff, m.csr_row_ptr, m.csr_col, m.csr_sim, m.attr_const,
+            self._loglik_buf,
+        )
+        linked = torch.zeros(E, dtype=torch.int64, device=self.device)
+        linked.scatter_add_(0, gs.rec_ent, torch.ones_like(gs.rec_ent))
+        isolates = (linked == 0).sum()
+        dist = gs.rec_dist.to(torch.int64)
+        agg_idx = (
+            torch.arange(A, device=self.device, dtype=torch.int64).view(1, A) * F
+            + gs.rec_file.to(torch.int64).view(R, 1)
+        ).reshape(-1)
+        agg = torch.zeros(A * F, dtype=torch.int64, device=self.device)
+        agg.scatter_add_(0, agg_idx, dist.reshape(-1))
+        ndist = dist.sum(dim=1)
+        hist = torch.zeros(A + 1, dtype=torch.int64, device=self.device)
+        hist.scatter_add_(0, ndist, torch.ones_like(ndist))
+
+        packed = torch.cat(
+            [
+                self._loglik_buf,
+                isolates.to(torch.float64).view(1),
+                agg.to(torch.float64),
+                hist.to(torch.float64),
+            ]
+        )
+        comm.all_reduce_sum_(packed)
+        host = packed.cpu().numpy()
+        err = int(self._err.cpu())
+        if err:
+            raise RuntimeError(f"{err} empty candidate sets in link update (invariant violated)")
+        out = SummaryVars(
+            num_isolates=int(round(host[1])),
+            log_likelihood=float(host[0]),
+            agg_distortions=host[2 : 2 + A * F].reshape(A, F).astype(np.int64),
+            rec_distortions=host[2 + A * F :].astype(np.int64),
+        )
+        from .cpu_engine import add_prior_terms
+
+        add_prior_terms(out, self.cache, state.dist_probs)
+        return out
+
+    def initial_summary(self, state: ChainState):
+        gs = self._gpu_state(state)
+        state.summary = self._summary_device(gs, state)
+
+    def linkage_structure(self, state: ChainState, rec_id_of=None):
+        self.sync_state(state)
+        return super().linkage_structure(state, rec_id_of)
+
+
+def cache_kc(cache, partitioner):
+    """How many power distributions to pre-cache per attribute (the
+    reference precaches 1..expectedMaxClusterSize, RecordsCache.scala:112)."""
+    # the cache was built with precache_powers = expectedMaxClusterSize
+    for ia in cache.indexed_attributes:
+        if not ia.is_constant:
+            return max(ia.index._max_cached_power, 1)
+    return 1

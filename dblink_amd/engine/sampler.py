@@ -57,6 +57,10 @@ def sample(
                 output_path, engine.cache, continue_chain=continue_chain
             )
 
+    def sync():
+        if hasattr(engine, "sync_state"):
+            engine.sync_state(state)
+
     def record_sample():
         nonlocal sample_ctr
         if write_output:
@@ -79,6 +83,7 @@ def sample(
             record_sample()
             sample_ctr += 1
         if checkpoint_interval and completed % checkpoint_interval == 0:
+            sync()
             state.save(output_path, rank=rank)
     dt = time.time() - t0
     iters = state.iteration - initial_iteration
@@ -91,5 +96,6 @@ def sample(
         if diagnostics_writer is not None:
             diagnostics_writer.close()
         comm.barrier()
+        sync()
         state.save(output_path, rank=rank)
     return state

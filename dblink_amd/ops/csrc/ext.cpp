@@ -1,0 +1,78 @@
+// Python bindings for the dblink_amd native extension.
+
+#include <torch/extension.h>
+
+#include <vector>
+
+namespace dblink {
+
+// sim_pairs_cpu.cpp
+std::vector<torch::Tensor> sim_pairs_cpu(torch::Tensor strs, torch::Tensor lens,
+                                         double threshold, double max_sim);
+
+// kernels.hip
+void link_update(torch::Tensor rec_values, torch::Tensor rec_dist,
+                 torch::Tensor rec_gid, torch::Tensor rec_part,
+                 torch::Tensor cand_lo, torch::Tensor cand_hi,
+                 torch::Tensor postings, torch::Tensor ent_values,
+                 torch::Tensor ent_ptr, torch::Tensor log_norm, torch::Tensor voff,
+                 torch::Tensor csr_row_ptr, torch::Tensor csr_col,
+                 torch::Tensor csr_sim, torch::Tensor attr_const, int64_t seed,
+                 int64_t iteration, torch::Tensor rec_ent_out,
+                 torch::Tensor rec_ent_in, torch::Tensor error_count);
+void link_update_dense(torch::Tensor rec_values, torch::Tensor rec_dist,
+                       torch::Tensor rec_gid, torch::Tensor rec_part,
+                       torch::Tensor rec_file, torch::Tensor ent_values,
+                       torch::Tensor ent_ptr, torch::Tensor theta, torch::Tensor phi,
+                       torch::Tensor norm_lin, torch::Tensor voff,
+                       torch::Tensor csr_row_ptr, torch::Tensor csr_col,
+                       torch::Tensor csr_sim, torch::Tensor attr_const,
+                       int64_t collapsed, int64_t seed, int64_t iteration,
+                       torch::Tensor rec_ent_out);
+void value_update(torch::Tensor rec_values, torch::Tensor rec_dist,
+                  torch::Tensor rec_file, torch::Tensor ent_rec_ptr,
+                  torch::Tensor ent_rec_idx, torch::Tensor ent_values,
+                  torch::Tensor theta, torch::Tensor phi, torch::Tensor log_phi,
+                  torch::Tensor norm_lin, torch::Tensor log_norm, torch::Tensor voff,
+                  torch::Tensor csr_row_ptr, torch::Tensor csr_col,
+                  torch::Tensor csr_sim, torch::Tensor phi_prob,
+                  torch::Tensor phi_alias, torch::Tensor pow_prob,
+                  torch::Tensor pow_alias, torch::Tensor pow_off,
+                  torch::Tensor log_pow_total, torch::Tensor attr_const, int64_t Kc,
+                  int64_t collapsed, int64_t sequential, int64_t seed,
+                  int64_t iteration, int64_t ent_id_base, torch::Tensor error_count);
+void distortion_update(torch::Tensor rec_values, torch::Tensor rec_dist,
+                       torch::Tensor rec_file, torch::Tensor rec_gid,
+                       torch::Tensor rec_ent, torch::Tensor ent_values,
+                       torch::Tensor theta, torch::Tensor phi,
+                       torch::Tensor norm_lin, torch::Tensor self_expsim,
+                       torch::Tensor voff, torch::Tensor attr_const, int64_t seed,
+                       int64_t iteration);
+void summary_loglik(torch::Tensor ent_values, torch::Tensor rec_values,
+                    torch::Tensor rec_dist, torch::Tensor rec_ent,
+                    torch::Tensor log_phi, torch::Tensor log_norm, torch::Tensor voff,
+                    torch::Tensor csr_row_ptr, torch::Tensor csr_col,
+                    torch::Tensor csr_sim, torch::Tensor attr_const,
+                    torch::Tensor out);
+void kd_descent(torch::Tensor ent_values, torch::Tensor node_kind,
+                torch::Tensor node_attr, torch::Tensor node_a, torch::Tensor node_b,
+                torch::Tensor rset, torch::Tensor ent_part_out);
+std::vector<torch::Tensor> sim_pairs_gpu(torch::Tensor strs, torch::Tensor lens,
+                                         double threshold, double max_sim);
+
+}  // namespace dblink
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "dblink_amd native ops (CDNA4 HIP kernels + host helpers)";
+  m.def("sim_pairs_cpu", &dblink::sim_pairs_cpu,
+        "banded Levenshtein sim-pair sweep (CPU/OpenMP)");
+  m.def("sim_pairs_gpu", &dblink::sim_pairs_gpu,
+        "banded Levenshtein sim-pair sweep (gfx950)");
+  m.def("link_update", &dblink::link_update, "K3/K4/K5 fused link update");
+  m.def("link_update_dense", &dblink::link_update_dense,
+        "dense link update (PCG-II / Gibbs-Sequential)");
+  m.def("value_update", &dblink::value_update, "K6 entity-value update");
+  m.def("distortion_update", &dblink::distortion_update, "K7 distortion resample");
+  m.def("summary_loglik", &dblink::summary_loglik, "K8 log-likelihood reduction");
+  m.def("kd_descent", &dblink::kd_descent, "K9a KD-tree partition reassignment");
+}

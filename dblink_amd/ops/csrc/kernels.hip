@@ -1,0 +1,949 @@
+// CDNA4 (gfx950) HIP kernels for the partitioned Gibbs sweep.
+//
+// Kernel inventory (SURVEY.md §2.4 K-list):
+//   K3+K4+K5  link_update           — posting-list intersection + per-candidate
+//                                     log-weights + fused Gumbel-max draw
+//             link_update_dense     — PCG-II collapsed / Gibbs-Sequential dense
+//                                     variants (LDS-tiled over entity values)
+//   K6        value_update          — per-(entity, attribute) collapsed /
+//                                     non-collapsed value draw with sparse
+//                                     perturbation weights merged in an LDS
+//                                     hash table; alias-table base draws
+//             value_update_seq      — brute-force dense-domain variant
+//   K7        distortion_update     — element-wise Bernoulli resample (Philox)
+//   K8        summary_loglik        — f64 block/atomic reduction of the
+//                                     log-likelihood terms
+//   K9a       kd_descent            — flat KD-tree partition reassignment
+//   K1 (GPU)  sim_pairs count/fill  — banded Levenshtein domain sweep
+//
+// One wave (64 lanes) owns one record / one (entity, attribute) pair; block
+// size 256 = 4 waves. All categorical draws use Gumbel-max with Philox
+// counter RNG (common.h), so results are independent of scheduling order.
+
+#include <ATen/cuda/CUDAContext.h>
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+
+#include "common.h"
+
+namespace dblink {
+
+#define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be a GPU tensor")
+#define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
+
+constexpr int MAX_ATTRS = 16;
+constexpr uint32_t PH_LINK = 1, PH_DIST = 2, PH_VALG = 3, PH_VALM = 4;
+
+// ---------------------------------------------------------------------------
+// K3+K4+K5: link update (PCG-I / Gibbs indexed path)
+// ---------------------------------------------------------------------------
+
+__global__ void link_update_kernel(
+    const int32_t* __restrict__ rec_values,  // [R, A]
+    const uint8_t* __restrict__ rec_dist,    // [R, A]
+    const int64_t* __restrict__ rec_gid,     // [R]
+    const int32_t* __restrict__ rec_part,    // [R]
+    const int64_t* __restrict__ cand_lo,     // [R, A]
+    const int64_t* __restrict__ cand_hi,     // [R, A]
+    const int32_t* __restrict__ postings,    // [E*A]
+    const int32_t* __restrict__ ent_values,  // [E, A]
+    const int64_t* __restrict__ ent_ptr,     // [P+1]
+    const float* __restrict__ log_norm,      // [Vtot]
+    const int64_t* __restrict__ voff,        // [A+1]
+    const int64_t* __restrict__ csr_row_ptr, // [Vtot+1]
+    const int32_t* __restrict__ csr_col,
+    const float* __restrict__ csr_sim,
+    const uint8_t* __restrict__ attr_const,  // [A]
+    int64_t R, int A,
+    uint64_t seed, uint32_t iteration,
+    int64_t* __restrict__ rec_ent_out,       // [R]
+    const int64_t* __restrict__ rec_ent_in,  // [R]
+    int* __restrict__ error_count) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t r = (int64_t)blockIdx.x * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
+  if (r >= R) return;
+
+  // Gather per-attribute candidate ranges (observed non-distorted) and the
+  // observed-distorted non-constant attributes (constant od attrs scale all
+  // weights equally and cancel under normalization).
+  int nd_n = 0, od_n = 0;
+  int nd_a[MAX_ATTRS];
+  int64_t nd_lo[MAX_ATTRS], nd_hi[MAX_ATTRS];
+  int od_a[MAX_ATTRS];
+  for (int a = 0; a < A; ++a) {
+    int32_t x = rec_values[r * A + a];
+    if (x < 0) continue;
+    if (!rec_dist[r * A + a]) {
+      nd_a[nd_n] = a;
+      nd_lo[nd_n] = cand_lo[r * A + a];
+      nd_hi[nd_n] = cand_hi[r * A + a];
+      ++nd_n;
+    } else if (!attr_const[a]) {
+      od_a[od_n++] = a;
+    }
+  }
+  // base = smallest candidate list
+  if (nd_n > 1) {
+    int best = 0;
+    int64_t best_sz = nd_hi[0] - nd_lo[0];
+    for (int i = 1; i < nd_n; ++i) {
+      int64_t sz = nd_hi[i] - nd_lo[i];
+      if (sz < best_sz) { best = i; best_sz = sz; }
+    }
+    if (best != 0) {
+      int ta = nd_a[0]; nd_a[0] = nd_a[best]; nd_a[best] = ta;
+      int64_t tl = nd_lo[0]; nd_lo[0] = nd_lo[best]; nd_lo[best] = tl;
+      int64_t th = nd_hi[0]; nd_hi[0] = nd_hi[best]; nd_hi[best] = th;
+    }
+  }
+
+  int64_t base_lo, base_n;
+  bool base_postings;
+  const int32_t p = rec_part[r];
+  if (nd_n == 0) {
+    base_lo = ent_ptr[p];
+    base_n = ent_ptr[p + 1] - base_lo;
+    base_postings = false;
+  } else {
+    base_lo = nd_lo[0];
+    base_n = nd_hi[0] - base_lo;
+    base_postings = true;
+  }
+
+  const uint64_t gid = (uint64_t)rec_gid[r];
+  float best_score = -INFINITY;
+  long long best_e = -1;
+  for (int64_t i = lane; i < base_n; i += WAVE) {
+    int32_t e = base_postings ? postings[base_lo + i] : (int32_t)(base_lo + i);
+    bool ok = true;
+    for (int j = 1; j < nd_n; ++j) {
+      if (!contains_i32(postings, nd_lo[j], nd_hi[j], e)) { ok = false; break; }
+    }
+    if (!ok) continue;
+    float logw = 0.0f;
+    for (int j = 0; j < od_n; ++j) {
+      const int a = od_a[j];
+      const int32_t x = rec_values[r * A + a];
+      const int32_t y = ent_values[(int64_t)e * A + a];
+      logw += log_norm[voff[a] + y] +
+              sim_lookup(csr_row_ptr, csr_col, csr_sim, voff[a] + x, y);
+    }
+    float g = gumbel_from_uniform(
+        philox_uniform(seed, iteration, PH_LINK, gid, (uint32_t)e));
+    float score = logw + g;
+    if (score > best_score) { best_score = score; best_e = e; }
+  }
+  wave_argmax(best_score, best_e);
+  if (lane == 0) {
+    if (best_e < 0) {  // empty candidate set: state invariant violated
+      atomicAdd(error_count, 1);
+      best_e = rec_ent_in[r];
+    }
+    rec_ent_out[r] = best_e;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Dense link update: PCG-II (collapsed) and Gibbs-Sequential
+// ---------------------------------------------------------------------------
+
+__global__ void link_update_dense_kernel(
+    const int32_t* __restrict__ rec_values, const uint8_t* __restrict__ rec_dist,
+    const int64_t* __restrict__ rec_gid, const int32_t* __restrict__ rec_part,
+    const int32_t* __restrict__ rec_file,
+    const int32_t* __restrict__ ent_values, const int64_t* __restrict__ ent_ptr,
+    const float* __restrict__ theta,       // [A, F]
+    const float* __restrict__ phi,         // [Vtot] linear probability
+    const float* __restrict__ norm_lin,    // [Vtot] linear 1/normalizer
+    const int64_t* __restrict__ voff, const int64_t* __restrict__ csr_row_ptr,
+    const int32_t* __restrict__ csr_col, const float* __restrict__ csr_sim,
+    const uint8_t* __restrict__ attr_const, int64_t R, int A, int F,
+    int collapsed,  // 1 = PCG-II weights, 0 = Gibbs-Sequential weights
+    uint64_t seed, uint32_t iteration,
+    int64_t* __restrict__ rec_ent_out) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t r = (int64_t)blockIdx.x * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
+  if (r >= R) return;
+  const int32_t p = rec_part[r];
+  const int64_t e0 = ent_ptr[p], e1 = ent_ptr[p + 1];
+  const int f = rec_file[r];
+  const uint64_t gid = (uint64_t)rec_gid[r];
+
+  float best_score = -INFINITY;
+  long long best_e = -1;
+  for (int64_t e = e0 + lane; e < e1; e += WAVE) {
+    float logw = 0.0f;
+    for (int a = 0; a < A; ++a) {
+      const int32_t x = rec_values[r * A + a];
+      if (x < 0) continue;
+      const int32_t y = ent_values[e * A + a];
+      if (collapsed) {
+        const float th = theta[a * F + f];
+        float like = phi[voff[a] + x];
+        if (!attr_const[a]) {
+          float s = sim_lookup(csr_row_ptr, csr_col, csr_sim, voff[a] + x, y);
+          like *= norm_lin[voff[a] + y] * __expf(s);
+        }
+        float w = (y == x ? 1.0f - th : 0.0f) + th * like;
+        logw += __logf(w);
+      } else {
+        // Gibbs-Sequential: non-distorted must match exactly
+        if (!rec_dist[r * A + a]) {
+          if (x != y) { logw = -INFINITY; break; }
+        } else if (!attr_const[a]) {
+          float s = sim_lookup(csr_row_ptr, csr_col, csr_sim, voff[a] + x, y);
+          logw += __logf(norm_lin[voff[a] + y]) + s;  // phi(x) constant: cancels
+        }
+      }
+    }
+    if (logw == -INFINITY) continue;
+    float g = gumbel_from_uniform(
+        philox_uniform(seed, iteration, PH_LINK, gid, (uint32_t)(e - e0)));
+    float score = logw + g;
+    if (score > best_score) { best_score = score; best_e = e; }
+  }
+  wave_argmax(best_score, best_e);
+  if (lane == 0) rec_ent_out[r] = best_e;  // weights always > 0 when collapsed
+}
+
+// ---------------------------------------------------------------------------
+// K6: value update
+// ---------------------------------------------------------------------------
+
+// LDS hash table per wave for sparse perturbation weights.
+constexpr int HASH_CAP = 1024;            // slots per wave
+constexpr int WAVES_PER_BLOCK_VAL = 4;    // 256 threads
+
+struct ValueArgs {
+  const int32_t* rec_values;
+  const uint8_t* rec_dist;
+  const int32_t* rec_file;
+  const int64_t* ent_rec_ptr;  // [E+1]
+  const int64_t* ent_rec_idx;  // [R] record rows grouped by entity
+  int32_t* ent_values;         // [E, A] in/out
+  const float* theta;          // [A, F]
+  const float* phi;            // [Vtot]
+  const float* log_phi;        // [Vtot]
+  const float* norm_lin;       // [Vtot]
+  const float* log_norm;       // [Vtot]
+  const int64_t* voff;         // [A+1]
+  const int64_t* csr_row_ptr;
+  const int32_t* csr_col;
+  const float* csr_sim;
+  const float* phi_prob;       // [Vtot] alias prob for phi
+  const int32_t* phi_alias;    // [Vtot]
+  const float* pow_prob;       // power-dist alias tables, concatenated
+  const int32_t* pow_alias;
+  const int64_t* pow_off;      // [A] offset into pow_* for k=1 (or -1)
+  const float* log_pow_total;  // [A * (Kc+1)] log Z_k (index a*(Kc+1)+k)
+  const uint8_t* attr_const;
+  int Kc;
+  int64_t E;
+  int A, F;
+  int collapsed;
+  uint64_t seed;
+  uint32_t iteration;
+  uint64_t ent_id_base;
+  int* error_count;
+};
+
+// Draw from p(v) ~ phi(v)*norm(v)^k by a dense Gumbel scan (rare path for
+// k > Kc where no alias table is cached). Also returns log Z_k via wave sum.
+__device__ int dense_power_draw(const ValueArgs& args, int a, int k, uint64_t elem,
+                                int lane, float* out_log_total) {
+  const int64_t v0 = args.voff[a], v1 = args.voff[a + 1];
+  float best = -INFINITY;
+  long long best_v = 0;
+  double total = 0.0;
+  for (int64_t v = v0 + lane; v < v1; v += WAVE) {
+    float lw = args.log_phi[v] + (float)k * args.log_norm[v];
+    total += exp((double)lw);
+    float g = gumbel_from_uniform(philox_uniform(args.seed, args.iteration, PH_VALG,
+                                                 elem, (uint32_t)(v - v0) | 0x40000000u));
+    if (lw + g > best) { best = lw + g; best_v = v - v0; }
+  }
+  total = wave_sum(total);
+  wave_argmax(best, best_v);
+  *out_log_total = (float)log(total);
+  return (int)best_v;
+}
+
+__global__ void __launch_bounds__(WAVES_PER_BLOCK_VAL * WAVE)
+value_update_kernel(ValueArgs args) {
+  __shared__ int32_t h_key[WAVES_PER_BLOCK_VAL][HASH_CAP];
+  __shared__ float h_val[WAVES_PER_BLOCK_VAL][HASH_CAP];
+
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = threadIdx.x / WAVE;
+  const int64_t pair = (int64_t)blockIdx.x * WAVES_PER_BLOCK_VAL + wave;
+  const int64_t total_pairs = args.E * args.A;
+  if (pair >= total_pairs) return;
+  const int64_t e = pair / args.A;
+  const int a = (int)(pair % args.A);
+  const bool is_const = args.attr_const[a];
+  const int64_t v0 = args.voff[a];
+  const int V = (int)(args.voff[a + 1] - v0);
+  const uint64_t elem = (args.ent_id_base + (uint64_t)e) * 32u + (uint64_t)a;
+
+  // ---- gather observed linked records -------------------------------------
+  const int64_t r_lo = args.ent_rec_ptr[e], r_hi = args.ent_rec_ptr[e + 1];
+  int k_obs = 0;
+  int32_t first_nondist = -1;  // first observed non-distorted value (record order)
+  for (int64_t i = r_lo; i < r_hi; ++i) {
+    const int64_t r = args.ent_rec_idx[i];
+    const int32_t x = args.rec_values[r * args.A + a];
+    if (x < 0) continue;
+    ++k_obs;
+    if (!args.collapsed && first_nondist < 0 && !args.rec_dist[r * args.A + a])
+      first_nondist = x;
+  }
+
+  // base distribution: phi for const attrs or k_obs == 0, else power dist k
+  auto base_draw = [&](uint32_t draw_tag) -> int {
+    float u1, u2;
+    philox_uniform2(args.seed, args.iteration, PH_VALM, elem, draw_tag, &u1, &u2);
+    if (is_const || k_obs == 0) {
+      return alias_draw(args.phi_prob + v0, args.phi_alias + v0, V, u1, u2);
+    }
+    if (k_obs <= args.Kc) {
+      const int64_t off = args.pow_off[a] + (int64_t)(k_obs - 1) * V;
+      return alias_draw(args.pow_prob + off, args.pow_alias + off, V, u1, u2);
+    }
+    float dummy;
+    return dense_power_draw(args, a, k_obs, elem, lane, &dummy);
+  };
+
+  // non-collapsed deterministic copy (GibbsUpdates.scala:619-630)
+  if (!args.collapsed && first_nondist >= 0) {
+    if (lane == 0) args.ent_values[e * args.A + a] = first_nondist;
+    return;
+  }
+
+  if (k_obs == 0) {
+    int v = base_draw(0xFFFF0000u);
+    if (lane == 0) args.ent_values[e * args.A + a] = (int32_t)v;
+    return;
+  }
+
+  if (!args.collapsed && is_const) {  // no perturbation for const non-collapsed
+    int v = base_draw(0xFFFF0000u);
+    if (lane == 0) args.ent_values[e * args.A + a] = (int32_t)v;
+    return;
+  }
+
+  // ---- perturbation weights in the LDS hash table -------------------------
+  // L_v = sum_r log f_r(v); final weight = base_prob(v) * (exp(L_v) - 1).
+  int32_t* keys = h_key[wave];
+  float* vals = h_val[wave];
+  for (int i = lane; i < HASH_CAP; i += WAVE) { keys[i] = -1; vals[i] = 0.0f; }
+  __builtin_amdgcn_wave_barrier();
+
+  // dense fallback decision: total row entries vs capacity
+  int64_t total_entries = 0;
+  if (!is_const) {
+    for (int64_t i = r_lo; i < r_hi; ++i) {
+      const int64_t r = args.ent_rec_idx[i];
+      const int32_t x = args.rec_values[r * args.A + a];
+      if (x < 0) continue;
+      total_entries += args.csr_row_ptr[v0 + x + 1] - args.csr_row_ptr[v0 + x];
+    }
+  } else {
+    total_entries = k_obs;
+  }
+  const bool dense = total_entries > (HASH_CAP * 3) / 4;
+
+  // log of normalized base probability of value v (local id)
+  const float log_z = (is_const || k_obs == 0)
+                          ? 0.0f
+                          : (k_obs <= args.Kc
+                                 ? args.log_pow_total[a * (args.Kc + 1) + k_obs]
+                                 : 0.0f);  // patched below for the rare path
+  float log_z_rare = 0.0f;
+  if (!is_const && k_obs > args.Kc) {
+    // need log Z_k for base_prob; compute by wave reduction
+    double tot = 0.0;
+    for (int64_t v = v0 + lane; v < v0 + V; v += WAVE)
+      tot += exp((double)(args.log_phi[v] + (float)k_obs * args.log_norm[v]));
+    log_z_rare = (float)log(wave_sum(tot));
+  }
+  auto log_base_prob = [&](int v_local) -> float {
+    const int64_t v = v0 + v_local;
+    if (is_const) return args.log_phi[v];
+    const float lz = (k_obs <= args.Kc) ? log_z : log_z_rare;
+    return args.log_phi[v] + (float)k_obs * args.log_norm[v] - lz;
+  };
+
+  double W = 0.0;            // total perturbation weight
+  float best = -INFINITY;    // gumbel-max over perturbation weights
+  long long best_v = -1;
+
+  if (!dense) {
+    // hash-accumulate log factors
+    for (int64_t i = r_lo; i < r_hi; ++i) {
+      const int64_t r = args.ent_rec_idx[i];
+      const int32_t x = args.rec_values[r * args.A + a];
+      if (x < 0) continue;
+      float self_extra = 0.0f;
+      if (args.collapsed) {
+        const float th = args.theta[a * args.F + args.rec_file[r]];
+        const float px = args.phi[v0 + x];
+        self_extra = (1.0f / th - 1.0f) /
+                     (is_const ? px : px * args.norm_lin[v0 + x]);
+      }
+      if (is_const) {
+        // single-entry row {x}: factor = 1 + self_extra
+        if (lane == 0) {
+          float logf_ = __logf(1.0f + self_extra);
+          // linear-probe insert
+          uint32_t h = ((uint32_t)x * 2654435761u) & (HASH_CAP - 1);
+          while (true) {
+            int32_t prev = atomicCAS(&keys[h], -1, x);
+            if (prev == -1 || prev == x) { atomicAdd(&vals[h], logf_); break; }
+            h = (h + 1) & (HASH_CAP - 1);
+          }
+        }
+      } else {
+        const int64_t row_lo = args.csr_row_ptr[v0 + x], row_hi = args.csr_row_ptr[v0 + x + 1];
+        for (int64_t j = row_lo + lane; j < row_hi; j += WAVE) {
+          const int32_t v = args.csr_col[j];
+          const float s = args.csr_sim[j];  // log expsim > 0
+          float factor_log = (v == x && self_extra > 0.0f)
+                                 ? __logf(__expf(s) + self_extra)
+                                 : s;
+          uint32_t h = ((uint32_t)v * 2654435761u) & (HASH_CAP - 1);
+          while (true) {
+            int32_t prev = atomicCAS(&keys[h], -1, v);
+            if (prev == -1 || prev == v) { atomicAdd(&vals[h], factor_log); break; }
+            h = (h + 1) & (HASH_CAP - 1);
+          }
+        }
+      }
+      __builtin_amdgcn_wave_barrier();
+    }
+    // transform + reduce + gumbel-max over occupied slots
+    for (int i = lane; i < HASH_CAP; i += WAVE) {
+      const int32_t v = keys[i];
+      if (v < 0) continue;
+      const float L = vals[i];
+      // log(exp(L) - 1) = L + log1p(-exp(-L)), stable for L > 0
+      const float log_expm1 = L + __logf(1.0f - __expf(-L));
+      const float logw = log_base_prob(v) + log_expm1;
+      W += exp((double)logw);
+      const float g = gumbel_from_uniform(
+          philox_uniform(args.seed, args.iteration, PH_VALG, elem, (uint32_t)v));
+      if (logw + g > best) { best = logw + g; best_v = v; }
+    }
+  } else {
+    // dense path: for every domain value accumulate log factors by searching
+    // each record's sim row (correct for any cluster size; O(V * k log row))
+    for (int v_local = lane; v_local < V; v_local += WAVE) {
+      float L = 0.0f;
+      for (int64_t i = r_lo; i < r_hi; ++i) {
+        const int64_t r = args.ent_rec_idx[i];
+        const int32_t x = args.rec_values[r * args.A + a];
+        if (x < 0) continue;
+        float self_extra = 0.0f;
+        if (args.collapsed) {
+          const float th = args.theta[a * args.F + args.rec_file[r]];
+          const float px = args.phi[v0 + x];
+          self_extra = (1.0f / th - 1.0f) /
+                       (is_const ? px : px * args.norm_lin[v0 + x]);
+        }
+        if (is_const) {
+          if (v_local == x) L += __logf(1.0f + self_extra);
+        } else {
+          const float s = sim_lookup(args.csr_row_ptr, args.csr_col, args.csr_sim,
+                                     v0 + x, v_local);
+          if (v_local == x && self_extra > 0.0f)
+            L += __logf(__expf(s) + self_extra);
+          else if (s != 0.0f)
+            L += s;
+        }
+      }
+      if (L > 0.0f) {
+        const float log_expm1 = L + __logf(1.0f - __expf(-L));
+        const float logw = log_base_prob(v_local) + log_expm1;
+        W += exp((double)logw);
+        const float g = gumbel_from_uniform(
+            philox_uniform(args.seed, args.iteration, PH_VALG, elem, (uint32_t)v_local));
+        if (logw + g > best) { best = logw + g; best_v = v_local; }
+      }
+    }
+  }
+  W = wave_sum(W);
+  wave_argmax(best, best_v);
+
+  // mixture between base and perturbation (GibbsUpdates.scala:593-597)
+  const float u = philox_uniform(args.seed, args.iteration, PH_VALM, elem, 0xFFFF0001u);
+  int v_new;
+  if ((double)u < 1.0 / (1.0 + W) || best_v < 0) {
+    v_new = base_draw(0xFFFF0002u);
+  } else {
+    v_new = (int)best_v;
+  }
+  if (lane == 0) args.ent_values[e * args.A + a] = (int32_t)v_new;
+}
+
+// Brute-force dense value update (Gibbs-Sequential, GibbsUpdates.scala:652-698)
+__global__ void value_update_seq_kernel(ValueArgs args) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = threadIdx.x / WAVE;
+  const int64_t pair = (int64_t)blockIdx.x * (blockDim.x / WAVE) + wave;
+  if (pair >= args.E * args.A) return;
+  const int64_t e = pair / args.A;
+  const int a = (int)(pair % args.A);
+  const bool is_const = args.attr_const[a];
+  const int64_t v0 = args.voff[a];
+  const int V = (int)(args.voff[a + 1] - v0);
+  const uint64_t elem = (args.ent_id_base + (uint64_t)e) * 32u + (uint64_t)a;
+
+  const int64_t r_lo = args.ent_rec_ptr[e], r_hi = args.ent_rec_ptr[e + 1];
+  int k_obs = 0;
+  int32_t first_nondist = -1;
+  for (int64_t i = r_lo; i < r_hi; ++i) {
+    const int64_t r = args.ent_rec_idx[i];
+    const int32_t x = args.rec_values[r * args.A + a];
+    if (x < 0) continue;
+    ++k_obs;
+    if (first_nondist < 0 && !args.rec_dist[r * args.A + a]) first_nondist = x;
+  }
+  if (first_nondist >= 0 && k_obs > 0) {
+    if (lane == 0) args.ent_values[e * args.A + a] = first_nondist;
+    return;
+  }
+  if (k_obs == 0 || is_const) {
+    float u1, u2;
+    philox_uniform2(args.seed, args.iteration, PH_VALM, elem, 0xFFFF0000u, &u1, &u2);
+    int v = alias_draw(args.phi_prob + v0, args.phi_alias + v0, V, u1, u2);
+    if (lane == 0) args.ent_values[e * args.A + a] = (int32_t)v;
+    return;
+  }
+  // dense scan: w(v) = phi(v) * prod_r [expsim(x_r, v) * norm(v) * phi(x_r)]
+  float best = -INFINITY;
+  long long best_v = 0;
+  for (int v_local = lane; v_local < V; v_local += WAVE) {
+    float lw = args.log_phi[v0 + v_local];
+    for (int64_t i = r_lo; i < r_hi; ++i) {
+      const int64_t r = args.ent_rec_idx[i];
+      const int32_t x = args.rec_values[r * args.A + a];
+      if (x < 0) continue;
+      lw += sim_lookup(args.csr_row_ptr, args.csr_col, args.csr_sim, v0 + x, v_local) +
+            args.log_norm[v0 + v_local] + args.log_phi[v0 + x];
+    }
+    const float g = gumbel_from_uniform(
+        philox_uniform(args.seed, args.iteration, PH_VALG, elem, (uint32_t)v_local));
+    if (lw + g > best) { best = lw + g; best_v = v_local; }
+  }
+  wave_argmax(best, best_v);
+  if (lane == 0) args.ent_values[e * args.A + a] = (int32_t)best_v;
+}
+
+// ---------------------------------------------------------------------------
+// K7: distortion update (element-wise over record-attribute pairs)
+// ---------------------------------------------------------------------------
+
+__global__ void distortion_update_kernel(
+    const int32_t* __restrict__ rec_values, uint8_t* __restrict__ rec_dist,
+    const int32_t* __restrict__ rec_file, const int64_t* __restrict__ rec_gid,
+    const int64_t* __restrict__ rec_ent, const int32_t* __restrict__ ent_values,
+    const float* __restrict__ theta, const float* __restrict__ phi,
+    const float* __restrict__ norm_lin, const float* __restrict__ self_expsim,
+    const int64_t* __restrict__ voff, const uint8_t* __restrict__ attr_const,
+    int64_t R, int A, int F, uint64_t seed, uint32_t iteration) {
+  const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= R * A) return;
+  const int64_t r = idx / A;
+  const int a = (int)(idx % A);
+  const int32_t x = rec_values[idx];
+  const float th = theta[a * F + rec_file[r]];
+  const float u = philox_uniform(seed, iteration, PH_DIST,
+                                 (uint64_t)rec_gid[r] * 32u + (uint64_t)a, 0);
+  uint8_t z;
+  if (x < 0) {
+    z = u < th;
+  } else {
+    const int32_t y = ent_values[rec_ent[r] * A + a];
+    if (x == y) {
+      float pr1 = th * phi[voff[a] + x];
+      if (!attr_const[a]) pr1 *= norm_lin[voff[a] + x] * self_expsim[voff[a] + x];
+      const float pr0 = 1.0f - th;
+      const float psum = pr0 + pr1;
+      const float pz = psum != 0.0f ? pr1 / psum : 0.0f;
+      z = u < pz;
+    } else {
+      z = 1;
+    }
+  }
+  rec_dist[idx] = z;
+}
+
+// ---------------------------------------------------------------------------
+// K8: log-likelihood reduction (entity priors + distorted record terms)
+// ---------------------------------------------------------------------------
+
+__global__ void summary_loglik_kernel(
+    const int32_t* __restrict__ ent_values, const int32_t* __restrict__ rec_values,
+    const uint8_t* __restrict__ rec_dist, const int64_t* __restrict__ rec_ent,
+    const float* __restrict__ log_phi, const float* __restrict__ log_norm,
+    const int64_t* __restrict__ voff, const int64_t* __restrict__ csr_row_ptr,
+    const int32_t* __restrict__ csr_col, const float* __restrict__ csr_sim,
+    const uint8_t* __restrict__ attr_const, int64_t E, int64_t R, int A,
+    double* __restrict__ out) {
+  const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t total = (E + R) * A;
+  double contrib = 0.0;
+  if (idx < E * A) {
+    const int64_t e = idx / A;
+    const int a = (int)(idx % A);
+    contrib = (double)log_phi[voff[a] + ent_values[e * A + a]];
+  } else if (idx < total) {
+    const int64_t j = idx - E * A;
+    const int64_t r = j / A;
+    const int a = (int)(j % A);
+    if (rec_dist[r * A + a]) {
+      const int32_t x = rec_values[r * A + a];
+      if (x >= 0) {
+        float lp = log_phi[voff[a] + x];
+        if (!attr_const[a]) {
+          const int32_t y = ent_values[rec_ent[r] * A + a];
+          lp += log_norm[voff[a] + y] +
+                sim_lookup(csr_row_ptr, csr_col, csr_sim, voff[a] + x, y);
+        }
+        contrib = (double)lp;
+      }
+    }
+  }
+  // block reduce then one atomic per block
+  __shared__ double partial[256];
+  partial[threadIdx.x] = contrib;
+  __syncthreads();
+  for (int s = blockDim.x / 2; s > 0; s >>= 1) {
+    if (threadIdx.x < s) partial[threadIdx.x] += partial[threadIdx.x + s];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) atomicAdd(out, partial[0]);
+}
+
+// ---------------------------------------------------------------------------
+// K9a: KD-tree descent (flat tree, partitioning.py as_flat layout)
+// ---------------------------------------------------------------------------
+
+__global__ void kd_descent_kernel(
+    const int32_t* __restrict__ ent_values, const int32_t* __restrict__ node_kind,
+    const int32_t* __restrict__ node_attr, const int32_t* __restrict__ node_a,
+    const int32_t* __restrict__ node_b, const int32_t* __restrict__ rset,
+    int64_t E, int A, int32_t* __restrict__ ent_part_out) {
+  const int64_t e = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (e >= E) return;
+  int nid = 0;
+  while (node_kind[nid] != 0) {
+    const int a = node_attr[nid];
+    const int32_t v = ent_values[e * A + a];
+    bool right;
+    if (node_kind[nid] == 1) {
+      right = v > node_a[nid];
+    } else {
+      const int lo = node_a[nid], n = node_b[nid];
+      right = contains_i32(rset, lo, lo + n, v);
+    }
+    nid = 2 * nid + (right ? 2 : 1);
+  }
+  ent_part_out[e] = node_a[nid];
+}
+
+// ---------------------------------------------------------------------------
+// K1 (GPU): banded Levenshtein sim-pair sweep over a value domain
+// ---------------------------------------------------------------------------
+
+// One thread per candidate pair (i, j); strings padded to max_len bytes.
+// Band pruning: pairs are pre-filtered by length difference on the host side
+// (sim > 0 requires d <= dmax(Li+Lj)), the kernel re-checks and computes the
+// exact banded DP.
+__device__ float lev_unit_sim(const uint8_t* a, int la, const uint8_t* b, int lb) {
+  // full DP over rows of b with rolling arrays in registers/scratch.
+  // strings are short (attribute values); cap at 64.
+  constexpr int MAXLEN = 64;
+  uint8_t prev[MAXLEN + 1];
+  uint8_t cur[MAXLEN + 1];
+  if (la + lb == 0) return 1.0f;
+  for (int j = 0; j <= lb; ++j) prev[j] = (uint8_t)j;
+  for (int i = 1; i <= la; ++i) {
+    cur[0] = (uint8_t)i;
+    const uint8_t ca = a[i - 1];
+    for (int j = 1; j <= lb; ++j) {
+      uint8_t cost = (ca == b[j - 1]) ? 0 : 1;
+      uint8_t m = prev[j] + 1;
+      uint8_t d = cur[j - 1] + 1;
+      uint8_t s = prev[j - 1] + cost;
+      if (d < m) m = d;
+      if (s < m) m = s;
+      cur[j] = m;
+    }
+    for (int j = 0; j <= lb; ++j) prev[j] = cur[j];
+  }
+  const float dist = (float)prev[lb];
+  return 1.0f - 2.0f * dist / ((float)(la + lb) + dist);
+}
+
+__global__ void sim_pairs_kernel(
+    const uint8_t* __restrict__ strs,  // [V, max_len]
+    const int32_t* __restrict__ lens,  // [V]
+    int V, int max_len, float threshold, float max_sim,
+    int64_t* __restrict__ row_counts,  // [V] (pass 1out) or row offsets (pass 2 in)
+    int32_t* __restrict__ out_col,     // pass 2
+    float* __restrict__ out_expsim,    // pass 2 (stores exp(sim))
+    int64_t* __restrict__ fill_pos,    // [V] atomic cursors (pass 2)
+    int fill) {
+  const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= (int64_t)V * V) return;
+  const int i = (int)(idx / V);
+  const int j = (int)(idx % V);
+  const int la = lens[i], lb = lens[j];
+  // length pruning: unit > thr/max requires d < (la+lb)*(1-u0)/(1+u0),
+  // and d >= |la-lb|.
+  const float u0 = threshold / max_sim;
+  const float dmax = (float)(la + lb) * (1.0f - u0) / (1.0f + u0);
+  if ((float)abs(la - lb) >= dmax && (la + lb) > 0) return;
+  const float unit = lev_unit_sim(strs + (int64_t)i * max_len, la,
+                                  strs + (int64_t)j * max_len, lb);
+  const float trans = max_sim / (max_sim - threshold) * (max_sim * unit - threshold);
+  if (trans <= 0.0f) return;  // expsim <= 1 filtered (AttributeIndex.scala:226)
+  if (!fill) {
+    atomicAdd((unsigned long long*)&row_counts[i], 1ull);
+  } else {
+    const int64_t pos = atomicAdd((unsigned long long*)&fill_pos[i], 1ull) + row_counts[i];
+    out_col[pos] = j;
+    out_expsim[pos] = __expf(trans);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// host launchers
+// ---------------------------------------------------------------------------
+
+static int64_t wave_grid(int64_t items, int waves_per_block) {
+  return (items + waves_per_block - 1) / waves_per_block;
+}
+
+void link_update(
+    torch::Tensor rec_values, torch::Tensor rec_dist, torch::Tensor rec_gid,
+    torch::Tensor rec_part, torch::Tensor cand_lo, torch::Tensor cand_hi,
+    torch::Tensor postings, torch::Tensor ent_values, torch::Tensor ent_ptr,
+    torch::Tensor log_norm, torch::Tensor voff, torch::Tensor csr_row_ptr,
+    torch::Tensor csr_col, torch::Tensor csr_sim, torch::Tensor attr_const,
+    int64_t seed, int64_t iteration, torch::Tensor rec_ent_out,
+    torch::Tensor rec_ent_in, torch::Tensor error_count) {
+  CHECK_GPU(rec_values);
+  const int64_t R = rec_values.size(0);
+  const int A = (int)rec_values.size(1);
+  TORCH_CHECK(A <= MAX_ATTRS, "at most ", MAX_ATTRS, " matching attributes supported");
+  if (R == 0) return;
+  constexpr int WPB = 4;
+  dim3 grid((unsigned)wave_grid(R, WPB));
+  hipLaunchKernelGGL(link_update_kernel, grid, dim3(WPB * WAVE), 0,
+                     at::cuda::getCurrentCUDAStream(),
+                     rec_values.data_ptr<int32_t>(), rec_dist.data_ptr<uint8_t>(),
+                     rec_gid.data_ptr<int64_t>(), rec_part.data_ptr<int32_t>(),
+                     cand_lo.data_ptr<int64_t>(), cand_hi.data_ptr<int64_t>(),
+                     postings.data_ptr<int32_t>(), ent_values.data_ptr<int32_t>(),
+                     ent_ptr.data_ptr<int64_t>(), log_norm.data_ptr<float>(),
+                     voff.data_ptr<int64_t>(), csr_row_ptr.data_ptr<int64_t>(),
+                     csr_col.data_ptr<int32_t>(), csr_sim.data_ptr<float>(),
+                     attr_const.data_ptr<uint8_t>(), R, A, (uint64_t)seed,
+                     (uint32_t)iteration, rec_ent_out.data_ptr<int64_t>(),
+                     rec_ent_in.data_ptr<int64_t>(), error_count.data_ptr<int>());
+}
+
+void link_update_dense(
+    torch::Tensor rec_values, torch::Tensor rec_dist, torch::Tensor rec_gid,
+    torch::Tensor rec_part, torch::Tensor rec_file, torch::Tensor ent_values,
+    torch::Tensor ent_ptr, torch::Tensor theta, torch::Tensor phi,
+    torch::Tensor norm_lin, torch::Tensor voff, torch::Tensor csr_row_ptr,
+    torch::Tensor csr_col, torch::Tensor csr_sim, torch::Tensor attr_const,
+    int64_t collapsed, int64_t seed, int64_t iteration, torch::Tensor rec_ent_out) {
+  const int64_t R = rec_values.size(0);
+  const int A = (int)rec_values.size(1);
+  const int F = (int)theta.size(1);
+  if (R == 0) return;
+  constexpr int WPB = 4;
+  dim3 grid((unsigned)wave_grid(R, WPB));
+  hipLaunchKernelGGL(link_update_dense_kernel, grid, dim3(WPB * WAVE), 0,
+                     at::cuda::getCurrentCUDAStream(),
+                     rec_values.data_ptr<int32_t>(), rec_dist.data_ptr<uint8_t>(),
+                     rec_gid.data_ptr<int64_t>(), rec_part.data_ptr<int32_t>(),
+                     rec_file.data_ptr<int32_t>(), ent_values.data_ptr<int32_t>(),
+                     ent_ptr.data_ptr<int64_t>(), theta.data_ptr<float>(),
+                     phi.data_ptr<float>(), norm_lin.data_ptr<float>(),
+                     voff.data_ptr<int64_t>(), csr_row_ptr.data_ptr<int64_t>(),
+                     csr_col.data_ptr<int32_t>(), csr_sim.data_ptr<float>(),
+                     attr_const.data_ptr<uint8_t>(), R, A, F, (int)collapsed,
+                     (uint64_t)seed, (uint32_t)iteration,
+                     rec_ent_out.data_ptr<int64_t>());
+}
+
+static ValueArgs make_value_args(
+    torch::Tensor rec_values, torch::Tensor rec_dist, torch::Tensor rec_file,
+    torch::Tensor ent_rec_ptr, torch::Tensor ent_rec_idx, torch::Tensor ent_values,
+    torch::Tensor theta, torch::Tensor phi, torch::Tensor log_phi,
+    torch::Tensor norm_lin, torch::Tensor log_norm, torch::Tensor voff,
+    torch::Tensor csr_row_ptr, torch::Tensor csr_col, torch::Tensor csr_sim,
+    torch::Tensor phi_prob, torch::Tensor phi_alias, torch::Tensor pow_prob,
+    torch::Tensor pow_alias, torch::Tensor pow_off, torch::Tensor log_pow_total,
+    torch::Tensor attr_const, int64_t Kc, int64_t collapsed, int64_t seed,
+    int64_t iteration, int64_t ent_id_base, torch::Tensor error_count) {
+  ValueArgs a;
+  a.rec_values = rec_values.data_ptr<int32_t>();
+  a.rec_dist = rec_dist.data_ptr<uint8_t>();
+  a.rec_file = rec_file.data_ptr<int32_t>();
+  a.ent_rec_ptr = ent_rec_ptr.data_ptr<int64_t>();
+  a.ent_rec_idx = ent_rec_idx.data_ptr<int64_t>();
+  a.ent_values = ent_values.data_ptr<int32_t>();
+  a.theta = theta.data_ptr<float>();
+  a.phi = phi.data_ptr<float>();
+  a.log_phi = log_phi.data_ptr<float>();
+  a.norm_lin = norm_lin.data_ptr<float>();
+  a.log_norm = log_norm.data_ptr<float>();
+  a.voff = voff.data_ptr<int64_t>();
+  a.csr_row_ptr = csr_row_ptr.data_ptr<int64_t>();
+  a.csr_col = csr_col.data_ptr<int32_t>();
+  a.csr_sim = csr_sim.data_ptr<float>();
+  a.phi_prob = phi_prob.data_ptr<float>();
+  a.phi_alias = phi_alias.data_ptr<int32_t>();
+  a.pow_prob = pow_prob.numel() ? pow_prob.data_ptr<float>() : nullptr;
+  a.pow_alias = pow_alias.numel() ? pow_alias.data_ptr<int32_t>() : nullptr;
+  a.pow_off = pow_off.data_ptr<int64_t>();
+  a.log_pow_total = log_pow_total.data_ptr<float>();
+  a.attr_const = attr_const.data_ptr<uint8_t>();
+  a.Kc = (int)Kc;
+  a.E = ent_values.size(0);
+  a.A = (int)ent_values.size(1);
+  a.F = (int)theta.size(1);
+  a.collapsed = (int)collapsed;
+  a.seed = (uint64_t)seed;
+  a.iteration = (uint32_t)iteration;
+  a.ent_id_base = (uint64_t)ent_id_base;
+  a.error_count = error_count.data_ptr<int>();
+  return a;
+}
+
+void value_update(
+    torch::Tensor rec_values, torch::Tensor rec_dist, torch::Tensor rec_file,
+    torch::Tensor ent_rec_ptr, torch::Tensor ent_rec_idx, torch::Tensor ent_values,
+    torch::Tensor theta, torch::Tensor phi, torch::Tensor log_phi,
+    torch::Tensor norm_lin, torch::Tensor log_norm, torch::Tensor voff,
+    torch::Tensor csr_row_ptr, torch::Tensor csr_col, torch::Tensor csr_sim,
+    torch::Tensor phi_prob, torch::Tensor phi_alias, torch::Tensor pow_prob,
+    torch::Tensor pow_alias, torch::Tensor pow_off, torch::Tensor log_pow_total,
+    torch::Tensor attr_const, int64_t Kc, int64_t collapsed, int64_t sequential,
+    int64_t seed, int64_t iteration, int64_t ent_id_base, torch::Tensor error_count) {
+  ValueArgs args = make_value_args(
+      rec_values, rec_dist, rec_file, ent_rec_ptr, ent_rec_idx, ent_values, theta,
+      phi, log_phi, norm_lin, log_norm, voff, csr_row_ptr, csr_col, csr_sim,
+      phi_prob, phi_alias, pow_prob, pow_alias, pow_off, log_pow_total, attr_const,
+      Kc, collapsed, seed, iteration, ent_id_base, error_count);
+  const int64_t pairs = args.E * args.A;
+  if (pairs == 0) return;
+  if (sequential) {
+    dim3 grid((unsigned)wave_grid(pairs, 4));
+    hipLaunchKernelGGL(value_update_seq_kernel, grid, dim3(4 * WAVE), 0,
+                       at::cuda::getCurrentCUDAStream(), args);
+  } else {
+    dim3 grid((unsigned)wave_grid(pairs, WAVES_PER_BLOCK_VAL));
+    hipLaunchKernelGGL(value_update_kernel, grid, dim3(WAVES_PER_BLOCK_VAL * WAVE), 0,
+                       at::cuda::getCurrentCUDAStream(), args);
+  }
+}
+
+void distortion_update(
+    torch::Tensor rec_values, torch::Tensor rec_dist, torch::Tensor rec_file,
+    torch::Tensor rec_gid, torch::Tensor rec_ent, torch::Tensor ent_values,
+    torch::Tensor theta, torch::Tensor phi, torch::Tensor norm_lin,
+    torch::Tensor self_expsim, torch::Tensor voff, torch::Tensor attr_const,
+    int64_t seed, int64_t iteration) {
+  const int64_t R = rec_values.size(0);
+  const int A = (int)rec_values.size(1);
+  const int F = (int)theta.size(1);
+  if (R == 0) return;
+  const int64_t total = R * A;
+  dim3 grid((unsigned)((total + 255) / 256));
+  hipLaunchKernelGGL(distortion_update_kernel, grid, dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(),
+                     rec_values.data_ptr<int32_t>(), rec_dist.data_ptr<uint8_t>(),
+                     rec_file.data_ptr<int32_t>(), rec_gid.data_ptr<int64_t>(),
+                     rec_ent.data_ptr<int64_t>(), ent_values.data_ptr<int32_t>(),
+                     theta.data_ptr<float>(), phi.data_ptr<float>(),
+                     norm_lin.data_ptr<float>(), self_expsim.data_ptr<float>(),
+                     voff.data_ptr<int64_t>(), attr_const.data_ptr<uint8_t>(),
+                     R, A, F, (uint64_t)seed, (uint32_t)iteration);
+}
+
+void summary_loglik(
+    torch::Tensor ent_values, torch::Tensor rec_values, torch::Tensor rec_dist,
+    torch::Tensor rec_ent, torch::Tensor log_phi, torch::Tensor log_norm,
+    torch::Tensor voff, torch::Tensor csr_row_ptr, torch::Tensor csr_col,
+    torch::Tensor csr_sim, torch::Tensor attr_const, torch::Tensor out) {
+  const int64_t E = ent_values.size(0);
+  const int64_t R = rec_values.size(0);
+  const int A = (int)ent_values.size(1);
+  const int64_t total = (E + R) * A;
+  if (total == 0) return;
+  dim3 grid((unsigned)((total + 255) / 256));
+  hipLaunchKernelGGL(summary_loglik_kernel, grid, dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(),
+                     ent_values.data_ptr<int32_t>(), rec_values.data_ptr<int32_t>(),
+                     rec_dist.data_ptr<uint8_t>(), rec_ent.data_ptr<int64_t>(),
+                     log_phi.data_ptr<float>(), log_norm.data_ptr<float>(),
+                     voff.data_ptr<int64_t>(), csr_row_ptr.data_ptr<int64_t>(),
+                     csr_col.data_ptr<int32_t>(), csr_sim.data_ptr<float>(),
+                     attr_const.data_ptr<uint8_t>(), E, R, A,
+                     out.data_ptr<double>());
+}
+
+void kd_descent(
+    torch::Tensor ent_values, torch::Tensor node_kind, torch::Tensor node_attr,
+    torch::Tensor node_a, torch::Tensor node_b, torch::Tensor rset,
+    torch::Tensor ent_part_out) {
+  const int64_t E = ent_values.size(0);
+  const int A = (int)ent_values.size(1);
+  if (E == 0) return;
+  dim3 grid((unsigned)((E + 255) / 256));
+  hipLaunchKernelGGL(kd_descent_kernel, grid, dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(),
+                     ent_values.data_ptr<int32_t>(), node_kind.data_ptr<int32_t>(),
+                     node_attr.data_ptr<int32_t>(), node_a.data_ptr<int32_t>(),
+                     node_b.data_ptr<int32_t>(), rset.data_ptr<int32_t>(), E, A,
+                     ent_part_out.data_ptr<int32_t>());
+}
+
+std::vector<torch::Tensor> sim_pairs_gpu(torch::Tensor strs, torch::Tensor lens,
+                                         double threshold, double max_sim) {
+  CHECK_GPU(strs);
+  const int V = (int)strs.size(0);
+  const int max_len = (int)strs.size(1);
+  TORCH_CHECK(max_len <= 64, "attribute values longer than 64 bytes unsupported");
+  auto opts_i64 = torch::TensorOptions().dtype(torch::kInt64).device(strs.device());
+  auto row_counts = torch::zeros({V}, opts_i64);
+  const int64_t total = (int64_t)V * V;
+  dim3 grid((unsigned)((total + 255) / 256));
+  hipLaunchKernelGGL(sim_pairs_kernel, grid, dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(),
+                     strs.data_ptr<uint8_t>(), lens.data_ptr<int32_t>(), V, max_len,
+                     (float)threshold, (float)max_sim,
+                     row_counts.data_ptr<int64_t>(), nullptr, nullptr, nullptr, 0);
+  auto row_ptr = torch::zeros({V + 1}, opts_i64);
+  row_ptr.slice(0, 1, V + 1) = torch::cumsum(row_counts, 0);
+  auto row_start = row_ptr.slice(0, 0, V).contiguous();
+  const int64_t nnz = row_ptr[V].item<int64_t>();
+  auto col = torch::empty({nnz}, torch::TensorOptions().dtype(torch::kInt32).device(strs.device()));
+  auto expsim = torch::empty({nnz}, torch::TensorOptions().dtype(torch::kFloat32).device(strs.device()));
+  auto fill_pos = torch::zeros({V}, opts_i64);
+  hipLaunchKernelGGL(sim_pairs_kernel, grid, dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(),
+                     strs.data_ptr<uint8_t>(), lens.data_ptr<int32_t>(), V, max_len,
+                     (float)threshold, (float)max_sim,
+                     row_start.data_ptr<int64_t>(), col.data_ptr<int32_t>(),
+                     expsim.data_ptr<float>(), fill_pos.data_ptr<int64_t>(), 1);
+  return {row_ptr, col, expsim};
+}
+
+}  // namespace dblink

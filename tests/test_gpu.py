@@ -1,0 +1,367 @@
+"""GPU kernel tests (run on MI355X via gpurun).
+
+Numerics strategy: the Gibbs kernels SAMPLE from discrete distributions, so
+parity with the CPU fp64 reference is checked distributionally — many
+independent replicas of one configuration in a single launch (each replica
+draws with its own Philox counters), empirical frequencies compared against
+exact fp64 probabilities computed in numpy. Deterministic kernels
+(kd_descent, summary_loglik, sim_pairs_gpu) are checked exactly / to fp32
+tolerance.
+"""
+
+import math
+
+import numpy as np
+import pytest
+
+torch = pytest.importorskip("torch")
+
+gpu = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from dblink_amd import ops
+
+    C = ops.native()
+    DEV = torch.device("cuda", 0)
+
+
+def _dev(arr, dtype):
+    return torch.as_tensor(np.ascontiguousarray(arr)).to(dtype).to(DEV)
+
+
+def make_model(device, values_weights=None, threshold=5.0, max_sim=10.0, Kc=10,
+               priors=(0.5, 50.0)):
+    """Small two-attribute model (1 const, 1 Levenshtein) on the GPU."""
+    from dblink_amd.engine.gpu_engine import GpuModel
+    from dblink_amd.models.attribute_index import _python_sim_pairs
+    from dblink_amd.models.records import Attribute, BetaShapeParameters, RecordsCache, RecordsTable
+    from dblink_amd.models.similarity import ConstantSimilarityFn, LevenshteinSimilarityFn
+
+    rng = np.random.default_rng(0)
+    names = ["ANNA", "ANNE", "ANNAH", "BOB", "BORB", "CLAIRE", "CLAIR", "DAVE"]
+    years = [str(y) for y in range(1950, 1960)]
+    rows = []
+    for i in range(400):
+        rows.append([str(rng.choice(years)), str(rng.choice(names))])
+    table = RecordsTable([str(i) for i in range(400)], ["0"] * 400, rows)
+    prior = BetaShapeParameters(*priors)
+    attrs = [
+        Attribute("year", ConstantSimilarityFn(), prior),
+        Attribute("name", LevenshteinSimilarityFn(threshold, max_sim), prior),
+    ]
+    cache = RecordsCache.build(table, attrs, max_cluster_size=Kc,
+                               pair_sweep=_python_sim_pairs)
+    model = GpuModel(cache, device, Kc)
+    return cache, model
+
+
+def tv_distance(emp, exact):
+    return 0.5 * np.abs(emp - exact).sum()
+
+
+@gpu
+def test_sim_pairs_gpu_matches_cpu():
+    from dblink_amd.models.attribute_index import _python_sim_pairs
+    from dblink_amd.models.similarity import LevenshteinSimilarityFn
+
+    fn = LevenshteinSimilarityFn(5.0, 10.0)
+    values = sorted({
+        "Australian Capital Territory", "New South Wales", "Northern Territory",
+        "Queensland", "South Australia", "Tasmania", "Victoria", "Western Australia",
+        "ANNA", "ANNE", "ANNAH", "BOB", "BORB", "",
+    })
+    ref = _python_sim_pairs(values, fn)
+    enc = [v.encode() for v in values]
+    lens = np.array([len(e) for e in enc], dtype=np.int32)
+    buf = np.zeros((len(enc), 64), dtype=np.uint8)
+    for i, e in enumerate(enc):
+        buf[i, : len(e)] = np.frombuffer(e, dtype=np.uint8)
+    row_ptr, col, expsim = C.sim_pairs_gpu(_dev(buf, torch.uint8), _dev(lens, torch.int32),
+                                           5.0, 10.0)
+    row_ptr = row_ptr.cpu().numpy()
+    np.testing.assert_array_equal(row_ptr, ref.row_ptr)
+    # rows may be filled in any order -> compare as sets per row
+    col = col.cpu().numpy()
+    expsim = expsim.cpu().numpy()
+    for v in range(len(values)):
+        lo, hi = row_ptr[v], row_ptr[v + 1]
+        got = dict(zip(col[lo:hi].tolist(), expsim[lo:hi].tolist()))
+        want = dict(zip(ref.col[ref.row_ptr[v]:ref.row_ptr[v+1]].tolist(),
+                        ref.expsim[ref.row_ptr[v]:ref.row_ptr[v+1]].tolist()))
+        assert set(got) == set(want)
+        for k in want:
+            assert got[k] == pytest.approx(want[k], rel=1e-5)
+
+
+@gpu
+def test_kd_descent_matches_cpu():
+    from dblink_amd.parallel.partitioning import KDTreePartitioner
+
+    rng = np.random.default_rng(2)
+    vals = rng.integers(0, 25, size=(4000, 3)).astype(np.int32)
+    p = KDTreePartitioner(3, [0, 1, 2]).fit(vals)
+    flat = p.as_flat()
+    out = torch.empty(4000, dtype=torch.int32, device=DEV)
+    rset = flat["rset"] if flat["rset"].size else np.zeros(1, np.int32)
+    C.kd_descent(_dev(vals, torch.int32), _dev(flat["kind"], torch.int32),
+                 _dev(flat["attr"], torch.int32), _dev(flat["a"], torch.int32),
+                 _dev(flat["b"], torch.int32), _dev(rset, torch.int32), out)
+    np.testing.assert_array_equal(out.cpu().numpy(), p.get_partition_ids(vals))
+
+
+@gpu
+def test_distortion_kernel_distribution():
+    """Empirical distortion rates vs exact Bernoulli probabilities
+    (GibbsUpdates.scala:324-359)."""
+    cache, model = make_model(DEV)
+    a_name = 1  # Levenshtein attribute
+    ia = cache.indexed_attributes[a_name]
+    N = 40000
+    theta = np.array([[0.05], [0.12]])
+    model.theta.copy_(torch.from_numpy(theta).float())
+
+    # three cases per record: missing(attr0), agree(attr1), disagree via values
+    x_name = 2  # some value id
+    rec_values = np.tile(np.array([[-1, x_name]], dtype=np.int32), (N, 1))
+    ent_values = np.tile(np.array([[0, x_name]], dtype=np.int32), (N, 1))
+    rec_ent = np.arange(N, dtype=np.int64)
+    rec_dist = np.zeros((N, 2), dtype=np.uint8)
+    C.distortion_update(
+        _dev(rec_values, torch.int32), (d := _dev(rec_dist, torch.uint8)),
+        _dev(np.zeros(N, np.int32), torch.int32),
+        _dev(np.arange(N, dtype=np.int64), torch.int64),
+        _dev(rec_ent, torch.int64), _dev(ent_values, torch.int32),
+        model.theta, model.phi, model.norm_lin, model.self_expsim, model.voff,
+        model.attr_const, 1234, 7,
+    )
+    z = d.cpu().numpy()
+    # attr 0 missing: P(z=1) = theta
+    assert z[:, 0].mean() == pytest.approx(0.05, abs=0.01)
+    # attr 1 agree: p = pr1/(pr0+pr1)
+    px = ia.index.probability_of(x_name)
+    pr1 = 0.12 * px * ia.index.sim_norms[x_name] * ia.index.exp_sim_of(x_name, x_name)
+    pr0 = 1 - 0.12
+    assert z[:, 1].mean() == pytest.approx(pr1 / (pr0 + pr1), abs=0.01)
+
+    # disagree: always distorted
+    ent_values[:, 1] = x_name + 1
+    C.distortion_update(
+        _dev(rec_values, torch.int32), (d := _dev(rec_dist, torch.uint8)),
+        _dev(np.zeros(N, np.int32), torch.int32),
+        _dev(np.arange(N, dtype=np.int64), torch.int64),
+        _dev(rec_ent, torch.int64), _dev(ent_values, torch.int32),
+        model.theta, model.phi, model.norm_lin, model.self_expsim, model.voff,
+        model.attr_const, 99, 3,
+    )
+    assert d.cpu().numpy()[:, 1].min() == 1
+
+
+@gpu
+def test_link_kernel_distribution():
+    """PCG-I link update: N identical records in one partition, empirical
+    entity-selection frequencies vs exact weights (GibbsUpdates.scala:398-430)."""
+    cache, model = make_model(DEV)
+    ia = cache.indexed_attributes[1]
+    idx = ia.index
+    E = 12
+    # entities: attr0 value all = 3 (so nd-attr0 matches everyone); attr1 varied
+    ent_vals = np.zeros((E, 2), dtype=np.int32)
+    ent_vals[:, 0] = 3
+    ent_vals[:, 1] = (np.arange(E) % idx.num_values).astype(np.int32)
+    x = 0  # record's name value
+    N = 30000
+    rec_values = np.tile(np.array([[3, x]], dtype=np.int32), (N, 1))
+    rec_dist = np.tile(np.array([[0, 1]], dtype=np.uint8), (N, 1))  # name distorted
+
+    # exact weights over candidates = all entities (single nd set = everyone)
+    w = np.array([
+        idx.sim_norms[y] * idx.exp_sim_of(x, int(y)) * idx.probability_of(x)
+        for y in ent_vals[:, 1]
+    ])
+    exact = w / w.sum()
+
+    # postings: key layout (part*A + a)*Vmax + v, a-major flatten like the engine
+    A, Vmax = 2, model.Vmax
+    keys = ((0 * A + np.repeat([0, 1], E)) * Vmax
+            + ent_vals.T.reshape(-1)).astype(np.int64)
+    order = np.argsort(keys, kind="stable")
+    sorted_keys = keys[order]
+    postings = (order % E).astype(np.int32)
+    qk = np.array([(0 * A + 0) * Vmax + 3, (0 * A + 1) * Vmax + x], dtype=np.int64)
+    lo = np.searchsorted(sorted_keys, qk, "left")
+    hi = np.searchsorted(sorted_keys, qk, "right")
+    cand_lo = np.tile(lo, (N, 1)).astype(np.int64)
+    cand_hi = np.tile(hi, (N, 1)).astype(np.int64)
+
+    out = torch.empty(N, dtype=torch.int64, device=DEV)
+    err = torch.zeros(1, dtype=torch.int32, device=DEV)
+    C.link_update(
+        _dev(rec_values, torch.int32), _dev(rec_dist, torch.uint8),
+        _dev(np.arange(N, dtype=np.int64), torch.int64),
+        _dev(np.zeros(N, np.int32), torch.int32),
+        _dev(cand_lo, torch.int64), _dev(cand_hi, torch.int64),
+        _dev(postings, torch.int32), _dev(ent_vals, torch.int32),
+        _dev(np.array([0, E], dtype=np.int64), torch.int64),
+        model.log_norm, model.voff, model.csr_row_ptr, model.csr_col, model.csr_sim,
+        model.attr_const, 4321, 11, out,
+        _dev(np.zeros(N, np.int64), torch.int64), err,
+    )
+    assert int(err.cpu()) == 0
+    sel = out.cpu().numpy()
+    emp = np.bincount(sel, minlength=E) / N
+    assert tv_distance(emp, exact) < 0.02, (emp, exact)
+
+
+@gpu
+def test_value_kernel_distribution():
+    """Collapsed value update: E identical single-record clusters; empirical
+    value frequencies vs the exact mixture P(v) = (base(v) + w_v) / (1 + W)
+    (GibbsUpdates.scala:576-599)."""
+    cache, model = make_model(DEV)
+    a = 1
+    ia = cache.indexed_attributes[a]
+    idx = ia.index
+    V = idx.num_values
+    x = 1
+    theta = np.array([[0.05], [0.08]])
+    model.theta.copy_(torch.from_numpy(theta).float())
+    E = 40000
+    # entity e linked to exactly one record with name value x (observed)
+    rec_values = np.tile(np.array([[3, x]], dtype=np.int32), (E, 1))
+    rec_dist = np.ones((E, 2), dtype=np.uint8)
+    ent_vals = np.zeros((E, 2), dtype=np.int32)
+    ent_rec_ptr = np.arange(E + 1, dtype=np.int64)
+    ent_rec_idx = np.arange(E, dtype=np.int64)
+
+    # exact distribution (k=1): base = phi*norm/Z1; pert per reference
+    base = idx.probs * idx.sim_norms / idx.sim_norm_total(1)
+    th = 0.08
+    px = idx.probability_of(x)
+    normx = idx.sim_norms[x]
+    cols, sims = idx.sim_index.row(x)
+    vw = {}
+    for v, es in zip(cols.tolist(), sims.tolist()):
+        f = es + (1 / th - 1) / (px * normx) if v == x else es
+        vw[v] = f
+    w = np.zeros(V)
+    for v, f in vw.items():
+        w[v] = base[v] * (f - 1.0)
+    W = w.sum()
+    exact = (base + w) / (1.0 + W)
+
+    ev = _dev(ent_vals, torch.int32)
+    err = torch.zeros(1, dtype=torch.int32, device=DEV)
+    C.value_update(
+        _dev(rec_values, torch.int32), _dev(rec_dist, torch.uint8),
+        _dev(np.zeros(E, np.int32), torch.int32),
+        _dev(ent_rec_ptr, torch.int64), _dev(ent_rec_idx, torch.int64),
+        ev, model.theta, model.phi, model.log_phi, model.norm_lin, model.log_norm,
+        model.voff, model.csr_row_ptr, model.csr_col, model.csr_sim,
+        model.phi_prob, model.phi_alias, model.pow_prob, model.pow_alias,
+        model.pow_off, model.log_pow_total, model.attr_const, model.Kc,
+        1, 0, 777, 5, 0, err,
+    )
+    got = ev.cpu().numpy()[:, a]
+    emp = np.bincount(got, minlength=V) / E
+    assert tv_distance(emp, exact) < 0.02, (emp[:10], exact[:10])
+
+
+@gpu
+def test_summary_loglik_matches_cpu():
+    from dblink_amd.engine.cpu_engine import compute_summary
+    from dblink_amd.engine.state import ChainState
+    from dblink_amd.models.distortion import DistortionProbs
+
+    cache, model = make_model(DEV)
+    rng = np.random.default_rng(5)
+    E, R, A = 300, 500, 2
+    Vs = [cache.indexed_attributes[a].index.num_values for a in range(A)]
+    ent_values = np.stack([rng.integers(0, Vs[a], E) for a in range(A)], 1).astype(np.int32)
+    rec_values = np.stack([rng.integers(-1, Vs[a], R) for a in range(A)], 1).astype(np.int32)
+    rec_ent = rng.integers(0, E, R).astype(np.int64)
+    rec_dist = (rng.random((R, A)) < 0.3).astype(np.uint8)
+    state = ChainState(
+        iteration=0, ent_values=ent_values, ent_part=np.zeros(E, np.int32),
+        rec_values=rec_values, rec_file=np.zeros(R, np.int32), rec_ent=rec_ent,
+        rec_dist=rec_dist, rec_gid=np.arange(R, dtype=np.int64),
+        dist_probs=DistortionProbs(np.full((A, 1), 0.05)), population_size=E,
+        start_seed=0, current_seed=0,
+    )
+    ref = compute_summary(state, cache, state.dist_probs)
+
+    out = torch.zeros(1, dtype=torch.float64, device=DEV)
+    C.summary_loglik(
+        _dev(ent_values, torch.int32), _dev(rec_values, torch.int32),
+        _dev(rec_dist, torch.uint8), _dev(rec_ent, torch.int64),
+        model.log_phi, model.log_norm, model.voff, model.csr_row_ptr,
+        model.csr_col, model.csr_sim, model.attr_const, out,
+    )
+    assert float(out.cpu()) == pytest.approx(ref.log_likelihood, rel=1e-5)
+
+
+@gpu
+def test_gpu_end_to_end_accuracy(tmp_path):
+    """Full GPU chain on synthetic RLdata-shaped data; F1/ARI oracle."""
+    import bench as b
+    from dblink_amd.analysis import chain as chain_q
+    from dblink_amd.analysis import metrics as metrics_m
+    from dblink_amd.engine import sampler as sampler_m
+    from dblink_amd.engine.cpu_engine import SamplerFlags
+    from dblink_amd.engine.gpu_engine import GpuEngine
+    from dblink_amd.engine.init import deterministic_init
+    from dblink_amd.parallel.partitioning import KDTreePartitioner
+    from dblink_amd.utils.synthdata import generate
+
+    n = 300
+    cols, header = generate(n, dup_fraction=0.1, seed=42)
+    cache, rec_values, rec_files = b.build_cache_and_records(n, seed=42)
+    partitioner = KDTreePartitioner(1, [3])
+    state = deterministic_init(rec_values, rec_files, np.arange(n, dtype=np.int64),
+                               cache, partitioner, seed=319158)
+    engine = GpuEngine(cache, partitioner, device=DEV)
+    engine.rec_id_of = lambda gid: str(gid + 1)  # rec_id = row + 1 in synthdata
+    engine.initial_summary(state)
+    out = str(tmp_path)
+    sampler_m.sample(engine, state, sample_size=100, output_path=out,
+                     burnin_interval=100, thinning_interval=4,
+                     checkpoint_interval=0, flags=SamplerFlags.for_sampler("PCG-I"))
+    table = chain_q.load_chain(out, 150)
+    smpc = chain_q.shared_most_probable_clusters(table)
+    truth = metrics_m.membership_to_clusters(
+        {cols["rec_id"][i]: cols["ent_id"][i] for i in range(n)}
+    )
+    pm = metrics_m.PairwiseMetrics.compute(smpc, truth)
+    ari = metrics_m.adjusted_rand_index(smpc, truth)
+    assert pm.f1score > 0.55, (pm.precision, pm.recall)
+    assert ari > 0.55
+
+
+@gpu
+def test_gpu_vs_cpu_posterior_band(tmp_path):
+    """GPU and CPU engines must land in the same posterior band on the same
+    data (log-likelihood trajectories agree within a few percent)."""
+    import bench as b
+    from dblink_amd.engine.cpu_engine import CpuEngine, SamplerFlags
+    from dblink_amd.engine.gpu_engine import GpuEngine
+    from dblink_amd.engine.init import deterministic_init
+    from dblink_amd.parallel.partitioning import KDTreePartitioner
+
+    n = 200
+    cache, rec_values, rec_files = b.build_cache_and_records(n, seed=9)
+    flags = SamplerFlags.for_sampler("PCG-I")
+
+    lls = {}
+    for kind in ("cpu", "gpu"):
+        partitioner = KDTreePartitioner(0, [])
+        state = deterministic_init(rec_values, rec_files, np.arange(n, dtype=np.int64),
+                                   cache, partitioner, seed=1)
+        eng = (CpuEngine(cache, partitioner) if kind == "cpu"
+               else GpuEngine(cache, partitioner, device=DEV))
+        eng.initial_summary(state)
+        tail = []
+        for i in range(60):
+            eng.step(state, flags)
+            if i >= 40:
+                tail.append(state.summary.log_likelihood)
+        lls[kind] = np.mean(tail)
+    assert abs(lls["cpu"] - lls["gpu"]) / abs(lls["cpu"]) < 0.03, lls
