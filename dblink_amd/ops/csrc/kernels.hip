@@ -336,6 +336,8 @@ value_update_kernel(ValueArgs args) {
   int32_t* keys = h_key[wave];
   float* vals = h_val[wave];
   for (int i = lane; i < HASH_CAP; i += WAVE) { keys[i] = -1; vals[i] = 0.0f; }
+  // drain LDS writes before other lanes' atomics may touch the slots
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
   __builtin_amdgcn_wave_barrier();
 
   // dense fallback decision: total row entries vs capacity
@@ -420,6 +422,9 @@ value_update_kernel(ValueArgs args) {
       }
       __builtin_amdgcn_wave_barrier();
     }
+    // drain all LDS atomics before cross-lane reads of the table
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_wave_barrier();
     // transform + reduce + gumbel-max over occupied slots
     for (int i = lane; i < HASH_CAP; i += WAVE) {
       const int32_t v = keys[i];
