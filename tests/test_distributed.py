@@ -1,0 +1,180 @@
+"""Multi-process (gloo, world_size=2) tests of the distributed path:
+migration all-to-all, summary all-reduce, and a short end-to-end chain.
+
+These exercise the SAME code path the 8-GPU RCCL run uses (comm.py switches
+backend only), per the reference's pseudocluster testing strategy
+(SURVEY.md §4)."""
+
+import json
+import os
+import subprocess
+import sys
+
+import numpy as np
+import pytest
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+WORKER = r"""
+import json, os, sys
+sys.path.insert(0, "__ROOT__")
+import numpy as np
+import torch.distributed as dist
+
+from dblink_amd.parallel import comm
+from dblink_amd.parallel.partitioning import KDTreePartitioner
+from dblink_amd.engine.init import deterministic_init
+from dblink_amd.engine.cpu_engine import CpuEngine, SamplerFlags
+from bench import build_cache_and_records
+
+rank, world, device = comm.init_from_env(backend="gloo")
+assert world == 2, world
+
+n = 240
+cache, rec_values, rec_files = build_cache_and_records(n, seed=11)
+partitioner = KDTreePartitioner(2, [3, 4])
+bounds = np.linspace(0, n, world + 1).astype(np.int64)
+lo, hi = int(bounds[rank]), int(bounds[rank + 1])
+state = deterministic_init(rec_values[lo:hi], rec_files[lo:hi],
+                           np.arange(lo, hi, dtype=np.int64), cache, partitioner,
+                           seed=5, rank=rank, world_size=world)
+
+# ownership invariant after init migration: pid % world == rank
+assert np.all(state.ent_part % world == rank), "entity placed on wrong rank"
+
+engine = CpuEngine(cache, partitioner, world_size=world, rank=rank)
+engine.initial_summary(state)
+flags = SamplerFlags.for_sampler("PCG-I")
+
+lls = []
+for i in range(25):
+    engine.step(state, flags)
+    assert np.all(state.ent_part % world == rank)
+    # rec_ent indices valid
+    assert state.rec_ent.min() >= 0 and state.rec_ent.max() < state.num_entities
+    lls.append(state.summary.log_likelihood)
+
+# conservation: global entity and record counts unchanged
+import torch
+t = torch.tensor([state.num_entities, state.num_records], dtype=torch.float64)
+comm.all_reduce_sum_(t)
+assert int(t[0]) == state.population_size, (int(t[0]), state.population_size)
+assert int(t[1]) == n
+
+# record gids form a partition of 0..n-1 globally
+gids = comm.all_gather_object(sorted(state.rec_gid.tolist()))
+if rank == 0:
+    allg = sorted(g for lst in gids for g in lst)
+    assert allg == list(range(n)), "records lost or duplicated in migration"
+    print(json.dumps({"ok": True, "ll": lls[-1], "isolates": state.summary.num_isolates}))
+dist.destroy_process_group()
+"""
+
+
+def _run_workers(script, world=2, timeout=600):
+    env = dict(os.environ)
+    env.update(
+        MASTER_ADDR="127.0.0.1",
+        MASTER_PORT=str(29500 + os.getpid() % 1000),
+        WORLD_SIZE=str(world),
+        GLOO_SOCKET_IFNAME="lo",
+    )
+    procs = []
+    for r in range(world):
+        e = dict(env, RANK=str(r), LOCAL_RANK=str(r))
+        procs.append(
+            subprocess.Popen(
+                [sys.executable, "-c", script], env=e,
+                stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+            )
+        )
+    outs = []
+    for p in procs:
+        out, _ = p.communicate(timeout=timeout)
+        outs.append(out)
+    for p, out in zip(procs, outs):
+        assert p.returncode == 0, f"worker failed:\n{out}"
+    return outs
+
+
+@pytest.mark.slow
+def test_two_rank_chain_conservation():
+    outs = _run_workers(WORKER.replace("__ROOT__", ROOT))
+    payload = None
+    for out in outs:
+        for line in out.splitlines():
+            if line.startswith("{"):
+                payload = json.loads(line)
+    assert payload and payload["ok"]
+    assert np.isfinite(payload["ll"])
+
+
+SINGLE_VS_TWO = r"""
+import json, os, sys
+sys.path.insert(0, "__ROOT__")
+import numpy as np
+import torch.distributed as dist
+
+from dblink_amd.parallel import comm
+from dblink_amd.parallel.partitioning import KDTreePartitioner
+from dblink_amd.engine.init import deterministic_init
+from dblink_amd.engine.cpu_engine import CpuEngine, SamplerFlags
+from bench import build_cache_and_records
+
+rank, world, device = comm.init_from_env(backend="gloo")
+n = 160
+cache, rec_values, rec_files = build_cache_and_records(n, seed=3)
+partitioner = KDTreePartitioner(1, [3])
+bounds = np.linspace(0, n, world + 1).astype(np.int64)
+lo, hi = int(bounds[rank]), int(bounds[rank + 1])
+state = deterministic_init(rec_values[lo:hi], rec_files[lo:hi],
+                           np.arange(lo, hi, dtype=np.int64), cache, partitioner,
+                           seed=2, rank=rank, world_size=world)
+engine = CpuEngine(cache, partitioner, world_size=world, rank=rank)
+engine.initial_summary(state)
+flags = SamplerFlags.for_sampler("PCG-I")
+tail = []
+for i in range(60):
+    engine.step(state, flags)
+    if i >= 40:
+        tail.append(state.summary.log_likelihood)
+if rank == 0:
+    print(json.dumps({"mean_ll": float(np.mean(tail))}))
+dist.destroy_process_group()
+"""
+
+
+@pytest.mark.slow
+def test_two_rank_posterior_matches_single_rank():
+    """The 2-rank chain must land in the same stationary log-likelihood band
+    as the 1-rank chain on identical data (same model, different RNG streams)."""
+    # single rank, in-process
+    sys.path.insert(0, ROOT)
+    from bench import build_cache_and_records
+    from dblink_amd.engine.cpu_engine import CpuEngine, SamplerFlags
+    from dblink_amd.engine.init import deterministic_init
+    from dblink_amd.parallel.partitioning import KDTreePartitioner
+
+    n = 160
+    cache, rec_values, rec_files = build_cache_and_records(n, seed=3)
+    partitioner = KDTreePartitioner(1, [3])
+    state = deterministic_init(rec_values, rec_files, np.arange(n, dtype=np.int64),
+                               cache, partitioner, seed=2)
+    engine = CpuEngine(cache, partitioner)
+    engine.initial_summary(state)
+    flags = SamplerFlags.for_sampler("PCG-I")
+    tail = []
+    for i in range(60):
+        engine.step(state, flags)
+        if i >= 40:
+            tail.append(state.summary.log_likelihood)
+    single = float(np.mean(tail))
+
+    outs = _run_workers(SINGLE_VS_TWO.replace("__ROOT__", ROOT))
+    two = None
+    for out in outs:
+        for line in out.splitlines():
+            if line.startswith("{"):
+                two = json.loads(line)["mean_ll"]
+    assert two is not None
+    assert abs(single - two) / abs(single) < 0.05, (single, two)
