@@ -311,58 +311,9 @@ class GpuEngine(CpuEngine):
     # ---- helpers -------------------------------------------------------------
 
     def _migrate_and_sort(self, gs: GpuStateTensors):
-        world = self.world_size
-        if world > 1 and comm.is_distributed():
-            dest_e = gs.ent_part.to(torch.int64) % world
-            order_e = torch.argsort(dest_e, stable=True)
-            inv_e = torch.empty_like(order_e)
-            inv_e[order_e] = torch.arange(order_e.numel(), device=self.device)
-            send_e = torch.bincount(dest_e, minlength=world)
-            dest_r = dest_e[gs.rec_ent]
-            order_r = torch.argsort(dest_r, stable=True)
-            send_r = torch.bincount(dest_r, minlength=world)
-            base = torch.cumsum(
-                torch.cat([torch.zeros(1, dtype=torch.int64, device=self.device), send_e[:-1]]), 0
-            )
-            pos_in_dest = inv_e - base[dest_e]
+        from ..parallel.migration import migrate_and_sort_tensors
 
-            send_e_l = [int(x) for x in send_e.cpu()]
-            send_r_l = [int(x) for x in send_r.cpu()]
-
-            new_ev, recv_e = comm.all_to_all_v(gs.ent_values[order_e].contiguous(), send_e_l)
-            new_ep, _ = comm.all_to_all_v(gs.ent_part[order_e].contiguous(), send_e_l)
-            rel = pos_in_dest[gs.rec_ent][order_r].contiguous()
-            new_rel, recv_r = comm.all_to_all_v(rel, send_r_l)
-            new_rv, _ = comm.all_to_all_v(gs.rec_values[order_r].contiguous(), send_r_l)
-            new_rf, _ = comm.all_to_all_v(gs.rec_file[order_r].contiguous(), send_r_l)
-            new_rd, _ = comm.all_to_all_v(gs.rec_dist[order_r].contiguous(), send_r_l)
-            new_rg, _ = comm.all_to_all_v(gs.rec_gid[order_r].contiguous(), send_r_l)
-
-            eoff = np.concatenate([[0], np.cumsum(recv_e)])
-            roff = np.concatenate([[0], np.cumsum(recv_r)])
-            new_re = torch.empty(int(roff[-1]), dtype=torch.int64, device=self.device)
-            for s in range(world):
-                new_re[int(roff[s]) : int(roff[s + 1])] = (
-                    new_rel[int(roff[s]) : int(roff[s + 1])] + int(eoff[s])
-                )
-            gs.ent_values, gs.ent_part = new_ev, new_ep
-            gs.rec_values, gs.rec_file, gs.rec_dist, gs.rec_gid = new_rv, new_rf, new_rd, new_rg
-            gs.rec_ent = new_re
-
-        # local re-sort by partition id (stable)
-        order = torch.argsort(gs.ent_part.to(torch.int64), stable=True)
-        inv = torch.empty_like(order)
-        inv[order] = torch.arange(order.numel(), device=self.device)
-        gs.ent_values = gs.ent_values[order].contiguous()
-        gs.ent_part = gs.ent_part[order].contiguous()
-        new_rec_ent = inv[gs.rec_ent]
-        rorder = torch.argsort(new_rec_ent, stable=True)
-        gs.rec_ent = new_rec_ent[rorder].contiguous()
-        gs.rec_values = gs.rec_values[rorder].contiguous()
-        gs.rec_file = gs.rec_file[rorder].contiguous()
-        gs.rec_dist = gs.rec_dist[rorder].contiguous()
-        gs.rec_gid = gs.rec_gid[rorder].contiguous()
-        gs.rec_part = gs.ent_part[gs.rec_ent].contiguous()
+        migrate_and_sort_tensors(gs, self.world_size)
 
     def _summary_device(self, gs: GpuStateTensors, state: ChainState) -> SummaryVars:
         m = self.model

@@ -178,3 +178,62 @@ def test_two_rank_posterior_matches_single_rank():
                 two = json.loads(line)["mean_ll"]
     assert two is not None
     assert abs(single - two) / abs(single) < 0.05, (single, two)
+
+
+TENSOR_MIGRATE = r"""
+import json, os, sys
+sys.path.insert(0, "__ROOT__")
+import numpy as np
+import torch
+import torch.distributed as dist
+from types import SimpleNamespace
+
+from dblink_amd.parallel import comm
+from dblink_amd.parallel.migration import migrate_and_sort_tensors
+
+rank, world, device = comm.init_from_env(backend="gloo")
+rng = np.random.default_rng(100 + rank)
+E, R, A, P = 50, 80, 3, 8
+gs = SimpleNamespace(
+    ent_values=torch.tensor(rng.integers(0, 9, (E, A)), dtype=torch.int32),
+    ent_part=torch.tensor(rng.integers(0, P, E), dtype=torch.int32),
+    rec_values=torch.tensor(rng.integers(-1, 9, (R, A)), dtype=torch.int32),
+    rec_file=torch.zeros(R, dtype=torch.int32),
+    rec_dist=torch.tensor(rng.integers(0, 2, (R, A)), dtype=torch.uint8),
+    rec_gid=torch.tensor(rank * R + np.arange(R), dtype=torch.int64),
+    rec_ent=torch.tensor(rng.integers(0, E, R), dtype=torch.int64),
+    rec_part=None,
+)
+# remember each record's entity VALUES so we can verify links survive the move
+before = {int(g): gs.ent_values[gs.rec_ent[i]].tolist() for i, g in enumerate(gs.rec_gid)}
+allbefore = comm.all_gather_object(before)
+merged = {}
+for d in allbefore:
+    merged.update(d)
+
+migrate_and_sort_tensors(gs, world)
+
+# ownership + sortedness invariants
+assert torch.all(gs.ent_part.to(torch.int64) % world == rank)
+assert torch.all(gs.ent_part[1:] >= gs.ent_part[:-1])
+assert torch.all(gs.rec_ent[1:] >= gs.rec_ent[:-1])
+assert torch.all(gs.rec_part == gs.ent_part[gs.rec_ent])
+# every record still points at an entity with its original values
+for i in range(gs.rec_gid.numel()):
+    g = int(gs.rec_gid[i])
+    assert gs.ent_values[gs.rec_ent[i]].tolist() == merged[g], g
+counts = comm.all_gather_object((gs.ent_values.shape[0], gs.rec_gid.numel()))
+if rank == 0:
+    assert sum(c[0] for c in counts) == world * E
+    assert sum(c[1] for c in counts) == world * R
+    print(json.dumps({"ok": True}))
+dist.destroy_process_group()
+"""
+
+
+@pytest.mark.slow
+def test_tensor_migration_preserves_links():
+    """migrate_and_sort_tensors (the GPU engine's migration) must preserve
+    every record->entity link across the all-to-all (gloo, CPU tensors)."""
+    outs = _run_workers(TENSOR_MIGRATE.replace("__ROOT__", ROOT))
+    assert any('"ok": true' in o for o in outs), outs
