@@ -62,11 +62,33 @@ def sim_pairs(values, similarity_fn):
             buf[i, : len(e)] = np.frombuffer(e, dtype=np.uint8)
         import torch
 
+        thr = float(similarity_fn.threshold)
+        msim = float(similarity_fn.max_similarity)
+        # Large domains: run the V x V sweep on the GPU (the reference's Spark
+        # cartesian analog, AttributeIndex.scala:222-231) — the quadratic pair
+        # filter is the startup bottleneck at V ~ 10^5.
+        if torch.cuda.is_available() and len(values) > 8192 and maxlen <= 64:
+            dev_buf = torch.from_numpy(buf).cuda()
+            if dev_buf.shape[1] < 64:
+                dev_buf = torch.nn.functional.pad(dev_buf, (0, 64 - dev_buf.shape[1]))
+            row_ptr, col, expsim = _C.sim_pairs_gpu(
+                dev_buf.contiguous(), torch.from_numpy(lens).cuda(), thr, msim
+            )
+            # atomically-filled rows are unordered; sort within rows by col
+            V = len(values)
+            counts = row_ptr[1:] - row_ptr[:-1]
+            row_of = torch.repeat_interleave(
+                torch.arange(V, device=col.device, dtype=torch.int64), counts
+            )
+            keys = row_of * V + col.to(torch.int64)
+            order = torch.argsort(keys)
+            return SimIndexCSR(
+                row_ptr.cpu().numpy(),
+                col[order].cpu().numpy(),
+                expsim[order].cpu().numpy().astype(np.float64),
+            )
         row_ptr, col, expsim = _C.sim_pairs_cpu(
-            torch.from_numpy(buf),
-            torch.from_numpy(lens),
-            float(similarity_fn.threshold),
-            float(similarity_fn.max_similarity),
+            torch.from_numpy(buf), torch.from_numpy(lens), thr, msim
         )
         return SimIndexCSR(row_ptr.numpy().astype(np.int64), col.numpy(), expsim.numpy())
 
