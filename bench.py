@@ -46,11 +46,8 @@ def build_cache_and_records(total_records, seed):
 
     cols, header = generate(total_records, dup_fraction=0.1, seed=seed)
     attr_names = ["by", "bm", "bd", "fname_c1", "lname_c1"]
-    values = [
-        [cols[a][i] if cols[a][i] != "NA" else None for a in attr_names]
-        for i in range(total_records)
-    ]
-    table = RecordsTable(cols["rec_id"], ["0"] * total_records, values)
+    columns = [np.where(cols[a] == "NA", None, cols[a]) for a in attr_names]
+    table = RecordsTable(cols["rec_id"], cols["file_id"], columns)
     prior = BetaShapeParameters(10.0, 1000.0)  # RLdata10000.conf:4
     attrs = [
         Attribute("by", ConstantSimilarityFn(), prior),

@@ -63,26 +63,22 @@ def deterministic_init(
 
     rng = np.random.Generator(np.random.Philox(key=seed + rank))
 
-    ent_values = np.full((E, A), -1, dtype=np.int32)
-    created = np.zeros(E, dtype=bool)
-    rec_ent = np.empty(R, dtype=np.int64)
-    rec_dist = np.zeros((R, A), dtype=np.uint8)
-    for i in range(R):
-        e = i % E
-        rec_ent[i] = e
-        if not created[e]:
-            for a in range(A):
-                v = rec_values[i, a]
-                if v >= 0:
-                    ent_values[e, a] = v
-                else:
-                    ent_values[e, a] = cache.indexed_attributes[a].index.draw(rng)
-            created[e] = True
-        rec_dist[i] = (rec_values[i] >= 0) & (rec_values[i] != ent_values[e])
-    for e in range(E):
-        if not created[e]:  # isolated entity: fully random values
-            for a in range(A):
-                ent_values[e, a] = cache.indexed_attributes[a].index.draw(rng)
+    # Entity e is created by its first linked record i = e (record i links to
+    # entity i % E, State.scala:269-281); vectorized over entities.
+    rec_ent = (np.arange(R, dtype=np.int64) % E) if E > 0 else np.empty(0, np.int64)
+    k = min(E, R)
+    ent_values = np.empty((E, A), dtype=np.int32)
+    ent_values[:k] = rec_values[:k]
+    for a in range(A):
+        dist = cache.indexed_attributes[a].index.distribution
+        miss = np.flatnonzero(ent_values[:k, a] < 0)
+        if miss.size:
+            ent_values[miss, a] = dist.sample(rng, miss.size)
+        if E > k:  # isolated entities: fully random values
+            ent_values[k:, a] = dist.sample(rng, E - k)
+    rec_dist = (
+        (rec_values >= 0) & (rec_values != ent_values[rec_ent])
+    ).astype(np.uint8)
 
     new_seed = seed + world_size
 

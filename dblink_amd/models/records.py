@@ -8,8 +8,9 @@ Capability parity with the reference:
   string -> value-id record transform; missing values encode as -1)
 - CSV loading with header + nullValue handling: ``Project.scala:173-180``
 
-The MI355X build holds records as dense numpy/torch int32 matrices
-(records x attributes) instead of an RDD of case classes.
+The MI355X build is columnar: records are numpy object columns, counted and
+dictionary-encoded with pandas factorize (C-speed hashing) so 10M-record
+datasets encode in seconds, then held as dense int32 matrices.
 """
 
 from __future__ import annotations
@@ -47,23 +48,37 @@ class Attribute:
 
 
 class RecordsTable:
-    """Raw records: string ids, file ids, string attribute values (None = missing)."""
+    """Raw records, columnar: string ids, file ids, and one object column per
+    matching attribute (None = missing)."""
 
-    def __init__(self, rec_ids, file_ids, values):
-        self.rec_ids = list(rec_ids)  # list[str]
-        self.file_ids = list(file_ids)  # list[str]
-        self.values = values  # list[list[str | None]] (num_records x num_attrs)
+    def __init__(self, rec_ids, file_ids, columns):
+        self.rec_ids = np.asarray(rec_ids, dtype=object)
+        self.file_ids = np.asarray(file_ids, dtype=object)
+        self.columns = [np.asarray(c, dtype=object) for c in columns]
+
+    @classmethod
+    def from_rows(cls, rec_ids, file_ids, rows):
+        """rows: list of per-record value lists (None = missing)."""
+        if rows:
+            cols = [np.array([r[a] for r in rows], dtype=object) for a in range(len(rows[0]))]
+        else:
+            cols = []
+        return cls(rec_ids, file_ids, cols)
 
     @property
     def num_records(self):
         return len(self.rec_ids)
 
+    @property
+    def num_attributes(self):
+        return len(self.columns)
+
 
 def load_csv(path, rec_id_col, file_id_col, attribute_names, null_value="NA", ent_id_col=None):
     """Load one or more CSV files into a RecordsTable.
 
-    ``path`` may be a single file or a comma-separated list of files. When
-    ``file_id_col`` is None, the file id of every record is "0"
+    ``path`` may be a single file or a comma-separated list of files / globs.
+    When ``file_id_col`` is None, the file id of every record is "0"
     (``State.scala:359-375``). Malformed rows (wrong column count) are dropped,
     matching Spark's DROPMALFORMED mode (``Project.scala:177``).
 
@@ -78,7 +93,7 @@ def load_csv(path, rec_id_col, file_id_col, attribute_names, null_value="NA", en
         expanded = sorted(glob.glob(p)) if any(ch in p for ch in "*?[") else [p]
         paths.extend(expanded)
 
-    rec_ids, file_ids, values, ent_ids = [], [], [], []
+    rec_ids, file_ids, rows, ent_ids = [], [], [], []
     for p in paths:
         with open(p, "r", encoding="utf-8", newline="") as f:
             reader = csv.reader(f)
@@ -97,10 +112,12 @@ def load_csv(path, rec_id_col, file_id_col, attribute_names, null_value="NA", en
                     continue  # DROPMALFORMED
                 rec_ids.append(row[rid_i])
                 file_ids.append(row[fid_i] if fid_i is not None else "0")
-                values.append([None if row[i] == null_value or row[i] == "" else row[i] for i in attr_is])
+                rows.append(
+                    [None if row[i] == null_value or row[i] == "" else row[i] for i in attr_is]
+                )
                 if eid_i is not None:
                     ent_ids.append(row[eid_i])
-    table = RecordsTable(rec_ids, file_ids, values)
+    table = RecordsTable.from_rows(rec_ids, file_ids, rows)
     return table, (ent_ids if ent_ids else None)
 
 
@@ -108,7 +125,7 @@ class RecordsCache:
     """Per-attribute domain indexes plus file-level statistics."""
 
     def __init__(self, indexed_attributes, file_sizes, missing_counts=None):
-        self.indexed_attributes = list(indexed_attributes)  # list[IndexedAttribute-like]
+        self.indexed_attributes = list(indexed_attributes)
         self.file_sizes = dict(file_sizes)  # {file_id: count}
         self.missing_counts = missing_counts or {}
         self.file_ids = sorted(self.file_sizes)  # stable order; id -> dense int
@@ -131,27 +148,24 @@ class RecordsCache:
 
     @classmethod
     def build(cls, table: RecordsTable, attributes, max_cluster_size: int, pair_sweep=None):
-        """One pass over the records for counts, then per-attribute index build
+        """Columnar counting pass, then per-attribute index build
         (``RecordsCache.scala:68-118``)."""
-        A = len(attributes)
-        file_sizes = {}
-        missing_counts = {}
-        value_counts = [dict() for _ in range(A)]
-        for fid, vals in zip(table.file_ids, table.values):
-            file_sizes[fid] = file_sizes.get(fid, 0) + 1
-            for a in range(A):
-                v = vals[a]
-                if v is None:
-                    key = (fid, a)
-                    missing_counts[key] = missing_counts.get(key, 0) + 1
-                else:
-                    vc = value_counts[a]
-                    vc[v] = vc.get(v, 0) + 1
+        import pandas as pd
 
+        file_ser = pd.Series(table.file_ids)
+        file_sizes = file_ser.value_counts().to_dict()
+
+        missing_counts = {}
         indexed = []
         for a, spec in enumerate(attributes):
+            ser = pd.Series(table.columns[a])
+            miss = ser.isna()
+            if miss.any():
+                for fid, cnt in file_ser[miss].value_counts().items():
+                    missing_counts[(fid, a)] = int(cnt)
+            counts = ser.value_counts()
             index = AttributeIndex(
-                {k: float(v) for k, v in value_counts[a].items()},
+                {k: float(v) for k, v in counts.items()},
                 spec.similarity_fn,
                 precache_powers=max_cluster_size,
                 pair_sweep=pair_sweep,
@@ -162,17 +176,25 @@ class RecordsCache:
     def transform_records(self, table: RecordsTable):
         """Encode string values to dense int32 value ids; missing -> -1
         (``RecordsCache.scala:120-134``). Also returns dense int32 file ids."""
+        import pandas as pd
+
         R, A = table.num_records, self.num_attributes
         out = np.full((R, A), -1, dtype=np.int32)
         for a in range(A):
+            codes, uniques = pd.factorize(pd.Series(table.columns[a]))
             sid = self.indexed_attributes[a].index._string_to_id
-            col = out[:, a]
-            for r, vals in enumerate(table.values):
-                v = vals[a]
-                if v is not None:
-                    col[r] = sid.get(v, -1)
-        files = np.array([self.file_id_to_int[f] for f in table.file_ids], dtype=np.int32)
-        return out, files
+            mapping = np.fromiter(
+                (sid.get(u, -1) for u in uniques), dtype=np.int32, count=len(uniques)
+            )
+            mapping = np.concatenate([mapping, [-1]])  # codes == -1 (missing)
+            out[:, a] = mapping[codes]
+        file_codes, file_uniques = pd.factorize(pd.Series(table.file_ids))
+        fmap = np.fromiter(
+            (self.file_id_to_int[u] for u in file_uniques), dtype=np.int32,
+            count=len(file_uniques),
+        )
+        files = fmap[file_codes]
+        return out, files.astype(np.int32)
 
 
 @dataclass

@@ -1,4 +1,4 @@
-"""Synthetic RLdata-shaped dataset generator.
+"""Synthetic RLdata-shaped dataset generator (vectorized).
 
 The reference ships RLdata500 / RLdata10000 (German name + birth-date fields,
 10% duplicates, distorted duplicates; ``README.md:25-34``). There is no
@@ -7,32 +7,46 @@ benchmarks and end-to-end tests generate datasets of the same SHAPE:
 
 - columns: fname_c1, lname_c1 (strings, Levenshtein-matched),
   by, bm, bd (birth year/month/day, constant-sim categorical)
-- `dup_fraction` of the records are duplicates of an earlier record with a
+- ``dup_fraction`` of the records are duplicates of an earlier record with a
   small number of attribute distortions (typo edits for names, redraws for
   dates)
-- ground-truth entity ids are included (`ent_id`) for evaluation
+- ground-truth entity ids are included (``ent_id``) for evaluation
 
-Deterministic given `seed`.
+Deterministic given ``seed``. Vectorized so 10M-record datasets build in
+seconds (the per-duplicate typo loop touches only the ~10% duplicates).
 """
 
 from __future__ import annotations
 
 import numpy as np
 
-# Frequency-skewed synthetic name pools (generated, not copied from RLdata).
+# Syllable pools for generated names (not copied from RLdata).
 _FIRST_SYLL = ["an", "be", "ca", "da", "el", "fi", "ga", "han", "in", "jo", "ka", "lu",
                "ma", "ni", "ol", "pe", "re", "sa", "to", "ul", "vi", "wa", "chris", "ste"]
 _LAST_SYLL = ["bach", "berg", "mann", "stein", "feld", "hof", "meier", "schmid", "mueller",
               "weber", "wagner", "becker", "koch", "richter", "wolf", "schroe", "neu", "lang"]
 
 
-def _make_pool(rng, syllables, n, min_parts=2, max_parts=3):
+def _make_pool(rng, syllables, n, min_parts=3):
+    """n distinct names; the part count grows until the combinatorial space
+    comfortably covers n (fixes the stall when n > |syllables|^3)."""
+    syl = np.array(syllables, dtype=object)
+    s = len(syllables)
+    max_parts = min_parts
+    cap = s ** max_parts
+    while cap < 4 * n:
+        max_parts += 1
+        cap *= s
     pool = set()
     while len(pool) < n:
-        k = rng.integers(min_parts, max_parts + 1)
-        name = "".join(rng.choice(syllables) for _ in range(k))
-        pool.add(name.upper())
-    return sorted(pool)
+        batch = max(8192, 2 * (n - len(pool)))
+        ks = rng.integers(min_parts, max_parts + 1, size=batch)
+        idx = rng.integers(0, s, size=(batch, max_parts))
+        for b in range(batch):
+            pool.add("".join(syl[idx[b, : ks[b]]]).upper())
+            if len(pool) >= n:
+                break
+    return np.array(sorted(pool), dtype=object)
 
 
 def _typo(rng, s):
@@ -61,11 +75,7 @@ def generate(
     num_files: int = 1,
     extra_string_attrs: int = 0,
 ):
-    """Return (columns dict, header list) for an RLdata-shaped dataset.
-
-    Entities get distinct ids; each duplicate record distorts 1-2 attributes
-    of its entity's true values.
-    """
+    """Return (columns dict of numpy object/str arrays, header list)."""
     rng = np.random.default_rng(seed)
     n_ent = int(round(num_records * (1.0 - dup_fraction)))
     n_dup = num_records - n_ent
@@ -88,52 +98,61 @@ def generate(
     ent_by = rng.integers(1900, 2000, size=n_ent)
     ent_bm = rng.integers(1, 13, size=n_ent)
     ent_bd = rng.integers(1, 29, size=n_ent)
-    extra = [
-        rng.choice(fpool, size=n_ent, p=fw) for _ in range(extra_string_attrs)
-    ]
+    extra = [rng.choice(fpool, size=n_ent, p=fw) for _ in range(extra_string_attrs)]
 
-    rows_ent = list(range(n_ent)) + list(rng.integers(0, n_ent, size=n_dup))
+    rows_ent = np.concatenate([np.arange(n_ent), rng.integers(0, n_ent, size=n_dup)])
     order = rng.permutation(num_records)
+    ent_of_row = rows_ent[order]
+    is_dup = order >= n_ent
 
-    cols = {k: [] for k in ["rec_id", "file_id", "ent_id", "fname_c1", "lname_c1", "by", "bm", "bd"]}
+    n_attr = 5 + extra_string_attrs
+    fn = ent_fname[ent_of_row].copy()
+    ln = ent_lname[ent_of_row].copy()
+    by = ent_by[ent_of_row].copy()
+    bm = ent_bm[ent_of_row].copy()
+    bd = ent_bd[ent_of_row].copy()
+    xv = [e[ent_of_row].copy() for e in extra]
+
+    # distort 1-2 attributes of each duplicate (loop touches only duplicates)
+    dup_rows = np.flatnonzero(is_dup)
+    n_dist = rng.integers(1, 3, size=dup_rows.size)
+    for row, nd in zip(dup_rows, n_dist):
+        targets = rng.choice(n_attr, size=nd, replace=False)
+        for t in targets:
+            if t == 0:
+                fn[row] = _typo(rng, fn[row])
+            elif t == 1:
+                ln[row] = _typo(rng, ln[row])
+            elif t == 2:
+                by[row] = rng.integers(1900, 2000)
+            elif t == 3:
+                bm[row] = rng.integers(1, 13)
+            elif t == 4:
+                bd[row] = rng.integers(1, 29)
+            else:
+                xv[t - 5][row] = _typo(rng, xv[t - 5][row])
+
+    cols = {
+        "rec_id": np.arange(1, num_records + 1).astype(str),
+        "file_id": (np.arange(num_records) % num_files).astype(str),
+        "ent_id": (ent_of_row + 1).astype(str),
+        "fname_c1": fn.astype(object),
+        "lname_c1": ln.astype(object),
+        "by": by.astype(str).astype(object),
+        "bm": bm.astype(str).astype(object),
+        "bd": bd.astype(str).astype(object),
+    }
     for j in range(extra_string_attrs):
-        cols[f"xattr{j}"] = []
+        cols[f"xattr{j}"] = xv[j].astype(object)
 
-    for out_i, idx in enumerate(order):
-        e = rows_ent[idx]
-        is_dup = idx >= n_ent
-        fn, ln = str(ent_fname[e]), str(ent_lname[e])
-        by, bm, bd = int(ent_by[e]), int(ent_bm[e]), int(ent_bd[e])
-        xvals = [str(extra[j][e]) for j in range(extra_string_attrs)]
-        if is_dup:
-            # distort 1-2 attributes
-            n_dist = int(rng.integers(1, 3))
-            targets = rng.choice(5 + extra_string_attrs, size=n_dist, replace=False)
-            for t in targets:
-                if t == 0:
-                    fn = _typo(rng, fn)
-                elif t == 1:
-                    ln = _typo(rng, ln)
-                elif t == 2:
-                    by = int(rng.integers(1900, 2000))
-                elif t == 3:
-                    bm = int(rng.integers(1, 13))
-                elif t == 4:
-                    bd = int(rng.integers(1, 29))
-                else:
-                    xvals[t - 5] = _typo(rng, xvals[t - 5])
-        vals = {"fname_c1": fn, "lname_c1": ln, "by": str(by), "bm": str(bm), "bd": str(bd)}
-        for j in range(extra_string_attrs):
-            vals[f"xattr{j}"] = xvals[j]
-        if missing_fraction > 0:
-            for k in list(vals):
-                if rng.random() < missing_fraction:
-                    vals[k] = "NA"
-        cols["rec_id"].append(str(out_i + 1))
-        cols["file_id"].append(str(out_i % num_files))
-        cols["ent_id"].append(str(e + 1))
-        for k, v in vals.items():
-            cols[k].append(v)
+    if missing_fraction > 0:
+        attr_cols = ["fname_c1", "lname_c1", "by", "bm", "bd"] + [
+            f"xattr{j}" for j in range(extra_string_attrs)
+        ]
+        for k in attr_cols:
+            mask = rng.random(num_records) < missing_fraction
+            c = cols[k]
+            c[mask] = "NA"
 
     header = ["rec_id", "file_id", "ent_id", "fname_c1", "lname_c1", "by", "bm", "bd"] + [
         f"xattr{j}" for j in range(extra_string_attrs)
@@ -146,5 +165,5 @@ def write_csv(path, num_records, **kw):
     with open(path, "w", encoding="utf-8") as f:
         f.write(",".join(header) + "\n")
         for i in range(num_records):
-            f.write(",".join(cols[h][i] for h in header) + "\n")
+            f.write(",".join(str(cols[h][i]) for h in header) + "\n")
     return path
