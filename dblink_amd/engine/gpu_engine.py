@@ -272,6 +272,16 @@ class GpuEngine(CpuEngine):
         ent_rec_idx = order
 
         # --- phase 2: value update (in place) --------------------------------
+        # split (entity, attribute) pairs: k_obs == 0 -> thread-per-pair base
+        # draw; k_obs >= 1 -> one wave each (perturbation sampling)
+        obs = (gs.rec_values >= 0)
+        kobs = torch.zeros(E * A, dtype=torch.int32, device=self.device)
+        pair_idx = (gs.rec_ent.view(R, 1) * A
+                    + torch.arange(A, device=self.device, dtype=torch.int64).view(1, A))
+        idxm = pair_idx.reshape(-1)[obs.reshape(-1)]
+        kobs.scatter_add_(0, idxm, torch.ones_like(idxm, dtype=torch.int32))
+        wave_pairs = torch.nonzero(kobs > 0).squeeze(1)
+        base_pairs = torch.nonzero(kobs == 0).squeeze(1)
         self.C.value_update(
             gs.rec_values, gs.rec_dist, gs.rec_file, ent_rec_ptr, ent_rec_idx,
             gs.ent_values, m.theta, m.phi, m.log_phi, m.norm_lin, m.log_norm,
@@ -279,6 +289,7 @@ class GpuEngine(CpuEngine):
             m.pow_prob, m.pow_alias, m.pow_off, m.log_pow_total, m.attr_const,
             m.Kc, 1 if flags.collapsed_entity_values else 0,
             1 if flags.sequential else 0, seed, it, self._ent_id_base, self._err,
+            wave_pairs, base_pairs,
         )
 
         # --- phase 3: distortion update --------------------------------------

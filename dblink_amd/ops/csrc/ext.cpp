@@ -40,7 +40,8 @@ void value_update(torch::Tensor rec_values, torch::Tensor rec_dist,
                   torch::Tensor pow_alias, torch::Tensor pow_off,
                   torch::Tensor log_pow_total, torch::Tensor attr_const, int64_t Kc,
                   int64_t collapsed, int64_t sequential, int64_t seed,
-                  int64_t iteration, int64_t ent_id_base, torch::Tensor error_count);
+                  int64_t iteration, int64_t ent_id_base, torch::Tensor error_count,
+                  torch::Tensor wave_pairs, torch::Tensor base_pairs);
 void distortion_update(torch::Tensor rec_values, torch::Tensor rec_dist,
                        torch::Tensor rec_file, torch::Tensor rec_gid,
                        torch::Tensor rec_ent, torch::Tensor ent_values,

@@ -215,6 +215,8 @@ constexpr int HASH_CAP = 1024;            // slots per wave
 constexpr int WAVES_PER_BLOCK_VAL = 4;    // 256 threads
 
 struct ValueArgs {
+  const int64_t* pair_list;    // [n_pairs] flattened (e*A + a) needing wave work
+  int64_t n_pairs;
   const int32_t* rec_values;
   const uint8_t* rec_dist;
   const int32_t* rec_file;
@@ -268,6 +270,23 @@ __device__ int dense_power_draw(const ValueArgs& args, int a, int k, uint64_t el
   return (int)best_v;
 }
 
+// Thread-per-pair base draws for empty clusters (k_obs == 0): the base
+// distribution is phi for every variant (GibbsUpdates.scala:584-588).
+__global__ void value_base_draw_kernel(ValueArgs args) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= args.n_pairs) return;
+  const int64_t pair = args.pair_list[i];
+  const int64_t e = pair / args.A;
+  const int a = (int)(pair % args.A);
+  const int64_t v0 = args.voff[a];
+  const int V = (int)(args.voff[a + 1] - v0);
+  const uint64_t elem = (args.ent_id_base + (uint64_t)e) * 32u + (uint64_t)a;
+  float u1, u2;
+  philox_uniform2(args.seed, args.iteration, PH_VALM, elem, 0xFFFF0000u, &u1, &u2);
+  const int v = alias_draw(args.phi_prob + v0, args.phi_alias + v0, V, u1, u2);
+  args.ent_values[e * args.A + a] = (int32_t)v;
+}
+
 __global__ void __launch_bounds__(WAVES_PER_BLOCK_VAL * WAVE)
 value_update_kernel(ValueArgs args) {
   __shared__ int32_t h_key[WAVES_PER_BLOCK_VAL][HASH_CAP];
@@ -275,9 +294,9 @@ value_update_kernel(ValueArgs args) {
 
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave = threadIdx.x / WAVE;
-  const int64_t pair = (int64_t)blockIdx.x * WAVES_PER_BLOCK_VAL + wave;
-  const int64_t total_pairs = args.E * args.A;
-  if (pair >= total_pairs) return;
+  const int64_t pidx = (int64_t)blockIdx.x * WAVES_PER_BLOCK_VAL + wave;
+  if (pidx >= args.n_pairs) return;
+  const int64_t pair = args.pair_list[pidx];
   const int64_t e = pair / args.A;
   const int a = (int)(pair % args.A);
   const bool is_const = args.attr_const[a];
@@ -288,11 +307,13 @@ value_update_kernel(ValueArgs args) {
   // ---- gather observed linked records -------------------------------------
   const int64_t r_lo = args.ent_rec_ptr[e], r_hi = args.ent_rec_ptr[e + 1];
   int k_obs = 0;
+  int64_t first_obs_r = -1;
   int32_t first_nondist = -1;  // first observed non-distorted value (record order)
   for (int64_t i = r_lo; i < r_hi; ++i) {
     const int64_t r = args.ent_rec_idx[i];
     const int32_t x = args.rec_values[r * args.A + a];
     if (x < 0) continue;
+    if (k_obs == 0) first_obs_r = r;
     ++k_obs;
     if (!args.collapsed && first_nondist < 0 && !args.rec_dist[r * args.A + a])
       first_nondist = x;
@@ -331,16 +352,84 @@ value_update_kernel(ValueArgs args) {
     return;
   }
 
-  // ---- perturbation weights in the LDS hash table -------------------------
+  // log of normalized base probability of value v (local id); log Z for the
+  // rare k > Kc case is computed on demand by a wave reduction.
+  float log_z_rare = 0.0f;
+  if (!is_const && k_obs > args.Kc) {
+    double tot = 0.0;
+    for (int64_t v = v0 + lane; v < v0 + V; v += WAVE)
+      tot += exp((double)(args.log_phi[v] + (float)k_obs * args.log_norm[v]));
+    log_z_rare = (float)log(wave_sum(tot));
+  }
+  const float log_z =
+      is_const ? 0.0f
+               : (k_obs <= args.Kc ? args.log_pow_total[a * (args.Kc + 1) + k_obs]
+                                   : log_z_rare);
+  auto log_base_prob = [&](int v_local) -> float {
+    const int64_t v = v0 + v_local;
+    if (is_const) return args.log_phi[v];
+    return args.log_phi[v] + (float)k_obs * args.log_norm[v] - log_z;
+  };
+  auto self_extra_of = [&](int64_t r, int32_t x) -> float {
+    if (!args.collapsed) return 0.0f;
+    const float th = args.theta[a * args.F + args.rec_file[r]];
+    const float px = args.phi[v0 + x];
+    return (1.0f / th - 1.0f) / (is_const ? px : px * args.norm_lin[v0 + x]);
+  };
+  auto finish = [&](double W, float best, long long best_v) {
+    const float u = philox_uniform(args.seed, args.iteration, PH_VALM, elem, 0xFFFF0001u);
+    int v_new;
+    if ((double)u < 1.0 / (1.0 + W) || best_v < 0) {
+      v_new = base_draw(0xFFFF0002u);
+    } else {
+      v_new = (int)best_v;
+    }
+    if (lane == 0) args.ent_values[e * args.A + a] = (int32_t)v_new;
+  };
+
+  // ---- k == 1 fast path: the perturbation support is one sim row ----------
+  if (k_obs == 1) {
+    const int64_t r = first_obs_r;
+    const int32_t x = args.rec_values[r * args.A + a];
+    const float se = self_extra_of(r, x);
+    double W = 0.0;
+    float best = -INFINITY;
+    long long best_v = -1;
+    if (is_const) {  // single support value {x}, factor = 1 + se
+      if (lane == 0 && se > 0.0f) {
+        const float L = __logf(1.0f + se);
+        const float logw = log_base_prob(x) + L + __logf(1.0f - __expf(-L));
+        W = exp((double)logw);
+        best = 0.0f;
+        best_v = x;
+      }
+      W = wave_sum(W);
+      best_v = __shfl(best_v, 0);
+    } else {
+      const int64_t row_lo = args.csr_row_ptr[v0 + x], row_hi = args.csr_row_ptr[v0 + x + 1];
+      for (int64_t jj = row_lo + lane; jj < row_hi; jj += WAVE) {
+        const int32_t v = args.csr_col[jj];
+        const float s = args.csr_sim[jj];
+        const float L = (v == x && se > 0.0f) ? __logf(__expf(s) + se) : s;
+        const float logw = log_base_prob(v) + L + __logf(1.0f - __expf(-L));
+        W += exp((double)logw);
+        const float g = gumbel_from_uniform(
+            philox_uniform(args.seed, args.iteration, PH_VALG, elem, (uint32_t)v));
+        if (logw + g > best) { best = logw + g; best_v = v; }
+      }
+      W = wave_sum(W);
+      wave_argmax(best, best_v);
+    }
+    finish(W, best, best_v);
+    return;
+  }
+
+  // ---- perturbation weights in the LDS hash table (k >= 2) ---------------
   // L_v = sum_r log f_r(v); final weight = base_prob(v) * (exp(L_v) - 1).
   int32_t* keys = h_key[wave];
   float* vals = h_val[wave];
-  for (int i = lane; i < HASH_CAP; i += WAVE) { keys[i] = -1; vals[i] = 0.0f; }
-  // drain LDS writes before other lanes' atomics may touch the slots
-  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-  __builtin_amdgcn_wave_barrier();
 
-  // dense fallback decision: total row entries vs capacity
+  // size the table (and its clear/scan cost) to the actual support
   int64_t total_entries = 0;
   if (!is_const) {
     for (int64_t i = r_lo; i < r_hi; ++i) {
@@ -353,55 +442,33 @@ value_update_kernel(ValueArgs args) {
     total_entries = k_obs;
   }
   const bool dense = total_entries > (HASH_CAP * 3) / 4;
-
-  // log of normalized base probability of value v (local id)
-  const float log_z = (is_const || k_obs == 0)
-                          ? 0.0f
-                          : (k_obs <= args.Kc
-                                 ? args.log_pow_total[a * (args.Kc + 1) + k_obs]
-                                 : 0.0f);  // patched below for the rare path
-  float log_z_rare = 0.0f;
-  if (!is_const && k_obs > args.Kc) {
-    // need log Z_k for base_prob; compute by wave reduction
-    double tot = 0.0;
-    for (int64_t v = v0 + lane; v < v0 + V; v += WAVE)
-      tot += exp((double)(args.log_phi[v] + (float)k_obs * args.log_norm[v]));
-    log_z_rare = (float)log(wave_sum(tot));
-  }
-  auto log_base_prob = [&](int v_local) -> float {
-    const int64_t v = v0 + v_local;
-    if (is_const) return args.log_phi[v];
-    const float lz = (k_obs <= args.Kc) ? log_z : log_z_rare;
-    return args.log_phi[v] + (float)k_obs * args.log_norm[v] - lz;
-  };
+  int tsize = 64;
+  while (tsize < 2 * (int)total_entries && tsize < HASH_CAP) tsize <<= 1;
 
   double W = 0.0;            // total perturbation weight
   float best = -INFINITY;    // gumbel-max over perturbation weights
   long long best_v = -1;
 
   if (!dense) {
+    for (int i = lane; i < tsize; i += WAVE) { keys[i] = -1; vals[i] = 0.0f; }
+    // drain LDS writes before other lanes' atomics may touch the slots
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_wave_barrier();
     // hash-accumulate log factors
     for (int64_t i = r_lo; i < r_hi; ++i) {
       const int64_t r = args.ent_rec_idx[i];
       const int32_t x = args.rec_values[r * args.A + a];
       if (x < 0) continue;
-      float self_extra = 0.0f;
-      if (args.collapsed) {
-        const float th = args.theta[a * args.F + args.rec_file[r]];
-        const float px = args.phi[v0 + x];
-        self_extra = (1.0f / th - 1.0f) /
-                     (is_const ? px : px * args.norm_lin[v0 + x]);
-      }
+      const float self_extra = self_extra_of(r, x);
       if (is_const) {
         // single-entry row {x}: factor = 1 + self_extra
         if (lane == 0) {
           float logf_ = __logf(1.0f + self_extra);
-          // linear-probe insert
-          uint32_t h = ((uint32_t)x * 2654435761u) & (HASH_CAP - 1);
+          uint32_t h = ((uint32_t)x * 2654435761u) & (tsize - 1);
           while (true) {
             int32_t prev = atomicCAS(&keys[h], -1, x);
             if (prev == -1 || prev == x) { atomicAdd(&vals[h], logf_); break; }
-            h = (h + 1) & (HASH_CAP - 1);
+            h = (h + 1) & (tsize - 1);
           }
         }
       } else {
@@ -412,11 +479,11 @@ value_update_kernel(ValueArgs args) {
           float factor_log = (v == x && self_extra > 0.0f)
                                  ? __logf(__expf(s) + self_extra)
                                  : s;
-          uint32_t h = ((uint32_t)v * 2654435761u) & (HASH_CAP - 1);
+          uint32_t h = ((uint32_t)v * 2654435761u) & (tsize - 1);
           while (true) {
             int32_t prev = atomicCAS(&keys[h], -1, v);
             if (prev == -1 || prev == v) { atomicAdd(&vals[h], factor_log); break; }
-            h = (h + 1) & (HASH_CAP - 1);
+            h = (h + 1) & (tsize - 1);
           }
         }
       }
@@ -426,7 +493,7 @@ value_update_kernel(ValueArgs args) {
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_wave_barrier();
     // transform + reduce + gumbel-max over occupied slots
-    for (int i = lane; i < HASH_CAP; i += WAVE) {
+    for (int i = lane; i < tsize; i += WAVE) {
       const int32_t v = keys[i];
       if (v < 0) continue;
       const float L = vals[i];
@@ -447,13 +514,7 @@ value_update_kernel(ValueArgs args) {
         const int64_t r = args.ent_rec_idx[i];
         const int32_t x = args.rec_values[r * args.A + a];
         if (x < 0) continue;
-        float self_extra = 0.0f;
-        if (args.collapsed) {
-          const float th = args.theta[a * args.F + args.rec_file[r]];
-          const float px = args.phi[v0 + x];
-          self_extra = (1.0f / th - 1.0f) /
-                       (is_const ? px : px * args.norm_lin[v0 + x]);
-        }
+        const float self_extra = self_extra_of(r, x);
         if (is_const) {
           if (v_local == x) L += __logf(1.0f + self_extra);
         } else {
@@ -479,14 +540,7 @@ value_update_kernel(ValueArgs args) {
   wave_argmax(best, best_v);
 
   // mixture between base and perturbation (GibbsUpdates.scala:593-597)
-  const float u = philox_uniform(args.seed, args.iteration, PH_VALM, elem, 0xFFFF0001u);
-  int v_new;
-  if ((double)u < 1.0 / (1.0 + W) || best_v < 0) {
-    v_new = base_draw(0xFFFF0002u);
-  } else {
-    v_new = (int)best_v;
-  }
-  if (lane == 0) args.ent_values[e * args.A + a] = (int32_t)v_new;
+  finish(W, best, best_v);
 }
 
 // Brute-force dense value update (Gibbs-Sequential, GibbsUpdates.scala:652-698)
@@ -494,7 +548,7 @@ __global__ void value_update_seq_kernel(ValueArgs args) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave = threadIdx.x / WAVE;
   const int64_t pair = (int64_t)blockIdx.x * (blockDim.x / WAVE) + wave;
-  if (pair >= args.E * args.A) return;
+  if (pair >= args.n_pairs) return;
   const int64_t e = pair / args.A;
   const int a = (int)(pair % args.A);
   const bool is_const = args.attr_const[a];
@@ -660,65 +714,104 @@ __global__ void kd_descent_kernel(
 // K1 (GPU): banded Levenshtein sim-pair sweep over a value domain
 // ---------------------------------------------------------------------------
 
-// One thread per candidate pair (i, j); strings padded to max_len bytes.
-// Band pruning: pairs are pre-filtered by length difference on the host side
-// (sim > 0 requires d <= dmax(Li+Lj)), the kernel re-checks and computes the
-// exact banded DP.
-__device__ float lev_unit_sim(const uint8_t* a, int la, const uint8_t* b, int lb) {
-  // full DP over rows of b with rolling arrays in registers/scratch.
-  // strings are short (attribute values); cap at 64.
-  constexpr int MAXLEN = 64;
-  uint8_t prev[MAXLEN + 1];
-  uint8_t cur[MAXLEN + 1];
-  if (la + lb == 0) return 1.0f;
-  for (int j = 0; j <= lb; ++j) prev[j] = (uint8_t)j;
-  for (int i = 1; i <= la; ++i) {
-    cur[0] = (uint8_t)i;
-    const uint8_t ca = a[i - 1];
-    for (int j = 1; j <= lb; ++j) {
-      uint8_t cost = (ca == b[j - 1]) ? 0 : 1;
-      uint8_t m = prev[j] + 1;
-      uint8_t d = cur[j - 1] + 1;
-      uint8_t s = prev[j - 1] + cost;
-      if (d < m) m = d;
-      if (s < m) m = s;
-      cur[j] = m;
-    }
-    for (int j = 0; j <= lb; ++j) prev[j] = cur[j];
+// One WAVE per value i: the wave builds a Myers bit-parallel Peq table for
+// string i in LDS once, then its 64 lanes sweep candidate values j from the
+// length-bucketed window (sim > 0 bounds |len_i - len_j|). Edit distance via
+// Hyyro's bit-vector recurrence: O(len_j) 64-bit ops per pair instead of the
+// O(len_i * len_j) scratch-array DP — ~25x fewer operations, no scratch.
+constexpr int SP_WAVES = 4;  // waves per block
+
+__device__ int myers_distance(const unsigned long long* Peq, int m,
+                              const uint8_t* b, int lb) {
+  if (m == 0) return lb;
+  int score = m;
+  unsigned long long Pv = ~0ull, Mv = 0ull;
+  const unsigned long long last = 1ull << (m - 1);
+  for (int j = 0; j < lb; ++j) {
+    const unsigned long long Eq = Peq[b[j]];
+    const unsigned long long Xv = Eq | Mv;
+    const unsigned long long Xh = (((Eq & Pv) + Pv) ^ Pv) | Eq;
+    unsigned long long Ph = Mv | ~(Xh | Pv);
+    unsigned long long Mh = Pv & Xh;
+    if (Ph & last) ++score;
+    if (Mh & last) --score;
+    Ph = (Ph << 1) | 1ull;
+    Mh <<= 1;
+    Pv = Mh | ~(Xv | Ph);
+    Mv = Ph & Xv;
   }
-  const float dist = (float)prev[lb];
-  return 1.0f - 2.0f * dist / ((float)(la + lb) + dist);
+  return score;
 }
 
-__global__ void sim_pairs_kernel(
-    const uint8_t* __restrict__ strs,  // [V, max_len]
+__global__ void __launch_bounds__(SP_WAVES * WAVE) sim_pairs_kernel(
+    const uint8_t* __restrict__ strs,  // [V, 64]
     const int32_t* __restrict__ lens,  // [V]
+    const int32_t* __restrict__ len_order,  // [V] value ids sorted by length
+    const int64_t* __restrict__ len_ptr,    // [66] bucket offsets by length
     int V, int max_len, float threshold, float max_sim,
-    int64_t* __restrict__ row_counts,  // [V] (pass 1out) or row offsets (pass 2 in)
+    int64_t* __restrict__ row_counts,  // [V] (pass 1 out) or row offsets (pass 2 in)
     int32_t* __restrict__ out_col,     // pass 2
     float* __restrict__ out_expsim,    // pass 2 (stores exp(sim))
     int64_t* __restrict__ fill_pos,    // [V] atomic cursors (pass 2)
     int fill) {
-  const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (idx >= (int64_t)V * V) return;
-  const int i = (int)(idx / V);
-  const int j = (int)(idx % V);
-  const int la = lens[i], lb = lens[j];
-  // length pruning: unit > thr/max requires d < (la+lb)*(1-u0)/(1+u0),
-  // and d >= |la-lb|.
+  __shared__ unsigned long long peq_block[SP_WAVES][256];
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = threadIdx.x / WAVE;
+  const int i = blockIdx.x * SP_WAVES + wave;
+  if (i >= V) return;
+  unsigned long long* Peq = peq_block[wave];
+  const int la = lens[i];
+  const uint8_t* a = strs + (int64_t)i * max_len;
+  for (int c = lane; c < 256; c += WAVE) Peq[c] = 0ull;
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_wave_barrier();
+  if (lane < la) atomicOr(&Peq[a[lane]], 1ull << lane);
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_wave_barrier();
+
   const float u0 = threshold / max_sim;
-  const float dmax = (float)(la + lb) * (1.0f - u0) / (1.0f + u0);
-  if ((float)abs(la - lb) >= dmax && (la + lb) > 0) return;
-  const float unit = lev_unit_sim(strs + (int64_t)i * max_len, la,
-                                  strs + (int64_t)j * max_len, lb);
-  const float trans = max_sim / (max_sim - threshold) * (max_sim * unit - threshold);
-  if (trans <= 0.0f) return;  // expsim <= 1 filtered (AttributeIndex.scala:226)
-  if (!fill) {
-    atomicAdd((unsigned long long*)&row_counts[i], 1ull);
+  const float scale = max_sim / (max_sim - threshold);
+  const float band = (1.0f - u0) / (1.0f + u0);
+  // length window: |la - lb| < band * (la + lb)
+  int lb_min, lb_max;
+  if (band >= 0.999f) {  // threshold ~ 0: no length pruning possible
+    lb_min = 0;
+    lb_max = max_len;
   } else {
-    const int64_t pos = atomicAdd((unsigned long long*)&fill_pos[i], 1ull) + row_counts[i];
-    out_col[pos] = j;
-    out_expsim[pos] = __expf(trans);
+    lb_min = (int)ceilf((1.0f - band) / (1.0f + band) * (float)la);
+    lb_max = (int)floorf((1.0f + band) / (1.0f - band) * (float)la);
+  }
+  if (la == 0) { lb_min = 0; lb_max = 0; }
+  if (lb_min < 0) lb_min = 0;
+  if (lb_max > max_len) lb_max = max_len;
+  const int64_t j0 = len_ptr[lb_min], j1 = len_ptr[lb_max + 1];
+
+  int64_t n_hits = 0;
+  for (int64_t jj = j0 + lane; jj < j1; jj += WAVE) {
+    const int j = len_order[jj];
+    const int lb = lens[j];
+    const int tot = la + lb;
+    float unit;
+    if (tot == 0) {
+      unit = 1.0f;
+    } else {
+      if ((float)abs(la - lb) >= band * (float)tot) continue;
+      const int d = myers_distance(Peq, la, strs + (int64_t)j * max_len, lb);
+      unit = 1.0f - 2.0f * (float)d / ((float)tot + (float)d);
+    }
+    const float trans = scale * (max_sim * unit - threshold);
+    if (trans <= 0.0f) continue;  // expsim <= 1 filtered (AttributeIndex.scala:226)
+    if (!fill) {
+      ++n_hits;
+    } else {
+      const int64_t pos = atomicAdd((unsigned long long*)&fill_pos[i], 1ull) + row_counts[i];
+      out_col[pos] = j;
+      out_expsim[pos] = __expf(trans);
+    }
+  }
+  if (!fill) {
+    double total = wave_sum((double)n_hits);
+    if (lane == 0) row_counts[i] = (int64_t)(total + 0.5);
   }
 }
 
@@ -840,20 +933,36 @@ void value_update(
     torch::Tensor phi_prob, torch::Tensor phi_alias, torch::Tensor pow_prob,
     torch::Tensor pow_alias, torch::Tensor pow_off, torch::Tensor log_pow_total,
     torch::Tensor attr_const, int64_t Kc, int64_t collapsed, int64_t sequential,
-    int64_t seed, int64_t iteration, int64_t ent_id_base, torch::Tensor error_count) {
+    int64_t seed, int64_t iteration, int64_t ent_id_base, torch::Tensor error_count,
+    torch::Tensor wave_pairs, torch::Tensor base_pairs) {
   ValueArgs args = make_value_args(
       rec_values, rec_dist, rec_file, ent_rec_ptr, ent_rec_idx, ent_values, theta,
       phi, log_phi, norm_lin, log_norm, voff, csr_row_ptr, csr_col, csr_sim,
       phi_prob, phi_alias, pow_prob, pow_alias, pow_off, log_pow_total, attr_const,
       Kc, collapsed, seed, iteration, ent_id_base, error_count);
-  const int64_t pairs = args.E * args.A;
-  if (pairs == 0) return;
   if (sequential) {
+    const int64_t pairs = args.E * args.A;
+    if (pairs == 0) return;
+    args.pair_list = nullptr;
+    args.n_pairs = pairs;
     dim3 grid((unsigned)wave_grid(pairs, 4));
     hipLaunchKernelGGL(value_update_seq_kernel, grid, dim3(4 * WAVE), 0,
                        at::cuda::getCurrentCUDAStream(), args);
-  } else {
-    dim3 grid((unsigned)wave_grid(pairs, WAVES_PER_BLOCK_VAL));
+    return;
+  }
+  // k_obs == 0 pairs: one thread each (base draw is phi for every variant)
+  if (base_pairs.numel() > 0) {
+    args.pair_list = base_pairs.data_ptr<int64_t>();
+    args.n_pairs = base_pairs.numel();
+    dim3 grid((unsigned)((args.n_pairs + 255) / 256));
+    hipLaunchKernelGGL(value_base_draw_kernel, grid, dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(), args);
+  }
+  // k_obs >= 1 pairs: one wave each
+  if (wave_pairs.numel() > 0) {
+    args.pair_list = wave_pairs.data_ptr<int64_t>();
+    args.n_pairs = wave_pairs.numel();
+    dim3 grid((unsigned)wave_grid(args.n_pairs, WAVES_PER_BLOCK_VAL));
     hipLaunchKernelGGL(value_update_kernel, grid, dim3(WAVES_PER_BLOCK_VAL * WAVE), 0,
                        at::cuda::getCurrentCUDAStream(), args);
   }
@@ -927,13 +1036,21 @@ std::vector<torch::Tensor> sim_pairs_gpu(torch::Tensor strs, torch::Tensor lens,
   const int max_len = (int)strs.size(1);
   TORCH_CHECK(max_len <= 64, "attribute values longer than 64 bytes unsupported");
   auto opts_i64 = torch::TensorOptions().dtype(torch::kInt64).device(strs.device());
+  // length buckets: candidates for value i live in a contiguous window of
+  // the length-sorted order
+  auto lens64 = lens.to(torch::kInt64);
+  auto len_order = torch::argsort(lens64, /*stable=*/true).to(torch::kInt32).contiguous();
+  auto sorted_lens = std::get<0>(torch::sort(lens64));
+  auto len_ptr = torch::searchsorted(
+      sorted_lens, torch::arange(66, opts_i64), /*out_int32=*/false, /*right=*/false)
+      .contiguous();
   auto row_counts = torch::zeros({V}, opts_i64);
-  const int64_t total = (int64_t)V * V;
-  dim3 grid((unsigned)((total + 255) / 256));
-  hipLaunchKernelGGL(sim_pairs_kernel, grid, dim3(256), 0,
+  dim3 grid((unsigned)((V + SP_WAVES - 1) / SP_WAVES));
+  hipLaunchKernelGGL(sim_pairs_kernel, grid, dim3(SP_WAVES * WAVE), 0,
                      at::cuda::getCurrentCUDAStream(),
-                     strs.data_ptr<uint8_t>(), lens.data_ptr<int32_t>(), V, max_len,
-                     (float)threshold, (float)max_sim,
+                     strs.data_ptr<uint8_t>(), lens.data_ptr<int32_t>(),
+                     len_order.data_ptr<int32_t>(), len_ptr.data_ptr<int64_t>(),
+                     V, max_len, (float)threshold, (float)max_sim,
                      row_counts.data_ptr<int64_t>(), nullptr, nullptr, nullptr, 0);
   auto row_ptr = torch::zeros({V + 1}, opts_i64);
   row_ptr.slice(0, 1, V + 1) = torch::cumsum(row_counts, 0);
@@ -942,10 +1059,11 @@ std::vector<torch::Tensor> sim_pairs_gpu(torch::Tensor strs, torch::Tensor lens,
   auto col = torch::empty({nnz}, torch::TensorOptions().dtype(torch::kInt32).device(strs.device()));
   auto expsim = torch::empty({nnz}, torch::TensorOptions().dtype(torch::kFloat32).device(strs.device()));
   auto fill_pos = torch::zeros({V}, opts_i64);
-  hipLaunchKernelGGL(sim_pairs_kernel, grid, dim3(256), 0,
+  hipLaunchKernelGGL(sim_pairs_kernel, grid, dim3(SP_WAVES * WAVE), 0,
                      at::cuda::getCurrentCUDAStream(),
-                     strs.data_ptr<uint8_t>(), lens.data_ptr<int32_t>(), V, max_len,
-                     (float)threshold, (float)max_sim,
+                     strs.data_ptr<uint8_t>(), lens.data_ptr<int32_t>(),
+                     len_order.data_ptr<int32_t>(), len_ptr.data_ptr<int64_t>(),
+                     V, max_len, (float)threshold, (float)max_sim,
                      row_start.data_ptr<int64_t>(), col.data_ptr<int32_t>(),
                      expsim.data_ptr<float>(), fill_pos.data_ptr<int64_t>(), 1);
   return {row_ptr, col, expsim};

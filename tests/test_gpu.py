@@ -260,6 +260,8 @@ def test_value_kernel_distribution():
         model.phi_prob, model.phi_alias, model.pow_prob, model.pow_alias,
         model.pow_off, model.log_pow_total, model.attr_const, model.Kc,
         1, 0, 777, 5, 0, err,
+        torch.arange(E * 2, dtype=torch.int64, device=DEV),
+        torch.empty(0, dtype=torch.int64, device=DEV),
     )
     got = ev.cpu().numpy()[:, a]
     emp = np.bincount(got, minlength=V) / E
