@@ -43,7 +43,7 @@ def make_model(device, values_weights=None, threshold=5.0, max_sim=10.0, Kc=10,
     rows = []
     for i in range(400):
         rows.append([str(rng.choice(years)), str(rng.choice(names))])
-    table = RecordsTable([str(i) for i in range(400)], ["0"] * 400, rows)
+    table = RecordsTable.from_rows([str(i) for i in range(400)], ["0"] * 400, rows)
     prior = BetaShapeParameters(*priors)
     attrs = [
         Attribute("year", ConstantSimilarityFn(), prior),
