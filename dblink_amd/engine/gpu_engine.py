@@ -256,11 +256,22 @@ class GpuEngine(CpuEngine):
                 1 if flags.collapsed_entity_ids else 0, seed, it, rec_ent_new,
             )
         else:
+            # records whose smallest candidate list is short run one-per-thread
+            cand_lo = cand_lo.contiguous()
+            cand_hi = cand_hi.contiguous()
+            nd = (gs.rec_values >= 0) & (gs.rec_dist == 0)
+            sizes = torch.where(nd, cand_hi - cand_lo, torch.full_like(cand_hi, 1 << 40))
+            min_sizes = sizes.amin(dim=1)
+            has_nd = nd.any(dim=1)
+            small = has_nd & (min_sizes <= 16)
+            small_list = torch.nonzero(small).squeeze(1)
+            wave_list = torch.nonzero(~small).squeeze(1)
             self.C.link_update(
                 gs.rec_values, gs.rec_dist, gs.rec_gid, gs.rec_part,
-                cand_lo.contiguous(), cand_hi.contiguous(), postings, gs.ent_values,
+                cand_lo, cand_hi, postings, gs.ent_values,
                 ent_ptr, m.log_norm, m.voff, m.csr_row_ptr, m.csr_col, m.csr_sim,
                 m.attr_const, seed, it, rec_ent_new, gs.rec_ent, self._err,
+                small_list, wave_list,
             )
         gs.rec_ent = rec_ent_new
 
@@ -280,8 +291,18 @@ class GpuEngine(CpuEngine):
                     + torch.arange(A, device=self.device, dtype=torch.int64).view(1, A))
         idxm = pair_idx.reshape(-1)[obs.reshape(-1)]
         kobs.scatter_add_(0, idxm, torch.ones_like(idxm, dtype=torch.int32))
-        wave_pairs = torch.nonzero(kobs > 0).squeeze(1)
         base_pairs = torch.nonzero(kobs == 0).squeeze(1)
+        # per-pair perturbation support size (sim-row length, 1 for const)
+        gv = m.voff[:-1].view(1, A) + gs.rec_values.clamp_min(0).to(torch.int64)
+        rowsz = (m.csr_row_ptr[gv + 1] - m.csr_row_ptr[gv]).to(torch.int32)
+        rowsz = torch.where(m.attr_const.view(1, A) > 0,
+                            torch.ones_like(rowsz), rowsz)
+        entries = torch.zeros(E * A, dtype=torch.int32, device=self.device)
+        entries.scatter_add_(0, pair_idx.reshape(-1)[obs.reshape(-1)],
+                             rowsz.reshape(-1)[obs.reshape(-1)])
+        k1 = (kobs == 1) & (entries <= 48)
+        k1_pairs = torch.nonzero(k1).squeeze(1)
+        wave_pairs = torch.nonzero((kobs >= 1) & ~k1).squeeze(1)
         self.C.value_update(
             gs.rec_values, gs.rec_dist, gs.rec_file, ent_rec_ptr, ent_rec_idx,
             gs.ent_values, m.theta, m.phi, m.log_phi, m.norm_lin, m.log_norm,
@@ -289,7 +310,7 @@ class GpuEngine(CpuEngine):
             m.pow_prob, m.pow_alias, m.pow_off, m.log_pow_total, m.attr_const,
             m.Kc, 1 if flags.collapsed_entity_values else 0,
             1 if flags.sequential else 0, seed, it, self._ent_id_base, self._err,
-            wave_pairs, base_pairs,
+            wave_pairs, base_pairs, k1_pairs,
         )
 
         # --- phase 3: distortion update --------------------------------------

@@ -19,7 +19,8 @@ void link_update(torch::Tensor rec_values, torch::Tensor rec_dist,
                  torch::Tensor csr_row_ptr, torch::Tensor csr_col,
                  torch::Tensor csr_sim, torch::Tensor attr_const, int64_t seed,
                  int64_t iteration, torch::Tensor rec_ent_out,
-                 torch::Tensor rec_ent_in, torch::Tensor error_count);
+                 torch::Tensor rec_ent_in, torch::Tensor error_count,
+                 torch::Tensor small_list, torch::Tensor wave_list);
 void link_update_dense(torch::Tensor rec_values, torch::Tensor rec_dist,
                        torch::Tensor rec_gid, torch::Tensor rec_part,
                        torch::Tensor rec_file, torch::Tensor ent_values,
@@ -41,7 +42,8 @@ void value_update(torch::Tensor rec_values, torch::Tensor rec_dist,
                   torch::Tensor log_pow_total, torch::Tensor attr_const, int64_t Kc,
                   int64_t collapsed, int64_t sequential, int64_t seed,
                   int64_t iteration, int64_t ent_id_base, torch::Tensor error_count,
-                  torch::Tensor wave_pairs, torch::Tensor base_pairs);
+                  torch::Tensor wave_pairs, torch::Tensor base_pairs,
+                  torch::Tensor k1_pairs);
 void distortion_update(torch::Tensor rec_values, torch::Tensor rec_dist,
                        torch::Tensor rec_file, torch::Tensor rec_gid,
                        torch::Tensor rec_ent, torch::Tensor ent_values,

@@ -54,14 +54,16 @@ __global__ void link_update_kernel(
     const int32_t* __restrict__ csr_col,
     const float* __restrict__ csr_sim,
     const uint8_t* __restrict__ attr_const,  // [A]
+    const int64_t* __restrict__ rec_list,    // [R] records to process (or null)
     int64_t R, int A,
     uint64_t seed, uint32_t iteration,
     int64_t* __restrict__ rec_ent_out,       // [R]
     const int64_t* __restrict__ rec_ent_in,  // [R]
     int* __restrict__ error_count) {
   const int lane = threadIdx.x & (WAVE - 1);
-  const int64_t r = (int64_t)blockIdx.x * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
+  int64_t r = (int64_t)blockIdx.x * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
   if (r >= R) return;
+  if (rec_list != nullptr) r = rec_list[r];
 
   // Gather per-attribute candidate ranges (observed non-distorted) and the
   // observed-distorted non-constant attributes (constant od attrs scale all
@@ -141,6 +143,71 @@ __global__ void link_update_kernel(
     }
     rec_ent_out[r] = best_e;
   }
+}
+
+// Thread-per-record link update for records whose smallest candidate list is
+// short (the common case once clusters localize): one thread walks the
+// intersection serially — 64x fewer wave slots than the wave path.
+__global__ void link_update_small_kernel(
+    const int64_t* __restrict__ rec_list, int64_t n_recs,
+    const int32_t* __restrict__ rec_values, const uint8_t* __restrict__ rec_dist,
+    const int64_t* __restrict__ rec_gid, const int64_t* __restrict__ cand_lo,
+    const int64_t* __restrict__ cand_hi, const int32_t* __restrict__ postings,
+    const int32_t* __restrict__ ent_values, const float* __restrict__ log_norm,
+    const int64_t* __restrict__ voff, const int64_t* __restrict__ csr_row_ptr,
+    const int32_t* __restrict__ csr_col, const float* __restrict__ csr_sim,
+    const uint8_t* __restrict__ attr_const, int A, uint64_t seed,
+    uint32_t iteration, int64_t* __restrict__ rec_ent_out,
+    const int64_t* __restrict__ rec_ent_in, int* __restrict__ error_count) {
+  const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= n_recs) return;
+  const int64_t r = rec_list[idx];
+
+  int nd_n = 0, od_n = 0;
+  int64_t nd_lo[MAX_ATTRS], nd_hi[MAX_ATTRS];
+  int od_a[MAX_ATTRS];
+  int64_t base_lo = 0, base_hi = 0, base_sz = INT64_MAX;
+  for (int a = 0; a < A; ++a) {
+    const int32_t x = rec_values[r * A + a];
+    if (x < 0) continue;
+    if (!rec_dist[r * A + a]) {
+      const int64_t lo = cand_lo[r * A + a], hi = cand_hi[r * A + a];
+      nd_lo[nd_n] = lo;
+      nd_hi[nd_n] = hi;
+      ++nd_n;
+      if (hi - lo < base_sz) { base_sz = hi - lo; base_lo = lo; base_hi = hi; }
+    } else if (!attr_const[a]) {
+      od_a[od_n++] = a;
+    }
+  }
+  const uint64_t gid = (uint64_t)rec_gid[r];
+  float best_score = -INFINITY;
+  long long best_e = -1;
+  for (int64_t i = base_lo; i < base_hi; ++i) {
+    const int32_t e = postings[i];
+    bool ok = true;
+    for (int j = 0; j < nd_n; ++j) {
+      if (nd_lo[j] == base_lo) continue;
+      if (!contains_i32(postings, nd_lo[j], nd_hi[j], e)) { ok = false; break; }
+    }
+    if (!ok) continue;
+    float logw = 0.0f;
+    for (int j = 0; j < od_n; ++j) {
+      const int a = od_a[j];
+      const int32_t x = rec_values[r * A + a];
+      const int32_t y = ent_values[(int64_t)e * A + a];
+      logw += log_norm[voff[a] + y] +
+              sim_lookup(csr_row_ptr, csr_col, csr_sim, voff[a] + x, y);
+    }
+    const float g = gumbel_from_uniform(
+        philox_uniform(seed, iteration, PH_LINK, gid, (uint32_t)e));
+    if (logw + g > best_score) { best_score = logw + g; best_e = e; }
+  }
+  if (best_e < 0) {
+    atomicAdd(error_count, 1);
+    best_e = rec_ent_in[r];
+  }
+  rec_ent_out[r] = best_e;
 }
 
 // ---------------------------------------------------------------------------
@@ -285,6 +352,89 @@ __global__ void value_base_draw_kernel(ValueArgs args) {
   philox_uniform2(args.seed, args.iteration, PH_VALM, elem, 0xFFFF0000u, &u1, &u2);
   const int v = alias_draw(args.phi_prob + v0, args.phi_alias + v0, V, u1, u2);
   args.ent_values[e * args.A + a] = (int32_t)v;
+}
+
+// Thread-per-pair value update for single-record clusters (k_obs == 1) with a
+// small sim row: the perturbation support is one row, scanned serially by one
+// thread — 64x fewer wave slots than the wave path for the dominant case.
+__global__ void value_update_k1_kernel(ValueArgs args) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= args.n_pairs) return;
+  const int64_t pair = args.pair_list[i];
+  const int64_t e = pair / args.A;
+  const int a = (int)(pair % args.A);
+  const bool is_const = args.attr_const[a];
+  const int64_t v0 = args.voff[a];
+  const int V = (int)(args.voff[a + 1] - v0);
+  const uint64_t elem = (args.ent_id_base + (uint64_t)e) * 32u + (uint64_t)a;
+
+  // locate the single observed linked record
+  const int64_t r_lo = args.ent_rec_ptr[e], r_hi = args.ent_rec_ptr[e + 1];
+  int64_t r = -1;
+  int32_t x = -1;
+  for (int64_t j = r_lo; j < r_hi; ++j) {
+    const int64_t rr = args.ent_rec_idx[j];
+    const int32_t xx = args.rec_values[rr * args.A + a];
+    if (xx >= 0) { r = rr; x = xx; break; }
+  }
+
+  // non-collapsed deterministic copy
+  if (!args.collapsed && !args.rec_dist[r * args.A + a]) {
+    args.ent_values[e * args.A + a] = x;
+    return;
+  }
+
+  auto base_draw = [&](uint32_t tag) -> int {
+    float u1, u2;
+    philox_uniform2(args.seed, args.iteration, PH_VALM, elem, tag, &u1, &u2);
+    if (is_const) return alias_draw(args.phi_prob + v0, args.phi_alias + v0, V, u1, u2);
+    const int64_t off = args.pow_off[a];  // k = 1 table
+    return alias_draw(args.pow_prob + off, args.pow_alias + off, V, u1, u2);
+  };
+  if (!args.collapsed && is_const) {
+    args.ent_values[e * args.A + a] = (int32_t)base_draw(0xFFFF0000u);
+    return;
+  }
+
+  const float log_z = is_const ? 0.0f : args.log_pow_total[a * (args.Kc + 1) + 1];
+  float se = 0.0f;
+  if (args.collapsed) {
+    const float th = args.theta[a * args.F + args.rec_file[r]];
+    const float px = args.phi[v0 + x];
+    se = (1.0f / th - 1.0f) / (is_const ? px : px * args.norm_lin[v0 + x]);
+  }
+  double W = 0.0;
+  float best = -INFINITY;
+  long long best_v = -1;
+  if (is_const) {
+    if (se > 0.0f) {
+      const float L = __logf(1.0f + se);
+      const float logw = args.log_phi[v0 + x] + L + __logf(1.0f - __expf(-L));
+      W = exp((double)logw);
+      best_v = x;
+    }
+  } else {
+    const int64_t row_lo = args.csr_row_ptr[v0 + x], row_hi = args.csr_row_ptr[v0 + x + 1];
+    for (int64_t jj = row_lo; jj < row_hi; ++jj) {
+      const int32_t v = args.csr_col[jj];
+      const float s = args.csr_sim[jj];
+      const float L = (v == x && se > 0.0f) ? __logf(__expf(s) + se) : s;
+      const float logw = args.log_phi[v0 + v] + args.log_norm[v0 + v] - log_z +
+                         L + __logf(1.0f - __expf(-L));
+      W += exp((double)logw);
+      const float g = gumbel_from_uniform(
+          philox_uniform(args.seed, args.iteration, PH_VALG, elem, (uint32_t)v));
+      if (logw + g > best) { best = logw + g; best_v = v; }
+    }
+  }
+  const float u = philox_uniform(args.seed, args.iteration, PH_VALM, elem, 0xFFFF0001u);
+  int v_new;
+  if ((double)u < 1.0 / (1.0 + W) || best_v < 0) {
+    v_new = base_draw(0xFFFF0002u);
+  } else {
+    v_new = (int)best_v;
+  }
+  args.ent_values[e * args.A + a] = (int32_t)v_new;
 }
 
 __global__ void __launch_bounds__(WAVES_PER_BLOCK_VAL * WAVE)
@@ -830,26 +980,49 @@ void link_update(
     torch::Tensor log_norm, torch::Tensor voff, torch::Tensor csr_row_ptr,
     torch::Tensor csr_col, torch::Tensor csr_sim, torch::Tensor attr_const,
     int64_t seed, int64_t iteration, torch::Tensor rec_ent_out,
-    torch::Tensor rec_ent_in, torch::Tensor error_count) {
+    torch::Tensor rec_ent_in, torch::Tensor error_count,
+    torch::Tensor small_list, torch::Tensor wave_list) {
   CHECK_GPU(rec_values);
   const int64_t R = rec_values.size(0);
   const int A = (int)rec_values.size(1);
   TORCH_CHECK(A <= MAX_ATTRS, "at most ", MAX_ATTRS, " matching attributes supported");
   if (R == 0) return;
   constexpr int WPB = 4;
-  dim3 grid((unsigned)wave_grid(R, WPB));
-  hipLaunchKernelGGL(link_update_kernel, grid, dim3(WPB * WAVE), 0,
-                     at::cuda::getCurrentCUDAStream(),
-                     rec_values.data_ptr<int32_t>(), rec_dist.data_ptr<uint8_t>(),
-                     rec_gid.data_ptr<int64_t>(), rec_part.data_ptr<int32_t>(),
-                     cand_lo.data_ptr<int64_t>(), cand_hi.data_ptr<int64_t>(),
-                     postings.data_ptr<int32_t>(), ent_values.data_ptr<int32_t>(),
-                     ent_ptr.data_ptr<int64_t>(), log_norm.data_ptr<float>(),
-                     voff.data_ptr<int64_t>(), csr_row_ptr.data_ptr<int64_t>(),
-                     csr_col.data_ptr<int32_t>(), csr_sim.data_ptr<float>(),
-                     attr_const.data_ptr<uint8_t>(), R, A, (uint64_t)seed,
-                     (uint32_t)iteration, rec_ent_out.data_ptr<int64_t>(),
-                     rec_ent_in.data_ptr<int64_t>(), error_count.data_ptr<int>());
+  const bool split = small_list.numel() > 0 || wave_list.numel() > 0;
+  if (!split || wave_list.numel() > 0) {
+    const int64_t n = split ? wave_list.numel() : R;
+    dim3 grid((unsigned)wave_grid(n, WPB));
+    hipLaunchKernelGGL(link_update_kernel, grid, dim3(WPB * WAVE), 0,
+                       at::cuda::getCurrentCUDAStream(),
+                       rec_values.data_ptr<int32_t>(), rec_dist.data_ptr<uint8_t>(),
+                       rec_gid.data_ptr<int64_t>(), rec_part.data_ptr<int32_t>(),
+                       cand_lo.data_ptr<int64_t>(), cand_hi.data_ptr<int64_t>(),
+                       postings.data_ptr<int32_t>(), ent_values.data_ptr<int32_t>(),
+                       ent_ptr.data_ptr<int64_t>(), log_norm.data_ptr<float>(),
+                       voff.data_ptr<int64_t>(), csr_row_ptr.data_ptr<int64_t>(),
+                       csr_col.data_ptr<int32_t>(), csr_sim.data_ptr<float>(),
+                       attr_const.data_ptr<uint8_t>(),
+                       split ? wave_list.data_ptr<int64_t>() : nullptr, n, A,
+                       (uint64_t)seed, (uint32_t)iteration,
+                       rec_ent_out.data_ptr<int64_t>(),
+                       rec_ent_in.data_ptr<int64_t>(), error_count.data_ptr<int>());
+  }
+  if (small_list.numel() > 0) {
+    const int64_t n = small_list.numel();
+    dim3 grid((unsigned)((n + 255) / 256));
+    hipLaunchKernelGGL(link_update_small_kernel, grid, dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(),
+                       small_list.data_ptr<int64_t>(), n,
+                       rec_values.data_ptr<int32_t>(), rec_dist.data_ptr<uint8_t>(),
+                       rec_gid.data_ptr<int64_t>(), cand_lo.data_ptr<int64_t>(),
+                       cand_hi.data_ptr<int64_t>(), postings.data_ptr<int32_t>(),
+                       ent_values.data_ptr<int32_t>(), log_norm.data_ptr<float>(),
+                       voff.data_ptr<int64_t>(), csr_row_ptr.data_ptr<int64_t>(),
+                       csr_col.data_ptr<int32_t>(), csr_sim.data_ptr<float>(),
+                       attr_const.data_ptr<uint8_t>(), A, (uint64_t)seed,
+                       (uint32_t)iteration, rec_ent_out.data_ptr<int64_t>(),
+                       rec_ent_in.data_ptr<int64_t>(), error_count.data_ptr<int>());
+  }
 }
 
 void link_update_dense(
@@ -934,7 +1107,7 @@ void value_update(
     torch::Tensor pow_alias, torch::Tensor pow_off, torch::Tensor log_pow_total,
     torch::Tensor attr_const, int64_t Kc, int64_t collapsed, int64_t sequential,
     int64_t seed, int64_t iteration, int64_t ent_id_base, torch::Tensor error_count,
-    torch::Tensor wave_pairs, torch::Tensor base_pairs) {
+    torch::Tensor wave_pairs, torch::Tensor base_pairs, torch::Tensor k1_pairs) {
   ValueArgs args = make_value_args(
       rec_values, rec_dist, rec_file, ent_rec_ptr, ent_rec_idx, ent_values, theta,
       phi, log_phi, norm_lin, log_norm, voff, csr_row_ptr, csr_col, csr_sim,
@@ -958,7 +1131,15 @@ void value_update(
     hipLaunchKernelGGL(value_base_draw_kernel, grid, dim3(256), 0,
                        at::cuda::getCurrentCUDAStream(), args);
   }
-  // k_obs >= 1 pairs: one wave each
+  // k_obs == 1 pairs with small sim rows: one thread each
+  if (k1_pairs.numel() > 0) {
+    args.pair_list = k1_pairs.data_ptr<int64_t>();
+    args.n_pairs = k1_pairs.numel();
+    dim3 grid((unsigned)((args.n_pairs + 255) / 256));
+    hipLaunchKernelGGL(value_update_k1_kernel, grid, dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(), args);
+  }
+  // remaining pairs: one wave each
   if (wave_pairs.numel() > 0) {
     args.pair_list = wave_pairs.data_ptr<int64_t>();
     args.n_pairs = wave_pairs.numel();

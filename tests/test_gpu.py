@@ -205,6 +205,8 @@ def test_link_kernel_distribution():
         model.log_norm, model.voff, model.csr_row_ptr, model.csr_col, model.csr_sim,
         model.attr_const, 4321, 11, out,
         _dev(np.zeros(N, np.int64), torch.int64), err,
+        torch.empty(0, dtype=torch.int64, device=DEV),
+        torch.empty(0, dtype=torch.int64, device=DEV),
     )
     assert int(err.cpu()) == 0
     sel = out.cpu().numpy()
@@ -261,6 +263,7 @@ def test_value_kernel_distribution():
         model.pow_off, model.log_pow_total, model.attr_const, model.Kc,
         1, 0, 777, 5, 0, err,
         torch.arange(E * 2, dtype=torch.int64, device=DEV),
+        torch.empty(0, dtype=torch.int64, device=DEV),
         torch.empty(0, dtype=torch.int64, device=DEV),
     )
     got = ev.cpu().numpy()[:, a]
