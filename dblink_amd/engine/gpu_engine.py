@@ -186,6 +186,9 @@ class GpuEngine(CpuEngine):
         self._gs = None
         self._ent_id_base = rank << 40
         self._err = torch.zeros(1, dtype=torch.int32, device=device)
+        import os
+        self.k1_row_cap = int(os.environ.get("DBLINK_K1_ROW_CAP", "48"))
+        self.debug_classes = os.environ.get("DBLINK_DEBUG_CLASSES", "") == "1"
         self._loglik_buf = torch.zeros(1, dtype=torch.float64, device=device)
 
     # ---- state residency -----------------------------------------------------
@@ -257,15 +260,20 @@ class GpuEngine(CpuEngine):
             )
         else:
             # records whose smallest candidate list is short run one-per-thread
+            # (skip the classification for small problems: extra launches cost
+            # more than they save below ~100k records)
             cand_lo = cand_lo.contiguous()
             cand_hi = cand_hi.contiguous()
-            nd = (gs.rec_values >= 0) & (gs.rec_dist == 0)
-            sizes = torch.where(nd, cand_hi - cand_lo, torch.full_like(cand_hi, 1 << 40))
-            min_sizes = sizes.amin(dim=1)
-            has_nd = nd.any(dim=1)
-            small = has_nd & (min_sizes <= 16)
-            small_list = torch.nonzero(small).squeeze(1)
-            wave_list = torch.nonzero(~small).squeeze(1)
+            if R >= 100_000:
+                nd = (gs.rec_values >= 0) & (gs.rec_dist == 0)
+                sizes = torch.where(nd, cand_hi - cand_lo, torch.full_like(cand_hi, 1 << 40))
+                min_sizes = sizes.amin(dim=1)
+                has_nd = nd.any(dim=1)
+                small = has_nd & (min_sizes <= 16)
+                small_list = torch.nonzero(small).squeeze(1)
+                wave_list = torch.nonzero(~small).squeeze(1)
+            else:
+                small_list = wave_list = torch.empty(0, dtype=torch.int64, device=self.device)
             self.C.link_update(
                 gs.rec_values, gs.rec_dist, gs.rec_gid, gs.rec_part,
                 cand_lo, cand_hi, postings, gs.ent_values,
@@ -292,17 +300,27 @@ class GpuEngine(CpuEngine):
         idxm = pair_idx.reshape(-1)[obs.reshape(-1)]
         kobs.scatter_add_(0, idxm, torch.ones_like(idxm, dtype=torch.int32))
         base_pairs = torch.nonzero(kobs == 0).squeeze(1)
-        # per-pair perturbation support size (sim-row length, 1 for const)
-        gv = m.voff[:-1].view(1, A) + gs.rec_values.clamp_min(0).to(torch.int64)
-        rowsz = (m.csr_row_ptr[gv + 1] - m.csr_row_ptr[gv]).to(torch.int32)
-        rowsz = torch.where(m.attr_const.view(1, A) > 0,
-                            torch.ones_like(rowsz), rowsz)
-        entries = torch.zeros(E * A, dtype=torch.int32, device=self.device)
-        entries.scatter_add_(0, pair_idx.reshape(-1)[obs.reshape(-1)],
-                             rowsz.reshape(-1)[obs.reshape(-1)])
-        k1 = (kobs == 1) & (entries <= 48)
-        k1_pairs = torch.nonzero(k1).squeeze(1)
-        wave_pairs = torch.nonzero((kobs >= 1) & ~k1).squeeze(1)
+        if R >= 100_000:
+            # per-pair perturbation support size (sim-row length, 1 for const)
+            gv = m.voff[:-1].view(1, A) + gs.rec_values.clamp_min(0).to(torch.int64)
+            rowsz = (m.csr_row_ptr[gv + 1] - m.csr_row_ptr[gv]).to(torch.int32)
+            rowsz = torch.where(m.attr_const.view(1, A) > 0,
+                                torch.ones_like(rowsz), rowsz)
+            entries = torch.zeros(E * A, dtype=torch.int32, device=self.device)
+            entries.scatter_add_(0, pair_idx.reshape(-1)[obs.reshape(-1)],
+                                 rowsz.reshape(-1)[obs.reshape(-1)])
+            k1 = (kobs == 1) & (entries <= self.k1_row_cap)
+            k1_pairs = torch.nonzero(k1).squeeze(1)
+            wave_pairs = torch.nonzero((kobs >= 1) & ~k1).squeeze(1)
+            if self.debug_classes:
+                import sys
+                print(f"[dblink classes] base={base_pairs.numel()} k1={k1_pairs.numel()} "
+                      f"wave={wave_pairs.numel()} entries_mean={entries[kobs>=1].float().mean().item():.1f} "
+                      f"entries_p99={entries[kobs>=1].float().quantile(0.99).item():.0f} "
+                      f"kobs_max={kobs.max().item()}", file=sys.stderr)
+        else:
+            k1_pairs = torch.empty(0, dtype=torch.int64, device=self.device)
+            wave_pairs = torch.nonzero(kobs >= 1).squeeze(1)
         self.C.value_update(
             gs.rec_values, gs.rec_dist, gs.rec_file, ent_rec_ptr, ent_rec_idx,
             gs.ent_values, m.theta, m.phi, m.log_phi, m.norm_lin, m.log_norm,
