@@ -97,6 +97,37 @@ class GpuModel:
                 log_pow_total[a, k] = np.log(ia.index.sim_norm_total(k))
             off += self.Kc * V
 
+        # Static k=1 perturbation tables: for a single linked record with
+        # value x, the perturbation weights are base(v)*(f(v)-1) where the
+        # self-term correction adds exactly (1/theta - 1)/Z1 of mass. So the
+        # whole distribution reduces to the PRECOMPUTED row quantities
+        #   raw_w1[j]   = phi(col)*norm(col)*(expsim-1)        (f64)
+        #   row prefix  = exclusive cumsum of raw_w1 within the row
+        #   rawsum[x]   = row total
+        # and a k=1 draw is one binary search (value_update_k1_kernel).
+        raw_list, excl_list, rawsum = [], [], np.zeros(self.Vtot, dtype=np.float64)
+        for a, ia in enumerate(attrs):
+            if ia.is_constant:
+                continue
+            si = ia.index.sim_index
+            pr = ia.index.probs
+            nm = ia.index.sim_norms
+            raw = pr[si.col] * nm[si.col] * (si.expsim - 1.0)
+            c = np.cumsum(raw)
+            counts = np.diff(si.row_ptr)
+            start = si.row_ptr[:-1]
+            base_c = np.where(start > 0, c[np.maximum(start - 1, 0)], 0.0)
+            excl = c - raw - np.repeat(base_c, counts)
+            ends = si.row_ptr[1:] - 1
+            sums = np.where(counts > 0, c[np.maximum(ends, 0)], 0.0) - base_c
+            raw_list.append(raw)
+            excl_list.append(excl)
+            rawsum[voff[a] : voff[a + 1]] = sums
+        csr_excl = np.concatenate(excl_list) if excl_list else np.empty(0)
+        z1 = np.array(
+            [1.0 if ia.is_constant else ia.index.sim_norm_total(1) for ia in attrs]
+        )
+
         self_expsim = np.concatenate(
             [
                 np.full(
@@ -130,6 +161,9 @@ class GpuModel:
             np.array([1 if ia.is_constant else 0 for ia in attrs], dtype=np.uint8), torch.uint8
         )
         self.self_expsim = dev(self_expsim, torch.float32)
+        self.csr_excl = dev(csr_excl, torch.float64)
+        self.csr_rawsum = dev(rawsum, torch.float64)
+        self.z1 = dev(z1, torch.float64)
         self.theta = torch.zeros((A, self.F), dtype=torch.float32, device=device)
 
 
@@ -301,23 +335,13 @@ class GpuEngine(CpuEngine):
         kobs.scatter_add_(0, idxm, torch.ones_like(idxm, dtype=torch.int32))
         base_pairs = torch.nonzero(kobs == 0).squeeze(1)
         if R >= 100_000:
-            # per-pair perturbation support size (sim-row length, 1 for const)
-            gv = m.voff[:-1].view(1, A) + gs.rec_values.clamp_min(0).to(torch.int64)
-            rowsz = (m.csr_row_ptr[gv + 1] - m.csr_row_ptr[gv]).to(torch.int32)
-            rowsz = torch.where(m.attr_const.view(1, A) > 0,
-                                torch.ones_like(rowsz), rowsz)
-            entries = torch.zeros(E * A, dtype=torch.int32, device=self.device)
-            entries.scatter_add_(0, pair_idx.reshape(-1)[obs.reshape(-1)],
-                                 rowsz.reshape(-1)[obs.reshape(-1)])
-            k1 = (kobs == 1) & (entries <= self.k1_row_cap)
+            k1 = kobs == 1
             k1_pairs = torch.nonzero(k1).squeeze(1)
-            wave_pairs = torch.nonzero((kobs >= 1) & ~k1).squeeze(1)
+            wave_pairs = torch.nonzero(kobs >= 2).squeeze(1)
             if self.debug_classes:
                 import sys
                 print(f"[dblink classes] base={base_pairs.numel()} k1={k1_pairs.numel()} "
-                      f"wave={wave_pairs.numel()} entries_mean={entries[kobs>=1].float().mean().item():.1f} "
-                      f"entries_p99={entries[kobs>=1].float().quantile(0.99).item():.0f} "
-                      f"kobs_max={kobs.max().item()}", file=sys.stderr)
+                      f"wave={wave_pairs.numel()} kobs_max={kobs.max().item()}", file=sys.stderr)
         else:
             k1_pairs = torch.empty(0, dtype=torch.int64, device=self.device)
             wave_pairs = torch.nonzero(kobs >= 1).squeeze(1)
@@ -328,7 +352,7 @@ class GpuEngine(CpuEngine):
             m.pow_prob, m.pow_alias, m.pow_off, m.log_pow_total, m.attr_const,
             m.Kc, 1 if flags.collapsed_entity_values else 0,
             1 if flags.sequential else 0, seed, it, self._ent_id_base, self._err,
-            wave_pairs, base_pairs, k1_pairs,
+            wave_pairs, base_pairs, k1_pairs, m.csr_excl, m.csr_rawsum, m.z1,
         )
 
         # --- phase 3: distortion update --------------------------------------

@@ -43,7 +43,8 @@ void value_update(torch::Tensor rec_values, torch::Tensor rec_dist,
                   int64_t collapsed, int64_t sequential, int64_t seed,
                   int64_t iteration, int64_t ent_id_base, torch::Tensor error_count,
                   torch::Tensor wave_pairs, torch::Tensor base_pairs,
-                  torch::Tensor k1_pairs);
+                  torch::Tensor k1_pairs, torch::Tensor csr_excl,
+                  torch::Tensor csr_rawsum, torch::Tensor z1);
 void distortion_update(torch::Tensor rec_values, torch::Tensor rec_dist,
                        torch::Tensor rec_file, torch::Tensor rec_gid,
                        torch::Tensor rec_ent, torch::Tensor ent_values,

@@ -284,6 +284,9 @@ constexpr int WAVES_PER_BLOCK_VAL = 4;    // 256 threads
 struct ValueArgs {
   const int64_t* pair_list;    // [n_pairs] flattened (e*A + a) needing wave work
   int64_t n_pairs;
+  const double* csr_excl;      // [nnz] exclusive row prefix of raw k=1 weights
+  const double* csr_rawsum;    // [Vtot] row totals of raw k=1 weights
+  const double* z1;            // [A] power-1 normalizer Z_1
   const int32_t* rec_values;
   const uint8_t* rec_dist;
   const int32_t* rec_file;
@@ -354,9 +357,13 @@ __global__ void value_base_draw_kernel(ValueArgs args) {
   args.ent_values[e * args.A + a] = (int32_t)v;
 }
 
-// Thread-per-pair value update for single-record clusters (k_obs == 1) with a
-// small sim row: the perturbation support is one row, scanned serially by one
-// thread — 64x fewer wave slots than the wave path for the dominant case.
+// Thread-per-pair value update for single-record clusters (k_obs == 1).
+// The perturbation distribution for one linked record with value x is STATIC
+// per (attribute, x) up to the self-term correction, which is exactly
+// (1/theta - 1) of extra raw mass on x. Using the precomputed raw row weights
+// raw_w1[j] = phi(col)*norm(col)*(expsim-1), their exclusive row prefix and
+// row totals, a draw is: one Philox call, a mixture test, and one binary
+// search over the row prefix — O(log row) regardless of row size.
 __global__ void value_update_k1_kernel(ValueArgs args) {
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= args.n_pairs) return;
@@ -384,55 +391,62 @@ __global__ void value_update_k1_kernel(ValueArgs args) {
     return;
   }
 
-  auto base_draw = [&](uint32_t tag) -> int {
-    float u1, u2;
-    philox_uniform2(args.seed, args.iteration, PH_VALM, elem, tag, &u1, &u2);
-    if (is_const) return alias_draw(args.phi_prob + v0, args.phi_alias + v0, V, u1, u2);
+  u32x4 rnd = philox4x32(args.seed, (uint32_t)elem, (uint32_t)(elem >> 32),
+                         args.iteration ^ (PH_VALM << 24), 0xFFFF0000u);
+  const double u_mix = ((double)rnd.x + 0.5) * 2.3283064365386963e-10;
+  const double u_sel = ((double)rnd.y + 0.5) * 2.3283064365386963e-10;
+  const float u_a1 = u32_to_uniform(rnd.z);
+  const float u_a2 = u32_to_uniform(rnd.w);
+
+  auto base_draw = [&]() -> int {
+    if (is_const)
+      return alias_draw(args.phi_prob + v0, args.phi_alias + v0, V, u_a1, u_a2);
     const int64_t off = args.pow_off[a];  // k = 1 table
-    return alias_draw(args.pow_prob + off, args.pow_alias + off, V, u1, u2);
+    return alias_draw(args.pow_prob + off, args.pow_alias + off, V, u_a1, u_a2);
   };
   if (!args.collapsed && is_const) {
-    args.ent_values[e * args.A + a] = (int32_t)base_draw(0xFFFF0000u);
+    args.ent_values[e * args.A + a] = (int32_t)base_draw();
     return;
   }
 
-  const float log_z = is_const ? 0.0f : args.log_pow_total[a * (args.Kc + 1) + 1];
-  float se = 0.0f;
+  // raw perturbation mass (in Z-scaled units): row total + self correction
+  double extra = 0.0;
   if (args.collapsed) {
     const float th = args.theta[a * args.F + args.rec_file[r]];
-    const float px = args.phi[v0 + x];
-    se = (1.0f / th - 1.0f) / (is_const ? px : px * args.norm_lin[v0 + x]);
+    extra = 1.0 / (double)th - 1.0;  // phi*norm*se collapses to (1/theta - 1)
   }
-  double W = 0.0;
-  float best = -INFINITY;
-  long long best_v = -1;
+  double raw_total, Wnorm;
   if (is_const) {
-    if (se > 0.0f) {
-      const float L = __logf(1.0f + se);
-      const float logw = args.log_phi[v0 + x] + L + __logf(1.0f - __expf(-L));
-      W = (double)__expf(logw);
-      best_v = x;
-    }
+    // row is {x}: raw weight = phi(x)*(factor-1) = phi(x)*se = (1/theta-1)
+    raw_total = extra;
+    Wnorm = extra <= 0.0 ? 0.0 : extra;  // base = phi (Z = 1), raw already phi-scaled
   } else {
-    const int64_t row_lo = args.csr_row_ptr[v0 + x], row_hi = args.csr_row_ptr[v0 + x + 1];
-    for (int64_t jj = row_lo; jj < row_hi; ++jj) {
-      const int32_t v = args.csr_col[jj];
-      const float s = args.csr_sim[jj];
-      const float L = (v == x && se > 0.0f) ? __logf(__expf(s) + se) : s;
-      const float logw = args.log_phi[v0 + v] + args.log_norm[v0 + v] - log_z +
-                         L + __logf(1.0f - __expf(-L));
-      W += (double)__expf(logw);
-      const float g = gumbel_from_uniform(
-          philox_uniform(args.seed, args.iteration, PH_VALG, elem, (uint32_t)v));
-      if (logw + g > best) { best = logw + g; best_v = v; }
-    }
+    raw_total = args.csr_rawsum[v0 + x] + extra;
+    Wnorm = raw_total / args.z1[a];
   }
-  const float u = philox_uniform(args.seed, args.iteration, PH_VALM, elem, 0xFFFF0001u);
+
   int v_new;
-  if ((double)u < 1.0 / (1.0 + W) || best_v < 0) {
-    v_new = base_draw(0xFFFF0002u);
+  if (u_mix < 1.0 / (1.0 + Wnorm) || raw_total <= 0.0) {
+    v_new = base_draw();
+  } else if (is_const) {
+    v_new = x;  // the only support value
   } else {
-    v_new = (int)best_v;
+    const double t = u_sel * raw_total;
+    const int64_t row_lo = args.csr_row_ptr[v0 + x], row_hi = args.csr_row_ptr[v0 + x + 1];
+    if (t >= args.csr_rawsum[v0 + x]) {
+      v_new = x;  // self-correction mass
+    } else {
+      // first entry whose exclusive prefix exceeds t, minus one
+      int64_t lo = row_lo, hi = row_hi;
+      while (lo < hi) {
+        const int64_t mid = (lo + hi) >> 1;
+        if (args.csr_excl[mid] <= t) lo = mid + 1; else hi = mid;
+      }
+      int64_t jidx = lo - 1;
+      if (jidx < row_lo) jidx = row_lo;
+      if (jidx >= row_hi) jidx = row_hi - 1;
+      v_new = args.csr_col[jidx];
+    }
   }
   args.ent_values[e * args.A + a] = (int32_t)v_new;
 }
@@ -1063,8 +1077,12 @@ static ValueArgs make_value_args(
     torch::Tensor phi_prob, torch::Tensor phi_alias, torch::Tensor pow_prob,
     torch::Tensor pow_alias, torch::Tensor pow_off, torch::Tensor log_pow_total,
     torch::Tensor attr_const, int64_t Kc, int64_t collapsed, int64_t seed,
-    int64_t iteration, int64_t ent_id_base, torch::Tensor error_count) {
+    int64_t iteration, int64_t ent_id_base, torch::Tensor error_count,
+    torch::Tensor csr_excl, torch::Tensor csr_rawsum, torch::Tensor z1) {
   ValueArgs a;
+  a.csr_excl = csr_excl.numel() ? csr_excl.data_ptr<double>() : nullptr;
+  a.csr_rawsum = csr_rawsum.data_ptr<double>();
+  a.z1 = z1.data_ptr<double>();
   a.rec_values = rec_values.data_ptr<int32_t>();
   a.rec_dist = rec_dist.data_ptr<uint8_t>();
   a.rec_file = rec_file.data_ptr<int32_t>();
@@ -1109,12 +1127,14 @@ void value_update(
     torch::Tensor pow_alias, torch::Tensor pow_off, torch::Tensor log_pow_total,
     torch::Tensor attr_const, int64_t Kc, int64_t collapsed, int64_t sequential,
     int64_t seed, int64_t iteration, int64_t ent_id_base, torch::Tensor error_count,
-    torch::Tensor wave_pairs, torch::Tensor base_pairs, torch::Tensor k1_pairs) {
+    torch::Tensor wave_pairs, torch::Tensor base_pairs, torch::Tensor k1_pairs,
+    torch::Tensor csr_excl, torch::Tensor csr_rawsum, torch::Tensor z1) {
   ValueArgs args = make_value_args(
       rec_values, rec_dist, rec_file, ent_rec_ptr, ent_rec_idx, ent_values, theta,
       phi, log_phi, norm_lin, log_norm, voff, csr_row_ptr, csr_col, csr_sim,
       phi_prob, phi_alias, pow_prob, pow_alias, pow_off, log_pow_total, attr_const,
-      Kc, collapsed, seed, iteration, ent_id_base, error_count);
+      Kc, collapsed, seed, iteration, ent_id_base, error_count, csr_excl,
+      csr_rawsum, z1);
   if (sequential) {
     const int64_t pairs = args.E * args.A;
     if (pairs == 0) return;

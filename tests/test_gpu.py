@@ -265,10 +265,31 @@ def test_value_kernel_distribution():
         torch.arange(E * 2, dtype=torch.int64, device=DEV),
         torch.empty(0, dtype=torch.int64, device=DEV),
         torch.empty(0, dtype=torch.int64, device=DEV),
+        model.csr_excl, model.csr_rawsum, model.z1,
     )
     got = ev.cpu().numpy()[:, a]
     emp = np.bincount(got, minlength=V) / E
     assert tv_distance(emp, exact) < 0.02, (emp[:10], exact[:10])
+
+    # same distribution via the table-based k1 thread kernel
+    ev2 = _dev(ent_vals, torch.int32)
+    C.value_update(
+        _dev(rec_values, torch.int32), _dev(rec_dist, torch.uint8),
+        _dev(np.zeros(E, np.int32), torch.int32),
+        _dev(ent_rec_ptr, torch.int64), _dev(ent_rec_idx, torch.int64),
+        ev2, model.theta, model.phi, model.log_phi, model.norm_lin, model.log_norm,
+        model.voff, model.csr_row_ptr, model.csr_col, model.csr_sim,
+        model.phi_prob, model.phi_alias, model.pow_prob, model.pow_alias,
+        model.pow_off, model.log_pow_total, model.attr_const, model.Kc,
+        1, 0, 777, 5, 0, err,
+        torch.empty(0, dtype=torch.int64, device=DEV),
+        torch.empty(0, dtype=torch.int64, device=DEV),
+        torch.arange(E * 2, dtype=torch.int64, device=DEV),
+        model.csr_excl, model.csr_rawsum, model.z1,
+    )
+    got2 = ev2.cpu().numpy()[:, a]
+    emp2 = np.bincount(got2, minlength=V) / E
+    assert tv_distance(emp2, exact) < 0.02, (emp2[:10], exact[:10])
 
 
 @gpu
