@@ -666,39 +666,56 @@ value_update_kernel(ValueArgs args) {
       // log(exp(L) - 1) = L + log1p(-exp(-L)), stable for L > 0
       const float log_expm1 = L + __logf(1.0f - __expf(-L));
       const float logw = log_base_prob(v) + log_expm1;
-      W += exp((double)logw);
+      W += (logw < 80.0f) ? (double)__expf(logw) : exp((double)logw);
       const float g = gumbel_from_uniform(
           philox_uniform(args.seed, args.iteration, PH_VALG, elem, (uint32_t)v));
       if (logw + g > best) { best = logw + g; best_v = v; }
     }
   } else {
-    // dense path: for every domain value accumulate log factors by searching
-    // each record's sim row (correct for any cluster size; O(V * k log row))
-    for (int v_local = lane; v_local < V; v_local += WAVE) {
-      float L = 0.0f;
-      for (int64_t i = r_lo; i < r_hi; ++i) {
-        const int64_t r = args.ent_rec_idx[i];
-        const int32_t x = args.rec_values[r * args.A + a];
-        if (x < 0) continue;
-        const float self_extra = self_extra_of(r, x);
-        if (is_const) {
-          if (v_local == x) L += __logf(1.0f + self_extra);
-        } else {
-          const float s = sim_lookup(args.csr_row_ptr, args.csr_col, args.csr_sim,
-                                     v0 + x, v_local);
-          if (v_local == x && self_extra > 0.0f)
-            L += __logf(__expf(s) + self_extra);
-          else if (s != 0.0f)
-            L += s;
+    // union-merge path (hash would overflow): iterate every row entry, but
+    // process a value only from the FIRST row containing it; full L_v comes
+    // from binary searches in the other rows. O(entries * k log row), far
+    // cheaper than a dense domain scan.
+    for (int64_t i = r_lo; i < r_hi; ++i) {
+      const int64_t r = args.ent_rec_idx[i];
+      const int32_t x = args.rec_values[r * args.A + a];
+      if (x < 0) continue;
+      const int64_t row_lo = args.csr_row_ptr[v0 + x], row_hi = args.csr_row_ptr[v0 + x + 1];
+      for (int64_t j = row_lo + lane; j < row_hi; j += WAVE) {
+        const int32_t v = args.csr_col[j];
+        // dedupe: skip if an earlier observed row already contains v
+        bool first = true;
+        for (int64_t i2 = r_lo; i2 < i && first; ++i2) {
+          const int64_t r2 = args.ent_rec_idx[i2];
+          const int32_t x2 = args.rec_values[r2 * args.A + a];
+          if (x2 < 0) continue;
+          const int64_t lo2 = args.csr_row_ptr[v0 + x2];
+          const int64_t hi2 = args.csr_row_ptr[v0 + x2 + 1];
+          if (contains_i32(args.csr_col, lo2, hi2, v)) first = false;
         }
-      }
-      if (L > 0.0f) {
+        if (!first) continue;
+        // accumulate log factors over ALL observed rows
+        float L = 0.0f;
+        for (int64_t i2 = r_lo; i2 < r_hi; ++i2) {
+          const int64_t r2 = args.ent_rec_idx[i2];
+          const int32_t x2 = args.rec_values[r2 * args.A + a];
+          if (x2 < 0) continue;
+          const float se2 = self_extra_of(r2, x2);
+          const float s2 = (i2 == i && x2 == x)
+                               ? args.csr_sim[j]
+                               : sim_lookup(args.csr_row_ptr, args.csr_col,
+                                            args.csr_sim, v0 + x2, v);
+          if (v == x2 && se2 > 0.0f)
+            L += __logf(__expf(s2) + se2);
+          else if (s2 != 0.0f)
+            L += s2;
+        }
         const float log_expm1 = L + __logf(1.0f - __expf(-L));
-        const float logw = log_base_prob(v_local) + log_expm1;
-        W += exp((double)logw);
+        const float logw = log_base_prob(v) + log_expm1;
+        W += (logw < 80.0f) ? (double)__expf(logw) : exp((double)logw);
         const float g = gumbel_from_uniform(
-            philox_uniform(args.seed, args.iteration, PH_VALG, elem, (uint32_t)v_local));
-        if (logw + g > best) { best = logw + g; best_v = v_local; }
+            philox_uniform(args.seed, args.iteration, PH_VALG, elem, (uint32_t)v));
+        if (logw + g > best) { best = logw + g; best_v = v; }
       }
     }
   }
