@@ -605,7 +605,7 @@ value_update_kernel(ValueArgs args) {
       total_entries += args.csr_row_ptr[v0 + x + 1] - args.csr_row_ptr[v0 + x];
     }
   } else {
-    total_entries = k_obs;
+    total_entries = k_obs < V ? k_obs : V;
   }
   const bool dense = total_entries > (HASH_CAP * 3) / 4;
   int tsize = 64;
@@ -680,6 +680,33 @@ value_update_kernel(ValueArgs args) {
       const int64_t r = args.ent_rec_idx[i];
       const int32_t x = args.rec_values[r * args.A + a];
       if (x < 0) continue;
+      if (is_const) {
+        // "row" is the single value {x}: process on its first occurrence
+        if (lane == 0) {
+          bool first = true;
+          for (int64_t i2 = r_lo; i2 < i && first; ++i2) {
+            const int64_t r2 = args.ent_rec_idx[i2];
+            if (args.rec_values[r2 * args.A + a] == x) first = false;
+          }
+          if (first) {
+            float L = 0.0f;
+            for (int64_t i2 = r_lo; i2 < r_hi; ++i2) {
+              const int64_t r2 = args.ent_rec_idx[i2];
+              if (args.rec_values[r2 * args.A + a] != x) continue;
+              L += __logf(1.0f + self_extra_of(r2, x));
+            }
+            if (L > 0.0f) {
+              const float log_expm1 = L + __logf(1.0f - __expf(-L));
+              const float logw = log_base_prob(x) + log_expm1;
+              W += (logw < 80.0f) ? (double)__expf(logw) : exp((double)logw);
+              const float g = gumbel_from_uniform(philox_uniform(
+                  args.seed, args.iteration, PH_VALG, elem, (uint32_t)x));
+              if (logw + g > best) { best = logw + g; best_v = x; }
+            }
+          }
+        }
+        continue;
+      }
       const int64_t row_lo = args.csr_row_ptr[v0 + x], row_hi = args.csr_row_ptr[v0 + x + 1];
       for (int64_t j = row_lo + lane; j < row_hi; j += WAVE) {
         const int32_t v = args.csr_col[j];
