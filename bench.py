@@ -34,7 +34,13 @@ import time
 import numpy as np
 
 
-def build_cache_and_records(total_records, seed):
+def build_cache_and_records(total_records, seed, schema="rldata", num_files=1):
+    """Synthetic data + cache for the BASELINE.json configs.
+
+    schema "rldata": 5 attrs (by, bm, bd constant; fname_c1, lname_c1
+    Levenshtein) — configs #1-#3 and #5. schema "strings8": 8 Levenshtein
+    string attributes — config #4 (10M-record HBM sizing).
+    """
     from dblink_amd.models.records import (
         Attribute,
         BetaShapeParameters,
@@ -44,18 +50,27 @@ def build_cache_and_records(total_records, seed):
     from dblink_amd.models.similarity import ConstantSimilarityFn, LevenshteinSimilarityFn
     from dblink_amd.utils.synthdata import generate
 
-    cols, header = generate(total_records, dup_fraction=0.1, seed=seed)
-    attr_names = ["by", "bm", "bd", "fname_c1", "lname_c1"]
+    extra = 6 if schema == "strings8" else 0
+    cols, header = generate(
+        total_records, dup_fraction=0.1, seed=seed, num_files=num_files,
+        extra_string_attrs=extra,
+    )
+    prior = BetaShapeParameters(10.0, 1000.0)  # RLdata10000.conf:4
+    lev = lambda: LevenshteinSimilarityFn(7.0, 10.0)
+    if schema == "strings8":
+        attr_names = ["fname_c1", "lname_c1"] + [f"xattr{j}" for j in range(6)]
+        attrs = [Attribute(n, lev(), prior) for n in attr_names]
+    else:
+        attr_names = ["by", "bm", "bd", "fname_c1", "lname_c1"]
+        attrs = [
+            Attribute("by", ConstantSimilarityFn(), prior),
+            Attribute("bm", ConstantSimilarityFn(), prior),
+            Attribute("bd", ConstantSimilarityFn(), prior),
+            Attribute("fname_c1", lev(), prior),
+            Attribute("lname_c1", lev(), prior),
+        ]
     columns = [np.where(cols[a] == "NA", None, cols[a]) for a in attr_names]
     table = RecordsTable(cols["rec_id"], cols["file_id"], columns)
-    prior = BetaShapeParameters(10.0, 1000.0)  # RLdata10000.conf:4
-    attrs = [
-        Attribute("by", ConstantSimilarityFn(), prior),
-        Attribute("bm", ConstantSimilarityFn(), prior),
-        Attribute("bd", ConstantSimilarityFn(), prior),
-        Attribute("fname_c1", LevenshteinSimilarityFn(7.0, 10.0), prior),
-        Attribute("lname_c1", LevenshteinSimilarityFn(7.0, 10.0), prior),
-    ]
     cache = RecordsCache.build(table, attrs, max_cluster_size=10)
     rec_values, rec_files = cache.transform_records(table)
     return cache, rec_values, rec_files
@@ -71,6 +86,10 @@ def main():
     ap.add_argument("--sampler", default="PCG-I")
     ap.add_argument("--seed", type=int, default=319158)
     ap.add_argument("--cpu", action="store_true", help="force the CPU engine")
+    ap.add_argument("--schema", default="rldata", choices=["rldata", "strings8"],
+                    help="rldata: 3 const + 2 Levenshtein attrs; strings8: 8 Levenshtein attrs")
+    ap.add_argument("--num-files", type=int, default=1,
+                    help="number of source files (per-file distortion probabilities)")
     args = ap.parse_args()
 
     sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
@@ -89,9 +108,12 @@ def main():
     total_partitions = args.partitions_per_gpu * n_gpus
     num_levels = max(0, int(round(math.log2(total_partitions))))
 
-    cache, rec_values, rec_files = build_cache_and_records(total_records, args.seed)
+    cache, rec_values, rec_files = build_cache_and_records(
+        total_records, args.seed, schema=args.schema, num_files=args.num_files
+    )
 
-    partitioner = KDTreePartitioner(num_levels, [3, 4])  # fname_c1, lname_c1
+    part_attrs = [3, 4] if args.schema == "rldata" else [0, 1]  # Levenshtein attrs
+    partitioner = KDTreePartitioner(num_levels, part_attrs)
     bounds = np.linspace(0, total_records, world + 1).astype(np.int64)
     lo, hi = int(bounds[rank]), int(bounds[rank + 1])
     state = deterministic_init(
@@ -153,6 +175,8 @@ def main():
                 "global_records": total_records,
                 "partitions": total_partitions,
                 "sampler": args.sampler,
+                "schema": args.schema,
+                "num_files": args.num_files,
                 "parallelism": f"entity-partitioned Gibbs, {n_gpus} rank(s) over RCCL",
                 "engine": "gpu" if use_gpu else "cpu",
             },

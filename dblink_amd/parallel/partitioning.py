@@ -239,10 +239,94 @@ class KDTreePartitioner:
 
 
 def partitioner_from_config(cfg, attribute_names):
-    """``Project.scala:219-229``: only KDTreePartitioner is supported."""
-    if cfg.get_string("name") != "KDTreePartitioner":
-        raise ValueError("unsupported partitioner: " + cfg.get_string("name"))
-    num_levels = cfg.get_int("parameters.numLevels")
-    names = cfg.get_string_list("parameters.matchingAttributes")
-    attr_ids = [list(attribute_names).index(n) for n in names]
-    return KDTreePartitioner(num_levels, attr_ids)
+    """``Project.scala:219-229``. KDTreePartitioner is the reference surface;
+    SimplePartitioner is additionally accepted (API-only in the reference)."""
+    name = cfg.get_string("name")
+    if name == "KDTreePartitioner":
+        num_levels = cfg.get_int("parameters.numLevels")
+        names = cfg.get_string_list("parameters.matchingAttributes")
+        attr_ids = [list(attribute_names).index(n) for n in names]
+        return KDTreePartitioner(num_levels, attr_ids)
+    if name == "SimplePartitioner":
+        attr = cfg.get_string("parameters.attribute")
+        return SimplePartitioner(
+            list(attribute_names).index(attr), cfg.get_int("parameters.numPartitions")
+        )
+    raise ValueError("unsupported partitioner: " + name)
+
+
+class LPTScheduler:
+    """Longest-processing-time greedy assignment of weighted jobs to
+    partitions (parity: ``partitioning/LPTScheduler.scala:38-85``)."""
+
+    def __init__(self, jobs, num_partitions):
+        """jobs: list of (job_key, weight)."""
+        if num_partitions <= 0:
+            raise ValueError("numPartitions must be positive")
+        self.num_partitions = num_partitions
+        loads = [0.0] * num_partitions
+        assignment = {}
+        for key, w in sorted(jobs, key=lambda kv: -kv[1]):
+            p = min(range(num_partitions), key=lambda i: loads[i])
+            loads[p] += w
+            assignment[key] = p
+        self.assignment = assignment
+        self.loads = loads
+
+    def partition_of(self, key):
+        return self.assignment[key]
+
+
+class SimplePartitioner:
+    """Blocks on a single attribute's value, LPT bin-packed into
+    ``num_partitions`` (parity: ``partitioning/SimplePartitioner.scala:33-64``;
+    API-only in the reference — not reachable from its config parser)."""
+
+    def __init__(self, attribute_id: int, num_partitions: int):
+        self.attribute_id = attribute_id
+        self._num_partitions = num_partitions
+        self.scheduler = None
+
+    @property
+    def num_partitions(self):
+        return self._num_partitions
+
+    def fit(self, values: np.ndarray, log=None):
+        col = np.asarray(values)[:, self.attribute_id]
+        uniq, counts = np.unique(col, return_counts=True)
+        self.scheduler = LPTScheduler(
+            list(zip(uniq.tolist(), counts.astype(float).tolist())), self._num_partitions
+        )
+        return self
+
+    def get_partition_id(self, values) -> int:
+        v = values[self.attribute_id]
+        return self.scheduler.assignment.get(int(v), 0)
+
+    def get_partition_ids(self, values: np.ndarray) -> np.ndarray:
+        col = np.asarray(values)[:, self.attribute_id]
+        out = np.zeros(col.shape[0], dtype=np.int32)
+        for i, v in enumerate(col):
+            out[i] = self.scheduler.assignment.get(int(v), 0)
+        return out
+
+    def mk_string(self):
+        return (
+            f"SimplePartitioner(attributeId={self.attribute_id}, "
+            f"numPartitions={self._num_partitions})"
+        )
+
+    def as_flat(self):
+        """Flat export: emulated as a single-level 'set' table is not possible
+        (arbitrary value -> partition map); the GPU engine detects the dense
+        map attribute instead."""
+        raise NotImplementedError(
+            "SimplePartitioner has no flat KD export; use value_map()"
+        )
+
+    def value_map(self, num_values: int) -> np.ndarray:
+        out = np.zeros(num_values, dtype=np.int32)
+        for v, p in self.scheduler.assignment.items():
+            if 0 <= v < num_values:
+                out[v] = p
+        return out

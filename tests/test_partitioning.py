@@ -91,3 +91,29 @@ def test_kdtree_deterministic():
     p1 = KDTreePartitioner(2, [0, 1]).fit(vals)
     p2 = KDTreePartitioner(2, [0, 1]).fit(vals)
     np.testing.assert_array_equal(p1.get_partition_ids(vals), p2.get_partition_ids(vals))
+
+
+def test_lpt_scheduler():
+    from dblink_amd.parallel.partitioning import LPTScheduler
+
+    jobs = [("a", 10.0), ("b", 8.0), ("c", 6.0), ("d", 5.0), ("e", 4.0), ("f", 3.0)]
+    s = LPTScheduler(jobs, 2)
+    assert abs(s.loads[0] - s.loads[1]) <= 2.0
+    assert set(s.assignment) == {"a", "b", "c", "d", "e", "f"}
+
+
+def test_simple_partitioner():
+    from dblink_amd.parallel.partitioning import SimplePartitioner
+
+    rng = np.random.default_rng(0)
+    vals = rng.integers(0, 12, size=(1000, 2)).astype(np.int32)
+    p = SimplePartitioner(1, 4).fit(vals)
+    pids = p.get_partition_ids(vals)
+    assert set(np.unique(pids)) <= set(range(4))
+    # same value always lands on the same partition (blocking invariant)
+    for v in range(12):
+        rows = vals[:, 1] == v
+        if rows.any():
+            assert len(set(pids[rows])) == 1
+    counts = np.bincount(pids, minlength=4)
+    assert counts.max() < 2.2 * counts.mean()
