@@ -207,6 +207,14 @@ class GpuEngine(CpuEngine):
     Inherits host-side pieces (theta update, summary packing/reduction,
     linkage structure) from CpuEngine; the Markov transition itself runs as
     HIP kernels.
+
+    Single-rank sweeps are captured into a hipGraph after a short warm-up:
+    the whole iteration (inverted-index sort, link/value/distortion kernels,
+    KD descent, partition re-sort, summary reduction) replays as one graph,
+    eliminating per-op launch and dispatch overhead. Kernels read
+    (seed, iteration) from a device control buffer so the captured graph
+    stays valid as the chain advances. Multi-rank runs stay eager (the
+    all-to-all exchange has data-dependent message sizes).
     """
 
     def __init__(self, cache, partitioner, world_size=1, rank=0, device=None):
@@ -221,9 +229,21 @@ class GpuEngine(CpuEngine):
         self._ent_id_base = rank << 40
         self._err = torch.zeros(1, dtype=torch.int32, device=device)
         import os
-        self.k1_row_cap = int(os.environ.get("DBLINK_K1_ROW_CAP", "48"))
+
         self.debug_classes = os.environ.get("DBLINK_DEBUG_CLASSES", "") == "1"
+        self._graphs_enabled = (
+            os.environ.get("DBLINK_GRAPHS", "1") != "0" and world_size == 1
+        )
         self._loglik_buf = torch.zeros(1, dtype=torch.float64, device=device)
+        A, F = self.model.A, self.model.F
+        self._packed = torch.zeros(2 + A * F + A + 1, dtype=torch.float64, device=device)
+        self._ctrl = torch.zeros(2, dtype=torch.int64, device=device)
+        self._ctrl_pin = torch.zeros(2, dtype=torch.int64, pin_memory=True)
+        self._theta_pin = torch.zeros((A, F), dtype=torch.float32, pin_memory=True)
+        self._empty_i64 = torch.empty(0, dtype=torch.int64, device=device)
+        self._graph = None
+        self._graph_key = None
+        self._graph_warm = 0
 
     # ---- state residency -----------------------------------------------------
 
@@ -250,19 +270,63 @@ class GpuEngine(CpuEngine):
 
     def step(self, state: ChainState, flags: SamplerFlags):
         gs = self._gpu_state(state)
-        m = self.model
 
         # theta from previous summary (host, Philox stream keyed by iteration)
         self._update_dist_probs(state)
-        m.theta.copy_(torch.from_numpy(state.dist_probs.probs).float())
+        self._theta_pin.copy_(torch.from_numpy(state.dist_probs.probs).float())
+        self._ctrl_pin[0] = state.current_seed
+        self._ctrl_pin[1] = state.iteration + 1
 
-        seed = state.current_seed
-        it = state.iteration + 1
+        key = (flags.collapsed_entity_ids, flags.collapsed_entity_values, flags.sequential)
+        if self._graphs_enabled:
+            if self._graph is not None and self._graph_key == key:
+                self._graph.replay()
+            elif self._graph_warm < 2 or self._graph_key not in (None, key):
+                if self._graph_key not in (None, key):
+                    self._graph = None
+                    self._graph_warm = 0
+                self._sweep_body(gs, flags)
+                self._graph_warm += 1
+                self._graph_key = None if self._graph is None else self._graph_key
+            else:
+                g = torch.cuda.CUDAGraph()
+                torch.cuda.synchronize()
+                with torch.cuda.graph(g):
+                    self._sweep_body(gs, flags)
+                self._graph = g
+                self._graph_key = key
+                g.replay()
+        else:
+            self._sweep_body(gs, flags)
+            if self.world_size > 1 and comm.is_distributed():
+                from ..parallel.migration import migrate_and_sort_tensors
+
+                migrate_and_sort_tensors(gs, self.world_size)
+                # rebuild static-buffer invariants after the exchange
+                self._pack_summary(gs)
+
+        state.current_seed += self.num_partitions
+        state.iteration += 1
+        state.summary = self._read_summary(state)
+        return state
+
+    def _sweep_body(self, gs: GpuStateTensors, flags: SamplerFlags):
+        """One full device-side iteration. Capture-safe for world_size == 1:
+        static input/output buffers (results copied back in place), control
+        values read from the device ctrl buffer, no host synchronisation."""
+        m = self.model
         A, E, R = m.A, gs.E, gs.R
+        dev = self.device
+        graph_safe = self._graphs_enabled
+
+        m.theta.copy_(self._theta_pin, non_blocking=True)
+        self._ctrl.copy_(self._ctrl_pin, non_blocking=True)
+        ctrl = self._ctrl
+        seed, it = 0, 0  # kernels read the ctrl buffer
 
         # --- inverted index (sorted postings) --------------------------------
         if not flags.sequential and not flags.collapsed_entity_ids:
-            arangeA = torch.arange(A, device=self.device, dtype=torch.int64).view(A, 1)
+            arangeA = torch.arange(A, device=dev, dtype=torch.int64).view(A, 1)
             keys = (
                 (gs.ent_part.to(torch.int64).view(1, E) * A + arangeA) * m.Vmax
                 + gs.ent_values.t().to(torch.int64)
@@ -271,16 +335,16 @@ class GpuEngine(CpuEngine):
             postings = (perm % E).to(torch.int32)
             qkeys = (
                 (gs.rec_part.to(torch.int64).view(R, 1) * A
-                 + torch.arange(A, device=self.device, dtype=torch.int64).view(1, A))
+                 + torch.arange(A, device=dev, dtype=torch.int64).view(1, A))
                 * m.Vmax
                 + gs.rec_values.clamp_min(0).to(torch.int64)
             ).reshape(-1)
-            cand_lo = torch.searchsorted(sorted_keys, qkeys, right=False).view(R, A)
-            cand_hi = torch.searchsorted(sorted_keys, qkeys, right=True).view(R, A)
+            cand_lo = torch.searchsorted(sorted_keys, qkeys, right=False).view(R, A).contiguous()
+            cand_hi = torch.searchsorted(sorted_keys, qkeys, right=True).view(R, A).contiguous()
 
         ent_ptr = torch.searchsorted(
             gs.ent_part.to(torch.int64).contiguous(),
-            torch.arange(self.num_partitions + 1, device=self.device, dtype=torch.int64),
+            torch.arange(self.num_partitions + 1, device=dev, dtype=torch.int64),
         )
 
         # --- phase 1: link update --------------------------------------------
@@ -290,15 +354,13 @@ class GpuEngine(CpuEngine):
                 gs.rec_values, gs.rec_dist, gs.rec_gid, gs.rec_part, gs.rec_file,
                 gs.ent_values, ent_ptr, m.theta, m.phi, m.norm_lin, m.voff,
                 m.csr_row_ptr, m.csr_col, m.csr_sim, m.attr_const,
-                1 if flags.collapsed_entity_ids else 0, seed, it, rec_ent_new,
+                1 if flags.collapsed_entity_ids else 0, seed, it, rec_ent_new, ctrl,
             )
         else:
-            # records whose smallest candidate list is short run one-per-thread
-            # (skip the classification for small problems: extra launches cost
-            # more than they save below ~100k records)
-            cand_lo = cand_lo.contiguous()
-            cand_hi = cand_hi.contiguous()
-            if R >= 100_000:
+            # records with a short smallest candidate list run one-per-thread;
+            # data-dependent list sizes are not graph-capturable, so graphs
+            # (small problems) keep the all-wave path
+            if R >= 100_000 and not graph_safe:
                 nd = (gs.rec_values >= 0) & (gs.rec_dist == 0)
                 sizes = torch.where(nd, cand_hi - cand_lo, torch.full_like(cand_hi, 1 << 40))
                 min_sizes = sizes.amin(dim=1)
@@ -307,44 +369,44 @@ class GpuEngine(CpuEngine):
                 small_list = torch.nonzero(small).squeeze(1)
                 wave_list = torch.nonzero(~small).squeeze(1)
             else:
-                small_list = wave_list = torch.empty(0, dtype=torch.int64, device=self.device)
+                small_list = wave_list = self._empty_i64
             self.C.link_update(
                 gs.rec_values, gs.rec_dist, gs.rec_gid, gs.rec_part,
                 cand_lo, cand_hi, postings, gs.ent_values,
                 ent_ptr, m.log_norm, m.voff, m.csr_row_ptr, m.csr_col, m.csr_sim,
                 m.attr_const, seed, it, rec_ent_new, gs.rec_ent, self._err,
-                small_list, wave_list,
+                small_list, wave_list, ctrl,
             )
-        gs.rec_ent = rec_ent_new
+        gs.rec_ent.copy_(rec_ent_new)
 
         # --- entity -> records CSR -------------------------------------------
         sorted_re, order = torch.sort(gs.rec_ent, stable=True)
         ent_rec_ptr = torch.searchsorted(
-            sorted_re, torch.arange(E + 1, device=self.device, dtype=torch.int64)
+            sorted_re, torch.arange(E + 1, device=dev, dtype=torch.int64)
         )
         ent_rec_idx = order
 
         # --- phase 2: value update (in place) --------------------------------
-        # split (entity, attribute) pairs: k_obs == 0 -> thread-per-pair base
-        # draw; k_obs >= 1 -> one wave each (perturbation sampling)
-        obs = (gs.rec_values >= 0)
-        kobs = torch.zeros(E * A, dtype=torch.int32, device=self.device)
+        obs = gs.rec_values >= 0
+        kobs = torch.zeros(E * A, dtype=torch.int32, device=dev)
         pair_idx = (gs.rec_ent.view(R, 1) * A
-                    + torch.arange(A, device=self.device, dtype=torch.int64).view(1, A))
-        idxm = pair_idx.reshape(-1)[obs.reshape(-1)]
-        kobs.scatter_add_(0, idxm, torch.ones_like(idxm, dtype=torch.int32))
-        base_pairs = torch.nonzero(kobs == 0).squeeze(1)
-        if R >= 100_000:
-            k1 = kobs == 1
-            k1_pairs = torch.nonzero(k1).squeeze(1)
+                    + torch.arange(A, device=dev, dtype=torch.int64).view(1, A))
+        idxm_all = pair_idx.reshape(-1)
+        kobs.scatter_add_(0, idxm_all, obs.reshape(-1).to(torch.int32))
+        if graph_safe:
+            # fixed shapes: every pair takes the wave kernel (it branches on
+            # k_obs internally)
+            base_pairs = k1_pairs = self._empty_i64
+            wave_pairs = torch.arange(E * A, device=dev, dtype=torch.int64)
+        else:
+            base_pairs = torch.nonzero(kobs == 0).squeeze(1)
+            k1_pairs = torch.nonzero(kobs == 1).squeeze(1)
             wave_pairs = torch.nonzero(kobs >= 2).squeeze(1)
             if self.debug_classes:
                 import sys
+
                 print(f"[dblink classes] base={base_pairs.numel()} k1={k1_pairs.numel()} "
                       f"wave={wave_pairs.numel()} kobs_max={kobs.max().item()}", file=sys.stderr)
-        else:
-            k1_pairs = torch.empty(0, dtype=torch.int64, device=self.device)
-            wave_pairs = torch.nonzero(kobs >= 1).squeeze(1)
         self.C.value_update(
             gs.rec_values, gs.rec_dist, gs.rec_file, ent_rec_ptr, ent_rec_idx,
             gs.ent_values, m.theta, m.phi, m.log_phi, m.norm_lin, m.log_norm,
@@ -352,14 +414,14 @@ class GpuEngine(CpuEngine):
             m.pow_prob, m.pow_alias, m.pow_off, m.log_pow_total, m.attr_const,
             m.Kc, 1 if flags.collapsed_entity_values else 0,
             1 if flags.sequential else 0, seed, it, self._ent_id_base, self._err,
-            wave_pairs, base_pairs, k1_pairs, m.csr_excl, m.csr_rawsum, m.z1,
+            wave_pairs, base_pairs, k1_pairs, m.csr_excl, m.csr_rawsum, m.z1, ctrl,
         )
 
         # --- phase 3: distortion update --------------------------------------
         self.C.distortion_update(
             gs.rec_values, gs.rec_dist, gs.rec_file, gs.rec_gid, gs.rec_ent,
             gs.ent_values, m.theta, m.phi, m.norm_lin, m.self_expsim, m.voff,
-            m.attr_const, seed, it,
+            m.attr_const, seed, it, ctrl,
         )
 
         # --- partition reassignment ------------------------------------------
@@ -369,27 +431,31 @@ class GpuEngine(CpuEngine):
             self.flat_tree["a"], self.flat_tree["b"], self.flat_tree["rset"],
             ent_part_new,
         )
-        gs.ent_part = ent_part_new
+        gs.ent_part.copy_(ent_part_new)
 
-        state.current_seed += self.num_partitions
-        state.iteration += 1
+        if self.world_size <= 1:
+            self._local_sort_static(gs)
+            self._pack_summary(gs)
 
-        # --- migration + local re-sort ---------------------------------------
-        self._migrate_and_sort(gs)
+    def _local_sort_static(self, gs: GpuStateTensors):
+        """Re-sort by partition id into the SAME (static) buffers."""
+        order = torch.argsort(gs.ent_part.to(torch.int64), stable=True)
+        inv = torch.empty_like(order)
+        inv[order] = torch.arange(order.numel(), device=self.device)
+        gs.ent_values.copy_(gs.ent_values[order])
+        gs.ent_part.copy_(gs.ent_part[order])
+        new_rec_ent = inv[gs.rec_ent]
+        rorder = torch.argsort(new_rec_ent, stable=True)
+        gs.rec_ent.copy_(new_rec_ent[rorder])
+        gs.rec_values.copy_(gs.rec_values[rorder])
+        gs.rec_file.copy_(gs.rec_file[rorder])
+        gs.rec_dist.copy_(gs.rec_dist[rorder])
+        gs.rec_gid.copy_(gs.rec_gid[rorder])
+        gs.rec_part.copy_(gs.ent_part[gs.rec_ent])
 
-        # --- summary ----------------------------------------------------------
-        summary = self._summary_device(gs, state)
-        state.summary = summary
-        return state
+    # ---- summary -------------------------------------------------------------
 
-    # ---- helpers -------------------------------------------------------------
-
-    def _migrate_and_sort(self, gs: GpuStateTensors):
-        from ..parallel.migration import migrate_and_sort_tensors
-
-        migrate_and_sort_tensors(gs, self.world_size)
-
-    def _summary_device(self, gs: GpuStateTensors, state: ChainState) -> SummaryVars:
+    def _pack_summary(self, gs: GpuStateTensors):
         m = self.model
         A, F = m.A, m.F
         E, R = gs.E, gs.R
@@ -412,34 +478,39 @@ class GpuEngine(CpuEngine):
         ndist = dist.sum(dim=1)
         hist = torch.zeros(A + 1, dtype=torch.int64, device=self.device)
         hist.scatter_add_(0, ndist, torch.ones_like(ndist))
+        self._packed[0:1].copy_(self._loglik_buf)
+        self._packed[1] = isolates.to(torch.float64)
+        self._packed[2 : 2 + A * F].copy_(agg.to(torch.float64))
+        self._packed[2 + A * F :].copy_(hist.to(torch.float64))
 
-        packed = torch.cat(
-            [
-                self._loglik_buf,
-                isolates.to(torch.float64).view(1),
-                agg.to(torch.float64),
-                hist.to(torch.float64),
-            ]
-        )
-        comm.all_reduce_sum_(packed)
+    def _read_summary(self, state: ChainState) -> SummaryVars:
+        from .cpu_engine import add_prior_terms
+
+        m = self.model
+        A, F = m.A, m.F
+        packed = self._packed
+        if self.world_size > 1 and comm.is_distributed():
+            packed = packed.clone()
+            comm.all_reduce_sum_(packed)
         host = packed.cpu().numpy()
         err = int(self._err.cpu())
         if err:
-            raise RuntimeError(f"{err} empty candidate sets in link update (invariant violated)")
+            raise RuntimeError(
+                f"{err} empty candidate sets in link update (invariant violated)"
+            )
         out = SummaryVars(
             num_isolates=int(round(host[1])),
             log_likelihood=float(host[0]),
             agg_distortions=host[2 : 2 + A * F].reshape(A, F).astype(np.int64),
             rec_distortions=host[2 + A * F :].astype(np.int64),
         )
-        from .cpu_engine import add_prior_terms
-
         add_prior_terms(out, self.cache, state.dist_probs)
         return out
 
     def initial_summary(self, state: ChainState):
         gs = self._gpu_state(state)
-        state.summary = self._summary_device(gs, state)
+        self._pack_summary(gs)
+        state.summary = self._read_summary(state)
 
     def linkage_structure(self, state: ChainState, rec_id_of=None):
         self.sync_state(state)
@@ -449,7 +520,6 @@ class GpuEngine(CpuEngine):
 def cache_kc(cache, partitioner):
     """How many power distributions to pre-cache per attribute (the
     reference precaches 1..expectedMaxClusterSize, RecordsCache.scala:112)."""
-    # the cache was built with precache_powers = expectedMaxClusterSize
     for ia in cache.indexed_attributes:
         if not ia.is_constant:
             return max(ia.index._max_cached_power, 1)

@@ -20,7 +20,8 @@ void link_update(torch::Tensor rec_values, torch::Tensor rec_dist,
                  torch::Tensor csr_sim, torch::Tensor attr_const, int64_t seed,
                  int64_t iteration, torch::Tensor rec_ent_out,
                  torch::Tensor rec_ent_in, torch::Tensor error_count,
-                 torch::Tensor small_list, torch::Tensor wave_list);
+                 torch::Tensor small_list, torch::Tensor wave_list,
+                 torch::Tensor ctrl);
 void link_update_dense(torch::Tensor rec_values, torch::Tensor rec_dist,
                        torch::Tensor rec_gid, torch::Tensor rec_part,
                        torch::Tensor rec_file, torch::Tensor ent_values,
@@ -29,7 +30,7 @@ void link_update_dense(torch::Tensor rec_values, torch::Tensor rec_dist,
                        torch::Tensor csr_row_ptr, torch::Tensor csr_col,
                        torch::Tensor csr_sim, torch::Tensor attr_const,
                        int64_t collapsed, int64_t seed, int64_t iteration,
-                       torch::Tensor rec_ent_out);
+                       torch::Tensor rec_ent_out, torch::Tensor ctrl);
 void value_update(torch::Tensor rec_values, torch::Tensor rec_dist,
                   torch::Tensor rec_file, torch::Tensor ent_rec_ptr,
                   torch::Tensor ent_rec_idx, torch::Tensor ent_values,
@@ -44,14 +45,14 @@ void value_update(torch::Tensor rec_values, torch::Tensor rec_dist,
                   int64_t iteration, int64_t ent_id_base, torch::Tensor error_count,
                   torch::Tensor wave_pairs, torch::Tensor base_pairs,
                   torch::Tensor k1_pairs, torch::Tensor csr_excl,
-                  torch::Tensor csr_rawsum, torch::Tensor z1);
+                  torch::Tensor csr_rawsum, torch::Tensor z1, torch::Tensor ctrl);
 void distortion_update(torch::Tensor rec_values, torch::Tensor rec_dist,
                        torch::Tensor rec_file, torch::Tensor rec_gid,
                        torch::Tensor rec_ent, torch::Tensor ent_values,
                        torch::Tensor theta, torch::Tensor phi,
                        torch::Tensor norm_lin, torch::Tensor self_expsim,
                        torch::Tensor voff, torch::Tensor attr_const, int64_t seed,
-                       int64_t iteration);
+                       int64_t iteration, torch::Tensor ctrl);
 void summary_loglik(torch::Tensor ent_values, torch::Tensor rec_values,
                     torch::Tensor rec_dist, torch::Tensor rec_ent,
                     torch::Tensor log_phi, torch::Tensor log_norm, torch::Tensor voff,

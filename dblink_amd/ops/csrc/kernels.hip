@@ -56,13 +56,14 @@ __global__ void link_update_kernel(
     const uint8_t* __restrict__ attr_const,  // [A]
     const int64_t* __restrict__ rec_list,    // [R] records to process (or null)
     int64_t R, int A,
-    uint64_t seed, uint32_t iteration,
+    uint64_t seed, uint32_t iteration, const int64_t* __restrict__ ctrl,
     int64_t* __restrict__ rec_ent_out,       // [R]
     const int64_t* __restrict__ rec_ent_in,  // [R]
     int* __restrict__ error_count) {
   const int lane = threadIdx.x & (WAVE - 1);
   int64_t r = (int64_t)blockIdx.x * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
   if (r >= R) return;
+  if (ctrl != nullptr) { seed = (uint64_t)ctrl[0]; iteration = (uint32_t)ctrl[1]; }
   if (rec_list != nullptr) r = rec_list[r];
 
   // Gather per-attribute candidate ranges (observed non-distorted) and the
@@ -157,10 +158,12 @@ __global__ void link_update_small_kernel(
     const int64_t* __restrict__ voff, const int64_t* __restrict__ csr_row_ptr,
     const int32_t* __restrict__ csr_col, const float* __restrict__ csr_sim,
     const uint8_t* __restrict__ attr_const, int A, uint64_t seed,
-    uint32_t iteration, int64_t* __restrict__ rec_ent_out,
+    uint32_t iteration, const int64_t* __restrict__ ctrl,
+    int64_t* __restrict__ rec_ent_out,
     const int64_t* __restrict__ rec_ent_in, int* __restrict__ error_count) {
   const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (idx >= n_recs) return;
+  if (ctrl != nullptr) { seed = (uint64_t)ctrl[0]; iteration = (uint32_t)ctrl[1]; }
   const int64_t r = rec_list[idx];
 
   int nd_n = 0, od_n = 0;
@@ -226,11 +229,12 @@ __global__ void link_update_dense_kernel(
     const int32_t* __restrict__ csr_col, const float* __restrict__ csr_sim,
     const uint8_t* __restrict__ attr_const, int64_t R, int A, int F,
     int collapsed,  // 1 = PCG-II weights, 0 = Gibbs-Sequential weights
-    uint64_t seed, uint32_t iteration,
+    uint64_t seed, uint32_t iteration, const int64_t* __restrict__ ctrl,
     int64_t* __restrict__ rec_ent_out) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int64_t r = (int64_t)blockIdx.x * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
   if (r >= R) return;
+  if (ctrl != nullptr) { seed = (uint64_t)ctrl[0]; iteration = (uint32_t)ctrl[1]; }
   const int32_t p = rec_part[r];
   const int64_t e0 = ent_ptr[p], e1 = ent_ptr[p + 1];
   const int f = rec_file[r];
@@ -282,6 +286,7 @@ constexpr int HASH_CAP = 1024;            // slots per wave
 constexpr int WAVES_PER_BLOCK_VAL = 4;    // 256 threads
 
 struct ValueArgs {
+  const int64_t* ctrl;         // [2] = {seed, iteration} device override (or null)
   const int64_t* pair_list;    // [n_pairs] flattened (e*A + a) needing wave work
   int64_t n_pairs;
   const double* csr_excl;      // [nnz] exclusive row prefix of raw k=1 weights
@@ -343,6 +348,10 @@ __device__ int dense_power_draw(const ValueArgs& args, int a, int k, uint64_t el
 // Thread-per-pair base draws for empty clusters (k_obs == 0): the base
 // distribution is phi for every variant (GibbsUpdates.scala:584-588).
 __global__ void value_base_draw_kernel(ValueArgs args) {
+  if (args.ctrl != nullptr) {
+    args.seed = (uint64_t)args.ctrl[0];
+    args.iteration = (uint32_t)args.ctrl[1];
+  }
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= args.n_pairs) return;
   const int64_t pair = args.pair_list[i];
@@ -365,6 +374,10 @@ __global__ void value_base_draw_kernel(ValueArgs args) {
 // row totals, a draw is: one Philox call, a mixture test, and one binary
 // search over the row prefix — O(log row) regardless of row size.
 __global__ void value_update_k1_kernel(ValueArgs args) {
+  if (args.ctrl != nullptr) {
+    args.seed = (uint64_t)args.ctrl[0];
+    args.iteration = (uint32_t)args.ctrl[1];
+  }
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= args.n_pairs) return;
   const int64_t pair = args.pair_list[i];
@@ -453,6 +466,10 @@ __global__ void value_update_k1_kernel(ValueArgs args) {
 
 __global__ void __launch_bounds__(WAVES_PER_BLOCK_VAL * WAVE)
 value_update_kernel(ValueArgs args) {
+  if (args.ctrl != nullptr) {
+    args.seed = (uint64_t)args.ctrl[0];
+    args.iteration = (uint32_t)args.ctrl[1];
+  }
   __shared__ int32_t h_key[WAVES_PER_BLOCK_VAL][HASH_CAP];
   __shared__ float h_val[WAVES_PER_BLOCK_VAL][HASH_CAP];
 
@@ -755,6 +772,10 @@ value_update_kernel(ValueArgs args) {
 
 // Brute-force dense value update (Gibbs-Sequential, GibbsUpdates.scala:652-698)
 __global__ void value_update_seq_kernel(ValueArgs args) {
+  if (args.ctrl != nullptr) {
+    args.seed = (uint64_t)args.ctrl[0];
+    args.iteration = (uint32_t)args.ctrl[1];
+  }
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave = threadIdx.x / WAVE;
   const int64_t pair = (int64_t)blockIdx.x * (blockDim.x / WAVE) + wave;
@@ -818,9 +839,11 @@ __global__ void distortion_update_kernel(
     const float* __restrict__ theta, const float* __restrict__ phi,
     const float* __restrict__ norm_lin, const float* __restrict__ self_expsim,
     const int64_t* __restrict__ voff, const uint8_t* __restrict__ attr_const,
-    int64_t R, int A, int F, uint64_t seed, uint32_t iteration) {
+    int64_t R, int A, int F, uint64_t seed, uint32_t iteration,
+    const int64_t* __restrict__ ctrl) {
   const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (idx >= R * A) return;
+  if (ctrl != nullptr) { seed = (uint64_t)ctrl[0]; iteration = (uint32_t)ctrl[1]; }
   const int64_t r = idx / A;
   const int a = (int)(idx % A);
   const int32_t x = rec_values[idx];
@@ -1041,7 +1064,8 @@ void link_update(
     torch::Tensor csr_col, torch::Tensor csr_sim, torch::Tensor attr_const,
     int64_t seed, int64_t iteration, torch::Tensor rec_ent_out,
     torch::Tensor rec_ent_in, torch::Tensor error_count,
-    torch::Tensor small_list, torch::Tensor wave_list) {
+    torch::Tensor small_list, torch::Tensor wave_list, torch::Tensor ctrl) {
+  const int64_t* ctrl_ptr = ctrl.numel() ? ctrl.data_ptr<int64_t>() : nullptr;
   CHECK_GPU(rec_values);
   const int64_t R = rec_values.size(0);
   const int A = (int)rec_values.size(1);
@@ -1063,7 +1087,7 @@ void link_update(
                        csr_col.data_ptr<int32_t>(), csr_sim.data_ptr<float>(),
                        attr_const.data_ptr<uint8_t>(),
                        split ? wave_list.data_ptr<int64_t>() : nullptr, n, A,
-                       (uint64_t)seed, (uint32_t)iteration,
+                       (uint64_t)seed, (uint32_t)iteration, ctrl_ptr,
                        rec_ent_out.data_ptr<int64_t>(),
                        rec_ent_in.data_ptr<int64_t>(), error_count.data_ptr<int>());
   }
@@ -1080,7 +1104,8 @@ void link_update(
                        voff.data_ptr<int64_t>(), csr_row_ptr.data_ptr<int64_t>(),
                        csr_col.data_ptr<int32_t>(), csr_sim.data_ptr<float>(),
                        attr_const.data_ptr<uint8_t>(), A, (uint64_t)seed,
-                       (uint32_t)iteration, rec_ent_out.data_ptr<int64_t>(),
+                       (uint32_t)iteration, ctrl_ptr,
+                       rec_ent_out.data_ptr<int64_t>(),
                        rec_ent_in.data_ptr<int64_t>(), error_count.data_ptr<int>());
   }
 }
@@ -1091,7 +1116,9 @@ void link_update_dense(
     torch::Tensor ent_ptr, torch::Tensor theta, torch::Tensor phi,
     torch::Tensor norm_lin, torch::Tensor voff, torch::Tensor csr_row_ptr,
     torch::Tensor csr_col, torch::Tensor csr_sim, torch::Tensor attr_const,
-    int64_t collapsed, int64_t seed, int64_t iteration, torch::Tensor rec_ent_out) {
+    int64_t collapsed, int64_t seed, int64_t iteration, torch::Tensor rec_ent_out,
+    torch::Tensor ctrl) {
+  const int64_t* ctrl_ptr = ctrl.numel() ? ctrl.data_ptr<int64_t>() : nullptr;
   const int64_t R = rec_values.size(0);
   const int A = (int)rec_values.size(1);
   const int F = (int)theta.size(1);
@@ -1108,7 +1135,7 @@ void link_update_dense(
                      voff.data_ptr<int64_t>(), csr_row_ptr.data_ptr<int64_t>(),
                      csr_col.data_ptr<int32_t>(), csr_sim.data_ptr<float>(),
                      attr_const.data_ptr<uint8_t>(), R, A, F, (int)collapsed,
-                     (uint64_t)seed, (uint32_t)iteration,
+                     (uint64_t)seed, (uint32_t)iteration, ctrl_ptr,
                      rec_ent_out.data_ptr<int64_t>());
 }
 
@@ -1124,6 +1151,7 @@ static ValueArgs make_value_args(
     int64_t iteration, int64_t ent_id_base, torch::Tensor error_count,
     torch::Tensor csr_excl, torch::Tensor csr_rawsum, torch::Tensor z1) {
   ValueArgs a;
+  a.ctrl = nullptr;
   a.csr_excl = csr_excl.numel() ? csr_excl.data_ptr<double>() : nullptr;
   a.csr_rawsum = csr_rawsum.data_ptr<double>();
   a.z1 = z1.data_ptr<double>();
@@ -1172,13 +1200,15 @@ void value_update(
     torch::Tensor attr_const, int64_t Kc, int64_t collapsed, int64_t sequential,
     int64_t seed, int64_t iteration, int64_t ent_id_base, torch::Tensor error_count,
     torch::Tensor wave_pairs, torch::Tensor base_pairs, torch::Tensor k1_pairs,
-    torch::Tensor csr_excl, torch::Tensor csr_rawsum, torch::Tensor z1) {
+    torch::Tensor csr_excl, torch::Tensor csr_rawsum, torch::Tensor z1,
+    torch::Tensor ctrl) {
   ValueArgs args = make_value_args(
       rec_values, rec_dist, rec_file, ent_rec_ptr, ent_rec_idx, ent_values, theta,
       phi, log_phi, norm_lin, log_norm, voff, csr_row_ptr, csr_col, csr_sim,
       phi_prob, phi_alias, pow_prob, pow_alias, pow_off, log_pow_total, attr_const,
       Kc, collapsed, seed, iteration, ent_id_base, error_count, csr_excl,
       csr_rawsum, z1);
+  args.ctrl = ctrl.numel() ? ctrl.data_ptr<int64_t>() : nullptr;
   if (sequential) {
     const int64_t pairs = args.E * args.A;
     if (pairs == 0) return;
@@ -1220,7 +1250,8 @@ void distortion_update(
     torch::Tensor rec_gid, torch::Tensor rec_ent, torch::Tensor ent_values,
     torch::Tensor theta, torch::Tensor phi, torch::Tensor norm_lin,
     torch::Tensor self_expsim, torch::Tensor voff, torch::Tensor attr_const,
-    int64_t seed, int64_t iteration) {
+    int64_t seed, int64_t iteration, torch::Tensor ctrl) {
+  const int64_t* ctrl_ptr = ctrl.numel() ? ctrl.data_ptr<int64_t>() : nullptr;
   const int64_t R = rec_values.size(0);
   const int A = (int)rec_values.size(1);
   const int F = (int)theta.size(1);
@@ -1235,7 +1266,7 @@ void distortion_update(
                      theta.data_ptr<float>(), phi.data_ptr<float>(),
                      norm_lin.data_ptr<float>(), self_expsim.data_ptr<float>(),
                      voff.data_ptr<int64_t>(), attr_const.data_ptr<uint8_t>(),
-                     R, A, F, (uint64_t)seed, (uint32_t)iteration);
+                     R, A, F, (uint64_t)seed, (uint32_t)iteration, ctrl_ptr);
 }
 
 void summary_loglik(

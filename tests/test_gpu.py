@@ -132,7 +132,7 @@ def test_distortion_kernel_distribution():
         _dev(np.arange(N, dtype=np.int64), torch.int64),
         _dev(rec_ent, torch.int64), _dev(ent_values, torch.int32),
         model.theta, model.phi, model.norm_lin, model.self_expsim, model.voff,
-        model.attr_const, 1234, 7,
+        model.attr_const, 1234, 7, torch.empty(0, dtype=torch.int64, device=DEV),
     )
     z = d.cpu().numpy()
     # attr 0 missing: P(z=1) = theta
@@ -151,7 +151,7 @@ def test_distortion_kernel_distribution():
         _dev(np.arange(N, dtype=np.int64), torch.int64),
         _dev(rec_ent, torch.int64), _dev(ent_values, torch.int32),
         model.theta, model.phi, model.norm_lin, model.self_expsim, model.voff,
-        model.attr_const, 99, 3,
+        model.attr_const, 99, 3, torch.empty(0, dtype=torch.int64, device=DEV),
     )
     assert d.cpu().numpy()[:, 1].min() == 1
 
@@ -205,6 +205,7 @@ def test_link_kernel_distribution():
         model.log_norm, model.voff, model.csr_row_ptr, model.csr_col, model.csr_sim,
         model.attr_const, 4321, 11, out,
         _dev(np.zeros(N, np.int64), torch.int64), err,
+        torch.empty(0, dtype=torch.int64, device=DEV),
         torch.empty(0, dtype=torch.int64, device=DEV),
         torch.empty(0, dtype=torch.int64, device=DEV),
     )
@@ -266,6 +267,7 @@ def test_value_kernel_distribution():
         torch.empty(0, dtype=torch.int64, device=DEV),
         torch.empty(0, dtype=torch.int64, device=DEV),
         model.csr_excl, model.csr_rawsum, model.z1,
+        torch.empty(0, dtype=torch.int64, device=DEV),
     )
     got = ev.cpu().numpy()[:, a]
     emp = np.bincount(got, minlength=V) / E
@@ -286,6 +288,7 @@ def test_value_kernel_distribution():
         torch.empty(0, dtype=torch.int64, device=DEV),
         torch.arange(E * 2, dtype=torch.int64, device=DEV),
         model.csr_excl, model.csr_rawsum, model.z1,
+        torch.empty(0, dtype=torch.int64, device=DEV),
     )
     got2 = ev2.cpu().numpy()[:, a]
     emp2 = np.bincount(got2, minlength=V) / E
