@@ -278,7 +278,10 @@ class GpuEngine(CpuEngine):
         self._ctrl_pin[1] = state.iteration + 1
 
         key = (flags.collapsed_entity_ids, flags.collapsed_entity_values, flags.sequential)
-        if self._graphs_enabled:
+        # graphs pay off where launch overhead dominates; big problems prefer
+        # the eager path with data-dependent kernel-path classification
+        use_graphs = self._graphs_enabled and gs.R < 100_000
+        if use_graphs:
             if self._graph is not None and self._graph_key == key:
                 self._graph.replay()
             elif self._graph_warm < 2 or self._graph_key not in (None, key):
@@ -297,7 +300,8 @@ class GpuEngine(CpuEngine):
                 self._graph_key = key
                 g.replay()
         else:
-            self._sweep_body(gs, flags)
+            self._graph = None
+            self._sweep_body(gs, flags, graph_safe=False)
             if self.world_size > 1 and comm.is_distributed():
                 from ..parallel.migration import migrate_and_sort_tensors
 
@@ -310,14 +314,13 @@ class GpuEngine(CpuEngine):
         state.summary = self._read_summary(state)
         return state
 
-    def _sweep_body(self, gs: GpuStateTensors, flags: SamplerFlags):
+    def _sweep_body(self, gs: GpuStateTensors, flags: SamplerFlags, graph_safe=True):
         """One full device-side iteration. Capture-safe for world_size == 1:
         static input/output buffers (results copied back in place), control
         values read from the device ctrl buffer, no host synchronisation."""
         m = self.model
         A, E, R = m.A, gs.E, gs.R
         dev = self.device
-        graph_safe = self._graphs_enabled
 
         m.theta.copy_(self._theta_pin, non_blocking=True)
         self._ctrl.copy_(self._ctrl_pin, non_blocking=True)

@@ -94,46 +94,76 @@ def migrate_and_sort_tensors(gs, world_size: int):
     ``gs`` carries torch tensors: ent_values, ent_part, rec_values, rec_file,
     rec_dist, rec_gid, rec_ent (+ rec_part, rebuilt here). Works on CPU
     (gloo tests) and GPU (RCCL all-to-all over xGMI) alike.
+
+    Payloads are packed into two int32 matrices (entities: values|part|nrec;
+    records: values|file|dist-bitmask|gid_lo|gid_hi) so a migration costs one
+    counts exchange + two all-to-all-v calls. Records of one source arrive
+    grouped behind their entity in order, so rec_ent is rebuilt from the
+    per-entity record counts.
     """
     import torch
 
     device = gs.ent_part.device
+    A = gs.ent_values.shape[1]
+    assert A <= 16, "distortion bitmask packing supports at most 16 attributes"
     if world_size > 1 and comm.is_distributed():
         dest_e = gs.ent_part.to(torch.int64) % world_size
         order_e = torch.argsort(dest_e, stable=True)
+        send_e = torch.bincount(dest_e, minlength=world_size)
         inv_e = torch.empty_like(order_e)
         inv_e[order_e] = torch.arange(order_e.numel(), device=device)
-        send_e = torch.bincount(dest_e, minlength=world_size)
-        dest_r = dest_e[gs.rec_ent]
-        order_r = torch.argsort(dest_r, stable=True)
-        send_r = torch.bincount(dest_r, minlength=world_size)
         base = torch.cumsum(
             torch.cat([torch.zeros(1, dtype=torch.int64, device=device), send_e[:-1]]), 0
         )
-        pos_in_dest = inv_e - base[dest_e]
+        pos_in_dest = inv_e - base[dest_e]  # entity position within its dest block
+        dest_r = dest_e[gs.rec_ent]
+        # group records behind their entity's position so receivers can rebuild
+        # rec_ent from per-entity counts (rec_ent need not be sorted here)
+        E_all = gs.ent_values.shape[0]
+        order_r = torch.argsort(dest_r * max(E_all, 1) + pos_in_dest[gs.rec_ent], stable=True)
+        send_r = torch.bincount(dest_r, minlength=world_size)
+
+        E, R = gs.ent_values.shape[0], gs.rec_values.shape[0]
+        nrec = torch.zeros(E, dtype=torch.int64, device=device)
+        nrec.scatter_add_(0, gs.rec_ent, torch.ones_like(gs.rec_ent))
+
+        ent_pack = torch.empty((E, A + 2), dtype=torch.int32, device=device)
+        ent_pack[:, :A] = gs.ent_values[order_e]
+        ent_pack[:, A] = gs.ent_part[order_e]
+        ent_pack[:, A + 1] = nrec[order_e].to(torch.int32)
+
+        weights = (1 << torch.arange(A, device=device, dtype=torch.int32))
+        distbits = (gs.rec_dist.to(torch.int32) * weights.view(1, A)).sum(dim=1)
+        rec_pack = torch.empty((R, A + 4), dtype=torch.int32, device=device)
+        rec_pack[:, :A] = gs.rec_values[order_r]
+        rec_pack[:, A] = gs.rec_file[order_r]
+        rec_pack[:, A + 1] = distbits[order_r]
+        gid = gs.rec_gid[order_r]
+        rec_pack[:, A + 2] = (gid & 0xFFFFFFFF).to(torch.int32)
+        rec_pack[:, A + 3] = (gid >> 32).to(torch.int32)
 
         send_e_l = [int(x) for x in send_e.cpu()]
         send_r_l = [int(x) for x in send_r.cpu()]
+        new_ep, recv_e = comm.all_to_all_v(ent_pack.contiguous(), send_e_l)
+        new_rp, recv_r = comm.all_to_all_v(rec_pack.contiguous(), send_r_l)
 
-        new_ev, recv_e = comm.all_to_all_v(gs.ent_values[order_e].contiguous(), send_e_l)
-        new_ep, _ = comm.all_to_all_v(gs.ent_part[order_e].contiguous(), send_e_l)
-        rel = pos_in_dest[gs.rec_ent][order_r].contiguous()
-        new_rel, recv_r = comm.all_to_all_v(rel, send_r_l)
-        new_rv, _ = comm.all_to_all_v(gs.rec_values[order_r].contiguous(), send_r_l)
-        new_rf, _ = comm.all_to_all_v(gs.rec_file[order_r].contiguous(), send_r_l)
-        new_rd, _ = comm.all_to_all_v(gs.rec_dist[order_r].contiguous(), send_r_l)
-        new_rg, _ = comm.all_to_all_v(gs.rec_gid[order_r].contiguous(), send_r_l)
-
-        eoff = np.concatenate([[0], np.cumsum(recv_e)])
-        roff = np.concatenate([[0], np.cumsum(recv_r)])
-        new_re = torch.empty(int(roff[-1]), dtype=torch.int64, device=device)
-        for s in range(world_size):
-            new_re[int(roff[s]) : int(roff[s + 1])] = (
-                new_rel[int(roff[s]) : int(roff[s + 1])] + int(eoff[s])
-            )
-        gs.ent_values, gs.ent_part = new_ev, new_ep
-        gs.rec_values, gs.rec_file, gs.rec_dist, gs.rec_gid = new_rv, new_rf, new_rd, new_rg
-        gs.rec_ent = new_re
+        gs.ent_values = new_ep[:, :A].contiguous()
+        gs.ent_part = new_ep[:, A].contiguous()
+        new_nrec = new_ep[:, A + 1].to(torch.int64)
+        gs.rec_values = new_rp[:, :A].contiguous()
+        gs.rec_file = new_rp[:, A].contiguous()
+        db = new_rp[:, A + 1].to(torch.int32)
+        gs.rec_dist = (
+            (db.view(-1, 1) >> torch.arange(A, device=device, dtype=torch.int32).view(1, A)) & 1
+        ).to(torch.uint8).contiguous()
+        lo = new_rp[:, A + 2].to(torch.int64) & 0xFFFFFFFF
+        hi = new_rp[:, A + 3].to(torch.int64)
+        gs.rec_gid = (lo | (hi << 32)).contiguous()
+        # records from each source arrive grouped behind their entities in
+        # entity order, and entity blocks concatenate in source order
+        gs.rec_ent = torch.repeat_interleave(
+            torch.arange(new_ep.shape[0], device=device, dtype=torch.int64), new_nrec
+        )
 
     # local re-sort by partition id (stable)
     order = torch.argsort(gs.ent_part.to(torch.int64), stable=True)
