@@ -108,9 +108,14 @@ def main():
     total_partitions = args.partitions_per_gpu * n_gpus
     num_levels = max(0, int(round(math.log2(total_partitions))))
 
+    t_init = time.time()
     cache, rec_values, rec_files = build_cache_and_records(
         total_records, args.seed, schema=args.schema, num_files=args.num_files
     )
+    if rank == 0:
+        print(f"[bench] data+cache built in {time.time() - t_init:.1f}s "
+              f"(V = {[ia.index.num_values for ia in cache.indexed_attributes]})",
+              file=sys.stderr)
 
     part_attrs = [3, 4] if args.schema == "rldata" else [0, 1]  # Levenshtein attrs
     partitioner = KDTreePartitioner(num_levels, part_attrs)
@@ -121,6 +126,8 @@ def main():
         cache, partitioner, args.seed, rank=rank, world_size=world,
     )
 
+    if rank == 0:
+        print(f"[bench] init state in {time.time() - t_init:.1f}s total", file=sys.stderr)
     if use_gpu:
         from dblink_amd.engine.gpu_engine import GpuEngine
 
