@@ -241,6 +241,7 @@ class GpuEngine(CpuEngine):
         self._ctrl_pin = torch.zeros(2, dtype=torch.int64, pin_memory=True)
         self._theta_pin = torch.zeros((A, F), dtype=torch.float32, pin_memory=True)
         self._empty_i64 = torch.empty(0, dtype=torch.int64, device=device)
+        self._counts = torch.zeros(1 + A * F + A + 1, dtype=torch.int64, device=device)
         self._graph = None
         self._graph_key = None
         self._graph_warm = 0
@@ -329,19 +330,12 @@ class GpuEngine(CpuEngine):
 
         # --- inverted index (sorted postings) --------------------------------
         if not flags.sequential and not flags.collapsed_entity_ids:
-            arangeA = torch.arange(A, device=dev, dtype=torch.int64).view(A, 1)
-            keys = (
-                (gs.ent_part.to(torch.int64).view(1, E) * A + arangeA) * m.Vmax
-                + gs.ent_values.t().to(torch.int64)
-            ).reshape(-1)
+            keys = torch.empty(A * E, dtype=torch.int64, device=dev)
+            qkeys = torch.empty(R * A, dtype=torch.int64, device=dev)
+            self.C.build_keys(gs.ent_part, gs.ent_values, gs.rec_part, gs.rec_values,
+                              m.Vmax, keys, qkeys)
             sorted_keys, perm = torch.sort(keys, stable=True)
             postings = (perm % E).to(torch.int32)
-            qkeys = (
-                (gs.rec_part.to(torch.int64).view(R, 1) * A
-                 + torch.arange(A, device=dev, dtype=torch.int64).view(1, A))
-                * m.Vmax
-                + gs.rec_values.clamp_min(0).to(torch.int64)
-            ).reshape(-1)
             cand_lo = torch.searchsorted(sorted_keys, qkeys, right=False).view(R, A).contiguous()
             cand_hi = torch.searchsorted(sorted_keys, qkeys, right=True).view(R, A).contiguous()
 
@@ -438,7 +432,9 @@ class GpuEngine(CpuEngine):
 
         if self.world_size <= 1:
             self._local_sort_static(gs)
-            self._pack_summary(gs)
+            # isolate COUNT is permutation-invariant, so the pre-sort CSR is
+            # still valid for the summary counts
+            self._pack_summary(gs, ent_rec_ptr=ent_rec_ptr)
 
     def _local_sort_static(self, gs: GpuStateTensors):
         """Re-sort by partition id into the SAME (static) buffers."""
@@ -458,33 +454,25 @@ class GpuEngine(CpuEngine):
 
     # ---- summary -------------------------------------------------------------
 
-    def _pack_summary(self, gs: GpuStateTensors):
+    def _pack_summary(self, gs: GpuStateTensors, ent_rec_ptr=None):
         m = self.model
-        A, F = m.A, m.F
-        E, R = gs.E, gs.R
+        E = gs.E
         self._loglik_buf.zero_()
+        self._counts.zero_()
         self.C.summary_loglik(
             gs.ent_values, gs.rec_values, gs.rec_dist, gs.rec_ent, m.log_phi,
             m.log_norm, m.voff, m.csr_row_ptr, m.csr_col, m.csr_sim, m.attr_const,
             self._loglik_buf,
         )
-        linked = torch.zeros(E, dtype=torch.int64, device=self.device)
-        linked.scatter_add_(0, gs.rec_ent, torch.ones_like(gs.rec_ent))
-        isolates = (linked == 0).sum()
-        dist = gs.rec_dist.to(torch.int64)
-        agg_idx = (
-            torch.arange(A, device=self.device, dtype=torch.int64).view(1, A) * F
-            + gs.rec_file.to(torch.int64).view(R, 1)
-        ).reshape(-1)
-        agg = torch.zeros(A * F, dtype=torch.int64, device=self.device)
-        agg.scatter_add_(0, agg_idx, dist.reshape(-1))
-        ndist = dist.sum(dim=1)
-        hist = torch.zeros(A + 1, dtype=torch.int64, device=self.device)
-        hist.scatter_add_(0, ndist, torch.ones_like(ndist))
-        self._packed[0:1].copy_(self._loglik_buf)
-        self._packed[1] = isolates.to(torch.float64)
-        self._packed[2 : 2 + A * F].copy_(agg.to(torch.float64))
-        self._packed[2 + A * F :].copy_(hist.to(torch.float64))
+        if ent_rec_ptr is None:
+            sorted_re, _ = torch.sort(gs.rec_ent)
+            ent_rec_ptr = torch.searchsorted(
+                sorted_re, torch.arange(E + 1, device=self.device, dtype=torch.int64)
+            )
+        self.C.summary_counts(
+            gs.rec_dist, gs.rec_file, ent_rec_ptr, E, self._counts,
+            self._loglik_buf, self._packed,
+        )
 
     def _read_summary(self, state: ChainState) -> SummaryVars:
         from .cpu_engine import add_prior_terms

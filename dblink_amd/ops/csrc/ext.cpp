@@ -59,6 +59,12 @@ void summary_loglik(torch::Tensor ent_values, torch::Tensor rec_values,
                     torch::Tensor csr_row_ptr, torch::Tensor csr_col,
                     torch::Tensor csr_sim, torch::Tensor attr_const,
                     torch::Tensor out);
+void build_keys(torch::Tensor ent_part, torch::Tensor ent_values,
+                torch::Tensor rec_part, torch::Tensor rec_values, int64_t Vmax,
+                torch::Tensor ekeys, torch::Tensor qkeys);
+void summary_counts(torch::Tensor rec_dist, torch::Tensor rec_file,
+                    torch::Tensor ent_rec_ptr, int64_t E, torch::Tensor counts,
+                    torch::Tensor loglik, torch::Tensor packed);
 void kd_descent(torch::Tensor ent_values, torch::Tensor node_kind,
                 torch::Tensor node_attr, torch::Tensor node_a, torch::Tensor node_b,
                 torch::Tensor rset, torch::Tensor ent_part_out);
@@ -80,4 +86,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("distortion_update", &dblink::distortion_update, "K7 distortion resample");
   m.def("summary_loglik", &dblink::summary_loglik, "K8 log-likelihood reduction");
   m.def("kd_descent", &dblink::kd_descent, "K9a KD-tree partition reassignment");
+  m.def("build_keys", &dblink::build_keys, "fused inverted-index key build");
+  m.def("summary_counts", &dblink::summary_counts, "fused summary counts + pack");
 }

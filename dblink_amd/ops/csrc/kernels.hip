@@ -917,6 +917,73 @@ __global__ void summary_loglik_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// Fused support kernels (small-problem launch-count reduction)
+// ---------------------------------------------------------------------------
+
+// Build inverted-index sort keys for entities and query keys for records in
+// one launch: key = (part * A + a) * Vmax + value (a-major for entities).
+__global__ void build_keys_kernel(
+    const int32_t* __restrict__ ent_part, const int32_t* __restrict__ ent_values,
+    const int32_t* __restrict__ rec_part, const int32_t* __restrict__ rec_values,
+    int64_t E, int64_t R, int A, int64_t Vmax,
+    int64_t* __restrict__ ekeys,   // [A * E] a-major
+    int64_t* __restrict__ qkeys) { // [R * A]
+  const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx < E * A) {
+    const int a = (int)(idx / E);
+    const int64_t e = idx % E;
+    ekeys[idx] = ((int64_t)ent_part[e] * A + a) * Vmax + ent_values[e * A + a];
+  } else if (idx < E * A + R * A) {
+    const int64_t j = idx - E * A;
+    const int64_t r = j / A;
+    const int a = (int)(j % A);
+    const int32_t v = rec_values[r * A + a];
+    qkeys[j] = ((int64_t)rec_part[r] * A + a) * Vmax + (v < 0 ? 0 : v);
+  }
+}
+
+// Summary counts in one pass: per-record distortion histogram + per
+// (attr, file) aggregates + isolate count (from the entity->record CSR).
+__global__ void summary_counts_kernel(
+    const uint8_t* __restrict__ rec_dist, const int32_t* __restrict__ rec_file,
+    const int64_t* __restrict__ ent_rec_ptr, int64_t E, int64_t R, int A, int F,
+    unsigned long long* __restrict__ counts) {  // [1 + A*F + A+1]: iso, agg, hist
+  const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  __shared__ int iso_partial[256];
+  int iso = 0;
+  if (idx < E) {
+    iso = (ent_rec_ptr[idx + 1] == ent_rec_ptr[idx]) ? 1 : 0;
+  } else if (idx < E + R) {
+    const int64_t r = idx - E;
+    int nd = 0;
+    const int f = rec_file[r];
+    for (int a = 0; a < A; ++a) {
+      if (rec_dist[r * A + a]) {
+        ++nd;
+        atomicAdd(&counts[1 + a * F + f], 1ull);
+      }
+    }
+    atomicAdd(&counts[1 + A * F + nd], 1ull);
+  }
+  iso_partial[threadIdx.x] = iso;
+  __syncthreads();
+  for (int s = blockDim.x / 2; s > 0; s >>= 1) {
+    if (threadIdx.x < s) iso_partial[threadIdx.x] += iso_partial[threadIdx.x + s];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0 && iso_partial[0]) atomicAdd(&counts[0], (unsigned long long)iso_partial[0]);
+}
+
+// Pack loglik + counts into the f64 summary buffer (one tiny launch).
+__global__ void summary_finalize_kernel(
+    const double* __restrict__ loglik, const unsigned long long* __restrict__ counts,
+    int n_counts, double* __restrict__ packed) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i == 0) packed[0] = loglik[0];
+  if (i < n_counts) packed[1 + i] = (double)counts[i];
+}
+
+// ---------------------------------------------------------------------------
 // K9a: KD-tree descent (flat tree, partitioning.py as_flat layout)
 // ---------------------------------------------------------------------------
 
@@ -1289,6 +1356,47 @@ void summary_loglik(
                      csr_col.data_ptr<int32_t>(), csr_sim.data_ptr<float>(),
                      attr_const.data_ptr<uint8_t>(), E, R, A,
                      out.data_ptr<double>());
+}
+
+void build_keys(torch::Tensor ent_part, torch::Tensor ent_values,
+                torch::Tensor rec_part, torch::Tensor rec_values, int64_t Vmax,
+                torch::Tensor ekeys, torch::Tensor qkeys) {
+  const int64_t E = ent_values.size(0);
+  const int64_t R = rec_values.size(0);
+  const int A = (int)ent_values.size(1);
+  const int64_t total = (E + R) * A;
+  if (total == 0) return;
+  dim3 grid((unsigned)((total + 255) / 256));
+  hipLaunchKernelGGL(build_keys_kernel, grid, dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(),
+                     ent_part.data_ptr<int32_t>(), ent_values.data_ptr<int32_t>(),
+                     rec_part.data_ptr<int32_t>(), rec_values.data_ptr<int32_t>(),
+                     E, R, A, Vmax, ekeys.data_ptr<int64_t>(),
+                     qkeys.data_ptr<int64_t>());
+}
+
+void summary_counts(torch::Tensor rec_dist, torch::Tensor rec_file,
+                    torch::Tensor ent_rec_ptr, int64_t E, torch::Tensor counts,
+                    torch::Tensor loglik, torch::Tensor packed) {
+  const int64_t R = rec_dist.size(0);
+  const int A = (int)rec_dist.size(1);
+  const int F = (int)((counts.numel() - 1 - (A + 1)) / A);
+  const int64_t total = E + R;
+  if (total > 0) {
+    dim3 grid((unsigned)((total + 255) / 256));
+    hipLaunchKernelGGL(summary_counts_kernel, grid, dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(),
+                       rec_dist.data_ptr<uint8_t>(), rec_file.data_ptr<int32_t>(),
+                       ent_rec_ptr.data_ptr<int64_t>(), E, R, A, F,
+                       (unsigned long long*)counts.data_ptr<int64_t>());
+  }
+  const int n_counts = (int)counts.numel();
+  dim3 g2((unsigned)((n_counts + 255) / 256));
+  hipLaunchKernelGGL(summary_finalize_kernel, g2, dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(),
+                     loglik.data_ptr<double>(),
+                     (const unsigned long long*)counts.data_ptr<int64_t>(),
+                     n_counts, packed.data_ptr<double>());
 }
 
 void kd_descent(
