@@ -947,12 +947,17 @@ __global__ void build_keys_kernel(
 __global__ void summary_counts_kernel(
     const uint8_t* __restrict__ rec_dist, const int32_t* __restrict__ rec_file,
     const int64_t* __restrict__ ent_rec_ptr, int64_t E, int64_t R, int A, int F,
+    int n_counts,
     unsigned long long* __restrict__ counts) {  // [1 + A*F + A+1]: iso, agg, hist
+  // stage all counters in LDS: one global atomic per counter per BLOCK
+  // (per-element global atomics on ~12 counters serialize at ~11 ns each)
+  extern __shared__ unsigned int block_counts[];
+  for (int i = threadIdx.x; i < n_counts; i += blockDim.x) block_counts[i] = 0u;
+  __syncthreads();
+
   const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  __shared__ int iso_partial[256];
-  int iso = 0;
   if (idx < E) {
-    iso = (ent_rec_ptr[idx + 1] == ent_rec_ptr[idx]) ? 1 : 0;
+    if (ent_rec_ptr[idx + 1] == ent_rec_ptr[idx]) atomicAdd(&block_counts[0], 1u);
   } else if (idx < E + R) {
     const int64_t r = idx - E;
     int nd = 0;
@@ -960,18 +965,15 @@ __global__ void summary_counts_kernel(
     for (int a = 0; a < A; ++a) {
       if (rec_dist[r * A + a]) {
         ++nd;
-        atomicAdd(&counts[1 + a * F + f], 1ull);
+        atomicAdd(&block_counts[1 + a * F + f], 1u);
       }
     }
-    atomicAdd(&counts[1 + A * F + nd], 1ull);
+    atomicAdd(&block_counts[1 + A * F + nd], 1u);
   }
-  iso_partial[threadIdx.x] = iso;
   __syncthreads();
-  for (int s = blockDim.x / 2; s > 0; s >>= 1) {
-    if (threadIdx.x < s) iso_partial[threadIdx.x] += iso_partial[threadIdx.x + s];
-    __syncthreads();
+  for (int i = threadIdx.x; i < n_counts; i += blockDim.x) {
+    if (block_counts[i]) atomicAdd(&counts[i], (unsigned long long)block_counts[i]);
   }
-  if (threadIdx.x == 0 && iso_partial[0]) atomicAdd(&counts[0], (unsigned long long)iso_partial[0]);
 }
 
 // Pack loglik + counts into the f64 summary buffer (one tiny launch).
@@ -1382,15 +1384,16 @@ void summary_counts(torch::Tensor rec_dist, torch::Tensor rec_file,
   const int A = (int)rec_dist.size(1);
   const int F = (int)((counts.numel() - 1 - (A + 1)) / A);
   const int64_t total = E + R;
+  const int n_counts = (int)counts.numel();
   if (total > 0) {
     dim3 grid((unsigned)((total + 255) / 256));
-    hipLaunchKernelGGL(summary_counts_kernel, grid, dim3(256), 0,
+    hipLaunchKernelGGL(summary_counts_kernel, grid, dim3(256),
+                       n_counts * sizeof(unsigned int),
                        at::cuda::getCurrentCUDAStream(),
                        rec_dist.data_ptr<uint8_t>(), rec_file.data_ptr<int32_t>(),
-                       ent_rec_ptr.data_ptr<int64_t>(), E, R, A, F,
+                       ent_rec_ptr.data_ptr<int64_t>(), E, R, A, F, n_counts,
                        (unsigned long long*)counts.data_ptr<int64_t>());
   }
-  const int n_counts = (int)counts.numel();
   dim3 g2((unsigned)((n_counts + 255) / 256));
   hipLaunchKernelGGL(summary_finalize_kernel, g2, dim3(256), 0,
                      at::cuda::getCurrentCUDAStream(),
