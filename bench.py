@@ -160,6 +160,13 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
     elapsed = float(t.cpu())
 
+    if rank == 0 and hasattr(engine, "phase_times") and getattr(engine, "phase_timers", False):
+        pt = engine.phase_times()
+        total = sum(pt.values()) or 1.0
+        detail = ", ".join(f"{k}={v/ (args.steps + args.warmup):.3f}ms({100*v/total:.0f}%)"
+                           for k, v in sorted(pt.items(), key=lambda kv: -kv[1]))
+        print(f"[bench] phase times per sweep: {detail}", file=sys.stderr)
+
     if rank == 0:
         value = args.steps / elapsed
         out = {

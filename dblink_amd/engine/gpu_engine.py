@@ -231,6 +231,9 @@ class GpuEngine(CpuEngine):
         import os
 
         self.debug_classes = os.environ.get("DBLINK_DEBUG_CLASSES", "") == "1"
+        # per-phase HIP-event timers (SURVEY.md §5.1 observability): eager mode
+        # only; accumulate totals, report via phase_times()
+        self.phase_timers = os.environ.get("DBLINK_PHASE_TIMERS", "") == "1"
         self._graphs_enabled = (
             os.environ.get("DBLINK_GRAPHS", "1") != "0" and world_size == 1
         )
@@ -245,6 +248,8 @@ class GpuEngine(CpuEngine):
         self._graph = None
         self._graph_key = None
         self._graph_warm = 0
+        self._phase_events = []  # (name, start_event, end_event)
+        self._phase_totals = {}
 
     # ---- state residency -----------------------------------------------------
 
@@ -315,6 +320,31 @@ class GpuEngine(CpuEngine):
         state.summary = self._read_summary(state)
         return state
 
+    def _mark(self, name, graph_safe):
+        """Record a named HIP event at a sweep-phase boundary (eager mode)."""
+        if not self.phase_timers or graph_safe:
+            return
+        ev = torch.cuda.Event(enable_timing=True)
+        ev.record()
+        self._phase_events.append((name, ev))
+
+    def phase_times(self):
+        """Drain recorded boundary events -> cumulative {phase: ms}."""
+        if self._phase_events:
+            torch.cuda.synchronize()
+            prev = None
+            for name, ev in self._phase_events:
+                if name == "start":
+                    prev = ev
+                    continue
+                if prev is not None:
+                    self._phase_totals[name] = (
+                        self._phase_totals.get(name, 0.0) + prev.elapsed_time(ev)
+                    )
+                prev = ev
+            self._phase_events = []
+        return dict(self._phase_totals)
+
     def _sweep_body(self, gs: GpuStateTensors, flags: SamplerFlags, graph_safe=True):
         """One full device-side iteration. Capture-safe for world_size == 1:
         static input/output buffers (results copied back in place), control
@@ -323,6 +353,7 @@ class GpuEngine(CpuEngine):
         A, E, R = m.A, gs.E, gs.R
         dev = self.device
 
+        self._mark("start", graph_safe)
         m.theta.copy_(self._theta_pin, non_blocking=True)
         self._ctrl.copy_(self._ctrl_pin, non_blocking=True)
         ctrl = self._ctrl
@@ -344,6 +375,7 @@ class GpuEngine(CpuEngine):
             torch.arange(self.num_partitions + 1, device=dev, dtype=torch.int64),
         )
 
+        self._mark("index", graph_safe)
         # --- phase 1: link update --------------------------------------------
         rec_ent_new = torch.empty_like(gs.rec_ent)
         if flags.sequential or flags.collapsed_entity_ids:
@@ -375,6 +407,7 @@ class GpuEngine(CpuEngine):
                 small_list, wave_list, ctrl,
             )
         gs.rec_ent.copy_(rec_ent_new)
+        self._mark("link", graph_safe)
 
         # --- entity -> records CSR -------------------------------------------
         sorted_re, order = torch.sort(gs.rec_ent, stable=True)
@@ -414,6 +447,7 @@ class GpuEngine(CpuEngine):
             wave_pairs, base_pairs, k1_pairs, m.csr_excl, m.csr_rawsum, m.z1, ctrl,
         )
 
+        self._mark("value", graph_safe)
         # --- phase 3: distortion update --------------------------------------
         self.C.distortion_update(
             gs.rec_values, gs.rec_dist, gs.rec_file, gs.rec_gid, gs.rec_ent,
@@ -421,6 +455,7 @@ class GpuEngine(CpuEngine):
             m.attr_const, seed, it, ctrl,
         )
 
+        self._mark("distortion", graph_safe)
         # --- partition reassignment ------------------------------------------
         ent_part_new = torch.empty_like(gs.ent_part)
         self.C.kd_descent(
@@ -429,12 +464,15 @@ class GpuEngine(CpuEngine):
             ent_part_new,
         )
         gs.ent_part.copy_(ent_part_new)
+        self._mark("kd", graph_safe)
 
         if self.world_size <= 1:
             self._local_sort_static(gs)
+            self._mark("sort", graph_safe)
             # isolate COUNT is permutation-invariant, so the pre-sort CSR is
             # still valid for the summary counts
             self._pack_summary(gs, ent_rec_ptr=ent_rec_ptr)
+            self._mark("summary", graph_safe)
 
     def _local_sort_static(self, gs: GpuStateTensors):
         """Re-sort by partition id into the SAME (static) buffers."""

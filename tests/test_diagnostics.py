@@ -1,0 +1,43 @@
+import numpy as np
+import pytest
+
+from dblink_amd.analysis.diagnostics import ess, read_diagnostics, summarize
+
+
+def test_ess_iid():
+    rng = np.random.default_rng(0)
+    x = rng.normal(size=4000)
+    e = ess(x)
+    assert 2500 < e <= 4000  # iid: ESS ~ n
+
+
+def test_ess_correlated():
+    rng = np.random.default_rng(1)
+    # AR(1) with phi = 0.9 -> tau ~ (1+phi)/(1-phi) = 19 -> ESS ~ n/19
+    n = 8000
+    x = np.empty(n)
+    x[0] = 0
+    for i in range(1, n):
+        x[i] = 0.9 * x[i - 1] + rng.normal()
+    e = ess(x)
+    assert 150 < e < 1200, e
+
+
+def test_ess_constant_and_tiny():
+    assert ess(np.ones(100)) == 100
+    assert ess(np.array([1.0, 2.0])) == 2
+
+
+def test_summarize_from_run(tmp_path):
+    # synthesize a diagnostics.csv
+    p = tmp_path / "diagnostics.csv"
+    rows = ["iteration,systemTime-ms,numObservedEntities,logLikelihood,popSize,aggDist-a,recDistortion-0,recDistortion-1"]
+    rng = np.random.default_rng(2)
+    t0 = 1_000_000
+    for i in range(200):
+        rows.append(f"{i},{t0 + i * 50},{90 + rng.integers(0, 5)},{-1000 + rng.normal():.3f},100,3,90,10")
+    p.write_text("\n".join(rows) + "\n")
+    out = summarize(str(tmp_path))
+    assert out["iterations_per_sec"] == pytest.approx(20.0, rel=0.01)
+    assert out["ess_logLikelihood"] > 50
+    assert "ess_logLikelihood_per_sec" in out
