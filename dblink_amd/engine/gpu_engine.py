@@ -284,9 +284,9 @@ class GpuEngine(CpuEngine):
         self._ctrl_pin[1] = state.iteration + 1
 
         key = (flags.collapsed_entity_ids, flags.collapsed_entity_values, flags.sequential)
-        # graphs pay off where launch overhead dominates; big problems prefer
-        # the eager path with data-dependent kernel-path classification
-        use_graphs = self._graphs_enabled and gs.R < 100_000
+        # classification is sync-free (device-side masks), so every
+        # single-rank problem size can be captured
+        use_graphs = self._graphs_enabled
         if use_graphs:
             if self._graph is not None and self._graph_key == key:
                 self._graph.replay()
@@ -387,24 +387,21 @@ class GpuEngine(CpuEngine):
             )
         else:
             # records with a short smallest candidate list run one-per-thread;
-            # data-dependent list sizes are not graph-capturable, so graphs
-            # (small problems) keep the all-wave path
-            if R >= 100_000 and not graph_safe:
+            # selection is a device-side mask (no host sync, graph-safe)
+            if R >= 50_000:
                 nd = (gs.rec_values >= 0) & (gs.rec_dist == 0)
                 sizes = torch.where(nd, cand_hi - cand_lo, torch.full_like(cand_hi, 1 << 40))
                 min_sizes = sizes.amin(dim=1)
                 has_nd = nd.any(dim=1)
-                small = has_nd & (min_sizes <= 16)
-                small_list = torch.nonzero(small).squeeze(1)
-                wave_list = torch.nonzero(~small).squeeze(1)
+                small_mask = (has_nd & (min_sizes <= 16)).to(torch.uint8).contiguous()
             else:
-                small_list = wave_list = self._empty_i64
+                small_mask = torch.empty(0, dtype=torch.uint8, device=dev)
             self.C.link_update(
                 gs.rec_values, gs.rec_dist, gs.rec_gid, gs.rec_part,
                 cand_lo, cand_hi, postings, gs.ent_values,
                 ent_ptr, m.log_norm, m.voff, m.csr_row_ptr, m.csr_col, m.csr_sim,
                 m.attr_const, seed, it, rec_ent_new, gs.rec_ent, self._err,
-                small_list, wave_list, ctrl,
+                small_mask, ctrl,
             )
         gs.rec_ent.copy_(rec_ent_new)
         self._mark("link", graph_safe)
@@ -421,22 +418,8 @@ class GpuEngine(CpuEngine):
         kobs = torch.zeros(E * A, dtype=torch.int32, device=dev)
         pair_idx = (gs.rec_ent.view(R, 1) * A
                     + torch.arange(A, device=dev, dtype=torch.int64).view(1, A))
-        idxm_all = pair_idx.reshape(-1)
-        kobs.scatter_add_(0, idxm_all, obs.reshape(-1).to(torch.int32))
-        if graph_safe:
-            # fixed shapes: every pair takes the wave kernel (it branches on
-            # k_obs internally)
-            base_pairs = k1_pairs = self._empty_i64
-            wave_pairs = torch.arange(E * A, device=dev, dtype=torch.int64)
-        else:
-            base_pairs = torch.nonzero(kobs == 0).squeeze(1)
-            k1_pairs = torch.nonzero(kobs == 1).squeeze(1)
-            wave_pairs = torch.nonzero(kobs >= 2).squeeze(1)
-            if self.debug_classes:
-                import sys
-
-                print(f"[dblink classes] base={base_pairs.numel()} k1={k1_pairs.numel()} "
-                      f"wave={wave_pairs.numel()} kobs_max={kobs.max().item()}", file=sys.stderr)
+        kobs.scatter_add_(0, pair_idx.reshape(-1), obs.reshape(-1).to(torch.int32))
+        # kernels self-select on kobs: no host-side pair lists, no sync
         self.C.value_update(
             gs.rec_values, gs.rec_dist, gs.rec_file, ent_rec_ptr, ent_rec_idx,
             gs.ent_values, m.theta, m.phi, m.log_phi, m.norm_lin, m.log_norm,
@@ -444,7 +427,8 @@ class GpuEngine(CpuEngine):
             m.pow_prob, m.pow_alias, m.pow_off, m.log_pow_total, m.attr_const,
             m.Kc, 1 if flags.collapsed_entity_values else 0,
             1 if flags.sequential else 0, seed, it, self._ent_id_base, self._err,
-            wave_pairs, base_pairs, k1_pairs, m.csr_excl, m.csr_rawsum, m.z1, ctrl,
+            self._empty_i64, self._empty_i64, self._empty_i64,
+            m.csr_excl, m.csr_rawsum, m.z1, ctrl, kobs,
         )
 
         self._mark("value", graph_safe)

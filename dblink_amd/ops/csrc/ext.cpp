@@ -20,8 +20,7 @@ void link_update(torch::Tensor rec_values, torch::Tensor rec_dist,
                  torch::Tensor csr_sim, torch::Tensor attr_const, int64_t seed,
                  int64_t iteration, torch::Tensor rec_ent_out,
                  torch::Tensor rec_ent_in, torch::Tensor error_count,
-                 torch::Tensor small_list, torch::Tensor wave_list,
-                 torch::Tensor ctrl);
+                 torch::Tensor small_mask, torch::Tensor ctrl);
 void link_update_dense(torch::Tensor rec_values, torch::Tensor rec_dist,
                        torch::Tensor rec_gid, torch::Tensor rec_part,
                        torch::Tensor rec_file, torch::Tensor ent_values,
@@ -45,7 +44,8 @@ void value_update(torch::Tensor rec_values, torch::Tensor rec_dist,
                   int64_t iteration, int64_t ent_id_base, torch::Tensor error_count,
                   torch::Tensor wave_pairs, torch::Tensor base_pairs,
                   torch::Tensor k1_pairs, torch::Tensor csr_excl,
-                  torch::Tensor csr_rawsum, torch::Tensor z1, torch::Tensor ctrl);
+                  torch::Tensor csr_rawsum, torch::Tensor z1, torch::Tensor ctrl,
+                  torch::Tensor kobs);
 void distortion_update(torch::Tensor rec_values, torch::Tensor rec_dist,
                        torch::Tensor rec_file, torch::Tensor rec_gid,
                        torch::Tensor rec_ent, torch::Tensor ent_values,

@@ -54,7 +54,7 @@ __global__ void link_update_kernel(
     const int32_t* __restrict__ csr_col,
     const float* __restrict__ csr_sim,
     const uint8_t* __restrict__ attr_const,  // [A]
-    const int64_t* __restrict__ rec_list,    // [R] records to process (or null)
+    const uint8_t* __restrict__ small_mask,  // [R] 1 = handled by small kernel (or null)
     int64_t R, int A,
     uint64_t seed, uint32_t iteration, const int64_t* __restrict__ ctrl,
     int64_t* __restrict__ rec_ent_out,       // [R]
@@ -63,8 +63,8 @@ __global__ void link_update_kernel(
   const int lane = threadIdx.x & (WAVE - 1);
   int64_t r = (int64_t)blockIdx.x * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
   if (r >= R) return;
+  if (small_mask != nullptr && small_mask[r]) return;
   if (ctrl != nullptr) { seed = (uint64_t)ctrl[0]; iteration = (uint32_t)ctrl[1]; }
-  if (rec_list != nullptr) r = rec_list[r];
 
   // Gather per-attribute candidate ranges (observed non-distorted) and the
   // observed-distorted non-constant attributes (constant od attrs scale all
@@ -150,7 +150,7 @@ __global__ void link_update_kernel(
 // short (the common case once clusters localize): one thread walks the
 // intersection serially — 64x fewer wave slots than the wave path.
 __global__ void link_update_small_kernel(
-    const int64_t* __restrict__ rec_list, int64_t n_recs,
+    const uint8_t* __restrict__ small_mask, int64_t n_recs,
     const int32_t* __restrict__ rec_values, const uint8_t* __restrict__ rec_dist,
     const int64_t* __restrict__ rec_gid, const int64_t* __restrict__ cand_lo,
     const int64_t* __restrict__ cand_hi, const int32_t* __restrict__ postings,
@@ -163,8 +163,9 @@ __global__ void link_update_small_kernel(
     const int64_t* __restrict__ rec_ent_in, int* __restrict__ error_count) {
   const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (idx >= n_recs) return;
+  if (!small_mask[idx]) return;
   if (ctrl != nullptr) { seed = (uint64_t)ctrl[0]; iteration = (uint32_t)ctrl[1]; }
-  const int64_t r = rec_list[idx];
+  const int64_t r = idx;
 
   int nd_n = 0, od_n = 0;
   int64_t nd_lo[MAX_ATTRS], nd_hi[MAX_ATTRS];
@@ -287,7 +288,8 @@ constexpr int WAVES_PER_BLOCK_VAL = 4;    // 256 threads
 
 struct ValueArgs {
   const int64_t* ctrl;         // [2] = {seed, iteration} device override (or null)
-  const int64_t* pair_list;    // [n_pairs] flattened (e*A + a) needing wave work
+  const int64_t* pair_list;    // [n_pairs] flattened (e*A + a) (or null)
+  const int32_t* kobs;         // [E*A] observed-linked counts for self-selection (or null)
   int64_t n_pairs;
   const double* csr_excl;      // [nnz] exclusive row prefix of raw k=1 weights
   const double* csr_rawsum;    // [Vtot] row totals of raw k=1 weights
@@ -354,7 +356,8 @@ __global__ void value_base_draw_kernel(ValueArgs args) {
   }
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= args.n_pairs) return;
-  const int64_t pair = args.pair_list[i];
+  const int64_t pair = args.pair_list ? args.pair_list[i] : i;
+  if (args.kobs && args.kobs[pair] != 0) return;
   const int64_t e = pair / args.A;
   const int a = (int)(pair % args.A);
   const int64_t v0 = args.voff[a];
@@ -380,7 +383,8 @@ __global__ void value_update_k1_kernel(ValueArgs args) {
   }
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= args.n_pairs) return;
-  const int64_t pair = args.pair_list[i];
+  const int64_t pair = args.pair_list ? args.pair_list[i] : i;
+  if (args.kobs && args.kobs[pair] != 1) return;
   const int64_t e = pair / args.A;
   const int a = (int)(pair % args.A);
   const bool is_const = args.attr_const[a];
@@ -464,6 +468,11 @@ __global__ void value_update_k1_kernel(ValueArgs args) {
   args.ent_values[e * args.A + a] = (int32_t)v_new;
 }
 
+constexpr int VAL_STRIDE = 8;  // pairs examined per wave in kobs mode
+
+__device__ void value_update_pair(const ValueArgs& args, int64_t pair, int lane,
+                                  int32_t* keys, float* vals);
+
 __global__ void __launch_bounds__(WAVES_PER_BLOCK_VAL * WAVE)
 value_update_kernel(ValueArgs args) {
   if (args.ctrl != nullptr) {
@@ -475,9 +484,22 @@ value_update_kernel(ValueArgs args) {
 
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave = threadIdx.x / WAVE;
-  const int64_t pidx = (int64_t)blockIdx.x * WAVES_PER_BLOCK_VAL + wave;
-  if (pidx >= args.n_pairs) return;
-  const int64_t pair = args.pair_list[pidx];
+  const int64_t widx = (int64_t)blockIdx.x * WAVES_PER_BLOCK_VAL + wave;
+  if (args.pair_list != nullptr) {
+    if (widx >= args.n_pairs) return;
+    value_update_pair(args, args.pair_list[widx], lane, h_key[wave], h_val[wave]);
+  } else {
+    // kobs self-selection: each wave examines VAL_STRIDE consecutive pairs
+    // and runs the (rare) k >= 2 ones serially
+    const int64_t p0 = widx * VAL_STRIDE;
+    for (int64_t pair = p0; pair < p0 + VAL_STRIDE && pair < args.n_pairs; ++pair) {
+      if (args.kobs[pair] >= 2) value_update_pair(args, pair, lane, h_key[wave], h_val[wave]);
+    }
+  }
+}
+
+__device__ void value_update_pair(const ValueArgs& args, int64_t pair, int lane,
+                                  int32_t* keys, float* vals) {
   const int64_t e = pair / args.A;
   const int a = (int)(pair % args.A);
   const bool is_const = args.attr_const[a];
@@ -609,8 +631,6 @@ value_update_kernel(ValueArgs args) {
 
   // ---- perturbation weights in the LDS hash table (k >= 2) ---------------
   // L_v = sum_r log f_r(v); final weight = base_prob(v) * (exp(L_v) - 1).
-  int32_t* keys = h_key[wave];
-  float* vals = h_val[wave];
 
   // size the table (and its clear/scan cost) to the actual support
   int64_t total_entries = 0;
@@ -1133,7 +1153,7 @@ void link_update(
     torch::Tensor csr_col, torch::Tensor csr_sim, torch::Tensor attr_const,
     int64_t seed, int64_t iteration, torch::Tensor rec_ent_out,
     torch::Tensor rec_ent_in, torch::Tensor error_count,
-    torch::Tensor small_list, torch::Tensor wave_list, torch::Tensor ctrl) {
+    torch::Tensor small_mask, torch::Tensor ctrl) {
   const int64_t* ctrl_ptr = ctrl.numel() ? ctrl.data_ptr<int64_t>() : nullptr;
   CHECK_GPU(rec_values);
   const int64_t R = rec_values.size(0);
@@ -1141,10 +1161,10 @@ void link_update(
   TORCH_CHECK(A <= MAX_ATTRS, "at most ", MAX_ATTRS, " matching attributes supported");
   if (R == 0) return;
   constexpr int WPB = 4;
-  const bool split = small_list.numel() > 0 || wave_list.numel() > 0;
-  if (!split || wave_list.numel() > 0) {
-    const int64_t n = split ? wave_list.numel() : R;
-    dim3 grid((unsigned)wave_grid(n, WPB));
+  const bool split = small_mask.numel() > 0;
+  const uint8_t* mask_ptr = split ? small_mask.data_ptr<uint8_t>() : nullptr;
+  {
+    dim3 grid((unsigned)wave_grid(R, WPB));
     hipLaunchKernelGGL(link_update_kernel, grid, dim3(WPB * WAVE), 0,
                        at::cuda::getCurrentCUDAStream(),
                        rec_values.data_ptr<int32_t>(), rec_dist.data_ptr<uint8_t>(),
@@ -1154,18 +1174,16 @@ void link_update(
                        ent_ptr.data_ptr<int64_t>(), log_norm.data_ptr<float>(),
                        voff.data_ptr<int64_t>(), csr_row_ptr.data_ptr<int64_t>(),
                        csr_col.data_ptr<int32_t>(), csr_sim.data_ptr<float>(),
-                       attr_const.data_ptr<uint8_t>(),
-                       split ? wave_list.data_ptr<int64_t>() : nullptr, n, A,
+                       attr_const.data_ptr<uint8_t>(), mask_ptr, R, A,
                        (uint64_t)seed, (uint32_t)iteration, ctrl_ptr,
                        rec_ent_out.data_ptr<int64_t>(),
                        rec_ent_in.data_ptr<int64_t>(), error_count.data_ptr<int>());
   }
-  if (small_list.numel() > 0) {
-    const int64_t n = small_list.numel();
-    dim3 grid((unsigned)((n + 255) / 256));
+  if (split) {
+    dim3 grid((unsigned)((R + 255) / 256));
     hipLaunchKernelGGL(link_update_small_kernel, grid, dim3(256), 0,
                        at::cuda::getCurrentCUDAStream(),
-                       small_list.data_ptr<int64_t>(), n,
+                       mask_ptr, R,
                        rec_values.data_ptr<int32_t>(), rec_dist.data_ptr<uint8_t>(),
                        rec_gid.data_ptr<int64_t>(), cand_lo.data_ptr<int64_t>(),
                        cand_hi.data_ptr<int64_t>(), postings.data_ptr<int32_t>(),
@@ -1221,6 +1239,8 @@ static ValueArgs make_value_args(
     torch::Tensor csr_excl, torch::Tensor csr_rawsum, torch::Tensor z1) {
   ValueArgs a;
   a.ctrl = nullptr;
+  a.kobs = nullptr;
+  a.pair_list = nullptr;
   a.csr_excl = csr_excl.numel() ? csr_excl.data_ptr<double>() : nullptr;
   a.csr_rawsum = csr_rawsum.data_ptr<double>();
   a.z1 = z1.data_ptr<double>();
@@ -1270,7 +1290,7 @@ void value_update(
     int64_t seed, int64_t iteration, int64_t ent_id_base, torch::Tensor error_count,
     torch::Tensor wave_pairs, torch::Tensor base_pairs, torch::Tensor k1_pairs,
     torch::Tensor csr_excl, torch::Tensor csr_rawsum, torch::Tensor z1,
-    torch::Tensor ctrl) {
+    torch::Tensor ctrl, torch::Tensor kobs) {
   ValueArgs args = make_value_args(
       rec_values, rec_dist, rec_file, ent_rec_ptr, ent_rec_idx, ent_values, theta,
       phi, log_phi, norm_lin, log_norm, voff, csr_row_ptr, csr_col, csr_sim,
@@ -1288,7 +1308,23 @@ void value_update(
                        at::cuda::getCurrentCUDAStream(), args);
     return;
   }
-  // k_obs == 0 pairs: one thread each (base draw is phi for every variant)
+  if (kobs.numel() > 0) {
+    // kobs self-selection: no host-side pair lists, no stream sync — the
+    // three kernels each cover every pair and act only on their class
+    args.kobs = kobs.data_ptr<int32_t>();
+    args.n_pairs = kobs.numel();
+    dim3 tgrid((unsigned)((args.n_pairs + 255) / 256));
+    hipLaunchKernelGGL(value_base_draw_kernel, tgrid, dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(), args);
+    hipLaunchKernelGGL(value_update_k1_kernel, tgrid, dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(), args);
+    const int64_t n_waves = (args.n_pairs + VAL_STRIDE - 1) / VAL_STRIDE;
+    dim3 wgrid((unsigned)wave_grid(n_waves, WAVES_PER_BLOCK_VAL));
+    hipLaunchKernelGGL(value_update_kernel, wgrid, dim3(WAVES_PER_BLOCK_VAL * WAVE), 0,
+                       at::cuda::getCurrentCUDAStream(), args);
+    return;
+  }
+  // explicit pair lists (tests / special cases)
   if (base_pairs.numel() > 0) {
     args.pair_list = base_pairs.data_ptr<int64_t>();
     args.n_pairs = base_pairs.numel();
@@ -1296,7 +1332,6 @@ void value_update(
     hipLaunchKernelGGL(value_base_draw_kernel, grid, dim3(256), 0,
                        at::cuda::getCurrentCUDAStream(), args);
   }
-  // k_obs == 1 pairs with small sim rows: one thread each
   if (k1_pairs.numel() > 0) {
     args.pair_list = k1_pairs.data_ptr<int64_t>();
     args.n_pairs = k1_pairs.numel();
@@ -1304,7 +1339,6 @@ void value_update(
     hipLaunchKernelGGL(value_update_k1_kernel, grid, dim3(256), 0,
                        at::cuda::getCurrentCUDAStream(), args);
   }
-  // remaining pairs: one wave each
   if (wave_pairs.numel() > 0) {
     args.pair_list = wave_pairs.data_ptr<int64_t>();
     args.n_pairs = wave_pairs.numel();

@@ -205,8 +205,7 @@ def test_link_kernel_distribution():
         model.log_norm, model.voff, model.csr_row_ptr, model.csr_col, model.csr_sim,
         model.attr_const, 4321, 11, out,
         _dev(np.zeros(N, np.int64), torch.int64), err,
-        torch.empty(0, dtype=torch.int64, device=DEV),
-        torch.empty(0, dtype=torch.int64, device=DEV),
+        torch.empty(0, dtype=torch.uint8, device=DEV),
         torch.empty(0, dtype=torch.int64, device=DEV),
     )
     assert int(err.cpu()) == 0
@@ -268,6 +267,7 @@ def test_value_kernel_distribution():
         torch.empty(0, dtype=torch.int64, device=DEV),
         model.csr_excl, model.csr_rawsum, model.z1,
         torch.empty(0, dtype=torch.int64, device=DEV),
+        torch.empty(0, dtype=torch.int32, device=DEV),
     )
     got = ev.cpu().numpy()[:, a]
     emp = np.bincount(got, minlength=V) / E
@@ -289,6 +289,7 @@ def test_value_kernel_distribution():
         torch.arange(E * 2, dtype=torch.int64, device=DEV),
         model.csr_excl, model.csr_rawsum, model.z1,
         torch.empty(0, dtype=torch.int64, device=DEV),
+        torch.empty(0, dtype=torch.int32, device=DEV),
     )
     got2 = ev2.cpu().numpy()[:, a]
     emp2 = np.bincount(got2, minlength=V) / E
