@@ -419,6 +419,38 @@ class GpuEngine(CpuEngine):
         pair_idx = (gs.rec_ent.view(R, 1) * A
                     + torch.arange(A, device=dev, dtype=torch.int64).view(1, A))
         kobs.scatter_add_(0, pair_idx.reshape(-1), obs.reshape(-1).to(torch.int32))
+        if getattr(self, "value_allwave", False):  # A/B debugging aid
+            wave_all = torch.arange(E * A, device=dev, dtype=torch.int64)
+            self.C.value_update(
+                gs.rec_values, gs.rec_dist, gs.rec_file, ent_rec_ptr, ent_rec_idx,
+                gs.ent_values, m.theta, m.phi, m.log_phi, m.norm_lin, m.log_norm,
+                m.voff, m.csr_row_ptr, m.csr_col, m.csr_sim, m.phi_prob, m.phi_alias,
+                m.pow_prob, m.pow_alias, m.pow_off, m.log_pow_total, m.attr_const,
+                m.Kc, 1 if flags.collapsed_entity_values else 0,
+                1 if flags.sequential else 0, seed, it, self._ent_id_base, self._err,
+                wave_all, self._empty_i64, self._empty_i64,
+                m.csr_excl, m.csr_rawsum, m.z1, ctrl,
+                torch.empty(0, dtype=torch.int32, device=dev),
+            )
+            self._mark("value", graph_safe)
+            self.C.distortion_update(
+                gs.rec_values, gs.rec_dist, gs.rec_file, gs.rec_gid, gs.rec_ent,
+                gs.ent_values, m.theta, m.phi, m.norm_lin, m.self_expsim, m.voff,
+                m.attr_const, seed, it, ctrl,
+            )
+            self._mark("distortion", graph_safe)
+            ent_part_new = torch.empty_like(gs.ent_part)
+            self.C.kd_descent(
+                gs.ent_values, self.flat_tree["kind"], self.flat_tree["attr"],
+                self.flat_tree["a"], self.flat_tree["b"], self.flat_tree["rset"],
+                ent_part_new,
+            )
+            gs.ent_part.copy_(ent_part_new)
+            self._mark("kd", graph_safe)
+            if self.world_size <= 1:
+                self._local_sort_static(gs)
+                self._pack_summary(gs, ent_rec_ptr=ent_rec_ptr)
+            return
         # kernels self-select on kobs: no host-side pair lists, no sync
         self.C.value_update(
             gs.rec_values, gs.rec_dist, gs.rec_file, ent_rec_ptr, ent_rec_idx,
