@@ -311,9 +311,15 @@ class GpuEngine(CpuEngine):
             if self.world_size > 1 and comm.is_distributed():
                 from ..parallel.migration import migrate_and_sort_tensors
 
-                migrate_and_sort_tensors(gs, self.world_size)
-                # rebuild static-buffer invariants after the exchange
+                # summary stats are partition-agnostic: pack them from the
+                # pre-migration state and overlap the small all-reduce (on
+                # its own communicator) with the migration all-to-all
                 self._pack_summary(gs)
+                work = comm.all_reduce_sum_async(self._packed)
+                migrate_and_sort_tensors(gs, self.world_size)
+                if work is not None:
+                    work.wait()
+                self._summary_reduced = True
 
         state.current_seed += self.num_partitions
         state.iteration += 1
@@ -534,9 +540,12 @@ class GpuEngine(CpuEngine):
         m = self.model
         A, F = m.A, m.F
         packed = self._packed
-        if self.world_size > 1 and comm.is_distributed():
+        if self.world_size > 1 and comm.is_distributed() and not getattr(
+            self, "_summary_reduced", False
+        ):
             packed = packed.clone()
             comm.all_reduce_sum_(packed)
+        self._summary_reduced = False
         host = packed.cpu().numpy()
         err = int(self._err.cpu())
         if err:

@@ -69,6 +69,29 @@ def all_reduce_sum_(t: torch.Tensor):
     return t
 
 
+_summary_group = None
+
+
+def summary_group():
+    """A second process group so the small summary all-reduce can overlap
+    with the migration all-to-all on the default group (separate
+    communicator -> separate stream on RCCL)."""
+    global _summary_group
+    if not is_distributed():
+        return None
+    if _summary_group is None:
+        _summary_group = dist.new_group(ranks=list(range(dist.get_world_size())))
+    return _summary_group
+
+
+def all_reduce_sum_async(t: torch.Tensor):
+    """Start an async SUM all-reduce on the summary group; returns the work
+    handle (or None when single-process)."""
+    if not is_distributed():
+        return None
+    return dist.all_reduce(t, op=dist.ReduceOp.SUM, group=summary_group(), async_op=True)
+
+
 def all_gather_object(obj):
     if not is_distributed():
         return [obj]

@@ -237,3 +237,33 @@ def test_tensor_migration_preserves_links():
     every record->entity link across the all-to-all (gloo, CPU tensors)."""
     outs = _run_workers(TENSOR_MIGRATE.replace("__ROOT__", ROOT))
     assert any('"ok": true' in o for o in outs), outs
+
+
+ASYNC_OVERLAP = r"""
+import json, sys
+sys.path.insert(0, "__ROOT__")
+import torch
+import torch.distributed as dist
+from dblink_amd.parallel import comm
+
+rank, world, device = comm.init_from_env(backend="gloo")
+# start async summary reduce on the second group, then run an all_to_all on
+# the default group before waiting (the GPU engine's overlap pattern)
+t = torch.full((8,), float(rank + 1), dtype=torch.float64)
+work = comm.all_reduce_sum_async(t)
+payload = torch.full((4, 3), rank, dtype=torch.int32)
+out, counts = comm.all_to_all_v(payload, [2] * world)
+work.wait()
+expect = sum(range(1, world + 1))
+assert torch.all(t == expect), t
+assert out.shape[0] == 2 * world
+if rank == 0:
+    print(json.dumps({"ok": True}))
+dist.destroy_process_group()
+"""
+
+
+@pytest.mark.slow
+def test_async_summary_reduce_overlaps_alltoall():
+    outs = _run_workers(ASYNC_OVERLAP.replace("__ROOT__", ROOT))
+    assert any('"ok": true' in o for o in outs), outs

@@ -473,3 +473,42 @@ def test_gpu_sampler_variants(sampler, tmp_path):
     assert s.rec_distortions.sum() == n
     engine.sync_state(state)
     assert state.rec_ent.min() >= 0 and state.rec_ent.max() < state.num_entities
+
+
+@gpu
+def test_gpu_project_pipeline(tmp_path):
+    """Full user path on GPU: HOCON config -> Project -> sample/summarize/
+    evaluate steps with engine 'auto' (selects the GPU engine)."""
+    import os
+    import sys
+
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    sys.path.insert(0, os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "tests"))
+    from test_end_to_end_cpu import CONF_TEMPLATE
+    from dblink_amd.api.project import Project, parse_steps
+    from dblink_amd.utils import hocon
+    from dblink_amd.utils.synthdata import write_csv
+
+    data = str(tmp_path / "data.csv")
+    write_csv(data, 250, dup_fraction=0.1, seed=12)
+    out = str(tmp_path / "results")
+    conf = CONF_TEMPLATE.format(
+        data=data, out=out, samples=30, burnin=20, thin=2, cutoff=30,
+        sampler="PCG-I", levels=1, part_attrs='"fname_c1"',
+    ).replace('engine : "cpu"', 'engine : "auto"')
+    cfg = hocon.parse_string(conf)
+    project = Project(cfg, rank=0, world_size=1)
+    os.makedirs(project.output_path, exist_ok=True)
+    for step in parse_steps(cfg, project):
+        step.execute()
+    from dblink_amd.engine.gpu_engine import GpuEngine
+
+    assert isinstance(project.engine(), GpuEngine), "auto engine must pick the GPU"
+    for f in ("diagnostics.csv", "evaluation-results.txt",
+              "cluster-size-distribution.csv", "driver-state"):
+        assert os.path.exists(os.path.join(out, f)), f
+    import numpy as np
+    with open(os.path.join(out, "diagnostics.csv")) as fh:
+        fh.readline()
+        ll = [float(line.split(",")[3]) for line in fh if line.strip()]
+    assert len(ll) >= 30 and all(np.isfinite(ll))
