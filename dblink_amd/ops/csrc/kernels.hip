@@ -85,18 +85,20 @@ __global__ void link_update_kernel(
       od_a[od_n++] = a;
     }
   }
-  // sort candidate lists ascending by size: the smallest is the iteration
-  // base and the remaining checks reject candidates most-selectively-first
-  for (int i = 1; i < nd_n; ++i) {
-    int a_i = nd_a[i];
-    int64_t lo_i = nd_lo[i], hi_i = nd_hi[i];
-    const int64_t sz = hi_i - lo_i;
-    int j = i - 1;
-    while (j >= 0 && (nd_hi[j] - nd_lo[j]) > sz) {
-      nd_a[j + 1] = nd_a[j]; nd_lo[j + 1] = nd_lo[j]; nd_hi[j + 1] = nd_hi[j];
-      --j;
+  // base = smallest candidate list (runtime-indexed shifting sorts would
+  // push these arrays to scratch — keep a single swap)
+  if (nd_n > 1) {
+    int best = 0;
+    int64_t best_sz = nd_hi[0] - nd_lo[0];
+    for (int i = 1; i < nd_n; ++i) {
+      int64_t sz = nd_hi[i] - nd_lo[i];
+      if (sz < best_sz) { best = i; best_sz = sz; }
     }
-    nd_a[j + 1] = a_i; nd_lo[j + 1] = lo_i; nd_hi[j + 1] = hi_i;
+    if (best != 0) {
+      int ta = nd_a[0]; nd_a[0] = nd_a[best]; nd_a[best] = ta;
+      int64_t tl = nd_lo[0]; nd_lo[0] = nd_lo[best]; nd_lo[best] = tl;
+      int64_t th = nd_hi[0]; nd_hi[0] = nd_hi[best]; nd_hi[best] = th;
+    }
   }
 
   int64_t base_lo, base_n;
@@ -118,7 +120,8 @@ __global__ void link_update_kernel(
   for (int64_t i = lane; i < base_n; i += WAVE) {
     int32_t e = base_postings ? postings[base_lo + i] : (int32_t)(base_lo + i);
     bool ok = true;
-    for (int j = 1; j < nd_n; ++j) {
+    for (int j = 0; j < nd_n; ++j) {
+      if (nd_lo[j] == base_lo) continue;
       if (!contains_i32(postings, nd_lo[j], nd_hi[j], e)) { ok = false; break; }
     }
     if (!ok) continue;
@@ -169,32 +172,28 @@ __global__ void link_update_small_kernel(
   int nd_n = 0, od_n = 0;
   int64_t nd_lo[MAX_ATTRS], nd_hi[MAX_ATTRS];
   int od_a[MAX_ATTRS];
+  int64_t base_lo = 0, base_hi = 0, base_sz = INT64_MAX;
   for (int a = 0; a < A; ++a) {
     const int32_t x = rec_values[r * A + a];
     if (x < 0) continue;
     if (!rec_dist[r * A + a]) {
       const int64_t lo = cand_lo[r * A + a], hi = cand_hi[r * A + a];
-      const int64_t sz = hi - lo;
-      int j = nd_n - 1;
-      while (j >= 0 && (nd_hi[j] - nd_lo[j]) > sz) {
-        nd_lo[j + 1] = nd_lo[j]; nd_hi[j + 1] = nd_hi[j];
-        --j;
-      }
-      nd_lo[j + 1] = lo; nd_hi[j + 1] = hi;
+      nd_lo[nd_n] = lo;
+      nd_hi[nd_n] = hi;
       ++nd_n;
+      if (hi - lo < base_sz) { base_sz = hi - lo; base_lo = lo; base_hi = hi; }
     } else if (!attr_const[a]) {
       od_a[od_n++] = a;
     }
   }
-  const int64_t base_lo = nd_n ? nd_lo[0] : 0;
-  const int64_t base_hi = nd_n ? nd_hi[0] : 0;
   const uint64_t gid = (uint64_t)rec_gid[r];
   float best_score = -INFINITY;
   long long best_e = -1;
   for (int64_t i = base_lo; i < base_hi; ++i) {
     const int32_t e = postings[i];
     bool ok = true;
-    for (int j = 1; j < nd_n; ++j) {
+    for (int j = 0; j < nd_n; ++j) {
+      if (nd_lo[j] == base_lo) continue;
       if (!contains_i32(postings, nd_lo[j], nd_hi[j], e)) { ok = false; break; }
     }
     if (!ok) continue;
