@@ -9,7 +9,6 @@ from __future__ import annotations
 import os
 from collections import Counter, defaultdict
 
-import numpy as np
 
 from ..engine.writers import read_linkage_chain
 

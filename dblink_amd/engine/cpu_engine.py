@@ -91,7 +91,7 @@ class _PartitionData:
 
     __slots__ = (
         "ent_values", "rec_values", "rec_file", "rec_dist", "rec_gid",
-        "rec_ent", "ent_recs", "inv_index",
+        "rec_ent", "ent_recs",
     )
 
 

@@ -25,7 +25,6 @@ import logging
 import numpy as np
 import torch
 
-from ..models.distortion import DistortionProbs
 from ..parallel import comm
 from .cpu_engine import CpuEngine, SamplerFlags
 from .state import ChainState, SummaryVars
@@ -105,7 +104,7 @@ class GpuModel:
         #   row prefix  = exclusive cumsum of raw_w1 within the row
         #   rawsum[x]   = row total
         # and a k=1 draw is one binary search (value_update_k1_kernel).
-        raw_list, excl_list, rawsum = [], [], np.zeros(self.Vtot, dtype=np.float64)
+        excl_list, rawsum = [], np.zeros(self.Vtot, dtype=np.float64)
         for a, ia in enumerate(attrs):
             if ia.is_constant:
                 continue
@@ -120,7 +119,6 @@ class GpuModel:
             excl = c - raw - np.repeat(base_c, counts)
             ends = si.row_ptr[1:] - 1
             sums = np.where(counts > 0, c[np.maximum(ends, 0)], 0.0) - base_c
-            raw_list.append(raw)
             excl_list.append(excl)
             rawsum[voff[a] : voff[a + 1]] = sums
         csr_excl = np.concatenate(excl_list) if excl_list else np.empty(0)
@@ -296,7 +294,6 @@ class GpuEngine(CpuEngine):
                     self._graph_warm = 0
                 self._sweep_body(gs, flags)
                 self._graph_warm += 1
-                self._graph_key = None if self._graph is None else self._graph_key
             else:
                 g = torch.cuda.CUDAGraph()
                 torch.cuda.synchronize()

@@ -18,7 +18,7 @@ import numpy as np
 
 from ..models.distortion import DistortionProbs
 from ..parallel import comm, migration
-from .state import ChainState, SummaryVars
+from .state import ChainState
 
 
 def entities_per_rank(num_recs_per_rank, population_size):

@@ -15,7 +15,6 @@ from __future__ import annotations
 import os
 import time
 
-import numpy as np
 import pyarrow as pa
 import pyarrow.parquet as pq
 

@@ -27,7 +27,7 @@ from bisect import bisect_left
 import numpy as np
 
 from .alias import AliasTable
-from .similarity import ConstantSimilarityFn, SimilarityFn
+from .similarity import SimilarityFn
 
 
 class SimIndexCSR:

@@ -13,7 +13,6 @@ Policy:
 
 from __future__ import annotations
 
-import math
 
 import numpy as np
 
