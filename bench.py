@@ -191,7 +191,10 @@ def main():
                 "sampler": args.sampler,
                 "schema": args.schema,
                 "num_files": args.num_files,
-                "parallelism": f"entity-partitioned Gibbs, {n_gpus} rank(s) over RCCL",
+                "parallelism": (
+                    f"entity-partitioned Gibbs, {n_gpus} rank(s) over "
+                    + ("RCCL" if use_gpu else "gloo")
+                ),
                 "engine": "gpu" if use_gpu else "cpu",
             },
         }

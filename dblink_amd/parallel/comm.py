@@ -107,6 +107,62 @@ def _backend_supports_all_to_all():
         return False
 
 
+def exchange_counts(*send_count_lists):
+    """Exchange K count vectors in ONE all_to_all (one sync instead of K).
+
+    Each argument is a length-world list; returns the matching recv lists.
+    """
+    if not is_distributed():
+        return tuple([int(c) for c in lst] for lst in send_count_lists)
+    world = dist.get_world_size()
+    K = len(send_count_lists)
+    sc = torch.tensor(
+        [[int(lst[p]) for lst in send_count_lists] for p in range(world)],
+        dtype=torch.int64,
+    ).reshape(-1)
+    rc = torch.empty_like(sc)
+    dist.all_to_all_single(rc, sc)
+    rc = rc.view(world, K).cpu()
+    return tuple([int(rc[p, k]) for p in range(world)] for k in range(K))
+
+
+def all_to_all_payload(tensor: torch.Tensor, send_counts, recv_counts):
+    """Row-slice exchange with counts already known (no counts round-trip)."""
+    if not is_distributed():
+        return tensor
+    world = dist.get_world_size()
+    out_shape = (sum(recv_counts),) + tuple(tensor.shape[1:])
+    out = torch.empty(out_shape, dtype=tensor.dtype, device=tensor.device)
+    if _backend_supports_all_to_all():
+        dist.all_to_all_single(out, tensor.contiguous(), recv_counts, send_counts)
+        return out
+    rank = dist.get_rank()
+    send_offsets = np.concatenate([[0], np.cumsum(send_counts)])
+    recv_offsets = np.concatenate([[0], np.cumsum(recv_counts)])
+    reqs = []
+    tensor = tensor.contiguous()
+    for peer in range(world):
+        if peer == rank:
+            continue
+        chunk = tensor[send_offsets[peer] : send_offsets[peer + 1]]
+        if chunk.numel():
+            reqs.append(dist.isend(chunk.clone(), dst=peer))
+    for peer in range(world):
+        if peer == rank:
+            out[recv_offsets[peer] : recv_offsets[peer + 1]] = tensor[
+                send_offsets[peer] : send_offsets[peer + 1]
+            ]
+            continue
+        dst = out[recv_offsets[peer] : recv_offsets[peer + 1]]
+        if dst.numel():
+            buf = dst.contiguous()
+            dist.recv(buf, src=peer)
+            dst.copy_(buf)
+    for r in reqs:
+        r.wait()
+    return out
+
+
 def all_to_all_v(tensor: torch.Tensor, send_counts, device=None):
     """Exchange row-slices of ``tensor`` between all ranks.
 

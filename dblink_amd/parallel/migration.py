@@ -144,8 +144,9 @@ def migrate_and_sort_tensors(gs, world_size: int):
 
         send_e_l = [int(x) for x in send_e.cpu()]
         send_r_l = [int(x) for x in send_r.cpu()]
-        new_ep, recv_e = comm.all_to_all_v(ent_pack.contiguous(), send_e_l)
-        new_rp, recv_r = comm.all_to_all_v(rec_pack.contiguous(), send_r_l)
+        recv_e, recv_r = comm.exchange_counts(send_e_l, send_r_l)
+        new_ep = comm.all_to_all_payload(ent_pack.contiguous(), send_e_l, recv_e)
+        new_rp = comm.all_to_all_payload(rec_pack.contiguous(), send_r_l, recv_r)
 
         gs.ent_values = new_ep[:, :A].contiguous()
         gs.ent_part = new_ep[:, A].contiguous()
