@@ -243,6 +243,20 @@ class GpuEngine(CpuEngine):
         self._theta_pin = torch.zeros((A, F), dtype=torch.float32, pin_memory=True)
         self._empty_i64 = torch.empty(0, dtype=torch.int64, device=device)
         self._counts = torch.zeros(1 + A * F + A + 1, dtype=torch.int64, device=device)
+        # word offsets of the per-(const attr, value) entity bitmaps; sized at
+        # first use (depends on the rank's entity count)
+        const_sizes = []
+        off = 0
+        const_off = np.full(A, -1, dtype=np.int64)
+        for a, ia in enumerate(cache.indexed_attributes):
+            if ia.is_constant:
+                const_off[a] = off
+                off += ia.index.num_values  # x words-per-value at runtime
+        self._const_voff = const_off  # value-slot offsets (multiply by words)
+        self._const_total_values = off
+        self._const_off_dev = None
+        self._bitmap = None
+        self._bm_words = 0
         self._graph = None
         self._graph_key = None
         self._graph_warm = 0
@@ -379,6 +393,25 @@ class GpuEngine(CpuEngine):
         )
 
         self._mark("index", graph_safe)
+        # per-sweep constant-attribute entity bitmaps (one bit test replaces a
+        # binary search in the link intersection)
+        if self._const_total_values > 0 and not flags.sequential and not flags.collapsed_entity_ids:
+            words = (E + 31) // 32
+            if self._bitmap is None or self._bm_words != words:
+                self._bm_words = words
+                self._bitmap = torch.zeros(
+                    self._const_total_values * words, dtype=torch.int32, device=dev
+                )
+                self._const_off_dev = torch.from_numpy(
+                    np.where(self._const_voff >= 0, self._const_voff * words, -1)
+                ).to(dev)
+            else:
+                self._bitmap.zero_()
+            self.C.build_const_bitmap(gs.ent_values, self._const_off_dev, words, self._bitmap)
+        elif self._const_off_dev is None:
+            self._const_off_dev = torch.full((A,), -1, dtype=torch.int64, device=dev)
+            self._bitmap = torch.zeros(1, dtype=torch.int32, device=dev)
+
         # --- phase 1: link update --------------------------------------------
         rec_ent_new = torch.empty_like(gs.rec_ent)
         if flags.sequential or flags.collapsed_entity_ids:
@@ -404,7 +437,7 @@ class GpuEngine(CpuEngine):
                 cand_lo, cand_hi, postings, gs.ent_values,
                 ent_ptr, m.log_norm, m.voff, m.csr_row_ptr, m.csr_col, m.csr_sim,
                 m.attr_const, seed, it, rec_ent_new, gs.rec_ent, self._err,
-                small_mask, ctrl,
+                small_mask, ctrl, self._bitmap, self._const_off_dev, self._bm_words,
             )
         gs.rec_ent.copy_(rec_ent_new)
         self._mark("link", graph_safe)

@@ -20,7 +20,10 @@ void link_update(torch::Tensor rec_values, torch::Tensor rec_dist,
                  torch::Tensor csr_sim, torch::Tensor attr_const, int64_t seed,
                  int64_t iteration, torch::Tensor rec_ent_out,
                  torch::Tensor rec_ent_in, torch::Tensor error_count,
-                 torch::Tensor small_mask, torch::Tensor ctrl);
+                 torch::Tensor small_mask, torch::Tensor ctrl,
+                 torch::Tensor bitmap, torch::Tensor const_off, int64_t bm_words);
+void build_const_bitmap(torch::Tensor ent_values, torch::Tensor const_off,
+                        int64_t words, torch::Tensor bitmap);
 void link_update_dense(torch::Tensor rec_values, torch::Tensor rec_dist,
                        torch::Tensor rec_gid, torch::Tensor rec_part,
                        torch::Tensor rec_file, torch::Tensor ent_values,
@@ -87,5 +90,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("summary_loglik", &dblink::summary_loglik, "K8 log-likelihood reduction");
   m.def("kd_descent", &dblink::kd_descent, "K9a KD-tree partition reassignment");
   m.def("build_keys", &dblink::build_keys, "fused inverted-index key build");
+  m.def("build_const_bitmap", &dblink::build_const_bitmap,
+        "per-sweep constant-attribute entity bitmaps");
   m.def("summary_counts", &dblink::summary_counts, "fused summary counts + pack");
 }

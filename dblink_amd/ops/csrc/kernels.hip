@@ -35,6 +35,31 @@ constexpr int MAX_ATTRS = 16;
 constexpr uint32_t PH_LINK = 1, PH_DIST = 2, PH_VALG = 3, PH_VALM = 4;
 
 // ---------------------------------------------------------------------------
+// Constant-attribute entity bitmaps: for small-domain (constant) attributes,
+// posting-list membership in the link intersection becomes one bit test
+// instead of a binary search. Rebuilt per sweep; word layout
+// [const_off[a] + v * words + (e >> 5)].
+// ---------------------------------------------------------------------------
+
+__global__ void const_bitmap_kernel(
+    const int32_t* __restrict__ ent_values, const int64_t* __restrict__ const_off,
+    int64_t E, int A, int64_t words, uint32_t* __restrict__ bitmap) {
+  const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= E * A) return;
+  const int a = (int)(idx / E);
+  if (const_off[a] < 0) return;
+  const int64_t e = idx % E;
+  const int32_t v = ent_values[e * A + a];
+  atomicOr(&bitmap[const_off[a] + (int64_t)v * words + (e >> 5)],
+           1u << (e & 31));
+}
+
+DBL_D bool bitmap_test(const uint32_t* __restrict__ bitmap, int64_t off,
+                       int64_t words, int32_t v, int32_t e) {
+  return (bitmap[off + (int64_t)v * words + (e >> 5)] >> (e & 31)) & 1u;
+}
+
+// ---------------------------------------------------------------------------
 // K3+K4+K5: link update (PCG-I / Gibbs indexed path)
 // ---------------------------------------------------------------------------
 
@@ -54,6 +79,9 @@ __global__ void link_update_kernel(
     const int32_t* __restrict__ csr_col,
     const float* __restrict__ csr_sim,
     const uint8_t* __restrict__ attr_const,  // [A]
+    const uint32_t* __restrict__ bitmap,     // const-attr entity bitmaps
+    const int64_t* __restrict__ const_off,   // [A] word offset or -1
+    int64_t bm_words,                        // words per value = ceil(E/32)
     const uint8_t* __restrict__ small_mask,  // [R] 1 = handled by small kernel (or null)
     int64_t R, int A,
     uint64_t seed, uint32_t iteration, const int64_t* __restrict__ ctrl,
@@ -71,6 +99,7 @@ __global__ void link_update_kernel(
   // weights equally and cancel under normalization).
   int nd_n = 0, od_n = 0;
   int nd_a[MAX_ATTRS];
+  int32_t nd_x[MAX_ATTRS];
   int64_t nd_lo[MAX_ATTRS], nd_hi[MAX_ATTRS];
   int od_a[MAX_ATTRS];
   for (int a = 0; a < A; ++a) {
@@ -78,6 +107,7 @@ __global__ void link_update_kernel(
     if (x < 0) continue;
     if (!rec_dist[r * A + a]) {
       nd_a[nd_n] = a;
+      nd_x[nd_n] = x;
       nd_lo[nd_n] = cand_lo[r * A + a];
       nd_hi[nd_n] = cand_hi[r * A + a];
       ++nd_n;
@@ -96,6 +126,7 @@ __global__ void link_update_kernel(
     }
     if (best != 0) {
       int ta = nd_a[0]; nd_a[0] = nd_a[best]; nd_a[best] = ta;
+      int32_t tx = nd_x[0]; nd_x[0] = nd_x[best]; nd_x[best] = tx;
       int64_t tl = nd_lo[0]; nd_lo[0] = nd_lo[best]; nd_lo[best] = tl;
       int64_t th = nd_hi[0]; nd_hi[0] = nd_hi[best]; nd_hi[best] = th;
     }
@@ -120,9 +151,12 @@ __global__ void link_update_kernel(
   for (int64_t i = lane; i < base_n; i += WAVE) {
     int32_t e = base_postings ? postings[base_lo + i] : (int32_t)(base_lo + i);
     bool ok = true;
-    for (int j = 0; j < nd_n; ++j) {
-      if (nd_lo[j] == base_lo) continue;
-      if (!contains_i32(postings, nd_lo[j], nd_hi[j], e)) { ok = false; break; }
+    for (int j = 1; j < nd_n; ++j) {  // index 0 is the iteration base
+      const int64_t coff = const_off[nd_a[j]];
+      const bool hit = (coff >= 0)
+                           ? bitmap_test(bitmap, coff, bm_words, nd_x[j], e)
+                           : contains_i32(postings, nd_lo[j], nd_hi[j], e);
+      if (!hit) { ok = false; break; }
     }
     if (!ok) continue;
     float logw = 0.0f;
@@ -159,7 +193,9 @@ __global__ void link_update_small_kernel(
     const int32_t* __restrict__ ent_values, const float* __restrict__ log_norm,
     const int64_t* __restrict__ voff, const int64_t* __restrict__ csr_row_ptr,
     const int32_t* __restrict__ csr_col, const float* __restrict__ csr_sim,
-    const uint8_t* __restrict__ attr_const, int A, uint64_t seed,
+    const uint8_t* __restrict__ attr_const,
+    const uint32_t* __restrict__ bitmap, const int64_t* __restrict__ const_off,
+    int64_t bm_words, int A, uint64_t seed,
     uint32_t iteration, const int64_t* __restrict__ ctrl,
     int64_t* __restrict__ rec_ent_out,
     const int64_t* __restrict__ rec_ent_in, int* __restrict__ error_count) {
@@ -170,6 +206,8 @@ __global__ void link_update_small_kernel(
   const int64_t r = idx;
 
   int nd_n = 0, od_n = 0;
+  int nd_a[MAX_ATTRS];
+  int32_t nd_x[MAX_ATTRS];
   int64_t nd_lo[MAX_ATTRS], nd_hi[MAX_ATTRS];
   int od_a[MAX_ATTRS];
   int64_t base_lo = 0, base_hi = 0, base_sz = INT64_MAX;
@@ -178,6 +216,8 @@ __global__ void link_update_small_kernel(
     if (x < 0) continue;
     if (!rec_dist[r * A + a]) {
       const int64_t lo = cand_lo[r * A + a], hi = cand_hi[r * A + a];
+      nd_a[nd_n] = a;
+      nd_x[nd_n] = x;
       nd_lo[nd_n] = lo;
       nd_hi[nd_n] = hi;
       ++nd_n;
@@ -194,7 +234,11 @@ __global__ void link_update_small_kernel(
     bool ok = true;
     for (int j = 0; j < nd_n; ++j) {
       if (nd_lo[j] == base_lo) continue;
-      if (!contains_i32(postings, nd_lo[j], nd_hi[j], e)) { ok = false; break; }
+      const int64_t coff = const_off[nd_a[j]];
+      const bool hit = (coff >= 0)
+                           ? bitmap_test(bitmap, coff, bm_words, nd_x[j], e)
+                           : contains_i32(postings, nd_lo[j], nd_hi[j], e);
+      if (!hit) { ok = false; break; }
     }
     if (!ok) continue;
     float logw = 0.0f;
@@ -1155,8 +1199,10 @@ void link_update(
     torch::Tensor csr_col, torch::Tensor csr_sim, torch::Tensor attr_const,
     int64_t seed, int64_t iteration, torch::Tensor rec_ent_out,
     torch::Tensor rec_ent_in, torch::Tensor error_count,
-    torch::Tensor small_mask, torch::Tensor ctrl) {
+    torch::Tensor small_mask, torch::Tensor ctrl, torch::Tensor bitmap,
+    torch::Tensor const_off, int64_t bm_words) {
   const int64_t* ctrl_ptr = ctrl.numel() ? ctrl.data_ptr<int64_t>() : nullptr;
+  const uint32_t* bm_ptr = (const uint32_t*)bitmap.data_ptr<int32_t>();
   CHECK_GPU(rec_values);
   const int64_t R = rec_values.size(0);
   const int A = (int)rec_values.size(1);
@@ -1176,7 +1222,8 @@ void link_update(
                        ent_ptr.data_ptr<int64_t>(), log_norm.data_ptr<float>(),
                        voff.data_ptr<int64_t>(), csr_row_ptr.data_ptr<int64_t>(),
                        csr_col.data_ptr<int32_t>(), csr_sim.data_ptr<float>(),
-                       attr_const.data_ptr<uint8_t>(), mask_ptr, R, A,
+                       attr_const.data_ptr<uint8_t>(), bm_ptr,
+                       const_off.data_ptr<int64_t>(), bm_words, mask_ptr, R, A,
                        (uint64_t)seed, (uint32_t)iteration, ctrl_ptr,
                        rec_ent_out.data_ptr<int64_t>(),
                        rec_ent_in.data_ptr<int64_t>(), error_count.data_ptr<int>());
@@ -1192,11 +1239,24 @@ void link_update(
                        ent_values.data_ptr<int32_t>(), log_norm.data_ptr<float>(),
                        voff.data_ptr<int64_t>(), csr_row_ptr.data_ptr<int64_t>(),
                        csr_col.data_ptr<int32_t>(), csr_sim.data_ptr<float>(),
-                       attr_const.data_ptr<uint8_t>(), A, (uint64_t)seed,
+                       attr_const.data_ptr<uint8_t>(), bm_ptr,
+                       const_off.data_ptr<int64_t>(), bm_words, A, (uint64_t)seed,
                        (uint32_t)iteration, ctrl_ptr,
                        rec_ent_out.data_ptr<int64_t>(),
                        rec_ent_in.data_ptr<int64_t>(), error_count.data_ptr<int>());
   }
+}
+
+void build_const_bitmap(torch::Tensor ent_values, torch::Tensor const_off,
+                        int64_t words, torch::Tensor bitmap) {
+  const int64_t E = ent_values.size(0);
+  const int A = (int)ent_values.size(1);
+  if (E == 0) return;
+  dim3 grid((unsigned)((E * A + 255) / 256));
+  hipLaunchKernelGGL(const_bitmap_kernel, grid, dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(),
+                     ent_values.data_ptr<int32_t>(), const_off.data_ptr<int64_t>(),
+                     E, A, words, (uint32_t*)bitmap.data_ptr<int32_t>());
 }
 
 void link_update_dense(
