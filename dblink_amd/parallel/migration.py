@@ -6,9 +6,10 @@ Replaces the reference's Spark shuffle (``GibbsUpdates.scala:144-150`` +
 together with its linked records.
 
 Implementation: rows are grouped by destination rank with one stable argsort,
-then exchanged with six all_to_all_v calls (entity meta/values, record
-values/file/dist/gid). On RCCL this is a direct pairwise all-to-all over
-xGMI; on gloo (CPU tests) it falls back to isend/irecv.
+payloads are packed into two int32 matrices, and a migration costs one
+counts exchange plus two all-to-all-v calls. On RCCL this is a direct
+pairwise all-to-all over xGMI; on gloo (CPU tests) it falls back to
+isend/irecv.
 """
 
 from __future__ import annotations

@@ -165,3 +165,35 @@ def test_sampler_variants_run(tmp_path, sampler):
         f.readline()
         loglik = [float(line.split(",")[3]) for line in f if line.strip()]
     assert len(loglik) >= 5 and all(np.isfinite(loglik))
+
+
+def test_two_file_project(tmp_path):
+    """Two source files via the fileIdentifier column: per-(attribute, file)
+    distortion probabilities (DistortionProbs.scala:27-44 semantics)."""
+    from dblink_amd.api.project import Project, SampleStep
+    from dblink_amd.utils import hocon
+    from dblink_amd.utils.synthdata import generate
+
+    n = 120
+    cols, header = generate(n, dup_fraction=0.1, seed=6, num_files=2)
+    data = tmp_path / "two.csv"
+    with open(data, "w") as f:
+        f.write(",".join(header) + "\n")
+        for i in range(n):
+            f.write(",".join(str(cols[h][i]) for h in header) + "\n")
+    conf = CONF_TEMPLATE.format(
+        data=str(data), out=str(tmp_path / "res"), samples=5, burnin=0, thin=1,
+        cutoff=0, sampler="PCG-I", levels=0, part_attrs="",
+    ).replace('recordIdentifier : "rec_id",',
+              'recordIdentifier : "rec_id",\n        fileIdentifier : "file_id",')
+    cfg = hocon.parse_string(conf)
+    project = Project(cfg, rank=0, world_size=1)
+    assert project.cache.num_files == 2
+    import os
+    os.makedirs(project.output_path, exist_ok=True)
+    SampleStep(project, sample_size=5, resume=False, checkpoint_interval=0).execute()
+    # theta matrix has a column per file and both were updated
+    state = project.saved_state()
+    assert state.dist_probs.probs.shape == (5, 2)
+    assert np.all(state.dist_probs.probs > 0) and np.all(state.dist_probs.probs < 1)
+    assert state.summary.agg_distortions.shape == (5, 2)
