@@ -257,6 +257,22 @@ class GpuEngine(CpuEngine):
         self._const_off_dev = None
         self._bitmap = None
         self._bm_words = 0
+        # constant-attribute pairs (composite postings shrink the link
+        # iteration base when only constants are non-distorted)
+        pairs = []
+        consts = [a for a, ia in enumerate(cache.indexed_attributes) if ia.is_constant]
+        for i in range(len(consts)):
+            for j in range(i + 1, len(consts)):
+                a1, a2 = consts[i], consts[j]
+                v1 = cache.indexed_attributes[a1].index.num_values
+                v2 = cache.indexed_attributes[a2].index.num_values
+                if v1 * v2 <= 65536:
+                    pairs.append((a1, a2, v2, v1 * v2))
+        self._pair_a1 = torch.tensor([p_[0] for p_ in pairs], dtype=torch.int32, device=device)
+        self._pair_a2 = torch.tensor([p_[1] for p_ in pairs], dtype=torch.int32, device=device)
+        self._pair_v2 = torch.tensor([p_[2] for p_ in pairs], dtype=torch.int32, device=device)
+        self._pair_vmax = max((p_[3] for p_ in pairs), default=0)
+        self._num_pairs = len(pairs)
         self._graph = None
         self._graph_key = None
         self._graph_warm = 0
@@ -378,14 +394,17 @@ class GpuEngine(CpuEngine):
 
         # --- inverted index (sorted postings) --------------------------------
         if not flags.sequential and not flags.collapsed_entity_ids:
-            keys = torch.empty(A * E, dtype=torch.int64, device=dev)
-            qkeys = torch.empty(R * A, dtype=torch.int64, device=dev)
+            T = A + self._num_pairs
+            vmax = max(m.Vmax, self._pair_vmax)
+            keys = torch.empty(T * E, dtype=torch.int64, device=dev)
+            qkeys = torch.empty(R * T, dtype=torch.int64, device=dev)
             self.C.build_keys(gs.ent_part, gs.ent_values, gs.rec_part, gs.rec_values,
-                              m.Vmax, keys, qkeys)
+                              self._pair_a1, self._pair_a2, self._pair_v2,
+                              vmax, keys, qkeys)
             sorted_keys, perm = torch.sort(keys, stable=True)
             postings = (perm % E).to(torch.int32)
-            cand_lo = torch.searchsorted(sorted_keys, qkeys, right=False).view(R, A).contiguous()
-            cand_hi = torch.searchsorted(sorted_keys, qkeys, right=True).view(R, A).contiguous()
+            cand_lo = torch.searchsorted(sorted_keys, qkeys, right=False).view(R, T).contiguous()
+            cand_hi = torch.searchsorted(sorted_keys, qkeys, right=True).view(R, T).contiguous()
 
         ent_ptr = torch.searchsorted(
             gs.ent_part.to(torch.int64).contiguous(),
@@ -426,7 +445,10 @@ class GpuEngine(CpuEngine):
             # selection is a device-side mask (no host sync, graph-safe)
             if R >= 50_000:
                 nd = (gs.rec_values >= 0) & (gs.rec_dist == 0)
-                sizes = torch.where(nd, cand_hi - cand_lo, torch.full_like(cand_hi, 1 << 40))
+                sizes = torch.where(
+                    nd, cand_hi[:, :A] - cand_lo[:, :A],
+                    torch.full_like(cand_hi[:, :A], 1 << 40),
+                )
                 min_sizes = sizes.amin(dim=1)
                 has_nd = nd.any(dim=1)
                 small_mask = (has_nd & (min_sizes <= 16)).to(torch.uint8).contiguous()
@@ -438,6 +460,7 @@ class GpuEngine(CpuEngine):
                 ent_ptr, m.log_norm, m.voff, m.csr_row_ptr, m.csr_col, m.csr_sim,
                 m.attr_const, seed, it, rec_ent_new, gs.rec_ent, self._err,
                 small_mask, ctrl, self._bitmap, self._const_off_dev, self._bm_words,
+                self._pair_a1, self._pair_a2,
             )
         gs.rec_ent.copy_(rec_ent_new)
         self._mark("link", graph_safe)

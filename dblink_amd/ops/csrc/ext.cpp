@@ -21,7 +21,8 @@ void link_update(torch::Tensor rec_values, torch::Tensor rec_dist,
                  int64_t iteration, torch::Tensor rec_ent_out,
                  torch::Tensor rec_ent_in, torch::Tensor error_count,
                  torch::Tensor small_mask, torch::Tensor ctrl,
-                 torch::Tensor bitmap, torch::Tensor const_off, int64_t bm_words);
+                 torch::Tensor bitmap, torch::Tensor const_off, int64_t bm_words,
+                 torch::Tensor pair_a1, torch::Tensor pair_a2);
 void build_const_bitmap(torch::Tensor ent_values, torch::Tensor const_off,
                         int64_t words, torch::Tensor bitmap);
 void link_update_dense(torch::Tensor rec_values, torch::Tensor rec_dist,
@@ -63,7 +64,9 @@ void summary_loglik(torch::Tensor ent_values, torch::Tensor rec_values,
                     torch::Tensor csr_sim, torch::Tensor attr_const,
                     torch::Tensor out);
 void build_keys(torch::Tensor ent_part, torch::Tensor ent_values,
-                torch::Tensor rec_part, torch::Tensor rec_values, int64_t Vmax,
+                torch::Tensor rec_part, torch::Tensor rec_values,
+                torch::Tensor pair_a1, torch::Tensor pair_a2,
+                torch::Tensor pair_v2, int64_t Vmax,
                 torch::Tensor ekeys, torch::Tensor qkeys);
 void summary_counts(torch::Tensor rec_dist, torch::Tensor rec_file,
                     torch::Tensor ent_rec_ptr, int64_t E, torch::Tensor counts,

@@ -82,6 +82,8 @@ __global__ void link_update_kernel(
     const uint32_t* __restrict__ bitmap,     // const-attr entity bitmaps
     const int64_t* __restrict__ const_off,   // [A] word offset or -1
     int64_t bm_words,                        // words per value = ceil(E/32)
+    const int32_t* __restrict__ pair_a1,     // [NP] const-pair pseudo slots
+    const int32_t* __restrict__ pair_a2, int NP,
     const uint8_t* __restrict__ small_mask,  // [R] 1 = handled by small kernel (or null)
     int64_t R, int A,
     uint64_t seed, uint32_t iteration, const int64_t* __restrict__ ctrl,
@@ -97,6 +99,7 @@ __global__ void link_update_kernel(
   // Gather per-attribute candidate ranges (observed non-distorted) and the
   // observed-distorted non-constant attributes (constant od attrs scale all
   // weights equally and cancel under normalization).
+  const int T = A + NP;
   int nd_n = 0, od_n = 0;
   int nd_a[MAX_ATTRS];
   int32_t nd_x[MAX_ATTRS];
@@ -108,12 +111,22 @@ __global__ void link_update_kernel(
     if (!rec_dist[r * A + a]) {
       nd_a[nd_n] = a;
       nd_x[nd_n] = x;
-      nd_lo[nd_n] = cand_lo[r * A + a];
-      nd_hi[nd_n] = cand_hi[r * A + a];
+      nd_lo[nd_n] = cand_lo[r * T + a];
+      nd_hi[nd_n] = cand_hi[r * T + a];
       ++nd_n;
     } else if (!attr_const[a]) {
       od_a[od_n++] = a;
     }
+  }
+  // constant-attribute PAIR postings: a smaller iteration base when both
+  // constituents are observed and non-distorted
+  int64_t pair_lo = 0, pair_sz = INT64_MAX;
+  for (int t = 0; t < NP; ++t) {
+    const int a1 = pair_a1[t], a2 = pair_a2[t];
+    if (rec_values[r * A + a1] < 0 || rec_values[r * A + a2] < 0) continue;
+    if (rec_dist[r * A + a1] || rec_dist[r * A + a2]) continue;
+    const int64_t lo = cand_lo[r * T + A + t], hi = cand_hi[r * T + A + t];
+    if (hi - lo < pair_sz) { pair_sz = hi - lo; pair_lo = lo; }
   }
   // base = smallest candidate list (runtime-indexed shifting sorts would
   // push these arrays to scratch — keep a single swap)
@@ -143,6 +156,10 @@ __global__ void link_update_kernel(
     base_lo = nd_lo[0];
     base_n = nd_hi[0] - base_lo;
     base_postings = true;
+    if (pair_sz < base_n) {  // pair base implies both constituents match
+      base_lo = pair_lo;
+      base_n = pair_sz;
+    }
   }
 
   const uint64_t gid = (uint64_t)rec_gid[r];
@@ -195,7 +212,8 @@ __global__ void link_update_small_kernel(
     const int32_t* __restrict__ csr_col, const float* __restrict__ csr_sim,
     const uint8_t* __restrict__ attr_const,
     const uint32_t* __restrict__ bitmap, const int64_t* __restrict__ const_off,
-    int64_t bm_words, int A, uint64_t seed,
+    int64_t bm_words, const int32_t* __restrict__ pair_a1,
+    const int32_t* __restrict__ pair_a2, int NP, int A, uint64_t seed,
     uint32_t iteration, const int64_t* __restrict__ ctrl,
     int64_t* __restrict__ rec_ent_out,
     const int64_t* __restrict__ rec_ent_in, int* __restrict__ error_count) {
@@ -205,6 +223,7 @@ __global__ void link_update_small_kernel(
   if (ctrl != nullptr) { seed = (uint64_t)ctrl[0]; iteration = (uint32_t)ctrl[1]; }
   const int64_t r = idx;
 
+  const int T = A + NP;
   int nd_n = 0, od_n = 0;
   int nd_a[MAX_ATTRS];
   int32_t nd_x[MAX_ATTRS];
@@ -215,7 +234,7 @@ __global__ void link_update_small_kernel(
     const int32_t x = rec_values[r * A + a];
     if (x < 0) continue;
     if (!rec_dist[r * A + a]) {
-      const int64_t lo = cand_lo[r * A + a], hi = cand_hi[r * A + a];
+      const int64_t lo = cand_lo[r * T + a], hi = cand_hi[r * T + a];
       nd_a[nd_n] = a;
       nd_x[nd_n] = x;
       nd_lo[nd_n] = lo;
@@ -225,6 +244,13 @@ __global__ void link_update_small_kernel(
     } else if (!attr_const[a]) {
       od_a[od_n++] = a;
     }
+  }
+  for (int t = 0; t < NP; ++t) {
+    const int a1 = pair_a1[t], a2 = pair_a2[t];
+    if (rec_values[r * A + a1] < 0 || rec_values[r * A + a2] < 0) continue;
+    if (rec_dist[r * A + a1] || rec_dist[r * A + a2]) continue;
+    const int64_t lo = cand_lo[r * T + A + t], hi = cand_hi[r * T + A + t];
+    if (hi - lo < base_sz) { base_sz = hi - lo; base_lo = lo; base_hi = hi; }
   }
   const uint64_t gid = (uint64_t)rec_gid[r];
   float best_score = -INFINITY;
@@ -991,20 +1017,40 @@ __global__ void summary_loglik_kernel(
 __global__ void build_keys_kernel(
     const int32_t* __restrict__ ent_part, const int32_t* __restrict__ ent_values,
     const int32_t* __restrict__ rec_part, const int32_t* __restrict__ rec_values,
-    int64_t E, int64_t R, int A, int64_t Vmax,
-    int64_t* __restrict__ ekeys,   // [A * E] a-major
-    int64_t* __restrict__ qkeys) { // [R * A]
+    const int32_t* __restrict__ pair_a1,   // [NP] first attr of pseudo slot
+    const int32_t* __restrict__ pair_a2,   // [NP] second attr
+    const int32_t* __restrict__ pair_v2,   // [NP] domain size of second attr
+    int64_t E, int64_t R, int A, int NP, int64_t Vmax,
+    int64_t* __restrict__ ekeys,   // [(A+NP) * E] slot-major
+    int64_t* __restrict__ qkeys) { // [R * (A+NP)]
   const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (idx < E * A) {
-    const int a = (int)(idx / E);
+  const int T = A + NP;
+  if (idx < E * T) {
+    const int t = (int)(idx / E);
     const int64_t e = idx % E;
-    ekeys[idx] = ((int64_t)ent_part[e] * A + a) * Vmax + ent_values[e * A + a];
-  } else if (idx < E * A + R * A) {
-    const int64_t j = idx - E * A;
-    const int64_t r = j / A;
-    const int a = (int)(j % A);
-    const int32_t v = rec_values[r * A + a];
-    qkeys[j] = ((int64_t)rec_part[r] * A + a) * Vmax + (v < 0 ? 0 : v);
+    int64_t v;
+    if (t < A) {
+      v = ent_values[e * A + t];
+    } else {
+      // composite value of a constant-attribute pair
+      v = (int64_t)ent_values[e * A + pair_a1[t - A]] * pair_v2[t - A]
+          + ent_values[e * A + pair_a2[t - A]];
+    }
+    ekeys[idx] = ((int64_t)ent_part[e] * T + t) * Vmax + v;
+  } else if (idx < E * T + R * T) {
+    const int64_t j = idx - E * T;
+    const int64_t r = j / T;
+    const int t = (int)(j % T);
+    int64_t v;
+    if (t < A) {
+      const int32_t x = rec_values[r * A + t];
+      v = x < 0 ? 0 : x;
+    } else {
+      const int32_t x1 = rec_values[r * A + pair_a1[t - A]];
+      const int32_t x2 = rec_values[r * A + pair_a2[t - A]];
+      v = (x1 < 0 || x2 < 0) ? 0 : (int64_t)x1 * pair_v2[t - A] + x2;
+    }
+    qkeys[j] = ((int64_t)rec_part[r] * T + t) * Vmax + v;
   }
 }
 
@@ -1200,9 +1246,11 @@ void link_update(
     int64_t seed, int64_t iteration, torch::Tensor rec_ent_out,
     torch::Tensor rec_ent_in, torch::Tensor error_count,
     torch::Tensor small_mask, torch::Tensor ctrl, torch::Tensor bitmap,
-    torch::Tensor const_off, int64_t bm_words) {
+    torch::Tensor const_off, int64_t bm_words, torch::Tensor pair_a1,
+    torch::Tensor pair_a2) {
   const int64_t* ctrl_ptr = ctrl.numel() ? ctrl.data_ptr<int64_t>() : nullptr;
   const uint32_t* bm_ptr = (const uint32_t*)bitmap.data_ptr<int32_t>();
+  const int NP = (int)pair_a1.numel();
   CHECK_GPU(rec_values);
   const int64_t R = rec_values.size(0);
   const int A = (int)rec_values.size(1);
@@ -1223,7 +1271,9 @@ void link_update(
                        voff.data_ptr<int64_t>(), csr_row_ptr.data_ptr<int64_t>(),
                        csr_col.data_ptr<int32_t>(), csr_sim.data_ptr<float>(),
                        attr_const.data_ptr<uint8_t>(), bm_ptr,
-                       const_off.data_ptr<int64_t>(), bm_words, mask_ptr, R, A,
+                       const_off.data_ptr<int64_t>(), bm_words,
+                       pair_a1.data_ptr<int32_t>(), pair_a2.data_ptr<int32_t>(),
+                       NP, mask_ptr, R, A,
                        (uint64_t)seed, (uint32_t)iteration, ctrl_ptr,
                        rec_ent_out.data_ptr<int64_t>(),
                        rec_ent_in.data_ptr<int64_t>(), error_count.data_ptr<int>());
@@ -1240,7 +1290,9 @@ void link_update(
                        voff.data_ptr<int64_t>(), csr_row_ptr.data_ptr<int64_t>(),
                        csr_col.data_ptr<int32_t>(), csr_sim.data_ptr<float>(),
                        attr_const.data_ptr<uint8_t>(), bm_ptr,
-                       const_off.data_ptr<int64_t>(), bm_words, A, (uint64_t)seed,
+                       const_off.data_ptr<int64_t>(), bm_words,
+                       pair_a1.data_ptr<int32_t>(), pair_a2.data_ptr<int32_t>(),
+                       NP, A, (uint64_t)seed,
                        (uint32_t)iteration, ctrl_ptr,
                        rec_ent_out.data_ptr<int64_t>(),
                        rec_ent_in.data_ptr<int64_t>(), error_count.data_ptr<int>());
@@ -1457,19 +1509,24 @@ void summary_loglik(
 }
 
 void build_keys(torch::Tensor ent_part, torch::Tensor ent_values,
-                torch::Tensor rec_part, torch::Tensor rec_values, int64_t Vmax,
+                torch::Tensor rec_part, torch::Tensor rec_values,
+                torch::Tensor pair_a1, torch::Tensor pair_a2,
+                torch::Tensor pair_v2, int64_t Vmax,
                 torch::Tensor ekeys, torch::Tensor qkeys) {
   const int64_t E = ent_values.size(0);
   const int64_t R = rec_values.size(0);
   const int A = (int)ent_values.size(1);
-  const int64_t total = (E + R) * A;
+  const int NP = (int)pair_a1.numel();
+  const int64_t total = (E + R) * (A + NP);
   if (total == 0) return;
   dim3 grid((unsigned)((total + 255) / 256));
   hipLaunchKernelGGL(build_keys_kernel, grid, dim3(256), 0,
                      at::cuda::getCurrentCUDAStream(),
                      ent_part.data_ptr<int32_t>(), ent_values.data_ptr<int32_t>(),
                      rec_part.data_ptr<int32_t>(), rec_values.data_ptr<int32_t>(),
-                     E, R, A, Vmax, ekeys.data_ptr<int64_t>(),
+                     pair_a1.data_ptr<int32_t>(), pair_a2.data_ptr<int32_t>(),
+                     pair_v2.data_ptr<int32_t>(),
+                     E, R, A, NP, Vmax, ekeys.data_ptr<int64_t>(),
                      qkeys.data_ptr<int64_t>());
 }
 

@@ -210,6 +210,8 @@ def test_link_kernel_distribution():
         torch.zeros(1, dtype=torch.int32, device=DEV),
         torch.full((2,), -1, dtype=torch.int64, device=DEV),
         0,
+        torch.empty(0, dtype=torch.int32, device=DEV),
+        torch.empty(0, dtype=torch.int32, device=DEV),
     )
     assert int(err.cpu()) == 0
     sel = out.cpu().numpy()
