@@ -147,6 +147,7 @@ __global__ void link_update_kernel(
 
   int64_t base_lo, base_n;
   bool base_postings;
+  int check_from = 1;  // nd[0] is the base unless a pair range replaces it
   const int32_t p = rec_part[r];
   if (nd_n == 0) {
     base_lo = ent_ptr[p];
@@ -156,9 +157,11 @@ __global__ void link_update_kernel(
     base_lo = nd_lo[0];
     base_n = nd_hi[0] - base_lo;
     base_postings = true;
-    if (pair_sz < base_n) {  // pair base implies both constituents match
+    if (pair_sz < base_n) {  // pair base implies both constituents match,
+                             // but every nd single must still be checked
       base_lo = pair_lo;
       base_n = pair_sz;
+      check_from = 0;
     }
   }
 
@@ -168,7 +171,7 @@ __global__ void link_update_kernel(
   for (int64_t i = lane; i < base_n; i += WAVE) {
     int32_t e = base_postings ? postings[base_lo + i] : (int32_t)(base_lo + i);
     bool ok = true;
-    for (int j = 1; j < nd_n; ++j) {  // index 0 is the iteration base
+    for (int j = check_from; j < nd_n; ++j) {
       const int64_t coff = const_off[nd_a[j]];
       const bool hit = (coff >= 0)
                            ? bitmap_test(bitmap, coff, bm_words, nd_x[j], e)
