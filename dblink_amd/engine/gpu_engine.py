@@ -258,9 +258,15 @@ class GpuEngine(CpuEngine):
         self._bitmap = None
         self._bm_words = 0
         # constant-attribute pairs (composite postings shrink the link
-        # iteration base when only constants are non-distorted)
+        # iteration base when only constants are non-distorted). Off by
+        # default: measured slightly net-negative on the RLdata schema (the
+        # larger posting sort outweighs the smaller bases); enable for
+        # constant-heavy schemas with DBLINK_CONST_PAIRS=1.
         pairs = []
-        consts = [a for a, ia in enumerate(cache.indexed_attributes) if ia.is_constant]
+        consts = (
+            [a for a, ia in enumerate(cache.indexed_attributes) if ia.is_constant]
+            if os.environ.get("DBLINK_CONST_PAIRS", "") == "1" else []
+        )
         for i in range(len(consts)):
             for j in range(i + 1, len(consts)):
                 a1, a2 = consts[i], consts[j]
