@@ -148,6 +148,7 @@ class Project:
                     self.cache, self.partitioner, world_size=self.world_size, rank=self.rank
                 )
             self._engine.rec_id_of = self.rec_id_of
+            self._engine.rec_ids_array = self.table.rec_ids
         return self._engine
 
     # ---- state ---------------------------------------------------------------

@@ -581,25 +581,33 @@ class CpuEngine:
 
     # -- linkage structure (State.getLinkageStructure) -----------------------
 
+    def linkage_arrays(self, state: ChainState):
+        """Vectorized linkage structure for this rank: returns
+        (partition_ids int32 [P_local], clusters-per-partition offsets
+        int64 [P_local+1], records-per-cluster offsets int64 [C+1],
+        record gids int64 [R]) — Arrow-ready, no per-cluster Python loops."""
+        pid = state.ent_part[state.rec_ent]
+        order = np.lexsort((state.rec_ent, pid))
+        sorted_pid = pid[order]
+        sorted_ent = state.rec_ent[order]
+        gids = state.rec_gid[order]
+        cb = np.flatnonzero(np.r_[True, sorted_ent[1:] != sorted_ent[:-1]])
+        cluster_offsets = np.r_[cb, len(sorted_ent)].astype(np.int64)
+        cluster_pid = sorted_pid[cb] if len(cb) else np.empty(0, np.int32)
+        pb = np.flatnonzero(np.r_[True, cluster_pid[1:] != cluster_pid[:-1]])             if len(cluster_pid) else np.empty(0, np.int64)
+        pid_list = cluster_pid[pb] if len(pb) else np.empty(0, np.int32)
+        pid_offsets = np.r_[pb, len(cluster_pid)].astype(np.int64)
+        return pid_list.astype(np.int32), pid_offsets, cluster_offsets, gids
+
     def linkage_structure(self, state: ChainState, rec_id_of=None):
-        """{pid -> list of clusters (lists of record-id strings)} for this rank."""
+        """{pid -> list of clusters (lists of record-id strings)} for this rank
+        (dict form, used by tests and small runs)."""
         if rec_id_of is None:
             rec_id_of = getattr(self, "rec_id_of", None) or (lambda gid: str(gid))
-        ent_ptr, rec_ptr = state.partition_offsets(self.num_partitions)
-        out = {}
-        # records sorted by entity; group by entity
-        order = np.argsort(state.rec_ent, kind="stable")
-        sorted_ent = state.rec_ent[order]
-        bounds = np.flatnonzero(np.r_[True, sorted_ent[1:] != sorted_ent[:-1]])
-        cluster_of_ent = {}
-        for bi, b in enumerate(bounds):
-            e_end = bounds[bi + 1] if bi + 1 < len(bounds) else len(sorted_ent)
-            e = int(sorted_ent[b])
-            gids = state.rec_gid[order[b:e_end]]
-            cluster_of_ent[e] = [rec_id_of(int(g)) for g in gids]
-        owned_parts = set(int(p) for p in np.unique(state.ent_part))
-        for p in owned_parts:
-            out[p] = []
-        for e, cluster in cluster_of_ent.items():
-            out[int(state.ent_part[e])].append(cluster)
+        pid_list, pid_offsets, cluster_offsets, gids = self.linkage_arrays(state)
+        out = {int(p): [] for p in np.unique(state.ent_part)}
+        for pi, p in enumerate(pid_list):
+            for ci in range(int(pid_offsets[pi]), int(pid_offsets[pi + 1])):
+                lo, hi = int(cluster_offsets[ci]), int(cluster_offsets[ci + 1])
+                out[int(p)].append([rec_id_of(int(g)) for g in gids[lo:hi]])
         return out

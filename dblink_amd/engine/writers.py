@@ -60,15 +60,44 @@ class LinkageChainWriter:
         if self._buffered >= self.buffer_size:
             self.flush()
 
+    def append_arrays(self, iteration, pid_list, pid_offsets, cluster_offsets,
+                      record_ids):
+        """Vectorized append: one (iteration, pid) row per partition with the
+        linkage structure built directly as Arrow nested lists (no per-cluster
+        Python objects) — record_ids is a string array aligned with the
+        flattened cluster layout."""
+        inner = pa.ListArray.from_arrays(
+            pa.array(cluster_offsets, type=pa.int32()), pa.array(record_ids, type=pa.string())
+        )
+        outer = pa.ListArray.from_arrays(pa.array(pid_offsets, type=pa.int32()), inner)
+        batch = pa.table(
+            {
+                "iteration": pa.array([int(iteration)] * len(pid_list), type=pa.int64()),
+                "partitionId": pa.array(pid_list, type=pa.int32()),
+                "linkageStructure": outer,
+            },
+            schema=self.SCHEMA,
+        )
+        self._batches = getattr(self, "_batches", [])
+        self._batches.append(batch)
+        self._buffered += 1
+        if self._buffered >= self.buffer_size:
+            self.flush()
+
     def flush(self):
-        if not self._rows["iteration"]:
+        batches = getattr(self, "_batches", [])
+        if not self._rows["iteration"] and not batches:
             self._buffered = 0
             return
-        table = pa.table(self._rows, schema=self.SCHEMA)
+        tables = list(batches)
+        if self._rows["iteration"]:
+            tables.append(pa.table(self._rows, schema=self.SCHEMA))
+        table = pa.concat_tables(tables) if len(tables) > 1 else tables[0]
         path = os.path.join(self.dir, f"part-r{self.rank:05d}-{self._file_ctr:05d}.parquet")
         pq.write_table(table, path)
         self._file_ctr += 1
         self._rows = {"iteration": [], "partitionId": [], "linkageStructure": []}
+        self._batches = []
         self._buffered = 0
 
     def close(self):
