@@ -65,7 +65,10 @@ def barrier():
 
 def all_reduce_sum_(t: torch.Tensor):
     if is_distributed():
-        dist.all_reduce(t, op=dist.ReduceOp.SUM)
+        src, home = _to_comm(t)
+        dist.all_reduce(src, op=dist.ReduceOp.SUM)
+        if home is not None:
+            t.copy_(src.to(home))
     return t
 
 
@@ -107,6 +110,26 @@ def _backend_supports_all_to_all():
         return False
 
 
+def _comm_device():
+    """Device collectives must run on: cuda for RCCL, cpu for gloo."""
+    try:
+        backend = dist.get_backend()
+    except Exception:
+        return None
+    if backend == "nccl":
+        return torch.device("cuda", torch.cuda.current_device())
+    if backend == "gloo":
+        return torch.device("cpu")
+    return None
+
+
+def _to_comm(t):
+    dev = _comm_device()
+    if dev is not None and t.device != dev:
+        return t.to(dev), t.device
+    return t, None
+
+
 def exchange_counts(*send_count_lists):
     """Exchange K count vectors in ONE all_to_all (one sync instead of K).
 
@@ -120,31 +143,28 @@ def exchange_counts(*send_count_lists):
         [[int(lst[p]) for lst in send_count_lists] for p in range(world)],
         dtype=torch.int64,
     ).reshape(-1)
+    dev = _comm_device()
+    if dev is not None:
+        sc = sc.to(dev)
     rc = torch.empty_like(sc)
     dist.all_to_all_single(rc, sc)
     rc = rc.view(world, K).cpu()
     return tuple([int(rc[p, k]) for p in range(world)] for k in range(K))
 
 
-def all_to_all_payload(tensor: torch.Tensor, send_counts, recv_counts):
-    """Row-slice exchange with counts already known (no counts round-trip)."""
-    if not is_distributed():
-        return tensor
-    world = dist.get_world_size()
-    out_shape = (sum(recv_counts),) + tuple(tensor.shape[1:])
-    out = torch.empty(out_shape, dtype=tensor.dtype, device=tensor.device)
-    if _backend_supports_all_to_all():
-        dist.all_to_all_single(out, tensor.contiguous(), recv_counts, send_counts)
-        return out
+def _fallback_exchange(tensor, send_counts, recv_counts, out):
+    """Pairwise isend/irecv exchange (gloo); chunks bridged to the comm
+    device (gloo needs CPU buffers even when state lives on the GPU)."""
     rank = dist.get_rank()
+    world = dist.get_world_size()
     send_offsets = np.concatenate([[0], np.cumsum(send_counts)])
     recv_offsets = np.concatenate([[0], np.cumsum(recv_counts)])
     reqs = []
-    tensor = tensor.contiguous()
+    src, _ = _to_comm(tensor.contiguous())
     for peer in range(world):
         if peer == rank:
             continue
-        chunk = tensor[send_offsets[peer] : send_offsets[peer + 1]]
+        chunk = src[send_offsets[peer] : send_offsets[peer + 1]]
         if chunk.numel():
             reqs.append(dist.isend(chunk.clone(), dst=peer))
     for peer in range(world):
@@ -155,12 +175,27 @@ def all_to_all_payload(tensor: torch.Tensor, send_counts, recv_counts):
             continue
         dst = out[recv_offsets[peer] : recv_offsets[peer + 1]]
         if dst.numel():
-            buf = dst.contiguous()
+            buf = torch.empty(dst.shape, dtype=dst.dtype, device=src.device)
             dist.recv(buf, src=peer)
-            dst.copy_(buf)
+            dst.copy_(buf.to(dst.device))
     for r in reqs:
         r.wait()
     return out
+
+
+def all_to_all_payload(tensor: torch.Tensor, send_counts, recv_counts):
+    """Row-slice exchange with counts already known (no counts round-trip)."""
+    if not is_distributed():
+        return tensor
+    world = dist.get_world_size()
+    out_shape = (sum(recv_counts),) + tuple(tensor.shape[1:])
+    if _backend_supports_all_to_all():
+        src, home = _to_comm(tensor.contiguous())
+        out = torch.empty(out_shape, dtype=tensor.dtype, device=src.device)
+        dist.all_to_all_single(out, src, recv_counts, send_counts)
+        return out.to(home) if home is not None else out
+    out = torch.empty(out_shape, dtype=tensor.dtype, device=tensor.device)
+    return _fallback_exchange(tensor, send_counts, recv_counts, out)
 
 
 def all_to_all_v(tensor: torch.Tensor, send_counts, device=None):
@@ -174,38 +209,18 @@ def all_to_all_v(tensor: torch.Tensor, send_counts, device=None):
         return tensor, [int(send_counts[0])]
     world = dist.get_world_size()
     send_counts = [int(c) for c in send_counts]
-    sc = torch.tensor(send_counts, dtype=torch.int64, device=tensor.device)
+    comm_dev = _comm_device()
+    sc = torch.tensor(send_counts, dtype=torch.int64,
+                      device=comm_dev if comm_dev is not None else tensor.device)
     rc = torch.empty_like(sc)
     dist.all_to_all_single(rc, sc)
     recv_counts = [int(x) for x in rc.cpu()]
     out_shape = (sum(recv_counts),) + tuple(tensor.shape[1:])
-    out = torch.empty(out_shape, dtype=tensor.dtype, device=tensor.device)
     if _backend_supports_all_to_all():
-        dist.all_to_all_single(out, tensor.contiguous(), recv_counts, send_counts)
-        return out, recv_counts
-    # gloo fallback: pairwise isend/irecv
-    rank = dist.get_rank()
-    send_offsets = np.concatenate([[0], np.cumsum(send_counts)])
-    recv_offsets = np.concatenate([[0], np.cumsum(recv_counts)])
-    reqs = []
-    tensor = tensor.contiguous()
-    for peer in range(world):
-        if peer == rank:
-            continue
-        chunk = tensor[send_offsets[peer] : send_offsets[peer + 1]]
-        if chunk.numel():
-            reqs.append(dist.isend(chunk.clone(), dst=peer))
-    for peer in range(world):
-        if peer == rank:
-            out[recv_offsets[peer] : recv_offsets[peer + 1]] = tensor[
-                send_offsets[peer] : send_offsets[peer + 1]
-            ]
-            continue
-        dst = out[recv_offsets[peer] : recv_offsets[peer + 1]]
-        if dst.numel():
-            buf = dst.contiguous()
-            dist.recv(buf, src=peer)
-            dst.copy_(buf)
-    for r in reqs:
-        r.wait()
+        src, home = _to_comm(tensor.contiguous())
+        out = torch.empty(out_shape, dtype=tensor.dtype, device=src.device)
+        dist.all_to_all_single(out, src, recv_counts, send_counts)
+        return (out.to(home) if home is not None else out), recv_counts
+    out = torch.empty(out_shape, dtype=tensor.dtype, device=tensor.device)
+    _fallback_exchange(tensor, send_counts, recv_counts, out)
     return out, recv_counts
