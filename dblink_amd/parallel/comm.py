@@ -37,12 +37,16 @@ def init_from_env(backend=None):
     world = int(os.environ["WORLD_SIZE"])
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
     if backend is None:
+        backend = os.environ.get("DBLINK_BACKEND")
+    if backend is None:
         backend = "nccl" if torch.cuda.is_available() else "gloo"
     if not dist.is_initialized():
         dist.init_process_group(backend=backend, timeout=datetime.timedelta(minutes=10))
     if torch.cuda.is_available():
-        torch.cuda.set_device(local_rank)
-        device = torch.device("cuda", local_rank)
+        # ranks may oversubscribe devices in shared-GPU test setups
+        dev_idx = local_rank % max(torch.cuda.device_count(), 1)
+        torch.cuda.set_device(dev_idx)
+        device = torch.device("cuda", dev_idx)
     else:
         device = torch.device("cpu")
     return rank, world, device
