@@ -193,7 +193,8 @@ def main():
                 "num_files": args.num_files,
                 "parallelism": (
                     f"entity-partitioned Gibbs, {n_gpus} rank(s) over "
-                    + ("RCCL" if use_gpu else "gloo")
+                    + (torch.distributed.get_backend().upper().replace("NCCL", "RCCL")
+                       if comm.is_distributed() else ("RCCL" if use_gpu else "local"))
                 ),
                 "engine": "gpu" if use_gpu else "cpu",
             },

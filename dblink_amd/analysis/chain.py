@@ -9,6 +9,7 @@ from __future__ import annotations
 import os
 from collections import Counter, defaultdict
 
+import numpy as np
 
 from ..engine.writers import read_linkage_chain
 
@@ -116,3 +117,84 @@ def load_chain(output_path, lower_iteration_cutoff=0):
     if table is None or table.num_rows == 0:
         return None
     return table
+
+
+def _flatten_chain(table):
+    """Flatten the nested linkage column -> (record_ids, cluster_gidx,
+    iteration_per_cluster, cluster_sizes). cluster_gidx is a dense global
+    cluster-instance index across all rows."""
+    import pyarrow as pa
+
+    col = table["linkageStructure"].combine_chunks()
+    if isinstance(col, pa.ChunkedArray):
+        col = col.combine_chunks()
+    iters = np.asarray(table["iteration"].to_numpy())
+    # outer level: clusters per (iteration, pid) row
+    outer = col
+    outer_offsets = np.asarray(outer.offsets)
+    clusters = outer.flatten()  # list<string> per cluster
+    inner_offsets = np.asarray(clusters.offsets)
+    record_ids = clusters.flatten().to_numpy(zero_copy_only=False)
+    n_clusters = len(clusters)
+    cluster_sizes = np.diff(inner_offsets)
+    # iteration of each cluster: repeat row iteration by clusters-per-row
+    clusters_per_row = np.diff(outer_offsets)
+    iter_per_cluster = np.repeat(iters, clusters_per_row)
+    cluster_gidx = np.repeat(np.arange(n_clusters, dtype=np.int64), cluster_sizes)
+    return record_ids, cluster_gidx, iter_per_cluster, cluster_sizes, inner_offsets
+
+
+def most_probable_clusters_fast(table):
+    """Vectorized MPC over large chains: clusters are identified by an
+    order-independent composite hash of their member record ids (two
+    independent 64-bit hash aggregates + size — collisions are negligible).
+    Returns {record_id -> (cluster member tuple, frequency)}."""
+    import pandas as pd
+
+    record_ids, cluster_gidx, iter_per_cluster, cluster_sizes, inner_offsets = (
+        _flatten_chain(table)
+    )
+    num_samples = len(np.unique(np.asarray(table["iteration"].to_numpy())))
+    rid_ser = pd.Series(record_ids)
+    h1 = pd.util.hash_pandas_object(rid_ser, index=False).to_numpy()
+    h2 = pd.util.hash_pandas_object(rid_ser + "#2", index=False).to_numpy()
+    df = pd.DataFrame({
+        "c": cluster_gidx,
+        "h1": h1.astype(np.uint64),
+        "h2": h2.astype(np.uint64),
+    })
+    agg = df.groupby("c", sort=False).agg(
+        s1=("h1", "sum"), s2=("h2", "sum"), n=("h1", "size")
+    )
+    # composite cluster-content key
+    key = (agg["s1"].to_numpy() ^ (agg["s2"].to_numpy() * np.uint64(0x9E3779B97F4A7C15))
+           ) + agg["n"].to_numpy().astype(np.uint64)
+    # frequency of each distinct cluster content
+    kdf = pd.DataFrame({"key": key})
+    counts = kdf.groupby("key", sort=False).size()
+    freq_per_cluster = counts.loc[kdf["key"]].to_numpy() / num_samples
+    # one representative instance per distinct key (first occurrence)
+    first_idx = kdf.drop_duplicates("key").index.to_numpy()
+    rep_of_key = dict(zip(kdf["key"].to_numpy()[first_idx], first_idx))
+    # per record: best (max freq) cluster among those containing it
+    rec_df = pd.DataFrame({
+        "rid": record_ids,
+        "key": kdf["key"].to_numpy()[cluster_gidx],
+        "freq": freq_per_cluster[cluster_gidx],
+    })
+    best = rec_df.loc[rec_df.groupby("rid", sort=False)["freq"].idxmax()]
+    out = {}
+    for rid, k, f in zip(best["rid"], best["key"], best["freq"]):
+        ci = rep_of_key[k]
+        lo, hi = inner_offsets[ci], inner_offsets[ci + 1]
+        out[rid] = (frozenset(record_ids[lo:hi].tolist()), float(f))
+    return out
+
+
+def shared_most_probable_clusters_fast(table):
+    """Vectorized sMPC (LinkageChain.scala:75-109 semantics)."""
+    mpc = most_probable_clusters_fast(table)
+    agg = defaultdict(set)
+    for rid, (cluster, _) in mpc.items():
+        agg[cluster].add(rid)
+    return [set(v) for v in agg.values()]

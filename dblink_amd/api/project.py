@@ -294,7 +294,11 @@ class EvaluateStep:
         else:
             table = chain_q.load_chain(self.p.output_path, self.cutoff)
             if table is not None:
-                smpc = chain_q.shared_most_probable_clusters(table)
+                smpc = (
+                    chain_q.shared_most_probable_clusters_fast(table)
+                    if table.num_rows > 1000
+                    else chain_q.shared_most_probable_clusters(table)
+                )
                 chain_q.save_clusters_csv(smpc, smpc_path)
             else:
                 log.error("No linkage chain")
@@ -342,7 +346,11 @@ class SummarizeStep:
             elif q == "partition-sizes":
                 chain_q.save_partition_sizes(chain_q.partition_sizes(table), self.p.output_path)
             elif q == "shared-most-probable-clusters":
-                smpc = chain_q.shared_most_probable_clusters(table)
+                smpc = (
+                    chain_q.shared_most_probable_clusters_fast(table)
+                    if table.num_rows > 1000
+                    else chain_q.shared_most_probable_clusters(table)
+                )
                 chain_q.save_clusters_csv(
                     smpc, os.path.join(self.p.output_path, "shared-most-probable-clusters.csv")
                 )

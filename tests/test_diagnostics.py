@@ -41,3 +41,42 @@ def test_summarize_from_run(tmp_path):
     assert out["iterations_per_sec"] == pytest.approx(20.0, rel=0.01)
     assert out["ess_logLikelihood"] > 50
     assert "ess_logLikelihood_per_sec" in out
+
+
+def test_fast_smpc_matches_reference_impl(tmp_path):
+    """Vectorized MPC/sMPC must agree with the direct Python implementation."""
+    import pyarrow as pa
+
+    from dblink_amd.analysis.chain import (
+        most_probable_clusters,
+        most_probable_clusters_fast,
+        shared_most_probable_clusters,
+        shared_most_probable_clusters_fast,
+    )
+
+    rng = np.random.default_rng(5)
+    rows = {"iteration": [], "partitionId": [], "linkageStructure": []}
+    rec = [f"r{i}" for i in range(30)]
+    for it in range(12):
+        # random clustering of 30 records into ~12 clusters, across 2 partitions
+        labels = rng.integers(0, 12, 30)
+        clusters = [[rec[i] for i in np.flatnonzero(labels == l)] for l in range(12)]
+        clusters = [c for c in clusters if c]
+        half = len(clusters) // 2
+        for pid, cl in ((0, clusters[:half]), (1, clusters[half:])):
+            rows["iteration"].append(it)
+            rows["partitionId"].append(pid)
+            rows["linkageStructure"].append(cl)
+    table = pa.table(rows, schema=pa.schema([
+        ("iteration", pa.int64()), ("partitionId", pa.int32()),
+        ("linkageStructure", pa.list_(pa.list_(pa.string()))),
+    ]))
+    ref = most_probable_clusters(table)
+    fast = most_probable_clusters_fast(table)
+    assert set(ref) == set(fast)
+    for rid in ref:
+        assert ref[rid][0] == fast[rid][0], rid
+        assert ref[rid][1] == pytest.approx(fast[rid][1])
+    s_ref = sorted(map(sorted, shared_most_probable_clusters(table)))
+    s_fast = sorted(map(sorted, shared_most_probable_clusters_fast(table)))
+    assert s_ref == s_fast
