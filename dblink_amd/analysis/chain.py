@@ -184,10 +184,15 @@ def most_probable_clusters_fast(table):
     })
     best = rec_df.loc[rec_df.groupby("rid", sort=False)["freq"].idxmax()]
     out = {}
+    fs_cache = {}
     for rid, k, f in zip(best["rid"], best["key"], best["freq"]):
-        ci = rep_of_key[k]
-        lo, hi = inner_offsets[ci], inner_offsets[ci + 1]
-        out[rid] = (frozenset(record_ids[lo:hi].tolist()), float(f))
+        fs = fs_cache.get(k)
+        if fs is None:
+            ci = rep_of_key[k]
+            lo, hi = inner_offsets[ci], inner_offsets[ci + 1]
+            fs = frozenset(record_ids[lo:hi].tolist())
+            fs_cache[k] = fs
+        out[rid] = (fs, float(f))
     return out
 
 

@@ -625,6 +625,42 @@ class GpuEngine(CpuEngine):
         self._pack_summary(gs)
         state.summary = self._read_summary(state)
 
+    def linkage_arrays(self, state: ChainState):
+        """Device-side linkage grouping (sort + boundaries on the GPU; only
+        offsets and gids come back to the host for the Parquet writer)."""
+        gs = self._gs
+        if gs is None:
+            return super().linkage_arrays(state)
+        E = gs.E
+        pid = gs.ent_part.to(torch.int64)[gs.rec_ent]
+        key = pid * (E + 1) + gs.rec_ent
+        order = torch.argsort(key)
+        sk = key[order]
+        gids = gs.rec_gid[order]
+        ne = sk.numel()
+        if ne == 0:
+            z = np.zeros(1, dtype=np.int64)
+            return np.empty(0, np.int32), z, z, np.empty(0, np.int64)
+        cb_mask = torch.ones_like(sk, dtype=torch.bool)
+        cb_mask[1:] = sk[1:] != sk[:-1]
+        cb = torch.nonzero(cb_mask).squeeze(1)
+        cluster_pid = (sk[cb] // (E + 1)).to(torch.int32)
+        pb_mask = torch.ones_like(cluster_pid, dtype=torch.bool)
+        pb_mask[1:] = cluster_pid[1:] != cluster_pid[:-1]
+        pb = torch.nonzero(pb_mask).squeeze(1)
+        cluster_offsets = torch.cat(
+            [cb, torch.tensor([ne], device=self.device)]
+        ).cpu().numpy()
+        pid_offsets = torch.cat(
+            [pb, torch.tensor([cluster_pid.numel()], device=self.device)]
+        ).cpu().numpy()
+        return (
+            cluster_pid[pb].cpu().numpy(),
+            pid_offsets.astype(np.int64),
+            cluster_offsets.astype(np.int64),
+            gids.cpu().numpy(),
+        )
+
     def linkage_structure(self, state: ChainState, rec_id_of=None):
         self.sync_state(state)
         return super().linkage_structure(state, rec_id_of)

@@ -64,7 +64,6 @@ def sample(
     def record_sample():
         nonlocal sample_ctr
         if write_output:
-            sync()
             pid_list, pid_offsets, cluster_offsets, gids = engine.linkage_arrays(state)
             resolver = getattr(engine, "rec_ids_array", None)
             if resolver is not None:
