@@ -169,15 +169,6 @@ class GpuStateTensors:
     """Device mirror of ChainState's mutable arrays."""
 
     FIELDS = ("ent_values", "ent_part", "rec_values", "rec_file", "rec_ent", "rec_dist", "rec_gid")
-    DTYPES = {
-        "ent_values": torch.int32,
-        "ent_part": torch.int32,
-        "rec_values": torch.int32,
-        "rec_file": torch.int32,
-        "rec_ent": torch.int64,
-        "rec_dist": torch.uint8,
-        "rec_gid": torch.int64,
-    }
 
     def __init__(self, state: ChainState, device):
         for f in self.FIELDS:

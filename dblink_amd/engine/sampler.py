@@ -65,18 +65,22 @@ def sample(
         nonlocal sample_ctr
         if write_output:
             pid_list, pid_offsets, cluster_offsets, gids = engine.linkage_arrays(state)
-            resolver = getattr(engine, "rec_ids_array", None)
-            if resolver is not None:
-                record_ids = resolver[gids]
+            dictionary = getattr(engine, "rec_ids_array", None)
+            if dictionary is not None:
+                # dictionary-encoded: indices are the gids, strings stored once
+                linkage_writer.append_arrays(
+                    state.iteration, pid_list, pid_offsets, cluster_offsets,
+                    gids, id_dictionary=dictionary,
+                )
             else:
                 rec_id_of = getattr(engine, "rec_id_of", None)
                 if rec_id_of is not None:
                     record_ids = [rec_id_of(int(g)) for g in gids]
                 else:
                     record_ids = gids.astype(str)
-            linkage_writer.append_arrays(
-                state.iteration, pid_list, pid_offsets, cluster_offsets, record_ids
-            )
+                linkage_writer.append_arrays(
+                    state.iteration, pid_list, pid_offsets, cluster_offsets, record_ids
+                )
             if diagnostics_writer is not None:
                 diagnostics_writer.write_row(state)
 

@@ -15,6 +15,7 @@ from __future__ import annotations
 import os
 import time
 
+import numpy as np
 import pyarrow as pa
 import pyarrow.parquet as pq
 
@@ -61,13 +62,23 @@ class LinkageChainWriter:
             self.flush()
 
     def append_arrays(self, iteration, pid_list, pid_offsets, cluster_offsets,
-                      record_ids):
+                      record_ids, id_dictionary=None):
         """Vectorized append: one (iteration, pid) row per partition with the
         linkage structure built directly as Arrow nested lists (no per-cluster
-        Python objects) — record_ids is a string array aligned with the
-        flattened cluster layout."""
+        Python objects). ``record_ids`` is either a string array aligned with
+        the flattened cluster layout, or — when ``id_dictionary`` is given —
+        an int index array into that dictionary (stored dictionary-encoded:
+        no per-sample string materialization at all)."""
+        if id_dictionary is not None:
+            if not hasattr(self, "_dict_arr") or self._dict_arr is None:
+                self._dict_arr = pa.array(id_dictionary, type=pa.string())
+            values = pa.DictionaryArray.from_arrays(
+                pa.array(np.asarray(record_ids, dtype=np.int32)), self._dict_arr
+            ).cast(pa.string())
+        else:
+            values = pa.array(record_ids, type=pa.string())
         inner = pa.ListArray.from_arrays(
-            pa.array(cluster_offsets, type=pa.int32()), pa.array(record_ids, type=pa.string())
+            pa.array(cluster_offsets, type=pa.int32()), values
         )
         outer = pa.ListArray.from_arrays(pa.array(pid_offsets, type=pa.int32()), inner)
         batch = pa.table(

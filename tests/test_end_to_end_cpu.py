@@ -197,3 +197,33 @@ def test_two_file_project(tmp_path):
     assert state.dist_probs.probs.shape == (5, 2)
     assert np.all(state.dist_probs.probs > 0) and np.all(state.dist_probs.probs < 1)
     assert state.summary.agg_distortions.shape == (5, 2)
+
+
+def test_missing_values_and_population_size(tmp_path):
+    """Missing values (NA) flow through init, link, value and distortion
+    updates; an explicit populationSize > numRecords creates isolates."""
+    from dblink_amd.api.project import Project, SampleStep
+    from dblink_amd.utils import hocon
+    from dblink_amd.utils.synthdata import write_csv
+
+    data = str(tmp_path / "data.csv")
+    write_csv(data, 150, dup_fraction=0.1, seed=3, missing_fraction=0.15)
+    out = str(tmp_path / "res")
+    conf = CONF_TEMPLATE.format(
+        data=data, out=out, samples=6, burnin=0, thin=1, cutoff=0,
+        sampler="PCG-I", levels=0, part_attrs="",
+    ).replace("randomSeed : 319158", "randomSeed : 319158\n    populationSize : 200")
+    cfg = hocon.parse_string(conf)
+    project = Project(cfg, rank=0, world_size=1)
+    # missing values encoded as -1
+    values, _ = project.encoded_records()
+    assert (values < 0).any()
+    assert project.cache.missing_counts  # per (file, attr) counts recorded
+    import os
+    os.makedirs(project.output_path, exist_ok=True)
+    SampleStep(project, sample_size=6, resume=False, checkpoint_interval=0).execute()
+    state = project.saved_state()
+    assert state.population_size == 200
+    assert state.num_entities == 200  # 150 records + 50 forced isolates
+    assert state.summary.num_isolates >= 50
+    assert np.isfinite(state.summary.log_likelihood)
