@@ -587,3 +587,51 @@ def test_gpu_two_rank_chain():
     for p, out in zip(procs, outs):
         assert p.returncode == 0, f"worker failed:\n{out}"
     assert any('"ok": true' in o for o in outs), outs
+
+
+@gpu
+def test_gpu_missing_values_chain(tmp_path):
+    """GPU chain over data with missing values (x = -1 flows through link
+    candidate generation, k_obs counting, distortions and summaries)."""
+    import bench as b
+    from dblink_amd.engine.cpu_engine import SamplerFlags
+    from dblink_amd.engine.gpu_engine import GpuEngine
+    from dblink_amd.engine.init import deterministic_init
+    from dblink_amd.models.records import RecordsCache, RecordsTable
+    from dblink_amd.parallel.partitioning import KDTreePartitioner
+    from dblink_amd.utils.synthdata import generate
+
+    n = 400
+    cols, _ = generate(n, dup_fraction=0.1, seed=23, missing_fraction=0.15)
+    from dblink_amd.models.records import Attribute, BetaShapeParameters
+    from dblink_amd.models.similarity import ConstantSimilarityFn, LevenshteinSimilarityFn
+
+    attr_names = ["by", "bm", "bd", "fname_c1", "lname_c1"]
+    columns = [np.where(cols[a] == "NA", None, cols[a]) for a in attr_names]
+    table = RecordsTable(cols["rec_id"], cols["file_id"], columns)
+    prior = BetaShapeParameters(0.5, 50.0)
+    attrs = [
+        Attribute("by", ConstantSimilarityFn(), prior),
+        Attribute("bm", ConstantSimilarityFn(), prior),
+        Attribute("bd", ConstantSimilarityFn(), prior),
+        Attribute("fname_c1", LevenshteinSimilarityFn(7.0, 10.0), prior),
+        Attribute("lname_c1", LevenshteinSimilarityFn(7.0, 10.0), prior),
+    ]
+    cache = RecordsCache.build(table, attrs, max_cluster_size=10)
+    rec_values, rec_files = cache.transform_records(table)
+    assert (rec_values < 0).any()
+
+    partitioner = KDTreePartitioner(1, [3])
+    state = deterministic_init(rec_values, rec_files, np.arange(n, dtype=np.int64),
+                               cache, partitioner, seed=2)
+    engine = GpuEngine(cache, partitioner, device=DEV)
+    engine.initial_summary(state)
+    flags = SamplerFlags.for_sampler("PCG-I")
+    for _ in range(20):
+        engine.step(state, flags)
+    assert np.isfinite(state.summary.log_likelihood)
+    engine.sync_state(state)
+    # missing values never mutate: align the state's missing mask by gid
+    aligned = np.zeros_like(rec_values, dtype=bool)
+    aligned[state.rec_gid] = state.rec_values < 0
+    np.testing.assert_array_equal(aligned, rec_values < 0)
