@@ -9,7 +9,7 @@ cache dependence).
 
 import os
 
-from setuptools import setup
+from setuptools import find_packages, setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
@@ -36,7 +36,7 @@ ext = CUDAExtension(
 setup(
     name="dblink_amd",
     version="0.1.0",
-    packages=["dblink_amd"],
+    packages=find_packages(include=["dblink_amd", "dblink_amd.*"]),
     ext_modules=[ext],
     cmdclass={"build_ext": BuildExtension.with_options(no_python_abi_suffix=False)},
 )
