@@ -227,3 +227,35 @@ def test_missing_values_and_population_size(tmp_path):
     assert state.num_entities == 200  # 150 records + 50 forced isolates
     assert state.summary.num_isolates >= 50
     assert np.isfinite(state.summary.log_likelihood)
+
+
+def test_copy_files_step(tmp_path):
+    """copy-files step parity (ProjectStep.scala:157-179): copies named
+    outputs (files and directories) to a destination, honoring overwrite."""
+    from dblink_amd.api.project import CopyFilesStep, Project
+    from dblink_amd.utils import hocon
+    from dblink_amd.utils.synthdata import write_csv
+
+    data = str(tmp_path / "d.csv")
+    write_csv(data, 50, seed=1)
+    out = str(tmp_path / "res")
+    conf = CONF_TEMPLATE.format(data=data, out=out, samples=2, burnin=0, thin=1,
+                                cutoff=0, sampler="PCG-I", levels=0, part_attrs="")
+    cfg = hocon.parse_string(conf)
+    project = Project(cfg)
+    os.makedirs(out, exist_ok=True)
+    (tmp_path / "res" / "diagnostics.csv").write_text("iteration\n0\n")
+    os.makedirs(tmp_path / "res" / "linkage-chain.parquet", exist_ok=True)
+    (tmp_path / "res" / "linkage-chain.parquet" / "p.parquet").write_text("x")
+    dest = str(tmp_path / "backup")
+    CopyFilesStep(project, ["diagnostics.csv", "linkage-chain.parquet",
+                            "missing.txt"], dest).execute()
+    assert os.path.exists(os.path.join(dest, "diagnostics.csv"))
+    assert os.path.exists(os.path.join(dest, "linkage-chain.parquet", "p.parquet"))
+    assert not os.path.exists(os.path.join(dest, "missing.txt"))
+    # no-overwrite: modify source, copy again without overwrite -> unchanged
+    (tmp_path / "res" / "diagnostics.csv").write_text("changed")
+    CopyFilesStep(project, ["diagnostics.csv"], dest, overwrite=False).execute()
+    assert open(os.path.join(dest, "diagnostics.csv")).read() == "iteration\n0\n"
+    CopyFilesStep(project, ["diagnostics.csv"], dest, overwrite=True).execute()
+    assert open(os.path.join(dest, "diagnostics.csv")).read() == "changed"

@@ -635,3 +635,68 @@ def test_gpu_missing_values_chain(tmp_path):
     aligned = np.zeros_like(rec_values, dtype=bool)
     aligned[state.rec_gid] = state.rec_values < 0
     np.testing.assert_array_equal(aligned, rec_values < 0)
+
+
+@gpu
+def test_value_kernel_large_cluster_rare_path():
+    """k_obs > Kc: the cached power tables don't cover the cluster size, so
+    the kernel computes log Z_k by wave reduction and draws the base via a
+    dense Gumbel scan (GibbsUpdates.scala computes the distribution on
+    demand). Empirical frequencies vs exact fp64 mixture probabilities."""
+    cache, model = make_model(DEV, Kc=3)  # force the rare path at k > 3
+    a = 1
+    ia = cache.indexed_attributes[a]
+    idx = ia.index
+    V = idx.num_values
+    x = 1
+    k = 5  # > Kc
+    theta = np.array([[0.05], [0.08]])
+    model.theta.copy_(torch.from_numpy(theta).float())
+    E = 30000
+    # each entity linked to k identical observed records
+    rec_values = np.tile(np.array([[3, x]], dtype=np.int32), (E * k, 1))
+    rec_dist = np.ones((E * k, 2), dtype=np.uint8)
+    ent_vals = np.zeros((E, 2), dtype=np.int32)
+    ent_rec_ptr = (np.arange(E + 1, dtype=np.int64) * k)
+    ent_rec_idx = np.arange(E * k, dtype=np.int64)
+
+    # exact distribution: base = phi * norm^k / Z_k; pert factors multiply
+    # across the k records (identical rows)
+    base_w = idx.probs * idx.sim_norms ** k
+    Zk = base_w.sum()
+    base = base_w / Zk
+    th = 0.08
+    px = idx.probability_of(x)
+    normx = idx.sim_norms[x]
+    cols, sims = idx.sim_index.row(x)
+    w = np.zeros(V)
+    for v, es in zip(cols.tolist(), sims.tolist()):
+        f = es + (1 / th - 1) / (px * normx) if v == x else es
+        w[v] = base[v] * (f ** k - 1.0)
+    W = w.sum()
+    exact = (base + w) / (1.0 + W)
+
+    ev = _dev(ent_vals, torch.int32)
+    err = torch.zeros(1, dtype=torch.int32, device=DEV)
+    kobs = torch.zeros(E * 2, dtype=torch.int32, device=DEV)
+    kobs.view(E, 2)[:, a] = k
+    # only attr `a` pairs are active (attr 0 stays k=0 -> phi draw, ignored)
+    C.value_update(
+        _dev(rec_values, torch.int32), _dev(rec_dist, torch.uint8),
+        _dev(np.zeros(E * k, np.int32), torch.int32),
+        _dev(ent_rec_ptr, torch.int64), _dev(ent_rec_idx, torch.int64),
+        ev, model.theta, model.phi, model.log_phi, model.norm_lin, model.log_norm,
+        model.voff, model.csr_row_ptr, model.csr_col, model.csr_sim,
+        model.phi_prob, model.phi_alias, model.pow_prob, model.pow_alias,
+        model.pow_off, model.log_pow_total, model.attr_const, model.Kc,
+        1, 0, 999, 4, 0, err,
+        torch.empty(0, dtype=torch.int64, device=DEV),
+        torch.empty(0, dtype=torch.int64, device=DEV),
+        torch.empty(0, dtype=torch.int64, device=DEV),
+        model.csr_excl, model.csr_rawsum, model.z1,
+        torch.empty(0, dtype=torch.int64, device=DEV),
+        kobs,
+    )
+    got = ev.cpu().numpy()[:, a]
+    emp = np.bincount(got, minlength=V) / E
+    assert tv_distance(emp, exact) < 0.025, (emp[:8], exact[:8])
