@@ -700,3 +700,109 @@ def test_value_kernel_large_cluster_rare_path():
     got = ev.cpu().numpy()[:, a]
     emp = np.bincount(got, minlength=V) / E
     assert tv_distance(emp, exact) < 0.025, (emp[:8], exact[:8])
+
+
+@gpu
+def test_link_kernel_parity_on_realistic_state():
+    """Link-update conditionals on a REAL mid-chain state: evolve a CPU chain,
+    freeze its entity table, then for records with diverse patterns (missing
+    values, distorted names, multiple candidates) compare GPU empirical
+    selection frequencies against the exact fp64 conditional probabilities
+    computed with the CPU oracle's code path."""
+    import bench as b
+    from dblink_amd.engine.cpu_engine import (
+        CpuEngine,
+        SamplerFlags,
+        _build_inverted_index,
+        _get_possible_entities,
+    )
+    from dblink_amd.engine.gpu_engine import GpuModel
+    from dblink_amd.engine.init import deterministic_init
+    from dblink_amd.parallel.partitioning import KDTreePartitioner
+
+    n = 250
+    cache, rec_values, rec_files = b.build_cache_and_records(n, seed=71)
+    partitioner = KDTreePartitioner(0, [])
+    state = deterministic_init(rec_values, rec_files, np.arange(n, dtype=np.int64),
+                               cache, partitioner, seed=8)
+    engine = CpuEngine(cache, partitioner)
+    engine.initial_summary(state)
+    for _ in range(15):
+        engine.step(state, SamplerFlags.for_sampler("PCG-I"))
+
+    attrs = cache.indexed_attributes
+    A = len(attrs)
+    E = state.num_entities
+    inv = _build_inverted_index(state.ent_values)
+
+    def exact_probs(r):
+        cands, obs_dist = _get_possible_entities(
+            state.rec_values[r], state.rec_dist[r], inv, E, None
+        )
+        w = np.ones(len(cands))
+        for a in obs_dist:
+            ia = attrs[a]
+            if ia.is_constant:
+                continue
+            x = int(state.rec_values[r, a])
+            y = state.ent_values[cands, a]
+            w *= ia.index.sim_norms[y] * np.array(
+                [ia.index.exp_sim_of(x, int(yy)) for yy in y]
+            )
+        return cands, w / w.sum()
+
+    # pick records with diverse patterns
+    picks = []
+    for r in range(n):
+        rv, rd = state.rec_values[r], state.rec_dist[r]
+        has_missing = (rv < 0).any()
+        name_dist = rd[3] or rd[4]
+        cands, p = exact_probs(r)
+        if len(cands) >= 2 and (has_missing or name_dist):
+            picks.append((r, cands, p))
+        if len(picks) == 3:
+            break
+    assert picks, "no multi-candidate records found in the frozen state"
+
+    model = GpuModel(cache, DEV, 10)
+    N = 20000
+    for r, cands, exact in picks:
+        rec_v = np.tile(state.rec_values[r], (N, 1)).astype(np.int32)
+        rec_d = np.tile(state.rec_dist[r], (N, 1)).astype(np.uint8)
+        # posting build identical to the engine (slot-major keys)
+        keys = ((0 * A + np.repeat(np.arange(A), E)) * model.Vmax
+                + state.ent_values.T.reshape(-1)).astype(np.int64)
+        order = np.argsort(keys, kind="stable")
+        sorted_keys = keys[order]
+        postings = (order % E).astype(np.int32)
+        qk = np.array([(0 * A + a) * model.Vmax + max(int(state.rec_values[r, a]), 0)
+                       for a in range(A)], dtype=np.int64)
+        lo = np.searchsorted(sorted_keys, qk, "left")
+        hi = np.searchsorted(sorted_keys, qk, "right")
+        out = torch.empty(N, dtype=torch.int64, device=DEV)
+        err = torch.zeros(1, dtype=torch.int32, device=DEV)
+        C.link_update(
+            _dev(rec_v, torch.int32), _dev(rec_d, torch.uint8),
+            _dev(np.arange(N, dtype=np.int64), torch.int64),
+            _dev(np.zeros(N, np.int32), torch.int32),
+            _dev(np.tile(lo, (N, 1)), torch.int64), _dev(np.tile(hi, (N, 1)), torch.int64),
+            _dev(postings, torch.int32), _dev(state.ent_values, torch.int32),
+            _dev(np.array([0, E], dtype=np.int64), torch.int64),
+            model.log_norm, model.voff, model.csr_row_ptr, model.csr_col,
+            model.csr_sim, model.attr_const, 9090, 2, out,
+            _dev(np.zeros(N, np.int64), torch.int64), err,
+            torch.empty(0, dtype=torch.uint8, device=DEV),
+            torch.empty(0, dtype=torch.int64, device=DEV),
+            torch.zeros(1, dtype=torch.int32, device=DEV),
+            torch.full((A,), -1, dtype=torch.int64, device=DEV),
+            0,
+            torch.empty(0, dtype=torch.int32, device=DEV),
+            torch.empty(0, dtype=torch.int32, device=DEV),
+        )
+        assert int(err.cpu()) == 0
+        sel = out.cpu().numpy()
+        emp = np.zeros(E)
+        np.add.at(emp, sel, 1.0 / N)
+        full = np.zeros(E)
+        full[cands] = exact
+        assert tv_distance(emp, full) < 0.025, (r, emp[cands][:6], exact[:6])
