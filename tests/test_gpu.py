@@ -735,47 +735,52 @@ def test_link_kernel_parity_on_realistic_state():
     E = state.num_entities
     inv = _build_inverted_index(state.ent_values)
 
-    def exact_probs(r):
-        cands, obs_dist = _get_possible_entities(
-            state.rec_values[r], state.rec_dist[r], inv, E, None
-        )
+    def exact_probs(rv, rd):
+        cands, obs_dist = _get_possible_entities(rv, rd, inv, E, None)
         w = np.ones(len(cands))
         for a in obs_dist:
             ia = attrs[a]
             if ia.is_constant:
                 continue
-            x = int(state.rec_values[r, a])
+            x = int(rv[a])
             y = state.ent_values[cands, a]
             w *= ia.index.sim_norms[y] * np.array(
                 [ia.index.exp_sim_of(x, int(yy)) for yy in y]
             )
         return cands, w / w.sum()
 
-    # pick records with diverse patterns
+    # At the frozen stationary state every record resolves to a singleton
+    # candidate set (a vacuous check), so craft the distortion-indicator
+    # input: keep exactly one attribute non-distorted (candidates = that
+    # value's posting list, weights from BOTH Levenshtein attrs), plus one
+    # all-distorted case exercising the nd_n==0 whole-partition scan.
     picks = []
-    for r in range(n):
-        rv, rd = state.rec_values[r], state.rec_dist[r]
-        has_missing = (rv < 0).any()
-        name_dist = rd[3] or rd[4]
-        cands, p = exact_probs(r)
-        if len(cands) >= 2 and (has_missing or name_dist):
-            picks.append((r, cands, p))
-        if len(picks) == 3:
-            break
-    assert picks, "no multi-candidate records found in the frozen state"
+    for r, keep in ((5, 1), (17, 2), (42, 3)):
+        rv = state.rec_values[r]
+        assert rv[keep] >= 0
+        rd = np.ones(A, dtype=bool)
+        rd[keep] = False
+        cands, p = exact_probs(rv, rd)
+        assert len(cands) >= 2
+        picks.append((r, rv, rd, cands, p))
+    rv = state.rec_values[7]
+    rd = np.ones(A, dtype=bool)
+    cands, p = exact_probs(rv, rd)
+    assert len(cands) == E
+    picks.append((7, rv, rd, cands, p))
 
     model = GpuModel(cache, DEV, 10)
     N = 20000
-    for r, cands, exact in picks:
-        rec_v = np.tile(state.rec_values[r], (N, 1)).astype(np.int32)
-        rec_d = np.tile(state.rec_dist[r], (N, 1)).astype(np.uint8)
+    for r, rv, rd, cands, exact in picks:
+        rec_v = np.tile(rv, (N, 1)).astype(np.int32)
+        rec_d = np.tile(rd, (N, 1)).astype(np.uint8)
         # posting build identical to the engine (slot-major keys)
         keys = ((0 * A + np.repeat(np.arange(A), E)) * model.Vmax
                 + state.ent_values.T.reshape(-1)).astype(np.int64)
         order = np.argsort(keys, kind="stable")
         sorted_keys = keys[order]
         postings = (order % E).astype(np.int32)
-        qk = np.array([(0 * A + a) * model.Vmax + max(int(state.rec_values[r, a]), 0)
+        qk = np.array([(0 * A + a) * model.Vmax + max(int(rv[a]), 0)
                        for a in range(A)], dtype=np.int64)
         lo = np.searchsorted(sorted_keys, qk, "left")
         hi = np.searchsorted(sorted_keys, qk, "right")
@@ -805,4 +810,8 @@ def test_link_kernel_parity_on_realistic_state():
         np.add.at(emp, sel, 1.0 / N)
         full = np.zeros(E)
         full[cands] = exact
-        assert tv_distance(emp, full) < 0.025, (r, emp[cands][:6], exact[:6])
+        # multinomial sampling noise: E[TV] ~ 0.5 * sum sqrt(p(1-p)/N)
+        noise = 0.5 * np.sum(np.sqrt(full * (1 - full) / N))
+        assert tv_distance(emp, full) < 3 * noise + 0.01, (
+            r, tv_distance(emp, full), noise, emp[cands][:6], exact[:6]
+        )
