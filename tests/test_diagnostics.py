@@ -1,7 +1,7 @@
 import numpy as np
 import pytest
 
-from dblink_amd.analysis.diagnostics import ess, read_diagnostics, summarize
+from dblink_amd.analysis.diagnostics import ess, summarize
 
 
 def test_ess_iid():

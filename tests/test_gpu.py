@@ -9,8 +9,6 @@ exact fp64 probabilities computed in numpy. Deterministic kernels
 tolerance.
 """
 
-import math
-
 import numpy as np
 import pytest
 

@@ -1,5 +1,3 @@
-import math
-
 import pytest
 
 from dblink_amd.analysis.baselines import exact_match_clusters, near_clusters
@@ -80,3 +78,12 @@ def test_near_clusters():
     clusters = near_clusters(recs, 1)
     merged = [c for c in clusters if len(c) > 1]
     assert {"r1", "r2"} in merged
+
+
+def test_clustering_metrics_report():
+    pred = [frozenset({1, 2}), frozenset({3})]
+    true = [frozenset({1, 2, 3})]
+    m = ClusteringMetrics.compute(pred, true)
+    assert m.adj_rand_index == pytest.approx(adjusted_rand_index(pred, true))
+    s = m.mk_string()
+    assert "Adj. Rand index" in s and str(m.adj_rand_index) in s

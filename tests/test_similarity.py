@@ -1,8 +1,6 @@
 """Golden-value similarity tests, constants cross-checked against the
 reference's SimilarityFnTest.scala:25-75."""
 
-import math
-
 import pytest
 
 from dblink_amd.models.similarity import (
