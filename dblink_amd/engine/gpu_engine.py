@@ -234,20 +234,12 @@ class GpuEngine(CpuEngine):
         self._theta_pin = torch.zeros((A, F), dtype=torch.float32, pin_memory=True)
         self._empty_i64 = torch.empty(0, dtype=torch.int64, device=device)
         self._counts = torch.zeros(1 + A * F + A + 1, dtype=torch.int64, device=device)
-        # word offsets of the per-(const attr, value) entity bitmaps; sized at
-        # first use (depends on the rank's entity count)
-        const_sizes = []
-        off = 0
-        const_off = np.full(A, -1, dtype=np.int64)
-        for a, ia in enumerate(cache.indexed_attributes):
-            if ia.is_constant:
-                const_off[a] = off
-                off += ia.index.num_values  # x words-per-value at runtime
-        self._const_voff = const_off  # value-slot offsets (multiply by words)
-        self._const_total_values = off
-        self._const_off_dev = None
-        self._bitmap = None
-        self._bm_words = 0
+        # counting-sort inverted-index buffers (dense per-key histogram /
+        # prefix / cursor over (partition, slot, value)); sized at first sweep
+        self._idx_counts = None
+        self._idx_ptr = None
+        self._idx_cursor = None
+        self._idx_nk = 0
         # constant-attribute pairs (composite postings shrink the link
         # iteration base when only constants are non-distorted). Off by
         # default: measured slightly net-negative on the RLdata schema (the
@@ -389,19 +381,48 @@ class GpuEngine(CpuEngine):
         ctrl = self._ctrl
         seed, it = 0, 0  # kernels read the ctrl buffer
 
-        # --- inverted index (sorted postings) --------------------------------
+        # --- inverted index (counting sort over (partition, slot, value)) ----
+        # Posting order within a key is arbitrary: the link kernels only
+        # enumerate ranges (membership checks are entity-value compares) and
+        # Gumbel draws are keyed by entity id, so no stable sort is needed.
+        # Ranges come straight off the dense key prefix — no searchsorted.
         if not flags.sequential and not flags.collapsed_entity_ids:
             T = A + self._num_pairs
             vmax = max(m.Vmax, self._pair_vmax)
-            keys = torch.empty(T * E, dtype=torch.int64, device=dev)
-            qkeys = torch.empty(R * T, dtype=torch.int64, device=dev)
-            self.C.build_keys(gs.ent_part, gs.ent_values, gs.rec_part, gs.rec_values,
-                              self._pair_a1, self._pair_a2, self._pair_v2,
-                              vmax, keys, qkeys)
-            sorted_keys, perm = torch.sort(keys, stable=True)
-            postings = (perm % E).to(torch.int32)
-            cand_lo = torch.searchsorted(sorted_keys, qkeys, right=False).view(R, T).contiguous()
-            cand_hi = torch.searchsorted(sorted_keys, qkeys, right=True).view(R, T).contiguous()
+            nk = self.num_partitions * T * vmax
+            if nk <= (1 << 28):  # dense counters (1 GiB cap; always true in practice)
+                if self._idx_nk != nk:
+                    self._idx_nk = nk
+                    self._idx_counts = torch.zeros(nk, dtype=torch.int32, device=dev)
+                    self._idx_ptr = torch.zeros(nk + 1, dtype=torch.int64, device=dev)
+                    self._idx_cursor = torch.empty(nk, dtype=torch.int32, device=dev)
+                else:
+                    self._idx_counts.zero_()
+                self.C.postings_hist(gs.ent_part, gs.ent_values, self._pair_a1,
+                                     self._pair_a2, self._pair_v2, vmax,
+                                     self._idx_counts)
+                torch.cumsum(self._idx_counts, 0, dtype=torch.int64,
+                             out=self._idx_ptr[1:])
+                self._idx_cursor.copy_(self._idx_ptr[:-1])
+                postings = torch.empty(T * E, dtype=torch.int32, device=dev)
+                self.C.postings_scatter(gs.ent_part, gs.ent_values, self._pair_a1,
+                                        self._pair_a2, self._pair_v2, vmax,
+                                        self._idx_cursor, postings)
+                cand_lo = torch.empty((R, T), dtype=torch.int64, device=dev)
+                cand_hi = torch.empty((R, T), dtype=torch.int64, device=dev)
+                self.C.cand_ranges(gs.rec_part, gs.rec_values, self._pair_a1,
+                                   self._pair_a2, self._pair_v2, self._idx_ptr,
+                                   vmax, cand_lo, cand_hi)
+            else:  # degenerate key space: radix sort + batched searchsorted
+                keys = torch.empty(T * E, dtype=torch.int64, device=dev)
+                qkeys = torch.empty(R * T, dtype=torch.int64, device=dev)
+                self.C.build_keys(gs.ent_part, gs.ent_values, gs.rec_part,
+                                  gs.rec_values, self._pair_a1, self._pair_a2,
+                                  self._pair_v2, vmax, keys, qkeys)
+                sorted_keys, perm = torch.sort(keys)
+                postings = (perm % E).to(torch.int32)
+                cand_lo = torch.searchsorted(sorted_keys, qkeys, right=False).view(R, T).contiguous()
+                cand_hi = torch.searchsorted(sorted_keys, qkeys, right=True).view(R, T).contiguous()
 
         ent_ptr = torch.searchsorted(
             gs.ent_part.to(torch.int64).contiguous(),
@@ -409,24 +430,6 @@ class GpuEngine(CpuEngine):
         )
 
         self._mark("index", graph_safe)
-        # per-sweep constant-attribute entity bitmaps (one bit test replaces a
-        # binary search in the link intersection)
-        if self._const_total_values > 0 and not flags.sequential and not flags.collapsed_entity_ids:
-            words = (E + 31) // 32
-            if self._bitmap is None or self._bm_words != words:
-                self._bm_words = words
-                self._bitmap = torch.zeros(
-                    self._const_total_values * words, dtype=torch.int32, device=dev
-                )
-                self._const_off_dev = torch.from_numpy(
-                    np.where(self._const_voff >= 0, self._const_voff * words, -1)
-                ).to(dev)
-            else:
-                self._bitmap.zero_()
-            self.C.build_const_bitmap(gs.ent_values, self._const_off_dev, words, self._bitmap)
-        elif self._const_off_dev is None:
-            self._const_off_dev = torch.full((A,), -1, dtype=torch.int64, device=dev)
-            self._bitmap = torch.zeros(1, dtype=torch.int32, device=dev)
 
         # --- phase 1: link update --------------------------------------------
         rec_ent_new = torch.empty_like(gs.rec_ent)
@@ -456,8 +459,7 @@ class GpuEngine(CpuEngine):
                 cand_lo, cand_hi, postings, gs.ent_values,
                 ent_ptr, m.log_norm, m.voff, m.csr_row_ptr, m.csr_col, m.csr_sim,
                 m.attr_const, seed, it, rec_ent_new, gs.rec_ent, self._err,
-                small_mask, ctrl, self._bitmap, self._const_off_dev, self._bm_words,
-                self._pair_a1, self._pair_a2,
+                small_mask, ctrl, self._pair_a1, self._pair_a2,
             )
         gs.rec_ent.copy_(rec_ent_new)
         self._mark("link", graph_safe)

@@ -21,10 +21,18 @@ void link_update(torch::Tensor rec_values, torch::Tensor rec_dist,
                  int64_t iteration, torch::Tensor rec_ent_out,
                  torch::Tensor rec_ent_in, torch::Tensor error_count,
                  torch::Tensor small_mask, torch::Tensor ctrl,
-                 torch::Tensor bitmap, torch::Tensor const_off, int64_t bm_words,
                  torch::Tensor pair_a1, torch::Tensor pair_a2);
-void build_const_bitmap(torch::Tensor ent_values, torch::Tensor const_off,
-                        int64_t words, torch::Tensor bitmap);
+void postings_hist(torch::Tensor ent_part, torch::Tensor ent_values,
+                   torch::Tensor pair_a1, torch::Tensor pair_a2,
+                   torch::Tensor pair_v2, int64_t Vmax, torch::Tensor counts);
+void postings_scatter(torch::Tensor ent_part, torch::Tensor ent_values,
+                      torch::Tensor pair_a1, torch::Tensor pair_a2,
+                      torch::Tensor pair_v2, int64_t Vmax, torch::Tensor cursor,
+                      torch::Tensor postings);
+void cand_ranges(torch::Tensor rec_part, torch::Tensor rec_values,
+                 torch::Tensor pair_a1, torch::Tensor pair_a2,
+                 torch::Tensor pair_v2, torch::Tensor ptr, int64_t Vmax,
+                 torch::Tensor cand_lo, torch::Tensor cand_hi);
 void link_update_dense(torch::Tensor rec_values, torch::Tensor rec_dist,
                        torch::Tensor rec_gid, torch::Tensor rec_part,
                        torch::Tensor rec_file, torch::Tensor ent_values,
@@ -93,7 +101,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("summary_loglik", &dblink::summary_loglik, "K8 log-likelihood reduction");
   m.def("kd_descent", &dblink::kd_descent, "K9a KD-tree partition reassignment");
   m.def("build_keys", &dblink::build_keys, "fused inverted-index key build");
-  m.def("build_const_bitmap", &dblink::build_const_bitmap,
-        "per-sweep constant-attribute entity bitmaps");
+  m.def("postings_hist", &dblink::postings_hist,
+        "counting-sort index: per-key histogram");
+  m.def("postings_scatter", &dblink::postings_scatter,
+        "counting-sort index: posting scatter");
+  m.def("cand_ranges", &dblink::cand_ranges,
+        "per-record candidate ranges from the dense key prefix");
   m.def("summary_counts", &dblink::summary_counts, "fused summary counts + pack");
 }

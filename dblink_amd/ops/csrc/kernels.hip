@@ -35,32 +35,14 @@ constexpr int MAX_ATTRS = 16;
 constexpr uint32_t PH_LINK = 1, PH_DIST = 2, PH_VALG = 3, PH_VALM = 4;
 
 // ---------------------------------------------------------------------------
-// Constant-attribute entity bitmaps: for small-domain (constant) attributes,
-// posting-list membership in the link intersection becomes one bit test
-// instead of a binary search. Rebuilt per sweep; word layout
-// [const_off[a] + v * words + (e >> 5)].
-// ---------------------------------------------------------------------------
-
-__global__ void const_bitmap_kernel(
-    const int32_t* __restrict__ ent_values, const int64_t* __restrict__ const_off,
-    int64_t E, int A, int64_t words, uint32_t* __restrict__ bitmap) {
-  const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (idx >= E * A) return;
-  const int a = (int)(idx / E);
-  if (const_off[a] < 0) return;
-  const int64_t e = idx % E;
-  const int32_t v = ent_values[e * A + a];
-  atomicOr(&bitmap[const_off[a] + (int64_t)v * words + (e >> 5)],
-           1u << (e & 31));
-}
-
-DBL_D bool bitmap_test(const uint32_t* __restrict__ bitmap, int64_t off,
-                       int64_t words, int32_t v, int32_t e) {
-  return (bitmap[off + (int64_t)v * words + (e >> 5)] >> (e & 31)) & 1u;
-}
-
-// ---------------------------------------------------------------------------
 // K3+K4+K5: link update (PCG-I / Gibbs indexed path)
+//
+// Posting lists are used ONLY to enumerate the smallest candidate set;
+// membership in every other non-distorted attribute's posting list is
+// equivalent to a direct entity-value comparison (the posting list of
+// (partition, attr, v) is exactly the entities whose attr equals v), so no
+// binary search / bitmap is needed and posting order is irrelevant — which
+// is what lets the index build be an unstable counting sort.
 // ---------------------------------------------------------------------------
 
 __global__ void link_update_kernel(
@@ -79,9 +61,6 @@ __global__ void link_update_kernel(
     const int32_t* __restrict__ csr_col,
     const float* __restrict__ csr_sim,
     const uint8_t* __restrict__ attr_const,  // [A]
-    const uint32_t* __restrict__ bitmap,     // const-attr entity bitmaps
-    const int64_t* __restrict__ const_off,   // [A] word offset or -1
-    int64_t bm_words,                        // words per value = ceil(E/32)
     const int32_t* __restrict__ pair_a1,     // [NP] const-pair pseudo slots
     const int32_t* __restrict__ pair_a2, int NP,
     const uint8_t* __restrict__ small_mask,  // [R] 1 = handled by small kernel (or null)
@@ -172,11 +151,7 @@ __global__ void link_update_kernel(
     int32_t e = base_postings ? postings[base_lo + i] : (int32_t)(base_lo + i);
     bool ok = true;
     for (int j = check_from; j < nd_n; ++j) {
-      const int64_t coff = const_off[nd_a[j]];
-      const bool hit = (coff >= 0)
-                           ? bitmap_test(bitmap, coff, bm_words, nd_x[j], e)
-                           : contains_i32(postings, nd_lo[j], nd_hi[j], e);
-      if (!hit) { ok = false; break; }
+      if (ent_values[(int64_t)e * A + nd_a[j]] != nd_x[j]) { ok = false; break; }
     }
     if (!ok) continue;
     float logw = 0.0f;
@@ -214,8 +189,7 @@ __global__ void link_update_small_kernel(
     const int64_t* __restrict__ voff, const int64_t* __restrict__ csr_row_ptr,
     const int32_t* __restrict__ csr_col, const float* __restrict__ csr_sim,
     const uint8_t* __restrict__ attr_const,
-    const uint32_t* __restrict__ bitmap, const int64_t* __restrict__ const_off,
-    int64_t bm_words, const int32_t* __restrict__ pair_a1,
+    const int32_t* __restrict__ pair_a1,
     const int32_t* __restrict__ pair_a2, int NP, int A, uint64_t seed,
     uint32_t iteration, const int64_t* __restrict__ ctrl,
     int64_t* __restrict__ rec_ent_out,
@@ -263,11 +237,7 @@ __global__ void link_update_small_kernel(
     bool ok = true;
     for (int j = 0; j < nd_n; ++j) {
       if (nd_lo[j] == base_lo) continue;
-      const int64_t coff = const_off[nd_a[j]];
-      const bool hit = (coff >= 0)
-                           ? bitmap_test(bitmap, coff, bm_words, nd_x[j], e)
-                           : contains_i32(postings, nd_lo[j], nd_hi[j], e);
-      if (!hit) { ok = false; break; }
+      if (ent_values[(int64_t)e * A + nd_a[j]] != nd_x[j]) { ok = false; break; }
     }
     if (!ok) continue;
     float logw = 0.0f;
@@ -1057,6 +1027,87 @@ __global__ void build_keys_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Counting-sort inverted index (replaces radix sort + batched searchsorted):
+//   1. postings_hist_kernel:    counts[key] += 1 over all (entity, slot)
+//   2. exclusive prefix (torch.cumsum, host-issued) -> dense ptr over keys
+//   3. postings_scatter_kernel: postings[cursor[key]++] = entity
+//   4. cand_ranges_kernel:      per (record, slot) lo = ptr[key], hi = ptr[key+1]
+// Posting order within one key is arbitrary (atomic cursors): the link
+// kernels only ENUMERATE a posting range — membership tests are direct
+// entity-value comparisons and the Gumbel-max draw is keyed by entity id,
+// so the sampled distribution is order-independent.
+// Key layout matches build_keys_kernel: (part * T + t) * Vmax + v.
+// ---------------------------------------------------------------------------
+
+DBL_D int64_t posting_slot_value(const int32_t* vals, int64_t row, int A, int t,
+                                 const int32_t* pair_a1, const int32_t* pair_a2,
+                                 const int32_t* pair_v2) {
+  if (t < A) return vals[row * A + t];
+  return (int64_t)vals[row * A + pair_a1[t - A]] * pair_v2[t - A]
+         + vals[row * A + pair_a2[t - A]];
+}
+
+__global__ void postings_hist_kernel(
+    const int32_t* __restrict__ ent_part, const int32_t* __restrict__ ent_values,
+    const int32_t* __restrict__ pair_a1, const int32_t* __restrict__ pair_a2,
+    const int32_t* __restrict__ pair_v2, int64_t E, int A, int NP, int64_t Vmax,
+    int32_t* __restrict__ counts) {
+  const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int T = A + NP;
+  if (idx >= E * T) return;
+  const int t = (int)(idx / E);
+  const int64_t e = idx % E;
+  const int64_t v = posting_slot_value(ent_values, e, A, t, pair_a1, pair_a2, pair_v2);
+  atomicAdd(&counts[((int64_t)ent_part[e] * T + t) * Vmax + v], 1);
+}
+
+__global__ void postings_scatter_kernel(
+    const int32_t* __restrict__ ent_part, const int32_t* __restrict__ ent_values,
+    const int32_t* __restrict__ pair_a1, const int32_t* __restrict__ pair_a2,
+    const int32_t* __restrict__ pair_v2, int64_t E, int A, int NP, int64_t Vmax,
+    int32_t* __restrict__ cursor,        // [NK] initialized to exclusive prefix
+    int32_t* __restrict__ postings) {    // [E * T]
+  const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int T = A + NP;
+  if (idx >= E * T) return;
+  const int t = (int)(idx / E);
+  const int64_t e = idx % E;
+  const int64_t v = posting_slot_value(ent_values, e, A, t, pair_a1, pair_a2, pair_v2);
+  const int32_t pos =
+      atomicAdd(&cursor[((int64_t)ent_part[e] * T + t) * Vmax + v], 1);
+  postings[pos] = (int32_t)e;
+}
+
+__global__ void cand_ranges_kernel(
+    const int32_t* __restrict__ rec_part, const int32_t* __restrict__ rec_values,
+    const int32_t* __restrict__ pair_a1, const int32_t* __restrict__ pair_a2,
+    const int32_t* __restrict__ pair_v2, const int64_t* __restrict__ ptr,
+    int64_t R, int A, int NP, int64_t Vmax,
+    int64_t* __restrict__ cand_lo, int64_t* __restrict__ cand_hi) {  // [R * T]
+  const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int T = A + NP;
+  if (idx >= R * T) return;
+  const int64_t r = idx / T;
+  const int t = (int)(idx % T);
+  int64_t v;
+  if (t < A) {
+    v = rec_values[r * A + t];
+  } else {
+    const int32_t x1 = rec_values[r * A + pair_a1[t - A]];
+    const int32_t x2 = rec_values[r * A + pair_a2[t - A]];
+    v = (x1 < 0 || x2 < 0) ? -1 : (int64_t)x1 * pair_v2[t - A] + x2;
+  }
+  if (v < 0) {  // missing value: empty range (never consulted as a constraint)
+    cand_lo[idx] = 0;
+    cand_hi[idx] = 0;
+    return;
+  }
+  const int64_t key = ((int64_t)rec_part[r] * T + t) * Vmax + v;
+  cand_lo[idx] = ptr[key];
+  cand_hi[idx] = ptr[key + 1];
+}
+
 // Summary counts in one pass: per-record distortion histogram + per
 // (attr, file) aggregates + isolate count (from the entity->record CSR).
 __global__ void summary_counts_kernel(
@@ -1248,11 +1299,9 @@ void link_update(
     torch::Tensor csr_col, torch::Tensor csr_sim, torch::Tensor attr_const,
     int64_t seed, int64_t iteration, torch::Tensor rec_ent_out,
     torch::Tensor rec_ent_in, torch::Tensor error_count,
-    torch::Tensor small_mask, torch::Tensor ctrl, torch::Tensor bitmap,
-    torch::Tensor const_off, int64_t bm_words, torch::Tensor pair_a1,
+    torch::Tensor small_mask, torch::Tensor ctrl, torch::Tensor pair_a1,
     torch::Tensor pair_a2) {
   const int64_t* ctrl_ptr = ctrl.numel() ? ctrl.data_ptr<int64_t>() : nullptr;
-  const uint32_t* bm_ptr = (const uint32_t*)bitmap.data_ptr<int32_t>();
   const int NP = (int)pair_a1.numel();
   CHECK_GPU(rec_values);
   const int64_t R = rec_values.size(0);
@@ -1273,8 +1322,7 @@ void link_update(
                        ent_ptr.data_ptr<int64_t>(), log_norm.data_ptr<float>(),
                        voff.data_ptr<int64_t>(), csr_row_ptr.data_ptr<int64_t>(),
                        csr_col.data_ptr<int32_t>(), csr_sim.data_ptr<float>(),
-                       attr_const.data_ptr<uint8_t>(), bm_ptr,
-                       const_off.data_ptr<int64_t>(), bm_words,
+                       attr_const.data_ptr<uint8_t>(),
                        pair_a1.data_ptr<int32_t>(), pair_a2.data_ptr<int32_t>(),
                        NP, mask_ptr, R, A,
                        (uint64_t)seed, (uint32_t)iteration, ctrl_ptr,
@@ -1292,8 +1340,7 @@ void link_update(
                        ent_values.data_ptr<int32_t>(), log_norm.data_ptr<float>(),
                        voff.data_ptr<int64_t>(), csr_row_ptr.data_ptr<int64_t>(),
                        csr_col.data_ptr<int32_t>(), csr_sim.data_ptr<float>(),
-                       attr_const.data_ptr<uint8_t>(), bm_ptr,
-                       const_off.data_ptr<int64_t>(), bm_words,
+                       attr_const.data_ptr<uint8_t>(),
                        pair_a1.data_ptr<int32_t>(), pair_a2.data_ptr<int32_t>(),
                        NP, A, (uint64_t)seed,
                        (uint32_t)iteration, ctrl_ptr,
@@ -1302,16 +1349,58 @@ void link_update(
   }
 }
 
-void build_const_bitmap(torch::Tensor ent_values, torch::Tensor const_off,
-                        int64_t words, torch::Tensor bitmap) {
+void postings_hist(torch::Tensor ent_part, torch::Tensor ent_values,
+                   torch::Tensor pair_a1, torch::Tensor pair_a2,
+                   torch::Tensor pair_v2, int64_t Vmax, torch::Tensor counts) {
   const int64_t E = ent_values.size(0);
   const int A = (int)ent_values.size(1);
-  if (E == 0) return;
-  dim3 grid((unsigned)((E * A + 255) / 256));
-  hipLaunchKernelGGL(const_bitmap_kernel, grid, dim3(256), 0,
+  const int NP = (int)pair_a1.numel();
+  const int64_t n = E * (A + NP);
+  if (n == 0) return;
+  dim3 grid((unsigned)((n + 255) / 256));
+  hipLaunchKernelGGL(postings_hist_kernel, grid, dim3(256), 0,
                      at::cuda::getCurrentCUDAStream(),
-                     ent_values.data_ptr<int32_t>(), const_off.data_ptr<int64_t>(),
-                     E, A, words, (uint32_t*)bitmap.data_ptr<int32_t>());
+                     ent_part.data_ptr<int32_t>(), ent_values.data_ptr<int32_t>(),
+                     pair_a1.data_ptr<int32_t>(), pair_a2.data_ptr<int32_t>(),
+                     pair_v2.data_ptr<int32_t>(), E, A, NP, Vmax,
+                     counts.data_ptr<int32_t>());
+}
+
+void postings_scatter(torch::Tensor ent_part, torch::Tensor ent_values,
+                      torch::Tensor pair_a1, torch::Tensor pair_a2,
+                      torch::Tensor pair_v2, int64_t Vmax, torch::Tensor cursor,
+                      torch::Tensor postings) {
+  const int64_t E = ent_values.size(0);
+  const int A = (int)ent_values.size(1);
+  const int NP = (int)pair_a1.numel();
+  const int64_t n = E * (A + NP);
+  if (n == 0) return;
+  dim3 grid((unsigned)((n + 255) / 256));
+  hipLaunchKernelGGL(postings_scatter_kernel, grid, dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(),
+                     ent_part.data_ptr<int32_t>(), ent_values.data_ptr<int32_t>(),
+                     pair_a1.data_ptr<int32_t>(), pair_a2.data_ptr<int32_t>(),
+                     pair_v2.data_ptr<int32_t>(), E, A, NP, Vmax,
+                     cursor.data_ptr<int32_t>(), postings.data_ptr<int32_t>());
+}
+
+void cand_ranges(torch::Tensor rec_part, torch::Tensor rec_values,
+                 torch::Tensor pair_a1, torch::Tensor pair_a2,
+                 torch::Tensor pair_v2, torch::Tensor ptr, int64_t Vmax,
+                 torch::Tensor cand_lo, torch::Tensor cand_hi) {
+  const int64_t R = rec_values.size(0);
+  const int A = (int)rec_values.size(1);
+  const int NP = (int)pair_a1.numel();
+  const int64_t n = R * (A + NP);
+  if (n == 0) return;
+  dim3 grid((unsigned)((n + 255) / 256));
+  hipLaunchKernelGGL(cand_ranges_kernel, grid, dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(),
+                     rec_part.data_ptr<int32_t>(), rec_values.data_ptr<int32_t>(),
+                     pair_a1.data_ptr<int32_t>(), pair_a2.data_ptr<int32_t>(),
+                     pair_v2.data_ptr<int32_t>(), ptr.data_ptr<int64_t>(),
+                     R, A, NP, Vmax,
+                     cand_lo.data_ptr<int64_t>(), cand_hi.data_ptr<int64_t>());
 }
 
 void link_update_dense(
