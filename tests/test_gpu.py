@@ -205,9 +205,6 @@ def test_link_kernel_distribution():
         _dev(np.zeros(N, np.int64), torch.int64), err,
         torch.empty(0, dtype=torch.uint8, device=DEV),
         torch.empty(0, dtype=torch.int64, device=DEV),
-        torch.zeros(1, dtype=torch.int32, device=DEV),
-        torch.full((2,), -1, dtype=torch.int64, device=DEV),
-        0,
         torch.empty(0, dtype=torch.int32, device=DEV),
         torch.empty(0, dtype=torch.int32, device=DEV),
     )
