@@ -27,7 +27,11 @@ def summarize(db_path):
 
 
 if __name__ == "__main__":
+    import os
+
     pattern = sys.argv[1] if len(sys.argv) > 1 else "gpurun_out/**/*.db"
+    if os.path.isdir(pattern):
+        pattern = os.path.join(pattern, "**", "*.db")
     for db in glob.glob(pattern, recursive=True):
         print(f"== {db} ==")
         print(summarize(db))
