@@ -1048,35 +1048,93 @@ DBL_D int64_t posting_slot_value(const int32_t* vals, int64_t row, int A, int t,
          + vals[row * A + pair_a2[t - A]];
 }
 
-__global__ void postings_hist_kernel(
+// Popular values (low-cardinality constant attributes) would serialize tens
+// of thousands of same-address global atomics, so both passes pre-aggregate
+// duplicate keys in an LDS open-addressing hash per block: one global atomic
+// per DISTINCT key per block. Keys fit int32 (the dense path is capped at
+// 2^28 counters), so the table uses 32-bit LDS CAS.
+constexpr int PB_THREADS = 1024;  // items per block == threads
+constexpr int PB_TBL = 2048;      // >= 2x max distinct keys per block
+
+DBL_D int pb_hash(int32_t key) {
+  return (int)(((uint32_t)key * 2654435761u) >> 16) & (PB_TBL - 1);
+}
+
+__global__ __launch_bounds__(PB_THREADS) void postings_hist_kernel(
     const int32_t* __restrict__ ent_part, const int32_t* __restrict__ ent_values,
     const int32_t* __restrict__ pair_a1, const int32_t* __restrict__ pair_a2,
     const int32_t* __restrict__ pair_v2, int64_t E, int A, int NP, int64_t Vmax,
     int32_t* __restrict__ counts) {
+  __shared__ int32_t h_key[PB_TBL];
+  __shared__ int32_t h_cnt[PB_TBL];
+  for (int i = threadIdx.x; i < PB_TBL; i += blockDim.x) {
+    h_key[i] = -1;
+    h_cnt[i] = 0;
+  }
+  __syncthreads();
   const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   const int T = A + NP;
-  if (idx >= E * T) return;
-  const int t = (int)(idx / E);
-  const int64_t e = idx % E;
-  const int64_t v = posting_slot_value(ent_values, e, A, t, pair_a1, pair_a2, pair_v2);
-  atomicAdd(&counts[((int64_t)ent_part[e] * T + t) * Vmax + v], 1);
+  if (idx < E * T) {
+    const int t = (int)(idx / E);
+    const int64_t e = idx % E;
+    const int64_t v = posting_slot_value(ent_values, e, A, t, pair_a1, pair_a2, pair_v2);
+    const int32_t key = (int32_t)(((int64_t)ent_part[e] * T + t) * Vmax + v);
+    int slot = pb_hash(key);
+    for (;;) {
+      const int32_t prev = atomicCAS(&h_key[slot], -1, key);
+      if (prev == -1 || prev == key) {
+        atomicAdd(&h_cnt[slot], 1);
+        break;
+      }
+      slot = (slot + 1) & (PB_TBL - 1);
+    }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < PB_TBL; i += blockDim.x)
+    if (h_key[i] >= 0) atomicAdd(&counts[h_key[i]], h_cnt[i]);
 }
 
-__global__ void postings_scatter_kernel(
+__global__ __launch_bounds__(PB_THREADS) void postings_scatter_kernel(
     const int32_t* __restrict__ ent_part, const int32_t* __restrict__ ent_values,
     const int32_t* __restrict__ pair_a1, const int32_t* __restrict__ pair_a2,
     const int32_t* __restrict__ pair_v2, int64_t E, int A, int NP, int64_t Vmax,
     int32_t* __restrict__ cursor,        // [NK] initialized to exclusive prefix
     int32_t* __restrict__ postings) {    // [E * T]
+  __shared__ int32_t h_key[PB_TBL];
+  __shared__ int32_t h_cnt[PB_TBL];
+  __shared__ int32_t h_base[PB_TBL];
+  for (int i = threadIdx.x; i < PB_TBL; i += blockDim.x) {
+    h_key[i] = -1;
+    h_cnt[i] = 0;
+  }
+  __syncthreads();
   const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   const int T = A + NP;
-  if (idx >= E * T) return;
-  const int t = (int)(idx / E);
-  const int64_t e = idx % E;
-  const int64_t v = posting_slot_value(ent_values, e, A, t, pair_a1, pair_a2, pair_v2);
-  const int32_t pos =
-      atomicAdd(&cursor[((int64_t)ent_part[e] * T + t) * Vmax + v], 1);
-  postings[pos] = (int32_t)e;
+  int my_slot = -1;
+  int32_t my_rank = 0;
+  int32_t e_out = 0;
+  if (idx < E * T) {
+    const int t = (int)(idx / E);
+    const int64_t e = idx % E;
+    e_out = (int32_t)e;
+    const int64_t v = posting_slot_value(ent_values, e, A, t, pair_a1, pair_a2, pair_v2);
+    const int32_t key = (int32_t)(((int64_t)ent_part[e] * T + t) * Vmax + v);
+    int slot = pb_hash(key);
+    for (;;) {
+      const int32_t prev = atomicCAS(&h_key[slot], -1, key);
+      if (prev == -1 || prev == key) {
+        my_rank = atomicAdd(&h_cnt[slot], 1);
+        my_slot = slot;
+        break;
+      }
+      slot = (slot + 1) & (PB_TBL - 1);
+    }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < PB_TBL; i += blockDim.x)
+    if (h_key[i] >= 0) h_base[i] = atomicAdd(&cursor[h_key[i]], h_cnt[i]);
+  __syncthreads();
+  if (my_slot >= 0) postings[h_base[my_slot] + my_rank] = e_out;
 }
 
 __global__ void cand_ranges_kernel(
@@ -1357,8 +1415,9 @@ void postings_hist(torch::Tensor ent_part, torch::Tensor ent_values,
   const int NP = (int)pair_a1.numel();
   const int64_t n = E * (A + NP);
   if (n == 0) return;
-  dim3 grid((unsigned)((n + 255) / 256));
-  hipLaunchKernelGGL(postings_hist_kernel, grid, dim3(256), 0,
+  TORCH_CHECK(counts.numel() <= INT32_MAX, "dense key space exceeds int32");
+  dim3 grid((unsigned)((n + PB_THREADS - 1) / PB_THREADS));
+  hipLaunchKernelGGL(postings_hist_kernel, grid, dim3(PB_THREADS), 0,
                      at::cuda::getCurrentCUDAStream(),
                      ent_part.data_ptr<int32_t>(), ent_values.data_ptr<int32_t>(),
                      pair_a1.data_ptr<int32_t>(), pair_a2.data_ptr<int32_t>(),
@@ -1375,8 +1434,9 @@ void postings_scatter(torch::Tensor ent_part, torch::Tensor ent_values,
   const int NP = (int)pair_a1.numel();
   const int64_t n = E * (A + NP);
   if (n == 0) return;
-  dim3 grid((unsigned)((n + 255) / 256));
-  hipLaunchKernelGGL(postings_scatter_kernel, grid, dim3(256), 0,
+  TORCH_CHECK(cursor.numel() <= INT32_MAX, "dense key space exceeds int32");
+  dim3 grid((unsigned)((n + PB_THREADS - 1) / PB_THREADS));
+  hipLaunchKernelGGL(postings_scatter_kernel, grid, dim3(PB_THREADS), 0,
                      at::cuda::getCurrentCUDAStream(),
                      ent_part.data_ptr<int32_t>(), ent_values.data_ptr<int32_t>(),
                      pair_a1.data_ptr<int32_t>(), pair_a2.data_ptr<int32_t>(),
