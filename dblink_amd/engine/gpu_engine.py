@@ -5,8 +5,9 @@ Design (MI355X-first, not a port of the Spark flow):
 - all chain state lives in HBM3E as flat SoA tensors sorted by partition id;
   per-partition segments are addressed by offset tables, so ONE kernel launch
   covers every partition this rank owns (256 CUs want >>256 workgroups)
-- the inverted index is rebuilt per sweep as a sorted posting array
-  (radix sort via torch.sort) + batched searchsorted for per-record ranges
+- the inverted index is rebuilt per sweep as a dense counting sort over
+  (partition, slot, value) keys with LDS-aggregated atomics; candidate
+  ranges are read straight off the key prefix
 - link / value / distortion updates run as the fused kernels of
   ``ops/csrc/kernels.hip`` with Philox counter RNG (order-independent draws)
 - summaries reduce on-device to a small packed tensor -> one RCCL all-reduce
