@@ -331,7 +331,7 @@ class GpuEngine(CpuEngine):
                 # summary stats are partition-agnostic: pack them from the
                 # pre-migration state and overlap the small all-reduce (on
                 # its own communicator) with the migration all-to-all
-                self._pack_summary(gs)
+                self._pack_summary(gs, loglik_done=True)
                 work = comm.all_reduce_sum_async(self._packed)
                 migrate_and_sort_tensors(gs, self.world_size)
                 if work is not None:
@@ -492,10 +492,12 @@ class GpuEngine(CpuEngine):
                 torch.empty(0, dtype=torch.int32, device=dev),
             )
             self._mark("value", graph_safe)
+            self._loglik_buf.zero_()
             self.C.distortion_update(
                 gs.rec_values, gs.rec_dist, gs.rec_file, gs.rec_gid, gs.rec_ent,
                 gs.ent_values, m.theta, m.phi, m.norm_lin, m.self_expsim, m.voff,
-                m.attr_const, seed, it, ctrl,
+                m.attr_const, seed, it, ctrl, m.log_phi, m.log_norm, m.csr_row_ptr,
+                m.csr_col, m.csr_sim, self._loglik_buf,
             )
             self._mark("distortion", graph_safe)
             ent_part_new = torch.empty_like(gs.ent_part)
@@ -508,7 +510,7 @@ class GpuEngine(CpuEngine):
             self._mark("kd", graph_safe)
             if self.world_size <= 1:
                 self._local_sort_static(gs)
-                self._pack_summary(gs, ent_rec_ptr=ent_rec_ptr)
+                self._pack_summary(gs, ent_rec_ptr=ent_rec_ptr, loglik_done=True)
             return
         # kernels self-select on kobs: no host-side pair lists, no sync
         self.C.value_update(
@@ -523,11 +525,13 @@ class GpuEngine(CpuEngine):
         )
 
         self._mark("value", graph_safe)
-        # --- phase 3: distortion update --------------------------------------
+        # --- phase 3: distortion update (log-likelihood fused in) ------------
+        self._loglik_buf.zero_()
         self.C.distortion_update(
             gs.rec_values, gs.rec_dist, gs.rec_file, gs.rec_gid, gs.rec_ent,
             gs.ent_values, m.theta, m.phi, m.norm_lin, m.self_expsim, m.voff,
-            m.attr_const, seed, it, ctrl,
+            m.attr_const, seed, it, ctrl, m.log_phi, m.log_norm, m.csr_row_ptr,
+            m.csr_col, m.csr_sim, self._loglik_buf,
         )
 
         self._mark("distortion", graph_safe)
@@ -546,7 +550,7 @@ class GpuEngine(CpuEngine):
             self._mark("sort", graph_safe)
             # isolate COUNT is permutation-invariant, so the pre-sort CSR is
             # still valid for the summary counts
-            self._pack_summary(gs, ent_rec_ptr=ent_rec_ptr)
+            self._pack_summary(gs, ent_rec_ptr=ent_rec_ptr, loglik_done=True)
             self._mark("summary", graph_safe)
 
     def _local_sort_static(self, gs: GpuStateTensors):
@@ -567,16 +571,17 @@ class GpuEngine(CpuEngine):
 
     # ---- summary -------------------------------------------------------------
 
-    def _pack_summary(self, gs: GpuStateTensors, ent_rec_ptr=None):
+    def _pack_summary(self, gs: GpuStateTensors, ent_rec_ptr=None, loglik_done=False):
         m = self.model
         E = gs.E
-        self._loglik_buf.zero_()
         self._counts.zero_()
-        self.C.summary_loglik(
-            gs.ent_values, gs.rec_values, gs.rec_dist, gs.rec_ent, m.log_phi,
-            m.log_norm, m.voff, m.csr_row_ptr, m.csr_col, m.csr_sim, m.attr_const,
-            self._loglik_buf,
-        )
+        if not loglik_done:  # standalone summary (no sweep ran this iteration)
+            self._loglik_buf.zero_()
+            self.C.summary_loglik(
+                gs.ent_values, gs.rec_values, gs.rec_dist, gs.rec_ent, m.log_phi,
+                m.log_norm, m.voff, m.csr_row_ptr, m.csr_col, m.csr_sim, m.attr_const,
+                self._loglik_buf,
+            )
         if ent_rec_ptr is None:
             sorted_re, _ = torch.sort(gs.rec_ent)
             ent_rec_ptr = torch.searchsorted(

@@ -64,7 +64,10 @@ void distortion_update(torch::Tensor rec_values, torch::Tensor rec_dist,
                        torch::Tensor theta, torch::Tensor phi,
                        torch::Tensor norm_lin, torch::Tensor self_expsim,
                        torch::Tensor voff, torch::Tensor attr_const, int64_t seed,
-                       int64_t iteration, torch::Tensor ctrl);
+                       int64_t iteration, torch::Tensor ctrl, torch::Tensor log_phi,
+                       torch::Tensor log_norm, torch::Tensor csr_row_ptr,
+                       torch::Tensor csr_col, torch::Tensor csr_sim,
+                       torch::Tensor loglik);
 void summary_loglik(torch::Tensor ent_values, torch::Tensor rec_values,
                     torch::Tensor rec_dist, torch::Tensor rec_ent,
                     torch::Tensor log_phi, torch::Tensor log_norm, torch::Tensor voff,

@@ -894,7 +894,11 @@ __global__ void value_update_seq_kernel(ValueArgs args) {
 }
 
 // ---------------------------------------------------------------------------
-// K7: distortion update (element-wise over record-attribute pairs)
+// K7 (+K8): distortion update, with the log-likelihood reduction fused in.
+// The distorted-record likelihood terms depend on exactly the data this
+// kernel just touched (x, y, z), and the entity prior terms ride along as
+// extra leading indices — fusing removes a second full pass over the state
+// (summary_loglik_kernel remains for the standalone initial summary).
 // ---------------------------------------------------------------------------
 
 __global__ void distortion_update_kernel(
@@ -905,33 +909,68 @@ __global__ void distortion_update_kernel(
     const float* __restrict__ norm_lin, const float* __restrict__ self_expsim,
     const int64_t* __restrict__ voff, const uint8_t* __restrict__ attr_const,
     int64_t R, int A, int F, uint64_t seed, uint32_t iteration,
-    const int64_t* __restrict__ ctrl) {
+    const int64_t* __restrict__ ctrl,
+    const float* __restrict__ log_phi, const float* __restrict__ log_norm,
+    const int64_t* __restrict__ csr_row_ptr, const int32_t* __restrict__ csr_col,
+    const float* __restrict__ csr_sim,
+    int64_t E,                      // entity prior rows (0 when not fusing)
+    double* __restrict__ loglik) {  // nullptr = distortion only
   const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (idx >= R * A) return;
   if (ctrl != nullptr) { seed = (uint64_t)ctrl[0]; iteration = (uint32_t)ctrl[1]; }
-  const int64_t r = idx / A;
-  const int a = (int)(idx % A);
-  const int32_t x = rec_values[idx];
-  const float th = theta[a * F + rec_file[r]];
-  const float u = philox_uniform(seed, iteration, PH_DIST,
-                                 (uint64_t)rec_gid[r] * 32u + (uint64_t)a, 0);
-  uint8_t z;
-  if (x < 0) {
-    z = u < th;
-  } else {
-    const int32_t y = ent_values[rec_ent[r] * A + a];
-    if (x == y) {
-      float pr1 = th * phi[voff[a] + x];
-      if (!attr_const[a]) pr1 *= norm_lin[voff[a] + x] * self_expsim[voff[a] + x];
-      const float pr0 = 1.0f - th;
-      const float psum = pr0 + pr1;
-      const float pz = psum != 0.0f ? pr1 / psum : 0.0f;
-      z = u < pz;
+  double contrib = 0.0;
+  if (idx < E * A) {
+    const int64_t e = idx / A;
+    const int a = (int)(idx % A);
+    contrib = (double)log_phi[voff[a] + ent_values[e * A + a]];
+  } else if (idx < (E + R) * A) {
+    const int64_t j = idx - E * A;
+    const int64_t r = j / A;
+    const int a = (int)(j % A);
+    const int32_t x = rec_values[j];
+    const float th = theta[a * F + rec_file[r]];
+    const float u = philox_uniform(seed, iteration, PH_DIST,
+                                   (uint64_t)rec_gid[r] * 32u + (uint64_t)a, 0);
+    uint8_t z;
+    if (x < 0) {
+      z = u < th;
     } else {
-      z = 1;
+      const int32_t y = ent_values[rec_ent[r] * A + a];
+      if (x == y) {
+        float pr1 = th * phi[voff[a] + x];
+        if (!attr_const[a]) pr1 *= norm_lin[voff[a] + x] * self_expsim[voff[a] + x];
+        const float pr0 = 1.0f - th;
+        const float psum = pr0 + pr1;
+        const float pz = psum != 0.0f ? pr1 / psum : 0.0f;
+        z = u < pz;
+      } else {
+        z = 1;
+      }
+    }
+    rec_dist[j] = z;
+    if (loglik != nullptr && z && x >= 0) {
+      float lp = log_phi[voff[a] + x];
+      if (!attr_const[a]) {
+        const int32_t y = ent_values[rec_ent[r] * A + a];
+        lp += log_norm[voff[a] + y] +
+              sim_lookup(csr_row_ptr, csr_col, csr_sim, voff[a] + x, y);
+      }
+      contrib = (double)lp;
     }
   }
-  rec_dist[idx] = z;
+  if (loglik == nullptr) return;
+  // wave shuffle reduce, then cross-wave LDS, one atomic per block
+  for (int off = WAVE / 2; off > 0; off >>= 1)
+    contrib += __shfl_down(contrib, off, WAVE);
+  __shared__ double wsum[256 / WAVE];
+  const int wid = threadIdx.x / WAVE;
+  if ((threadIdx.x & (WAVE - 1)) == 0) wsum[wid] = contrib;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    double s = 0.0;
+    const int nw = (int)(blockDim.x / WAVE);
+    for (int i = 0; i < nw; ++i) s += wsum[i];
+    if (s != 0.0) atomicAdd(loglik, s);
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -1619,13 +1658,17 @@ void distortion_update(
     torch::Tensor rec_gid, torch::Tensor rec_ent, torch::Tensor ent_values,
     torch::Tensor theta, torch::Tensor phi, torch::Tensor norm_lin,
     torch::Tensor self_expsim, torch::Tensor voff, torch::Tensor attr_const,
-    int64_t seed, int64_t iteration, torch::Tensor ctrl) {
+    int64_t seed, int64_t iteration, torch::Tensor ctrl, torch::Tensor log_phi,
+    torch::Tensor log_norm, torch::Tensor csr_row_ptr, torch::Tensor csr_col,
+    torch::Tensor csr_sim, torch::Tensor loglik) {
   const int64_t* ctrl_ptr = ctrl.numel() ? ctrl.data_ptr<int64_t>() : nullptr;
+  const bool fuse = loglik.numel() > 0;
   const int64_t R = rec_values.size(0);
   const int A = (int)rec_values.size(1);
   const int F = (int)theta.size(1);
+  const int64_t E = fuse ? ent_values.size(0) : 0;
   if (R == 0) return;
-  const int64_t total = R * A;
+  const int64_t total = (E + R) * A;
   dim3 grid((unsigned)((total + 255) / 256));
   hipLaunchKernelGGL(distortion_update_kernel, grid, dim3(256), 0,
                      at::cuda::getCurrentCUDAStream(),
@@ -1635,7 +1678,11 @@ void distortion_update(
                      theta.data_ptr<float>(), phi.data_ptr<float>(),
                      norm_lin.data_ptr<float>(), self_expsim.data_ptr<float>(),
                      voff.data_ptr<int64_t>(), attr_const.data_ptr<uint8_t>(),
-                     R, A, F, (uint64_t)seed, (uint32_t)iteration, ctrl_ptr);
+                     R, A, F, (uint64_t)seed, (uint32_t)iteration, ctrl_ptr,
+                     log_phi.data_ptr<float>(), log_norm.data_ptr<float>(),
+                     csr_row_ptr.data_ptr<int64_t>(), csr_col.data_ptr<int32_t>(),
+                     csr_sim.data_ptr<float>(), E,
+                     fuse ? loglik.data_ptr<double>() : nullptr);
 }
 
 void summary_loglik(

@@ -131,6 +131,8 @@ def test_distortion_kernel_distribution():
         _dev(rec_ent, torch.int64), _dev(ent_values, torch.int32),
         model.theta, model.phi, model.norm_lin, model.self_expsim, model.voff,
         model.attr_const, 1234, 7, torch.empty(0, dtype=torch.int64, device=DEV),
+        model.log_phi, model.log_norm, model.csr_row_ptr, model.csr_col,
+        model.csr_sim, torch.empty(0, dtype=torch.float64, device=DEV),
     )
     z = d.cpu().numpy()
     # attr 0 missing: P(z=1) = theta
@@ -150,6 +152,8 @@ def test_distortion_kernel_distribution():
         _dev(rec_ent, torch.int64), _dev(ent_values, torch.int32),
         model.theta, model.phi, model.norm_lin, model.self_expsim, model.voff,
         model.attr_const, 99, 3, torch.empty(0, dtype=torch.int64, device=DEV),
+        model.log_phi, model.log_norm, model.csr_row_ptr, model.csr_col,
+        model.csr_sim, torch.empty(0, dtype=torch.float64, device=DEV),
     )
     assert d.cpu().numpy()[:, 1].min() == 1
 
