@@ -75,74 +75,46 @@ __global__ void link_update_kernel(
   if (small_mask != nullptr && small_mask[r]) return;
   if (ctrl != nullptr) { seed = (uint64_t)ctrl[0]; iteration = (uint32_t)ctrl[1]; }
 
-  // Gather per-attribute candidate ranges (observed non-distorted) and the
-  // observed-distorted non-constant attributes (constant od attrs scale all
-  // weights equally and cancel under normalization).
+  // Attribute classification as BITMASKS, not arrays: runtime-indexed local
+  // arrays spill to scratch (measured 208 B/lane = a scratch round-trip per
+  // membership probe). All rec_values/voff reads below are wave-uniform, so
+  // they compile to scalar-cache loads and cost nothing per candidate.
+  // Constant observed-distorted attrs scale all weights equally and cancel
+  // under normalization, so od_mask keeps only non-constant ones.
   const int T = A + NP;
-  int nd_n = 0, od_n = 0;
-  int nd_a[MAX_ATTRS];
-  int32_t nd_x[MAX_ATTRS];
-  int64_t nd_lo[MAX_ATTRS], nd_hi[MAX_ATTRS];
-  int od_a[MAX_ATTRS];
+  uint32_t nd_mask = 0, od_mask = 0;
   for (int a = 0; a < A; ++a) {
-    int32_t x = rec_values[r * A + a];
+    const int32_t x = rec_values[r * A + a];
     if (x < 0) continue;
-    if (!rec_dist[r * A + a]) {
-      nd_a[nd_n] = a;
-      nd_x[nd_n] = x;
-      nd_lo[nd_n] = cand_lo[r * T + a];
-      nd_hi[nd_n] = cand_hi[r * T + a];
-      ++nd_n;
-    } else if (!attr_const[a]) {
-      od_a[od_n++] = a;
-    }
+    if (!rec_dist[r * A + a]) nd_mask |= 1u << a;
+    else if (!attr_const[a]) od_mask |= 1u << a;
   }
-  // constant-attribute PAIR postings: a smaller iteration base when both
-  // constituents are observed and non-distorted
-  int64_t pair_lo = 0, pair_sz = INT64_MAX;
+  // base = smallest candidate list among nd attrs and (optional) const-pair
+  // pseudo slots; a pair base implies both constituents match, but every nd
+  // single must still be value-checked (check_mask keeps them all).
+  int base_attr = -1;
+  int64_t base_lo = 0, base_n = INT64_MAX;
+  for (uint32_t m = nd_mask; m;) {
+    const int a = __ffs(m) - 1;
+    m &= m - 1;
+    const int64_t lo = cand_lo[r * T + a], n = cand_hi[r * T + a] - lo;
+    if (n < base_n) { base_n = n; base_lo = lo; base_attr = a; }
+  }
   for (int t = 0; t < NP; ++t) {
     const int a1 = pair_a1[t], a2 = pair_a2[t];
-    if (rec_values[r * A + a1] < 0 || rec_values[r * A + a2] < 0) continue;
-    if (rec_dist[r * A + a1] || rec_dist[r * A + a2]) continue;
-    const int64_t lo = cand_lo[r * T + A + t], hi = cand_hi[r * T + A + t];
-    if (hi - lo < pair_sz) { pair_sz = hi - lo; pair_lo = lo; }
+    if (((nd_mask >> a1) & 1u) == 0 || ((nd_mask >> a2) & 1u) == 0) continue;
+    const int64_t lo = cand_lo[r * T + A + t], n = cand_hi[r * T + A + t] - lo;
+    if (n < base_n) { base_n = n; base_lo = lo; base_attr = -1; }
   }
-  // base = smallest candidate list (runtime-indexed shifting sorts would
-  // push these arrays to scratch — keep a single swap)
-  if (nd_n > 1) {
-    int best = 0;
-    int64_t best_sz = nd_hi[0] - nd_lo[0];
-    for (int i = 1; i < nd_n; ++i) {
-      int64_t sz = nd_hi[i] - nd_lo[i];
-      if (sz < best_sz) { best = i; best_sz = sz; }
-    }
-    if (best != 0) {
-      int ta = nd_a[0]; nd_a[0] = nd_a[best]; nd_a[best] = ta;
-      int32_t tx = nd_x[0]; nd_x[0] = nd_x[best]; nd_x[best] = tx;
-      int64_t tl = nd_lo[0]; nd_lo[0] = nd_lo[best]; nd_lo[best] = tl;
-      int64_t th = nd_hi[0]; nd_hi[0] = nd_hi[best]; nd_hi[best] = th;
-    }
-  }
-
-  int64_t base_lo, base_n;
-  bool base_postings;
-  int check_from = 1;  // nd[0] is the base unless a pair range replaces it
-  const int32_t p = rec_part[r];
-  if (nd_n == 0) {
+  bool base_postings = true;
+  if (nd_mask == 0) {
+    const int32_t p = rec_part[r];
     base_lo = ent_ptr[p];
     base_n = ent_ptr[p + 1] - base_lo;
     base_postings = false;
-  } else {
-    base_lo = nd_lo[0];
-    base_n = nd_hi[0] - base_lo;
-    base_postings = true;
-    if (pair_sz < base_n) {  // pair base implies both constituents match,
-                             // but every nd single must still be checked
-      base_lo = pair_lo;
-      base_n = pair_sz;
-      check_from = 0;
-    }
   }
+  const uint32_t check_mask =
+      base_attr >= 0 ? (nd_mask & ~(1u << base_attr)) : nd_mask;
 
   const uint64_t gid = (uint64_t)rec_gid[r];
   float best_score = -INFINITY;
@@ -150,17 +122,23 @@ __global__ void link_update_kernel(
   for (int64_t i = lane; i < base_n; i += WAVE) {
     int32_t e = base_postings ? postings[base_lo + i] : (int32_t)(base_lo + i);
     bool ok = true;
-    for (int j = check_from; j < nd_n; ++j) {
-      if (ent_values[(int64_t)e * A + nd_a[j]] != nd_x[j]) { ok = false; break; }
+    for (uint32_t m = check_mask; m;) {
+      const int a = __ffs(m) - 1;
+      m &= m - 1;
+      if (ent_values[(int64_t)e * A + a] != rec_values[r * A + a]) {
+        ok = false;
+        break;
+      }
     }
     if (!ok) continue;
     float logw = 0.0f;
-    for (int j = 0; j < od_n; ++j) {
-      const int a = od_a[j];
-      const int32_t x = rec_values[r * A + a];
+    for (uint32_t m = od_mask; m;) {
+      const int a = __ffs(m) - 1;
+      m &= m - 1;
       const int32_t y = ent_values[(int64_t)e * A + a];
       logw += log_norm[voff[a] + y] +
-              sim_lookup(csr_row_ptr, csr_col, csr_sim, voff[a] + x, y);
+              sim_lookup(csr_row_ptr, csr_col, csr_sim,
+                         voff[a] + rec_values[r * A + a], y);
     }
     float g = gumbel_from_uniform(
         philox_uniform(seed, iteration, PH_LINK, gid, (uint32_t)e));
@@ -200,53 +178,54 @@ __global__ void link_update_small_kernel(
   if (ctrl != nullptr) { seed = (uint64_t)ctrl[0]; iteration = (uint32_t)ctrl[1]; }
   const int64_t r = idx;
 
+  // Bitmask classification (see link_update_kernel: local arrays spill)
   const int T = A + NP;
-  int nd_n = 0, od_n = 0;
-  int nd_a[MAX_ATTRS];
-  int32_t nd_x[MAX_ATTRS];
-  int64_t nd_lo[MAX_ATTRS], nd_hi[MAX_ATTRS];
-  int od_a[MAX_ATTRS];
-  int64_t base_lo = 0, base_hi = 0, base_sz = INT64_MAX;
+  uint32_t nd_mask = 0, od_mask = 0;
   for (int a = 0; a < A; ++a) {
     const int32_t x = rec_values[r * A + a];
     if (x < 0) continue;
-    if (!rec_dist[r * A + a]) {
-      const int64_t lo = cand_lo[r * T + a], hi = cand_hi[r * T + a];
-      nd_a[nd_n] = a;
-      nd_x[nd_n] = x;
-      nd_lo[nd_n] = lo;
-      nd_hi[nd_n] = hi;
-      ++nd_n;
-      if (hi - lo < base_sz) { base_sz = hi - lo; base_lo = lo; base_hi = hi; }
-    } else if (!attr_const[a]) {
-      od_a[od_n++] = a;
-    }
+    if (!rec_dist[r * A + a]) nd_mask |= 1u << a;
+    else if (!attr_const[a]) od_mask |= 1u << a;
+  }
+  int base_attr = -1;
+  int64_t base_lo = 0, base_sz = INT64_MAX;
+  for (uint32_t m = nd_mask; m;) {
+    const int a = __ffs(m) - 1;
+    m &= m - 1;
+    const int64_t lo = cand_lo[r * T + a], n = cand_hi[r * T + a] - lo;
+    if (n < base_sz) { base_sz = n; base_lo = lo; base_attr = a; }
   }
   for (int t = 0; t < NP; ++t) {
     const int a1 = pair_a1[t], a2 = pair_a2[t];
-    if (rec_values[r * A + a1] < 0 || rec_values[r * A + a2] < 0) continue;
-    if (rec_dist[r * A + a1] || rec_dist[r * A + a2]) continue;
-    const int64_t lo = cand_lo[r * T + A + t], hi = cand_hi[r * T + A + t];
-    if (hi - lo < base_sz) { base_sz = hi - lo; base_lo = lo; base_hi = hi; }
+    if (((nd_mask >> a1) & 1u) == 0 || ((nd_mask >> a2) & 1u) == 0) continue;
+    const int64_t lo = cand_lo[r * T + A + t], n = cand_hi[r * T + A + t] - lo;
+    if (n < base_sz) { base_sz = n; base_lo = lo; base_attr = -1; }
   }
+  const uint32_t check_mask =
+      base_attr >= 0 ? (nd_mask & ~(1u << base_attr)) : nd_mask;
   const uint64_t gid = (uint64_t)rec_gid[r];
   float best_score = -INFINITY;
   long long best_e = -1;
-  for (int64_t i = base_lo; i < base_hi; ++i) {
+  for (int64_t i = base_lo; i < base_lo + base_sz; ++i) {
     const int32_t e = postings[i];
     bool ok = true;
-    for (int j = 0; j < nd_n; ++j) {
-      if (nd_lo[j] == base_lo) continue;
-      if (ent_values[(int64_t)e * A + nd_a[j]] != nd_x[j]) { ok = false; break; }
+    for (uint32_t m = check_mask; m;) {
+      const int a = __ffs(m) - 1;
+      m &= m - 1;
+      if (ent_values[(int64_t)e * A + a] != rec_values[r * A + a]) {
+        ok = false;
+        break;
+      }
     }
     if (!ok) continue;
     float logw = 0.0f;
-    for (int j = 0; j < od_n; ++j) {
-      const int a = od_a[j];
-      const int32_t x = rec_values[r * A + a];
+    for (uint32_t m = od_mask; m;) {
+      const int a = __ffs(m) - 1;
+      m &= m - 1;
       const int32_t y = ent_values[(int64_t)e * A + a];
       logw += log_norm[voff[a] + y] +
-              sim_lookup(csr_row_ptr, csr_col, csr_sim, voff[a] + x, y);
+              sim_lookup(csr_row_ptr, csr_col, csr_sim,
+                         voff[a] + rec_values[r * A + a], y);
     }
     const float g = gumbel_from_uniform(
         philox_uniform(seed, iteration, PH_LINK, gid, (uint32_t)e));
