@@ -178,6 +178,14 @@ class GpuStateTensors:
         self.rec_part = self.ent_part[self.rec_ent].contiguous()
         self.device = device
 
+    @classmethod
+    def like(cls, other):
+        o = cls.__new__(cls)
+        for f in cls.FIELDS + ("rec_part",):
+            setattr(o, f, torch.empty_like(getattr(other, f)))
+        o.device = other.device
+        return o
+
     def to_host(self, state: ChainState):
         for f in self.FIELDS:
             setattr(state, f, getattr(self, f).cpu().numpy())
@@ -263,9 +271,11 @@ class GpuEngine(CpuEngine):
         self._pair_v2 = torch.tensor([p_[2] for p_ in pairs], dtype=torch.int32, device=device)
         self._pair_vmax = max((p_[3] for p_ in pairs), default=0)
         self._num_pairs = len(pairs)
-        self._graph = None
+        self._graphs = {}
         self._graph_key = None
         self._graph_warm = 0
+        self._flip = 0
+        self._gs_alt = None
         self._phase_events = []  # (name, start_event, end_event)
         self._phase_totals = {}
 
@@ -302,30 +312,40 @@ class GpuEngine(CpuEngine):
         self._ctrl_pin[1] = state.iteration + 1
 
         key = (flags.collapsed_entity_ids, flags.collapsed_entity_values, flags.sequential)
-        # classification is sync-free (device-side masks), so every
-        # single-rank problem size can be captured
-        use_graphs = self._graphs_enabled
-        if use_graphs:
-            if self._graph is not None and self._graph_key == key:
-                self._graph.replay()
-            elif self._graph_warm < 2 or self._graph_key not in (None, key):
+        if self.world_size <= 1:
+            # double-buffered state: the end-of-sweep partition re-sort
+            # gathers directly into the other buffer set; graphs are captured
+            # per direction (A->B and B->A) and replayed alternately
+            if self._gs_alt is None:
+                self._gs_alt = GpuStateTensors.like(gs)
+            out = self._gs_alt
+            use_graphs = self._graphs_enabled
+            if use_graphs:
+                gkey = (key, self._flip)
                 if self._graph_key not in (None, key):
-                    self._graph = None
+                    self._graphs = {}
                     self._graph_warm = 0
-                self._sweep_body(gs, flags)
-                self._graph_warm += 1
-            else:
-                g = torch.cuda.CUDAGraph()
-                torch.cuda.synchronize()
-                with torch.cuda.graph(g):
-                    self._sweep_body(gs, flags)
-                self._graph = g
                 self._graph_key = key
-                g.replay()
+                g = self._graphs.get(gkey)
+                if g is not None:
+                    g.replay()
+                elif self._graph_warm < 2:
+                    self._sweep_body(gs, flags, gs_out=out)
+                    self._graph_warm += 1
+                else:
+                    g = torch.cuda.CUDAGraph()
+                    torch.cuda.synchronize()
+                    with torch.cuda.graph(g):
+                        self._sweep_body(gs, flags, gs_out=out)
+                    self._graphs[gkey] = g
+                    g.replay()
+            else:
+                self._sweep_body(gs, flags, graph_safe=False, gs_out=out)
+            self._gs, self._gs_alt = out, gs
+            self._flip ^= 1
         else:
-            self._graph = None
             self._sweep_body(gs, flags, graph_safe=False)
-            if self.world_size > 1 and comm.is_distributed():
+            if comm.is_distributed():
                 from ..parallel.migration import migrate_and_sort_tensors
 
                 # summary stats are partition-agnostic: pack them from the
@@ -368,7 +388,8 @@ class GpuEngine(CpuEngine):
             self._phase_events = []
         return dict(self._phase_totals)
 
-    def _sweep_body(self, gs: GpuStateTensors, flags: SamplerFlags, graph_safe=True):
+    def _sweep_body(self, gs: GpuStateTensors, flags: SamplerFlags, graph_safe=True,
+                    gs_out=None):
         """One full device-side iteration. Capture-safe for world_size == 1:
         static input/output buffers (results copied back in place), control
         values read from the device ctrl buffer, no host synchronisation."""
@@ -508,9 +529,7 @@ class GpuEngine(CpuEngine):
             )
             gs.ent_part.copy_(ent_part_new)
             self._mark("kd", graph_safe)
-            if self.world_size <= 1:
-                self._local_sort_static(gs)
-                self._pack_summary(gs, ent_rec_ptr=ent_rec_ptr, loglik_done=True)
+            self._finish_local(gs, gs_out, ent_rec_ptr, graph_safe)
             return
         # kernels self-select on kobs: no host-side pair lists, no sync
         self.C.value_update(
@@ -545,13 +564,48 @@ class GpuEngine(CpuEngine):
         gs.ent_part.copy_(ent_part_new)
         self._mark("kd", graph_safe)
 
-        if self.world_size <= 1:
+        self._finish_local(gs, gs_out, ent_rec_ptr, graph_safe)
+
+    def _finish_local(self, gs, gs_out, ent_rec_ptr, graph_safe):
+        """Single-rank sweep tail: re-sort by partition id and pack the
+        summary. isolate COUNT is permutation-invariant, so the pre-sort CSR
+        is still valid for the summary counts."""
+        if self.world_size > 1:
+            return
+        if gs_out is None:
             self._local_sort_static(gs)
-            self._mark("sort", graph_safe)
-            # isolate COUNT is permutation-invariant, so the pre-sort CSR is
-            # still valid for the summary counts
-            self._pack_summary(gs, ent_rec_ptr=ent_rec_ptr, loglik_done=True)
-            self._mark("summary", graph_safe)
+            target = gs
+        else:
+            self._sort_into(gs, gs_out)
+            target = gs_out
+        self._mark("sort", graph_safe)
+        self._pack_summary(target, ent_rec_ptr=ent_rec_ptr, loglik_done=True)
+        self._mark("summary", graph_safe)
+
+    def _sort_into(self, gs: GpuStateTensors, out: GpuStateTensors):
+        """Re-sort by partition id, gathering straight into the OTHER buffer
+        set (double buffering removes the copy-back traffic of the static
+        single-buffer variant; two alternating hipGraphs capture A->B and
+        B->A)."""
+        E, A = gs.E, self.model.A
+        dev = self.device
+        order = torch.argsort(gs.ent_part.to(torch.int64), stable=True)
+        inv = torch.empty_like(order)
+        inv[order] = torch.arange(order.numel(), device=dev)
+        torch.gather(gs.ent_part, 0, order, out=out.ent_part)
+        torch.gather(gs.ent_values, 0, order.view(E, 1).expand(E, A),
+                     out=out.ent_values)
+        new_rec_ent = inv[gs.rec_ent]
+        rorder = torch.argsort(new_rec_ent, stable=True)
+        R = rorder.numel()
+        torch.gather(new_rec_ent, 0, rorder, out=out.rec_ent)
+        torch.gather(gs.rec_values, 0, rorder.view(R, 1).expand(R, A),
+                     out=out.rec_values)
+        torch.gather(gs.rec_dist, 0, rorder.view(R, 1).expand(R, A),
+                     out=out.rec_dist)
+        torch.gather(gs.rec_file, 0, rorder, out=out.rec_file)
+        torch.gather(gs.rec_gid, 0, rorder, out=out.rec_gid)
+        torch.gather(out.ent_part, 0, out.rec_ent, out=out.rec_part)
 
     def _local_sort_static(self, gs: GpuStateTensors):
         """Re-sort by partition id into the SAME (static) buffers."""
