@@ -466,14 +466,9 @@ class GpuEngine(CpuEngine):
             # records with a short smallest candidate list run one-per-thread;
             # selection is a device-side mask (no host sync, graph-safe)
             if R >= 50_000:
-                nd = (gs.rec_values >= 0) & (gs.rec_dist == 0)
-                sizes = torch.where(
-                    nd, cand_hi[:, :A] - cand_lo[:, :A],
-                    torch.full_like(cand_hi[:, :A], 1 << 40),
-                )
-                min_sizes = sizes.amin(dim=1)
-                has_nd = nd.any(dim=1)
-                small_mask = (has_nd & (min_sizes <= 16)).to(torch.uint8).contiguous()
+                small_mask = torch.empty(R, dtype=torch.uint8, device=dev)
+                self.C.classify_small(gs.rec_values, gs.rec_dist, cand_lo,
+                                      cand_hi, self._num_pairs, 16, small_mask)
             else:
                 small_mask = torch.empty(0, dtype=torch.uint8, device=dev)
             self.C.link_update(

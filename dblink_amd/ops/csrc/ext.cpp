@@ -33,6 +33,9 @@ void cand_ranges(torch::Tensor rec_part, torch::Tensor rec_values,
                  torch::Tensor pair_a1, torch::Tensor pair_a2,
                  torch::Tensor pair_v2, torch::Tensor ptr, int64_t Vmax,
                  torch::Tensor cand_lo, torch::Tensor cand_hi);
+void classify_small(torch::Tensor rec_values, torch::Tensor rec_dist,
+                    torch::Tensor cand_lo, torch::Tensor cand_hi, int64_t NP,
+                    int64_t threshold, torch::Tensor small_mask);
 void link_update_dense(torch::Tensor rec_values, torch::Tensor rec_dist,
                        torch::Tensor rec_gid, torch::Tensor rec_part,
                        torch::Tensor rec_file, torch::Tensor ent_values,
@@ -110,5 +113,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "counting-sort index: posting scatter");
   m.def("cand_ranges", &dblink::cand_ranges,
         "per-record candidate ranges from the dense key prefix");
+  m.def("classify_small", &dblink::classify_small,
+        "mark records with a short smallest candidate list");
   m.def("summary_counts", &dblink::summary_counts, "fused summary counts + pack");
 }

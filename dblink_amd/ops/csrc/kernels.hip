@@ -1155,6 +1155,26 @@ __global__ __launch_bounds__(PB_THREADS) void postings_scatter_kernel(
   if (my_slot >= 0) postings[h_base[my_slot] + my_rank] = e_out;
 }
 
+// One thread per record: mark records whose smallest non-distorted candidate
+// list is short (handled serially by link_update_small_kernel). Replaces a
+// six-op torch mask pipeline over [R, A].
+__global__ void classify_small_kernel(
+    const int32_t* __restrict__ rec_values, const uint8_t* __restrict__ rec_dist,
+    const int64_t* __restrict__ cand_lo, const int64_t* __restrict__ cand_hi,
+    int64_t R, int A, int NP, int64_t threshold,
+    uint8_t* __restrict__ small_mask) {
+  const int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (r >= R) return;
+  const int T = A + NP;
+  int64_t best = INT64_MAX;
+  for (int a = 0; a < A; ++a) {
+    if (rec_values[r * A + a] < 0 || rec_dist[r * A + a]) continue;
+    const int64_t n = cand_hi[r * T + a] - cand_lo[r * T + a];
+    if (n < best) best = n;
+  }
+  small_mask[r] = best != INT64_MAX && best <= threshold;
+}
+
 __global__ void cand_ranges_kernel(
     const int32_t* __restrict__ rec_part, const int32_t* __restrict__ rec_values,
     const int32_t* __restrict__ pair_a1, const int32_t* __restrict__ pair_a2,
@@ -1460,6 +1480,20 @@ void postings_scatter(torch::Tensor ent_part, torch::Tensor ent_values,
                      pair_a1.data_ptr<int32_t>(), pair_a2.data_ptr<int32_t>(),
                      pair_v2.data_ptr<int32_t>(), E, A, NP, Vmax,
                      cursor.data_ptr<int32_t>(), postings.data_ptr<int32_t>());
+}
+
+void classify_small(torch::Tensor rec_values, torch::Tensor rec_dist,
+                    torch::Tensor cand_lo, torch::Tensor cand_hi, int64_t NP,
+                    int64_t threshold, torch::Tensor small_mask) {
+  const int64_t R = rec_values.size(0);
+  const int A = (int)rec_values.size(1);
+  if (R == 0) return;
+  dim3 grid((unsigned)((R + 255) / 256));
+  hipLaunchKernelGGL(classify_small_kernel, grid, dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(),
+                     rec_values.data_ptr<int32_t>(), rec_dist.data_ptr<uint8_t>(),
+                     cand_lo.data_ptr<int64_t>(), cand_hi.data_ptr<int64_t>(),
+                     R, A, (int)NP, threshold, small_mask.data_ptr<uint8_t>());
 }
 
 void cand_ranges(torch::Tensor rec_part, torch::Tensor rec_values,
