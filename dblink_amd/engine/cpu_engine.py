@@ -564,6 +564,11 @@ class CpuEngine:
     def step(self, state: ChainState, flags: SamplerFlags):
         from ..parallel import migration
 
+        # The GPU engine keeps records in stable identity order; the CPU
+        # sweep slices positionally by partition, so canonicalize once when
+        # picking up such a state (idempotent for CPU-produced states).
+        if not getattr(state, "cpu_sorted", False):
+            state.sort_by_partition()
         self._update_dist_probs(state)
         sweep(state, self.cache, self.partitioner, flags, self.num_partitions)
         migration.migrate(state, self.world_size)

@@ -180,15 +180,24 @@ class GpuStateTensors:
 
     @classmethod
     def like(cls, other):
+        """Second buffer set for the double-buffered single-rank sweep.
+        Records never move on a single rank, so the four fields that are
+        only written element-wise (or not at all) are SHARED, not cloned —
+        the per-sweep re-sort permutes entities and remaps rec_ent only."""
         o = cls.__new__(cls)
-        for f in cls.FIELDS + ("rec_part",):
+        for f in ("ent_values", "ent_part", "rec_ent", "rec_part"):
             setattr(o, f, torch.empty_like(getattr(other, f)))
+        for f in ("rec_values", "rec_file", "rec_dist", "rec_gid"):
+            setattr(o, f, getattr(other, f))
         o.device = other.device
         return o
 
     def to_host(self, state: ChainState):
         for f in self.FIELDS:
             setattr(state, f, getattr(self, f).cpu().numpy())
+        # single-rank GPU states keep records in identity order; the CPU
+        # engine re-sorts lazily if it ever picks this state up
+        state.cpu_sorted = False
 
     @property
     def E(self):
@@ -578,10 +587,10 @@ class GpuEngine(CpuEngine):
         self._mark("summary", graph_safe)
 
     def _sort_into(self, gs: GpuStateTensors, out: GpuStateTensors):
-        """Re-sort by partition id, gathering straight into the OTHER buffer
-        set (double buffering removes the copy-back traffic of the static
-        single-buffer variant; two alternating hipGraphs capture A->B and
-        B->A)."""
+        """Re-sort ENTITIES by partition id, gathering straight into the
+        other buffer set (two alternating hipGraphs capture A->B and B->A).
+        Records keep stable identity order on a single rank: kernels address
+        them through rec_ent / rec_part, so only those remap per sweep."""
         E, A = gs.E, self.model.A
         dev = self.device
         order = torch.argsort(gs.ent_part.to(torch.int64), stable=True)
@@ -590,16 +599,8 @@ class GpuEngine(CpuEngine):
         torch.gather(gs.ent_part, 0, order, out=out.ent_part)
         torch.gather(gs.ent_values, 0, order.view(E, 1).expand(E, A),
                      out=out.ent_values)
-        new_rec_ent = inv[gs.rec_ent]
-        rorder = torch.argsort(new_rec_ent, stable=True)
-        R = rorder.numel()
-        torch.gather(new_rec_ent, 0, rorder, out=out.rec_ent)
-        torch.gather(gs.rec_values, 0, rorder.view(R, 1).expand(R, A),
-                     out=out.rec_values)
-        torch.gather(gs.rec_dist, 0, rorder.view(R, 1).expand(R, A),
-                     out=out.rec_dist)
-        torch.gather(gs.rec_file, 0, rorder, out=out.rec_file)
-        torch.gather(gs.rec_gid, 0, rorder, out=out.rec_gid)
+        # records stay put: remap their entity ids through the permutation
+        torch.gather(inv, 0, gs.rec_ent, out=out.rec_ent)
         torch.gather(out.ent_part, 0, out.rec_ent, out=out.rec_part)
 
     def _local_sort_static(self, gs: GpuStateTensors):

@@ -82,6 +82,7 @@ class ChainState:
         Keeps rec_ent consistent. Stable, so within-partition order is
         preserved (matters for reproducibility of per-partition RNG streams).
         """
+        self.cpu_sorted = True
         order = np.argsort(self.ent_part, kind="stable")
         inv = np.empty_like(order)
         inv[order] = np.arange(order.size)
@@ -192,5 +193,11 @@ class ChainState:
             ),
             rng_state=driver.get("rng_state"),
         )
-        state.sort_by_partition()
+        # A single-rank resume must reproduce the saved engine order verbatim
+        # (the GPU engine keeps records in stable identity order; re-sorting
+        # would make a resumed chain diverge bitwise from a continuing one).
+        # Only a world-size change needs re-sorting: concatenated shards are
+        # not globally partition-sorted.
+        if len(mine) != 1 or world_size != len(shards):
+            state.sort_by_partition()
         return state
