@@ -244,7 +244,7 @@ class GpuEngine(CpuEngine):
         self._graphs_enabled = (
             os.environ.get("DBLINK_GRAPHS", "1") != "0" and world_size == 1
         )
-        self._loglik_buf = torch.zeros(1, dtype=torch.float64, device=device)
+        self._loglik_buf = torch.zeros(256, dtype=torch.float64, device=device)
         A, F = self.model.A, self.model.F
         self._packed = torch.zeros(2 + A * F + A + 1, dtype=torch.float64, device=device)
         self._ctrl = torch.zeros(2, dtype=torch.int64, device=device)

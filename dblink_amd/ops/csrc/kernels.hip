@@ -948,7 +948,9 @@ __global__ void distortion_update_kernel(
     double s = 0.0;
     const int nw = (int)(blockDim.x / WAVE);
     for (int i = 0; i < nw; ++i) s += wsum[i];
-    if (s != 0.0) atomicAdd(loglik, s);
+    // spread over 256 slots: every block adding one global f64 serializes
+    // (~10 ns per same-address atomic x tens of thousands of blocks)
+    if (s != 0.0) atomicAdd(&loglik[blockIdx.x & 255], s);
   }
 }
 
@@ -996,7 +998,7 @@ __global__ void summary_loglik_kernel(
     if (threadIdx.x < s) partial[threadIdx.x] += partial[threadIdx.x + s];
     __syncthreads();
   }
-  if (threadIdx.x == 0) atomicAdd(out, partial[0]);
+  if (threadIdx.x == 0) atomicAdd(&out[blockIdx.x & 255], partial[0]);
 }
 
 // ---------------------------------------------------------------------------
@@ -1243,7 +1245,11 @@ __global__ void summary_finalize_kernel(
     const double* __restrict__ loglik, const unsigned long long* __restrict__ counts,
     int n_counts, double* __restrict__ packed) {
   const int i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i == 0) packed[0] = loglik[0];
+  if (i == 0) {
+    double s = 0.0;
+    for (int j = 0; j < 256; ++j) s += loglik[j];
+    packed[0] = s;
+  }
   if (i < n_counts) packed[1 + i] = (double)counts[i];
 }
 
@@ -1676,6 +1682,7 @@ void distortion_update(
     torch::Tensor csr_sim, torch::Tensor loglik) {
   const int64_t* ctrl_ptr = ctrl.numel() ? ctrl.data_ptr<int64_t>() : nullptr;
   const bool fuse = loglik.numel() > 0;
+  TORCH_CHECK(!fuse || loglik.numel() >= 256, "loglik buffer must have 256 slots");
   const int64_t R = rec_values.size(0);
   const int A = (int)rec_values.size(1);
   const int F = (int)theta.size(1);
@@ -1703,6 +1710,7 @@ void summary_loglik(
     torch::Tensor rec_ent, torch::Tensor log_phi, torch::Tensor log_norm,
     torch::Tensor voff, torch::Tensor csr_row_ptr, torch::Tensor csr_col,
     torch::Tensor csr_sim, torch::Tensor attr_const, torch::Tensor out) {
+  TORCH_CHECK(out.numel() >= 256, "loglik buffer must have 256 slots");
   const int64_t E = ent_values.size(0);
   const int64_t R = rec_values.size(0);
   const int A = (int)ent_values.size(1);

@@ -323,14 +323,14 @@ def test_summary_loglik_matches_cpu():
     )
     ref = compute_summary(state, cache, state.dist_probs)
 
-    out = torch.zeros(1, dtype=torch.float64, device=DEV)
+    out = torch.zeros(256, dtype=torch.float64, device=DEV)  # 256-slot spread
     C.summary_loglik(
         _dev(ent_values, torch.int32), _dev(rec_values, torch.int32),
         _dev(rec_dist, torch.uint8), _dev(rec_ent, torch.int64),
         model.log_phi, model.log_norm, model.voff, model.csr_row_ptr,
         model.csr_col, model.csr_sim, model.attr_const, out,
     )
-    assert float(out.cpu()) == pytest.approx(ref.log_likelihood, rel=1e-5)
+    assert float(out.sum().cpu()) == pytest.approx(ref.log_likelihood, rel=1e-5)
 
 
 @gpu
