@@ -22,6 +22,7 @@ written or the state is saved.
 from __future__ import annotations
 
 import logging
+import os
 
 import numpy as np
 import torch
@@ -235,8 +236,6 @@ class GpuEngine(CpuEngine):
         self._gs = None
         self._ent_id_base = rank << 40
         self._err = torch.zeros(1, dtype=torch.int32, device=device)
-        import os
-
         self.debug_classes = os.environ.get("DBLINK_DEBUG_CLASSES", "") == "1"
         # per-phase HIP-event timers (SURVEY.md §5.1 observability): eager mode
         # only; accumulate totals, report via phase_times()
@@ -328,7 +327,13 @@ class GpuEngine(CpuEngine):
             if self._gs_alt is None:
                 self._gs_alt = GpuStateTensors.like(gs)
             out = self._gs_alt
-            use_graphs = self._graphs_enabled
+            # graph replay saves ~5 us x ~100 launches; at multi-million-record
+            # scales a sweep is tens of ms (capture of the multi-GB workspace
+            # pools also proved fragile there), so capture only where the
+            # launch overhead actually shows
+            use_graphs = self._graphs_enabled and gs.R <= int(
+                os.environ.get("DBLINK_GRAPH_MAX_RECORDS", "4000000")
+            )
             if use_graphs:
                 gkey = (key, self._flip)
                 if self._graph_key not in (None, key):
