@@ -114,3 +114,72 @@ def test_cross_variant_posterior_consistency():
     vals = list(results.values())
     for v in vals[1:]:
         assert abs(v - vals[0]) <= max(1.5, 0.35 * max(vals[0], 1e-9)), results
+
+
+def test_load_preserves_saved_order_same_world(tmp_path):
+    """A same-world resume must reproduce the saved array order VERBATIM:
+    the GPU engine keeps records in stable identity order, and a resumed
+    chain only matches a continuing one bitwise if load() does not reorder
+    (State.scala:122-193 two-file contract; ordering contract is ours)."""
+    state, engine = _make_state_and_engine(n=100, seed=9)
+    engine.step(state, SamplerFlags.for_sampler("PCG-I"))
+    # scramble records into a non-partition-sorted order before saving
+    rng = np.random.default_rng(0)
+    perm = rng.permutation(state.rec_values.shape[0])
+    for f in ("rec_values", "rec_file", "rec_ent", "rec_dist", "rec_gid"):
+        setattr(state, f, getattr(state, f)[perm])
+    state.cpu_sorted = False
+    state.save(str(tmp_path))
+    loaded = ChainState.load(str(tmp_path))
+    for f in ("ent_values", "ent_part", "rec_values", "rec_file", "rec_ent",
+              "rec_dist", "rec_gid"):
+        np.testing.assert_array_equal(getattr(loaded, f), getattr(state, f))
+
+
+def test_load_resorts_on_world_change(tmp_path):
+    """Re-sharding (world-size change) concatenates shards, which are not
+    globally partition-sorted - load must canonicalize then."""
+    state, engine = _make_state_and_engine(n=80, seed=5)
+    # fake a 2-rank save: split entities/records across two shards
+    half = state.num_entities // 2
+    rmask = state.rec_ent < half
+    for rank, (emask, rm) in enumerate(
+        [(slice(0, half), rmask), (slice(half, None), ~rmask)]
+    ):
+        shard = ChainState(
+            iteration=state.iteration,
+            ent_values=state.ent_values[emask],
+            ent_part=state.ent_part[emask],
+            rec_values=state.rec_values[rm],
+            rec_file=state.rec_file[rm],
+            rec_ent=state.rec_ent[rm] - (half if rank == 1 else 0),
+            rec_dist=state.rec_dist[rm],
+            rec_gid=state.rec_gid[rm],
+            dist_probs=state.dist_probs,
+            population_size=state.population_size,
+            start_seed=state.start_seed,
+            current_seed=state.current_seed,
+            summary=state.summary,
+        )
+        shard.save(str(tmp_path), rank=rank)
+    loaded = ChainState.load(str(tmp_path), rank=0, world_size=1)
+    assert loaded.num_records == state.num_records
+    assert (np.diff(loaded.ent_part) >= 0).all(), "entities not partition-sorted"
+    # conservation: same multiset of record gids
+    np.testing.assert_array_equal(np.sort(loaded.rec_gid), np.sort(state.rec_gid))
+
+
+def test_cpu_engine_canonicalizes_unsorted_state():
+    """CpuEngine.step must sort a GPU-ordered (identity-order) state once
+    before slicing partitions positionally."""
+    state, engine = _make_state_and_engine(n=100, seed=11)
+    rng = np.random.default_rng(3)
+    perm = rng.permutation(state.rec_values.shape[0])
+    for f in ("rec_values", "rec_file", "rec_ent", "rec_dist", "rec_gid"):
+        setattr(state, f, getattr(state, f)[perm])
+    state.cpu_sorted = False
+    engine.step(state, SamplerFlags.for_sampler("PCG-I"))
+    assert np.isfinite(state.summary.log_likelihood)
+    # records must now be grouped behind partition-sorted entities
+    rec_part = state.ent_part[state.rec_ent]
+    assert (np.diff(rec_part) >= 0).all()
