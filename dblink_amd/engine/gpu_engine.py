@@ -405,8 +405,9 @@ class GpuEngine(CpuEngine):
     def _sweep_body(self, gs: GpuStateTensors, flags: SamplerFlags, graph_safe=True,
                     gs_out=None):
         """One full device-side iteration. Capture-safe for world_size == 1:
-        static input/output buffers (results copied back in place), control
-        values read from the device ctrl buffer, no host synchronisation."""
+        control values read from the device ctrl buffer, no host
+        synchronisation; with ``gs_out`` the final re-sort gathers into the
+        other buffer set (double buffering - see _sort_into)."""
         m = self.model
         A, E, R = m.A, gs.E, gs.R
         dev = self.device
