@@ -582,14 +582,9 @@ class GpuEngine(CpuEngine):
         is still valid for the summary counts."""
         if self.world_size > 1:
             return
-        if gs_out is None:
-            self._local_sort_static(gs)
-            target = gs
-        else:
-            self._sort_into(gs, gs_out)
-            target = gs_out
+        self._sort_into(gs, gs_out)
         self._mark("sort", graph_safe)
-        self._pack_summary(target, ent_rec_ptr=ent_rec_ptr, loglik_done=True)
+        self._pack_summary(gs_out, ent_rec_ptr=ent_rec_ptr, loglik_done=True)
         self._mark("summary", graph_safe)
 
     def _sort_into(self, gs: GpuStateTensors, out: GpuStateTensors):
@@ -608,22 +603,6 @@ class GpuEngine(CpuEngine):
         # records stay put: remap their entity ids through the permutation
         torch.gather(inv, 0, gs.rec_ent, out=out.rec_ent)
         torch.gather(out.ent_part, 0, out.rec_ent, out=out.rec_part)
-
-    def _local_sort_static(self, gs: GpuStateTensors):
-        """Re-sort by partition id into the SAME (static) buffers."""
-        order = torch.argsort(gs.ent_part.to(torch.int64), stable=True)
-        inv = torch.empty_like(order)
-        inv[order] = torch.arange(order.numel(), device=self.device)
-        gs.ent_values.copy_(gs.ent_values[order])
-        gs.ent_part.copy_(gs.ent_part[order])
-        new_rec_ent = inv[gs.rec_ent]
-        rorder = torch.argsort(new_rec_ent, stable=True)
-        gs.rec_ent.copy_(new_rec_ent[rorder])
-        gs.rec_values.copy_(gs.rec_values[rorder])
-        gs.rec_file.copy_(gs.rec_file[rorder])
-        gs.rec_dist.copy_(gs.rec_dist[rorder])
-        gs.rec_gid.copy_(gs.rec_gid[rorder])
-        gs.rec_part.copy_(gs.ent_part[gs.rec_ent])
 
     # ---- summary -------------------------------------------------------------
 
