@@ -110,23 +110,41 @@ def _build_inverted_index(ent_values):
     return inv
 
 
-def _get_possible_entities(rec_vals, rec_dist, inv_index, num_entities, const_mask):
+def _get_possible_entities(rec_vals, rec_dist, inv_index, num_entities, const_mask,
+                           ent_values=None):
     """Set intersection over observed, non-distorted attributes
-    (GibbsUpdates.scala:473-530). Returns (candidate ndarray, obs-distorted attr list)."""
+    (GibbsUpdates.scala:473-530). Returns (candidate ndarray, obs-distorted attr list).
+
+    With ``ent_values`` the intersection enumerates the SMALLEST posting list
+    and filters by direct value comparison — the posting list of (a, v) is
+    exactly the entities whose a equals v, so membership needs no set ops.
+    Yields the identical candidate array (same set, same order) as the
+    isin-based intersection, just ~10x cheaper.
+    """
     obs_dist = []
     sets = []
+    attrs = []
     A = len(rec_vals)
     for a in range(A):
         v = rec_vals[a]
         if v >= 0:
             if not rec_dist[a]:
                 sets.append(inv_index.get((a, int(v)), _EMPTY_I64))
+                attrs.append(a)
             else:
                 obs_dist.append(a)
     if not sets:
         return np.arange(num_entities), obs_dist
     if len(sets) == 1:
         return sets[0], obs_dist
+    if ent_values is not None:
+        base = min(range(len(sets)), key=lambda i: len(sets[i]))
+        result = sets[base]
+        for i, a in enumerate(attrs):
+            if i == base or result.size == 0:
+                continue
+            result = result[ent_values[result, a] == rec_vals[a]]
+        return result, obs_dist
     sets.sort(key=len)
     result = sets[0]
     for s in sets[1:]:
@@ -205,7 +223,8 @@ def sweep_partition(
 
 def _update_entity_id(rng, rv, rd, ent_values, inv_index, attrs):
     """Non-collapsed indexed link update (GibbsUpdates.scala:398-430)."""
-    cands, obs_dist = _get_possible_entities(rv, rd, inv_index, ent_values.shape[0], None)
+    cands, obs_dist = _get_possible_entities(rv, rd, inv_index, ent_values.shape[0], None,
+                                             ent_values=ent_values)
     if len(cands) == 0:
         raise RuntimeError("empty candidate set: state invariant violated")
     if not obs_dist:
@@ -220,7 +239,7 @@ def _update_entity_id(rng, rv, rd, ent_values, inv_index, attrs):
         else:
             y = ent_values[cands, a]
             norms = ia.index.sim_norms[y]
-            es = np.array([ia.index.exp_sim_of(x, int(yy)) for yy in y])
+            es = ia.index.exp_sim_many(x, y)
             weights *= norms * es * px
     idx = _sample_weights(rng, weights)
     return int(cands[idx])
@@ -244,7 +263,7 @@ def _update_entity_id_collapsed(rng, rv, ent_values, attrs, dist_probs, file_id)
             weights *= agree + theta * px
         else:
             norms = ia.index.sim_norms[y]
-            es = np.array([ia.index.exp_sim_of(x, int(yy)) for yy in y])
+            es = ia.index.exp_sim_many(x, y)
             weights *= agree + theta * px * norms * es
     return _sample_weights(rng, weights)
 

@@ -51,6 +51,17 @@ class SimIndexCSR:
             return float(self.expsim[i])
         return 1.0
 
+    def lookup_row_many(self, v: int, ws) -> np.ndarray:
+        """Vectorized ``lookup(v, w)`` for an array of ``w`` (same values)."""
+        lo, hi = int(self.row_ptr[v]), int(self.row_ptr[v + 1])
+        cols = self.col[lo:hi]
+        idx = np.searchsorted(cols, ws)
+        idx_c = np.minimum(idx, hi - lo - 1) if hi > lo else idx * 0
+        hit = (hi > lo) & (cols[idx_c] == ws)
+        out = np.ones(len(ws), dtype=np.float64)
+        out[hit] = self.expsim[lo + idx_c[hit]]
+        return out
+
     @property
     def nnz(self):
         return int(self.col.size)
@@ -183,6 +194,12 @@ class AttributeIndex:
         if self.is_constant:
             return 1.0
         return self.sim_index.lookup(v1, v2)
+
+    def exp_sim_many(self, v1: int, v2s) -> np.ndarray:
+        """Vectorized ``exp_sim_of(v1, .)`` over an int array (same values)."""
+        if self.is_constant:
+            return np.ones(len(v2s))
+        return self.sim_index.lookup_row_many(v1, np.asarray(v2s))
 
     def sim_norm_dist(self, power: int) -> AliasTable:
         """Distribution p(v) ∝ phi(v) * norm(v)^power."""
