@@ -155,6 +155,7 @@ def _get_possible_entities(rec_vals, rec_dist, inv_index, num_entities, const_ma
 
 
 _EMPTY_I64 = np.empty(0, dtype=np.int64)
+_FAST_VALUE = True
 
 
 def sweep_partition(
@@ -321,6 +322,33 @@ def _update_entity_value_collapsed(rng, a, ia, part, linked, dist_probs):
     base, kind = _base_distribution(ia, len(obs))
     if not obs:
         return _alias_draw(rng, base)
+    if (_FAST_VALUE and len(obs) == 1 and not ia.is_constant
+            and ia.index.sim_row_len(int(part.rec_values[obs[0], a])) >= 128):
+        # vectorized single-record case (the common one): one sim row, same
+        # arithmetic and draw sequence as the dict path below (bitwise-equal;
+        # _FAST_VALUE exists so tests can force the reference dict path)
+        r = obs[0]
+        theta = dist_probs(a, int(part.rec_file[r]))
+        x = int(part.rec_values[r, a])
+        px = ia.index.probability_of(x)
+        normx = ia.index.sim_norms[x]
+        cols, sims = ia.index.sim_index.row(x)
+        w = sims.astype(np.float64, copy=True)
+        pos = int(np.searchsorted(cols, x))
+        if pos < len(cols) and cols[pos] == x:
+            w[pos] = sims[pos] + (1.0 / theta - 1.0) / (px * normx)
+        k = kind[1]
+        weights = (ia.index.probs[cols] * ia.index.sim_norms[cols] ** k
+                   / ia.index.sim_norm_total(k)) * (w - 1.0)
+        cum = np.cumsum(weights)
+        total = float(cum[-1]) if weights.size else 0.0
+        if rng.random() < 1.0 / (1.0 + total):
+            return _alias_draw(rng, base)
+        if total <= 0.0 or not math.isfinite(total):
+            raise ValueError("zero or non-finite probability mass")
+        u = rng.random() * total
+        i = int(np.searchsorted(cum, u, side="right"))
+        return int(cols[min(i, len(cols) - 1)])
     vw = {}
     for r in obs:
         theta = dist_probs(a, int(part.rec_file[r]))
@@ -399,16 +427,8 @@ def _update_distortions(rng, part, r, y, attrs, dist_probs):
         if x < 0:
             part.rec_dist[r, a] = rng.random() < theta
         elif x == int(y[a]):
-            ia = attrs[a]
-            if ia.is_constant:
-                pr1 = theta * ia.index.probability_of(x)
-            else:
-                pr1 = (
-                    theta
-                    * ia.index.probability_of(x)
-                    * ia.index.sim_norms[x]
-                    * ia.index.exp_sim_of(x, x)
-                )
+            # self_mass = phi(x) [* norm(x) * expsim(x, x)] — static per index
+            pr1 = theta * attrs[a].index.self_mass[x]
             pr0 = 1.0 - theta
             p = pr1 / (pr1 + pr0) if (pr1 + pr0) != 0.0 else 0.0
             part.rec_dist[r, a] = rng.random() < p

@@ -195,6 +195,25 @@ class AttributeIndex:
             return 1.0
         return self.sim_index.lookup(v1, v2)
 
+    def sim_row_len(self, v: int) -> int:
+        if self.is_constant:
+            return 0
+        return int(self.sim_index.row_ptr[v + 1] - self.sim_index.row_ptr[v])
+
+    @property
+    def self_mass(self) -> np.ndarray:
+        """phi(v) [* norm(v) * expsim(v, v)] — the agreement mass used by the
+        distortion conditional (GibbsUpdates.scala:332-346); static, cached."""
+        got = getattr(self, "_self_mass", None)
+        if got is None:
+            if self.is_constant:
+                got = self.probs.copy()
+            else:
+                es = np.array([self.exp_sim_of(v, v) for v in range(self.num_values)])
+                got = self.probs * self.sim_norms * es
+            self._self_mass = got
+        return got
+
     def exp_sim_many(self, v1: int, v2s) -> np.ndarray:
         """Vectorized ``exp_sim_of(v1, .)`` over an int array (same values)."""
         if self.is_constant:
