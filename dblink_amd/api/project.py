@@ -156,7 +156,16 @@ class Project:
     def saved_state(self):
         if ChainState.exists(self.output_path):
             log.info("Resuming from saved state at %s", self.output_path)
-            return ChainState.load(self.output_path, rank=self.rank, world_size=self.world_size)
+            state = ChainState.load(self.output_path, rank=self.rank,
+                                    world_size=self.world_size)
+            # restore the FITTED partition function saved with the chain —
+            # a freshly parsed partitioner is unfit (single leaf) and would
+            # silently collapse the partition structure on resume
+            saved = getattr(state, "saved_partitioner", None)
+            if saved is not None:
+                self.partitioner = saved
+                self._engine = None
+            return state
         return None
 
     def generate_initial_state(self):

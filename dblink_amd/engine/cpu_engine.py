@@ -158,6 +158,12 @@ _EMPTY_I64 = np.empty(0, dtype=np.int64)
 _FAST_VALUE = True
 
 
+def _cpu_fast_enabled():
+    import os
+
+    return os.environ.get("DBLINK_CPU_FAST", "1") != "0"
+
+
 def sweep_partition(
     rng,
     part,  # _PartitionData
@@ -508,7 +514,24 @@ def sweep(
 
     Does NOT exchange clusters between ranks (see parallel.migration) and does
     not compute summaries. Advances ``current_seed`` by num_partitions.
+
+    The flagship PCG-I variant runs through the vectorized whole-rank sweep
+    (``cpu_fast.sweep_fast``) unless DBLINK_CPU_FAST=0; the per-record loops
+    below remain the numerical oracle and serve the other variants.
     """
+    if (_cpu_fast_enabled() and flags.collapsed_entity_values
+            and not flags.collapsed_entity_ids and not flags.sequential):
+        from ..parallel import comm
+        from .cpu_fast import sweep_fast
+
+        rank = comm.rank_world()[0] if comm.is_distributed() else 0
+        sweep_fast(state, cache, partitioner, num_partitions, rank=rank)
+        return
+    if state.num_entities and int(state.ent_part.max()) >= num_partitions:
+        raise RuntimeError(
+            "state has partition ids beyond the partitioner's range "
+            "(resuming with an unfitted partitioner?)"
+        )
     ent_ptr, rec_ptr = state.partition_offsets(num_partitions)
     new_rec_ent = np.empty_like(state.rec_ent)
     for p in range(num_partitions):

@@ -1,0 +1,309 @@
+"""Vectorized CPU sweep for the PCG-I sampler (the flagship variant).
+
+Mirrors the GPU engine's phase structure in numpy — whole-rank flat arrays,
+counter-based Philox draws keyed by (seed, iteration, phase, element id) —
+instead of the per-record python loops of ``cpu_engine.sweep_partition``.
+The per-record implementation remains the numerical oracle (and serves the
+other sampler variants); this path produces the same conditional
+distributions (checked by the cross-variant posterior-consistency and
+fast-vs-reference tests) but a different — still deterministic — draw
+sequence.
+
+Parity: link update ``GibbsUpdates.scala:398-430, 473-530``; collapsed value
+update ``:576-599``; distortion update ``:324-359``.
+"""
+
+from __future__ import annotations
+
+import numpy as np
+
+_PH_LINK = np.uint32(1)
+_PH_DIST = np.uint32(2)
+_PH_VAL = np.uint32(3)
+
+_M0 = np.uint64(0xD2511F53)
+_M1 = np.uint64(0xCD9E8D57)
+_W0 = np.uint32(0x9E3779B9)
+_W1 = np.uint32(0xBB67AE85)
+
+
+def _philox_uniform(seed, iteration, phase, ids, draw, rank=0):
+    """Philox4x32-10 keyed uniform in (0, 1) for an int64 id array."""
+    ids = np.asarray(ids, dtype=np.uint64)
+    c0 = ids.astype(np.uint32)
+    c1 = (ids >> np.uint64(32)).astype(np.uint32)
+    c2 = np.full_like(c0, np.uint32(draw))
+    c3 = np.full_like(c0, np.uint32(iteration))
+    k0 = np.full_like(c0, np.uint32(np.uint64(seed) & np.uint64(0xFFFFFFFF)))
+    k1 = np.full_like(
+        c0,
+        np.uint32((np.uint64(seed) >> np.uint64(32)) & np.uint64(0xFFFFFFFF))
+        ^ np.uint32((int(phase) * 0x9E3779B1) & 0xFFFFFFFF)
+        ^ np.uint32((rank * 0x85EBCA6B) & 0xFFFFFFFF),
+    )
+    for _ in range(10):
+        p0 = c0.astype(np.uint64) * _M0
+        p1 = c2.astype(np.uint64) * _M1
+        hi0 = (p0 >> np.uint64(32)).astype(np.uint32)
+        lo0 = p0.astype(np.uint32)
+        hi1 = (p1 >> np.uint64(32)).astype(np.uint32)
+        lo1 = p1.astype(np.uint32)
+        c0, c1, c2, c3 = hi1 ^ c1 ^ k0, lo1, hi0 ^ c3 ^ k1, lo0
+        k0 = k0 + _W0
+        k1 = k1 + _W1
+    return (c0.astype(np.float64) + 0.5) * (2.0 ** -32)
+
+
+def _alias_draw_vec(u1, u2, prob, alias):
+    n = len(prob)
+    idx = np.minimum((u1 * n).astype(np.int64), n - 1)
+    return np.where(u2 < prob[idx], idx, alias[idx])
+
+
+def _ragged_expand(lengths):
+    """(flat->group index, group offsets) for ragged group lengths."""
+    off = np.zeros(len(lengths) + 1, dtype=np.int64)
+    np.cumsum(lengths, out=off[1:])
+    grp = np.repeat(np.arange(len(lengths)), lengths)
+    return grp, off
+
+
+class _FastModel:
+    """Per-cache static arrays for the vectorized sweep (built once)."""
+
+    def __init__(self, cache):
+        import scipy.sparse as sp
+
+        attrs = cache.indexed_attributes
+        self.attrs = attrs
+        self.A = len(attrs)
+        self.Vmax = max(ia.index.num_values for ia in attrs)
+        self.log_norms = [
+            None if ia.is_constant else np.log(ia.index.sim_norms) for ia in attrs
+        ]
+        self.sim_csr = []
+        for ia in attrs:
+            if ia.is_constant:
+                self.sim_csr.append(None)
+            else:
+                si = ia.index.sim_index
+                V = ia.index.num_values
+                self.sim_csr.append(
+                    sp.csr_matrix((si.expsim, si.col, si.row_ptr), shape=(V, V))
+                )
+        self.phi_tables = [ia.index.distribution for ia in attrs]
+        self.pow1_tables = [
+            None if ia.is_constant else ia.index.sim_norm_dist(1) for ia in attrs
+        ]
+        self.pow1_w = []  # per attr: weights(v) = phi*norm/Z1 precomputed
+        for ia in attrs:
+            if ia.is_constant:
+                self.pow1_w.append(None)
+            else:
+                self.pow1_w.append(
+                    ia.index.probs * ia.index.sim_norms / ia.index.sim_norm_total(1)
+                )
+
+
+def get_fast_model(cache):
+    got = getattr(cache, "_fast_model", None)
+    if got is None:
+        got = _FastModel(cache)
+        cache._fast_model = got
+    return got
+
+
+def sweep_fast(state, cache, partitioner, num_partitions, rank=0):
+    """One PCG-I sweep over the rank's (partition-sorted) state, in place.
+
+    Same contract as ``cpu_engine.sweep``: advances current_seed by
+    num_partitions and increments the iteration counter.
+    """
+    fm = get_fast_model(cache)
+    attrs, A = fm.attrs, fm.A
+    seed = int(state.current_seed)
+    it = int(state.iteration) + 1
+    E = state.num_entities
+    R = state.num_records
+    ev = state.ent_values
+    rv = state.rec_values
+    rdist = state.rec_dist.astype(bool)
+    obs = rv >= 0
+    theta = state.dist_probs.probs  # [A, F]
+    theta_ra = theta[np.arange(A)[None, :], state.rec_file[:, None]]  # [R, A]
+
+    # ---- phase 1: link update (indexed, Gumbel-max) -------------------------
+    if E and int(state.ent_part.max()) >= num_partitions:
+        raise RuntimeError(
+            "state has partition ids beyond the partitioner's range "
+            "(resuming with an unfitted partitioner?)"
+        )
+    ent_ptr = np.searchsorted(state.ent_part, np.arange(num_partitions + 1))
+    keys = (
+        (state.ent_part[:, None].astype(np.int64) * A + np.arange(A)[None, :])
+        * fm.Vmax
+        + ev
+    ).reshape(-1)
+    order = np.argsort(keys)
+    skeys = keys[order]
+    postings = (order // A).astype(np.int64)  # row-major [E, A] flatten
+    rec_part = state.ent_part[state.rec_ent]
+    qkeys = (
+        (rec_part[:, None].astype(np.int64) * A + np.arange(A)[None, :]) * fm.Vmax
+        + np.maximum(rv, 0)
+    )
+    lo = np.searchsorted(skeys, qkeys.reshape(-1)).reshape(R, A)
+    hi = np.searchsorted(skeys, qkeys.reshape(-1), side="right").reshape(R, A)
+    nd = obs & ~rdist
+    sizes = np.where(nd, hi - lo, np.int64(1) << 60)
+    base_a = np.argmin(sizes, axis=1)
+    rows = np.arange(R)
+    has_nd = nd.any(axis=1)
+    base_n = np.where(has_nd, sizes[rows, base_a], ent_ptr[rec_part + 1] - ent_ptr[rec_part])
+    base_lo = np.where(has_nd, lo[rows, base_a], ent_ptr[rec_part])
+
+    if (base_n == 0).any():
+        raise RuntimeError("empty candidate set: state invariant violated")
+    rec_of, off = _ragged_expand(base_n)
+    pos = np.arange(off[-1]) - off[rec_of]
+    cand = np.where(
+        has_nd[rec_of],
+        postings[np.minimum(base_lo[rec_of] + pos, len(postings) - 1)],
+        base_lo[rec_of] + pos,
+    )
+    ok = np.ones(len(cand), dtype=bool)
+    logw = np.zeros(len(cand))
+    for a in range(A):
+        m = nd[rec_of, a] & ((a != base_a[rec_of]) | ~has_nd[rec_of])
+        if m.any():
+            ok[m] &= ev[cand[m], a] == rv[rec_of[m], a]
+        if attrs[a].is_constant:
+            continue
+        od_m = obs[rec_of, a] & rdist[rec_of, a]
+        if od_m.any():
+            y = ev[cand[od_m], a]
+            x = rv[rec_of[od_m], a]
+            es = np.asarray(fm.sim_csr[a][x, y]).ravel()
+            es = np.where(es > 0.0, es, 1.0)
+            logw[od_m] += fm.log_norms[a][y] + np.log(es)
+    u = _philox_uniform(seed, it, _PH_LINK, state.rec_gid[rec_of] * 64 + (cand & 63),
+                        cand >> 6, rank)
+    score = np.where(ok, logw - np.log(-np.log(u)), -np.inf)
+    # segmented argmax: sort by (record, score) and take the last per record
+    perm = np.lexsort((score, rec_of))
+    last = off[1:] - 1
+    best_flat = perm[last]
+    if not np.isfinite(score[best_flat]).all():
+        raise RuntimeError("empty candidate set: state invariant violated")
+    state.rec_ent = cand[best_flat].astype(np.int64)
+
+    # ---- phase 2: collapsed entity-value update -----------------------------
+    kobs = np.zeros((E, A), dtype=np.int64)
+    np.add.at(kobs, state.rec_ent, obs.astype(np.int64))
+    ea_ids = (np.arange(E)[:, None] * A + np.arange(A)[None, :])
+    u_mix = _philox_uniform(seed, it, _PH_VAL, ea_ids.reshape(-1), 0, rank).reshape(E, A)
+    u_a1 = _philox_uniform(seed, it, _PH_VAL, ea_ids.reshape(-1), 1, rank).reshape(E, A)
+    u_a2 = _philox_uniform(seed, it, _PH_VAL, ea_ids.reshape(-1), 2, rank).reshape(E, A)
+    u_sel = _philox_uniform(seed, it, _PH_VAL, ea_ids.reshape(-1), 3, rank).reshape(E, A)
+
+    # single observed record per (e, a): its row index
+    first_rec = np.full((E, A), -1, dtype=np.int64)
+    rr = np.arange(R)
+    for a in range(A):
+        o = obs[:, a]
+        first_rec[state.rec_ent[o], a] = rr[o]  # k==1 groups get THE record
+
+    new_ev = ev.copy()
+    fallback = []
+    for a in range(A):
+        ia = attrs[a]
+        k = kobs[:, a]
+        k0 = k == 0
+        if k0.any():  # phi draw
+            t = fm.phi_tables[a]
+            new_ev[k0, a] = _alias_draw_vec(u_a1[k0, a], u_a2[k0, a], t.prob, t.alias)
+        k1 = k == 1
+        if k1.any():
+            e_idx = np.flatnonzero(k1)
+            r1 = first_rec[e_idx, a]
+            x = rv[r1, a]
+            th = theta_ra[r1, a]
+            if ia.is_constant:
+                # closed form: P(base) = theta, else keep x
+                take_base = u_mix[e_idx, a] < th
+                tb = e_idx[take_base]
+                t = fm.phi_tables[a]
+                new_ev[tb, a] = _alias_draw_vec(u_a1[tb, a], u_a2[tb, a], t.prob, t.alias)
+                new_ev[e_idx[~take_base], a] = x[~take_base]
+            else:
+                si = ia.index.sim_index
+                rlo = si.row_ptr[x]
+                rlen = si.row_ptr[x + 1] - rlo
+                grp, goff = _ragged_expand(rlen)
+                fcol = si.col[rlo[grp] + (np.arange(goff[-1]) - goff[grp])]
+                w = si.expsim[rlo[grp] + (np.arange(goff[-1]) - goff[grp])].copy()
+                px = ia.index.probs[x]
+                normx = ia.index.sim_norms[x]
+                delta = (1.0 / th - 1.0) / (px * normx)
+                self_m = fcol == x[grp]
+                w[self_m] += delta[grp[self_m]]
+                wgt = fm.pow1_w[a][fcol] * (w - 1.0)
+                c = np.cumsum(wgt)
+                tot = c[goff[1:] - 1] - np.where(goff[:-1] > 0, c[goff[:-1] - 1], 0.0)
+                take_base = u_mix[e_idx, a] < 1.0 / (1.0 + tot)
+                tb = e_idx[take_base]
+                t1 = fm.pow1_tables[a]
+                new_ev[tb, a] = _alias_draw_vec(u_a1[tb, a], u_a2[tb, a], t1.prob, t1.alias)
+                pert = ~take_base
+                if pert.any():
+                    basec = np.where(goff[:-1] > 0, c[goff[:-1] - 1], 0.0)[pert]
+                    target = basec + u_sel[e_idx[pert], a] * tot[pert]
+                    j = np.searchsorted(c, target, side="right")
+                    j = np.minimum(j, goff[1:][pert] - 1)
+                    new_ev[e_idx[pert], a] = fcol[j]
+        k2 = k >= 2
+        if k2.any():
+            for e in np.flatnonzero(k2):
+                fallback.append((int(e), a))
+
+    if fallback:
+        from . import cpu_engine as ce
+
+        class _P:  # lightweight partition view over the whole rank
+            pass
+
+        part = _P()
+        part.rec_values = rv
+        part.rec_file = state.rec_file
+        part.rec_dist = state.rec_dist
+        ent_recs = [[] for _ in range(E)]
+        for r in range(R):
+            ent_recs[state.rec_ent[r]].append(r)
+        for e, a in fallback:
+            key = ((seed & ((1 << 63) - 1)) << 64) | (it << 40) | (e * A + a) | (rank << 36)
+            rng = np.random.Generator(np.random.Philox(key=key))
+            new_ev[e, a] = ce._update_entity_value_collapsed(
+                rng, a, attrs[a], part, ent_recs[e], state.dist_probs
+            )
+    state.ent_values = new_ev
+
+    # ---- phase 3: distortion update -----------------------------------------
+    y_link = state.ent_values[state.rec_ent]  # [R, A]
+    ra_ids = (state.rec_gid[:, None] * 32 + np.arange(A)[None, :]).reshape(-1)
+    u_d = _philox_uniform(seed, it, _PH_DIST, ra_ids, 0, rank).reshape(R, A)
+    self_mass = np.stack(
+        [ia.index.self_mass[np.maximum(rv[:, a], 0)] for a, ia in enumerate(attrs)], 1
+    )
+    pr1 = theta_ra * self_mass
+    p_agree = pr1 / (pr1 + (1.0 - theta_ra))
+    z = np.where(
+        ~obs,
+        u_d < theta_ra,
+        np.where(rv == y_link, u_d < p_agree, True),
+    )
+    state.rec_dist = z.astype(np.uint8)
+
+    # ---- partition reassignment + bookkeeping -------------------------------
+    state.ent_part = partitioner.get_partition_ids(state.ent_values).astype(np.int32)
+    state.current_seed += num_partitions
+    state.iteration += 1

@@ -100,7 +100,8 @@ def sample(
             sample_ctr += 1
         if checkpoint_interval and completed % checkpoint_interval == 0:
             sync()
-            state.save(output_path, rank=rank)
+            state.save(output_path, rank=rank,
+                       extra={"partitioner": engine.partitioner})
     dt = time.time() - t0
     iters = state.iteration - initial_iteration
     log.info(
@@ -113,5 +114,6 @@ def sample(
             diagnostics_writer.close()
         comm.barrier()
         sync()
-        state.save(output_path, rank=rank)
+        state.save(output_path, rank=rank,
+                   extra={"partitioner": engine.partitioner})
     return state

@@ -193,6 +193,9 @@ class ChainState:
             ),
             rng_state=driver.get("rng_state"),
         )
+        # the fitted partition function travels with the driver state
+        # (parity: State.scala broadcasts/saves the PartitionFunction)
+        state.saved_partitioner = driver.get("partitioner")
         # A single-rank resume must reproduce the saved engine order verbatim
         # (the GPU engine keeps records in stable identity order; re-sorting
         # would make a resumed chain diverge bitwise from a continuing one).
