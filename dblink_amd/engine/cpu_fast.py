@@ -27,8 +27,9 @@ _W0 = np.uint32(0x9E3779B9)
 _W1 = np.uint32(0xBB67AE85)
 
 
-def _philox_uniform(seed, iteration, phase, ids, draw, rank=0):
-    """Philox4x32-10 keyed uniform in (0, 1) for an int64 id array."""
+def _philox_uniform4(seed, iteration, phase, ids, draw, rank=0):
+    """Philox4x32-10 keyed uniforms in (0, 1): FOUR independent streams per
+    id (the four 32-bit output words) for the price of one keyed evaluation."""
     ids = np.asarray(ids, dtype=np.uint64)
     c0 = ids.astype(np.uint32)
     c1 = (ids >> np.uint64(32)).astype(np.uint32)
@@ -51,7 +52,12 @@ def _philox_uniform(seed, iteration, phase, ids, draw, rank=0):
         c0, c1, c2, c3 = hi1 ^ c1 ^ k0, lo1, hi0 ^ c3 ^ k1, lo0
         k0 = k0 + _W0
         k1 = k1 + _W1
-    return (c0.astype(np.float64) + 0.5) * (2.0 ** -32)
+    to_u = lambda c: (c.astype(np.float64) + 0.5) * (2.0 ** -32)  # noqa: E731
+    return to_u(c0), to_u(c1), to_u(c2), to_u(c3)
+
+
+def _philox_uniform(seed, iteration, phase, ids, draw, rank=0):
+    return _philox_uniform4(seed, iteration, phase, ids, draw, rank)[0]
 
 
 def _alias_draw_vec(u1, u2, prob, alias):
@@ -201,10 +207,10 @@ def sweep_fast(state, cache, partitioner, num_partitions, rank=0):
     kobs = np.zeros((E, A), dtype=np.int64)
     np.add.at(kobs, state.rec_ent, obs.astype(np.int64))
     ea_ids = (np.arange(E)[:, None] * A + np.arange(A)[None, :])
-    u_mix = _philox_uniform(seed, it, _PH_VAL, ea_ids.reshape(-1), 0, rank).reshape(E, A)
-    u_a1 = _philox_uniform(seed, it, _PH_VAL, ea_ids.reshape(-1), 1, rank).reshape(E, A)
-    u_a2 = _philox_uniform(seed, it, _PH_VAL, ea_ids.reshape(-1), 2, rank).reshape(E, A)
-    u_sel = _philox_uniform(seed, it, _PH_VAL, ea_ids.reshape(-1), 3, rank).reshape(E, A)
+    u_mix, u_a1, u_a2, u_sel = (
+        u.reshape(E, A)
+        for u in _philox_uniform4(seed, it, _PH_VAL, ea_ids.reshape(-1), 0, rank)
+    )
 
     # single observed record per (e, a): its row index
     first_rec = np.full((E, A), -1, dtype=np.int64)
@@ -276,14 +282,16 @@ def sweep_fast(state, cache, partitioner, num_partitions, rank=0):
         part.rec_values = rv
         part.rec_file = state.rec_file
         part.rec_dist = state.rec_dist
-        ent_recs = [[] for _ in range(E)]
-        for r in range(R):
-            ent_recs[state.rec_ent[r]].append(r)
+        rorder = np.argsort(state.rec_ent, kind="stable")
+        starts = np.searchsorted(state.rec_ent[rorder], np.arange(E + 1))
+        # one sweep-keyed stream, consumed in deterministic (e, a) order
+        rng = np.random.Generator(
+            np.random.Philox(key=((seed & ((1 << 63) - 1)) << 64) | (it << 32) | rank)
+        )
         for e, a in fallback:
-            key = ((seed & ((1 << 63) - 1)) << 64) | (it << 40) | (e * A + a) | (rank << 36)
-            rng = np.random.Generator(np.random.Philox(key=key))
+            linked = rorder[starts[e]:starts[e + 1]]
             new_ev[e, a] = ce._update_entity_value_collapsed(
-                rng, a, attrs[a], part, ent_recs[e], state.dist_probs
+                rng, a, attrs[a], part, linked, state.dist_probs
             )
     state.ent_values = new_ev
 
