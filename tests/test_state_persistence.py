@@ -220,3 +220,101 @@ def test_fast_sweep_matches_reference_posterior():
     ll_s, pairs_s = run(False)
     assert abs(ll_f - ll_s) / abs(ll_s) < 0.02, (ll_f, ll_s)
     assert abs(pairs_f - pairs_s) <= max(2.5, 0.4 * pairs_s), (pairs_f, pairs_s)
+
+
+def test_fast_link_conditional_matches_exact():
+    """The vectorized link update's selection frequencies against exact fp64
+    conditional probabilities on a frozen mid-chain state (same methodology
+    as the GPU kernel tests; the rec_ent drawn in phase 1 is unchanged by
+    the later phases, so full sweeps expose the link conditional)."""
+    import copy
+    import os
+    import sys
+
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    from bench import build_cache_and_records
+    from dblink_amd.engine.cpu_engine import (
+        _build_inverted_index,
+        _get_possible_entities,
+    )
+    from dblink_amd.engine.cpu_fast import sweep_fast
+
+    n = 150
+    cache, rv, rf = build_cache_and_records(n, seed=71)
+    partitioner = KDTreePartitioner(0, [])
+    state0 = deterministic_init(rv, rf, np.arange(n, dtype=np.int64), cache,
+                                partitioner, seed=8)
+    engine = CpuEngine(cache, partitioner)
+    engine.initial_summary(state0)
+    os.environ["DBLINK_CPU_FAST"] = "0"
+    try:
+        for _ in range(8):
+            engine.step(state0, SamplerFlags.for_sampler("PCG-I"))
+    finally:
+        os.environ.pop("DBLINK_CPU_FAST", None)
+
+    # craft multi-candidate conditionals: one non-distorted attribute each
+    attrs = cache.indexed_attributes
+    A = len(attrs)
+    E = state0.num_entities
+    picks = [(5, 1), (17, 2), (42, 3)]
+    rd = state0.rec_dist.copy()
+    for r, keep in picks:
+        rd[r] = 1
+        rd[r, keep] = 0
+    state0.rec_dist = rd
+    inv = _build_inverted_index(state0.ent_values)
+
+    def exact(r):
+        cands, od = _get_possible_entities(
+            state0.rec_values[r], state0.rec_dist[r], inv, E, None,
+            ent_values=state0.ent_values,
+        )
+        w = np.ones(len(cands))
+        for a in od:
+            ia = attrs[a]
+            if ia.is_constant:
+                continue
+            x = int(state0.rec_values[r, a])
+            y = state0.ent_values[cands, a]
+            w *= ia.index.sim_norms[y] * ia.index.exp_sim_many(x, y)
+        return cands, w / w.sum()
+
+    N = 1500
+    counts = {r: np.zeros(E) for r, _ in picks}
+    for i in range(N):
+        st = copy.deepcopy(state0)
+        st.current_seed = 90000 + 11 * i
+        st.iteration = i
+        sweep_fast(st, cache, partitioner, 1)
+        for r in counts:
+            counts[r][st.rec_ent[r]] += 1
+    for r, _ in picks:
+        cands, p = exact(r)
+        assert len(cands) >= 2
+        emp = counts[r] / N
+        full = np.zeros(E)
+        full[cands] = p
+        noise = 0.5 * np.sum(np.sqrt(full * (1 - full) / N))
+        tv = 0.5 * np.abs(emp - full).sum()
+        assert tv < 3 * noise + 0.015, (r, tv, noise)
+
+
+def test_fast_philox_uniformity():
+    """Counter-based Philox stream quality: uniformity and independence
+    across ids, draw indices and iterations."""
+    from dblink_amd.engine.cpu_fast import _philox_uniform, _philox_uniform4
+
+    ids = np.arange(100000, dtype=np.int64)
+    u0 = _philox_uniform(12345, 3, np.uint32(1), ids, 0)
+    assert abs(u0.mean() - 0.5) < 0.005
+    assert abs(u0.std() - (1 / 12) ** 0.5) < 0.005
+    # Kolmogorov-Smirnov style max deviation
+    assert np.abs(np.sort(u0) - np.arange(len(u0)) / len(u0)).max() < 0.01
+    words = _philox_uniform4(12345, 3, np.uint32(1), ids, 0)
+    for i in range(4):
+        for j in range(i + 1, 4):
+            assert abs(np.corrcoef(words[i], words[j])[0, 1]) < 0.02
+    u_next_iter = _philox_uniform(12345, 4, np.uint32(1), ids, 0)
+    assert abs(np.corrcoef(u0, u_next_iter)[0, 1]) < 0.02
+    assert abs(np.corrcoef(u0[:-1], u0[1:])[0, 1]) < 0.02
