@@ -86,6 +86,10 @@ class Project:
 
     @staticmethod
     def _parse_matching_attributes(cfg_list):
+        if not cfg_list:
+            raise ValueError(
+                "dblink.data.matchingAttributes must contain at least one attribute"
+            )
         out = []
         for c in cfg_list:
             sim = similarity_fn_from_config(c.get_config("similarityFunction"))

@@ -105,3 +105,28 @@ def test_unquoted_strings_and_numbers():
     assert cfg.get_double("y") == 2.5
     assert cfg.get_int("z") == -3
     assert cfg.get("w") is None
+
+
+def test_project_rejects_empty_matching_attributes(tmp_path):
+    from dblink_amd.api.project import Project
+    from dblink_amd.utils import hocon as h
+
+    cfg = h.parse_string(
+        """
+        dblink : {
+          data : { path : "x.csv", recordIdentifier : "id",
+                   matchingAttributes : [] }
+          randomSeed : 1
+          expectedMaxClusterSize : 4
+          partitioner : {name : "KDTreePartitioner",
+                         parameters : {numLevels : 0, matchingAttributes : []}}
+          outputPath : "o/"
+          checkpointPath : "c/"
+          steps : []
+        }
+        """
+    )
+    import pytest
+
+    with pytest.raises(ValueError, match="matchingAttributes"):
+        Project(cfg, rank=0, world_size=1)
