@@ -98,17 +98,19 @@ class _FastModel:
                     sp.csr_matrix((si.expsim, si.col, si.row_ptr), shape=(V, V))
                 )
         self.phi_tables = [ia.index.distribution for ia in attrs]
-        self.pow1_tables = [
-            None if ia.is_constant else ia.index.sim_norm_dist(1) for ia in attrs
-        ]
-        self.pow1_w = []  # per attr: weights(v) = phi*norm/Z1 precomputed
+        # cached power-dist normalizers Z_k, k = 0..kmax (Z_0 unused)
+        self.kmax = max(1, min(16, max(
+            getattr(ia.index, "_max_cached_power", 1) for ia in attrs
+        )))
+        self.pow_totals = []
         for ia in attrs:
             if ia.is_constant:
-                self.pow1_w.append(None)
+                self.pow_totals.append(None)
             else:
-                self.pow1_w.append(
-                    ia.index.probs * ia.index.sim_norms / ia.index.sim_norm_total(1)
-                )
+                self.pow_totals.append(np.array(
+                    [1.0] + [ia.index.sim_norm_total(kv)
+                             for kv in range(1, self.kmax + 1)]
+                ))
 
 
 def get_fast_model(cache):
@@ -212,65 +214,138 @@ def sweep_fast(state, cache, partitioner, num_partitions, rank=0):
         for u in _philox_uniform4(seed, it, _PH_VAL, ea_ids.reshape(-1), 0, rank)
     )
 
-    # single observed record per (e, a): its row index
-    first_rec = np.full((E, A), -1, dtype=np.int64)
-    rr = np.arange(R)
-    for a in range(A):
-        o = obs[:, a]
-        first_rec[state.rec_ent[o], a] = rr[o]  # k==1 groups get THE record
-
+    # observed linked records grouped per entity (stable record order)
     new_ev = ev.copy()
     fallback = []
     for a in range(A):
         ia = attrs[a]
         k = kobs[:, a]
         k0 = k == 0
-        if k0.any():  # phi draw
+        if k0.any():  # no observed copies: draw from phi
             t = fm.phi_tables[a]
             new_ev[k0, a] = _alias_draw_vec(u_a1[k0, a], u_a2[k0, a], t.prob, t.alias)
+        kcap = fm.kmax
         k1 = k == 1
         if k1.any():
-            e_idx = np.flatnonzero(k1)
-            r1 = first_rec[e_idx, a]
-            x = rv[r1, a]
-            th = theta_ra[r1, a]
+            # single observed copy (the common case): one sim row, no merge
+            e1 = np.flatnonzero(k1)
+            mo = obs[:, a]
+            rr_o = np.flatnonzero(mo)
+            re_o = state.rec_ent[rr_o]
+            o2 = np.argsort(re_o, kind="stable")
+            r1 = rr_o[o2[np.searchsorted(re_o[o2], e1)]]
+            x1 = rv[r1, a]
+            th1 = theta_ra[r1, a]
             if ia.is_constant:
-                # closed form: P(base) = theta, else keep x
-                take_base = u_mix[e_idx, a] < th
-                tb = e_idx[take_base]
+                # total = 1/theta - 1  =>  P(base) = theta, else keep x
+                take_base = u_mix[e1, a] < th1
                 t = fm.phi_tables[a]
-                new_ev[tb, a] = _alias_draw_vec(u_a1[tb, a], u_a2[tb, a], t.prob, t.alias)
-                new_ev[e_idx[~take_base], a] = x[~take_base]
+                tb = e1[take_base]
+                new_ev[tb, a] = _alias_draw_vec(u_a1[tb, a], u_a2[tb, a],
+                                                t.prob, t.alias)
+                new_ev[e1[~take_base], a] = x1[~take_base]
             else:
                 si = ia.index.sim_index
-                rlo = si.row_ptr[x]
-                rlen = si.row_ptr[x + 1] - rlo
+                rlen = si.row_ptr[x1 + 1] - si.row_ptr[x1]
                 grp, goff = _ragged_expand(rlen)
-                fcol = si.col[rlo[grp] + (np.arange(goff[-1]) - goff[grp])]
-                w = si.expsim[rlo[grp] + (np.arange(goff[-1]) - goff[grp])].copy()
-                px = ia.index.probs[x]
-                normx = ia.index.sim_norms[x]
-                delta = (1.0 / th - 1.0) / (px * normx)
-                self_m = fcol == x[grp]
-                w[self_m] += delta[grp[self_m]]
-                wgt = fm.pow1_w[a][fcol] * (w - 1.0)
+                flat_i = si.row_ptr[x1[grp]] + (np.arange(goff[-1]) - goff[grp])
+                fcol = si.col[flat_i]
+                w = si.expsim[flat_i].copy()
+                pxn = ia.index.probs[x1] * ia.index.sim_norms[x1]
+                selfm = fcol == x1[grp]
+                gsf = grp[selfm]
+                w[selfm] += (1.0 / th1[gsf] - 1.0) / pxn[gsf]
+                wgt = (ia.index.probs[fcol] * ia.index.sim_norms[fcol]
+                       / fm.pow_totals[a][1]) * (w - 1.0)
                 c = np.cumsum(wgt)
-                tot = c[goff[1:] - 1] - np.where(goff[:-1] > 0, c[goff[:-1] - 1], 0.0)
-                take_base = u_mix[e_idx, a] < 1.0 / (1.0 + tot)
-                tb = e_idx[take_base]
-                t1 = fm.pow1_tables[a]
-                new_ev[tb, a] = _alias_draw_vec(u_a1[tb, a], u_a2[tb, a], t1.prob, t1.alias)
+                basec = np.where(goff[:-1] > 0, c[goff[:-1] - 1], 0.0)
+                tot = c[goff[1:] - 1] - basec
+                take_base = u_mix[e1, a] < 1.0 / (1.0 + tot)
+                t1 = ia.index.sim_norm_dist(1)
+                tb = e1[take_base]
+                new_ev[tb, a] = _alias_draw_vec(u_a1[tb, a], u_a2[tb, a],
+                                                t1.prob, t1.alias)
                 pert = ~take_base
                 if pert.any():
-                    basec = np.where(goff[:-1] > 0, c[goff[:-1] - 1], 0.0)[pert]
-                    target = basec + u_sel[e_idx[pert], a] * tot[pert]
+                    target = basec[pert] + u_sel[e1[pert], a] * tot[pert]
                     j = np.searchsorted(c, target, side="right")
                     j = np.minimum(j, goff[1:][pert] - 1)
-                    new_ev[e_idx[pert], a] = fcol[j]
-        k2 = k >= 2
-        if k2.any():
-            for e in np.flatnonzero(k2):
-                fallback.append((int(e), a))
+                    new_ev[e1[pert], a] = fcol[j]
+        km = (k >= 2) & (k <= kcap)
+        for e in np.flatnonzero(k > kcap):  # beyond the cached powers: oracle path
+            fallback.append((int(e), a))
+        if not km.any():
+            continue
+        e_idx = np.flatnonzero(km)
+        kk = k[e_idx]
+        # ragged (pair -> its observed records)
+        mo = obs[:, a]
+        rr_o = np.flatnonzero(mo)
+        re_o = state.rec_ent[rr_o]
+        o2 = np.argsort(re_o, kind="stable")
+        sre = re_o[o2]
+        gstart = np.searchsorted(sre, e_idx)
+        pgrp, poff = _ragged_expand(kk)
+        ridx = rr_o[o2[gstart[pgrp] + (np.arange(poff[-1]) - poff[pgrp])]]
+        x = rv[ridx, a]
+        th = theta_ra[ridx, a]
+        # each record contributes its sim row (constants: the singleton {x})
+        # with the closed-form self-term added at v == x
+        if ia.is_constant:
+            fcol = x
+            px = ia.index.probs[x]
+            w = 1.0 + (1.0 / th - 1.0) / px
+            pair_of = pgrp
+        else:
+            si = ia.index.sim_index
+            rlen = si.row_ptr[x + 1] - si.row_ptr[x]
+            ggrp, goff2 = _ragged_expand(rlen)
+            flat_i = si.row_ptr[x[ggrp]] + (np.arange(goff2[-1]) - goff2[ggrp])
+            fcol = si.col[flat_i]
+            w = si.expsim[flat_i].copy()
+            pxn = ia.index.probs[x] * ia.index.sim_norms[x]
+            selfm = fcol == x[ggrp]
+            gs_ = ggrp[selfm]
+            w[selfm] += (1.0 / th[gs_] - 1.0) / pxn[gs_]
+            pair_of = pgrp[ggrp]
+        # union-combine: product of the records' factors per (pair, value)
+        so = np.lexsort((fcol, pair_of))
+        pc, cc, wc = pair_of[so], fcol[so], w[so]
+        runs = np.flatnonzero(np.r_[True, (pc[1:] != pc[:-1]) | (cc[1:] != cc[:-1])])
+        vw = np.multiply.reduceat(wc, runs)
+        ucol = cc[runs]
+        upair = pc[runs]
+        # base distribution: p_k(v) = phi(v) * norm(v)^k / Z_k (phi for consts)
+        if ia.is_constant:
+            basep = ia.index.probs[ucol]
+        else:
+            kw = kk[upair].astype(np.float64)
+            basep = (ia.index.probs[ucol] * ia.index.sim_norms[ucol] ** kw
+                     / fm.pow_totals[a][kk[upair]])
+        weight = basep * (vw - 1.0)
+        c = np.cumsum(weight)
+        pstart = np.flatnonzero(np.r_[True, upair[1:] != upair[:-1]])
+        pend = np.r_[pstart[1:], len(upair)]
+        basec = np.where(pstart > 0, c[pstart - 1], 0.0)
+        tot = c[pend - 1] - basec
+        take_base = u_mix[e_idx, a] < 1.0 / (1.0 + tot)
+        if ia.is_constant:
+            t = fm.phi_tables[a]
+            tb = e_idx[take_base]
+            new_ev[tb, a] = _alias_draw_vec(u_a1[tb, a], u_a2[tb, a], t.prob, t.alias)
+        else:
+            for kv in np.unique(kk[take_base]):
+                sel = take_base & (kk == kv)
+                tb = e_idx[sel]
+                t = ia.index.sim_norm_dist(int(kv))
+                new_ev[tb, a] = _alias_draw_vec(u_a1[tb, a], u_a2[tb, a],
+                                                t.prob, t.alias)
+        pert = ~take_base
+        if pert.any():
+            target = basec[pert] + u_sel[e_idx[pert], a] * tot[pert]
+            j = np.searchsorted(c, target, side="right")
+            j = np.minimum(j, pend[pert] - 1)
+            new_ev[e_idx[pert], a] = ucol[j]
 
     if fallback:
         from . import cpu_engine as ce
@@ -288,7 +363,7 @@ def sweep_fast(state, cache, partitioner, num_partitions, rank=0):
         rng = np.random.Generator(
             np.random.Philox(key=((seed & ((1 << 63) - 1)) << 64) | (it << 32) | rank)
         )
-        for e, a in fallback:
+        for e, a in sorted(fallback):
             linked = rorder[starts[e]:starts[e + 1]]
             new_ev[e, a] = ce._update_entity_value_collapsed(
                 rng, a, attrs[a], part, linked, state.dist_probs
