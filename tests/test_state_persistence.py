@@ -318,3 +318,77 @@ def test_fast_philox_uniformity():
     u_next_iter = _philox_uniform(12345, 4, np.uint32(1), ids, 0)
     assert abs(np.corrcoef(u0, u_next_iter)[0, 1]) < 0.02
     assert abs(np.corrcoef(u0[:-1], u0[1:])[0, 1]) < 0.02
+
+
+def test_fast_value_k2_conditional_matches_exact():
+    """The union-merge collapsed value update (k=2 cluster) against the exact
+    mixture P(v) = basep_2(v) * vw(v) / (1 + T): link is pinned by unique
+    constant values so the value draw is the only randomness observed."""
+    import copy
+    import os
+    import sys
+
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    from bench import build_cache_and_records
+    from dblink_amd.engine.cpu_fast import sweep_fast
+
+    n = 60
+    cache, rv, rf = build_cache_and_records(n, seed=9)
+    partitioner = KDTreePartitioner(0, [])
+    state0 = deterministic_init(rv, rf, np.arange(n, dtype=np.int64), cache,
+                                partitioner, seed=9)
+    attrs = cache.indexed_attributes
+    a_t = 3  # Levenshtein attribute under test
+    ia = attrs[a_t].index
+
+    # pin records 0 and 1 to entity 0: unique constant values 0/1/2 on the
+    # target entity only, records agree and are non-distorted there
+    state0.ent_values[0, 0:3] = [0, 1, 2]
+    state0.ent_values[1:, 0] = 3  # no other entity shares attr-0 value 0
+    for r in (0, 1):
+        state0.rec_values[r, 0:3] = [0, 1, 2]
+        state0.rec_dist[r, 0:3] = 0
+        state0.rec_dist[r, 3:] = 1
+    x1, x2 = 4, 5
+    state0.rec_values[0, a_t] = x1
+    state0.rec_values[1, a_t] = x2
+    # no other record may link to entity 0: give them a different attr-0 value
+    state0.rec_values[2:, 0] = 3
+    state0.rec_dist[2:, 0] = 0
+
+    theta = state0.dist_probs
+
+    def exact_probs():
+        V = ia.num_values
+        vw = np.ones(V)
+        touched = np.zeros(V, dtype=bool)
+        for r, x in ((0, x1), (1, x2)):
+            th = theta(a_t, int(state0.rec_file[r]))
+            cols, sims = ia.sim_index.row(x)
+            w = sims.copy()
+            pos = int(np.searchsorted(cols, x))
+            if pos < len(cols) and cols[pos] == x:
+                w[pos] = sims[pos] + (1.0 / th - 1.0) / (
+                    ia.probs[x] * ia.sim_norms[x]
+                )
+            vw[cols] *= w
+            touched[cols] = True
+        basep = ia.probs * ia.sim_norms ** 2 / ia.sim_norm_total(2)
+        wgt = np.where(touched, basep * (vw - 1.0), 0.0)
+        T = wgt.sum()
+        return (basep + wgt) / (1.0 + T)
+
+    exact = exact_probs()
+    N = 8000
+    counts = np.zeros(ia.num_values)
+    for i in range(N):
+        st = copy.deepcopy(state0)
+        st.current_seed = 77000 + 5 * i
+        st.iteration = i
+        sweep_fast(st, cache, partitioner, 1)
+        assert st.rec_ent[0] == st.rec_ent[1]  # pinned cluster
+        counts[st.ent_values[st.rec_ent[0], a_t]] += 1
+    emp = counts / N
+    noise = 0.5 * np.sum(np.sqrt(exact * (1 - exact) / N))
+    tv = 0.5 * np.abs(emp - exact).sum()
+    assert tv < 3 * noise + 0.01, (tv, noise)
