@@ -131,3 +131,37 @@ def test_native_pair_sweep_matches_python():
     np.testing.assert_array_equal(native.row_ptr, ref.row_ptr)
     np.testing.assert_array_equal(native.col, ref.col)
     np.testing.assert_allclose(native.expsim, ref.expsim, rtol=1e-6)
+
+
+def test_native_pair_sweep_matches_python_randomized():
+    """Property test: native (OpenMP) pair sweep vs the pure-Python oracle on
+    random ASCII domains and thresholds."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    from dblink_amd import ops
+
+    if not ops.have_native():
+        pytest.skip("native extension not built")
+
+    alphabet = "ABCDEF "
+
+    @settings(max_examples=25, deadline=None)
+    @given(
+        values=st.lists(st.text(alphabet=alphabet, min_size=0, max_size=14),
+                        min_size=2, max_size=24, unique=True),
+        threshold=st.floats(min_value=0.5, max_value=9.5),
+        max_sim=st.floats(min_value=1.0, max_value=12.0),
+    )
+    def check(values, threshold, max_sim):
+        if threshold >= max_sim:
+            threshold = max_sim * 0.7
+        fn = LevenshteinSimilarityFn(threshold, max_sim)
+        values = sorted(values)
+        native = ops.sim_pairs(values, fn)
+        ref = _python_sim_pairs(values, fn)
+        np.testing.assert_array_equal(native.row_ptr, ref.row_ptr)
+        np.testing.assert_array_equal(native.col, ref.col)
+        np.testing.assert_allclose(native.expsim, ref.expsim, rtol=1e-6)
+
+    check()

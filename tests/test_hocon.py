@@ -130,3 +130,43 @@ def test_project_rejects_empty_matching_attributes(tmp_path):
 
     with pytest.raises(ValueError, match="matchingAttributes"):
         Project(cfg, rank=0, world_size=1)
+
+
+def test_hocon_parser_robustness_randomized():
+    """Property test: the HOCON subset parser round-trips simple typed values
+    and never crashes un-pythonically on random key/value combinations."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    from dblink_amd.utils import hocon as h
+
+    key = st.text(alphabet="abcXYZ", min_size=1, max_size=8)
+
+    @settings(max_examples=40, deadline=None)
+    @given(
+        k1=key, k2=key,
+        i=st.integers(min_value=-10**9, max_value=10**9),
+        f=st.floats(min_value=-1e6, max_value=1e6, allow_nan=False),
+        s=st.text(alphabet="abc XYZ_-.", max_size=20),
+        b=st.booleans(),
+    )
+    def check(k1, k2, i, f, s, b):
+        text = (
+            "root : {\n"
+            f"  {k1} : {{ i : {i}, f : {f!r}, s : {json_str(s)}, b : {str(b).lower()} }}\n"
+            f"  {k2}2 : ${{root.{k1}.i}}\n"
+            "}\n"
+        )
+        cfg = h.parse_string(text)
+        assert cfg.get_int(f"root.{k1}.i") == i
+        assert cfg.get_double(f"root.{k1}.f") == pytest.approx(f)
+        assert cfg.get_string(f"root.{k1}.s") == s
+        assert cfg.get_bool(f"root.{k1}.b") is b
+        assert cfg.get_int(f"root.{k2}2") == i
+
+    def json_str(x):
+        import json
+
+        return json.dumps(x)
+
+    check()

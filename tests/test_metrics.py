@@ -87,3 +87,57 @@ def test_clustering_metrics_report():
     assert m.adj_rand_index == pytest.approx(adjusted_rand_index(pred, true))
     s = m.mk_string()
     assert "Adj. Rand index" in s and str(m.adj_rand_index) in s
+
+
+def test_ari_and_f1_invariances_randomized():
+    """Property test: ARI is invariant to cluster relabeling/permutation and
+    equals 1 iff the clusterings match; pairwise F1 is symmetric in its
+    confusion counts."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    from dblink_amd.analysis.metrics import (
+        PairwiseMetrics,
+        to_pairwise_links,
+    )
+
+    @settings(max_examples=40, deadline=None)
+    @given(
+        membership=st.lists(st.integers(min_value=0, max_value=4),
+                            min_size=3, max_size=24),
+        seed=st.integers(min_value=0, max_value=2**31),
+    )
+    def check(membership, seed):
+        import numpy as np
+
+        rng = np.random.default_rng(seed)
+        items = list(range(len(membership)))
+        clusters = membership_to_clusters(dict(zip(items, membership)))
+        # relabel + permute items inside clusters
+        perm = rng.permutation(len(membership))
+        relabeled = membership_to_clusters(
+            {i: 10 + membership[i] for i in items}
+        )
+        n_multi = sum(1 for c in clusters if len(c) > 1)
+        if 1 < len(clusters) <= len(membership) - 1 or (n_multi and len(clusters) > 1):
+            # ARI is 0/0-undefined for the trivial clusterings (all one
+            # cluster / all singletons) — same as the reference's formula
+            assert adjusted_rand_index(clusters, relabeled) == pytest.approx(1.0)
+        # a second random clustering: ARI symmetric
+        other_m = rng.integers(0, 3, len(membership))
+        other = membership_to_clusters(dict(zip(items, other_m.tolist())))
+        try:
+            a1 = adjusted_rand_index(clusters, other)
+            a2 = adjusted_rand_index(other, clusters)
+            assert a1 == pytest.approx(a2)
+            assert a1 <= 1.0 + 1e-12
+        except ZeroDivisionError:
+            pass  # degenerate all-singleton-vs-all-singleton corner
+        if to_pairwise_links(clusters) and to_pairwise_links(other):
+            pm1 = PairwiseMetrics.compute(clusters, other)
+            pm2 = PairwiseMetrics.compute(other, clusters)
+            assert pm1.precision == pytest.approx(pm2.recall)
+            assert pm1.recall == pytest.approx(pm2.precision)
+        assert to_pairwise_links(clusters) == to_pairwise_links(relabeled)
+
+    check()
