@@ -28,7 +28,7 @@ CONF = """dblink : {{
 
     data : {{
         path : "{out}/records.csv"
-        recordIdentifier : "rec_id",
+        recordIdentifier : "rec_id",{file_id_line}
         entityIdentifier : "ent_id"
         nullValue : "NA"
         matchingAttributes : [
@@ -79,14 +79,19 @@ def main():
     ap.add_argument("--burnin", type=int, default=100)
     ap.add_argument("--thin", type=int, default=10)
     ap.add_argument("--seed", type=int, default=42)
+    ap.add_argument("--files", type=int, default=1,
+                    help="spread records over N source files (record linkage "
+                         "with per-file distortion probabilities)")
     args = ap.parse_args()
 
     os.makedirs(args.out, exist_ok=True)
     write_csv(os.path.join(args.out, "records.csv"), args.records,
-              dup_fraction=0.1, seed=args.seed)
+              dup_fraction=0.1, seed=args.seed, num_files=args.files)
     part_attrs = '"fname_c1"' if args.levels > 0 else ""
+    file_id_line = '\n        fileIdentifier : "file_id",' if args.files > 1 else ""
     conf = CONF.format(out=args.out, levels=args.levels, part_attrs=part_attrs,
                        samples=args.samples, burnin=args.burnin, thin=args.thin,
+                       file_id_line=file_id_line,
                        cutoff=args.burnin + (args.samples * args.thin) // 2)
     with open(os.path.join(args.out, "project.conf"), "w") as f:
         f.write(conf)

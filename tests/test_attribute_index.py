@@ -165,3 +165,21 @@ def test_native_pair_sweep_matches_python_randomized():
         np.testing.assert_allclose(native.expsim, ref.expsim, rtol=1e-6)
 
     check()
+
+
+def test_build_inverted_index_unit():
+    """(attr, value) -> ascending entity lists, duplicate values grouped
+    (parity: EntityInvertedIndexTest.scala:6-37)."""
+    from dblink_amd.engine.cpu_engine import _build_inverted_index
+
+    ev = np.array(
+        [[0, 5], [1, 5], [0, 6], [2, 5], [0, 5]],
+        dtype=np.int32,
+    )
+    inv = _build_inverted_index(ev)
+    np.testing.assert_array_equal(inv[(0, 0)], [0, 2, 4])
+    np.testing.assert_array_equal(inv[(0, 1)], [1])
+    np.testing.assert_array_equal(inv[(0, 2)], [3])
+    np.testing.assert_array_equal(inv[(1, 5)], [0, 1, 3, 4])
+    np.testing.assert_array_equal(inv[(1, 6)], [2])
+    assert (0, 3) not in inv
