@@ -28,11 +28,10 @@ from dblink_amd.engine.cpu_engine import CpuEngine, SamplerFlags
 from bench import build_cache_and_records
 
 rank, world, device = comm.init_from_env(backend="gloo")
-assert world == 2, world
 
 n = 240
 cache, rec_values, rec_files = build_cache_and_records(n, seed=11)
-partitioner = KDTreePartitioner(2, [3, 4])
+partitioner = KDTreePartitioner(__LEVELS__, [3, 4, 0][:__LEVELS__])
 bounds = np.linspace(0, n, world + 1).astype(np.int64)
 lo, hi = int(bounds[rank]), int(bounds[rank + 1])
 state = deterministic_init(rec_values[lo:hi], rec_files[lo:hi],
@@ -99,7 +98,7 @@ def _run_workers(script, world=2, timeout=600):
 
 @pytest.mark.slow
 def test_two_rank_chain_conservation():
-    outs = _run_workers(WORKER.replace("__ROOT__", ROOT))
+    outs = _run_workers(WORKER.replace("__ROOT__", ROOT).replace("__LEVELS__", "2"))
     payload = None
     for out in outs:
         for line in out.splitlines():
@@ -145,6 +144,16 @@ dist.destroy_process_group()
 
 
 @pytest.mark.slow
+@pytest.mark.slow
+def test_four_rank_chain_conservation():
+    """Four ranks over eight KD partitions: ownership, conservation and gid
+    coverage invariants at a world size with multi-partition-per-rank
+    migration fan-out (same path as the 8-GPU run)."""
+    outs = _run_workers(WORKER.replace("__ROOT__", ROOT).replace("__LEVELS__", "3"),
+                        world=4)
+    assert any('"ok": true' in o for o in outs), outs
+
+
 def test_two_rank_posterior_matches_single_rank():
     """The 2-rank chain must land in the same stationary log-likelihood band
     as the 1-rank chain on identical data (same model, different RNG streams)."""
