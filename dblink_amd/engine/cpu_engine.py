@@ -442,6 +442,21 @@ def _update_distortions(rng, part, r, y, attrs, dist_probs):
             part.rec_dist[r, a] = True
 
 
+def _exp_sim_pairs(index, xs, ys):
+    """Vectorized exp_sim over elementwise (x, y) pairs (sparse CSR lookup;
+    values not in x's similarity row have exp-similarity 1)."""
+    import scipy.sparse as sp
+
+    si = index.sim_index
+    mat = getattr(index, "_scipy_csr", None)
+    if mat is None:
+        V = index.num_values
+        mat = sp.csr_matrix((si.expsim, si.col, si.row_ptr), shape=(V, V))
+        index._scipy_csr = mat
+    es = np.asarray(mat[np.asarray(xs), np.asarray(ys)]).ravel()
+    return np.where(es > 0.0, es, 1.0)
+
+
 def compute_summary(state: ChainState, cache, dist_probs) -> SummaryVars:
     """Summary variables (GibbsUpdates.scala:219-301) for the LOCAL shard.
 
@@ -479,9 +494,7 @@ def compute_summary(state: ChainState, cache, dist_probs) -> SummaryVars:
                 loglik += float(np.sum(np.log(ia.index.probs[xo])))
             else:
                 y = state.ent_values[state.rec_ent[d], a][obs]
-                es = np.array(
-                    [ia.index.exp_sim_of(int(xx), int(yy)) for xx, yy in zip(xo, y)]
-                )
+                es = _exp_sim_pairs(ia.index, xo, y)
                 loglik += float(
                     np.sum(np.log(ia.index.probs[xo] * ia.index.sim_norms[y] * es))
                 )
