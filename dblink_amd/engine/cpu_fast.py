@@ -214,7 +214,9 @@ def sweep_fast(state, cache, partitioner, num_partitions, rank=0):
         for u in _philox_uniform4(seed, it, _PH_VAL, ea_ids.reshape(-1), 0, rank)
     )
 
-    # observed linked records grouped per entity (stable record order)
+    # records in entity order, computed once; per attribute the observed
+    # subset stays entity-sorted after masking
+    rorder_all = np.argsort(state.rec_ent, kind="stable")
     new_ev = ev.copy()
     fallback = []
     for a in range(A):
@@ -229,11 +231,8 @@ def sweep_fast(state, cache, partitioner, num_partitions, rank=0):
         if k1.any():
             # single observed copy (the common case): one sim row, no merge
             e1 = np.flatnonzero(k1)
-            mo = obs[:, a]
-            rr_o = np.flatnonzero(mo)
-            re_o = state.rec_ent[rr_o]
-            o2 = np.argsort(re_o, kind="stable")
-            r1 = rr_o[o2[np.searchsorted(re_o[o2], e1)]]
+            rr_s = rorder_all[obs[rorder_all, a]]
+            r1 = rr_s[np.searchsorted(state.rec_ent[rr_s], e1)]
             x1 = rv[r1, a]
             th1 = theta_ra[r1, a]
             if ia.is_constant:
@@ -278,15 +277,11 @@ def sweep_fast(state, cache, partitioner, num_partitions, rank=0):
             continue
         e_idx = np.flatnonzero(km)
         kk = k[e_idx]
-        # ragged (pair -> its observed records)
-        mo = obs[:, a]
-        rr_o = np.flatnonzero(mo)
-        re_o = state.rec_ent[rr_o]
-        o2 = np.argsort(re_o, kind="stable")
-        sre = re_o[o2]
-        gstart = np.searchsorted(sre, e_idx)
+        # ragged (pair -> its observed records), entity-sorted
+        rr_s = rorder_all[obs[rorder_all, a]]
+        gstart = np.searchsorted(state.rec_ent[rr_s], e_idx)
         pgrp, poff = _ragged_expand(kk)
-        ridx = rr_o[o2[gstart[pgrp] + (np.arange(poff[-1]) - poff[pgrp])]]
+        ridx = rr_s[gstart[pgrp] + (np.arange(poff[-1]) - poff[pgrp])]
         x = rv[ridx, a]
         th = theta_ra[ridx, a]
         # each record contributes its sim row (constants: the singleton {x})
@@ -357,14 +352,13 @@ def sweep_fast(state, cache, partitioner, num_partitions, rank=0):
         part.rec_values = rv
         part.rec_file = state.rec_file
         part.rec_dist = state.rec_dist
-        rorder = np.argsort(state.rec_ent, kind="stable")
-        starts = np.searchsorted(state.rec_ent[rorder], np.arange(E + 1))
+        starts = np.searchsorted(state.rec_ent[rorder_all], np.arange(E + 1))
         # one sweep-keyed stream, consumed in deterministic (e, a) order
         rng = np.random.Generator(
             np.random.Philox(key=((seed & ((1 << 63) - 1)) << 64) | (it << 32) | rank)
         )
         for e, a in sorted(fallback):
-            linked = rorder[starts[e]:starts[e + 1]]
+            linked = rorder_all[starts[e]:starts[e + 1]]
             new_ev[e, a] = ce._update_entity_value_collapsed(
                 rng, a, attrs[a], part, linked, state.dist_probs
             )
