@@ -392,3 +392,58 @@ def test_fast_value_k2_conditional_matches_exact():
     noise = 0.5 * np.sum(np.sqrt(exact * (1 - exact) / N))
     tv = 0.5 * np.abs(emp - exact).sum()
     assert tv < 3 * noise + 0.01, (tv, noise)
+
+
+def test_fast_distortion_conditional_matches_exact():
+    """Vectorized distortion resample: empirical z rates for the three cases
+    (missing / agree / disagree) vs the exact Bernoulli probabilities, on a
+    pinned single-record cluster so y is deterministic."""
+    import copy
+    import os
+    import sys
+
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    from bench import build_cache_and_records
+    from dblink_amd.engine.cpu_fast import sweep_fast
+
+    n = 50
+    cache, rv, rf = build_cache_and_records(n, seed=12)
+    partitioner = KDTreePartitioner(0, [])
+    state0 = deterministic_init(rv, rf, np.arange(n, dtype=np.int64), cache,
+                                partitioner, seed=12)
+    attrs = cache.indexed_attributes
+    # pin record 0 alone on entity 0 via a unique constant value
+    state0.ent_values[0, 0] = 0
+    state0.ent_values[1:, 0] = 3
+    state0.rec_values[0, 0] = 0
+    state0.rec_dist[0, :] = 0
+    state0.rec_values[1:, 0] = 3
+    state0.rec_dist[1:, 0] = 0
+    # cases on record 0: attr 1 missing, attr 4 observed
+    state0.rec_values[0, 1] = -1
+    x4 = int(state0.rec_values[0, 4])
+    assert x4 >= 0
+
+    N = 5000
+    z1 = z4 = 0
+    th = None
+    for i in range(N):
+        st = copy.deepcopy(state0)
+        st.current_seed = 31000 + 3 * i
+        st.iteration = i
+        sweep_fast(st, cache, partitioner, 1)
+        row = int(np.flatnonzero(st.rec_gid == 0)[0])
+        e = int(st.rec_ent[row])
+        z1 += int(st.rec_dist[row, 1])
+        # agree case only when the entity kept the record's value
+        if st.ent_values[e, 4] == x4:
+            z4 += int(st.rec_dist[row, 4])
+            if th is None:
+                th = st.dist_probs(4, int(st.rec_file[row]))
+    theta1 = state0.dist_probs(1, int(state0.rec_file[0]))
+    assert z1 / N == pytest.approx(theta1, abs=0.02)
+    ia = attrs[4].index
+    pr1 = th * ia.self_mass[x4]
+    p_agree = pr1 / (pr1 + (1 - th))
+    # nearly every replica keeps the value (k=1 self term dominates)
+    assert z4 / N == pytest.approx(p_agree, abs=0.02)
