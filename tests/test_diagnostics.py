@@ -80,3 +80,48 @@ def test_fast_smpc_matches_reference_impl(tmp_path):
     s_ref = sorted(map(sorted, shared_most_probable_clusters(table)))
     s_fast = sorted(map(sorted, shared_most_probable_clusters_fast(table)))
     assert s_ref == s_fast
+
+
+def test_sampler_burnin_thinning_record_rule(tmp_path):
+    """Exact recorded-iteration set under burn-in + thinning (the reference's
+    ``>= burninInterval`` record rule, Sampler.scala:92-115): with burnin=5,
+    thin=3, sampleSize=4 and iteration 0 also recorded only when burnin=0."""
+    import os
+    import sys
+
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    import pyarrow.parquet  # noqa: F401
+    from bench import build_cache_and_records
+    from dblink_amd.analysis.chain import load_chain
+    from dblink_amd.engine import sampler as sampler_m
+    from dblink_amd.engine.cpu_engine import CpuEngine, SamplerFlags
+    from dblink_amd.engine.init import deterministic_init
+    from dblink_amd.parallel.partitioning import KDTreePartitioner
+
+    n = 40
+    cache, rv, rf = build_cache_and_records(n, seed=6)
+    partitioner = KDTreePartitioner(0, [])
+    state = deterministic_init(rv, rf, np.arange(n, dtype=np.int64), cache,
+                               partitioner, seed=6)
+    engine = CpuEngine(cache, partitioner)
+    engine.rec_ids_array = np.array([str(i) for i in range(n)], dtype=object)
+    engine.initial_summary(state)
+    out = str(tmp_path / "o1")
+    sampler_m.sample(engine, state, 4, out, burnin_interval=5,
+                     thinning_interval=3, checkpoint_interval=0,
+                     flags=SamplerFlags.for_sampler("PCG-I"))
+    got = sorted(set(load_chain(out)["iteration"].to_pylist()))
+    assert got == [5, 8, 11, 14], got
+
+    # burnin = 0 records the INITIAL state too (iteration 0)
+    state2 = deterministic_init(rv, rf, np.arange(n, dtype=np.int64), cache,
+                                partitioner, seed=6)
+    engine2 = CpuEngine(cache, partitioner)
+    engine2.rec_ids_array = engine.rec_ids_array
+    engine2.initial_summary(state2)
+    out2 = str(tmp_path / "o2")
+    sampler_m.sample(engine2, state2, 3, out2, burnin_interval=0,
+                     thinning_interval=2, checkpoint_interval=0,
+                     flags=SamplerFlags.for_sampler("PCG-I"))
+    got2 = sorted(set(load_chain(out2)["iteration"].to_pylist()))
+    assert got2 == [0, 2, 4, 6], got2
