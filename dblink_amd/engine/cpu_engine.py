@@ -164,6 +164,11 @@ def _cpu_fast_enabled():
     return os.environ.get("DBLINK_CPU_FAST", "1") != "0"
 
 
+import os as _os  # noqa: E402
+
+_PHASE_TIMERS = {} if _os.environ.get("DBLINK_PHASE_TIMERS", "") == "1" else None
+
+
 def sweep_partition(
     rng,
     part,  # _PartitionData
@@ -538,7 +543,8 @@ def sweep(
         from .cpu_fast import sweep_fast
 
         rank = comm.rank_world()[0] if comm.is_distributed() else 0
-        sweep_fast(state, cache, partitioner, num_partitions, rank=rank)
+        sweep_fast(state, cache, partitioner, num_partitions, rank=rank,
+                   timers=_PHASE_TIMERS)
         return
     if state.num_entities and int(state.ent_part.max()) >= num_partitions:
         raise RuntimeError(
@@ -588,6 +594,7 @@ class CpuEngine:
         self.world_size = world_size
         self.rank = rank
         self.num_partitions = partitioner.num_partitions
+        self.phase_timers = _PHASE_TIMERS is not None
 
     # -- theta ---------------------------------------------------------------
 
@@ -652,6 +659,11 @@ class CpuEngine:
         add_prior_terms(summary, self.cache, state.dist_probs)
         state.summary = summary
         return state
+
+    def phase_times(self):
+        """Cumulative {phase: ms} from the vectorized sweep
+        (DBLINK_PHASE_TIMERS=1; same contract as GpuEngine.phase_times)."""
+        return dict(_PHASE_TIMERS or {})
 
     def initial_summary(self, state: ChainState):
         local = compute_summary(state, self.cache, state.dist_probs)

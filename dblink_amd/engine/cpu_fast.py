@@ -121,12 +121,24 @@ def get_fast_model(cache):
     return got
 
 
-def sweep_fast(state, cache, partitioner, num_partitions, rank=0):
+def sweep_fast(state, cache, partitioner, num_partitions, rank=0, timers=None):
     """One PCG-I sweep over the rank's (partition-sorted) state, in place.
 
     Same contract as ``cpu_engine.sweep``: advances current_seed by
-    num_partitions and increments the iteration counter.
+    num_partitions and increments the iteration counter. ``timers`` (a dict)
+    accumulates per-phase wall-clock ms when given (DBLINK_PHASE_TIMERS=1).
     """
+    import time as _time
+
+    def _mark(name, _t=[None]):
+        if timers is None:
+            return
+        now = _time.perf_counter()
+        if _t[0] is not None and name is not None:
+            timers[name] = timers.get(name, 0.0) + (now - _t[0]) * 1000.0
+        _t[0] = now
+
+    _mark(None)
     fm = get_fast_model(cache)
     attrs, A = fm.attrs, fm.A
     seed = int(state.current_seed)
@@ -204,6 +216,7 @@ def sweep_fast(state, cache, partitioner, num_partitions, rank=0):
     if not np.isfinite(score[best_flat]).all():
         raise RuntimeError("empty candidate set: state invariant violated")
     state.rec_ent = cand[best_flat].astype(np.int64)
+    _mark("link")
 
     # ---- phase 2: collapsed entity-value update -----------------------------
     kobs = np.zeros((E, A), dtype=np.int64)
@@ -363,6 +376,7 @@ def sweep_fast(state, cache, partitioner, num_partitions, rank=0):
                 rng, a, attrs[a], part, linked, state.dist_probs
             )
     state.ent_values = new_ev
+    _mark("value")
 
     # ---- phase 3: distortion update -----------------------------------------
     y_link = state.ent_values[state.rec_ent]  # [R, A]
@@ -379,8 +393,10 @@ def sweep_fast(state, cache, partitioner, num_partitions, rank=0):
         np.where(rv == y_link, u_d < p_agree, True),
     )
     state.rec_dist = z.astype(np.uint8)
+    _mark("distortion")
 
     # ---- partition reassignment + bookkeeping -------------------------------
     state.ent_part = partitioner.get_partition_ids(state.ent_values).astype(np.int32)
     state.current_seed += num_partitions
     state.iteration += 1
+    _mark("kd")
