@@ -29,35 +29,57 @@ _W1 = np.uint32(0xBB67AE85)
 
 def _philox_uniform4(seed, iteration, phase, ids, draw, rank=0):
     """Philox4x32-10 keyed uniforms in (0, 1): FOUR independent streams per
-    id (the four 32-bit output words) for the price of one keyed evaluation."""
-    ids = np.asarray(ids, dtype=np.uint64)
-    c0 = ids.astype(np.uint32)
-    c1 = (ids >> np.uint64(32)).astype(np.uint32)
-    c2 = np.full_like(c0, np.uint32(draw))
-    c3 = np.full_like(c0, np.uint32(iteration))
-    k0 = np.full_like(c0, np.uint32(np.uint64(seed) & np.uint64(0xFFFFFFFF)))
-    k1 = np.full_like(
-        c0,
-        np.uint32((np.uint64(seed) >> np.uint64(32)) & np.uint64(0xFFFFFFFF))
-        ^ np.uint32((int(phase) * 0x9E3779B1) & 0xFFFFFFFF)
-        ^ np.uint32((rank * 0x85EBCA6B) & 0xFFFFFFFF),
+    id (the four 32-bit output words) for the price of one keyed evaluation.
+    State lives in masked uint64 lanes with preallocated scratch — the 10
+    rounds run allocation-free."""
+    ids = np.ascontiguousarray(ids, dtype=np.uint64)
+    n = len(ids)
+    M32 = np.uint64(0xFFFFFFFF)
+    c0 = ids & M32
+    c1 = ids >> np.uint64(32)
+    c2 = np.full(n, np.uint64(np.uint32(draw)))
+    c3 = np.full(n, np.uint64(np.uint32(iteration)))
+    k0 = np.uint64(np.uint64(seed) & M32)
+    k1 = np.uint64(
+        int((np.uint64(seed) >> np.uint64(32)) & M32)
+        ^ ((int(phase) * 0x9E3779B1) & 0xFFFFFFFF)
+        ^ ((rank * 0x85EBCA6B) & 0xFFFFFFFF)
     )
+    W0, W1 = np.uint64(0x9E3779B9), np.uint64(0xBB67AE85)
+    M0, M1 = _M0, _M1
+    p0 = np.empty(n, dtype=np.uint64)
+    p1 = np.empty(n, dtype=np.uint64)
+    t = np.empty(n, dtype=np.uint64)
     for _ in range(10):
-        p0 = c0.astype(np.uint64) * _M0
-        p1 = c2.astype(np.uint64) * _M1
-        hi0 = (p0 >> np.uint64(32)).astype(np.uint32)
-        lo0 = p0.astype(np.uint32)
-        hi1 = (p1 >> np.uint64(32)).astype(np.uint32)
-        lo1 = p1.astype(np.uint32)
-        c0, c1, c2, c3 = hi1 ^ c1 ^ k0, lo1, hi0 ^ c3 ^ k1, lo0
-        k0 = k0 + _W0
-        k1 = k1 + _W1
-    to_u = lambda c: (c.astype(np.float64) + 0.5) * (2.0 ** -32)  # noqa: E731
+        np.multiply(c0, M0, out=p0)
+        np.multiply(c2, M1, out=p1)
+        # c0' = hi1 ^ c1 ^ k0 ; c1' = lo1 ; c2' = hi0 ^ c3 ^ k1 ; c3' = lo0
+        np.right_shift(p1, np.uint64(32), out=t)
+        np.bitwise_xor(t, c1, out=t)
+        np.bitwise_xor(t, k0, out=t)
+        np.bitwise_and(p1, M32, out=c1)
+        c0, t = t, c0
+        np.right_shift(p0, np.uint64(32), out=t)
+        np.bitwise_xor(t, c3, out=t)
+        np.bitwise_xor(t, k1, out=t)
+        np.bitwise_and(p0, M32, out=c3)
+        c2, t = t, c2
+        k0 = (k0 + W0) & M32
+        k1 = (k1 + W1) & M32
+    to_u = lambda c: ((c & M32).astype(np.float64) + 0.5) * (2.0 ** -32)  # noqa: E731
     return to_u(c0), to_u(c1), to_u(c2), to_u(c3)
 
 
 def _philox_uniform(seed, iteration, phase, ids, draw, rank=0):
     return _philox_uniform4(seed, iteration, phase, ids, draw, rank)[0]
+
+
+def _philox_dense(seed, iteration, phase, n, rank=0):
+    """n iid uniforms keyed by dense position: one Philox evaluation yields
+    four output words, so positions are packed four per counter."""
+    m = (n + 3) // 4
+    w = _philox_uniform4(seed, iteration, phase, np.arange(m, dtype=np.int64), 0, rank)
+    return np.stack(w, axis=1).reshape(-1)[:n]
 
 
 def _alias_draw_vec(u1, u2, prob, alias):
@@ -206,8 +228,10 @@ def sweep_fast(state, cache, partitioner, num_partitions, rank=0, timers=None):
             es = np.asarray(fm.sim_csr[a][x, y]).ravel()
             es = np.where(es > 0.0, es, 1.0)
             logw[od_m] += fm.log_norms[a][y] + np.log(es)
-    u = _philox_uniform(seed, it, _PH_LINK, state.rec_gid[rec_of] * 64 + (cand & 63),
-                        cand >> 6, rank)
+    # Gumbel draws keyed by flat position: the candidate list order is a
+    # deterministic function of the state, so position-keyed iid uniforms
+    # give the same conditional distribution as entity-keyed ones
+    u = _philox_dense(seed, it, _PH_LINK, len(cand), rank)
     score = np.where(ok, logw - np.log(-np.log(u)), -np.inf)
     # segmented argmax: sort by (record, score) and take the last per record
     perm = np.lexsort((score, rec_of))
@@ -380,8 +404,7 @@ def sweep_fast(state, cache, partitioner, num_partitions, rank=0, timers=None):
 
     # ---- phase 3: distortion update -----------------------------------------
     y_link = state.ent_values[state.rec_ent]  # [R, A]
-    ra_ids = (state.rec_gid[:, None] * 32 + np.arange(A)[None, :]).reshape(-1)
-    u_d = _philox_uniform(seed, it, _PH_DIST, ra_ids, 0, rank).reshape(R, A)
+    u_d = _philox_dense(seed, it, _PH_DIST, R * A, rank).reshape(R, A)
     self_mass = np.stack(
         [ia.index.self_mass[np.maximum(rv[:, a], 0)] for a, ia in enumerate(attrs)], 1
     )
