@@ -533,18 +533,19 @@ def sweep(
     Does NOT exchange clusters between ranks (see parallel.migration) and does
     not compute summaries. Advances ``current_seed`` by num_partitions.
 
-    The flagship PCG-I variant runs through the vectorized whole-rank sweep
-    (``cpu_fast.sweep_fast``) unless DBLINK_CPU_FAST=0; the per-record loops
-    below remain the numerical oracle and serve the other variants.
+    The indexed variants (PCG-I and plain Gibbs) run through the vectorized
+    whole-rank sweep (``cpu_fast.sweep_fast``) unless DBLINK_CPU_FAST=0; the
+    per-record loops below remain the numerical oracle and serve PCG-II and
+    Gibbs-Sequential.
     """
-    if (_cpu_fast_enabled() and flags.collapsed_entity_values
-            and not flags.collapsed_entity_ids and not flags.sequential):
+    if (_cpu_fast_enabled() and not flags.collapsed_entity_ids
+            and not flags.sequential):
         from ..parallel import comm
         from .cpu_fast import sweep_fast
 
         rank = comm.rank_world()[0] if comm.is_distributed() else 0
         sweep_fast(state, cache, partitioner, num_partitions, rank=rank,
-                   timers=_PHASE_TIMERS)
+                   timers=_PHASE_TIMERS, collapsed=flags.collapsed_entity_values)
         return
     if state.num_entities and int(state.ent_part.max()) >= num_partitions:
         raise RuntimeError(

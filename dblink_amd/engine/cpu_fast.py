@@ -143,8 +143,14 @@ def get_fast_model(cache):
     return got
 
 
-def sweep_fast(state, cache, partitioner, num_partitions, rank=0, timers=None):
-    """One PCG-I sweep over the rank's (partition-sorted) state, in place.
+def sweep_fast(state, cache, partitioner, num_partitions, rank=0, timers=None,
+               collapsed=True):
+    """One PCG-I (``collapsed=True``) or plain-Gibbs sweep over the rank's
+    (partition-sorted) state, in place. The variants share the indexed link
+    and distortion updates; the value update differs — Gibbs copies the
+    (unique) value of any linked non-distorted record and samples the
+    perturbation WITHOUT the collapsed ``(1/theta - 1)`` self term
+    (GibbsUpdates.scala:605-646 vs :576-599).
 
     Same contract as ``cpu_engine.sweep``: advances current_seed by
     num_partitions and increments the iteration counter. ``timers`` (a dict)
@@ -256,9 +262,27 @@ def sweep_fast(state, cache, partitioner, num_partitions, rank=0, timers=None):
     rorder_all = np.argsort(state.rec_ent, kind="stable")
     new_ev = ev.copy()
     fallback = []
+    if not collapsed:
+        # Gibbs: a non-distorted observed copy pins the value (all such
+        # copies agree with the entity by chain invariant)
+        ndm = obs & ~rdist
+        nd_count = np.zeros((E, A), dtype=np.int64)
+        np.add.at(nd_count, state.rec_ent, ndm.astype(np.int64))
+        copy_val = np.zeros((E, A), dtype=rv.dtype)
+        rr_all = np.arange(R)
+        for a in range(A):
+            m = ndm[:, a]
+            copy_val[state.rec_ent[m], a] = rv[m, a]
     for a in range(A):
         ia = attrs[a]
-        k = kobs[:, a]
+        k = kobs[:, a].copy()
+        if not collapsed:
+            pinned = nd_count[:, a] > 0
+            new_ev[pinned, a] = copy_val[pinned, a]
+            k[pinned] = -1  # handled; excluded from every draw branch below
+            if ia.is_constant:
+                # constant attr, observed but all distorted: plain phi draw
+                k[k >= 1] = 0
         k0 = k == 0
         if k0.any():  # no observed copies: draw from phi
             t = fm.phi_tables[a]
@@ -273,7 +297,8 @@ def sweep_fast(state, cache, partitioner, num_partitions, rank=0, timers=None):
             x1 = rv[r1, a]
             th1 = theta_ra[r1, a]
             if ia.is_constant:
-                # total = 1/theta - 1  =>  P(base) = theta, else keep x
+                # collapsed closed form: total = 1/theta - 1 => P(base) = theta
+                # (the non-collapsed constant case never reaches here)
                 take_base = u_mix[e1, a] < th1
                 t = fm.phi_tables[a]
                 tb = e1[take_base]
@@ -287,10 +312,11 @@ def sweep_fast(state, cache, partitioner, num_partitions, rank=0, timers=None):
                 flat_i = si.row_ptr[x1[grp]] + (np.arange(goff[-1]) - goff[grp])
                 fcol = si.col[flat_i]
                 w = si.expsim[flat_i].copy()
-                pxn = ia.index.probs[x1] * ia.index.sim_norms[x1]
-                selfm = fcol == x1[grp]
-                gsf = grp[selfm]
-                w[selfm] += (1.0 / th1[gsf] - 1.0) / pxn[gsf]
+                if collapsed:
+                    pxn = ia.index.probs[x1] * ia.index.sim_norms[x1]
+                    selfm = fcol == x1[grp]
+                    gsf = grp[selfm]
+                    w[selfm] += (1.0 / th1[gsf] - 1.0) / pxn[gsf]
                 wgt = (ia.index.probs[fcol] * ia.index.sim_norms[fcol]
                        / fm.pow_totals[a][1]) * (w - 1.0)
                 c = np.cumsum(wgt)
@@ -326,7 +352,7 @@ def sweep_fast(state, cache, partitioner, num_partitions, rank=0, timers=None):
         if ia.is_constant:
             fcol = x
             px = ia.index.probs[x]
-            w = 1.0 + (1.0 / th - 1.0) / px
+            w = 1.0 + (1.0 / th - 1.0) / px  # collapsed only (see gate above)
             pair_of = pgrp
         else:
             si = ia.index.sim_index
@@ -335,10 +361,11 @@ def sweep_fast(state, cache, partitioner, num_partitions, rank=0, timers=None):
             flat_i = si.row_ptr[x[ggrp]] + (np.arange(goff2[-1]) - goff2[ggrp])
             fcol = si.col[flat_i]
             w = si.expsim[flat_i].copy()
-            pxn = ia.index.probs[x] * ia.index.sim_norms[x]
-            selfm = fcol == x[ggrp]
-            gs_ = ggrp[selfm]
-            w[selfm] += (1.0 / th[gs_] - 1.0) / pxn[gs_]
+            if collapsed:
+                pxn = ia.index.probs[x] * ia.index.sim_norms[x]
+                selfm = fcol == x[ggrp]
+                gs_ = ggrp[selfm]
+                w[selfm] += (1.0 / th[gs_] - 1.0) / pxn[gs_]
             pair_of = pgrp[ggrp]
         # union-combine: product of the records' factors per (pair, value)
         so = np.lexsort((fcol, pair_of))
@@ -396,9 +423,14 @@ def sweep_fast(state, cache, partitioner, num_partitions, rank=0, timers=None):
         )
         for e, a in sorted(fallback):
             linked = rorder_all[starts[e]:starts[e + 1]]
-            new_ev[e, a] = ce._update_entity_value_collapsed(
-                rng, a, attrs[a], part, linked, state.dist_probs
-            )
+            if collapsed:
+                new_ev[e, a] = ce._update_entity_value_collapsed(
+                    rng, a, attrs[a], part, linked, state.dist_probs
+                )
+            else:
+                new_ev[e, a] = ce._update_entity_value(
+                    rng, a, attrs[a], part, linked
+                )
     state.ent_values = new_ev
     _mark("value")
 

@@ -222,6 +222,42 @@ def test_fast_sweep_matches_reference_posterior():
     assert abs(pairs_f - pairs_s) <= max(2.5, 0.4 * pairs_s), (pairs_f, pairs_s)
 
 
+def test_fast_sweep_matches_reference_posterior_gibbs():
+    """Same fast-vs-reference band for the plain Gibbs variant (value update
+    = non-distorted copy / perturbation without the collapsed self term)."""
+    import os
+    import sys
+
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    from bench import build_cache_and_records
+
+    def run(fast, iters=260, n=150, seed=5):
+        os.environ["DBLINK_CPU_FAST"] = "1" if fast else "0"
+        try:
+            cache, rv, rf = build_cache_and_records(n, seed=seed)
+            partitioner = KDTreePartitioner(0, [])
+            state = deterministic_init(rv, rf, np.arange(n, dtype=np.int64),
+                                       cache, partitioner, seed=seed)
+            engine = CpuEngine(cache, partitioner)
+            engine.initial_summary(state)
+            flags = SamplerFlags.for_sampler("Gibbs")
+            lls, pairs = [], []
+            for i in range(iters):
+                engine.step(state, flags)
+                if i >= iters // 2:
+                    lls.append(state.summary.log_likelihood)
+                    c = np.bincount(state.rec_ent, minlength=state.num_entities)
+                    pairs.append(int(np.sum(c * (c - 1) // 2)))
+            return float(np.mean(lls)), float(np.mean(pairs))
+        finally:
+            os.environ.pop("DBLINK_CPU_FAST", None)
+
+    ll_f, pairs_f = run(True)
+    ll_s, pairs_s = run(False)
+    assert abs(ll_f - ll_s) / abs(ll_s) < 0.02, (ll_f, ll_s)
+    assert abs(pairs_f - pairs_s) <= max(2.5, 0.4 * pairs_s), (pairs_f, pairs_s)
+
+
 def test_fast_link_conditional_matches_exact():
     """The vectorized link update's selection frequencies against exact fp64
     conditional probabilities on a frozen mid-chain state (same methodology
