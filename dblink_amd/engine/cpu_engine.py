@@ -538,15 +538,26 @@ def sweep(
     per-record loops below remain the numerical oracle and serve PCG-II and
     Gibbs-Sequential.
     """
-    if (_cpu_fast_enabled() and not flags.collapsed_entity_ids
-            and not flags.sequential):
-        from ..parallel import comm
-        from .cpu_fast import sweep_fast
+    if _cpu_fast_enabled() and not flags.sequential:
+        dense_ok = True
+        if flags.collapsed_entity_ids:
+            # PCG-II is dense over each partition's entities (like the
+            # reference); only vectorize when the largest R_p x E_p block
+            # is affordable, else fall through to the per-record loops
+            ec = np.bincount(state.ent_part, minlength=num_partitions)
+            rc = np.bincount(state.ent_part[state.rec_ent], minlength=num_partitions)
+            dense_ok = (flags.collapsed_entity_values
+                        and int((ec.astype(np.int64) * rc).max()) <= 30_000_000)
+        if not flags.collapsed_entity_ids or dense_ok:
+            from ..parallel import comm
+            from .cpu_fast import sweep_fast
 
-        rank = comm.rank_world()[0] if comm.is_distributed() else 0
-        sweep_fast(state, cache, partitioner, num_partitions, rank=rank,
-                   timers=_PHASE_TIMERS, collapsed=flags.collapsed_entity_values)
-        return
+            rank = comm.rank_world()[0] if comm.is_distributed() else 0
+            sweep_fast(state, cache, partitioner, num_partitions, rank=rank,
+                       timers=_PHASE_TIMERS,
+                       collapsed=flags.collapsed_entity_values,
+                       collapsed_ids=flags.collapsed_entity_ids)
+            return
     if state.num_entities and int(state.ent_part.max()) >= num_partitions:
         raise RuntimeError(
             "state has partition ids beyond the partitioner's range "

@@ -483,3 +483,38 @@ def test_fast_distortion_conditional_matches_exact():
     p_agree = pr1 / (pr1 + (1 - th))
     # nearly every replica keeps the value (k=1 self term dominates)
     assert z4 / N == pytest.approx(p_agree, abs=0.02)
+
+
+def test_fast_sweep_matches_reference_posterior_pcg2():
+    """Fast-vs-reference band for PCG-II (dense collapsed link update)."""
+    import os
+    import sys
+
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    from bench import build_cache_and_records
+
+    def run(fast, iters=200, n=120, seed=5):
+        os.environ["DBLINK_CPU_FAST"] = "1" if fast else "0"
+        try:
+            cache, rv, rf = build_cache_and_records(n, seed=seed)
+            partitioner = KDTreePartitioner(1, [3])
+            state = deterministic_init(rv, rf, np.arange(n, dtype=np.int64),
+                                       cache, partitioner, seed=seed)
+            engine = CpuEngine(cache, partitioner)
+            engine.initial_summary(state)
+            flags = SamplerFlags.for_sampler("PCG-II")
+            lls, pairs = [], []
+            for i in range(iters):
+                engine.step(state, flags)
+                if i >= iters // 2:
+                    lls.append(state.summary.log_likelihood)
+                    c = np.bincount(state.rec_ent, minlength=state.num_entities)
+                    pairs.append(int(np.sum(c * (c - 1) // 2)))
+            return float(np.mean(lls)), float(np.mean(pairs))
+        finally:
+            os.environ.pop("DBLINK_CPU_FAST", None)
+
+    ll_f, pairs_f = run(True)
+    ll_s, pairs_s = run(False)
+    assert abs(ll_f - ll_s) / abs(ll_s) < 0.02, (ll_f, ll_s)
+    assert abs(pairs_f - pairs_s) <= max(2.5, 0.4 * pairs_s), (pairs_f, pairs_s)
