@@ -161,9 +161,10 @@ def _link_dense_collapsed(state, fm, theta_ra, num_partitions, seed, it, rank):
         Rp, Ep = r1 - r0, e1 - e0
         if Rp == 0:
             continue
-        # linear-space weight product: one log at the end instead of one per
-        # attribute (w >= theta*phi*norm ~ 1e-6, so >= 40 attrs fit in f64)
-        wprod = np.ones((Rp, Ep))
+        # linear-space weight product in f32 (matching the GPU kernels'
+        # weight precision; the dense block is memory-bandwidth-bound), one
+        # log avoided entirely via the inverse-CDF draw below
+        wprod = np.ones((Rp, Ep), dtype=np.float32)
         for a in range(A):
             ia = attrs[a]
             xo = rv[r0:r1, a]
@@ -171,9 +172,9 @@ def _link_dense_collapsed(state, fm, theta_ra, num_partitions, seed, it, rank):
             if not obs.any():
                 continue
             y = ev[e0:e1, a]
-            th = theta_ra[r0:r1, a]
-            px = ia.index.probs[np.maximum(xo, 0)]
-            agree = (xo[:, None] == y[None, :]) * (1.0 - th)[:, None]
+            th = theta_ra[r0:r1, a].astype(np.float32)
+            px = ia.index.probs[np.maximum(xo, 0)].astype(np.float32)
+            agree = (xo[:, None] == y[None, :]) * (np.float32(1.0) - th)[:, None]
             if ia.is_constant:
                 w = agree + (th * px)[:, None]
             else:
@@ -192,15 +193,17 @@ def _link_dense_collapsed(state, fm, theta_ra, num_partitions, seed, it, rank):
                 cnt = np.searchsorted(ys, fcol, side="right") - a_lo
                 g2, off2 = _ragged_expand(cnt)
                 epos = eorder[a_lo[g2] + (np.arange(off2[-1]) - off2[g2])]
-                es = np.ones((Rp, Ep))
+                es = np.ones((Rp, Ep), dtype=np.float32)
                 es[grp[g2], epos] = fval[g2]
-                w = agree + (th * px)[:, None] * ia.index.sim_norms[y][None, :] * es
+                norms32 = ia.index.sim_norms[y].astype(np.float32)
+                w = agree + (th * px)[:, None] * norms32[None, :] * es
             if obs.all():
                 wprod *= w
             else:
                 wprod[obs] *= w[obs]
-        # inverse-CDF categorical per record (one uniform per record)
-        cum = np.cumsum(wprod, axis=1)
+        # inverse-CDF categorical per record (one uniform per record);
+        # f64 cumulative sum keeps the CDF monotone
+        cum = np.cumsum(wprod, axis=1, dtype=np.float64)
         target = u_rec[r0:r1] * cum[:, -1]
         sel = (cum < target[:, None]).sum(axis=1)
         new_rec_ent[r0:r1] = e0 + np.minimum(sel, Ep - 1)
