@@ -52,13 +52,27 @@ def sim_pairs(values, similarity_fn):
         n = len(values)
         return SimIndexCSR(np.zeros(n + 1, dtype=np.int64), np.empty(0, np.int32), np.empty(0))
 
+    # Edit distance must be over CHARACTERS (the reference uses JVM strings,
+    # SimilarityFn.scala:92-98), but the native kernels compare bytes. Encode
+    # through a per-domain character vocabulary: one byte per character, so
+    # byte-level Levenshtein == character-level. Domains with > 255 distinct
+    # characters (not seen in practice) fall back to the python oracle.
+    charset = sorted({ch for v in values for ch in v})
     if _C is not None and hasattr(_C, "sim_pairs_cpu"):
-        enc = [v.encode("utf-8", "surrogatepass") for v in values]
-        lens = np.array([len(e) for e in enc], dtype=np.int32)
+        lens_list = [len(v) for v in values]
+        if len(charset) <= 255:
+            vocab = {ch: i + 1 for i, ch in enumerate(charset)}
+            rows = [[vocab[ch] for ch in v] for v in values]
+        else:
+            # > 255 distinct characters: keep the native pass on UTF-8 bytes
+            # (a byte-level distance; avoids the quadratic python fallback)
+            rows = [list(v.encode("utf-8", "surrogatepass")) for v in values]
+            lens_list = [len(r) for r in rows]
+        lens = np.array(lens_list, dtype=np.int32)
         maxlen = int(lens.max()) if len(lens) else 0
-        buf = np.zeros((len(enc), max(maxlen, 1)), dtype=np.uint8)
-        for i, e in enumerate(enc):
-            buf[i, : len(e)] = np.frombuffer(e, dtype=np.uint8)
+        buf = np.zeros((len(values), max(maxlen, 1)), dtype=np.uint8)
+        for i, r in enumerate(rows):
+            buf[i, : len(r)] = r
         import torch
 
         thr = float(similarity_fn.threshold)

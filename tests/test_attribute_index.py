@@ -144,7 +144,7 @@ def test_native_pair_sweep_matches_python_randomized():
     if not ops.have_native():
         pytest.skip("native extension not built")
 
-    alphabet = "ABCDEF "
+    alphabet = "ABCDEF \u00dc\u00d6\u0141"  # incl. non-ASCII (char-level distance)
 
     @settings(max_examples=25, deadline=None)
     @given(
