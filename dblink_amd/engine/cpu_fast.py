@@ -1,16 +1,18 @@
-"""Vectorized CPU sweep for the PCG-I sampler (the flagship variant).
+"""Vectorized CPU sweeps for PCG-I (flagship), plain Gibbs and PCG-II.
 
 Mirrors the GPU engine's phase structure in numpy — whole-rank flat arrays,
 counter-based Philox draws keyed by (seed, iteration, phase, element id) —
 instead of the per-record python loops of ``cpu_engine.sweep_partition``.
-The per-record implementation remains the numerical oracle (and serves the
-other sampler variants); this path produces the same conditional
-distributions (checked by the cross-variant posterior-consistency and
-fast-vs-reference tests) but a different — still deterministic — draw
+The per-record implementation remains the numerical oracle (and serves
+Gibbs-Sequential plus oversized PCG-II partitions); this path produces the
+same conditional distributions (checked by the cross-variant
+posterior-consistency and fast-vs-reference band tests plus per-phase
+conditional TV tests) but a different — still deterministic — draw
 sequence.
 
-Parity: link update ``GibbsUpdates.scala:398-430, 473-530``; collapsed value
-update ``:576-599``; distortion update ``:324-359``.
+Parity: indexed link update ``GibbsUpdates.scala:398-430, 473-530``; dense
+collapsed link ``:363-395``; value updates ``:576-599`` (collapsed) /
+``:605-646`` (plain); distortion update ``:324-359``.
 """
 
 from __future__ import annotations
@@ -358,7 +360,6 @@ def _value_and_rest(state, fm, partitioner, num_partitions, rank, seed, it,
         nd_count = np.zeros((E, A), dtype=np.int64)
         np.add.at(nd_count, state.rec_ent, ndm.astype(np.int64))
         copy_val = np.zeros((E, A), dtype=rv.dtype)
-        rr_all = np.arange(R)
         for a in range(A):
             m = ndm[:, a]
             copy_val[state.rec_ent[m], a] = rv[m, a]
