@@ -165,8 +165,11 @@ def _link_dense_collapsed(state, fm, theta_ra, num_partitions, seed, it, rank):
             continue
         # linear-space weight product in f32 (matching the GPU kernels'
         # weight precision; the dense block is memory-bandwidth-bound), one
-        # log avoided entirely via the inverse-CDF draw below
-        wprod = np.ones((Rp, Ep), dtype=np.float32)
+        # log avoided entirely via the inverse-CDF draw below. Per-attr
+        # weights can be ~1e-8, so many attributes could underflow f32 —
+        # fall back to f64 beyond 8 attributes (product >= 1e-64).
+        wdt = np.float32 if A <= 8 else np.float64
+        wprod = np.ones((Rp, Ep), dtype=wdt)
         for a in range(A):
             ia = attrs[a]
             xo = rv[r0:r1, a]
@@ -174,9 +177,9 @@ def _link_dense_collapsed(state, fm, theta_ra, num_partitions, seed, it, rank):
             if not obs.any():
                 continue
             y = ev[e0:e1, a]
-            th = theta_ra[r0:r1, a].astype(np.float32)
-            px = ia.index.probs[np.maximum(xo, 0)].astype(np.float32)
-            agree = (xo[:, None] == y[None, :]) * (np.float32(1.0) - th)[:, None]
+            th = theta_ra[r0:r1, a].astype(wdt)
+            px = ia.index.probs[np.maximum(xo, 0)].astype(wdt)
+            agree = (xo[:, None] == y[None, :]) * (wdt(1.0) - th)[:, None]
             if ia.is_constant:
                 w = agree + (th * px)[:, None]
             else:
@@ -195,10 +198,10 @@ def _link_dense_collapsed(state, fm, theta_ra, num_partitions, seed, it, rank):
                 cnt = np.searchsorted(ys, fcol, side="right") - a_lo
                 g2, off2 = _ragged_expand(cnt)
                 epos = eorder[a_lo[g2] + (np.arange(off2[-1]) - off2[g2])]
-                es = np.ones((Rp, Ep), dtype=np.float32)
+                es = np.ones((Rp, Ep), dtype=wdt)
                 es[grp[g2], epos] = fval[g2]
-                norms32 = ia.index.sim_norms[y].astype(np.float32)
-                w = agree + (th * px)[:, None] * norms32[None, :] * es
+                norms_w = ia.index.sim_norms[y].astype(wdt)
+                w = agree + (th * px)[:, None] * norms_w[None, :] * es
             if obs.all():
                 wprod *= w
             else:
