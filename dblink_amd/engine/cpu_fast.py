@@ -273,15 +273,25 @@ def sweep_fast(state, cache, partitioner, num_partitions, rank=0, timers=None,
         + ev
     ).reshape(-1)
     order = np.argsort(keys)
-    skeys = keys[order]
     postings = (order // A).astype(np.int64)  # row-major [E, A] flatten
     rec_part = state.ent_part[state.rec_ent]
     qkeys = (
         (rec_part[:, None].astype(np.int64) * A + np.arange(A)[None, :]) * fm.Vmax
         + np.maximum(rv, 0)
     )
-    lo = np.searchsorted(skeys, qkeys.reshape(-1)).reshape(R, A)
-    hi = np.searchsorted(skeys, qkeys.reshape(-1), side="right").reshape(R, A)
+    nk = num_partitions * A * fm.Vmax
+    if nk <= (1 << 26):
+        # dense key prefix: candidate ranges by direct lookup (the GPU
+        # engine's counting-sort index, in numpy)
+        ptr = np.zeros(nk + 1, dtype=np.int64)
+        np.cumsum(np.bincount(keys, minlength=nk), out=ptr[1:])
+        qflat = qkeys.reshape(-1)
+        lo = ptr[qflat].reshape(R, A)
+        hi = ptr[qflat + 1].reshape(R, A)
+    else:
+        skeys = keys[order]
+        lo = np.searchsorted(skeys, qkeys.reshape(-1)).reshape(R, A)
+        hi = np.searchsorted(skeys, qkeys.reshape(-1), side="right").reshape(R, A)
     nd = obs & ~rdist
     sizes = np.where(nd, hi - lo, np.int64(1) << 60)
     base_a = np.argmin(sizes, axis=1)
