@@ -1,3 +1,6 @@
+"""CPU-vs-GPU end-to-end F1 comparison on one synthetic dataset
+(validation utility used during development; run on a GPU box)."""
+
 import sys, numpy as np
 sys.path.insert(0, "/root/repo")
 import torch

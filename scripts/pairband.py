@@ -1,3 +1,6 @@
+"""Linked-pair posterior band: CPU vs GPU (and the all-wave debug value
+path) on one dataset — the quick stationary-behavior cross-check."""
+
 import sys, numpy as np
 sys.path.insert(0, "/root/repo")
 import torch
