@@ -51,24 +51,22 @@ def migrate(state, world_size: int, device=None):
     # entities; receiver offsets by its own entity-block boundaries.
     pos_in_dest = inv_e - np.concatenate([[0], np.cumsum(send_counts_e)])[dest_e]
 
-    def a2a(arr_np, dtype):
+    def a2a(arr_np, counts_list):
         t = torch.from_numpy(np.ascontiguousarray(arr_np))
-        out, counts = comm.all_to_all_v(t, send_counts_list, device)
+        out, counts = comm.all_to_all_v(t, counts_list, device)
         return out.numpy(), counts
 
-    send_counts_list = [int(c) for c in send_counts_e]
-    ent_sorted_vals = state.ent_values[order_e]
-    ent_sorted_part = state.ent_part[order_e]
-    new_ent_values, recv_counts_e = a2a(ent_sorted_vals, np.int32)
-    new_ent_part, _ = a2a(ent_sorted_part, np.int32)
+    counts_e = [int(c) for c in send_counts_e]
+    new_ent_values, recv_counts_e = a2a(state.ent_values[order_e], counts_e)
+    new_ent_part, _ = a2a(state.ent_part[order_e], counts_e)
 
-    send_counts_list = [int(c) for c in send_counts_r]
+    counts_r = [int(c) for c in send_counts_r]
     rec_ent_local = pos_in_dest[state.rec_ent][order_r].astype(np.int64)
-    new_rec_entlocal, recv_counts_r = a2a(rec_ent_local, np.int64)
-    new_rec_values, _ = a2a(state.rec_values[order_r], np.int32)
-    new_rec_file, _ = a2a(state.rec_file[order_r], np.int32)
-    new_rec_dist, _ = a2a(state.rec_dist[order_r], np.uint8)
-    new_rec_gid, _ = a2a(state.rec_gid[order_r], np.int64)
+    new_rec_entlocal, recv_counts_r = a2a(rec_ent_local, counts_r)
+    new_rec_values, _ = a2a(state.rec_values[order_r], counts_r)
+    new_rec_file, _ = a2a(state.rec_file[order_r], counts_r)
+    new_rec_dist, _ = a2a(state.rec_dist[order_r], counts_r)
+    new_rec_gid, _ = a2a(state.rec_gid[order_r], counts_r)
 
     # rebuild rec_ent: received records from rank s refer to entity block s
     ent_block_offsets = np.concatenate([[0], np.cumsum(recv_counts_e)])
