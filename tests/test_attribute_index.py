@@ -183,3 +183,25 @@ def test_build_inverted_index_unit():
     np.testing.assert_array_equal(inv[(1, 5)], [0, 1, 3, 4])
     np.testing.assert_array_equal(inv[(1, 6)], [2])
     assert (0, 3) not in inv
+
+
+def test_native_pair_sweep_wide_charset_fallback():
+    """Domains with > 255 distinct characters take the UTF-8 byte pass —
+    it must still produce a structurally valid, symmetric index."""
+    from dblink_amd import ops
+
+    if not ops.have_native():
+        pytest.skip("native extension not built")
+    # 300 distinct CJK-range characters + a few similar Latin names
+    wide = ["".join(chr(0x4E00 + 7 * i + j) for j in range(3)) for i in range(300)]
+    values = sorted(set(wide) | {"ANNA", "ANNE", "BOB"})
+    fn = LevenshteinSimilarityFn(5.0, 10.0)
+    idx = ops.sim_pairs(values, fn)
+    V = len(values)
+    assert idx.row_ptr.shape == (V + 1,)
+    assert (np.diff(idx.row_ptr) >= 0).all()
+    ia = values.index("ANNA")
+    ib = values.index("ANNE")
+    # ANNA ~ ANNE must be mutually similar whatever the byte encoding
+    assert ib in idx.col[idx.row_ptr[ia]:idx.row_ptr[ia + 1]]
+    assert ia in idx.col[idx.row_ptr[ib]:idx.row_ptr[ib + 1]]
