@@ -264,3 +264,15 @@ def test_copy_files_step(tmp_path):
     assert open(os.path.join(dest, "diagnostics.csv")).read() == "iteration\n0\n"
     CopyFilesStep(project, ["diagnostics.csv"], dest, overwrite=True).execute()
     assert open(os.path.join(dest, "diagnostics.csv")).read() == "changed"
+
+
+def test_run_manifest_contents(tmp_path):
+    """run.txt carries the project manifest (Run.scala:38-42 /
+    Project.scala:58-96): data path, attributes with similarity functions
+    and priors, partitioner, output paths."""
+    project, out = run_project(tmp_path, n_records=60, samples=3, burnin=0,
+                               thin=1, cutoff=0)
+    txt = open(os.path.join(out, "run.txt")).read()
+    for needle in ("fname_c1", "LevenshteinSimilarityFn", "ConstantSimilarityFn",
+                   "KDTreePartitioner", "beta", "results"):
+        assert needle.lower() in txt.lower(), (needle, txt)
