@@ -8,3 +8,10 @@ migration and summary reductions.
 """
 
 __version__ = "0.1.0"
+
+
+def run_config(path):
+    """Execute a HOCON project config (the CLI's entry point, importable)."""
+    from .api.project import run_config as _run
+
+    return _run(path)

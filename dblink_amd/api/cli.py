@@ -18,6 +18,10 @@ def main(argv=None):
     )
     parser.add_argument("config", help="path to HOCON configuration file")
     parser.add_argument("-v", "--verbose", action="store_true")
+    from .. import __version__
+
+    parser.add_argument("--version", action="version",
+                        version=f"dblink_amd {__version__}")
     args = parser.parse_args(argv)
 
     logging.basicConfig(
