@@ -18,6 +18,9 @@ def main(argv=None):
     )
     parser.add_argument("config", help="path to HOCON configuration file")
     parser.add_argument("-v", "--verbose", action="store_true")
+    parser.add_argument("--check", action="store_true",
+                        help="validate the configuration (and that the data "
+                             "files and columns exist) without running")
     from .. import __version__
 
     parser.add_argument("--version", action="version",
@@ -28,6 +31,10 @@ def main(argv=None):
         level=logging.DEBUG if args.verbose else logging.INFO,
         format="%(asctime)s %(levelname)s %(name)s: %(message)s",
     )
+    if args.check:
+        from .project import check_config
+
+        return check_config(args.config)
     from .project import run_config
 
     run_config(args.config)

@@ -467,6 +467,36 @@ def parse_steps(config: hocon.Config, project: Project):
     return steps
 
 
+def check_config(path):
+    """Validate a project config without running: parse the HOCON, construct
+    the Project (attribute/partitioner/step validation) and load the data,
+    reporting problems instead of raising. Returns a process exit code."""
+    problems = []
+    try:
+        cfg = hocon.parse_file(path)
+        project = Project(cfg, rank=0, world_size=1)
+        steps = parse_steps(cfg, project)
+    except Exception as exc:  # configuration-level failure
+        print(f"INVALID: {exc}")
+        return 1
+    n = 0
+    try:
+        table = project.table
+        n = table.num_records
+        if n == 0:
+            problems.append("data loaded but contains zero records")
+    except Exception as exc:
+        problems.append(f"data loading failed: {exc}")
+    if problems:
+        for msg in problems:
+            print(f"INVALID: {msg}")
+        return 1
+    print(f"OK: {len(project.matching_attributes)} matching attributes, "
+          f"{n} records, {len(steps)} steps "
+          f"({', '.join(type(s).__name__ for s in steps)})")
+    return 0
+
+
 def run_config(config_path, rank=None, world_size=None, device=None):
     """CLI entry body (``Run.scala:27-50``)."""
     cfg = hocon.parse_file(config_path)

@@ -1,6 +1,7 @@
 """HOCON parser tests, including the exact shape of the reference's example
 configs (examples/RLdata500.conf structure)."""
 
+import os
 import textwrap
 
 import pytest
@@ -170,3 +171,43 @@ def test_hocon_parser_robustness_randomized():
         return json.dumps(x)
 
     check()
+
+
+def test_cli_check_mode(tmp_path):
+    """`python -m dblink_amd --check conf` validates without running."""
+    import subprocess
+    import sys
+
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    import os as _os
+
+    from dblink_amd.utils.synthdata import write_csv
+
+    data = str(tmp_path / "d.csv")
+    write_csv(data, 50, dup_fraction=0.1, seed=1)
+    conf = tmp_path / "p.conf"
+    conf.write_text(
+        """
+        dblink : {
+          data : { path : \"""" + data + """\", recordIdentifier : "rec_id",
+                   matchingAttributes : [
+                     {name : "fname_c1",
+                      similarityFunction : {name : "LevenshteinSimilarityFn",
+                        parameters : {threshold : 7.0, maxSimilarity : 10.0}},
+                      distortionPrior : {alpha : 0.5, beta : 50.0}} ] }
+          randomSeed : 1
+          partitioner : {name : "KDTreePartitioner",
+                         parameters : {numLevels : 0, matchingAttributes : []}}
+          outputPath : \"""" + str(tmp_path / "o") + """/\"
+          checkpointPath : \"""" + str(tmp_path / "c") + """/\"
+          steps : [{name : "summarize", parameters :
+                    {lowerIterationCutoff : 0, quantities : ["partition-sizes"]}}]
+        }
+        """
+    )
+    from dblink_amd.api.cli import main as cli_main
+
+    assert cli_main(["--check", str(conf)]) == 0
+    conf2 = tmp_path / "bad.conf"
+    conf2.write_text(conf.read_text().replace(data, str(tmp_path / "nope.csv")))
+    assert cli_main(["--check", str(conf2)]) == 1
