@@ -141,3 +141,31 @@ def test_ari_and_f1_invariances_randomized():
         assert to_pairwise_links(clusters) == to_pairwise_links(relabeled)
 
     check()
+
+
+def test_distortion_probs_unit():
+    """Prior-mean init, call semantics, out-of-range errors, Beta update
+    bounds (DistortionProbsTest.scala:24-46 analog)."""
+    import numpy as np
+
+    from dblink_amd.models.distortion import DistortionProbs, update_dist_probs
+    from dblink_amd.models.records import BetaShapeParameters
+
+    priors = [BetaShapeParameters(1.0, 9.0), BetaShapeParameters(0.5, 49.5)]
+    dp = DistortionProbs.from_prior_mean(priors, num_files=3)
+    assert dp(0, 0) == pytest.approx(0.1)
+    assert dp(1, 2) == pytest.approx(0.01)
+    with pytest.raises(IndexError):
+        dp(5, 0)
+    with pytest.raises(IndexError):
+        dp(0, 7)
+    rng = np.random.default_rng(0)
+    agg = np.array([[3, 0, 1], [0, 0, 0]], dtype=np.int64)
+    sizes = np.array([10, 10, 10], dtype=np.int64)
+    out = update_dist_probs(agg, priors, sizes, rng)
+    assert out.probs.shape == (2, 3)
+    assert ((out.probs > 0) & (out.probs < 1)).all()
+    # more observed distortions pull theta up
+    many = update_dist_probs(np.array([[9, 0, 0], [0, 0, 0]]), priors, sizes,
+                             np.random.default_rng(1))
+    assert many.probs[0, 0] > out.probs[0, 1]
