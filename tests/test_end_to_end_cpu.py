@@ -276,3 +276,34 @@ def test_run_manifest_contents(tmp_path):
     for needle in ("fname_c1", "LevenshteinSimilarityFn", "ConstantSimilarityFn",
                    "KDTreePartitioner", "beta", "results"):
         assert needle.lower() in txt.lower(), (needle, txt)
+
+
+def test_summarize_smpc_quantity(tmp_path):
+    """The summarize step's shared-most-probable-clusters quantity writes the
+    sMPC CSV without an evaluate step (ProjectSteps supportedSummaryQuantities)."""
+    data = str(tmp_path / "d.csv")
+    write_csv(data, 120, dup_fraction=0.1, seed=3)
+    out = str(tmp_path / "res")
+    conf = CONF_TEMPLATE.format(
+        data=data, out=out, samples=10, burnin=0, thin=1, cutoff=0,
+        sampler="PCG-I", levels=0, part_attrs="",
+    ).replace(
+        'quantities : ["cluster-size-distribution", "partition-sizes"]',
+        'quantities : ["shared-most-probable-clusters"]',
+    )
+    conf_path = tmp_path / "p.conf"
+    conf_path.write_text(conf)
+    cfg = hocon.parse_file(str(conf_path))
+    project = Project(cfg, rank=0, world_size=1)
+    os.makedirs(project.output_path, exist_ok=True)
+    steps = parse_steps(cfg, project)
+    for step in steps[:2]:  # sample + summarize only
+        step.execute()
+    smpc = os.path.join(out, "shared-most-probable-clusters.csv")
+    assert os.path.exists(smpc)
+    from dblink_amd.analysis.chain import read_clusters_csv
+
+    clusters = read_clusters_csv(smpc)
+    covered = set().union(*clusters) if clusters else set()
+    assert len(covered) == 120  # sMPC covers every record exactly once
+    assert sum(len(c) for c in clusters) == 120
