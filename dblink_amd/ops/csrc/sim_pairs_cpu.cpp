@@ -83,9 +83,13 @@ std::vector<torch::Tensor> sim_pairs_cpu(torch::Tensor strs, torch::Tensor lens,
         unit = 1.0 - 2.0 * (double)d / ((double)tot + d);
       }
       const double trans = scale * (max_sim * unit - threshold);
-      if (trans > 0.0) {
+      // inclusion criterion matches the python oracle exactly: exp(sim) > 1
+      // in f64 (knife-edge sims ~1e-17 round exp() to exactly 1.0 and are
+      // excluded on both sides; they would carry zero weight anyway)
+      const double es = std::exp(trans);
+      if (es > 1.0) {
         cols[i].push_back((int32_t)j);
-        sims[i].push_back((float)std::exp(trans));
+        sims[i].push_back((float)es);
       }
     }
   }
