@@ -125,3 +125,49 @@ def test_sampler_burnin_thinning_record_rule(tmp_path):
                      flags=SamplerFlags.for_sampler("PCG-I"))
     got2 = sorted(set(load_chain(out2)["iteration"].to_pylist()))
     assert got2 == [0, 2, 4, 6], got2
+
+
+def test_mpc_fast_equals_slow_randomized():
+    """Property test: the hash-aggregated MPC/sMPC equals the reference
+    per-cluster implementation on random synthetic chains."""
+    import pyarrow as pa
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    from dblink_amd.analysis.chain import (
+        most_probable_clusters,
+        most_probable_clusters_fast,
+        shared_most_probable_clusters,
+        shared_most_probable_clusters_fast,
+    )
+    from dblink_amd.engine.writers import LinkageChainWriter
+
+    def make_table(memberships):
+        rows = {"iteration": [], "partitionId": [], "linkageStructure": []}
+        for it, m in enumerate(memberships):
+            clusters = {}
+            for rid, c in enumerate(m):
+                clusters.setdefault(int(c), []).append(str(rid))
+            rows["iteration"].append(it)
+            rows["partitionId"].append(0)
+            rows["linkageStructure"].append(list(clusters.values()))
+        return pa.table(rows, schema=LinkageChainWriter.SCHEMA)
+
+    @settings(max_examples=30, deadline=None)
+    @given(st.integers(min_value=2, max_value=12),
+           st.integers(min_value=1, max_value=8),
+           st.integers(min_value=0, max_value=2**31))
+    def check(n, iters, seed):
+        rng = np.random.default_rng(seed)
+        memberships = [rng.integers(0, max(2, n // 2), n) for _ in range(iters)]
+        t = make_table(memberships)
+        slow = most_probable_clusters(t)
+        fast = most_probable_clusters_fast(t)
+        assert set(slow) == set(fast)
+        for rid in slow:
+            assert slow[rid][0] == fast[rid][0]
+            assert abs(slow[rid][1] - fast[rid][1]) < 1e-9
+        assert ({frozenset(c) for c in shared_most_probable_clusters(t)}
+                == {frozenset(c) for c in shared_most_probable_clusters_fast(t)})
+
+    check()
