@@ -205,3 +205,41 @@ def test_native_pair_sweep_wide_charset_fallback():
     # ANNA ~ ANNE must be mutually similar whatever the byte encoding
     assert ib in idx.col[idx.row_ptr[ia]:idx.row_ptr[ia + 1]]
     assert ia in idx.col[idx.row_ptr[ib]:idx.row_ptr[ib + 1]]
+
+
+def test_index_invariants_randomized():
+    """Property test: normalizers equal the brute-force 1/sum phi*expsim and
+    cached power distributions are exactly phi*norm^k normalized."""
+    import math
+
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    @settings(max_examples=25, deadline=None)
+    @given(
+        values=st.lists(st.text(alphabet="ABC", min_size=0, max_size=8),
+                        min_size=2, max_size=12, unique=True),
+        weights=st.lists(st.floats(min_value=0.1, max_value=50.0),
+                         min_size=2, max_size=12),
+        threshold=st.floats(min_value=0.5, max_value=8.0),
+    )
+    def check(values, weights, threshold):
+        k = min(len(values), len(weights))
+        values, weights = sorted(values)[:k], weights[:k]
+        if k < 2:
+            return
+        fn = LevenshteinSimilarityFn(threshold, 10.0)
+        idx = AttributeIndex(dict(zip(values, weights)), fn, precache_powers=3,
+                             pair_sweep=_python_sim_pairs)
+        V = idx.num_values
+        for v in range(V):
+            tot = sum(idx.probs[w] * math.exp(fn.similarity(values[w], values[v]))
+                      for w in range(V))
+            assert abs(idx.sim_norms[v] - 1.0 / tot) < 1e-9 / tot + 1e-12
+        for kpow in (1, 2, 3):
+            d = idx.sim_norm_dist(kpow)
+            w = idx.probs * idx.sim_norms ** kpow
+            np.testing.assert_allclose(d.probs, w / w.sum(), rtol=1e-9)
+            assert abs(idx.sim_norm_total(kpow) - w.sum()) < 1e-9 * w.sum()
+
+    check()
