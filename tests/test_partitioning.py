@@ -117,3 +117,47 @@ def test_simple_partitioner():
             assert len(set(pids[rows])) == 1
     counts = np.bincount(pids, minlength=4)
     assert counts.max() < 2.2 * counts.mean()
+
+
+def test_kd_partitioner_randomized():
+    """Property test: fitted KD partition ids stay in range (seen and unseen
+    values), and the flat export descends to the same leaf as the tree."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    @settings(max_examples=40, deadline=None)
+    @given(
+        n=st.integers(min_value=2, max_value=300),
+        a=st.integers(min_value=1, max_value=4),
+        levels=st.integers(min_value=0, max_value=4),
+        vmax=st.integers(min_value=2, max_value=40),
+        seed=st.integers(min_value=0, max_value=2**31),
+    )
+    def check(n, a, levels, vmax, seed):
+        rng = np.random.default_rng(seed)
+        vals = rng.integers(0, vmax, size=(n, a)).astype(np.int32)
+        p = KDTreePartitioner(levels, list(range(a)) * max(1, levels))
+        p.fit(vals)
+        pids = p.get_partition_ids(vals)
+        assert pids.min() >= 0 and pids.max() < p.num_partitions <= 2 ** levels
+        flat = p.as_flat()
+
+        def descend(row):
+            nid = 0
+            while flat["kind"][nid] != 0:
+                attr = flat["attr"][nid]
+                if flat["kind"][nid] == 1:
+                    go_right = row[attr] > flat["a"][nid]
+                else:
+                    lo, ln = flat["a"][nid], flat["b"][nid]
+                    go_right = row[attr] in set(flat["rset"][lo:lo + ln].tolist())
+                nid = 2 * nid + (2 if go_right else 1)
+            return flat["a"][nid]
+
+        for i in rng.integers(0, n, min(15, n)):
+            assert descend(vals[i]) == pids[i]
+        unseen = rng.integers(0, vmax + 5, size=(8, a)).astype(np.int32)
+        up = p.get_partition_ids(unseen)
+        assert up.min() >= 0 and up.max() < p.num_partitions
+
+    check()
