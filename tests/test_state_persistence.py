@@ -518,3 +518,51 @@ def test_fast_sweep_matches_reference_posterior_pcg2():
     ll_s, pairs_s = run(False)
     assert abs(ll_f - ll_s) / abs(ll_s) < 0.02, (ll_f, ll_s)
     assert abs(pairs_f - pairs_s) <= max(2.5, 0.4 * pairs_s), (pairs_f, pairs_s)
+
+
+def test_save_load_roundtrip_randomized():
+    """Property test: arbitrary states round-trip the two-file contract
+    verbatim (same-world load preserves order bitwise)."""
+    import tempfile
+
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    from dblink_amd.engine.state import SummaryVars
+    from dblink_amd.models.distortion import DistortionProbs
+
+    @settings(max_examples=20, deadline=None)
+    @given(
+        E=st.integers(min_value=1, max_value=40),
+        R=st.integers(min_value=1, max_value=60),
+        A=st.integers(min_value=1, max_value=5),
+        F=st.integers(min_value=1, max_value=3),
+        seed=st.integers(min_value=0, max_value=2**31),
+    )
+    def check(E, R, A, F, seed):
+        rng = np.random.default_rng(seed)
+        s = ChainState(
+            iteration=int(rng.integers(0, 1000)),
+            ent_values=rng.integers(0, 50, (E, A)).astype(np.int32),
+            ent_part=np.sort(rng.integers(0, 4, E)).astype(np.int32),
+            rec_values=rng.integers(-1, 50, (R, A)).astype(np.int32),
+            rec_file=rng.integers(0, F, R).astype(np.int32),
+            rec_ent=rng.integers(0, E, R).astype(np.int64),
+            rec_dist=rng.integers(0, 2, (R, A)).astype(np.uint8),
+            rec_gid=rng.permutation(R).astype(np.int64),
+            dist_probs=DistortionProbs(rng.random((A, F))),
+            population_size=E,
+            start_seed=seed, current_seed=seed + 7,
+            summary=SummaryVars(1, -2.5, np.zeros((A, F), np.int64),
+                                np.zeros(A + 1, np.int64)),
+        )
+        with tempfile.TemporaryDirectory() as d:
+            s.save(d)
+            t = ChainState.load(d)
+        for f in ("ent_values", "ent_part", "rec_values", "rec_file",
+                  "rec_ent", "rec_dist", "rec_gid"):
+            np.testing.assert_array_equal(getattr(s, f), getattr(t, f))
+        np.testing.assert_array_equal(s.dist_probs.probs, t.dist_probs.probs)
+        assert (s.iteration, s.current_seed) == (t.iteration, t.current_seed)
+
+    check()
