@@ -566,3 +566,44 @@ def test_save_load_roundtrip_randomized():
         assert (s.iteration, s.current_seed) == (t.iteration, t.current_seed)
 
     check()
+
+
+def test_chain_invariants_randomized():
+    """Property test: along random chains of all vectorized samplers —
+    record conservation, index ranges, finite likelihood, and the model
+    invariant that an observed disagreement implies a distortion flag."""
+    import os
+    import sys
+
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    from bench import build_cache_and_records
+
+    @settings(max_examples=10, deadline=None)
+    @given(
+        n=st.integers(min_value=10, max_value=100),
+        levels=st.integers(min_value=0, max_value=2),
+        sampler=st.sampled_from(["PCG-I", "Gibbs", "PCG-II"]),
+        seed=st.integers(min_value=0, max_value=10000),
+    )
+    def check(n, levels, sampler, seed):
+        cache, rv, rf = build_cache_and_records(n, seed=seed % 50)
+        part = KDTreePartitioner(levels, [3, 4][:max(1, levels)] if levels else [])
+        state = deterministic_init(rv, rf, np.arange(n, dtype=np.int64),
+                                   cache, part, seed=seed)
+        engine = CpuEngine(cache, part)
+        engine.initial_summary(state)
+        flags = SamplerFlags.for_sampler(sampler)
+        for _ in range(5):
+            engine.step(state, flags)
+            assert (np.sort(state.rec_gid) == np.arange(n)).all()
+            assert 0 <= state.rec_ent.min() and state.rec_ent.max() < state.num_entities
+            assert np.isfinite(state.summary.log_likelihood)
+            y = state.ent_values[state.rec_ent]
+            obs = state.rec_values >= 0
+            disagree = obs & (state.rec_values != y)
+            assert (state.rec_dist[disagree] == 1).all()
+
+    check()
