@@ -18,7 +18,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from dblink_amd.utils.synthdata import write_csv
 
 CONF = """dblink : {{
-    lowDistortion : {{alpha : 0.5, beta : 50.0}}
+    lowDistortion : {{alpha : {alpha}, beta : {beta}}}
 
     constSimFn : {{ name : "ConstantSimilarityFn" }}
     levSimFn : {{
@@ -89,9 +89,14 @@ def main():
               dup_fraction=0.1, seed=args.seed, num_files=args.files)
     part_attrs = '"fname_c1"' if args.levels > 0 else ""
     file_id_line = '\n        fileIdentifier : "file_id",' if args.files > 1 else ""
+    # distortion prior concentration scales with data size, mirroring the
+    # reference's example configs (RLdata500.conf Beta(0.5, 50) at n=500,
+    # RLdata10000.conf Beta(10, 1000)): a weak prior at large n lets the
+    # chain drift to a high-distortion mode
+    alpha, beta = (10.0, 1000.0) if args.records >= 5000 else (0.5, 50.0)
     conf = CONF.format(out=args.out, levels=args.levels, part_attrs=part_attrs,
                        samples=args.samples, burnin=args.burnin, thin=args.thin,
-                       file_id_line=file_id_line,
+                       file_id_line=file_id_line, alpha=alpha, beta=beta,
                        cutoff=args.burnin + (args.samples * args.thin) // 2)
     with open(os.path.join(args.out, "project.conf"), "w") as f:
         f.write(conf)
