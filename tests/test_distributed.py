@@ -43,7 +43,7 @@ assert np.all(state.ent_part % world == rank), "entity placed on wrong rank"
 
 engine = CpuEngine(cache, partitioner, world_size=world, rank=rank)
 engine.initial_summary(state)
-flags = SamplerFlags.for_sampler("PCG-I")
+flags = SamplerFlags.for_sampler(os.environ.get("DBLINK_TEST_SAMPLER", "PCG-I"))
 
 lls = []
 for i in range(25):
@@ -144,6 +144,21 @@ dist.destroy_process_group()
 
 
 @pytest.mark.slow
+@pytest.mark.slow
+@pytest.mark.parametrize("sampler", ["PCG-II", "Gibbs"])
+def test_two_rank_variants_conservation(sampler):
+    """The dense (PCG-II) and plain-Gibbs fast paths under real
+    multi-process migration: same conservation/ownership invariants."""
+    import os as _os
+
+    _os.environ["DBLINK_TEST_SAMPLER"] = sampler
+    try:
+        outs = _run_workers(WORKER.replace("__ROOT__", ROOT).replace("__LEVELS__", "2"))
+        assert any('"ok": true' in o for o in outs), outs
+    finally:
+        _os.environ.pop("DBLINK_TEST_SAMPLER", None)
+
+
 @pytest.mark.slow
 def test_four_rank_chain_conservation():
     """Four ranks over eight KD partitions: ownership, conservation and gid
