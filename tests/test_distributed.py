@@ -291,3 +291,64 @@ dist.destroy_process_group()
 def test_async_summary_reduce_overlaps_alltoall():
     outs = _run_workers(ASYNC_OVERLAP.replace("__ROOT__", ROOT))
     assert any('"ok": true' in o for o in outs), outs
+
+
+PROJECT_WORKER = r"""
+import sys
+sys.path.insert(0, "__ROOT__")
+from dblink_amd.api.cli import main
+raise SystemExit(main(["__CONF__"]))
+"""
+
+
+@pytest.mark.slow
+def test_multi_rank_project_resume(tmp_path):
+    """Full project flow at world 2 through the CLI twice (fresh + resume):
+    per-rank parquet parts, partition structure and record coverage survive
+    the resume (exercises the saved-partitioner restore at world > 1)."""
+    import sys as _sys
+
+    _sys.path.insert(0, ROOT)
+    from dblink_amd.analysis.chain import load_chain
+    from dblink_amd.utils.synthdata import write_csv
+
+    data = str(tmp_path / "d.csv")
+    write_csv(data, 300, dup_fraction=0.1, seed=4)
+    out = str(tmp_path / "res")
+    conf = tmp_path / "p.conf"
+    conf_text = """
+    dblink : {
+      lowDistortion : {alpha : 0.5, beta : 50.0}
+      data : { path : "%s", recordIdentifier : "rec_id",
+               entityIdentifier : "ent_id", nullValue : "NA",
+               matchingAttributes : [
+        {name : "by", similarityFunction : {name : "ConstantSimilarityFn"}, distortionPrior : ${dblink.lowDistortion}},
+        {name : "bm", similarityFunction : {name : "ConstantSimilarityFn"}, distortionPrior : ${dblink.lowDistortion}},
+        {name : "fname_c1", similarityFunction : {name : "LevenshteinSimilarityFn", parameters : {threshold : 7.0, maxSimilarity : 10.0}}, distortionPrior : ${dblink.lowDistortion}},
+        {name : "lname_c1", similarityFunction : {name : "LevenshteinSimilarityFn", parameters : {threshold : 7.0, maxSimilarity : 10.0}}, distortionPrior : ${dblink.lowDistortion}} ] }
+      randomSeed : 7
+      engine : "cpu"
+      partitioner : {name : "KDTreePartitioner",
+                     parameters : {numLevels : 2, matchingAttributes : ["fname_c1", "lname_c1"]}}
+      outputPath : "%s/"
+      checkpointPath : "%s/ckpt/"
+      steps : [{name : "sample", parameters : {
+        sampleSize : 6, burninInterval : 4, thinningInterval : 2,
+        resume : %s, sampler : "PCG-I", checkpointInterval : 0}}]
+    }
+    """
+    script = PROJECT_WORKER.replace("__ROOT__", ROOT).replace("__CONF__", str(conf))
+    conf.write_text(conf_text % (data, out, out, "false"))
+    _run_workers(script, world=2)
+    first = max(load_chain(out)["iteration"].to_pylist())
+    conf.write_text(conf_text % (data, out, out, "true"))
+    _run_workers(script, world=2)
+    t = load_chain(out)
+    assert max(t["iteration"].to_pylist()) > first
+    assert sorted(set(t["partitionId"].to_pylist())) == [0, 1, 2, 3]
+    import collections
+
+    per_iter = collections.Counter()
+    for row in t.to_pylist():
+        per_iter[row["iteration"]] += sum(len(c) for c in row["linkageStructure"])
+    assert set(per_iter.values()) == {300}
