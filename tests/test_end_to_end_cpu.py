@@ -307,3 +307,34 @@ def test_summarize_smpc_quantity(tmp_path):
     covered = set().union(*clusters) if clusters else set()
     assert len(covered) == 120  # sMPC covers every record exactly once
     assert sum(len(c) for c in clusters) == 120
+
+
+REFERENCE_RLDATA500 = "/root/reference/examples/RLdata500.csv"
+
+
+@pytest.mark.slow
+@pytest.mark.skipif(not os.path.exists(REFERENCE_RLDATA500),
+                    reason="reference RLdata500 not available")
+def test_real_rldata500_with_reference_config(tmp_path):
+    """THE parity benchmark: the reference's shipped RLdata500 dataset with
+    its RLdata500.conf VERBATIM (only file paths redirected). The d-blink
+    methodology reports ~0.9 pairwise F1 on this dataset; a correct
+    implementation must land there."""
+    src = open("/root/reference/examples/RLdata500.conf").read()
+    out = str(tmp_path / "rl500")
+    conf = (src
+            .replace("./examples/RLdata500.csv", REFERENCE_RLDATA500)
+            .replace("./examples/RLdata500_results/", out + "/"))
+    conf_path = tmp_path / "rl500.conf"
+    conf_path.write_text(conf)
+    cfg = hocon.parse_file(str(conf_path))
+    project = Project(cfg, rank=0, world_size=1)
+    os.makedirs(project.output_path, exist_ok=True)
+    for step in parse_steps(cfg, project):
+        step.execute()
+    txt = open(os.path.join(out, "evaluation-results.txt")).read()
+    f1 = float([l for l in txt.splitlines() if "F1-score" in l][0].split(":")[1])
+    ari = float([l for l in txt.splitlines() if "Rand" in l][0].split(":")[1])
+    # measured 0.907/0.907 (seed fixed by the config); generous floor
+    assert f1 > 0.8, txt
+    assert ari > 0.8, txt
