@@ -148,7 +148,14 @@ def get_fast_model(cache):
 def _link_dense_collapsed(state, fm, theta_ra, num_partitions, seed, it, rank):
     """PCG-II link update: distortions integrated out, dense over each
     partition's entities (GibbsUpdates.scala:363-395). Quadratic like the
-    reference; the caller size-gates it."""
+    reference; the caller size-gates it.
+
+    Works entirely in per-shape CACHED workspaces with in-place updates —
+    at R_p x E_p ~ 25M elements, per-sweep 100 MB temporaries previously
+    spent more time in the allocator (madvise churn) than in compute.
+    Weight products stay in f32 (matching the GPU kernels' precision; f64
+    beyond 8 attributes for underflow headroom); the agreement bonus is
+    applied by sparse scatter instead of a dense mask product."""
     attrs, A = fm.attrs, fm.A
     ev = state.ent_values
     rv = state.rec_values
@@ -157,19 +164,35 @@ def _link_dense_collapsed(state, fm, theta_ra, num_partitions, seed, it, rank):
     rec_ptr = np.searchsorted(rec_part, np.arange(num_partitions + 1))
     u_rec = _philox_dense(seed, it, _PH_LINK, state.num_records, rank)
     new_rec_ent = np.empty(state.num_records, dtype=np.int64)
+    wdt = np.float32 if A <= 8 else np.float64
+    ws = getattr(fm, "_dense_ws", None)
+    if ws is None:
+        ws = fm._dense_ws = {}
     for pid in range(num_partitions):
         r0, r1 = int(rec_ptr[pid]), int(rec_ptr[pid + 1])
         e0, e1 = int(ent_ptr[pid]), int(ent_ptr[pid + 1])
         Rp, Ep = r1 - r0, e1 - e0
         if Rp == 0:
             continue
-        # linear-space weight product in f32 (matching the GPU kernels'
-        # weight precision; the dense block is memory-bandwidth-bound), one
-        # log avoided entirely via the inverse-CDF draw below. Per-attr
-        # weights can be ~1e-8, so many attributes could underflow f32 —
-        # fall back to f64 beyond 8 attributes (product >= 1e-64).
-        wdt = np.float32 if A <= 8 else np.float64
-        wprod = np.ones((Rp, Ep), dtype=wdt)
+        # Two arithmetic styles by block size: plain numpy expressions give
+        # the allocator brk-reusable ~25 MB temporaries and run fastest for
+        # mid-size partitions, but at >= ~50 MB per temporary glibc switches
+        # to mmap/munmap per allocation and the sweep drowns in page-table
+        # churn — there the cached in-place workspace wins by an order of
+        # magnitude.
+        big = Rp * Ep > 8_000_000
+        if big:
+            key = (Rp, Ep)
+            if key not in ws:
+                ws[key] = (np.empty((Rp, Ep), dtype=wdt),   # weight product
+                           np.empty((Rp, Ep), dtype=wdt),   # per-attr work
+                           np.empty((Rp, Ep), dtype=wdt),   # agree scratch
+                           np.empty((Rp, Ep), dtype=bool),  # agreement mask
+                           np.empty((Rp, Ep), dtype=np.float64))  # CDF
+            wprod, work, scr, eq, cum = ws[key]
+            wprod.fill(wdt(1.0))
+        else:
+            wprod = np.ones((Rp, Ep), dtype=wdt)
         for a in range(A):
             ia = attrs[a]
             xo = rv[r0:r1, a]
@@ -179,10 +202,8 @@ def _link_dense_collapsed(state, fm, theta_ra, num_partitions, seed, it, rank):
             y = ev[e0:e1, a]
             th = theta_ra[r0:r1, a].astype(wdt)
             px = ia.index.probs[np.maximum(xo, 0)].astype(wdt)
-            agree = (xo[:, None] == y[None, :]) * (wdt(1.0) - th)[:, None]
-            if ia.is_constant:
-                w = agree + (th * px)[:, None]
-            else:
+            base_r = th * px
+            if not ia.is_constant:
                 si = ia.index.sim_index
                 # dense exp-sim block: ragged fill of each record's sim row
                 # onto the partition's entities, grouped by entity value
@@ -198,17 +219,36 @@ def _link_dense_collapsed(state, fm, theta_ra, num_partitions, seed, it, rank):
                 cnt = np.searchsorted(ys, fcol, side="right") - a_lo
                 g2, off2 = _ragged_expand(cnt)
                 epos = eorder[a_lo[g2] + (np.arange(off2[-1]) - off2[g2])]
-                es = np.ones((Rp, Ep), dtype=wdt)
-                es[grp[g2], epos] = fval[g2]
                 norms_w = ia.index.sim_norms[y].astype(wdt)
-                w = agree + (th * px)[:, None] * norms_w[None, :] * es
-            if obs.all():
-                wprod *= w
+            if big:
+                if ia.is_constant:
+                    np.copyto(work, base_r[:, None])
+                else:
+                    work.fill(wdt(1.0))
+                    work[grp[g2], epos] = fval[g2]
+                    work *= norms_w[None, :]
+                    work *= base_r[:, None]
+                np.equal(xo[:, None], y[None, :], out=eq)
+                np.multiply(eq, (wdt(1.0) - th)[:, None], out=scr)
+                work += scr
             else:
-                wprod[obs] *= w[obs]
+                agree = (xo[:, None] == y[None, :]) * (wdt(1.0) - th)[:, None]
+                if ia.is_constant:
+                    work = agree + base_r[:, None]
+                else:
+                    es = np.ones((Rp, Ep), dtype=wdt)
+                    es[grp[g2], epos] = fval[g2]
+                    work = agree + base_r[:, None] * norms_w[None, :] * es
+            if obs.all():
+                wprod *= work
+            else:
+                wprod[obs] *= work[obs]
         # inverse-CDF categorical per record (one uniform per record);
         # f64 cumulative sum keeps the CDF monotone
-        cum = np.cumsum(wprod, axis=1, dtype=np.float64)
+        if big:
+            np.cumsum(wprod, axis=1, out=cum)
+        else:
+            cum = np.cumsum(wprod, axis=1, dtype=np.float64)
         target = u_rec[r0:r1] * cum[:, -1]
         sel = (cum < target[:, None]).sum(axis=1)
         new_rec_ent[r0:r1] = e0 + np.minimum(sel, Ep - 1)
