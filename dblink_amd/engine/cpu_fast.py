@@ -29,6 +29,13 @@ _W0 = np.uint32(0x9E3779B9)
 _W1 = np.uint32(0xBB67AE85)
 
 
+# Dense PCG-II blocks larger than this switch from expression arithmetic
+# (fast while glibc serves the temporaries from brk) to a cached in-place
+# workspace (above ~50 MB per temporary glibc mmaps/munmaps every
+# allocation and page-table churn dominates the sweep).
+_DENSE_WS_THRESHOLD = 8_000_000
+
+
 def _philox_uniform4(seed, iteration, phase, ids, draw, rank=0):
     """Philox4x32-10 keyed uniforms in (0, 1): FOUR independent streams per
     id (the four 32-bit output words) for the price of one keyed evaluation.
@@ -180,7 +187,7 @@ def _link_dense_collapsed(state, fm, theta_ra, num_partitions, seed, it, rank):
         # to mmap/munmap per allocation and the sweep drowns in page-table
         # churn — there the cached in-place workspace wins by an order of
         # magnitude.
-        big = Rp * Ep > 8_000_000
+        big = Rp * Ep > _DENSE_WS_THRESHOLD
         if big:
             key = (Rp, Ep)
             if key not in ws:

@@ -520,6 +520,45 @@ def test_fast_sweep_matches_reference_posterior_pcg2():
     assert abs(pairs_f - pairs_s) <= max(2.5, 0.4 * pairs_s), (pairs_f, pairs_s)
 
 
+def test_pcg2_dense_workspace_path_matches_expression_path(monkeypatch):
+    """The dense PCG-II link update has two arithmetic paths selected by
+    block size (expression temporaries vs cached in-place workspace, see
+    cpu_fast._DENSE_WS_THRESHOLD). Force each path over the same chain and
+    check they sample the same posterior (the float32 op order differs
+    slightly, so parity is a band, not bitwise)."""
+    import os
+    import sys
+
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    from bench import build_cache_and_records
+
+    from dblink_amd.engine import cpu_fast
+
+    def run(threshold, iters=200, n=120, seed=5):
+        monkeypatch.setattr(cpu_fast, "_DENSE_WS_THRESHOLD", threshold)
+        cache, rv, rf = build_cache_and_records(n, seed=seed)
+        partitioner = KDTreePartitioner(1, [3])
+        state = deterministic_init(rv, rf, np.arange(n, dtype=np.int64),
+                                   cache, partitioner, seed=seed)
+        engine = CpuEngine(cache, partitioner)
+        engine.initial_summary(state)
+        flags = SamplerFlags.for_sampler("PCG-II")
+        lls, pairs = [], []
+        for i in range(iters):
+            engine.step(state, flags)
+            if i >= iters // 2:
+                lls.append(state.summary.log_likelihood)
+                c = np.bincount(state.rec_ent, minlength=state.num_entities)
+                pairs.append(int(np.sum(c * (c - 1) // 2)))
+        return float(np.mean(lls)), float(np.mean(pairs))
+
+    ll_big, pairs_big = run(0)            # every block takes the workspace path
+    ll_small, pairs_small = run(1 << 60)  # every block takes the expression path
+    assert abs(ll_big - ll_small) / abs(ll_small) < 0.02, (ll_big, ll_small)
+    assert abs(pairs_big - pairs_small) <= max(2.5, 0.4 * pairs_small), (
+        pairs_big, pairs_small)
+
+
 def test_save_load_roundtrip_randomized():
     """Property test: arbitrary states round-trip the two-file contract
     verbatim (same-world load preserves order bitwise)."""
