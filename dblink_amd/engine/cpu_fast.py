@@ -119,15 +119,45 @@ class _FastModel:
             None if ia.is_constant else np.log(ia.index.sim_norms) for ia in attrs
         ]
         self.sim_csr = []
+        # flat sorted (x * V + y) keys with log(expsim) values: pair lookups
+        # become one searchsorted over the nnz array (the scipy fancy-index
+        # path builds intermediate matrices and costs ~3x more)
+        self.sim_keys = []
+        self.sim_logvals = []
+        self.num_values = [ia.index.num_values for ia in attrs]
         for ia in attrs:
             if ia.is_constant:
                 self.sim_csr.append(None)
+                self.sim_keys.append(None)
+                self.sim_logvals.append(None)
             else:
                 si = ia.index.sim_index
                 V = ia.index.num_values
                 self.sim_csr.append(
                     sp.csr_matrix((si.expsim, si.col, si.row_ptr), shape=(V, V))
                 )
+                rows = np.repeat(np.arange(V, dtype=np.int64),
+                                 np.diff(si.row_ptr))
+                keys = rows * V + si.col.astype(np.int64)
+                order = np.argsort(keys)
+                self.sim_keys.append(keys[order])
+                self.sim_logvals.append(
+                    np.log(si.expsim.astype(np.float64))[order])
+        # dense [V, V] log link-weight tables, log_norms[y] folded in, for
+        # small domains (the common RLdata-like case): one gather replaces
+        # a searchsorted per (record, candidate) pair
+        self.sim_logdense = []
+        for ia in attrs:
+            if ia.is_constant or ia.index.num_values > 2048:
+                self.sim_logdense.append(None)
+            else:
+                si = ia.index.sim_index
+                V = ia.index.num_values
+                tbl = np.tile(np.log(ia.index.sim_norms), (V, 1))
+                rows = np.repeat(np.arange(V, dtype=np.int64),
+                                 np.diff(si.row_ptr))
+                tbl[rows, si.col] += np.log(si.expsim.astype(np.float64))
+                self.sim_logdense.append(tbl)
         self.phi_tables = [ia.index.distribution for ia in attrs]
         # cached power-dist normalizers Z_k, k = 0..kmax (Z_0 unused)
         self.kmax = max(1, min(16, max(
@@ -356,32 +386,46 @@ def sweep_fast(state, cache, partitioner, num_partitions, rank=0, timers=None,
         postings[np.minimum(base_lo[rec_of] + pos, len(postings) - 1)],
         base_lo[rec_of] + pos,
     )
-    ok = np.ones(len(cand), dtype=bool)
-    logw = np.zeros(len(cand))
+    n_cand = len(cand)
+    evc = ev.astype(np.int32)[cand]  # [n, A] candidate entity values
+    rvc = rv.astype(np.int32)[rec_of]
+    # non-distorted observed attributes (other than the base posting list)
+    # must agree exactly; the per-record mask is [R, A], gathered once
+    mask_ra = nd & (
+        (np.arange(A)[None, :] != base_a[:, None]) | ~has_nd[:, None]
+    )
+    ok = ~np.any(mask_ra[rec_of] & (evc != rvc), axis=1)
+    logw = np.zeros(n_cand)
+    odm = obs & rdist
     for a in range(A):
-        m = nd[rec_of, a] & ((a != base_a[rec_of]) | ~has_nd[rec_of])
-        if m.any():
-            ok[m] &= ev[cand[m], a] == rv[rec_of[m], a]
         if attrs[a].is_constant:
             continue
-        od_m = obs[rec_of, a] & rdist[rec_of, a]
+        od_m = odm[rec_of, a]
         if od_m.any():
-            y = ev[cand[od_m], a]
-            x = rv[rec_of[od_m], a]
-            es = np.asarray(fm.sim_csr[a][x, y]).ravel()
-            es = np.where(es > 0.0, es, 1.0)
-            logw[od_m] += fm.log_norms[a][y] + np.log(es)
+            y = evc[od_m, a]
+            x = rvc[od_m, a]
+            ld = fm.sim_logdense[a]
+            if ld is not None:
+                logw[od_m] += ld[x, y]
+            else:
+                qk = x.astype(np.int64) * fm.num_values[a] + y
+                fk = fm.sim_keys[a]
+                pos = np.minimum(np.searchsorted(fk, qk), len(fk) - 1)
+                hit = fk[pos] == qk
+                logw[od_m] += fm.log_norms[a][y] + np.where(
+                    hit, fm.sim_logvals[a][pos], 0.0)
     # Gumbel draws keyed by flat position: the candidate list order is a
     # deterministic function of the state, so position-keyed iid uniforms
     # give the same conditional distribution as entity-keyed ones
-    u = _philox_dense(seed, it, _PH_LINK, len(cand), rank)
+    u = _philox_dense(seed, it, _PH_LINK, n_cand, rank)
     score = np.where(ok, logw - np.log(-np.log(u)), -np.inf)
-    # segmented argmax: sort by (record, score) and take the last per record
-    perm = np.lexsort((score, rec_of))
-    last = off[1:] - 1
-    best_flat = perm[last]
-    if not np.isfinite(score[best_flat]).all():
+    # segmented argmax via two reduceats (max value, then first position
+    # attaining it; Gumbel ties have measure zero)
+    segmax = np.maximum.reduceat(score, off[:-1])
+    if not np.isfinite(segmax).all():
         raise RuntimeError("empty candidate set: state invariant violated")
+    at_max = np.where(score == segmax[rec_of], np.arange(n_cand), n_cand)
+    best_flat = np.minimum.reduceat(at_max, off[:-1])
     state.rec_ent = cand[best_flat].astype(np.int64)
     _mark("link")
 
