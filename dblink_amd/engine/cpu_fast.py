@@ -35,6 +35,11 @@ _W1 = np.uint32(0xBB67AE85)
 # allocation and page-table churn dominates the sweep).
 _DENSE_WS_THRESHOLD = 8_000_000
 
+# Attribute domains up to this size get dense [V, V] log link-weight tables
+# (see _FastModel); larger domains use searchsorted over flat sim keys.
+# The two paths are bitwise-identical (log_norms[y] + log(expsim) either way).
+_DENSE_LOGSIM_MAX_V = 2048
+
 
 def _philox_uniform4(seed, iteration, phase, ids, draw, rank=0):
     """Philox4x32-10 keyed uniforms in (0, 1): FOUR independent streams per
@@ -148,7 +153,7 @@ class _FastModel:
         # a searchsorted per (record, candidate) pair
         self.sim_logdense = []
         for ia in attrs:
-            if ia.is_constant or ia.index.num_values > 2048:
+            if ia.is_constant or ia.index.num_values > _DENSE_LOGSIM_MAX_V:
                 self.sim_logdense.append(None)
             else:
                 si = ia.index.sim_index

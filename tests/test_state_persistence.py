@@ -559,6 +559,42 @@ def test_pcg2_dense_workspace_path_matches_expression_path(monkeypatch):
         pairs_big, pairs_small)
 
 
+def test_link_logweight_dense_table_bitwise_matches_searchsorted(monkeypatch):
+    """The indexed link phase computes log weights either from dense [V, V]
+    tables (small domains) or searchsorted over flat sim keys; both compute
+    log_norms[y] + log(expsim(x, y)) in f64, so forcing either path over the
+    same chain must give bitwise-identical states."""
+    import copy
+    import os
+    import sys
+
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    from bench import build_cache_and_records
+
+    from dblink_amd.engine import cpu_fast
+
+    def run(max_v, iters=40, n=150, seed=9):
+        monkeypatch.setattr(cpu_fast, "_DENSE_LOGSIM_MAX_V", max_v)
+        cache, rv, rf = build_cache_and_records(n, seed=seed)
+        cache._fast_model = None  # rebuild under the patched gate
+        partitioner = KDTreePartitioner(1, [3])
+        state = deterministic_init(rv, rf, np.arange(n, dtype=np.int64),
+                                   cache, partitioner, seed=seed)
+        engine = CpuEngine(cache, partitioner)
+        engine.initial_summary(state)
+        flags = SamplerFlags.for_sampler("PCG-I")
+        for _ in range(iters):
+            engine.step(state, flags)
+        return state
+
+    s_dense = run(1 << 30)  # every attribute gets a dense table
+    s_flat = run(0)         # every attribute takes the searchsorted path
+    assert np.array_equal(s_dense.rec_ent, s_flat.rec_ent)
+    assert np.array_equal(s_dense.ent_values, s_flat.ent_values)
+    assert np.array_equal(s_dense.rec_dist, s_flat.rec_dist)
+    assert s_dense.summary.log_likelihood == s_flat.summary.log_likelihood
+
+
 def test_save_load_roundtrip_randomized():
     """Property test: arbitrary states round-trip the two-file contract
     verbatim (same-world load preserves order bitwise)."""
