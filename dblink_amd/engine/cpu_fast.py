@@ -114,8 +114,6 @@ class _FastModel:
     """Per-cache static arrays for the vectorized sweep (built once)."""
 
     def __init__(self, cache):
-        import scipy.sparse as sp
-
         attrs = cache.indexed_attributes
         self.attrs = attrs
         self.A = len(attrs)
@@ -123,7 +121,6 @@ class _FastModel:
         self.log_norms = [
             None if ia.is_constant else np.log(ia.index.sim_norms) for ia in attrs
         ]
-        self.sim_csr = []
         # flat sorted (x * V + y) keys with log(expsim) values: pair lookups
         # become one searchsorted over the nnz array (the scipy fancy-index
         # path builds intermediate matrices and costs ~3x more)
@@ -132,15 +129,11 @@ class _FastModel:
         self.num_values = [ia.index.num_values for ia in attrs]
         for ia in attrs:
             if ia.is_constant:
-                self.sim_csr.append(None)
                 self.sim_keys.append(None)
                 self.sim_logvals.append(None)
             else:
                 si = ia.index.sim_index
                 V = ia.index.num_values
-                self.sim_csr.append(
-                    sp.csr_matrix((si.expsim, si.col, si.row_ptr), shape=(V, V))
-                )
                 rows = np.repeat(np.arange(V, dtype=np.int64),
                                  np.diff(si.row_ptr))
                 keys = rows * V + si.col.astype(np.int64)
