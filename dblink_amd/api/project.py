@@ -250,12 +250,19 @@ class Project:
 class SampleStep:
     def __init__(self, project, sample_size, burnin_interval=0, thinning_interval=1,
                  resume=True, sampler="PCG-I", checkpoint_interval=20):
-        assert sample_size > 0 and burnin_interval >= 0 and thinning_interval >= 0
+        # matching Sampler.scala's require(): a non-positive thinning interval
+        # is a config error, not something to silently coerce
+        if sample_size <= 0:
+            raise ValueError("`sampleSize` must be positive.")
+        if burnin_interval < 0:
+            raise ValueError("`burninInterval` must be non-negative.")
+        if thinning_interval <= 0:
+            raise ValueError("`thinningInterval` must be positive.")
         assert sampler in SUPPORTED_SAMPLERS, f"sampler must be one of {SUPPORTED_SAMPLERS}"
         self.p = project
         self.sample_size = sample_size
         self.burnin_interval = burnin_interval
-        self.thinning_interval = max(thinning_interval, 1)
+        self.thinning_interval = thinning_interval
         self.resume = resume
         self.sampler = sampler
         self.checkpoint_interval = checkpoint_interval

@@ -698,10 +698,16 @@ class CpuEngine:
         cb = np.flatnonzero(np.r_[True, sorted_ent[1:] != sorted_ent[:-1]])
         cluster_offsets = np.r_[cb, len(sorted_ent)].astype(np.int64)
         cluster_pid = sorted_pid[cb] if len(cb) else np.empty(0, np.int32)
-        pb = np.flatnonzero(np.r_[True, cluster_pid[1:] != cluster_pid[:-1]])             if len(cluster_pid) else np.empty(0, np.int64)
-        pid_list = cluster_pid[pb] if len(pb) else np.empty(0, np.int32)
-        pid_offsets = np.r_[pb, len(cluster_pid)].astype(np.int64)
-        return pid_list.astype(np.int32), pid_offsets, cluster_offsets, gids
+        # every owned partition (one with entities) gets a row, even when all
+        # its entities are isolated — the reference's getLinkageStructure
+        # emits an empty cluster list per partition (State.scala:102-112)
+        owned = np.unique(state.ent_part)
+        counts = np.zeros(owned.size, dtype=np.int64)
+        if cluster_pid.size:
+            uniq, ccounts = np.unique(cluster_pid, return_counts=True)
+            counts[np.searchsorted(owned, uniq)] = ccounts
+        pid_offsets = np.r_[0, np.cumsum(counts)].astype(np.int64)
+        return owned.astype(np.int32), pid_offsets, cluster_offsets, gids
 
     def linkage_structure(self, state: ChainState, rec_id_of=None):
         """{pid -> list of clusters (lists of record-id strings)} for this rank

@@ -287,7 +287,54 @@ def _link_dense_collapsed(state, fm, theta_ra, num_partitions, seed, it, rank):
         target = u_rec[r0:r1] * cum[:, -1]
         sel = (cum < target[:, None]).sum(axis=1)
         new_rec_ent[r0:r1] = e0 + np.minimum(sel, Ep - 1)
+        # underflow guard: a record disagreeing with every entity across
+        # several rare-valued attributes can drive all f32 products to 0.0,
+        # which would silently select entity 0 — recompute those rows in
+        # log-space f64 (the true weights are strictly positive: every term
+        # has the theta*phi(x) floor)
+        bad = np.flatnonzero(~np.isfinite(cum[:, -1]) | (cum[:, -1] <= 0.0))
+        for i in bad:
+            lw = _dense_row_logweights(state, fm, theta_ra, r0 + i, e0, e1)
+            w = np.exp(lw - lw.max())
+            c = np.cumsum(w)
+            tgt = u_rec[r0 + i] * c[-1]
+            new_rec_ent[r0 + i] = e0 + min(int((c < tgt).sum()), Ep - 1)
     return new_rec_ent
+
+
+def _dense_row_logweights(state, fm, theta_ra, r, e0, e1):
+    """f64 log-weights of one record against its partition's entities
+    (the PCG-II product of GibbsUpdates.scala:370-393, in log space)."""
+    attrs = fm.attrs
+    ev = state.ent_values
+    rv = state.rec_values
+    Ep = e1 - e0
+    lw = np.zeros(Ep, dtype=np.float64)
+    for a, ia in enumerate(attrs):
+        x = int(rv[r, a])
+        if x < 0:
+            continue
+        y = ev[e0:e1, a]
+        th = float(theta_ra[r, a])
+        base = th * float(ia.index.probs[x])
+        if ia.is_constant:
+            w = base + (y == x) * (1.0 - th)
+        else:
+            si = ia.index.sim_index
+            es = np.ones(Ep, dtype=np.float64)
+            row = slice(int(si.row_ptr[x]), int(si.row_ptr[x + 1]))
+            cols, vals = si.col[row], si.expsim[row]
+            # scatter this record-value's sparse sim row onto matching entities
+            eorder = np.argsort(y, kind="stable")
+            ys = y[eorder]
+            lo = np.searchsorted(ys, cols)
+            cnt = np.searchsorted(ys, cols, side="right") - lo
+            grp, goff = _ragged_expand(cnt)
+            epos = eorder[lo[grp] + (np.arange(goff[-1]) - goff[grp])]
+            es[epos] = vals[grp]
+            w = base * ia.index.sim_norms[y] * es + (y == x) * (1.0 - th)
+        lw += np.log(w)
+    return lw
 
 
 def sweep_fast(state, cache, partitioner, num_partitions, rank=0, timers=None,

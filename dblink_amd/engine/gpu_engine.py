@@ -678,19 +678,22 @@ class GpuEngine(CpuEngine):
         cb_mask = torch.ones_like(sk, dtype=torch.bool)
         cb_mask[1:] = sk[1:] != sk[:-1]
         cb = torch.nonzero(cb_mask).squeeze(1)
-        cluster_pid = (sk[cb] // (E + 1)).to(torch.int32)
-        pb_mask = torch.ones_like(cluster_pid, dtype=torch.bool)
-        pb_mask[1:] = cluster_pid[1:] != cluster_pid[:-1]
-        pb = torch.nonzero(pb_mask).squeeze(1)
+        cluster_pid = (sk[cb] // (E + 1)).to(torch.int64)
+        # every owned partition gets a row (empty list when all its entities
+        # are isolated), matching the reference's getLinkageStructure
+        owned = torch.unique(gs.ent_part.to(torch.int64))
+        counts = torch.zeros(owned.numel(), dtype=torch.int64, device=self.device)
+        counts.scatter_add_(0, torch.searchsorted(owned, cluster_pid),
+                            torch.ones_like(cluster_pid))
+        pid_offsets = torch.zeros(owned.numel() + 1, dtype=torch.int64,
+                                  device=self.device)
+        torch.cumsum(counts, 0, out=pid_offsets[1:])
         cluster_offsets = torch.cat(
             [cb, torch.tensor([ne], device=self.device)]
         ).cpu().numpy()
-        pid_offsets = torch.cat(
-            [pb, torch.tensor([cluster_pid.numel()], device=self.device)]
-        ).cpu().numpy()
         return (
-            cluster_pid[pb].cpu().numpy(),
-            pid_offsets.astype(np.int64),
+            owned.to(torch.int32).cpu().numpy(),
+            pid_offsets.cpu().numpy().astype(np.int64),
             cluster_offsets.astype(np.int64),
             gids.cpu().numpy(),
         )

@@ -100,7 +100,7 @@ def sample(
             sample_ctr += 1
         if checkpoint_interval and completed % checkpoint_interval == 0:
             sync()
-            state.save(output_path, rank=rank,
+            state.save(output_path, rank=rank, world_size=engine.world_size,
                        extra={"partitioner": engine.partitioner})
     dt = time.time() - t0
     iters = state.iteration - initial_iteration
@@ -114,6 +114,6 @@ def sample(
             diagnostics_writer.close()
         comm.barrier()
         sync()
-        state.save(output_path, rank=rank,
+        state.save(output_path, rank=rank, world_size=engine.world_size,
                    extra={"partitioner": engine.partitioner})
     return state

@@ -47,6 +47,17 @@ class SummaryVars:
         )
 
 
+def _shard_rank(fname):
+    """Rank number of a partitions-state shard filename, or None."""
+    prefix, suffix = "partitions-state-rank", ".npz"
+    if fname.startswith(prefix) and fname.endswith(suffix):
+        try:
+            return int(fname[len(prefix):-len(suffix)])
+        except ValueError:
+            return None
+    return None
+
+
 @dataclass
 class ChainState:
     iteration: int
@@ -105,7 +116,7 @@ class ChainState:
 
     # ---- persistence (two-file contract, State.scala:122-193) ----------------
 
-    def save(self, output_path, rank=0, extra=None):
+    def save(self, output_path, rank=0, world_size=None, extra=None):
         os.makedirs(output_path, exist_ok=True)
         driver = {
             "iteration": self.iteration,
@@ -121,11 +132,20 @@ class ChainState:
             },
             "rng_state": self.rng_state,
         }
+        if world_size is not None:
+            driver["num_shards"] = int(world_size)
         if extra:
             driver.update(extra)
         if rank == 0:
             with open(os.path.join(output_path, "driver-state"), "wb") as f:
                 pickle.dump(driver, f)
+            if world_size is not None:
+                # a previous run in the same outputPath may have used a larger
+                # world size; its extra shard files must not leak into a resume
+                for fname in list(os.listdir(output_path)):
+                    r = _shard_rank(fname)
+                    if r is not None and r >= world_size:
+                        os.remove(os.path.join(output_path, fname))
         np.savez(
             os.path.join(output_path, f"partitions-state-rank{rank:05d}.npz"),
             ent_values=self.ent_values,
@@ -150,8 +170,18 @@ class ChainState:
         with open(os.path.join(output_path, "driver-state"), "rb") as f:
             driver = pickle.load(f)
         shards = sorted(
-            f for f in os.listdir(output_path) if f.startswith("partitions-state-rank")
+            f for f in os.listdir(output_path) if _shard_rank(f) is not None
         )
+        num_shards = driver.get("num_shards")
+        if num_shards is not None:
+            # only read the shards the saving run actually wrote; anything
+            # beyond is stale debris from an earlier, wider run
+            shards = [s for s in shards if _shard_rank(s) < num_shards]
+            if len(shards) != num_shards:
+                raise FileNotFoundError(
+                    f"saved state in {output_path} expects {num_shards} partition "
+                    f"shard(s) but found {len(shards)}"
+                )
         # Re-shard if world size changed: each rank takes every k-th shard and
         # re-sorts; partition ownership is re-established by the next migration.
         mine = [s for i, s in enumerate(shards) if i % world_size == rank]

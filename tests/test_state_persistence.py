@@ -682,3 +682,46 @@ def test_chain_invariants_randomized():
             assert (state.rec_dist[disagree] == 1).all()
 
     check()
+
+
+def test_save_prunes_stale_wider_run_shards(tmp_path):
+    """Resuming in an outputPath previously used by a WIDER run must not mix
+    stale shards in: save(world_size=N) records the shard count and deletes
+    rank files >= N; load honors the recorded count."""
+    state, engine = _make_state_and_engine(n=60, seed=3)
+    # simulate debris from an earlier 3-rank run
+    for r in range(3):
+        state.save(str(tmp_path), rank=r, world_size=3)
+    n_rec = state.num_records
+    # new 1-rank run saves into the same path
+    state.save(str(tmp_path), rank=0, world_size=1)
+    import os
+    shard_files = [f for f in os.listdir(str(tmp_path))
+                   if f.startswith("partitions-state-rank")]
+    assert shard_files == ["partitions-state-rank00000.npz"]
+    loaded = ChainState.load(str(tmp_path), rank=0, world_size=1)
+    assert loaded.num_records == n_rec
+
+
+def test_load_fails_on_missing_shard(tmp_path):
+    state, _ = _make_state_and_engine(n=40, seed=7)
+    state.save(str(tmp_path), rank=0, world_size=2)
+    with pytest.raises(FileNotFoundError):
+        ChainState.load(str(tmp_path), rank=0, world_size=2)
+
+
+def test_linkage_arrays_emits_empty_partitions():
+    """A partition whose entities are all isolated still gets a linkage-chain
+    row (empty cluster list), like the reference's getLinkageStructure."""
+    state, engine = _make_state_and_engine(n=60, seed=2)
+    # force every record into partition-0 entities; partition 1 keeps
+    # entities but no linked records
+    p0_entities = np.flatnonzero(state.ent_part == 0)
+    assert p0_entities.size > 0 and (state.ent_part == 1).any()
+    state.rec_ent = np.full(state.num_records, p0_entities[0], dtype=np.int64)
+    state.sort_by_partition()
+    pid_list, pid_offsets, cluster_offsets, gids = engine.linkage_arrays(state)
+    assert list(pid_list) == [0, 1]
+    # partition 1 has zero clusters
+    assert pid_offsets[2] - pid_offsets[1] == 0
+    assert pid_offsets[1] - pid_offsets[0] >= 1
