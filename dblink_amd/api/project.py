@@ -474,11 +474,53 @@ def parse_steps(config: hocon.Config, project: Project):
     return steps
 
 
+def native_limit_problems(project, table=None):
+    """Native-path (HIP kernel) limits, checked against the actual data.
+
+    Returns (errors, warnings). Errors are limits the GPU engine cannot run
+    past (kernels.hip MAX_ATTRS; migration distortion bitmask); warnings are
+    documented degradations (CPU fallback of the V x V similarity sweep above
+    64-byte values; byte-level edit distance above 255 distinct characters).
+    """
+    errors, warnings = [], []
+    A = len(project.matching_attributes)
+    if A > 16:
+        errors.append(
+            f"{A} matching attributes exceed the native-path limit of 16 "
+            "(kernels.hip MAX_ATTRS / migration distortion bitmask); "
+            "reduce the attribute count or run with dblink.engine = cpu"
+        )
+    if table is not None:
+        for ai, attr in enumerate(project.matching_attributes):
+            if attr.is_constant:
+                continue
+            vals = [v for v in set(table.columns[ai].tolist()) if v is not None]
+            charset = {ch for v in vals for ch in v}
+            maxlen = max((len(v) for v in vals), default=0)
+            if len(charset) > 255:
+                warnings.append(
+                    f"attribute {attr.name!r} has {len(charset)} distinct "
+                    "characters (> 255): the native similarity pass computes "
+                    "BYTE-level (UTF-8) edit distance there, which can differ "
+                    "from the reference's character-level distance "
+                    "(SimilarityFn.scala:92-98) for multi-byte characters"
+                )
+            elif maxlen > 64:
+                warnings.append(
+                    f"attribute {attr.name!r} has values up to {maxlen} "
+                    "characters (> 64): the V x V similarity sweep runs on "
+                    "the CPU (OpenMP) pass instead of the GPU kernel — "
+                    "identical results, slower index build"
+                )
+    return errors, warnings
+
+
 def check_config(path):
     """Validate a project config without running: parse the HOCON, construct
     the Project (attribute/partitioner/step validation) and load the data,
     reporting problems instead of raising. Returns a process exit code."""
     problems = []
+    warnings = []
     try:
         cfg = hocon.parse_file(path)
         project = Project(cfg, rank=0, world_size=1)
@@ -487,6 +529,7 @@ def check_config(path):
         print(f"INVALID: {exc}")
         return 1
     n = 0
+    table = None
     try:
         table = project.table
         n = table.num_records
@@ -494,6 +537,11 @@ def check_config(path):
             problems.append("data loaded but contains zero records")
     except Exception as exc:
         problems.append(f"data loading failed: {exc}")
+    errs, warns = native_limit_problems(project, table)
+    problems.extend(errs)
+    warnings.extend(warns)
+    for msg in warnings:
+        print(f"WARNING: {msg}")
     if problems:
         for msg in problems:
             print(f"INVALID: {msg}")
@@ -518,6 +566,8 @@ def run_config(config_path, rank=None, world_size=None, device=None):
     if rank == 0:
         lines = ["Scheduled steps", "---------------"] + ["  * " + s.mk_string() for s in steps]
         log.info("\n".join(lines))
+        for msg in native_limit_problems(project, project.table)[1]:
+            log.warning(msg)
     for step in steps:
         step.execute()
     return project

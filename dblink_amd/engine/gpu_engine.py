@@ -48,6 +48,11 @@ class GpuModel:
         self.device = device
         attrs = cache.indexed_attributes
         A = len(attrs)
+        if A > 16:
+            raise ValueError(
+                f"{A} matching attributes exceed the GPU engine's limit of 16 "
+                "(kernels.hip MAX_ATTRS); use dblink.engine = cpu"
+            )
         self.A = A
         self.F = cache.num_files
         self.Kc = int(max_cluster_size)

@@ -66,6 +66,14 @@ def sim_pairs(values, similarity_fn):
         else:
             # > 255 distinct characters: keep the native pass on UTF-8 bytes
             # (a byte-level distance; avoids the quadratic python fallback)
+            import logging
+
+            logging.getLogger("dblink_amd.ops").warning(
+                "attribute domain has %d distinct characters (> 255): computing "
+                "BYTE-level (UTF-8) edit distance, which can differ from the "
+                "reference's character-level distance for multi-byte characters",
+                len(charset),
+            )
             rows = [list(v.encode("utf-8", "surrogatepass")) for v in values]
             lens_list = [len(r) for r in rows]
         lens = np.array(lens_list, dtype=np.int32)
