@@ -291,6 +291,41 @@ class GpuEngine(CpuEngine):
         self._gs_alt = None
         self._phase_events = []  # (name, start_event, end_event)
         self._phase_totals = {}
+        self._hist_ready = False
+        self._summary_host = None
+        # overlapped migration (migrants-only async all-to-all with the
+        # posting build and summary readback hidden under it) is the default
+        # multi-rank path; DBLINK_OVERLAP=0 selects the eager reference path
+        self._overlap = os.environ.get("DBLINK_OVERLAP", "1") != "0"
+
+    def _ensure_idx_buffers(self, zero=False):
+        """Size (and optionally zero) the dense counting-sort buffers for the
+        inverted index; shared by the sweep body and the overlapped-migration
+        histogram prebuild."""
+        T = self.model.A + self._num_pairs
+        vmax = max(self.model.Vmax, self._pair_vmax)
+        nk = self.num_partitions * T * vmax
+        if self._idx_nk != nk:
+            self._idx_nk = nk
+            self._idx_counts = torch.zeros(nk, dtype=torch.int32, device=self.device)
+            self._idx_ptr = torch.zeros(nk + 1, dtype=torch.int64, device=self.device)
+            self._idx_cursor = torch.empty(nk, dtype=torch.int32, device=self.device)
+        elif zero:
+            self._idx_counts.zero_()
+        return nk
+
+    def _hist_add(self, ent_part, ent_values, reset=False):
+        """Accumulate (partition, slot, value) posting counts for a subset of
+        entities — the overlapped-migration prebuild of the next sweep's
+        inverted index (atomic adds, so stay-home and arrival contributions
+        compose in any order)."""
+        self._ensure_idx_buffers(zero=reset)
+        vmax = max(self.model.Vmax, self._pair_vmax)
+        self.C.postings_hist(ent_part, ent_values, self._pair_a1,
+                             self._pair_a2, self._pair_v2, vmax,
+                             self._idx_counts)
+        if not reset:
+            self._hist_ready = True
 
     # ---- state residency -----------------------------------------------------
 
@@ -365,16 +400,37 @@ class GpuEngine(CpuEngine):
         else:
             self._sweep_body(gs, flags, graph_safe=False)
             if comm.is_distributed():
-                from ..parallel.migration import migrate_and_sort_tensors
+                from ..parallel.migration import (
+                    migrate_and_sort_tensors,
+                    migrate_overlapped,
+                )
 
                 # summary stats are partition-agnostic: pack them from the
                 # pre-migration state and overlap the small all-reduce (on
                 # its own communicator) with the migration all-to-all
                 self._pack_summary(gs, loglik_done=True)
                 work = comm.all_reduce_sum_async(self._packed)
-                migrate_and_sort_tensors(gs, self.world_size)
-                if work is not None:
-                    work.wait()
+                if self._overlap:
+                    hist_add = None
+                    if (not flags.sequential and not flags.collapsed_entity_ids
+                            and self._ensure_idx_buffers() <= (1 << 28)):
+                        hist_add = self._hist_add
+
+                    def during_flight():
+                        # payloads in the air: land the summary reduction and
+                        # stage the host readback that feeds the next theta
+                        # draw (the one required host sync of the loop)
+                        if work is not None:
+                            work.wait()
+                        self._summary_host = self._packed.cpu().numpy().copy()
+
+                    migrate_overlapped(gs, self.world_size, self.rank,
+                                       during_flight=during_flight,
+                                       hist_add=hist_add)
+                else:
+                    migrate_and_sort_tensors(gs, self.world_size)
+                    if work is not None:
+                        work.wait()
                 self._summary_reduced = True
 
         state.current_seed += self.num_partitions
@@ -422,6 +478,11 @@ class GpuEngine(CpuEngine):
         self._ctrl.copy_(self._ctrl_pin, non_blocking=True)
         ctrl = self._ctrl
         seed, it = 0, 0  # kernels read the ctrl buffer
+        # posting histogram possibly pre-built during the previous sweep's
+        # migration overlap; consume the flag unconditionally (a sampler
+        # switch must not leave stale counts armed)
+        hist_pre = self._hist_ready
+        self._hist_ready = False
 
         # --- inverted index (counting sort over (partition, slot, value)) ----
         # Posting order within a key is arbitrary: the link kernels only
@@ -433,16 +494,11 @@ class GpuEngine(CpuEngine):
             vmax = max(m.Vmax, self._pair_vmax)
             nk = self.num_partitions * T * vmax
             if nk <= (1 << 28):  # dense counters (1 GiB cap; always true in practice)
-                if self._idx_nk != nk:
-                    self._idx_nk = nk
-                    self._idx_counts = torch.zeros(nk, dtype=torch.int32, device=dev)
-                    self._idx_ptr = torch.zeros(nk + 1, dtype=torch.int64, device=dev)
-                    self._idx_cursor = torch.empty(nk, dtype=torch.int32, device=dev)
-                else:
-                    self._idx_counts.zero_()
-                self.C.postings_hist(gs.ent_part, gs.ent_values, self._pair_a1,
-                                     self._pair_a2, self._pair_v2, vmax,
-                                     self._idx_counts)
+                if not hist_pre:
+                    self._ensure_idx_buffers(zero=True)
+                    self.C.postings_hist(gs.ent_part, gs.ent_values, self._pair_a1,
+                                         self._pair_a2, self._pair_v2, vmax,
+                                         self._idx_counts)
                 torch.cumsum(self._idx_counts, 0, dtype=torch.int64,
                              out=self._idx_ptr[1:])
                 self._idx_cursor.copy_(self._idx_ptr[:-1])
@@ -644,7 +700,11 @@ class GpuEngine(CpuEngine):
             packed = packed.clone()
             comm.all_reduce_sum_(packed)
         self._summary_reduced = False
-        host = packed.cpu().numpy()
+        if self._summary_host is not None:
+            # readback already staged during the migration overlap window
+            host, self._summary_host = self._summary_host, None
+        else:
+            host = packed.cpu().numpy()
         err = int(self._err.cpu())
         if err:
             raise RuntimeError(

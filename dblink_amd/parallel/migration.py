@@ -87,6 +87,183 @@ def migrate(state, world_size: int, device=None):
     return state
 
 
+def migrate_overlapped(gs, world_size: int, rank: int, during_flight=None,
+                       hist_add=None):
+    """Overlapped, migrant-only migration (the per-sweep RCCL all-to-all over
+    xGMI): only entities whose destination rank changed —
+    with their records — travel; the counts exchange and both payload
+    exchanges run async on the communicator's stream while the compute
+    stream packs payloads, sorts the stay-home block, and (via ``hist_add``)
+    pre-builds the NEXT sweep's posting histogram; ``during_flight`` runs on
+    the host while payloads are in the air (the engine uses it for the
+    summary readback that feeds the next theta draw).
+
+    Bitwise-identical post-state to ``migrate_and_sort_tensors``: the merged
+    pre-sort entity order is [arrivals from ranks < rank | stay-home |
+    arrivals from ranks > rank] (what the dense all-to-all would produce),
+    followed by the same stable sort by partition id.
+
+    Replaces the reference's Spark shuffle (GibbsUpdates.scala:144-150).
+    """
+    import torch
+    import torch.distributed as dist
+
+    device = gs.ent_part.device
+    A = gs.ent_values.shape[1]
+    assert A <= 16, "distortion bitmask packing supports at most 16 attributes"
+    if world_size <= 1 or not comm.is_distributed():
+        return migrate_and_sort_tensors(gs, world_size)
+
+    E, R = gs.ent_values.shape[0], gs.rec_values.shape[0]
+    dest_e = gs.ent_part.to(torch.int64) % world_size
+    mig_mask = dest_e != rank
+    mig_idx = torch.nonzero(mig_mask).squeeze(1)       # original order
+    stay_idx = torch.nonzero(~mig_mask).squeeze(1)
+    mdest = dest_e[mig_idx]
+    morder = torch.argsort(mdest, stable=True)
+    mig_sorted = mig_idx[morder]                       # grouped by dest
+    send_e = torch.bincount(mdest, minlength=world_size)
+
+    dest_r = dest_e[gs.rec_ent]
+    rmig_idx = torch.nonzero(dest_r != rank).squeeze(1)
+    send_r = torch.bincount(dest_r[rmig_idx], minlength=world_size)
+
+    # ---- async counts exchange (equal splits, stays on the comm device) ----
+    comm_dev = comm._comm_device() or device
+    sc = torch.stack([send_e, send_r], dim=1).reshape(-1).to(comm_dev)
+    rc = torch.empty_like(sc)
+    counts_work = dist.all_to_all_single(rc, sc, async_op=True)
+
+    # ---- pack migrant payloads on the compute stream ------------------------
+    nrec = torch.zeros(E, dtype=torch.int64, device=device)
+    nrec.scatter_add_(0, gs.rec_ent, torch.ones_like(gs.rec_ent))
+
+    n_mig = mig_sorted.numel()
+    ent_pack = torch.empty((n_mig, A + 2), dtype=torch.int32, device=device)
+    ent_pack[:, :A] = gs.ent_values[mig_sorted]
+    ent_pack[:, A] = gs.ent_part[mig_sorted]
+    ent_pack[:, A + 1] = nrec[mig_sorted].to(torch.int32)
+
+    # migrant records sorted by (dest, entity position within dest block) so
+    # receivers can rebuild rec_ent from the per-entity record counts
+    base = torch.cumsum(
+        torch.cat([torch.zeros(1, dtype=torch.int64, device=device), send_e[:-1]]), 0
+    )
+    pos_of_ent = torch.empty(E, dtype=torch.int64, device=device)
+    pos_of_ent[mig_sorted] = (
+        torch.arange(n_mig, device=device, dtype=torch.int64) - base[mdest[morder]]
+    )
+    rkey = dest_r[rmig_idx] * max(E, 1) + pos_of_ent[gs.rec_ent[rmig_idx]]
+    rmig_sorted = rmig_idx[torch.argsort(rkey, stable=True)]
+
+    n_rmig = rmig_sorted.numel()
+    weights = 1 << torch.arange(A, device=device, dtype=torch.int32)
+    rec_pack = torch.empty((n_rmig, A + 4), dtype=torch.int32, device=device)
+    rec_pack[:, :A] = gs.rec_values[rmig_sorted]
+    rec_pack[:, A] = gs.rec_file[rmig_sorted]
+    rec_pack[:, A + 1] = (
+        gs.rec_dist[rmig_sorted].to(torch.int32) * weights.view(1, A)
+    ).sum(dim=1)
+    gid = gs.rec_gid[rmig_sorted]
+    rec_pack[:, A + 2] = (gid & 0xFFFFFFFF).to(torch.int32)
+    rec_pack[:, A + 3] = (gid >> 32).to(torch.int32)
+
+    # stay-home block (original relative order, like the dense path's self
+    # block); posting histogram for the next sweep can start right away
+    stay_vals = gs.ent_values[stay_idx].contiguous()
+    stay_part = gs.ent_part[stay_idx].contiguous()
+    if hist_add is not None:
+        hist_add(stay_part, stay_vals, reset=True)
+
+    # ---- counts land: allocate and launch the payload exchanges -------------
+    counts_work.wait()
+    both = torch.stack([sc, rc]).cpu()  # ONE small D2H for all count lists
+    send_el = [int(x) for x in both[0, 0::2]]
+    send_rl = [int(x) for x in both[0, 1::2]]
+    recv_el = [int(x) for x in both[1, 0::2]]
+    recv_rl = [int(x) for x in both[1, 1::2]]
+
+    ep_src, ep_home = comm._to_comm(ent_pack)
+    rp_src, _ = comm._to_comm(rec_pack)
+    arr_ep = torch.empty((sum(recv_el), A + 2), dtype=torch.int32, device=ep_src.device)
+    arr_rp = torch.empty((sum(recv_rl), A + 4), dtype=torch.int32, device=rp_src.device)
+    work_e = dist.all_to_all_single(arr_ep, ep_src, recv_el, send_el, async_op=True)
+    work_r = dist.all_to_all_single(arr_rp, rp_src, recv_rl, send_rl, async_op=True)
+
+    if during_flight is not None:
+        during_flight()
+
+    work_e.wait()
+    work_r.wait()
+    if ep_home is not None:  # gloo bridging for CUDA-resident tests
+        arr_ep = arr_ep.to(ep_home)
+        arr_rp = arr_rp.to(ep_home)
+
+    # ---- merge: [arrivals < rank | stay | arrivals > rank], stable sort -----
+    n_arr = arr_ep.shape[0]
+    n_stay = stay_idx.numel()
+    arr_before = int(sum(recv_el[:rank]))  # arrivals from lower ranks
+    arr_vals = arr_ep[:, :A]
+    arr_part = arr_ep[:, A]
+    arr_nrec = arr_ep[:, A + 1].to(torch.int64)
+    if hist_add is not None:
+        hist_add(arr_part.contiguous(), arr_vals.contiguous(), reset=False)
+
+    E_new = n_stay + n_arr
+    merged_vals = torch.empty((E_new, A), dtype=torch.int32, device=device)
+    merged_part = torch.empty(E_new, dtype=torch.int32, device=device)
+    merged_vals[:arr_before] = arr_vals[:arr_before]
+    merged_part[:arr_before] = arr_part[:arr_before]
+    merged_vals[arr_before : arr_before + n_stay] = stay_vals
+    merged_part[arr_before : arr_before + n_stay] = stay_part
+    merged_vals[arr_before + n_stay :] = arr_vals[arr_before:]
+    merged_part[arr_before + n_stay :] = arr_part[arr_before:]
+
+    order = torch.argsort(merged_part.to(torch.int64), stable=True)
+    inv = torch.empty_like(order)
+    inv[order] = torch.arange(E_new, device=device)
+    gs.ent_values = merged_vals[order].contiguous()
+    gs.ent_part = merged_part[order].contiguous()
+
+    # ---- records: concat [stay | arrivals], final entity ids, stable sort ---
+    stay_r = torch.nonzero(dest_r == rank).squeeze(1)
+    # stay entity -> merged pre-sort index
+    stay_pos = torch.empty(E, dtype=torch.int64, device=device)
+    stay_pos[stay_idx] = torch.arange(n_stay, device=device, dtype=torch.int64)
+    p_stay = arr_before + stay_pos[gs.rec_ent[stay_r]]
+    # arrival entity k (global, source-ascending) -> merged pre-sort index
+    k = torch.arange(n_arr, device=device, dtype=torch.int64)
+    p_of_arr_ent = torch.where(k < arr_before, k, k + n_stay)
+    p_arr = torch.repeat_interleave(p_of_arr_ent, arr_nrec)
+
+    new_rec_ent = inv[torch.cat([p_stay, p_arr])]
+    rorder = torch.argsort(new_rec_ent, stable=True)
+    n_rstay = stay_r.numel()
+    ro_stay = rorder < n_rstay
+
+    def merge_rec(stay_col, arr_col, dtype):
+        out = torch.empty((rorder.numel(),) + tuple(stay_col.shape[1:]),
+                          dtype=dtype, device=device)
+        out[ro_stay] = stay_col[rorder[ro_stay]].to(dtype)
+        out[~ro_stay] = arr_col[rorder[~ro_stay] - n_rstay].to(dtype)
+        return out.contiguous()
+
+    db = arr_rp[:, A + 1]
+    arr_dist = (
+        (db.view(-1, 1) >> torch.arange(A, device=device, dtype=torch.int32).view(1, A)) & 1
+    ).to(torch.uint8)
+    arr_gid = (arr_rp[:, A + 2].to(torch.int64) & 0xFFFFFFFF) | (
+        arr_rp[:, A + 3].to(torch.int64) << 32
+    )
+    gs.rec_ent = new_rec_ent[rorder].contiguous()
+    gs.rec_values = merge_rec(gs.rec_values[stay_r], arr_rp[:, :A], torch.int32)
+    gs.rec_file = merge_rec(gs.rec_file[stay_r], arr_rp[:, A], torch.int32)
+    gs.rec_dist = merge_rec(gs.rec_dist[stay_r], arr_dist, torch.uint8)
+    gs.rec_gid = merge_rec(gs.rec_gid[stay_r], arr_gid, torch.int64)
+    gs.rec_part = gs.ent_part[gs.rec_ent].contiguous()
+    return gs
+
+
 def migrate_and_sort_tensors(gs, world_size: int):
     """Tensor-resident variant of ``migrate`` used by the GPU engine.
 

@@ -542,7 +542,10 @@ lo, hi = int(bounds[rank]), int(bounds[rank + 1])
 state = deterministic_init(rec_values[lo:hi], rec_files[lo:hi],
                            np.arange(lo, hi, dtype=np.int64), cache, partitioner,
                            seed=3, rank=rank, world_size=world)
+import copy
+state_eager = copy.deepcopy(state)
 engine = GpuEngine(cache, partitioner, world_size=world, rank=rank, device=device)
+assert engine._overlap, "overlapped migration must be the default"
 engine.initial_summary(state)
 flags = SamplerFlags.for_sampler("PCG-I")
 for i in range(12):
@@ -550,6 +553,17 @@ for i in range(12):
 engine.sync_state(state)
 assert np.all(state.ent_part % world == rank), "ownership violated"
 assert np.isfinite(state.summary.log_likelihood)
+
+# posterior parity: the overlapped path must replay the eager path bitwise
+engine2 = GpuEngine(cache, partitioner, world_size=world, rank=rank, device=device)
+engine2._overlap = False
+engine2.initial_summary(state_eager)
+for i in range(12):
+    engine2.step(state_eager, flags)
+engine2.sync_state(state_eager)
+for f in ("ent_values", "ent_part", "rec_values", "rec_ent", "rec_dist", "rec_gid"):
+    assert np.array_equal(getattr(state, f), getattr(state_eager, f)), f
+assert state.summary.log_likelihood == state_eager.summary.log_likelihood
 gids = comm.all_gather_object(sorted(state.rec_gid.tolist()))
 t = torch.tensor([state.num_entities], dtype=torch.float64)
 comm.all_reduce_sum_(t)
