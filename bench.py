@@ -195,6 +195,12 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
     elapsed = float(t.cpu())
 
+    if rank == 0 and getattr(engine, "_heavy_stats", None) is not None:
+        hs = engine._heavy_stats.cpu().numpy()
+        if hs[0]:
+            print(f"[bench] heavy link sampler: {int(hs[0])} A* iterations, "
+                  f"{int(hs[1])} full-scan fallbacks (all sweeps)", file=sys.stderr)
+
     phase_ms = None
     if rank == 0 and hasattr(engine, "phase_times") and getattr(engine, "phase_timers", False):
         pt = engine.phase_times()

@@ -293,6 +293,14 @@ class GpuEngine(CpuEngine):
         self._phase_totals = {}
         self._hist_ready = False
         self._summary_host = None
+        # hierarchical A* link sampler for huge candidate sets (stationary
+        # high-distortion regime); DBLINK_HEAVY=0 reverts to pure scans
+        self._heavy_thresh = (
+            int(os.environ.get("DBLINK_HEAVY_THRESH", "512"))
+            if os.environ.get("DBLINK_HEAVY", "1") != "0" else 0
+        )
+        self._heavy_ratio_cap = 256
+        self._heavy_stats = torch.zeros(2, dtype=torch.int64, device=device)
         # overlapped migration (migrants-only async all-to-all with the
         # posting build and summary readback hidden under it) is the default
         # multi-rank path; DBLINK_OVERLAP=0 selects the eager reference path
@@ -484,16 +492,23 @@ class GpuEngine(CpuEngine):
         hist_pre = self._hist_ready
         self._hist_ready = False
 
+        ent_ptr = torch.searchsorted(
+            gs.ent_part.to(torch.int64).contiguous(),
+            torch.arange(self.num_partitions + 1, device=dev, dtype=torch.int64),
+        )
+
         # --- inverted index (counting sort over (partition, slot, value)) ----
         # Posting order within a key is arbitrary: the link kernels only
         # enumerate ranges (membership checks are entity-value compares) and
         # Gumbel draws are keyed by entity id, so no stable sort is needed.
         # Ranges come straight off the dense key prefix — no searchsorted.
+        dense_idx = False
         if not flags.sequential and not flags.collapsed_entity_ids:
             T = A + self._num_pairs
             vmax = max(m.Vmax, self._pair_vmax)
             nk = self.num_partitions * T * vmax
             if nk <= (1 << 28):  # dense counters (1 GiB cap; always true in practice)
+                dense_idx = True
                 if not hist_pre:
                     self._ensure_idx_buffers(zero=True)
                     self.C.postings_hist(gs.ent_part, gs.ent_values, self._pair_a1,
@@ -522,11 +537,6 @@ class GpuEngine(CpuEngine):
                 cand_lo = torch.searchsorted(sorted_keys, qkeys, right=False).view(R, T).contiguous()
                 cand_hi = torch.searchsorted(sorted_keys, qkeys, right=True).view(R, T).contiguous()
 
-        ent_ptr = torch.searchsorted(
-            gs.ent_part.to(torch.int64).contiguous(),
-            torch.arange(self.num_partitions + 1, device=dev, dtype=torch.int64),
-        )
-
         self._mark("index", graph_safe)
 
         # --- phase 1: link update --------------------------------------------
@@ -539,21 +549,30 @@ class GpuEngine(CpuEngine):
                 1 if flags.collapsed_entity_ids else 0, seed, it, rec_ent_new, ctrl,
             )
         else:
-            # records with a short smallest candidate list run one-per-thread;
-            # selection is a device-side mask (no host sync, graph-safe)
-            if R >= 50_000:
-                small_mask = torch.empty(R, dtype=torch.uint8, device=dev)
-                self.C.classify_small(gs.rec_values, gs.rec_dist, cand_lo,
-                                      cand_hi, self._num_pairs, 16, small_mask)
-            else:
-                small_mask = torch.empty(0, dtype=torch.uint8, device=dev)
+            # three-way routing, all device-side (no host sync, graph-safe):
+            # mode 0 wave scan, mode 1 thread scan (short candidate lists),
+            # mode 2 hierarchical A* sampler (huge partition-slice candidate
+            # sets — the stationary high-distortion regime)
+            heavy_th = self._heavy_thresh if dense_idx else 0
+            mode_mask = torch.empty(R, dtype=torch.uint8, device=dev)
+            self.C.classify_modes(gs.rec_values, gs.rec_dist, gs.rec_part,
+                                  ent_ptr, cand_lo, cand_hi, self._num_pairs,
+                                  16, heavy_th, self._heavy_ratio_cap, mode_mask)
             self.C.link_update(
                 gs.rec_values, gs.rec_dist, gs.rec_gid, gs.rec_part,
                 cand_lo, cand_hi, postings, gs.ent_values,
                 ent_ptr, m.log_norm, m.voff, m.csr_row_ptr, m.csr_col, m.csr_sim,
                 m.attr_const, seed, it, rec_ent_new, gs.rec_ent, self._err,
-                small_mask, ctrl, self._pair_a1, self._pair_a2,
+                mode_mask, ctrl, self._pair_a1, self._pair_a2,
             )
+            if heavy_th > 0:
+                self.C.link_update_heavy(
+                    mode_mask, gs.rec_values, gs.rec_dist, gs.rec_gid,
+                    gs.rec_part, gs.ent_values, ent_ptr, m.log_norm, m.voff,
+                    m.csr_row_ptr, m.csr_col, m.csr_sim, m.attr_const,
+                    postings, self._idx_ptr, vmax, self._num_pairs, seed, it,
+                    ctrl, rec_ent_new, gs.rec_ent, self._err, self._heavy_stats,
+                )
         gs.rec_ent.copy_(rec_ent_new)
         self._mark("link", graph_safe)
 

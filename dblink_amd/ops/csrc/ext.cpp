@@ -33,9 +33,22 @@ void cand_ranges(torch::Tensor rec_part, torch::Tensor rec_values,
                  torch::Tensor pair_a1, torch::Tensor pair_a2,
                  torch::Tensor pair_v2, torch::Tensor ptr, int64_t Vmax,
                  torch::Tensor cand_lo, torch::Tensor cand_hi);
-void classify_small(torch::Tensor rec_values, torch::Tensor rec_dist,
+void classify_modes(torch::Tensor rec_values, torch::Tensor rec_dist,
+                    torch::Tensor rec_part, torch::Tensor ent_ptr,
                     torch::Tensor cand_lo, torch::Tensor cand_hi, int64_t NP,
-                    int64_t threshold, torch::Tensor small_mask);
+                    int64_t small_threshold, int64_t heavy_threshold,
+                    int64_t heavy_ratio_cap, torch::Tensor mode);
+void link_update_heavy(torch::Tensor mode, torch::Tensor rec_values,
+                       torch::Tensor rec_dist, torch::Tensor rec_gid,
+                       torch::Tensor rec_part, torch::Tensor ent_values,
+                       torch::Tensor ent_ptr, torch::Tensor log_norm,
+                       torch::Tensor voff, torch::Tensor csr_row_ptr,
+                       torch::Tensor csr_col, torch::Tensor csr_sim,
+                       torch::Tensor attr_const, torch::Tensor postings,
+                       torch::Tensor idx_ptr, int64_t Vmax, int64_t NP,
+                       int64_t seed, int64_t iteration, torch::Tensor ctrl,
+                       torch::Tensor rec_ent_out, torch::Tensor rec_ent_in,
+                       torch::Tensor error_count, torch::Tensor stats);
 void link_update_dense(torch::Tensor rec_values, torch::Tensor rec_dist,
                        torch::Tensor rec_gid, torch::Tensor rec_part,
                        torch::Tensor rec_file, torch::Tensor ent_values,
@@ -113,7 +126,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "counting-sort index: posting scatter");
   m.def("cand_ranges", &dblink::cand_ranges,
         "per-record candidate ranges from the dense key prefix");
-  m.def("classify_small", &dblink::classify_small,
-        "mark records with a short smallest candidate list");
+  m.def("classify_modes", &dblink::classify_modes,
+        "route records to the wave / thread / hierarchical link paths");
+  m.def("link_update_heavy", &dblink::link_update_heavy,
+        "hierarchical (A*) Gumbel-max link update for huge candidate sets");
   m.def("summary_counts", &dblink::summary_counts, "fused summary counts + pack");
 }

@@ -32,7 +32,8 @@ namespace dblink {
 #define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
 
 constexpr int MAX_ATTRS = 16;
-constexpr uint32_t PH_LINK = 1, PH_DIST = 2, PH_VALG = 3, PH_VALM = 4;
+constexpr uint32_t PH_LINK = 1, PH_DIST = 2, PH_VALG = 3, PH_VALM = 4,
+                   PH_LINKH = 7, PH_LINKF = 8;
 
 // ---------------------------------------------------------------------------
 // K3+K4+K5: link update (PCG-I / Gibbs indexed path)
@@ -174,7 +175,7 @@ __global__ void link_update_small_kernel(
     const int64_t* __restrict__ rec_ent_in, int* __restrict__ error_count) {
   const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (idx >= n_recs) return;
-  if (!small_mask[idx]) return;
+  if (small_mask[idx] != 1) return;
   if (ctrl != nullptr) { seed = (uint64_t)ctrl[0]; iteration = (uint32_t)ctrl[1]; }
   const int64_t r = idx;
 
@@ -236,6 +237,229 @@ __global__ void link_update_small_kernel(
     best_e = rec_ent_in[r];
   }
   rec_ent_out[r] = best_e;
+}
+
+// ---------------------------------------------------------------------------
+// Heavy link update: hierarchical (A*) Gumbel-max sampler.
+//
+// In the model's stationary regime the distortion probabilities are large
+// (~0.2-0.6 under the shipped demo priors, measured on real RLdata10000), so
+// many records have few observed NON-distorted attributes: their candidate
+// set is a whole low-cardinality posting range (thousands of entities) or
+// the entire partition (GibbsUpdates.scala:398-430 with an empty
+// obsNonDistorted set). Scanning those ranges is what made sweeps grow
+// ~100x between burn-in and stationarity. This kernel draws the SAME
+// categorical (exactly — validated statistically against the scan path)
+// in O(|similar set| + tens) per record:
+//
+//   score(e) = t(e) + sim(e) + gumbel_e,   t(e) = sum_od lognorm(y_e,a) <= 0
+//
+//   1. exact Gumbel-max over S_r = entities similar to any observed-distorted
+//      record value (enumerated via the sim-row's posting segments) that pass
+//      the non-distorted equality checks; per-entity keyed gumbels.
+//   2. the complement (sim == 0) via A* sampling with a UNIFORM proposal
+//      (q = 0 >= t since every log-normalizer is <= 0): successive maxima
+//      G_1 > G_2 > ... of the pool's gumbels via truncated-Gumbel draws;
+//      each drawn entity is excluded (Z -= 1) and, if it passes the checks
+//      and is not similar, scores t(e) + G_k. Stop when G <= current best
+//      (no remaining candidate can win: t <= 0).
+//   3. winner = argmax of the two halves. Independent keyed RNG streams per
+//      half keep the combined draw an exact categorical sample.
+//
+// All draws are keyed by (seed, iteration, record gid, counter), so results
+// are deterministic and independent of posting order. If the A* loop hits
+// its exclusion cap (rare; counted in stats[1]), the record falls back to a
+// deterministic full scan with a FRESH key stream (PH_LINKF) — exact
+// regardless of what the truncated chain did.
+// ---------------------------------------------------------------------------
+
+constexpr int HEAVY_EXC_CAP = 128;   // A* exclusions per record before fallback
+constexpr int HEAVY_WAVES = 4;
+
+__global__ void link_update_heavy_kernel(
+    const uint8_t* __restrict__ mode,        // [R] 2 = heavy
+    const int32_t* __restrict__ rec_values, const uint8_t* __restrict__ rec_dist,
+    const int64_t* __restrict__ rec_gid, const int32_t* __restrict__ rec_part,
+    const int32_t* __restrict__ ent_values, const int64_t* __restrict__ ent_ptr,
+    const float* __restrict__ log_norm, const int64_t* __restrict__ voff,
+    const int64_t* __restrict__ csr_row_ptr, const int32_t* __restrict__ csr_col,
+    const float* __restrict__ csr_sim, const uint8_t* __restrict__ attr_const,
+    const int32_t* __restrict__ postings, const int64_t* __restrict__ idx_ptr,
+    int64_t Vmax, int NP, int64_t R, int A,
+    uint64_t seed, uint32_t iteration, const int64_t* __restrict__ ctrl,
+    int64_t* __restrict__ rec_ent_out, const int64_t* __restrict__ rec_ent_in,
+    int* __restrict__ error_count,
+    unsigned long long* __restrict__ stats) { // [2] = {astar_iters, fallbacks} or null
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int64_t r = (int64_t)blockIdx.x * HEAVY_WAVES + wid;
+  if (r >= R || mode[r] != 2) return;
+  if (ctrl != nullptr) { seed = (uint64_t)ctrl[0]; iteration = (uint32_t)ctrl[1]; }
+
+  const int T = A + NP;
+  uint32_t nd_mask = 0, od_mask = 0;
+  for (int a = 0; a < A; ++a) {
+    const int32_t x = rec_values[r * A + a];
+    if (x < 0) continue;
+    if (!rec_dist[r * A + a]) nd_mask |= 1u << a;
+    else if (!attr_const[a]) od_mask |= 1u << a;
+  }
+  const int32_t p = rec_part[r];
+  const int64_t e0 = ent_ptr[p], e1 = ent_ptr[p + 1];
+  const int64_t n_pool = e1 - e0;
+  const uint64_t gid = (uint64_t)rec_gid[r];
+
+  // ---- 1. exact scan over the similar set S_r ----------------------------
+  float best_f = -INFINITY;
+  long long best_e = -1;
+  for (uint32_t m = od_mask; m;) {
+    const int a = __ffs(m) - 1;
+    m &= m - 1;
+    const int64_t row = voff[a] + rec_values[r * A + a];
+    const int64_t j0 = csr_row_ptr[row], j1 = csr_row_ptr[row + 1];
+    for (int64_t j = j0; j < j1; ++j) {
+      const int32_t c = csr_col[j];
+      const int64_t key = ((int64_t)p * T + a) * Vmax + c;
+      const int64_t lo = idx_ptr[key], hi = idx_ptr[key + 1];
+      for (int64_t i = lo + lane; i < hi; i += WAVE) {
+        const int32_t e = postings[i];
+        bool ok = true;
+        for (uint32_t mm = nd_mask; mm;) {
+          const int aa = __ffs(mm) - 1;
+          mm &= mm - 1;
+          if (ent_values[(int64_t)e * A + aa] != rec_values[r * A + aa]) {
+            ok = false;
+            break;
+          }
+        }
+        if (!ok) continue;
+        float logw = 0.0f;
+        for (uint32_t mo = od_mask; mo;) {
+          const int ao = __ffs(mo) - 1;
+          mo &= mo - 1;
+          const int32_t y = ent_values[(int64_t)e * A + ao];
+          logw += log_norm[voff[ao] + y] +
+                  sim_lookup(csr_row_ptr, csr_col, csr_sim,
+                             voff[ao] + rec_values[r * A + ao], y);
+        }
+        const float g = gumbel_from_uniform(
+            philox_uniform(seed, iteration, PH_LINK, gid, (uint32_t)e));
+        if (logw + g > best_f) { best_f = logw + g; best_e = e; }
+      }
+    }
+  }
+  wave_argmax(best_f, best_e);
+
+  // ---- 2. A* over the complement (uniform proposal; lane 0 serial) -------
+  __shared__ int32_t excl_s[HEAVY_WAVES][HEAVY_EXC_CAP];
+  __shared__ long long winner_s[HEAVY_WAVES];
+  __shared__ int fell_back_s[HEAVY_WAVES];
+  if (lane == 0) {
+    int32_t* excl = excl_s[wid];
+    double best = (best_e >= 0) ? (double)best_f : -INFINITY;
+    long long bE = best_e;
+    double Z = (double)n_pool;
+    double b = 0.0;
+    bool first = true;
+    uint32_t ctr = 0;
+    int n_exc = 0;
+    bool fell_back = false;
+    unsigned long long iters = 0;
+    while (Z > 0.5) {
+      const double logZ = log(Z);
+      double G;
+      {
+        const double u = (double)philox_uniform(seed, iteration, PH_LINKH, gid, ctr++);
+        G = first ? logZ - log(-log(u)) : logZ - log(exp(logZ - b) - log(u));
+        first = false;
+      }
+      b = G;
+      if (G <= best) break;
+      // categorical (uniform) draw over pool \ excluded
+      int32_t e = -1;
+      for (int tries = 0; tries < 4 * HEAVY_EXC_CAP; ++tries) {
+        const float u = philox_uniform(seed, iteration, PH_LINKH, gid, ctr++);
+        int64_t cand = e0 + (int64_t)(u * (float)n_pool);
+        if (cand >= e1) cand = e1 - 1;
+        bool hit = false;
+        for (int k = 0; k < n_exc; ++k)
+          if (excl[k] == (int32_t)cand) { hit = true; break; }
+        if (!hit) { e = (int32_t)cand; break; }
+      }
+      if (e < 0 || n_exc >= HEAVY_EXC_CAP) { fell_back = true; break; }
+      excl[n_exc++] = e;
+      Z -= 1.0;
+      ++iters;
+      bool ok = true;
+      for (uint32_t mm = nd_mask; mm;) {
+        const int aa = __ffs(mm) - 1;
+        mm &= mm - 1;
+        if (ent_values[(int64_t)e * A + aa] != rec_values[r * A + aa]) {
+          ok = false;
+          break;
+        }
+      }
+      if (!ok) continue;
+      double t = 0.0;
+      bool similar = false;
+      for (uint32_t mo = od_mask; mo;) {
+        const int ao = __ffs(mo) - 1;
+        mo &= mo - 1;
+        const int32_t y = ent_values[(int64_t)e * A + ao];
+        const float s = sim_lookup(csr_row_ptr, csr_col, csr_sim,
+                                   voff[ao] + rec_values[r * A + ao], y);
+        if (s != 0.0f) { similar = true; break; }  // scored exactly in S_r
+        t += (double)log_norm[voff[ao] + y];
+      }
+      if (similar) continue;
+      const double s_true = t + G;  // uniform proposal: gumbel realization == G
+      if (s_true > best) { best = s_true; bE = e; }
+    }
+    winner_s[wid] = bE;
+    fell_back_s[wid] = fell_back ? 1 : 0;
+    if (stats != nullptr) {
+      atomicAdd(&stats[0], iters);
+      if (fell_back) atomicAdd(&stats[1], 1ull);
+    }
+  }
+  __syncthreads();
+
+  // ---- 3. rare fallback: deterministic full scan, fresh key stream -------
+  if (fell_back_s[wid]) {
+    float bf = -INFINITY;
+    long long be = -1;
+    for (int64_t e = e0 + lane; e < e1; e += WAVE) {
+      bool ok = true;
+      for (uint32_t mm = nd_mask; mm;) {
+        const int aa = __ffs(mm) - 1;
+        mm &= mm - 1;
+        if (ent_values[e * A + aa] != rec_values[r * A + aa]) { ok = false; break; }
+      }
+      if (!ok) continue;
+      float logw = 0.0f;
+      for (uint32_t mo = od_mask; mo;) {
+        const int ao = __ffs(mo) - 1;
+        mo &= mo - 1;
+        const int32_t y = ent_values[e * A + ao];
+        logw += log_norm[voff[ao] + y] +
+                sim_lookup(csr_row_ptr, csr_col, csr_sim,
+                           voff[ao] + rec_values[r * A + ao], y);
+      }
+      const float g = gumbel_from_uniform(
+          philox_uniform(seed, iteration, PH_LINKF, gid, (uint32_t)(e - e0)));
+      if (logw + g > bf) { bf = logw + g; be = e; }
+    }
+    wave_argmax(bf, be);
+    if (lane == 0) winner_s[wid] = be;
+  }
+  if (lane == 0) {
+    long long w = winner_s[wid];
+    if (w < 0) {  // empty candidate set: state invariant violated
+      atomicAdd(error_count, 1);
+      w = rec_ent_in[r];
+    }
+    rec_ent_out[r] = w;
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -1157,24 +1381,42 @@ __global__ __launch_bounds__(PB_THREADS) void postings_scatter_kernel(
   if (my_slot >= 0) postings[h_base[my_slot] + my_rank] = e_out;
 }
 
-// One thread per record: mark records whose smallest non-distorted candidate
-// list is short (handled serially by link_update_small_kernel). Replaces a
-// six-op torch mask pipeline over [R, A].
-__global__ void classify_small_kernel(
+// One thread per record: route each record to one of three link paths.
+//   mode 0 = wave scan (link_update_kernel)
+//   mode 1 = thread scan (link_update_small_kernel, short candidate list)
+//   mode 2 = hierarchical sampler (link_update_heavy_kernel): records whose
+//            candidate set is a large partition slice — no observed
+//            non-distorted attribute (pool = whole partition), or exactly one
+//            with a big posting range — where an exact-scan would touch
+//            thousands of entities. Guarded by pool/base ratio so the
+//            uniform-proposal rejection stays cheap.
+__global__ void classify_modes_kernel(
     const int32_t* __restrict__ rec_values, const uint8_t* __restrict__ rec_dist,
+    const int32_t* __restrict__ rec_part, const int64_t* __restrict__ ent_ptr,
     const int64_t* __restrict__ cand_lo, const int64_t* __restrict__ cand_hi,
-    int64_t R, int A, int NP, int64_t threshold,
-    uint8_t* __restrict__ small_mask) {
+    int64_t R, int A, int NP, int64_t small_threshold, int64_t heavy_threshold,
+    int64_t heavy_ratio_cap, uint8_t* __restrict__ mode) {
   const int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (r >= R) return;
   const int T = A + NP;
   int64_t best = INT64_MAX;
+  int nd_count = 0;
   for (int a = 0; a < A; ++a) {
     if (rec_values[r * A + a] < 0 || rec_dist[r * A + a]) continue;
+    ++nd_count;
     const int64_t n = cand_hi[r * T + a] - cand_lo[r * T + a];
     if (n < best) best = n;
   }
-  small_mask[r] = best != INT64_MAX && best <= threshold;
+  uint8_t m = 0;
+  if (heavy_threshold > 0) {
+    const int32_t p = rec_part[r];
+    const int64_t pool = ent_ptr[p + 1] - ent_ptr[p];
+    if (nd_count == 0 && pool > heavy_threshold) m = 2;
+    else if (nd_count == 1 && best > heavy_threshold &&
+             pool <= heavy_ratio_cap * best) m = 2;
+  }
+  if (m == 0 && best != INT64_MAX && best <= small_threshold) m = 1;
+  mode[r] = m;
 }
 
 __global__ void cand_ranges_kernel(
@@ -1488,18 +1730,53 @@ void postings_scatter(torch::Tensor ent_part, torch::Tensor ent_values,
                      cursor.data_ptr<int32_t>(), postings.data_ptr<int32_t>());
 }
 
-void classify_small(torch::Tensor rec_values, torch::Tensor rec_dist,
+void classify_modes(torch::Tensor rec_values, torch::Tensor rec_dist,
+                    torch::Tensor rec_part, torch::Tensor ent_ptr,
                     torch::Tensor cand_lo, torch::Tensor cand_hi, int64_t NP,
-                    int64_t threshold, torch::Tensor small_mask) {
+                    int64_t small_threshold, int64_t heavy_threshold,
+                    int64_t heavy_ratio_cap, torch::Tensor mode) {
   const int64_t R = rec_values.size(0);
   const int A = (int)rec_values.size(1);
   if (R == 0) return;
   dim3 grid((unsigned)((R + 255) / 256));
-  hipLaunchKernelGGL(classify_small_kernel, grid, dim3(256), 0,
+  hipLaunchKernelGGL(classify_modes_kernel, grid, dim3(256), 0,
                      at::cuda::getCurrentCUDAStream(),
                      rec_values.data_ptr<int32_t>(), rec_dist.data_ptr<uint8_t>(),
+                     rec_part.data_ptr<int32_t>(), ent_ptr.data_ptr<int64_t>(),
                      cand_lo.data_ptr<int64_t>(), cand_hi.data_ptr<int64_t>(),
-                     R, A, (int)NP, threshold, small_mask.data_ptr<uint8_t>());
+                     R, A, (int)NP, small_threshold, heavy_threshold,
+                     heavy_ratio_cap, mode.data_ptr<uint8_t>());
+}
+
+void link_update_heavy(
+    torch::Tensor mode, torch::Tensor rec_values, torch::Tensor rec_dist,
+    torch::Tensor rec_gid, torch::Tensor rec_part, torch::Tensor ent_values,
+    torch::Tensor ent_ptr, torch::Tensor log_norm, torch::Tensor voff,
+    torch::Tensor csr_row_ptr, torch::Tensor csr_col, torch::Tensor csr_sim,
+    torch::Tensor attr_const, torch::Tensor postings, torch::Tensor idx_ptr,
+    int64_t Vmax, int64_t NP, int64_t seed, int64_t iteration, torch::Tensor ctrl,
+    torch::Tensor rec_ent_out, torch::Tensor rec_ent_in,
+    torch::Tensor error_count, torch::Tensor stats) {
+  const int64_t* ctrl_ptr = ctrl.numel() ? ctrl.data_ptr<int64_t>() : nullptr;
+  const int64_t R = rec_values.size(0);
+  const int A = (int)rec_values.size(1);
+  if (R == 0) return;
+  dim3 grid((unsigned)wave_grid(R, HEAVY_WAVES));
+  hipLaunchKernelGGL(link_update_heavy_kernel, grid, dim3(HEAVY_WAVES * WAVE), 0,
+                     at::cuda::getCurrentCUDAStream(),
+                     mode.data_ptr<uint8_t>(), rec_values.data_ptr<int32_t>(),
+                     rec_dist.data_ptr<uint8_t>(), rec_gid.data_ptr<int64_t>(),
+                     rec_part.data_ptr<int32_t>(), ent_values.data_ptr<int32_t>(),
+                     ent_ptr.data_ptr<int64_t>(), log_norm.data_ptr<float>(),
+                     voff.data_ptr<int64_t>(), csr_row_ptr.data_ptr<int64_t>(),
+                     csr_col.data_ptr<int32_t>(), csr_sim.data_ptr<float>(),
+                     attr_const.data_ptr<uint8_t>(), postings.data_ptr<int32_t>(),
+                     idx_ptr.data_ptr<int64_t>(), Vmax, (int)NP, R, A,
+                     (uint64_t)seed, (uint32_t)iteration, ctrl_ptr,
+                     rec_ent_out.data_ptr<int64_t>(), rec_ent_in.data_ptr<int64_t>(),
+                     error_count.data_ptr<int>(),
+                     stats.numel() ? (unsigned long long*)stats.data_ptr<int64_t>()
+                                   : nullptr);
 }
 
 void cand_ranges(torch::Tensor rec_part, torch::Tensor rec_values,

@@ -825,3 +825,80 @@ def test_link_kernel_parity_on_realistic_state():
         assert tv_distance(emp, full) < 3 * noise + 0.01, (
             r, tv_distance(emp, full), noise, emp[cands][:6], exact[:6]
         )
+
+
+@gpu
+def test_link_heavy_kernel_distribution():
+    """Hierarchical A* link sampler (link_update_heavy_kernel) vs exact
+    weights: class-U records (no non-distorted attr -> pool = whole
+    partition) and single-nd records (one big posting range). The sampler
+    must draw the same categorical as the scan path
+    (GibbsUpdates.scala:398-430), in O(|similar set| + tens) work."""
+    cache, model = make_model(DEV)
+    ia = cache.indexed_attributes[1]
+    idx = ia.index
+    rng = np.random.default_rng(7)
+    E = 1500
+    ent_vals = np.zeros((E, 2), dtype=np.int32)
+    ent_vals[:800, 0] = 3                       # posting(year=3) = 800 > 512
+    ent_vals[800:, 0] = rng.integers(0, 3, E - 800)
+    ent_vals[:, 1] = rng.integers(0, idx.num_values, E).astype(np.int32)
+
+    A, T, Vmax = 2, 2, model.Vmax
+    nk = 1 * T * Vmax
+    keys = ((np.repeat([0, 1], E)) * Vmax + ent_vals.T.reshape(-1)).astype(np.int64)
+    counts = np.bincount(keys, minlength=nk)
+    idx_ptr = np.zeros(nk + 1, dtype=np.int64)
+    idx_ptr[1:] = np.cumsum(counts)
+    order = np.argsort(keys, kind="stable")
+    postings = (order % E).astype(np.int32)
+
+    x = 0  # record's (distorted) name value
+    for case, dist_row, rec_row in (
+        ("classU", [1, 1], [3, x]),      # no nd attr: pool = all entities
+        ("single-nd", [0, 1], [3, x]),   # nd = {year}: base = posting(year=3)
+    ):
+        N = 40000
+        rec_values = np.tile(np.array([rec_row], dtype=np.int32), (N, 1))
+        rec_dist = np.tile(np.array([dist_row], dtype=np.uint8), (N, 1))
+        w = np.array([
+            idx.sim_norms[y] * idx.exp_sim_of(x, int(y)) for y in ent_vals[:, 1]
+        ])
+        if case == "single-nd":
+            w = w * (ent_vals[:, 0] == 3)
+        exact = w / w.sum()
+
+        qk = np.array([0 * Vmax + rec_row[0], 1 * Vmax + x], dtype=np.int64)
+        cand_lo = np.tile(idx_ptr[qk], (N, 1)).astype(np.int64)
+        cand_hi = np.tile(idx_ptr[qk + 1], (N, 1)).astype(np.int64)
+
+        mode = torch.empty(N, dtype=torch.uint8, device=DEV)
+        ent_ptr = _dev(np.array([0, E], dtype=np.int64), torch.int64)
+        rec_part = _dev(np.zeros(N, np.int32), torch.int32)
+        rv = _dev(rec_values, torch.int32)
+        rd = _dev(rec_dist, torch.uint8)
+        C.classify_modes(rv, rd, rec_part, ent_ptr,
+                         _dev(cand_lo, torch.int64), _dev(cand_hi, torch.int64),
+                         0, 16, 512, 256, mode)
+        assert int(mode.to(torch.int64).min().cpu()) == 2, case
+
+        out = torch.empty(N, dtype=torch.int64, device=DEV)
+        err = torch.zeros(1, dtype=torch.int32, device=DEV)
+        stats = torch.zeros(2, dtype=torch.int64, device=DEV)
+        C.link_update_heavy(
+            mode, rv, rd,
+            _dev(np.arange(N, dtype=np.int64), torch.int64), rec_part,
+            _dev(ent_vals, torch.int32), ent_ptr,
+            model.log_norm, model.voff, model.csr_row_ptr, model.csr_col,
+            model.csr_sim, model.attr_const,
+            _dev(postings, torch.int32), _dev(idx_ptr, torch.int64),
+            Vmax, 0, 4321, 11, torch.empty(0, dtype=torch.int64, device=DEV),
+            out, _dev(np.zeros(N, np.int64), torch.int64), err, stats)
+        assert int(err.cpu()) == 0, case
+        sel = out.cpu().numpy()
+        emp = np.bincount(sel, minlength=E) / N
+        assert tv_distance(emp, exact) < 0.02, (case, tv_distance(emp, exact))
+        st = stats.cpu().numpy()
+        # the whole point: tens of A* iterations per record, few fallbacks
+        assert st[0] / N < 200, ("A* iterations per record too high", st)
+        assert st[1] / N < 0.01, ("too many full-scan fallbacks", st)
