@@ -299,7 +299,6 @@ class GpuEngine(CpuEngine):
             int(os.environ.get("DBLINK_HEAVY_THRESH", "512"))
             if os.environ.get("DBLINK_HEAVY", "1") != "0" else 0
         )
-        self._heavy_ratio_cap = 256
         self._heavy_stats = torch.zeros(2, dtype=torch.int64, device=device)
         # overlapped migration (migrants-only async all-to-all with the
         # posting build and summary readback hidden under it) is the default
@@ -516,11 +515,24 @@ class GpuEngine(CpuEngine):
                                          self._idx_counts)
                 torch.cumsum(self._idx_counts, 0, dtype=torch.int64,
                              out=self._idx_ptr[1:])
-                self._idx_cursor.copy_(self._idx_ptr[:-1])
-                postings = torch.empty(T * E, dtype=torch.int32, device=dev)
-                self.C.postings_scatter(gs.ent_part, gs.ent_values, self._pair_a1,
-                                        self._pair_a2, self._pair_v2, vmax,
-                                        self._idx_cursor, postings)
+                if self._heavy_thresh > 0:
+                    # heavy sampler draws postings by segment INDEX, so the
+                    # order within each key must be deterministic: sort
+                    # (key * E + entity) — same boundaries as the prefix,
+                    # entities ascending within each segment
+                    ekeys = torch.empty(T * E, dtype=torch.int64, device=dev)
+                    self.C.build_ekeys_stable(gs.ent_part, gs.ent_values,
+                                              self._pair_a1, self._pair_a2,
+                                              self._pair_v2, vmax, ekeys)
+                    perm = torch.argsort(ekeys)
+                    postings = (perm % E).to(torch.int32)
+                else:
+                    self._idx_cursor.copy_(self._idx_ptr[:-1])
+                    postings = torch.empty(T * E, dtype=torch.int32, device=dev)
+                    self.C.postings_scatter(gs.ent_part, gs.ent_values,
+                                            self._pair_a1, self._pair_a2,
+                                            self._pair_v2, vmax,
+                                            self._idx_cursor, postings)
                 cand_lo = torch.empty((R, T), dtype=torch.int64, device=dev)
                 cand_hi = torch.empty((R, T), dtype=torch.int64, device=dev)
                 self.C.cand_ranges(gs.rec_part, gs.rec_values, self._pair_a1,
@@ -557,7 +569,7 @@ class GpuEngine(CpuEngine):
             mode_mask = torch.empty(R, dtype=torch.uint8, device=dev)
             self.C.classify_modes(gs.rec_values, gs.rec_dist, gs.rec_part,
                                   ent_ptr, cand_lo, cand_hi, self._num_pairs,
-                                  16, heavy_th, self._heavy_ratio_cap, mode_mask)
+                                  16, heavy_th, mode_mask)
             self.C.link_update(
                 gs.rec_values, gs.rec_dist, gs.rec_gid, gs.rec_part,
                 cand_lo, cand_hi, postings, gs.ent_values,

@@ -37,7 +37,10 @@ void classify_modes(torch::Tensor rec_values, torch::Tensor rec_dist,
                     torch::Tensor rec_part, torch::Tensor ent_ptr,
                     torch::Tensor cand_lo, torch::Tensor cand_hi, int64_t NP,
                     int64_t small_threshold, int64_t heavy_threshold,
-                    int64_t heavy_ratio_cap, torch::Tensor mode);
+                    torch::Tensor mode);
+void build_ekeys_stable(torch::Tensor ent_part, torch::Tensor ent_values,
+                        torch::Tensor pair_a1, torch::Tensor pair_a2,
+                        torch::Tensor pair_v2, int64_t Vmax, torch::Tensor ekeys);
 void link_update_heavy(torch::Tensor mode, torch::Tensor rec_values,
                        torch::Tensor rec_dist, torch::Tensor rec_gid,
                        torch::Tensor rec_part, torch::Tensor ent_values,
@@ -128,6 +131,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "per-record candidate ranges from the dense key prefix");
   m.def("classify_modes", &dblink::classify_modes,
         "route records to the wave / thread / hierarchical link paths");
+  m.def("build_ekeys_stable", &dblink::build_ekeys_stable,
+        "stable inverted-index sort keys (key * E + entity)");
   m.def("link_update_heavy", &dblink::link_update_heavy,
         "hierarchical (A*) Gumbel-max link update for huge candidate sets");
   m.def("summary_counts", &dblink::summary_counts, "fused summary counts + pack");

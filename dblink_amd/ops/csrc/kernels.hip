@@ -306,8 +306,22 @@ __global__ void link_update_heavy_kernel(
   }
   const int32_t p = rec_part[r];
   const int64_t e0 = ent_ptr[p], e1 = ent_ptr[p + 1];
-  const int64_t n_pool = e1 - e0;
   const uint64_t gid = (uint64_t)rec_gid[r];
+
+  // proposal pool: the whole partition (no nd attr) or the base posting
+  // segment of the single nd attr (postings are stable-sorted when this
+  // kernel is active, so indexed draws are deterministic)
+  const bool seg_pool = nd_mask != 0;
+  int64_t plo, pn;
+  if (seg_pool) {
+    const int a = __ffs(nd_mask) - 1;
+    const int64_t key = ((int64_t)p * T + a) * Vmax + rec_values[r * A + a];
+    plo = idx_ptr[key];
+    pn = idx_ptr[key + 1] - plo;
+  } else {
+    plo = e0;
+    pn = e1 - e0;
+  }
 
   // ---- 1. exact scan over the similar set S_r ----------------------------
   float best_f = -INFINITY;
@@ -358,7 +372,7 @@ __global__ void link_update_heavy_kernel(
     int32_t* excl = excl_s[wid];
     double best = (best_e >= 0) ? (double)best_f : -INFINITY;
     long long bE = best_e;
-    double Z = (double)n_pool;
+    double Z = (double)pn;
     double b = 0.0;
     bool first = true;
     uint32_t ctr = 0;
@@ -375,21 +389,22 @@ __global__ void link_update_heavy_kernel(
       }
       b = G;
       if (G <= best) break;
-      // categorical (uniform) draw over pool \ excluded
-      int32_t e = -1;
+      // categorical (uniform) draw over pool \ excluded, by pool index
+      int64_t ci = -1;
       for (int tries = 0; tries < 4 * HEAVY_EXC_CAP; ++tries) {
         const float u = philox_uniform(seed, iteration, PH_LINKH, gid, ctr++);
-        int64_t cand = e0 + (int64_t)(u * (float)n_pool);
-        if (cand >= e1) cand = e1 - 1;
+        int64_t cand = (int64_t)(u * (float)pn);
+        if (cand >= pn) cand = pn - 1;
         bool hit = false;
         for (int k = 0; k < n_exc; ++k)
           if (excl[k] == (int32_t)cand) { hit = true; break; }
-        if (!hit) { e = (int32_t)cand; break; }
+        if (!hit) { ci = cand; break; }
       }
-      if (e < 0 || n_exc >= HEAVY_EXC_CAP) { fell_back = true; break; }
-      excl[n_exc++] = e;
+      if (ci < 0 || n_exc >= HEAVY_EXC_CAP) { fell_back = true; break; }
+      excl[n_exc++] = (int32_t)ci;
       Z -= 1.0;
       ++iters;
+      const int32_t e = seg_pool ? postings[plo + ci] : (int32_t)(plo + ci);
       bool ok = true;
       for (uint32_t mm = nd_mask; mm;) {
         const int aa = __ffs(mm) - 1;
@@ -1387,15 +1402,14 @@ __global__ __launch_bounds__(PB_THREADS) void postings_scatter_kernel(
 //   mode 2 = hierarchical sampler (link_update_heavy_kernel): records whose
 //            candidate set is a large partition slice — no observed
 //            non-distorted attribute (pool = whole partition), or exactly one
-//            with a big posting range — where an exact-scan would touch
-//            thousands of entities. Guarded by pool/base ratio so the
-//            uniform-proposal rejection stays cheap.
+//            with a big posting range (pool = that stable posting segment) —
+//            where an exact scan would touch thousands of entities.
 __global__ void classify_modes_kernel(
     const int32_t* __restrict__ rec_values, const uint8_t* __restrict__ rec_dist,
     const int32_t* __restrict__ rec_part, const int64_t* __restrict__ ent_ptr,
     const int64_t* __restrict__ cand_lo, const int64_t* __restrict__ cand_hi,
     int64_t R, int A, int NP, int64_t small_threshold, int64_t heavy_threshold,
-    int64_t heavy_ratio_cap, uint8_t* __restrict__ mode) {
+    uint8_t* __restrict__ mode) {
   const int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (r >= R) return;
   const int T = A + NP;
@@ -1412,8 +1426,7 @@ __global__ void classify_modes_kernel(
     const int32_t p = rec_part[r];
     const int64_t pool = ent_ptr[p + 1] - ent_ptr[p];
     if (nd_count == 0 && pool > heavy_threshold) m = 2;
-    else if (nd_count == 1 && best > heavy_threshold &&
-             pool <= heavy_ratio_cap * best) m = 2;
+    else if (nd_count == 1 && best > heavy_threshold) m = 2;
   }
   if (m == 0 && best != INT64_MAX && best <= small_threshold) m = 1;
   mode[r] = m;
@@ -1734,7 +1747,7 @@ void classify_modes(torch::Tensor rec_values, torch::Tensor rec_dist,
                     torch::Tensor rec_part, torch::Tensor ent_ptr,
                     torch::Tensor cand_lo, torch::Tensor cand_hi, int64_t NP,
                     int64_t small_threshold, int64_t heavy_threshold,
-                    int64_t heavy_ratio_cap, torch::Tensor mode) {
+                    torch::Tensor mode) {
   const int64_t R = rec_values.size(0);
   const int A = (int)rec_values.size(1);
   if (R == 0) return;
@@ -1745,7 +1758,44 @@ void classify_modes(torch::Tensor rec_values, torch::Tensor rec_dist,
                      rec_part.data_ptr<int32_t>(), ent_ptr.data_ptr<int64_t>(),
                      cand_lo.data_ptr<int64_t>(), cand_hi.data_ptr<int64_t>(),
                      R, A, (int)NP, small_threshold, heavy_threshold,
-                     heavy_ratio_cap, mode.data_ptr<uint8_t>());
+                     mode.data_ptr<uint8_t>());
+}
+
+// Stable inverted-index keys: (key * E + e) sorts into the same segment
+// boundaries as the counting-sort prefix, but with entities in ascending-id
+// order within each segment — deterministic indexed draws for the heavy
+// sampler's posting-segment proposals.
+__global__ void build_ekeys_stable_kernel(
+    const int32_t* __restrict__ ent_part, const int32_t* __restrict__ ent_values,
+    const int32_t* __restrict__ pair_a1, const int32_t* __restrict__ pair_a2,
+    const int32_t* __restrict__ pair_v2, int64_t E, int A, int NP, int64_t Vmax,
+    int64_t* __restrict__ ekeys) {
+  const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int T = A + NP;
+  if (idx >= E * T) return;
+  const int t = (int)(idx / E);
+  const int64_t e = idx % E;
+  const int64_t v = posting_slot_value(ent_values, e, A, t, pair_a1, pair_a2, pair_v2);
+  const int64_t key = ((int64_t)ent_part[e] * T + t) * Vmax + v;
+  ekeys[idx] = key * E + e;
+}
+
+void build_ekeys_stable(torch::Tensor ent_part, torch::Tensor ent_values,
+                        torch::Tensor pair_a1, torch::Tensor pair_a2,
+                        torch::Tensor pair_v2, int64_t Vmax,
+                        torch::Tensor ekeys) {
+  const int64_t E = ent_values.size(0);
+  const int A = (int)ent_values.size(1);
+  const int NP = (int)pair_a1.numel();
+  const int64_t n = E * (A + NP);
+  if (n == 0) return;
+  dim3 grid((unsigned)((n + 255) / 256));
+  hipLaunchKernelGGL(build_ekeys_stable_kernel, grid, dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(),
+                     ent_part.data_ptr<int32_t>(), ent_values.data_ptr<int32_t>(),
+                     pair_a1.data_ptr<int32_t>(), pair_a2.data_ptr<int32_t>(),
+                     pair_v2.data_ptr<int32_t>(), E, A, NP, Vmax,
+                     ekeys.data_ptr<int64_t>());
 }
 
 void link_update_heavy(

@@ -879,7 +879,7 @@ def test_link_heavy_kernel_distribution():
         rd = _dev(rec_dist, torch.uint8)
         C.classify_modes(rv, rd, rec_part, ent_ptr,
                          _dev(cand_lo, torch.int64), _dev(cand_hi, torch.int64),
-                         0, 16, 512, 256, mode)
+                         0, 16, 512, mode)
         assert int(mode.to(torch.int64).min().cpu()) == 2, case
 
         out = torch.empty(N, dtype=torch.int64, device=DEV)
@@ -897,8 +897,16 @@ def test_link_heavy_kernel_distribution():
         assert int(err.cpu()) == 0, case
         sel = out.cpu().numpy()
         emp = np.bincount(sel, minlength=E) / N
-        assert tv_distance(emp, exact) < 0.02, (case, tv_distance(emp, exact))
+        # per-entity TV at 1500 cells is dominated by multinomial noise
+        # (~0.033 even for a perfect sampler at N=40000); entities sharing a
+        # (year-match, name-value) profile are exchangeable, so aggregate
+        # onto those groups where the noise floor is negligible
+        group = ent_vals[:, 1].astype(np.int64) * 2 + (ent_vals[:, 0] == 3)
+        ng = int(group.max()) + 1
+        emp_g = np.bincount(group, weights=emp, minlength=ng)
+        exact_g = np.bincount(group, weights=exact, minlength=ng)
+        assert tv_distance(emp_g, exact_g) < 0.01, (case, tv_distance(emp_g, exact_g))
         st = stats.cpu().numpy()
-        # the whole point: tens of A* iterations per record, few fallbacks
-        assert st[0] / N < 200, ("A* iterations per record too high", st)
+        # the whole point: few A* iterations per record, no fallback storms
+        assert st[0] / N < 100, ("A* iterations per record too high", st)
         assert st[1] / N < 0.01, ("too many full-scan fallbacks", st)
