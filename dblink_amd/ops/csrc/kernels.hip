@@ -324,6 +324,11 @@ __global__ void link_update_heavy_kernel(
   }
 
   // ---- 1. exact scan over the similar set S_r ----------------------------
+  // The sim-row's posting segments average only a few entries, so a
+  // segment-at-a-time wave loop would idle ~60 of 64 lanes. Instead each
+  // 64-column chunk is flattened warp-cooperatively: a register prefix scan
+  // of the segment lengths, then lanes stride the flat space and locate
+  // their owning column by a shuffle-based binary search.
   float best_f = -INFINITY;
   long long best_e = -1;
   for (uint32_t m = od_mask; m;) {
@@ -331,11 +336,36 @@ __global__ void link_update_heavy_kernel(
     m &= m - 1;
     const int64_t row = voff[a] + rec_values[r * A + a];
     const int64_t j0 = csr_row_ptr[row], j1 = csr_row_ptr[row + 1];
-    for (int64_t j = j0; j < j1; ++j) {
-      const int32_t c = csr_col[j];
-      const int64_t key = ((int64_t)p * T + a) * Vmax + c;
-      const int64_t lo = idx_ptr[key], hi = idx_ptr[key + 1];
-      for (int64_t i = lo + lane; i < hi; i += WAVE) {
+    for (int64_t cbase = j0; cbase < j1; cbase += WAVE) {
+      const int64_t myj = cbase + lane;
+      long long lo = 0;
+      int n = 0;
+      if (myj < j1) {
+        const int64_t key = ((int64_t)p * T + a) * Vmax + csr_col[myj];
+        lo = idx_ptr[key];
+        n = (int)(idx_ptr[key + 1] - lo);
+      }
+      int pre = n;  // inclusive prefix of segment lengths across lanes
+#pragma unroll
+      for (int off = 1; off < WAVE; off <<= 1) {
+        const int v = __shfl_up(pre, off);
+        if (lane >= off) pre += v;
+      }
+      const int total = __shfl(pre, WAVE - 1);
+      const int excl = pre - n;
+      for (int sb = 0; sb < total; sb += WAVE) {
+        const int s = sb + lane;
+        // owning lane = last lane with excl <= s (shuffle binary search);
+        // every lane participates in the shuffles (uniform loop bound)
+        int fl = 0;
+#pragma unroll
+        for (int step = WAVE / 2; step; step >>= 1) {
+          const int cand = fl + step;
+          const int ce = __shfl(excl, cand < WAVE ? cand : WAVE - 1);
+          if (cand < WAVE && ce <= s) fl = cand;
+        }
+        const int64_t i = __shfl(lo, fl) + (s - __shfl(excl, fl));
+        if (s >= total) continue;
         const int32_t e = postings[i];
         bool ok = true;
         for (uint32_t mm = nd_mask; mm;) {
