@@ -92,6 +92,12 @@ def main():
                     help="rldata: 3 const + 2 Levenshtein attrs; strings8: 8 Levenshtein attrs")
     ap.add_argument("--num-files", type=int, default=1,
                     help="number of source files (per-file distortion probabilities)")
+    ap.add_argument("--dump-state", default=None, metavar="DIR",
+                    help="run the warmup sweeps, save the chain state, exit "
+                         "(for profiling the stationary regime)")
+    ap.add_argument("--init-state", default=None, metavar="DIR",
+                    help="resume the chain from a saved state instead of "
+                         "deterministic init")
     args = ap.parse_args()
 
     sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
@@ -121,12 +127,20 @@ def main():
 
     part_attrs = [3, 4] if args.schema == "rldata" else [0, 1]  # Levenshtein attrs
     partitioner = KDTreePartitioner(num_levels, part_attrs)
-    bounds = np.linspace(0, total_records, world + 1).astype(np.int64)
-    lo, hi = int(bounds[rank]), int(bounds[rank + 1])
-    state = deterministic_init(
-        rec_values[lo:hi], rec_files[lo:hi], np.arange(lo, hi, dtype=np.int64),
-        cache, partitioner, args.seed, rank=rank, world_size=world,
-    )
+    if args.init_state:
+        from dblink_amd.engine.state import ChainState
+
+        state = ChainState.load(args.init_state, rank=rank, world_size=world)
+        saved = getattr(state, "saved_partitioner", None)
+        if saved is not None:
+            partitioner = saved
+    else:
+        bounds = np.linspace(0, total_records, world + 1).astype(np.int64)
+        lo, hi = int(bounds[rank]), int(bounds[rank + 1])
+        state = deterministic_init(
+            rec_values[lo:hi], rec_files[lo:hi], np.arange(lo, hi, dtype=np.int64),
+            cache, partitioner, args.seed, rank=rank, world_size=world,
+        )
 
     if rank == 0:
         print(f"[bench] init state in {time.time() - t_init:.1f}s total", file=sys.stderr)
@@ -150,6 +164,16 @@ def main():
         engine.step(state, flags)
     sync()
     t_w = time.time() - t_w
+
+    if args.dump_state:
+        if hasattr(engine, "sync_state"):
+            engine.sync_state(state)
+        state.save(args.dump_state, rank=rank, world_size=world,
+                   extra={"partitioner": engine.partitioner})
+        if rank == 0:
+            print(f"[bench] state after {args.warmup} sweeps saved to "
+                  f"{args.dump_state}", file=sys.stderr)
+        return
 
     if args.steps is None:
         # calibrate on a WARM probe (the warmup itself includes first-call
