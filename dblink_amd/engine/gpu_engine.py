@@ -321,6 +321,22 @@ class GpuEngine(CpuEngine):
             self._idx_counts.zero_()
         return nk
 
+    def _sort_bufs(self, n, E):
+        """Persistent radix-sort buffers (keys double-buffer, the static
+        entity-id pattern, output postings, rocprim workspace)."""
+        if getattr(self, "_sb_key", None) != (n, E):
+            tb = int(self.C.radix_sort_pairs_temp_bytes(n))
+            dev = self.device
+            self._sb = (
+                torch.empty(n, dtype=torch.int64, device=dev),
+                torch.empty(n, dtype=torch.int64, device=dev),
+                (torch.arange(n, device=dev, dtype=torch.int64) % E).to(torch.int32),
+                torch.empty(n, dtype=torch.int32, device=dev),
+                torch.empty(tb, dtype=torch.uint8, device=dev),
+            )
+            self._sb_key = (n, E)
+        return self._sb
+
     def _hist_add(self, ent_part, ent_values, reset=False):
         """Accumulate (partition, slot, value) posting counts for a subset of
         entities — the overlapped-migration prebuild of the next sweep's
@@ -519,13 +535,17 @@ class GpuEngine(CpuEngine):
                     # heavy sampler draws postings by segment INDEX, so the
                     # order within each key must be deterministic: sort
                     # (key * E + entity) — same boundaries as the prefix,
-                    # entities ascending within each segment
-                    ekeys = torch.empty(T * E, dtype=torch.int64, device=dev)
+                    # entities ascending within each segment. rocprim with a
+                    # persistent workspace (torch.sort's allocator-managed
+                    # workspace faults under hipGraph replay).
+                    ekeys, keys_out, vals_in, postings, temp = self._sort_bufs(
+                        T * E, E)
                     self.C.build_ekeys_stable(gs.ent_part, gs.ent_values,
                                               self._pair_a1, self._pair_a2,
                                               self._pair_v2, vmax, ekeys)
-                    perm = torch.argsort(ekeys)
-                    postings = (perm % E).to(torch.int32)
+                    end_bit = min(64, int(nk * E).bit_length())
+                    self.C.radix_sort_pairs_i64_i32(ekeys, keys_out, vals_in,
+                                                    postings, end_bit, temp)
                 else:
                     self._idx_cursor.copy_(self._idx_ptr[:-1])
                     postings = torch.empty(T * E, dtype=torch.int32, device=dev)

@@ -41,6 +41,10 @@ void classify_modes(torch::Tensor rec_values, torch::Tensor rec_dist,
 void build_ekeys_stable(torch::Tensor ent_part, torch::Tensor ent_values,
                         torch::Tensor pair_a1, torch::Tensor pair_a2,
                         torch::Tensor pair_v2, int64_t Vmax, torch::Tensor ekeys);
+int64_t radix_sort_pairs_temp_bytes(int64_t n);
+void radix_sort_pairs_i64_i32(torch::Tensor keys_in, torch::Tensor keys_out,
+                              torch::Tensor vals_in, torch::Tensor vals_out,
+                              int64_t end_bit, torch::Tensor temp);
 void link_update_heavy(torch::Tensor mode, torch::Tensor rec_values,
                        torch::Tensor rec_dist, torch::Tensor rec_gid,
                        torch::Tensor rec_part, torch::Tensor ent_values,
@@ -133,6 +137,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "route records to the wave / thread / hierarchical link paths");
   m.def("build_ekeys_stable", &dblink::build_ekeys_stable,
         "stable inverted-index sort keys (key * E + entity)");
+  m.def("radix_sort_pairs_temp_bytes", &dblink::radix_sort_pairs_temp_bytes,
+        "rocprim radix-sort workspace size");
+  m.def("radix_sort_pairs_i64_i32", &dblink::radix_sort_pairs_i64_i32,
+        "hipGraph-safe rocprim radix sort (persistent workspace)");
   m.def("link_update_heavy", &dblink::link_update_heavy,
         "hierarchical (A*) Gumbel-max link update for huge candidate sets");
   m.def("summary_counts", &dblink::summary_counts, "fused summary counts + pack");

@@ -22,6 +22,7 @@
 
 #include <ATen/cuda/CUDAContext.h>
 #include <hip/hip_runtime.h>
+#include <rocprim/device/device_radix_sort.hpp>
 #include <torch/extension.h>
 
 #include "common.h"
@@ -1808,6 +1809,33 @@ __global__ void build_ekeys_stable_kernel(
   const int64_t v = posting_slot_value(ent_values, e, A, t, pair_a1, pair_a2, pair_v2);
   const int64_t key = ((int64_t)ent_part[e] * T + t) * Vmax + v;
   ekeys[idx] = key * E + e;
+}
+
+// hipGraph-safe radix sort: torch.sort allocates its workspace through the
+// caching allocator, which faults on replay when capture records multi-GB
+// pools (the r01 ">4M records" graph bug). These wrappers let the engine own
+// ONE persistent workspace and keep every launch on the capture stream.
+int64_t radix_sort_pairs_temp_bytes(int64_t n) {
+  size_t bytes = 0;
+  rocprim::radix_sort_pairs(
+      nullptr, bytes, (const int64_t*)nullptr, (int64_t*)nullptr,
+      (const int32_t*)nullptr, (int32_t*)nullptr, (size_t)n);
+  return (int64_t)bytes;
+}
+
+void radix_sort_pairs_i64_i32(torch::Tensor keys_in, torch::Tensor keys_out,
+                              torch::Tensor vals_in, torch::Tensor vals_out,
+                              int64_t end_bit, torch::Tensor temp) {
+  const size_t n = (size_t)keys_in.numel();
+  if (n == 0) return;
+  size_t bytes = (size_t)temp.numel();
+  hipError_t err = rocprim::radix_sort_pairs(
+      temp.data_ptr(), bytes, keys_in.data_ptr<int64_t>(),
+      keys_out.data_ptr<int64_t>(), vals_in.data_ptr<int32_t>(),
+      vals_out.data_ptr<int32_t>(), n, 0, (unsigned)end_bit,
+      at::cuda::getCurrentCUDAStream());
+  TORCH_CHECK(err == hipSuccess, "radix_sort_pairs failed: ",
+              hipGetErrorString(err));
 }
 
 void build_ekeys_stable(torch::Tensor ent_part, torch::Tensor ent_values,
