@@ -531,7 +531,8 @@ class GpuEngine(CpuEngine):
                                          self._idx_counts)
                 torch.cumsum(self._idx_counts, 0, dtype=torch.int64,
                              out=self._idx_ptr[1:])
-                if self._heavy_thresh > 0:
+                if self._heavy_thresh > 0 or os.environ.get(
+                        "DBLINK_FORCE_SORT", "") == "1":
                     # heavy sampler draws postings by segment INDEX, so the
                     # order within each key must be deterministic: sort
                     # (key * E + entity) — same boundaries as the prefix,
