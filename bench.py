@@ -221,8 +221,9 @@ def main():
 
     if rank == 0 and getattr(engine, "_heavy_stats", None) is not None:
         hs = engine._heavy_stats.cpu().numpy()
-        if hs[0]:
-            print(f"[bench] heavy link sampler: {int(hs[0])} A* iterations, "
+        if hs[3]:
+            print(f"[bench] heavy link sampler: {int(hs[3])} records, "
+                  f"{int(hs[2])} similar-set visits, {int(hs[0])} A* iterations, "
                   f"{int(hs[1])} full-scan fallbacks (all sweeps)", file=sys.stderr)
 
     phase_ms = None

@@ -299,7 +299,13 @@ class GpuEngine(CpuEngine):
             int(os.environ.get("DBLINK_HEAVY_THRESH", "512"))
             if os.environ.get("DBLINK_HEAVY", "1") != "0" else 0
         )
-        self._heavy_stats = torch.zeros(2, dtype=torch.int64, device=device)
+        # the hierarchical sampler pays when candidate scans dominate (big
+        # partitions); hipGraphs pay when kernel launches dominate (small,
+        # sub-ms sweeps). Stream-workspace sorts (rocprim/torch) fault under
+        # hipGraph replay at scale, so the two are mutually exclusive: heavy
+        # routing engages above this record count and turns graphs off.
+        self._heavy_min_records = int(os.environ.get("DBLINK_HEAVY_MIN", "50000"))
+        self._heavy_stats = torch.zeros(4, dtype=torch.int64, device=device)
         # overlapped migration (migrants-only async all-to-all with the
         # posting build and summary readback hidden under it) is the default
         # multi-rank path; DBLINK_OVERLAP=0 selects the eager reference path
@@ -320,6 +326,9 @@ class GpuEngine(CpuEngine):
         elif zero:
             self._idx_counts.zero_()
         return nk
+
+    def _heavy_active(self, R):
+        return self._heavy_thresh > 0 and R >= self._heavy_min_records
 
     def _sort_bufs(self, n, E):
         """Persistent radix-sort buffers (keys double-buffer, the static
@@ -391,11 +400,14 @@ class GpuEngine(CpuEngine):
                 self._gs_alt = GpuStateTensors.like(gs)
             out = self._gs_alt
             # graph replay saves ~5 us x ~100 launches; at multi-million-record
-            # scales a sweep is tens of ms (capture of the multi-GB workspace
-            # pools also proved fragile there), so capture only where the
+            # scales a sweep is tens of ms and the heavy-path sort is not
+            # graph-safe (rocprim/torch stream-workspace sorts fault under
+            # replay — the r01 ">4M records" fault), so capture only where
             # launch overhead actually shows
-            use_graphs = self._graphs_enabled and gs.R <= int(
-                os.environ.get("DBLINK_GRAPH_MAX_RECORDS", "4000000")
+            use_graphs = (
+                self._graphs_enabled
+                and gs.R <= int(os.environ.get("DBLINK_GRAPH_MAX_RECORDS", "4000000"))
+                and not self._heavy_active(gs.R)
             )
             if use_graphs:
                 gkey = (key, self._flip)
@@ -531,8 +543,8 @@ class GpuEngine(CpuEngine):
                                          self._idx_counts)
                 torch.cumsum(self._idx_counts, 0, dtype=torch.int64,
                              out=self._idx_ptr[1:])
-                if self._heavy_thresh > 0 or os.environ.get(
-                        "DBLINK_FORCE_SORT", "") == "1":
+                if self._heavy_active(R) or \
+                        os.environ.get("DBLINK_FORCE_SORT", "") == "1":
                     # heavy sampler draws postings by segment INDEX, so the
                     # order within each key must be deterministic: sort
                     # (key * E + entity) — same boundaries as the prefix,
@@ -586,7 +598,8 @@ class GpuEngine(CpuEngine):
             # mode 0 wave scan, mode 1 thread scan (short candidate lists),
             # mode 2 hierarchical A* sampler (huge partition-slice candidate
             # sets — the stationary high-distortion regime)
-            heavy_th = self._heavy_thresh if dense_idx else 0
+            heavy_th = self._heavy_thresh if (
+                dense_idx and self._heavy_active(R)) else 0
             mode_mask = torch.empty(R, dtype=torch.uint8, device=dev)
             self.C.classify_modes(gs.rec_values, gs.rec_dist, gs.rec_part,
                                   ent_ptr, cand_lo, cand_hi, self._num_pairs,

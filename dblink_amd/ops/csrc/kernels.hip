@@ -290,7 +290,8 @@ __global__ void link_update_heavy_kernel(
     uint64_t seed, uint32_t iteration, const int64_t* __restrict__ ctrl,
     int64_t* __restrict__ rec_ent_out, const int64_t* __restrict__ rec_ent_in,
     int* __restrict__ error_count,
-    unsigned long long* __restrict__ stats) { // [2] = {astar_iters, fallbacks} or null
+    unsigned long long* __restrict__ stats) {
+  // stats[4] (or null): {A* iterations, fallbacks, S_r visits, heavy records}
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
   const int64_t r = (int64_t)blockIdx.x * HEAVY_WAVES + wid;
@@ -341,10 +342,12 @@ __global__ void link_update_heavy_kernel(
       const int64_t myj = cbase + lane;
       long long lo = 0;
       int n = 0;
+      float simv = 0.0f;
       if (myj < j1) {
         const int64_t key = ((int64_t)p * T + a) * Vmax + csr_col[myj];
         lo = idx_ptr[key];
         n = (int)(idx_ptr[key + 1] - lo);
+        simv = csr_sim[myj];
       }
       int pre = n;  // inclusive prefix of segment lengths across lanes
 #pragma unroll
@@ -366,6 +369,7 @@ __global__ void link_update_heavy_kernel(
           if (cand < WAVE && ce <= s) fl = cand;
         }
         const int64_t i = __shfl(lo, fl) + (s - __shfl(excl, fl));
+        const float dsim = __shfl(simv, fl);  // sim of the enumerated attr
         if (s >= total) continue;
         const int32_t e = postings[i];
         bool ok = true;
@@ -384,13 +388,16 @@ __global__ void link_update_heavy_kernel(
           mo &= mo - 1;
           const int32_t y = ent_values[(int64_t)e * A + ao];
           logw += log_norm[voff[ao] + y] +
-                  sim_lookup(csr_row_ptr, csr_col, csr_sim,
-                             voff[ao] + rec_values[r * A + ao], y);
+                  (ao == a ? dsim
+                           : sim_lookup(csr_row_ptr, csr_col, csr_sim,
+                                        voff[ao] + rec_values[r * A + ao], y));
         }
         const float g = gumbel_from_uniform(
             philox_uniform(seed, iteration, PH_LINK, gid, (uint32_t)e));
         if (logw + g > best_f) { best_f = logw + g; best_e = e; }
       }
+      if (lane == 0 && stats != nullptr)
+        atomicAdd(&stats[2], (unsigned long long)total);
     }
   }
   wave_argmax(best_f, best_e);
@@ -466,6 +473,7 @@ __global__ void link_update_heavy_kernel(
     if (stats != nullptr) {
       atomicAdd(&stats[0], iters);
       if (fell_back) atomicAdd(&stats[1], 1ull);
+      atomicAdd(&stats[3], 1ull);  // heavy records processed
     }
   }
   __syncthreads();
