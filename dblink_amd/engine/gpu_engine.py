@@ -133,6 +133,23 @@ class GpuModel:
             [1.0 if ia.is_constant else ia.index.sim_norm_total(1) for ia in attrs]
         )
 
+        # High-similarity sub-index for the hierarchical link sampler: only
+        # entries with sim >= tau are enumerated exactly; lower-sim entities
+        # ride in the A* complement under the proposal shift |od| * tau.
+        # Exact for ANY tau (A* needs only an upper bound) — tau trades the
+        # exact-scan size against A* iterations.
+        self.heavy_tau = float(os.environ.get("DBLINK_HEAVY_TAU", "1.5"))
+        # filter on the f32 values the kernel compares, so membership in the
+        # big index and the kernel's pool-rejection test agree exactly
+        big_mask = csr_sim.astype(np.float32) >= np.float32(self.heavy_tau)
+        big_counts = np.zeros(self.Vtot, dtype=np.int64)
+        rows_of = np.repeat(np.arange(self.Vtot), np.diff(row_ptr))
+        np.add.at(big_counts, rows_of[big_mask], 1)
+        row_ptr_big = np.zeros(self.Vtot + 1, dtype=np.int64)
+        row_ptr_big[1:] = np.cumsum(big_counts)
+        col_big = csr_col[big_mask]
+        sim_big = csr_sim[big_mask]
+
         self_expsim = np.concatenate(
             [
                 np.full(
@@ -154,6 +171,9 @@ class GpuModel:
         self.csr_row_ptr = dev(row_ptr, torch.int64)
         self.csr_col = dev(csr_col, torch.int32)
         self.csr_sim = dev(csr_sim, torch.float32)
+        self.csr_row_ptr_big = dev(row_ptr_big, torch.int64)
+        self.csr_col_big = dev(col_big, torch.int32)
+        self.csr_sim_big = dev(sim_big, torch.float32)
         self.phi_prob = dev(phi_prob, torch.float32)
         self.phi_alias = dev(phi_alias, torch.int32)
         self.pow_prob = dev(np.concatenate(pow_prob) if pow_prob else np.empty(0), torch.float32)
@@ -616,8 +636,10 @@ class GpuEngine(CpuEngine):
                     mode_mask, gs.rec_values, gs.rec_dist, gs.rec_gid,
                     gs.rec_part, gs.ent_values, ent_ptr, m.log_norm, m.voff,
                     m.csr_row_ptr, m.csr_col, m.csr_sim, m.attr_const,
-                    postings, self._idx_ptr, vmax, self._num_pairs, seed, it,
-                    ctrl, rec_ent_new, gs.rec_ent, self._err, self._heavy_stats,
+                    m.csr_row_ptr_big, m.csr_col_big, m.csr_sim_big,
+                    m.heavy_tau, postings, self._idx_ptr, vmax,
+                    self._num_pairs, seed, it, ctrl, rec_ent_new, gs.rec_ent,
+                    self._err, self._heavy_stats,
                 )
         gs.rec_ent.copy_(rec_ent_new)
         self._mark("link", graph_safe)

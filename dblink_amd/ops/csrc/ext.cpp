@@ -51,7 +51,9 @@ void link_update_heavy(torch::Tensor mode, torch::Tensor rec_values,
                        torch::Tensor ent_ptr, torch::Tensor log_norm,
                        torch::Tensor voff, torch::Tensor csr_row_ptr,
                        torch::Tensor csr_col, torch::Tensor csr_sim,
-                       torch::Tensor attr_const, torch::Tensor postings,
+                       torch::Tensor attr_const, torch::Tensor csr_row_ptr_big,
+                       torch::Tensor csr_col_big, torch::Tensor csr_sim_big,
+                       double tau, torch::Tensor postings,
                        torch::Tensor idx_ptr, int64_t Vmax, int64_t NP,
                        int64_t seed, int64_t iteration, torch::Tensor ctrl,
                        torch::Tensor rec_ent_out, torch::Tensor rec_ent_in,

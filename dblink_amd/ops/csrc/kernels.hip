@@ -274,7 +274,8 @@ __global__ void link_update_small_kernel(
 // regardless of what the truncated chain did.
 // ---------------------------------------------------------------------------
 
-constexpr int HEAVY_EXC_CAP = 128;   // A* exclusions per record before fallback
+constexpr int HEAVY_EXC_CAP = 512;   // A* exclusions per record before fallback
+constexpr int HEAVY_HASH = 1024;     // LDS exclusion hash slots per wave
 constexpr int HEAVY_WAVES = 4;
 
 __global__ void link_update_heavy_kernel(
@@ -285,6 +286,9 @@ __global__ void link_update_heavy_kernel(
     const float* __restrict__ log_norm, const int64_t* __restrict__ voff,
     const int64_t* __restrict__ csr_row_ptr, const int32_t* __restrict__ csr_col,
     const float* __restrict__ csr_sim, const uint8_t* __restrict__ attr_const,
+    const int64_t* __restrict__ csr_row_ptr_big,
+    const int32_t* __restrict__ csr_col_big, const float* __restrict__ csr_sim_big,
+    float tau,
     const int32_t* __restrict__ postings, const int64_t* __restrict__ idx_ptr,
     int64_t Vmax, int NP, int64_t R, int A,
     uint64_t seed, uint32_t iteration, const int64_t* __restrict__ ctrl,
@@ -337,17 +341,17 @@ __global__ void link_update_heavy_kernel(
     const int a = __ffs(m) - 1;
     m &= m - 1;
     const int64_t row = voff[a] + rec_values[r * A + a];
-    const int64_t j0 = csr_row_ptr[row], j1 = csr_row_ptr[row + 1];
+    const int64_t j0 = csr_row_ptr_big[row], j1 = csr_row_ptr_big[row + 1];
     for (int64_t cbase = j0; cbase < j1; cbase += WAVE) {
       const int64_t myj = cbase + lane;
       long long lo = 0;
       int n = 0;
       float simv = 0.0f;
       if (myj < j1) {
-        const int64_t key = ((int64_t)p * T + a) * Vmax + csr_col[myj];
+        const int64_t key = ((int64_t)p * T + a) * Vmax + csr_col_big[myj];
         lo = idx_ptr[key];
         n = (int)(idx_ptr[key + 1] - lo);
-        simv = csr_sim[myj];
+        simv = csr_sim_big[myj];
       }
       int pre = n;  // inclusive prefix of segment lengths across lanes
 #pragma unroll
@@ -402,12 +406,18 @@ __global__ void link_update_heavy_kernel(
   }
   wave_argmax(best_f, best_e);
 
-  // ---- 2. A* over the complement (uniform proposal; lane 0 serial) -------
-  __shared__ int32_t excl_s[HEAVY_WAVES][HEAVY_EXC_CAP];
+  // ---- 2. A* over the complement (lane 0 serial) -------------------------
+  // Proposal: uniform with constant shift q = |od| * tau — every pool entity
+  // (all big-sims excluded) has score t + sum(small sims) + g <= q + g, so
+  // the successive proposal maxima q + G_k bound the remaining candidates.
+  __shared__ int32_t exclh_s[HEAVY_WAVES][HEAVY_HASH];
   __shared__ long long winner_s[HEAVY_WAVES];
   __shared__ int fell_back_s[HEAVY_WAVES];
+  int32_t* exclh = exclh_s[wid];
+  for (int i = lane; i < HEAVY_HASH; i += WAVE) exclh[i] = -1;
+  // wave-lockstep: lane 0 reads below see these LDS writes
   if (lane == 0) {
-    int32_t* excl = excl_s[wid];
+    const double q_shift = (double)tau * (double)__popc(od_mask);
     double best = (best_e >= 0) ? (double)best_f : -INFINITY;
     long long bE = best_e;
     double Z = (double)pn;
@@ -426,20 +436,25 @@ __global__ void link_update_heavy_kernel(
         first = false;
       }
       b = G;
-      if (G <= best) break;
-      // categorical (uniform) draw over pool \ excluded, by pool index
+      if (G + q_shift <= best) break;
+      // categorical (uniform) draw over pool \ excluded, by pool index;
+      // the single-writer LDS hash both tests and records exclusions
       int64_t ci = -1;
       for (int tries = 0; tries < 4 * HEAVY_EXC_CAP; ++tries) {
         const float u = philox_uniform(seed, iteration, PH_LINKH, gid, ctr++);
         int64_t cand = (int64_t)(u * (float)pn);
         if (cand >= pn) cand = pn - 1;
-        bool hit = false;
-        for (int k = 0; k < n_exc; ++k)
-          if (excl[k] == (int32_t)cand) { hit = true; break; }
-        if (!hit) { ci = cand; break; }
+        int h = (int)(((uint32_t)cand * 2654435761u) >> 16) & (HEAVY_HASH - 1);
+        bool fresh = false;
+        for (;;) {
+          const int32_t cur = exclh[h];
+          if (cur == (int32_t)cand) break;          // already excluded
+          if (cur == -1) { exclh[h] = (int32_t)cand; fresh = true; break; }
+          h = (h + 1) & (HEAVY_HASH - 1);
+        }
+        if (fresh) { ci = cand; break; }
       }
-      if (ci < 0 || n_exc >= HEAVY_EXC_CAP) { fell_back = true; break; }
-      excl[n_exc++] = (int32_t)ci;
+      if (ci < 0 || ++n_exc > HEAVY_EXC_CAP) { fell_back = true; break; }
       Z -= 1.0;
       ++iters;
       const int32_t e = seg_pool ? postings[plo + ci] : (int32_t)(plo + ci);
@@ -454,18 +469,18 @@ __global__ void link_update_heavy_kernel(
       }
       if (!ok) continue;
       double t = 0.0;
-      bool similar = false;
+      bool in_big = false;
       for (uint32_t mo = od_mask; mo;) {
         const int ao = __ffs(mo) - 1;
         mo &= mo - 1;
         const int32_t y = ent_values[(int64_t)e * A + ao];
         const float s = sim_lookup(csr_row_ptr, csr_col, csr_sim,
                                    voff[ao] + rec_values[r * A + ao], y);
-        if (s != 0.0f) { similar = true; break; }  // scored exactly in S_r
-        t += (double)log_norm[voff[ao] + y];
+        if (s >= tau) { in_big = true; break; }  // scored exactly in S_r
+        t += (double)(log_norm[voff[ao] + y] + s);
       }
-      if (similar) continue;
-      const double s_true = t + G;  // uniform proposal: gumbel realization == G
+      if (in_big) continue;
+      const double s_true = t + G;  // gumbel realization of this draw == G
       if (s_true > best) { best = s_true; bE = e; }
     }
     winner_s[wid] = bE;
@@ -1869,7 +1884,9 @@ void link_update_heavy(
     torch::Tensor rec_gid, torch::Tensor rec_part, torch::Tensor ent_values,
     torch::Tensor ent_ptr, torch::Tensor log_norm, torch::Tensor voff,
     torch::Tensor csr_row_ptr, torch::Tensor csr_col, torch::Tensor csr_sim,
-    torch::Tensor attr_const, torch::Tensor postings, torch::Tensor idx_ptr,
+    torch::Tensor attr_const, torch::Tensor csr_row_ptr_big,
+    torch::Tensor csr_col_big, torch::Tensor csr_sim_big, double tau,
+    torch::Tensor postings, torch::Tensor idx_ptr,
     int64_t Vmax, int64_t NP, int64_t seed, int64_t iteration, torch::Tensor ctrl,
     torch::Tensor rec_ent_out, torch::Tensor rec_ent_in,
     torch::Tensor error_count, torch::Tensor stats) {
@@ -1886,7 +1903,10 @@ void link_update_heavy(
                      ent_ptr.data_ptr<int64_t>(), log_norm.data_ptr<float>(),
                      voff.data_ptr<int64_t>(), csr_row_ptr.data_ptr<int64_t>(),
                      csr_col.data_ptr<int32_t>(), csr_sim.data_ptr<float>(),
-                     attr_const.data_ptr<uint8_t>(), postings.data_ptr<int32_t>(),
+                     attr_const.data_ptr<uint8_t>(),
+                     csr_row_ptr_big.data_ptr<int64_t>(),
+                     csr_col_big.data_ptr<int32_t>(), csr_sim_big.data_ptr<float>(),
+                     (float)tau, postings.data_ptr<int32_t>(),
                      idx_ptr.data_ptr<int64_t>(), Vmax, (int)NP, R, A,
                      (uint64_t)seed, (uint32_t)iteration, ctrl_ptr,
                      rec_ent_out.data_ptr<int64_t>(), rec_ent_in.data_ptr<int64_t>(),

@@ -884,13 +884,15 @@ def test_link_heavy_kernel_distribution():
 
         out = torch.empty(N, dtype=torch.int64, device=DEV)
         err = torch.zeros(1, dtype=torch.int32, device=DEV)
-        stats = torch.zeros(2, dtype=torch.int64, device=DEV)
+        stats = torch.zeros(4, dtype=torch.int64, device=DEV)
         C.link_update_heavy(
             mode, rv, rd,
             _dev(np.arange(N, dtype=np.int64), torch.int64), rec_part,
             _dev(ent_vals, torch.int32), ent_ptr,
             model.log_norm, model.voff, model.csr_row_ptr, model.csr_col,
             model.csr_sim, model.attr_const,
+            model.csr_row_ptr_big, model.csr_col_big, model.csr_sim_big,
+            model.heavy_tau,
             _dev(postings, torch.int32), _dev(idx_ptr, torch.int64),
             Vmax, 0, 4321, 11, torch.empty(0, dtype=torch.int64, device=DEV),
             out, _dev(np.zeros(N, np.int64), torch.int64), err, stats)
