@@ -138,7 +138,7 @@ class GpuModel:
         # ride in the A* complement under the proposal shift |od| * tau.
         # Exact for ANY tau (A* needs only an upper bound) — tau trades the
         # exact-scan size against A* iterations.
-        self.heavy_tau = float(os.environ.get("DBLINK_HEAVY_TAU", "1.5"))
+        self.heavy_tau = float(os.environ.get("DBLINK_HEAVY_TAU", "1.0"))
         # filter on the f32 values the kernel compares, so membership in the
         # big index and the kernel's pool-rejection test agree exactly
         big_mask = csr_sim.astype(np.float32) >= np.float32(self.heavy_tau)
