@@ -787,11 +787,17 @@ __global__ void value_update_k1_kernel(ValueArgs args) {
 
 constexpr int VAL_STRIDE = 8;  // pairs examined per wave in kobs mode
 
+// VMODE selects which k >= 2 perturbation path a kernel instantiation
+// carries: 0 = both (explicit pair lists / tests), 1 = LDS-hash only,
+// 2 = union-merge only. The split halves the register pressure of the hot
+// kobs-mode kernels (the monolithic kernel sat at 118 VGPRs = 4 waves/SIMD).
+template <int VMODE>
 __device__ void value_update_pair(const ValueArgs& args, int64_t pair, int lane,
                                   int32_t* keys, float* vals);
 
+template <int VMODE>
 __global__ void __launch_bounds__(WAVES_PER_BLOCK_VAL * WAVE)
-value_update_kernel(ValueArgs args) {
+value_update_kernel_t(ValueArgs args) {
   if (args.ctrl != nullptr) {
     args.seed = (uint64_t)args.ctrl[0];
     args.iteration = (uint32_t)args.ctrl[1];
@@ -804,17 +810,19 @@ value_update_kernel(ValueArgs args) {
   const int64_t widx = (int64_t)blockIdx.x * WAVES_PER_BLOCK_VAL + wave;
   if (args.pair_list != nullptr) {
     if (widx >= args.n_pairs) return;
-    value_update_pair(args, args.pair_list[widx], lane, h_key[wave], h_val[wave]);
+    value_update_pair<VMODE>(args, args.pair_list[widx], lane, h_key[wave], h_val[wave]);
   } else {
     // kobs self-selection: each wave examines VAL_STRIDE consecutive pairs
     // and runs the (rare) k >= 2 ones serially
     const int64_t p0 = widx * VAL_STRIDE;
     for (int64_t pair = p0; pair < p0 + VAL_STRIDE && pair < args.n_pairs; ++pair) {
-      if (args.kobs[pair] >= 2) value_update_pair(args, pair, lane, h_key[wave], h_val[wave]);
+      if (args.kobs[pair] >= 2)
+        value_update_pair<VMODE>(args, pair, lane, h_key[wave], h_val[wave]);
     }
   }
 }
 
+template <int VMODE>
 __device__ void value_update_pair(const ValueArgs& args, int64_t pair, int lane,
                                   int32_t* keys, float* vals) {
   const int64_t e = pair / args.A;
@@ -962,6 +970,8 @@ __device__ void value_update_pair(const ValueArgs& args, int64_t pair, int lane,
     total_entries = k_obs < V ? k_obs : V;
   }
   const bool dense = total_entries > (HASH_CAP * 3) / 4;
+  if (VMODE == 1 && dense) return;   // the merge kernel's pair
+  if (VMODE == 2 && !dense) return;  // the hash kernel's pair
   int tsize = 64;
   while (tsize < 2 * (int)total_entries && tsize < HASH_CAP) tsize <<= 1;
 
@@ -969,7 +979,7 @@ __device__ void value_update_pair(const ValueArgs& args, int64_t pair, int lane,
   float best = -INFINITY;    // gumbel-max over perturbation weights
   long long best_v = -1;
 
-  if (!dense) {
+  if (VMODE != 2 && !dense) {
     for (int i = lane; i < tsize; i += WAVE) { keys[i] = -1; vals[i] = 0.0f; }
     // drain LDS writes before other lanes' atomics may touch the slots
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -1025,7 +1035,7 @@ __device__ void value_update_pair(const ValueArgs& args, int64_t pair, int lane,
           philox_uniform(args.seed, args.iteration, PH_VALG, elem, (uint32_t)v));
       if (logw + g > best) { best = logw + g; best_v = v; }
     }
-  } else {
+  } else if (VMODE != 1) {
     // union-merge path (hash would overflow): iterate every row entry, but
     // process a value only from the FIRST row containing it; full L_v comes
     // from binary searches in the other rows. O(entries * k log row), far
@@ -2047,7 +2057,8 @@ void value_update(
   }
   if (kobs.numel() > 0) {
     // kobs self-selection: no host-side pair lists, no stream sync — the
-    // three kernels each cover every pair and act only on their class
+    // four kernels each cover every pair and act only on their class
+    // (k=0 base, k=1, k>=2 hash, k>=2 union-merge)
     args.kobs = kobs.data_ptr<int32_t>();
     args.n_pairs = kobs.numel();
     dim3 tgrid((unsigned)((args.n_pairs + 255) / 256));
@@ -2057,7 +2068,11 @@ void value_update(
                        at::cuda::getCurrentCUDAStream(), args);
     const int64_t n_waves = (args.n_pairs + VAL_STRIDE - 1) / VAL_STRIDE;
     dim3 wgrid((unsigned)wave_grid(n_waves, WAVES_PER_BLOCK_VAL));
-    hipLaunchKernelGGL(value_update_kernel, wgrid, dim3(WAVES_PER_BLOCK_VAL * WAVE), 0,
+    hipLaunchKernelGGL(value_update_kernel_t<1>, wgrid,
+                       dim3(WAVES_PER_BLOCK_VAL * WAVE), 0,
+                       at::cuda::getCurrentCUDAStream(), args);
+    hipLaunchKernelGGL(value_update_kernel_t<2>, wgrid,
+                       dim3(WAVES_PER_BLOCK_VAL * WAVE), 0,
                        at::cuda::getCurrentCUDAStream(), args);
     return;
   }
@@ -2080,7 +2095,7 @@ void value_update(
     args.pair_list = wave_pairs.data_ptr<int64_t>();
     args.n_pairs = wave_pairs.numel();
     dim3 grid((unsigned)wave_grid(args.n_pairs, WAVES_PER_BLOCK_VAL));
-    hipLaunchKernelGGL(value_update_kernel, grid, dim3(WAVES_PER_BLOCK_VAL * WAVE), 0,
+    hipLaunchKernelGGL(value_update_kernel_t<0>, grid, dim3(WAVES_PER_BLOCK_VAL * WAVE), 0,
                        at::cuda::getCurrentCUDAStream(), args);
   }
 }
