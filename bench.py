@@ -226,6 +226,13 @@ def main():
                   f"{int(hs[2])} similar-set visits, {int(hs[0])} A* iterations, "
                   f"{int(hs[1])} full-scan fallbacks (all sweeps)", file=sys.stderr)
 
+    if rank == 0 and getattr(engine, "_value_stats", None) is not None:
+        vs = engine._value_stats.cpu().numpy()
+        print(f"[bench] value k>=2: {int(vs[0])} pairs ({int(vs[1])} hash / "
+              f"{int(vs[2])} merge / {int(vs[6])} k>Kc), sum_entries={int(vs[3])} "
+              f"(merge {int(vs[7])}), sum_tsize={int(vs[4])}, sum_k={int(vs[5])} "
+              f"(all sweeps)", file=sys.stderr)
+
     phase_ms = None
     if rank == 0 and hasattr(engine, "phase_times") and getattr(engine, "phase_timers", False):
         pt = engine.phase_times()

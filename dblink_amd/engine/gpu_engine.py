@@ -326,6 +326,10 @@ class GpuEngine(CpuEngine):
         # routing engages above this record count and turns graphs off.
         self._heavy_min_records = int(os.environ.get("DBLINK_HEAVY_MIN", "50000"))
         self._heavy_stats = torch.zeros(4, dtype=torch.int64, device=device)
+        self._value_stats = None
+        if os.environ.get("DBLINK_VALUE_STATS", "") == "1":
+            self._value_stats = torch.zeros(8, dtype=torch.int64, device=device)
+            self.C.set_value_stats(self._value_stats)
         # overlapped migration (migrants-only async all-to-all with the
         # posting build and summary readback hidden under it) is the default
         # multi-rank path; DBLINK_OVERLAP=0 selects the eager reference path

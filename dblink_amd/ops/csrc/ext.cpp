@@ -104,6 +104,7 @@ void build_keys(torch::Tensor ent_part, torch::Tensor ent_values,
                 torch::Tensor pair_a1, torch::Tensor pair_a2,
                 torch::Tensor pair_v2, int64_t Vmax,
                 torch::Tensor ekeys, torch::Tensor qkeys);
+void set_value_stats(torch::Tensor t);
 void summary_counts(torch::Tensor rec_dist, torch::Tensor rec_file,
                     torch::Tensor ent_rec_ptr, int64_t E, torch::Tensor counts,
                     torch::Tensor loglik, torch::Tensor packed);
@@ -146,4 +147,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("link_update_heavy", &dblink::link_update_heavy,
         "hierarchical (A*) Gumbel-max link update for huge candidate sets");
   m.def("summary_counts", &dblink::summary_counts, "fused summary counts + pack");
+  m.def("set_value_stats", &dblink::set_value_stats,
+        "install the optional value-phase work-counter buffer");
 }

@@ -641,6 +641,9 @@ struct ValueArgs {
   uint32_t iteration;
   uint64_t ent_id_base;
   int* error_count;
+  // optional [8]: {k2_pairs, hash_pairs, merge_pairs, sum_entries,
+  //                sum_tsize, sum_kobs, rare_pairs, merge_entries}
+  unsigned long long* stats;
 };
 
 // Draw from p(v) ~ phi(v)*norm(v)^k by a dense Gumbel scan (rare path for
@@ -974,6 +977,16 @@ __device__ void value_update_pair(const ValueArgs& args, int64_t pair, int lane,
   if (VMODE == 2 && !dense) return;  // the hash kernel's pair
   int tsize = 64;
   while (tsize < 2 * (int)total_entries && tsize < HASH_CAP) tsize <<= 1;
+  if (args.stats != nullptr && lane == 0) {
+    atomicAdd(&args.stats[0], 1ull);
+    atomicAdd(&args.stats[dense ? 2 : 1], 1ull);
+    atomicAdd(&args.stats[3], (unsigned long long)total_entries);
+    atomicAdd(&args.stats[4], (unsigned long long)(dense ? 0 : tsize));
+    atomicAdd(&args.stats[5], (unsigned long long)k_obs);
+    if (!is_const && k_obs > args.Kc) atomicAdd(&args.stats[6], 1ull);
+    if (dense)
+      atomicAdd(&args.stats[7], (unsigned long long)total_entries);
+  }
 
   double W = 0.0;            // total perturbation weight
   float best = -INFINITY;    // gumbel-max over perturbation weights
@@ -1973,6 +1986,21 @@ void link_update_dense(
                      rec_ent_out.data_ptr<int64_t>());
 }
 
+// Opt-in value-phase work counters (DBLink_VALUE_STATS): set once from
+// python; the tensor is held so the device pointer stays alive.
+static unsigned long long* g_value_stats = nullptr;
+static torch::Tensor g_value_stats_keep;
+
+void set_value_stats(torch::Tensor t) {
+  if (t.numel()) {
+    g_value_stats_keep = t;
+    g_value_stats = (unsigned long long*)t.data_ptr<int64_t>();
+  } else {
+    g_value_stats = nullptr;
+    g_value_stats_keep = torch::Tensor();
+  }
+}
+
 static ValueArgs make_value_args(
     torch::Tensor rec_values, torch::Tensor rec_dist, torch::Tensor rec_file,
     torch::Tensor ent_rec_ptr, torch::Tensor ent_rec_idx, torch::Tensor ent_values,
@@ -1988,6 +2016,7 @@ static ValueArgs make_value_args(
   a.ctrl = nullptr;
   a.kobs = nullptr;
   a.pair_list = nullptr;
+  a.stats = g_value_stats;
   a.csr_excl = csr_excl.numel() ? csr_excl.data_ptr<double>() : nullptr;
   a.csr_rawsum = csr_rawsum.data_ptr<double>();
   a.z1 = z1.data_ptr<double>();
