@@ -602,6 +602,7 @@ __global__ void link_update_dense_kernel(
 // LDS hash table per wave for sparse perturbation weights.
 constexpr int HASH_CAP = 1024;            // slots per wave
 constexpr int WAVES_PER_BLOCK_VAL = 4;    // 256 threads
+constexpr int VAL_DMAX = 16;              // distinct (value, file) groups per pair
 
 struct ValueArgs {
   const int64_t* ctrl;         // [2] = {seed, iteration} device override (or null)
@@ -796,7 +797,7 @@ constexpr int VAL_STRIDE = 8;  // pairs examined per wave in kobs mode
 // kobs-mode kernels (the monolithic kernel sat at 118 VGPRs = 4 waves/SIMD).
 template <int VMODE>
 __device__ void value_update_pair(const ValueArgs& args, int64_t pair, int lane,
-                                  int32_t* keys, float* vals);
+                                  int32_t* keys, float* vals, int32_t* gbuf);
 
 template <int VMODE>
 __global__ void __launch_bounds__(WAVES_PER_BLOCK_VAL * WAVE)
@@ -807,27 +808,30 @@ value_update_kernel_t(ValueArgs args) {
   }
   __shared__ int32_t h_key[WAVES_PER_BLOCK_VAL][HASH_CAP];
   __shared__ float h_val[WAVES_PER_BLOCK_VAL][HASH_CAP];
+  __shared__ int32_t g_buf[WAVES_PER_BLOCK_VAL][3 * VAL_DMAX + 1];
 
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave = threadIdx.x / WAVE;
   const int64_t widx = (int64_t)blockIdx.x * WAVES_PER_BLOCK_VAL + wave;
   if (args.pair_list != nullptr) {
     if (widx >= args.n_pairs) return;
-    value_update_pair<VMODE>(args, args.pair_list[widx], lane, h_key[wave], h_val[wave]);
+    value_update_pair<VMODE>(args, args.pair_list[widx], lane, h_key[wave],
+                             h_val[wave], g_buf[wave]);
   } else {
     // kobs self-selection: each wave examines VAL_STRIDE consecutive pairs
     // and runs the (rare) k >= 2 ones serially
     const int64_t p0 = widx * VAL_STRIDE;
     for (int64_t pair = p0; pair < p0 + VAL_STRIDE && pair < args.n_pairs; ++pair) {
       if (args.kobs[pair] >= 2)
-        value_update_pair<VMODE>(args, pair, lane, h_key[wave], h_val[wave]);
+        value_update_pair<VMODE>(args, pair, lane, h_key[wave], h_val[wave],
+                                 g_buf[wave]);
     }
   }
 }
 
 template <int VMODE>
 __device__ void value_update_pair(const ValueArgs& args, int64_t pair, int lane,
-                                  int32_t* keys, float* vals) {
+                                  int32_t* keys, float* vals, int32_t* gbuf) {
   const int64_t e = pair / args.A;
   const int a = (int)(pair % args.A);
   const bool is_const = args.attr_const[a];
@@ -957,20 +961,83 @@ __device__ void value_update_pair(const ValueArgs& args, int64_t pair, int lane,
     return;
   }
 
-  // ---- perturbation weights in the LDS hash table (k >= 2) ---------------
-  // L_v = sum_r log f_r(v); final weight = base_prob(v) * (exp(L_v) - 1).
-
-  // size the table (and its clear/scan cost) to the actual support
-  int64_t total_entries = 0;
-  if (!is_const) {
+  // ---- perturbation weights (k >= 2): grouped by distinct value ----------
+  // A cluster's linked records mostly agree, so group them by distinct
+  // (value, file): m records sharing x contribute the factor (e^s + se)^m,
+  // i.e. one row pass with a multiplier instead of m passes — the union
+  // support is sum of DISTINCT rows, not k rows. L_v = sum_g m_g log f_g(v);
+  // final weight = base_prob(v) * (exp(L_v) - 1). More than VAL_DMAX
+  // distinct groups (rare) falls back to per-record units.
+  int d = 0;
+  bool gover = false;
+  if (lane == 0) {
     for (int64_t i = r_lo; i < r_hi; ++i) {
       const int64_t r = args.ent_rec_idx[i];
       const int32_t x = args.rec_values[r * args.A + a];
       if (x < 0) continue;
-      total_entries += args.csr_row_ptr[v0 + x + 1] - args.csr_row_ptr[v0 + x];
+      const int32_t f = args.collapsed ? args.rec_file[r] : 0;
+      int g = 0;
+      for (; g < d; ++g)
+        if (gbuf[g] == x && gbuf[VAL_DMAX + g] == f) break;
+      if (g < d) {
+        gbuf[2 * VAL_DMAX + g] += 1;
+      } else if (d < VAL_DMAX) {
+        gbuf[d] = x;
+        gbuf[VAL_DMAX + d] = f;
+        gbuf[2 * VAL_DMAX + d] = 1;
+        ++d;
+      } else {
+        gover = true;
+        break;
+      }
     }
-  } else {
-    total_entries = k_obs < V ? k_obs : V;
+    gbuf[3 * VAL_DMAX] = gover ? -1 : d;
+  }
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_wave_barrier();
+  gover = gbuf[3 * VAL_DMAX] < 0;
+  d = gover ? 0 : gbuf[3 * VAL_DMAX];
+
+  // self-extra for a group (or record unit): the collapsed (1/theta - 1)
+  // correction, constant within a (value, file) group
+  auto se_of = [&](int32_t x, int32_t f) -> float {
+    if (!args.collapsed) return 0.0f;
+    const float th = args.theta[a * args.F + f];
+    const float px = args.phi[v0 + x];
+    return (1.0f / th - 1.0f) / (is_const ? px : px * args.norm_lin[v0 + x]);
+  };
+  const int n_units = gover ? k_obs : d;
+  // unit u -> (x, multiplicity, file); gover scans the record list
+  auto unit_of = [&](int u, int32_t* x, int32_t* m_, int32_t* f) {
+    if (!gover) {
+      *x = gbuf[u];
+      *f = gbuf[VAL_DMAX + u];
+      *m_ = gbuf[2 * VAL_DMAX + u];
+      return;
+    }
+    int seen = 0;
+    for (int64_t i = r_lo; i < r_hi; ++i) {
+      const int64_t r = args.ent_rec_idx[i];
+      const int32_t xx = args.rec_values[r * args.A + a];
+      if (xx < 0) continue;
+      if (seen++ == u) {
+        *x = xx;
+        *m_ = 1;
+        *f = args.collapsed ? args.rec_file[r] : 0;
+        return;
+      }
+    }
+  };
+
+  // size the table (and its clear/scan cost) to the actual support
+  int64_t total_entries = 0;
+  for (int u = 0; u < n_units; ++u) {
+    int32_t ux, um, uf;
+    unit_of(u, &ux, &um, &uf);
+    if (is_const)
+      total_entries += 1;
+    else
+      total_entries += args.csr_row_ptr[v0 + ux + 1] - args.csr_row_ptr[v0 + ux];
   }
   const bool dense = total_entries > (HASH_CAP * 3) / 4;
   if (VMODE == 1 && dense) return;   // the merge kernel's pair
@@ -997,31 +1064,32 @@ __device__ void value_update_pair(const ValueArgs& args, int64_t pair, int lane,
     // drain LDS writes before other lanes' atomics may touch the slots
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_wave_barrier();
-    // hash-accumulate log factors
-    for (int64_t i = r_lo; i < r_hi; ++i) {
-      const int64_t r = args.ent_rec_idx[i];
-      const int32_t x = args.rec_values[r * args.A + a];
-      if (x < 0) continue;
-      const float self_extra = self_extra_of(r, x);
+    // hash-accumulate log factors per unit
+    for (int u = 0; u < n_units; ++u) {
+      int32_t ux, um, uf;
+      unit_of(u, &ux, &um, &uf);
+      const float self_extra = se_of(ux, uf);
+      const float fm = (float)um;
       if (is_const) {
-        // single-entry row {x}: factor = 1 + self_extra
+        // single-entry row {x}: factor = (1 + self_extra)^m
         if (lane == 0) {
-          float logf_ = __logf(1.0f + self_extra);
-          uint32_t h = ((uint32_t)x * 2654435761u) & (tsize - 1);
+          float logf_ = fm * __logf(1.0f + self_extra);
+          uint32_t h = ((uint32_t)ux * 2654435761u) & (tsize - 1);
           while (true) {
-            int32_t prev = atomicCAS(&keys[h], -1, x);
-            if (prev == -1 || prev == x) { atomicAdd(&vals[h], logf_); break; }
+            int32_t prev = atomicCAS(&keys[h], -1, ux);
+            if (prev == -1 || prev == ux) { atomicAdd(&vals[h], logf_); break; }
             h = (h + 1) & (tsize - 1);
           }
         }
       } else {
-        const int64_t row_lo = args.csr_row_ptr[v0 + x], row_hi = args.csr_row_ptr[v0 + x + 1];
+        const int64_t row_lo = args.csr_row_ptr[v0 + ux];
+        const int64_t row_hi = args.csr_row_ptr[v0 + ux + 1];
         for (int64_t j = row_lo + lane; j < row_hi; j += WAVE) {
           const int32_t v = args.csr_col[j];
           const float s = args.csr_sim[j];  // log expsim > 0
-          float factor_log = (v == x && self_extra > 0.0f)
-                                 ? __logf(__expf(s) + self_extra)
-                                 : s;
+          float factor_log = (v == ux && self_extra > 0.0f)
+                                 ? fm * __logf(__expf(s) + self_extra)
+                                 : fm * s;
           uint32_t h = ((uint32_t)v * 2654435761u) & (tsize - 1);
           while (true) {
             int32_t prev = atomicCAS(&keys[h], -1, v);
@@ -1049,70 +1117,68 @@ __device__ void value_update_pair(const ValueArgs& args, int64_t pair, int lane,
       if (logw + g > best) { best = logw + g; best_v = v; }
     }
   } else if (VMODE != 1) {
-    // union-merge path (hash would overflow): iterate every row entry, but
-    // process a value only from the FIRST row containing it; full L_v comes
-    // from binary searches in the other rows. O(entries * k log row), far
-    // cheaper than a dense domain scan.
-    for (int64_t i = r_lo; i < r_hi; ++i) {
-      const int64_t r = args.ent_rec_idx[i];
-      const int32_t x = args.rec_values[r * args.A + a];
-      if (x < 0) continue;
+    // union-merge path (hash would overflow): iterate every entry of every
+    // DISTINCT row, but process a value only from the FIRST unit containing
+    // it; full L_v comes from binary searches in the other units' rows.
+    for (int u = 0; u < n_units; ++u) {
+      int32_t ux, um, uf;
+      unit_of(u, &ux, &um, &uf);
       if (is_const) {
-        // "row" is the single value {x}: process on its first occurrence
         if (lane == 0) {
           bool first = true;
-          for (int64_t i2 = r_lo; i2 < i && first; ++i2) {
-            const int64_t r2 = args.ent_rec_idx[i2];
-            if (args.rec_values[r2 * args.A + a] == x) first = false;
+          for (int u2 = 0; u2 < u && first; ++u2) {
+            int32_t x2, m2, f2;
+            unit_of(u2, &x2, &m2, &f2);
+            if (x2 == ux) first = false;  // only possible when gover
           }
           if (first) {
             float L = 0.0f;
-            for (int64_t i2 = r_lo; i2 < r_hi; ++i2) {
-              const int64_t r2 = args.ent_rec_idx[i2];
-              if (args.rec_values[r2 * args.A + a] != x) continue;
-              L += __logf(1.0f + self_extra_of(r2, x));
+            for (int u2 = u; u2 < n_units; ++u2) {
+              int32_t x2, m2, f2;
+              unit_of(u2, &x2, &m2, &f2);
+              if (x2 != ux) continue;
+              L += (float)m2 * __logf(1.0f + se_of(x2, f2));
             }
             if (L > 0.0f) {
               const float log_expm1 = L + __logf(1.0f - __expf(-L));
-              const float logw = log_base_prob(x) + log_expm1;
+              const float logw = log_base_prob(ux) + log_expm1;
               W += (logw < 80.0f) ? (double)__expf(logw) : exp((double)logw);
               const float g = gumbel_from_uniform(philox_uniform(
-                  args.seed, args.iteration, PH_VALG, elem, (uint32_t)x));
-              if (logw + g > best) { best = logw + g; best_v = x; }
+                  args.seed, args.iteration, PH_VALG, elem, (uint32_t)ux));
+              if (logw + g > best) { best = logw + g; best_v = ux; }
             }
           }
         }
         continue;
       }
-      const int64_t row_lo = args.csr_row_ptr[v0 + x], row_hi = args.csr_row_ptr[v0 + x + 1];
+      const int64_t row_lo = args.csr_row_ptr[v0 + ux];
+      const int64_t row_hi = args.csr_row_ptr[v0 + ux + 1];
       for (int64_t j = row_lo + lane; j < row_hi; j += WAVE) {
         const int32_t v = args.csr_col[j];
-        // dedupe: skip if an earlier observed row already contains v
+        // dedupe: skip if an earlier unit's row already contains v
         bool first = true;
-        for (int64_t i2 = r_lo; i2 < i && first; ++i2) {
-          const int64_t r2 = args.ent_rec_idx[i2];
-          const int32_t x2 = args.rec_values[r2 * args.A + a];
-          if (x2 < 0) continue;
+        for (int u2 = 0; u2 < u && first; ++u2) {
+          int32_t x2, m2, f2;
+          unit_of(u2, &x2, &m2, &f2);
           const int64_t lo2 = args.csr_row_ptr[v0 + x2];
           const int64_t hi2 = args.csr_row_ptr[v0 + x2 + 1];
           if (contains_i32(args.csr_col, lo2, hi2, v)) first = false;
         }
         if (!first) continue;
-        // accumulate log factors over ALL observed rows
+        // accumulate log factors over ALL units
         float L = 0.0f;
-        for (int64_t i2 = r_lo; i2 < r_hi; ++i2) {
-          const int64_t r2 = args.ent_rec_idx[i2];
-          const int32_t x2 = args.rec_values[r2 * args.A + a];
-          if (x2 < 0) continue;
-          const float se2 = self_extra_of(r2, x2);
-          const float s2 = (i2 == i && x2 == x)
+        for (int u2 = 0; u2 < n_units; ++u2) {
+          int32_t x2, m2, f2;
+          unit_of(u2, &x2, &m2, &f2);
+          const float se2 = se_of(x2, f2);
+          const float s2 = (u2 == u)
                                ? args.csr_sim[j]
                                : sim_lookup(args.csr_row_ptr, args.csr_col,
                                             args.csr_sim, v0 + x2, v);
           if (v == x2 && se2 > 0.0f)
-            L += __logf(__expf(s2) + se2);
+            L += (float)m2 * __logf(__expf(s2) + se2);
           else if (s2 != 0.0f)
-            L += s2;
+            L += (float)m2 * s2;
         }
         const float log_expm1 = L + __logf(1.0f - __expf(-L));
         const float logw = log_base_prob(v) + log_expm1;
