@@ -105,6 +105,10 @@ void build_keys(torch::Tensor ent_part, torch::Tensor ent_values,
                 torch::Tensor pair_v2, int64_t Vmax,
                 torch::Tensor ekeys, torch::Tensor qkeys);
 void set_value_stats(torch::Tensor t);
+void mfma_score_bench(torch::Tensor rcode, torch::Tensor rbonus,
+                      torch::Tensor ecode, int64_t K, torch::Tensor score);
+void scalar_score_bench(torch::Tensor rcode, torch::Tensor rbonus,
+                        torch::Tensor ecode, torch::Tensor score);
 void summary_counts(torch::Tensor rec_dist, torch::Tensor rec_file,
                     torch::Tensor ent_rec_ptr, int64_t E, torch::Tensor counts,
                     torch::Tensor loglik, torch::Tensor packed);
@@ -149,4 +153,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("summary_counts", &dblink::summary_counts, "fused summary counts + pack");
   m.def("set_value_stats", &dblink::set_value_stats,
         "install the optional value-phase work-counter buffer");
+  m.def("mfma_score_bench", &dblink::mfma_score_bench,
+        "MFMA one-hot categorical scorer (experiment)");
+  m.def("scalar_score_bench", &dblink::scalar_score_bench,
+        "LDS scalar categorical scorer (experiment)");
 }

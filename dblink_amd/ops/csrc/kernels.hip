@@ -1652,6 +1652,124 @@ __global__ void summary_finalize_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// MFMA experiment: the most matrix-shaped hot computation is
+// the dense categorical-agreement block of the PCG-II link weights —
+// score[r, e] = sum_a bonus_a(x_{r,a}) * 1{x_{r,a} == y_{e,a}} over the
+// constant attributes. Two implementations of the same R x E scorer:
+//   scalar_score_bench: LDS-staged entity codes, wave-per-record compare/add
+//   mfma_score_bench:   one-hot formulation U[r,k] * H[k,e] on the bf16
+//                       matrix cores (mfma_f32_16x16x32_bf16, K padded to a
+//                       multiple of 32), fragments synthesized on the fly
+//                       from the compact codes (no dense one-hot in memory)
+// The measured comparison lives in profiles/README.md.
+// ---------------------------------------------------------------------------
+
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+DBL_D short f32_to_bf16(float f) {
+  union { float f; uint32_t u; } c = {f};
+  return (short)(c.u >> 16);  // truncate (bench precision is bf16 anyway)
+}
+
+// score tile per WAVE: 16 records x 16 entities, K-loop over one-hot domain
+__global__ void mfma_score_kernel(
+    const int32_t* __restrict__ rcode,   // [R, 3] concatenated-domain codes
+    const float* __restrict__ rbonus,    // [R, 3] per-record attr bonuses
+    const int32_t* __restrict__ ecode,   // [E, 3]
+    int R, int E, int K,                 // K = padded one-hot width
+    float* __restrict__ score) {         // [R, E]
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = threadIdx.x / WAVE;
+  const int tiles_e = E / 16;
+  const int tile = blockIdx.x * (blockDim.x / WAVE) + wave;
+  const int tr = tile / tiles_e, te = tile % tiles_e;
+  if (tr * 16 >= R) return;
+
+  // A: row i = lane & 15 (record), 8 elements at k = k0 + (lane>>4)*8 + j
+  // B: col j = lane & 15 (entity), same k layout
+  // C/D: col = lane & 15, row = (lane>>4)*4 + reg
+  const int r = tr * 16 + (lane & 15);
+  const int e = te * 16 + (lane & 15);
+  int32_t rc0 = rcode[r * 3], rc1 = rcode[r * 3 + 1], rc2 = rcode[r * 3 + 2];
+  float rb0 = rbonus[r * 3], rb1 = rbonus[r * 3 + 1], rb2 = rbonus[r * 3 + 2];
+  int32_t ec0 = ecode[e * 3], ec1 = ecode[e * 3 + 1], ec2 = ecode[e * 3 + 2];
+
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  for (int k0 = 0; k0 < K; k0 += 32) {
+    bf16x8 fa, fb;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int k = k0 + (lane >> 4) * 8 + j;
+      float av = (k == rc0) ? rb0 : (k == rc1) ? rb1 : (k == rc2) ? rb2 : 0.0f;
+      fa[j] = f32_to_bf16(av);
+      fb[j] = f32_to_bf16((k == ec0 || k == ec1 || k == ec2) ? 1.0f : 0.0f);
+    }
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(fa, fb, acc, 0, 0, 0);
+  }
+#pragma unroll
+  for (int reg = 0; reg < 4; ++reg) {
+    const int row = (lane >> 4) * 4 + reg;
+    score[(int64_t)(tr * 16 + row) * E + te * 16 + (lane & 15)] = acc[reg];
+  }
+}
+
+// Wave-per-record scalar scorer over LDS-staged packed entity codes.
+constexpr int SCALAR_ETILE = 1024;
+
+__global__ void scalar_score_kernel(
+    const int32_t* __restrict__ rcode, const float* __restrict__ rbonus,
+    const int32_t* __restrict__ ecode, int R, int E,
+    float* __restrict__ score) {
+  __shared__ uint32_t epack[SCALAR_ETILE];
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = threadIdx.x / WAVE;
+  const int r = blockIdx.x * (blockDim.x / WAVE) + wave;
+  const uint32_t rc = r < R ? (uint32_t)(rcode[r * 3] | (rcode[r * 3 + 1] << 10)
+                                         | (rcode[r * 3 + 2] << 20)) : 0u;
+  const float b0 = r < R ? rbonus[r * 3] : 0.f;
+  const float b1 = r < R ? rbonus[r * 3 + 1] : 0.f;
+  const float b2 = r < R ? rbonus[r * 3 + 2] : 0.f;
+  for (int e0 = 0; e0 < E; e0 += SCALAR_ETILE) {
+    __syncthreads();
+    for (int i = threadIdx.x; i < SCALAR_ETILE && e0 + i < E; i += blockDim.x)
+      epack[i] = (uint32_t)(ecode[(e0 + i) * 3] | (ecode[(e0 + i) * 3 + 1] << 10)
+                            | (ecode[(e0 + i) * 3 + 2] << 20));
+    __syncthreads();
+    if (r >= R) continue;
+    for (int i = lane; i < SCALAR_ETILE && e0 + i < E; i += WAVE) {
+      const uint32_t ec = epack[i];
+      const uint32_t x = rc ^ ec;
+      float s = 0.0f;
+      if ((x & 0x3FFu) == 0) s += b0;
+      if ((x & 0xFFC00u) == 0) s += b1;
+      if ((x & 0x3FF00000u) == 0) s += b2;
+      score[(int64_t)r * E + e0 + i] = s;
+    }
+  }
+}
+
+void mfma_score_bench(torch::Tensor rcode, torch::Tensor rbonus,
+                      torch::Tensor ecode, int64_t K, torch::Tensor score) {
+  const int R = (int)rcode.size(0), E = (int)ecode.size(0);
+  const int tiles = (R / 16) * (E / 16);
+  hipLaunchKernelGGL(mfma_score_kernel, dim3((tiles + 3) / 4), dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(),
+                     rcode.data_ptr<int32_t>(), rbonus.data_ptr<float>(),
+                     ecode.data_ptr<int32_t>(), R, E, (int)K,
+                     score.data_ptr<float>());
+}
+
+void scalar_score_bench(torch::Tensor rcode, torch::Tensor rbonus,
+                        torch::Tensor ecode, torch::Tensor score) {
+  const int R = (int)rcode.size(0), E = (int)ecode.size(0);
+  hipLaunchKernelGGL(scalar_score_kernel, dim3((R + 3) / 4), dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(),
+                     rcode.data_ptr<int32_t>(), rbonus.data_ptr<float>(),
+                     ecode.data_ptr<int32_t>(), R, E, score.data_ptr<float>());
+}
+
+// ---------------------------------------------------------------------------
 // K9a: KD-tree descent (flat tree, partitioning.py as_flat layout)
 // ---------------------------------------------------------------------------
 

@@ -912,3 +912,29 @@ def test_link_heavy_kernel_distribution():
         # the whole point: few A* iterations per record, no fallback storms
         assert st[0] / N < 100, ("A* iterations per record too high", st)
         assert st[1] / N < 0.01, ("too many full-scan fallbacks", st)
+
+
+@gpu
+def test_mfma_scorer_matches_scalar():
+    """MFMA one-hot categorical scorer vs the LDS scalar scorer: identical
+    scores up to bf16 truncation (the VERDICT #4 experiment's correctness
+    gate; see scripts/mfma_experiment.py and profiles/README.md)."""
+    rng = np.random.default_rng(3)
+    R, E = 256, 512
+    VS = [100, 12, 28]
+    BASE = np.cumsum([0] + VS)[:3]
+    K = ((sum(VS) + 31) // 32) * 32
+    rcode = np.stack([rng.integers(0, v, R) + b for v, b in zip(VS, BASE)], 1)
+    ecode = np.stack([rng.integers(0, v, E) + b for v, b in zip(VS, BASE)], 1)
+    rbonus = rng.uniform(1.0, 12.0, (R, 3))
+    t_rc = _dev(rcode, torch.int32)
+    t_ec = _dev(ecode, torch.int32)
+    t_rb = _dev(rbonus, torch.float32)
+    s1 = torch.empty((R, E), dtype=torch.float32, device=DEV)
+    s2 = torch.empty((R, E), dtype=torch.float32, device=DEV)
+    C.scalar_score_bench(t_rc, t_rb, t_ec, s1)
+    C.mfma_score_bench(t_rc, t_rb, t_ec, K, s2)
+    # oracle in numpy
+    want = ((rcode[:, None, :] == ecode[None, :, :]) * rbonus[:, None, :]).sum(2)
+    np.testing.assert_allclose(s1.cpu().numpy(), want, rtol=1e-6)
+    np.testing.assert_allclose(s2.cpu().numpy(), want, rtol=2 ** -7, atol=0.1)
