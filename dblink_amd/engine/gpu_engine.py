@@ -624,10 +624,16 @@ class GpuEngine(CpuEngine):
             # sets — the stationary high-distortion regime)
             heavy_th = self._heavy_thresh if (
                 dense_idx and self._heavy_active(R)) else 0
-            mode_mask = torch.empty(R, dtype=torch.uint8, device=dev)
-            self.C.classify_modes(gs.rec_values, gs.rec_dist, gs.rec_part,
-                                  ent_ptr, cand_lo, cand_hi, self._num_pairs,
-                                  16, heavy_th, mode_mask)
+            if R >= 50_000 or heavy_th > 0:
+                mode_mask = torch.empty(R, dtype=torch.uint8, device=dev)
+                self.C.classify_modes(gs.rec_values, gs.rec_dist, gs.rec_part,
+                                      ent_ptr, cand_lo, cand_hi, self._num_pairs,
+                                      16, heavy_th, mode_mask)
+            else:
+                # small problems keep every record on the wave path: splitting
+                # 10k records across three kernels leaves the chip idle
+                # (~150 waves total) and costs more than it saves
+                mode_mask = torch.empty(0, dtype=torch.uint8, device=dev)
             self.C.link_update(
                 gs.rec_values, gs.rec_dist, gs.rec_gid, gs.rec_part,
                 cand_lo, cand_hi, postings, gs.ent_values,

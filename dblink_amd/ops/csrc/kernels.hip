@@ -643,7 +643,7 @@ struct ValueArgs {
   uint64_t ent_id_base;
   int* error_count;
   // optional [8]: {k2_pairs, hash_pairs, merge_pairs, sum_entries,
-  //                sum_tsize, sum_kobs, rare_pairs, merge_entries}
+  //                sum_units, sum_kobs, rare_pairs, merge_entries}
   unsigned long long* stats;
 };
 
@@ -1048,7 +1048,7 @@ __device__ void value_update_pair(const ValueArgs& args, int64_t pair, int lane,
     atomicAdd(&args.stats[0], 1ull);
     atomicAdd(&args.stats[dense ? 2 : 1], 1ull);
     atomicAdd(&args.stats[3], (unsigned long long)total_entries);
-    atomicAdd(&args.stats[4], (unsigned long long)(dense ? 0 : tsize));
+    atomicAdd(&args.stats[4], (unsigned long long)n_units);
     atomicAdd(&args.stats[5], (unsigned long long)k_obs);
     if (!is_const && k_obs > args.Kc) atomicAdd(&args.stats[6], 1ull);
     if (dense)
