@@ -6,6 +6,18 @@
 
 namespace dblink {
 
+// link_dense_cpu.cpp
+torch::Tensor pcg2_link_cpu(torch::Tensor rec_values, torch::Tensor rec_file,
+                            torch::Tensor rec_part, torch::Tensor ent_values,
+                            torch::Tensor ent_ptr, torch::Tensor theta,
+                            torch::Tensor phi, torch::Tensor norm,
+                            torch::Tensor log_norm, torch::Tensor voff,
+                            torch::Tensor csr_row_ptr, torch::Tensor csr_col,
+                            torch::Tensor csr_expsim, torch::Tensor attr_const,
+                            std::vector<torch::Tensor> post_perm,
+                            std::vector<torch::Tensor> post_ptr,
+                            torch::Tensor u_rec);
+
 // sim_pairs_cpu.cpp
 std::vector<torch::Tensor> sim_pairs_cpu(torch::Tensor strs, torch::Tensor lens,
                                          double threshold, double max_sim);
@@ -122,6 +134,8 @@ std::vector<torch::Tensor> sim_pairs_gpu(torch::Tensor strs, torch::Tensor lens,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "dblink_amd native ops (CDNA4 HIP kernels + host helpers)";
+  m.def("pcg2_link_cpu", &dblink::pcg2_link_cpu,
+        "PCG-II dense link update (OpenMP, f64 log-space)");
   m.def("sim_pairs_cpu", &dblink::sim_pairs_cpu,
         "banded Levenshtein sim-pair sweep (CPU/OpenMP)");
   m.def("sim_pairs_gpu", &dblink::sim_pairs_gpu,

@@ -23,6 +23,7 @@ ext = CUDAExtension(
     sources=[
         os.path.join(CSRC, "ext.cpp"),
         os.path.join(CSRC, "sim_pairs_cpu.cpp"),
+        os.path.join(CSRC, "link_dense_cpu.cpp"),
         os.path.join(CSRC, "kernels.hip"),
     ],
     include_dirs=[CSRC],

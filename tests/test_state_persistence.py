@@ -534,6 +534,8 @@ def test_pcg2_dense_workspace_path_matches_expression_path(monkeypatch):
 
     from dblink_amd.engine import cpu_fast
 
+    monkeypatch.setenv("DBLINK_NATIVE_PCG2", "0")  # exercise the numpy paths
+
     def run(threshold, iters=200, n=120, seed=5):
         monkeypatch.setattr(cpu_fast, "_DENSE_WS_THRESHOLD", threshold)
         cache, rv, rf = build_cache_and_records(n, seed=seed)
@@ -725,3 +727,43 @@ def test_linkage_arrays_emits_empty_partitions():
     # partition 1 has zero clusters
     assert pid_offsets[2] - pid_offsets[1] == 0
     assert pid_offsets[1] - pid_offsets[0] >= 1
+
+
+def test_pcg2_native_link_matches_numpy_posterior(monkeypatch):
+    """The OpenMP f64 PCG-II link kernel (link_dense_cpu.cpp) and the numpy
+    fast path sample the same posterior (draw realizations differ: f64
+    log-space vs f32 products)."""
+    import os
+    import sys
+
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    from bench import build_cache_and_records
+    from dblink_amd import ops
+
+    if not ops.have_native():
+        pytest.skip("native extension not built")
+
+    def run(native, iters=200, n=120, seed=5):
+        monkeypatch.setenv("DBLINK_NATIVE_PCG2", "1" if native else "0")
+        cache, rv, rf = build_cache_and_records(n, seed=seed)
+        cache._fast_model = None  # fresh model per run
+        delattr(cache, "_fast_model")
+        partitioner = KDTreePartitioner(1, [3])
+        state = deterministic_init(rv, rf, np.arange(n, dtype=np.int64),
+                                   cache, partitioner, seed=seed)
+        engine = CpuEngine(cache, partitioner)
+        engine.initial_summary(state)
+        flags = SamplerFlags.for_sampler("PCG-II")
+        lls, pairs = [], []
+        for i in range(iters):
+            engine.step(state, flags)
+            if i >= iters // 2:
+                lls.append(state.summary.log_likelihood)
+                c = np.bincount(state.rec_ent, minlength=state.num_entities)
+                pairs.append(int(np.sum(c * (c - 1) // 2)))
+        return float(np.mean(lls)), float(np.mean(pairs))
+
+    ll_n, pairs_n = run(True)
+    ll_p, pairs_p = run(False)
+    assert abs(ll_n - ll_p) / abs(ll_p) < 0.02, (ll_n, ll_p)
+    assert abs(pairs_n - pairs_p) <= max(2.5, 0.4 * pairs_p), (pairs_n, pairs_p)
