@@ -219,7 +219,56 @@ def _pcg2_native_model(fm):
         "const": np.array([1 if ia.is_constant else 0 for ia in attrs],
                           dtype=np.uint8),
     }
+    with np.errstate(divide="ignore"):
+        got["log_expsim"] = np.ascontiguousarray(np.log(got["expsim"]))
     return got
+
+
+def _link_indexed_native(state, fm, num_partitions, seed, it):
+    """OpenMP indexed PCG-I/Gibbs link update (pcg1_link_cpu); None when the
+    native extension is unavailable or disabled."""
+    import os as _os
+
+    if _os.environ.get("DBLINK_NATIVE_PCG1", "1") == "0":
+        return None
+    from .. import ops
+
+    if not ops.have_native() or not hasattr(ops.native(), "pcg1_link_cpu"):
+        return None
+    import torch
+
+    C = ops.native()
+    nm = _pcg2_native_model(fm)
+    ent_ptr = np.searchsorted(
+        state.ent_part, np.arange(num_partitions + 1)).astype(np.int64)
+    perms, ptrs = [], []
+    for a, ia in enumerate(fm.attrs):
+        Va = ia.index.num_values
+        keys = state.ent_part.astype(np.int64) * Va + state.ent_values[:, a]
+        perm = np.argsort(keys, kind="stable").astype(np.int32)
+        cnt = np.bincount(keys, minlength=num_partitions * Va)
+        ptr = np.zeros(num_partitions * Va + 1, np.int64)
+        ptr[1:] = np.cumsum(cnt)
+        perms.append(torch.from_numpy(perm))
+        ptrs.append(torch.from_numpy(ptr))
+    rec_part = state.ent_part[state.rec_ent].astype(np.int32)
+    t = torch.from_numpy
+    out, n_empty = C.pcg1_link_cpu(
+        t(np.ascontiguousarray(state.rec_values)),
+        t(np.ascontiguousarray(state.rec_dist)),
+        t(np.ascontiguousarray(rec_part)),
+        t(np.ascontiguousarray(state.rec_gid)),
+        t(np.ascontiguousarray(state.rec_ent)),
+        t(np.ascontiguousarray(state.ent_values)),
+        t(ent_ptr), t(nm["log_norm"]), t(nm["voff"]), t(nm["row_ptr"]),
+        t(nm["col"]), t(nm["log_expsim"]), t(nm["const"]),
+        perms, ptrs, int(seed), int(it),
+    )
+    if n_empty:
+        raise RuntimeError(
+            f"{n_empty} empty candidate sets in link update (invariant violated)"
+        )
+    return out.numpy()
 
 
 def _link_dense_collapsed_native(state, fm, num_partitions, seed, it, rank):
@@ -478,6 +527,13 @@ def sweep_fast(state, cache, partitioner, num_partitions, rank=0, timers=None,
         state.rec_ent = _link_dense_collapsed(
             state, fm, theta_ra, num_partitions, seed, it, rank
         )
+        _mark("link")
+        return _value_and_rest(state, fm, partitioner, num_partitions, rank,
+                               seed, it, theta_ra, obs, rdist, collapsed,
+                               _mark)
+    native = _link_indexed_native(state, fm, num_partitions, seed, it)
+    if native is not None:
+        state.rec_ent = native
         _mark("link")
         return _value_and_rest(state, fm, partitioner, num_partitions, rank,
                                seed, it, theta_ra, obs, rdist, collapsed,

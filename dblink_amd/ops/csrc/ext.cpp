@@ -18,6 +18,14 @@ torch::Tensor pcg2_link_cpu(torch::Tensor rec_values, torch::Tensor rec_file,
                             std::vector<torch::Tensor> post_ptr,
                             torch::Tensor u_rec);
 
+std::tuple<torch::Tensor, int64_t> pcg1_link_cpu(
+    torch::Tensor rec_values, torch::Tensor rec_dist, torch::Tensor rec_part,
+    torch::Tensor rec_gid, torch::Tensor rec_ent_in, torch::Tensor ent_values,
+    torch::Tensor ent_ptr, torch::Tensor log_norm, torch::Tensor voff,
+    torch::Tensor csr_row_ptr, torch::Tensor csr_col, torch::Tensor log_expsim,
+    torch::Tensor attr_const, std::vector<torch::Tensor> post_perm,
+    std::vector<torch::Tensor> post_ptr, int64_t seed, int64_t iteration);
+
 // sim_pairs_cpu.cpp
 std::vector<torch::Tensor> sim_pairs_cpu(torch::Tensor strs, torch::Tensor lens,
                                          double threshold, double max_sim);
@@ -134,6 +142,8 @@ std::vector<torch::Tensor> sim_pairs_gpu(torch::Tensor strs, torch::Tensor lens,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "dblink_amd native ops (CDNA4 HIP kernels + host helpers)";
+  m.def("pcg1_link_cpu", &dblink::pcg1_link_cpu,
+        "indexed PCG-I/Gibbs link update (OpenMP, keyed Philox gumbels)");
   m.def("pcg2_link_cpu", &dblink::pcg2_link_cpu,
         "PCG-II dense link update (OpenMP, f64 log-space)");
   m.def("sim_pairs_cpu", &dblink::sim_pairs_cpu,

@@ -135,3 +135,117 @@ torch::Tensor pcg2_link_cpu(
 }
 
 }  // namespace dblink
+
+#include "common.h"
+
+namespace dblink {
+
+// Indexed (PCG-I / Gibbs) link update, threaded over records: candidates are
+// the smallest non-distorted posting segment (GibbsUpdates.scala:398-430,
+// 473-530), weights over observed-distorted non-constant attributes, draws
+// by Gumbel-max with Philox keyed (seed, iteration, record gid, entity) —
+// deterministic regardless of thread schedule.
+std::tuple<torch::Tensor, int64_t> pcg1_link_cpu(
+    torch::Tensor rec_values, torch::Tensor rec_dist, torch::Tensor rec_part,
+    torch::Tensor rec_gid, torch::Tensor rec_ent_in, torch::Tensor ent_values,
+    torch::Tensor ent_ptr, torch::Tensor log_norm, torch::Tensor voff,
+    torch::Tensor csr_row_ptr, torch::Tensor csr_col, torch::Tensor log_expsim,
+    torch::Tensor attr_const, std::vector<torch::Tensor> post_perm,
+    std::vector<torch::Tensor> post_ptr, int64_t seed, int64_t iteration) {
+  const int64_t R = rec_values.size(0);
+  const int A = (int)rec_values.size(1);
+  auto out = torch::empty({R}, torch::TensorOptions().dtype(torch::kInt64));
+
+  const int32_t* rv = rec_values.data_ptr<int32_t>();
+  const uint8_t* rd = rec_dist.data_ptr<uint8_t>();
+  const int32_t* rp = rec_part.data_ptr<int32_t>();
+  const int64_t* rg = rec_gid.data_ptr<int64_t>();
+  const int64_t* rin = rec_ent_in.data_ptr<int64_t>();
+  const int32_t* ev = ent_values.data_ptr<int32_t>();
+  const int64_t* eptr = ent_ptr.data_ptr<int64_t>();
+  const double* lnrm = log_norm.data_ptr<double>();
+  const int64_t* vo = voff.data_ptr<int64_t>();
+  const int64_t* rptr = csr_row_ptr.data_ptr<int64_t>();
+  const int32_t* rcol = csr_col.data_ptr<int32_t>();
+  const double* rsim = log_expsim.data_ptr<double>();
+  const uint8_t* cst = attr_const.data_ptr<uint8_t>();
+  int64_t* o = out.data_ptr<int64_t>();
+
+  std::vector<const int32_t*> pp(A);
+  std::vector<const int64_t*> pq(A);
+  std::vector<int64_t> Va(A);
+  for (int a = 0; a < A; ++a) {
+    pp[a] = post_perm[a].data_ptr<int32_t>();
+    pq[a] = post_ptr[a].data_ptr<int64_t>();
+    Va[a] = vo[a + 1] - vo[a];
+  }
+  int64_t n_empty = 0;
+
+#pragma omp parallel reduction(+ : n_empty)
+  {
+#pragma omp for schedule(dynamic, 32)
+    for (int64_t r = 0; r < R; ++r) {
+      const int64_t p = rp[r];
+      uint32_t nd_mask = 0, od_mask = 0;
+      for (int a = 0; a < A; ++a) {
+        const int32_t x = rv[r * A + a];
+        if (x < 0) continue;
+        if (!rd[r * A + a]) nd_mask |= 1u << a;
+        else if (!cst[a]) od_mask |= 1u << a;
+      }
+      int base_a = -1;
+      int64_t blo = 0, bn = INT64_MAX;
+      for (uint32_t m = nd_mask; m;) {
+        const int a = __builtin_ctz(m);
+        m &= m - 1;
+        const int64_t key = p * Va[a] + rv[r * A + a];
+        const int64_t lo = pq[a][key], n = pq[a][key + 1] - lo;
+        if (n < bn) { bn = n; blo = lo; base_a = a; }
+      }
+      if (base_a < 0) {  // no observed non-distorted attr: whole partition
+        blo = eptr[p];
+        bn = eptr[p + 1] - blo;
+      }
+      const uint32_t check = base_a >= 0 ? (nd_mask & ~(1u << base_a)) : 0u;
+      const uint64_t gid = (uint64_t)rg[r];
+      double best = -INFINITY;
+      int64_t best_e = -1;
+      for (int64_t i = 0; i < bn; ++i) {
+        const int64_t e = base_a >= 0 ? (int64_t)pp[base_a][blo + i] : blo + i;
+        bool okc = true;
+        for (uint32_t m = check; m;) {
+          const int a = __builtin_ctz(m);
+          m &= m - 1;
+          if (ev[e * A + a] != rv[r * A + a]) { okc = false; break; }
+        }
+        if (!okc) continue;
+        double logw = 0.0;
+        for (uint32_t m = od_mask; m;) {
+          const int a = __builtin_ctz(m);
+          m &= m - 1;
+          const int32_t y = ev[e * A + a];
+          logw += lnrm[vo[a] + y];
+          const int64_t row = vo[a] + rv[r * A + a];
+          int64_t lo2 = rptr[row], hi2 = rptr[row + 1];
+          while (lo2 < hi2) {
+            const int64_t mid = (lo2 + hi2) >> 1;
+            if (rcol[mid] < y) lo2 = mid + 1; else hi2 = mid;
+          }
+          if (lo2 < rptr[row + 1] && rcol[lo2] == y) logw += rsim[lo2];
+        }
+        const float u = philox_uniform((uint64_t)seed, (uint32_t)iteration, 1u,
+                                       gid, (uint32_t)e);
+        const double s = logw - std::log(-std::log((double)u));
+        if (s > best) { best = s; best_e = e; }
+      }
+      if (best_e < 0) {
+        ++n_empty;
+        best_e = rin[r];
+      }
+      o[r] = best_e;
+    }
+  }
+  return {out, n_empty};
+}
+
+}  // namespace dblink
