@@ -190,6 +190,53 @@ class GpuModel:
         self.csr_rawsum = dev(rawsum, torch.float64)
         self.z1 = dev(z1, torch.float64)
         self.theta = torch.zeros((A, self.F), dtype=torch.float32, device=device)
+        self._build_value_ktables(device)
+
+    def _build_value_ktables(self, device):
+        """Per-(value, k) perturbation tables for k in [2, kmax]: the raw
+        weights phi(c) norm(c)^k (e^{k s} - 1) of each sim-row entry, as
+        exclusive row prefixes + row totals (f64). Lets the dominant k >= 2
+        value-update class (all linked records sharing one value; measured
+        d~1.1-1.6 mean distinct values) draw by ONE binary search instead of
+        an O(k row) wave merge. Memory-capped by DBLINK_KTAB_MAX_BYTES."""
+        import torch
+
+        self.ktab_excl = self.ktab_rawsum = None
+        self.ktab_max = 0
+        nnz = int(self.csr_col.numel())
+        if nnz == 0 or self.Kc < 2:
+            return
+        cap = int(os.environ.get("DBLINK_KTAB_MAX_BYTES", str(48 << 30)))
+        levels = min(self.Kc - 1, max(0, cap // (16 * (nnz + self.Vtot))))
+        if levels < 1:
+            return
+        kmax = 1 + levels
+        row_counts = (self.csr_row_ptr[1:] - self.csr_row_ptr[:-1])
+        row_of = torch.repeat_interleave(
+            torch.arange(self.Vtot, device=device, dtype=torch.int64), row_counts)
+        a_of = torch.searchsorted(self.voff, row_of, right=True) - 1
+        gcol = self.voff[a_of] + self.csr_col.to(torch.int64)
+        lp = self.log_phi.double()
+        ln = self.log_norm.double()
+        sim = self.csr_sim.double()
+        base_log = lp[gcol] + ln[gcol]  # reused; k-scaled below
+        ln_g = ln[gcol]
+        start = self.csr_row_ptr[:-1]
+        ends = self.csr_row_ptr[1:] - 1
+        excl = torch.empty((levels, nnz), dtype=torch.float64, device=device)
+        rawsum = torch.zeros((levels, self.Vtot), dtype=torch.float64, device=device)
+        for k in range(2, kmax + 1):
+            raw = torch.exp(base_log + (k - 1) * ln_g) * torch.expm1(k * sim)
+            c = torch.cumsum(raw, 0)
+            base_c = torch.where(start > 0, c[torch.clamp(start - 1, min=0)],
+                                 torch.zeros((), dtype=torch.float64, device=device))
+            excl[k - 2] = c - raw - torch.repeat_interleave(base_c, row_counts)
+            rs = torch.where(row_counts > 0, c[torch.clamp(ends, min=0)],
+                             torch.zeros((), dtype=torch.float64, device=device)) - base_c
+            rawsum[k - 2] = rs
+        self.ktab_excl = excl
+        self.ktab_rawsum = rawsum
+        self.ktab_max = kmax
 
 
 class GpuStateTensors:
@@ -257,6 +304,16 @@ class GpuEngine(CpuEngine):
         self.device = device
         self.C = _require_native()
         self.model = GpuModel(cache, device, cache_kc(cache, partitioner))
+        if self.model.ktab_max >= 2 and os.environ.get("DBLINK_KTAB", "1") != "0":
+            self.C.set_value_ktables(
+                self.model.ktab_excl, self.model.ktab_rawsum,
+                self.model.self_expsim, self.model.ktab_max,
+                int(self.model.csr_col.numel()))
+        else:
+            self.C.set_value_ktables(
+                torch.empty(0, dtype=torch.float64),
+                torch.empty(0, dtype=torch.float64),
+                torch.empty(0, dtype=torch.float32), 0, 0)
         self.flat_tree = None
         self._gs = None
         self._ent_id_base = rank << 40

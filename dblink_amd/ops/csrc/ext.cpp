@@ -125,6 +125,8 @@ void build_keys(torch::Tensor ent_part, torch::Tensor ent_values,
                 torch::Tensor pair_v2, int64_t Vmax,
                 torch::Tensor ekeys, torch::Tensor qkeys);
 void set_value_stats(torch::Tensor t);
+void set_value_ktables(torch::Tensor excl, torch::Tensor rawsum,
+                       torch::Tensor self_expsim, int64_t kmax, int64_t nnz);
 void mfma_score_bench(torch::Tensor rcode, torch::Tensor rbonus,
                       torch::Tensor ecode, int64_t K, torch::Tensor score);
 void scalar_score_bench(torch::Tensor rcode, torch::Tensor rbonus,
@@ -177,6 +179,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("summary_counts", &dblink::summary_counts, "fused summary counts + pack");
   m.def("set_value_stats", &dblink::set_value_stats,
         "install the optional value-phase work-counter buffer");
+  m.def("set_value_ktables", &dblink::set_value_ktables,
+        "install the k>=2 single-value perturbation tables");
   m.def("mfma_score_bench", &dblink::mfma_score_bench,
         "MFMA one-hot categorical scorer (experiment)");
   m.def("scalar_score_bench", &dblink::scalar_score_bench,

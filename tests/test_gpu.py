@@ -938,3 +938,65 @@ def test_mfma_scorer_matches_scalar():
     want = ((rcode[:, None, :] == ecode[None, :, :]) * rbonus[:, None, :]).sum(2)
     np.testing.assert_allclose(s1.cpu().numpy(), want, rtol=1e-6)
     np.testing.assert_allclose(s2.cpu().numpy(), want, rtol=2 ** -7, atol=0.1)
+
+
+@gpu
+def test_value_kd1_kernel_distribution():
+    """k=2 single-shared-value clusters (the kd1 table path): empirical value
+    frequencies vs the exact mixture P(v) = (base2(v) + w(v)) / (1 + W) with
+    w(v) = base2(v) (f(v)^2 - 1), f(x) boosted by the collapsed self term."""
+    cache, model = make_model(DEV)
+    assert model.ktab_max >= 2
+    a = 1
+    ia = cache.indexed_attributes[a]
+    idx = ia.index
+    x = 0
+    th = 0.1
+    model.theta.copy_(torch.full((2, 1), th))
+    N = 40000
+    # N entities, each with two linked records of value x (file 0)
+    rec_values = np.tile(np.array([[-1, x]], dtype=np.int32), (2 * N, 1))
+    rec_dist = np.ones((2 * N, 2), dtype=np.uint8)
+    ent_vals = np.zeros((N, 2), dtype=np.int32)
+    ent_rec_ptr = np.arange(0, 2 * N + 1, 2, dtype=np.int64)
+    ent_rec_idx = np.arange(2 * N, dtype=np.int64)
+    kobs = np.zeros((N, 2), dtype=np.int32)
+    kobs[:, a] = 2
+
+    C.set_value_ktables(model.ktab_excl, model.ktab_rawsum, model.self_expsim,
+                        model.ktab_max, int(model.csr_col.numel()))
+    err = torch.zeros(1, dtype=torch.int32, device=DEV)
+    ev = _dev(ent_vals, torch.int32)
+    empty64 = torch.empty(0, dtype=torch.int64, device=DEV)
+    C.value_update(
+        _dev(rec_values, torch.int32), _dev(rec_dist, torch.uint8),
+        _dev(np.zeros(2 * N, np.int32), torch.int32),
+        _dev(ent_rec_ptr, torch.int64), _dev(ent_rec_idx, torch.int64), ev,
+        model.theta, model.phi, model.log_phi, model.norm_lin, model.log_norm,
+        model.voff, model.csr_row_ptr, model.csr_col, model.csr_sim,
+        model.phi_prob, model.phi_alias, model.pow_prob, model.pow_alias,
+        model.pow_off, model.log_pow_total, model.attr_const, model.Kc,
+        1, 0, 777, 5, 0, err, empty64, empty64, empty64,
+        model.csr_excl, model.csr_rawsum, model.z1, empty64,
+        _dev(kobs.reshape(-1), torch.int32))
+    sel = ev.cpu().numpy()[:, a]
+
+    V = idx.num_values
+    phi = idx.probs
+    norms = idx.sim_norms
+    z2 = idx.sim_norm_total(2)
+    base2 = phi * norms ** 2 / z2
+    w = np.zeros(V)
+    si = idx.sim_index
+    se = (1.0 / th - 1.0) / (phi[x] * norms[x])
+    for j in range(si.row_ptr[x], si.row_ptr[x + 1]):
+        c = si.col[j]
+        f = si.expsim[j] + (se if c == x else 0.0)
+        w[c] = base2[c] * (f * f - 1.0)
+    W = w.sum()
+    exact = (base2 + w) / (1.0 + W)
+    emp = np.bincount(sel, minlength=V) / N
+    assert tv_distance(emp, exact) < 0.02, tv_distance(emp, exact)
+    # reset globals so later direct-kernel tests see a clean slate
+    z = torch.empty(0, dtype=torch.float64)
+    C.set_value_ktables(z, z, torch.empty(0, dtype=torch.float32), 0, 0)
