@@ -445,3 +445,13 @@ def test_overlapped_migration_eight_rank_soak():
         assert any('"ok": true' in o for o in outs), outs
     finally:
         _os.environ.pop("DBLINK_TEST_ROUNDS", None)
+
+
+@pytest.mark.slow
+def test_eight_rank_chain_soak():
+    """Eight ranks over sixteen KD partitions — the 8-GPU node's fan-out
+    shape: ownership, conservation and gid-coverage invariants held across
+    25 full sweeps with per-sweep migration (gloo)."""
+    outs = _run_workers(WORKER.replace("__ROOT__", ROOT).replace("__LEVELS__", "4"),
+                        world=8, timeout=900)
+    assert any('"ok": true' in o for o in outs), outs
