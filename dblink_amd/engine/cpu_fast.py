@@ -224,6 +224,73 @@ def _pcg2_native_model(fm):
     return got
 
 
+def _value_native(state, fm, collapsed, seed, it, rank, rorder_all):
+    """OpenMP entity-value update (value_update_cpu): same Philox streams and
+    f64 op order as the numpy block below (draws agree except ulp-boundary
+    CDF ties). Returns (new_ev, fallback pair list) or None when the native
+    extension is unavailable or disabled."""
+    import os as _os
+
+    if _os.environ.get("DBLINK_NATIVE_VALUE", "1") == "0":
+        return None
+    from .. import ops
+
+    if not ops.have_native() or not hasattr(ops.native(), "value_update_cpu"):
+        return None
+    import torch
+
+    C = ops.native()
+    nm = _pcg2_native_model(fm)
+    vt = getattr(fm, "_value_native_tabs", None)
+    if vt is None:
+        attrs = fm.attrs
+        phi_prob, phi_alias, pow_prob, pow_alias = [], [], [], []
+        ptot = np.ones((fm.A, fm.kmax + 1))
+        for a, ia in enumerate(attrs):
+            t = fm.phi_tables[a]
+            phi_prob.append(torch.from_numpy(
+                np.ascontiguousarray(t.prob.astype(np.float64))))
+            phi_alias.append(torch.from_numpy(
+                np.ascontiguousarray(t.alias.astype(np.int64))))
+            if ia.is_constant:
+                pow_prob.append(torch.empty(0, dtype=torch.float64))
+                pow_alias.append(torch.empty(0, dtype=torch.int64))
+            else:
+                pp = np.concatenate([
+                    ia.index.sim_norm_dist(kv).prob.astype(np.float64)
+                    for kv in range(1, fm.kmax + 1)])
+                pa = np.concatenate([
+                    ia.index.sim_norm_dist(kv).alias.astype(np.int64)
+                    for kv in range(1, fm.kmax + 1)])
+                pow_prob.append(torch.from_numpy(np.ascontiguousarray(pp)))
+                pow_alias.append(torch.from_numpy(np.ascontiguousarray(pa)))
+                ptot[a, :] = fm.pow_totals[a][: fm.kmax + 1]
+        vt = fm._value_native_tabs = (
+            phi_prob, phi_alias, pow_prob, pow_alias,
+            torch.from_numpy(np.ascontiguousarray(ptot)))
+    phi_prob, phi_alias, pow_prob, pow_alias, ptot_t = vt
+    starts = np.searchsorted(
+        state.rec_ent[rorder_all], np.arange(state.num_entities + 1)
+    ).astype(np.int64)
+    t = torch.from_numpy
+    new_ev, fb = C.value_update_cpu(
+        t(np.ascontiguousarray(state.rec_values)),
+        t(np.ascontiguousarray(state.rec_file.astype(np.int32))),
+        t(np.ascontiguousarray(state.rec_dist)),
+        t(np.ascontiguousarray(state.rec_ent)),
+        t(np.ascontiguousarray(rorder_all.astype(np.int64))), t(starts),
+        t(np.ascontiguousarray(state.ent_values)),
+        t(np.ascontiguousarray(state.dist_probs.probs)),
+        t(nm["phi"]), t(nm["norm"]), t(nm["voff"]), t(nm["row_ptr"]),
+        t(nm["col"]), t(nm["expsim"]), t(nm["const"]),
+        phi_prob, phi_alias, pow_prob, pow_alias, ptot_t,
+        int(fm.kmax), 1 if collapsed else 0, int(seed), int(it), int(rank),
+    )
+    A = fm.A
+    fallback = [(int(p) // A, int(p) % A) for p in fb.numpy()]
+    return new_ev.numpy(), fallback
+
+
 def _link_indexed_native(state, fm, num_partitions, seed, it):
     """OpenMP indexed PCG-I/Gibbs link update (pcg1_link_cpu); None when the
     native extension is unavailable or disabled."""
@@ -639,161 +706,167 @@ def _value_and_rest(state, fm, partitioner, num_partitions, rank, seed, it,
     rv = state.rec_values
 
     # ---- phase 2: collapsed entity-value update -----------------------------
-    kobs = np.zeros((E, A), dtype=np.int64)
-    np.add.at(kobs, state.rec_ent, obs.astype(np.int64))
-    ea_ids = (np.arange(E)[:, None] * A + np.arange(A)[None, :])
-    u_mix, u_a1, u_a2, u_sel = (
-        u.reshape(E, A)
-        for u in _philox_uniform4(seed, it, _PH_VAL, ea_ids.reshape(-1), 0, rank)
-    )
-
-    # records in entity order, computed once; per attribute the observed
-    # subset stays entity-sorted after masking
+    # records in entity order, computed once (native + numpy paths and the
+    # oracle fallback all address records through it)
     rorder_all = np.argsort(state.rec_ent, kind="stable")
-    new_ev = ev.copy()
-    fallback = []
-    if not collapsed:
-        # Gibbs: a non-distorted observed copy pins the value (all such
-        # copies agree with the entity by chain invariant)
-        ndm = obs & ~rdist
-        nd_count = np.zeros((E, A), dtype=np.int64)
-        np.add.at(nd_count, state.rec_ent, ndm.astype(np.int64))
-        copy_val = np.zeros((E, A), dtype=rv.dtype)
-        for a in range(A):
-            m = ndm[:, a]
-            copy_val[state.rec_ent[m], a] = rv[m, a]
-    for a in range(A):
-        ia = attrs[a]
-        k = kobs[:, a].copy()
+    native = _value_native(state, fm, collapsed, seed, it, rank, rorder_all)
+    if native is not None:
+        new_ev, fallback = native
+    else:
+        # ---- phase 2: collapsed entity-value update -----------------------------
+        kobs = np.zeros((E, A), dtype=np.int64)
+        np.add.at(kobs, state.rec_ent, obs.astype(np.int64))
+        ea_ids = (np.arange(E)[:, None] * A + np.arange(A)[None, :])
+        u_mix, u_a1, u_a2, u_sel = (
+            u.reshape(E, A)
+            for u in _philox_uniform4(seed, it, _PH_VAL, ea_ids.reshape(-1), 0, rank)
+        )
+
+        new_ev = ev.copy()
+        fallback = []
         if not collapsed:
-            pinned = nd_count[:, a] > 0
-            new_ev[pinned, a] = copy_val[pinned, a]
-            k[pinned] = -1  # handled; excluded from every draw branch below
-            if ia.is_constant:
-                # constant attr, observed but all distorted: plain phi draw
-                k[k >= 1] = 0
-        k0 = k == 0
-        if k0.any():  # no observed copies: draw from phi
-            t = fm.phi_tables[a]
-            new_ev[k0, a] = _alias_draw_vec(u_a1[k0, a], u_a2[k0, a], t.prob, t.alias)
-        kcap = fm.kmax
-        k1 = k == 1
-        if k1.any():
-            # single observed copy (the common case): one sim row, no merge
-            e1 = np.flatnonzero(k1)
-            rr_s = rorder_all[obs[rorder_all, a]]
-            r1 = rr_s[np.searchsorted(state.rec_ent[rr_s], e1)]
-            x1 = rv[r1, a]
-            th1 = theta_ra[r1, a]
-            if ia.is_constant:
-                # collapsed closed form: total = 1/theta - 1 => P(base) = theta
-                # (the non-collapsed constant case never reaches here)
-                take_base = u_mix[e1, a] < th1
+            # Gibbs: a non-distorted observed copy pins the value (all such
+            # copies agree with the entity by chain invariant)
+            ndm = obs & ~rdist
+            nd_count = np.zeros((E, A), dtype=np.int64)
+            np.add.at(nd_count, state.rec_ent, ndm.astype(np.int64))
+            copy_val = np.zeros((E, A), dtype=rv.dtype)
+            for a in range(A):
+                m = ndm[:, a]
+                copy_val[state.rec_ent[m], a] = rv[m, a]
+        for a in range(A):
+            ia = attrs[a]
+            k = kobs[:, a].copy()
+            if not collapsed:
+                pinned = nd_count[:, a] > 0
+                new_ev[pinned, a] = copy_val[pinned, a]
+                k[pinned] = -1  # handled; excluded from every draw branch below
+                if ia.is_constant:
+                    # constant attr, observed but all distorted: plain phi draw
+                    k[k >= 1] = 0
+            k0 = k == 0
+            if k0.any():  # no observed copies: draw from phi
                 t = fm.phi_tables[a]
-                tb = e1[take_base]
-                new_ev[tb, a] = _alias_draw_vec(u_a1[tb, a], u_a2[tb, a],
-                                                t.prob, t.alias)
-                new_ev[e1[~take_base], a] = x1[~take_base]
+                new_ev[k0, a] = _alias_draw_vec(u_a1[k0, a], u_a2[k0, a], t.prob, t.alias)
+            kcap = fm.kmax
+            k1 = k == 1
+            if k1.any():
+                # single observed copy (the common case): one sim row, no merge
+                e1 = np.flatnonzero(k1)
+                rr_s = rorder_all[obs[rorder_all, a]]
+                r1 = rr_s[np.searchsorted(state.rec_ent[rr_s], e1)]
+                x1 = rv[r1, a]
+                th1 = theta_ra[r1, a]
+                if ia.is_constant:
+                    # collapsed closed form: total = 1/theta - 1 => P(base) = theta
+                    # (the non-collapsed constant case never reaches here)
+                    take_base = u_mix[e1, a] < th1
+                    t = fm.phi_tables[a]
+                    tb = e1[take_base]
+                    new_ev[tb, a] = _alias_draw_vec(u_a1[tb, a], u_a2[tb, a],
+                                                    t.prob, t.alias)
+                    new_ev[e1[~take_base], a] = x1[~take_base]
+                else:
+                    si = ia.index.sim_index
+                    rlen = si.row_ptr[x1 + 1] - si.row_ptr[x1]
+                    grp, goff = _ragged_expand(rlen)
+                    flat_i = si.row_ptr[x1[grp]] + (np.arange(goff[-1]) - goff[grp])
+                    fcol = si.col[flat_i]
+                    w = si.expsim[flat_i].copy()
+                    if collapsed:
+                        pxn = ia.index.probs[x1] * ia.index.sim_norms[x1]
+                        selfm = fcol == x1[grp]
+                        gsf = grp[selfm]
+                        w[selfm] += (1.0 / th1[gsf] - 1.0) / pxn[gsf]
+                    wgt = (ia.index.probs[fcol] * ia.index.sim_norms[fcol]
+                           / fm.pow_totals[a][1]) * (w - 1.0)
+                    c = np.cumsum(wgt)
+                    basec = np.where(goff[:-1] > 0, c[goff[:-1] - 1], 0.0)
+                    tot = c[goff[1:] - 1] - basec
+                    take_base = u_mix[e1, a] < 1.0 / (1.0 + tot)
+                    t1 = ia.index.sim_norm_dist(1)
+                    tb = e1[take_base]
+                    new_ev[tb, a] = _alias_draw_vec(u_a1[tb, a], u_a2[tb, a],
+                                                    t1.prob, t1.alias)
+                    pert = ~take_base
+                    if pert.any():
+                        target = basec[pert] + u_sel[e1[pert], a] * tot[pert]
+                        j = np.searchsorted(c, target, side="right")
+                        j = np.minimum(j, goff[1:][pert] - 1)
+                        new_ev[e1[pert], a] = fcol[j]
+            km = (k >= 2) & (k <= kcap)
+            for e in np.flatnonzero(k > kcap):  # beyond the cached powers: oracle path
+                fallback.append((int(e), a))
+            if not km.any():
+                continue
+            e_idx = np.flatnonzero(km)
+            kk = k[e_idx]
+            # ragged (pair -> its observed records), entity-sorted
+            rr_s = rorder_all[obs[rorder_all, a]]
+            gstart = np.searchsorted(state.rec_ent[rr_s], e_idx)
+            pgrp, poff = _ragged_expand(kk)
+            ridx = rr_s[gstart[pgrp] + (np.arange(poff[-1]) - poff[pgrp])]
+            x = rv[ridx, a]
+            th = theta_ra[ridx, a]
+            # each record contributes its sim row (constants: the singleton {x})
+            # with the closed-form self-term added at v == x
+            if ia.is_constant:
+                fcol = x
+                px = ia.index.probs[x]
+                w = 1.0 + (1.0 / th - 1.0) / px  # collapsed only (see gate above)
+                pair_of = pgrp
             else:
                 si = ia.index.sim_index
-                rlen = si.row_ptr[x1 + 1] - si.row_ptr[x1]
-                grp, goff = _ragged_expand(rlen)
-                flat_i = si.row_ptr[x1[grp]] + (np.arange(goff[-1]) - goff[grp])
+                rlen = si.row_ptr[x + 1] - si.row_ptr[x]
+                ggrp, goff2 = _ragged_expand(rlen)
+                flat_i = si.row_ptr[x[ggrp]] + (np.arange(goff2[-1]) - goff2[ggrp])
                 fcol = si.col[flat_i]
                 w = si.expsim[flat_i].copy()
                 if collapsed:
-                    pxn = ia.index.probs[x1] * ia.index.sim_norms[x1]
-                    selfm = fcol == x1[grp]
-                    gsf = grp[selfm]
-                    w[selfm] += (1.0 / th1[gsf] - 1.0) / pxn[gsf]
-                wgt = (ia.index.probs[fcol] * ia.index.sim_norms[fcol]
-                       / fm.pow_totals[a][1]) * (w - 1.0)
-                c = np.cumsum(wgt)
-                basec = np.where(goff[:-1] > 0, c[goff[:-1] - 1], 0.0)
-                tot = c[goff[1:] - 1] - basec
-                take_base = u_mix[e1, a] < 1.0 / (1.0 + tot)
-                t1 = ia.index.sim_norm_dist(1)
-                tb = e1[take_base]
-                new_ev[tb, a] = _alias_draw_vec(u_a1[tb, a], u_a2[tb, a],
-                                                t1.prob, t1.alias)
-                pert = ~take_base
-                if pert.any():
-                    target = basec[pert] + u_sel[e1[pert], a] * tot[pert]
-                    j = np.searchsorted(c, target, side="right")
-                    j = np.minimum(j, goff[1:][pert] - 1)
-                    new_ev[e1[pert], a] = fcol[j]
-        km = (k >= 2) & (k <= kcap)
-        for e in np.flatnonzero(k > kcap):  # beyond the cached powers: oracle path
-            fallback.append((int(e), a))
-        if not km.any():
-            continue
-        e_idx = np.flatnonzero(km)
-        kk = k[e_idx]
-        # ragged (pair -> its observed records), entity-sorted
-        rr_s = rorder_all[obs[rorder_all, a]]
-        gstart = np.searchsorted(state.rec_ent[rr_s], e_idx)
-        pgrp, poff = _ragged_expand(kk)
-        ridx = rr_s[gstart[pgrp] + (np.arange(poff[-1]) - poff[pgrp])]
-        x = rv[ridx, a]
-        th = theta_ra[ridx, a]
-        # each record contributes its sim row (constants: the singleton {x})
-        # with the closed-form self-term added at v == x
-        if ia.is_constant:
-            fcol = x
-            px = ia.index.probs[x]
-            w = 1.0 + (1.0 / th - 1.0) / px  # collapsed only (see gate above)
-            pair_of = pgrp
-        else:
-            si = ia.index.sim_index
-            rlen = si.row_ptr[x + 1] - si.row_ptr[x]
-            ggrp, goff2 = _ragged_expand(rlen)
-            flat_i = si.row_ptr[x[ggrp]] + (np.arange(goff2[-1]) - goff2[ggrp])
-            fcol = si.col[flat_i]
-            w = si.expsim[flat_i].copy()
-            if collapsed:
-                pxn = ia.index.probs[x] * ia.index.sim_norms[x]
-                selfm = fcol == x[ggrp]
-                gs_ = ggrp[selfm]
-                w[selfm] += (1.0 / th[gs_] - 1.0) / pxn[gs_]
-            pair_of = pgrp[ggrp]
-        # union-combine: product of the records' factors per (pair, value)
-        so = np.lexsort((fcol, pair_of))
-        pc, cc, wc = pair_of[so], fcol[so], w[so]
-        runs = np.flatnonzero(np.r_[True, (pc[1:] != pc[:-1]) | (cc[1:] != cc[:-1])])
-        vw = np.multiply.reduceat(wc, runs)
-        ucol = cc[runs]
-        upair = pc[runs]
-        # base distribution: p_k(v) = phi(v) * norm(v)^k / Z_k (phi for consts)
-        if ia.is_constant:
-            basep = ia.index.probs[ucol]
-        else:
-            kw = kk[upair].astype(np.float64)
-            basep = (ia.index.probs[ucol] * ia.index.sim_norms[ucol] ** kw
-                     / fm.pow_totals[a][kk[upair]])
-        weight = basep * (vw - 1.0)
-        c = np.cumsum(weight)
-        pstart = np.flatnonzero(np.r_[True, upair[1:] != upair[:-1]])
-        pend = np.r_[pstart[1:], len(upair)]
-        basec = np.where(pstart > 0, c[pstart - 1], 0.0)
-        tot = c[pend - 1] - basec
-        take_base = u_mix[e_idx, a] < 1.0 / (1.0 + tot)
-        if ia.is_constant:
-            t = fm.phi_tables[a]
-            tb = e_idx[take_base]
-            new_ev[tb, a] = _alias_draw_vec(u_a1[tb, a], u_a2[tb, a], t.prob, t.alias)
-        else:
-            for kv in np.unique(kk[take_base]):
-                sel = take_base & (kk == kv)
-                tb = e_idx[sel]
-                t = ia.index.sim_norm_dist(int(kv))
-                new_ev[tb, a] = _alias_draw_vec(u_a1[tb, a], u_a2[tb, a],
-                                                t.prob, t.alias)
-        pert = ~take_base
-        if pert.any():
-            target = basec[pert] + u_sel[e_idx[pert], a] * tot[pert]
-            j = np.searchsorted(c, target, side="right")
-            j = np.minimum(j, pend[pert] - 1)
-            new_ev[e_idx[pert], a] = ucol[j]
+                    pxn = ia.index.probs[x] * ia.index.sim_norms[x]
+                    selfm = fcol == x[ggrp]
+                    gs_ = ggrp[selfm]
+                    w[selfm] += (1.0 / th[gs_] - 1.0) / pxn[gs_]
+                pair_of = pgrp[ggrp]
+            # union-combine: product of the records' factors per (pair, value)
+            so = np.lexsort((fcol, pair_of))
+            pc, cc, wc = pair_of[so], fcol[so], w[so]
+            runs = np.flatnonzero(np.r_[True, (pc[1:] != pc[:-1]) | (cc[1:] != cc[:-1])])
+            vw = np.multiply.reduceat(wc, runs)
+            ucol = cc[runs]
+            upair = pc[runs]
+            # base distribution: p_k(v) = phi(v) * norm(v)^k / Z_k (phi for consts)
+            if ia.is_constant:
+                basep = ia.index.probs[ucol]
+            else:
+                kw = kk[upair].astype(np.float64)
+                basep = (ia.index.probs[ucol] * ia.index.sim_norms[ucol] ** kw
+                         / fm.pow_totals[a][kk[upair]])
+            weight = basep * (vw - 1.0)
+            c = np.cumsum(weight)
+            pstart = np.flatnonzero(np.r_[True, upair[1:] != upair[:-1]])
+            pend = np.r_[pstart[1:], len(upair)]
+            basec = np.where(pstart > 0, c[pstart - 1], 0.0)
+            tot = c[pend - 1] - basec
+            take_base = u_mix[e_idx, a] < 1.0 / (1.0 + tot)
+            if ia.is_constant:
+                t = fm.phi_tables[a]
+                tb = e_idx[take_base]
+                new_ev[tb, a] = _alias_draw_vec(u_a1[tb, a], u_a2[tb, a], t.prob, t.alias)
+            else:
+                for kv in np.unique(kk[take_base]):
+                    sel = take_base & (kk == kv)
+                    tb = e_idx[sel]
+                    t = ia.index.sim_norm_dist(int(kv))
+                    new_ev[tb, a] = _alias_draw_vec(u_a1[tb, a], u_a2[tb, a],
+                                                    t.prob, t.alias)
+            pert = ~take_base
+            if pert.any():
+                target = basec[pert] + u_sel[e_idx[pert], a] * tot[pert]
+                j = np.searchsorted(c, target, side="right")
+                j = np.minimum(j, pend[pert] - 1)
+                new_ev[e_idx[pert], a] = ucol[j]
+
 
     if fallback:
         from . import cpu_engine as ce

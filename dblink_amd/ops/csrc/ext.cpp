@@ -26,6 +26,17 @@ std::tuple<torch::Tensor, int64_t> pcg1_link_cpu(
     torch::Tensor attr_const, std::vector<torch::Tensor> post_perm,
     std::vector<torch::Tensor> post_ptr, int64_t seed, int64_t iteration);
 
+std::tuple<torch::Tensor, torch::Tensor> value_update_cpu(
+    torch::Tensor rec_values, torch::Tensor rec_file, torch::Tensor rec_dist,
+    torch::Tensor rec_ent, torch::Tensor rorder, torch::Tensor starts,
+    torch::Tensor ent_values, torch::Tensor theta, torch::Tensor phi,
+    torch::Tensor norm, torch::Tensor voff, torch::Tensor csr_row_ptr,
+    torch::Tensor csr_col, torch::Tensor csr_expsim, torch::Tensor attr_const,
+    std::vector<torch::Tensor> phi_prob, std::vector<torch::Tensor> phi_alias,
+    std::vector<torch::Tensor> pow_prob, std::vector<torch::Tensor> pow_alias,
+    torch::Tensor pow_totals, int64_t kmax, int64_t collapsed, int64_t seed,
+    int64_t iteration, int64_t rank);
+
 // sim_pairs_cpu.cpp
 std::vector<torch::Tensor> sim_pairs_cpu(torch::Tensor strs, torch::Tensor lens,
                                          double threshold, double max_sim);
@@ -144,6 +155,8 @@ std::vector<torch::Tensor> sim_pairs_gpu(torch::Tensor strs, torch::Tensor lens,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "dblink_amd native ops (CDNA4 HIP kernels + host helpers)";
+  m.def("value_update_cpu", &dblink::value_update_cpu,
+        "entity-value update (OpenMP, bitwise-matches the numpy fast path)");
   m.def("pcg1_link_cpu", &dblink::pcg1_link_cpu,
         "indexed PCG-I/Gibbs link update (OpenMP, keyed Philox gumbels)");
   m.def("pcg2_link_cpu", &dblink::pcg2_link_cpu,

@@ -12,8 +12,10 @@
 // Draws use the same per-record Philox uniforms as the numpy path
 // (inverse CDF over the normalized weights).
 
+#include <omp.h>
 #include <torch/extension.h>
 
+#include <algorithm>
 #include <cmath>
 #include <cstdint>
 #include <vector>
@@ -246,6 +248,226 @@ std::tuple<torch::Tensor, int64_t> pcg1_link_cpu(
     }
   }
   return {out, n_empty};
+}
+
+}  // namespace dblink
+
+namespace dblink {
+
+// CPU Philox4x32-10 with the cpu_fast keying (phase folded into the key's
+// high word, counter = (id_lo, id_hi, draw, iteration)) — bitwise-identical
+// to cpu_fast._philox_uniform4.
+static inline void philox_cpu4(uint64_t seed, uint32_t iteration, uint32_t phase,
+                               uint64_t id, uint32_t draw, int rank, double u[4]) {
+  uint32_t c0 = (uint32_t)id, c1 = (uint32_t)(id >> 32), c2 = draw, c3 = iteration;
+  uint32_t k0 = (uint32_t)seed;
+  uint32_t k1 = (uint32_t)(seed >> 32) ^ (uint32_t)(phase * 0x9E3779B1u) ^
+                (uint32_t)((uint32_t)rank * 0x85EBCA6Bu);
+  for (int r = 0; r < 10; ++r) {
+    const uint64_t p0 = (uint64_t)c0 * 0xD2511F53ull;
+    const uint64_t p1 = (uint64_t)c2 * 0xCD9E8D57ull;
+    const uint32_t n0 = (uint32_t)(p1 >> 32) ^ c1 ^ k0;
+    const uint32_t n1 = (uint32_t)p1;
+    const uint32_t n2 = (uint32_t)(p0 >> 32) ^ c3 ^ k1;
+    const uint32_t n3 = (uint32_t)p0;
+    c0 = n0; c1 = n1; c2 = n2; c3 = n3;
+    k0 += 0x9E3779B9u; k1 += 0xBB67AE85u;
+  }
+  const double s = ldexp(1.0, -32);
+  u[0] = ((double)c0 + 0.5) * s;
+  u[1] = ((double)c1 + 0.5) * s;
+  u[2] = ((double)c2 + 0.5) * s;
+  u[3] = ((double)c3 + 0.5) * s;
+}
+
+static inline int alias_draw_cpu(double u1, double u2, const double* prob,
+                                 const int64_t* alias, int64_t n) {
+  int64_t idx = (int64_t)(u1 * (double)n);
+  if (idx > n - 1) idx = n - 1;
+  return (int)(u2 < prob[idx] ? idx : alias[idx]);
+}
+
+// The collapsed / non-collapsed entity-value update (K6,
+// GibbsUpdates.scala:576-727), threaded over entities. Bitwise-identical to
+// the numpy fast path (same Philox streams, same f64 op order); pairs with
+// k_obs beyond the cached power tables are returned for the python oracle
+// fallback.
+std::tuple<torch::Tensor, torch::Tensor> value_update_cpu(
+    torch::Tensor rec_values, torch::Tensor rec_file, torch::Tensor rec_dist,
+    torch::Tensor rec_ent, torch::Tensor rorder, torch::Tensor starts,
+    torch::Tensor ent_values, torch::Tensor theta, torch::Tensor phi,
+    torch::Tensor norm, torch::Tensor voff, torch::Tensor csr_row_ptr,
+    torch::Tensor csr_col, torch::Tensor csr_expsim, torch::Tensor attr_const,
+    std::vector<torch::Tensor> phi_prob, std::vector<torch::Tensor> phi_alias,
+    std::vector<torch::Tensor> pow_prob, std::vector<torch::Tensor> pow_alias,
+    torch::Tensor pow_totals, int64_t kmax, int64_t collapsed, int64_t seed,
+    int64_t iteration, int64_t rank) {
+  const int64_t E = ent_values.size(0);
+  const int A = (int)ent_values.size(1);
+  const int F = (int)theta.size(1);
+  auto new_ev = ent_values.clone();
+  const int32_t* rv = rec_values.data_ptr<int32_t>();
+  const int32_t* rf = rec_file.data_ptr<int32_t>();
+  const uint8_t* rd = rec_dist.data_ptr<uint8_t>();
+  const int64_t* ro = rorder.data_ptr<int64_t>();
+  const int64_t* st = starts.data_ptr<int64_t>();
+  int32_t* ev = new_ev.data_ptr<int32_t>();
+  const double* th_ = theta.data_ptr<double>();
+  const double* phi_ = phi.data_ptr<double>();
+  const double* nrm = norm.data_ptr<double>();
+  const int64_t* vo = voff.data_ptr<int64_t>();
+  const int64_t* rptr = csr_row_ptr.data_ptr<int64_t>();
+  const int32_t* rcol = csr_col.data_ptr<int32_t>();
+  const double* rexp = csr_expsim.data_ptr<double>();
+  const uint8_t* cst = attr_const.data_ptr<uint8_t>();
+  const double* ptot = pow_totals.data_ptr<double>();  // [A, kmax+1]
+  std::vector<const double*> php(A), pwp(A);
+  std::vector<const int64_t*> pha(A), pwa(A);
+  for (int a = 0; a < A; ++a) {
+    php[a] = phi_prob[a].data_ptr<double>();
+    pha[a] = phi_alias[a].data_ptr<int64_t>();
+    pwp[a] = pow_prob[a].numel() ? pow_prob[a].data_ptr<double>() : nullptr;
+    pwa[a] = pow_alias[a].numel() ? pow_alias[a].data_ptr<int64_t>() : nullptr;
+  }
+  std::vector<std::vector<int64_t>> fb_per_thread(64);
+
+#pragma omp parallel
+  {
+    const int tid = omp_get_thread_num() & 63;
+    std::vector<std::pair<int32_t, double>> buf;   // (col, factor) entries
+    std::vector<double> cum;
+#pragma omp for schedule(dynamic, 64)
+    for (int64_t e = 0; e < E; ++e) {
+      for (int a = 0; a < A; ++a) {
+        const int64_t V = vo[a + 1] - vo[a];
+        double u[4];
+        philox_cpu4((uint64_t)seed, (uint32_t)iteration, 3u,
+                    (uint64_t)(e * A + a), 0u, (int)rank, u);
+        const double u_mix = u[0], u_a1 = u[1], u_a2 = u[2], u_sel = u[3];
+        // gather observed linked records for this attribute
+        int k = 0;
+        int32_t pinned = -1;
+        buf.clear();
+        for (int64_t j = st[e]; j < st[e + 1]; ++j) {
+          const int64_t r = ro[j];
+          const int32_t x = rv[r * A + a];
+          if (x < 0) continue;
+          ++k;
+          if (!collapsed && !rd[r * A + a]) pinned = x;  // last write wins
+          buf.emplace_back(x, (double)rf[r]);  // (value, file) of obs records
+        }
+        if (!collapsed && pinned >= 0) { ev[e * A + a] = pinned; continue; }
+        int keff = k;
+        if (!collapsed && cst[a] && keff >= 1) keff = 0;  // plain phi draw
+        if (keff == 0) {
+          ev[e * A + a] = alias_draw_cpu(u_a1, u_a2, php[a], pha[a], V);
+          continue;
+        }
+        if (keff == 1) {
+          const int32_t x = buf[0].first;
+          const double th = th_[a * F + (int)buf[0].second];
+          if (cst[a]) {  // collapsed closed form: P(base) = theta
+            ev[e * A + a] = u_mix < th
+                ? alias_draw_cpu(u_a1, u_a2, php[a], pha[a], V)
+                : x;
+            continue;
+          }
+          const double Z1 = ptot[a * (kmax + 1) + 1];
+          const int64_t lo = rptr[vo[a] + x], hi = rptr[vo[a] + x + 1];
+          cum.clear();
+          double c = 0.0;
+          for (int64_t j = lo; j < hi; ++j) {
+            double w = rexp[j];
+            if (collapsed && rcol[j] == x)
+              w += (1.0 / th - 1.0) / (phi_[vo[a] + x] * nrm[vo[a] + x]);
+            const double wgt =
+                (phi_[vo[a] + rcol[j]] * nrm[vo[a] + rcol[j]] / Z1) * (w - 1.0);
+            c += wgt;
+            cum.push_back(c);
+          }
+          const double tot = c;
+          if (u_mix < 1.0 / (1.0 + tot)) {
+            ev[e * A + a] = alias_draw_cpu(u_a1, u_a2, pwp[a], pwa[a], V);
+          } else {
+            const double target = u_sel * tot;
+            int64_t j = std::upper_bound(cum.begin(), cum.end(), target) -
+                        cum.begin();
+            if (j > (int64_t)cum.size() - 1) j = (int64_t)cum.size() - 1;
+            ev[e * A + a] = rcol[lo + j];
+          }
+          continue;
+        }
+        if (keff > kmax) {  // beyond cached powers: python oracle fallback
+          fb_per_thread[tid].push_back(e * A + a);
+          continue;
+        }
+        // ---- k >= 2: union-combine the records' factor rows --------------
+        // entries in record order; stable sort by col keeps the product
+        // order identical to numpy's lexsort + multiply.reduceat
+        const int64_t n_obs = (int64_t)buf.size();
+        std::vector<std::pair<int32_t, double>> ents;
+        for (int64_t m = 0; m < n_obs; ++m) {
+          const int32_t x = buf[m].first;
+          const double th = th_[a * F + (int)buf[m].second];
+          if (cst[a]) {
+            ents.emplace_back(x, 1.0 + (1.0 / th - 1.0) / phi_[vo[a] + x]);
+          } else {
+            const int64_t lo = rptr[vo[a] + x], hi = rptr[vo[a] + x + 1];
+            for (int64_t j = lo; j < hi; ++j) {
+              double w = rexp[j];
+              if (collapsed && rcol[j] == x)
+                w += (1.0 / th - 1.0) / (phi_[vo[a] + x] * nrm[vo[a] + x]);
+              ents.emplace_back(rcol[j], w);
+            }
+          }
+        }
+        std::stable_sort(ents.begin(), ents.end(),
+                         [](const std::pair<int32_t, double>& p,
+                            const std::pair<int32_t, double>& q) {
+                           return p.first < q.first;
+                         });
+        // run-reduce products per distinct col, then weights + CDF
+        cum.clear();
+        std::vector<int32_t> ucol;
+        const double Zk = cst[a] ? 1.0 : ptot[a * (kmax + 1) + keff];
+        double c = 0.0;
+        for (size_t i = 0; i < ents.size();) {
+          const int32_t v = ents[i].first;
+          double vw = ents[i].second;
+          for (++i; i < ents.size() && ents[i].first == v; ++i) vw *= ents[i].second;
+          double basep;
+          if (cst[a]) basep = phi_[vo[a] + v];
+          else
+            basep = phi_[vo[a] + v] * std::pow(nrm[vo[a] + v], (double)keff) / Zk;
+          c += basep * (vw - 1.0);
+          cum.push_back(c);
+          ucol.push_back(v);
+        }
+        const double tot = c;
+        if (u_mix < 1.0 / (1.0 + tot)) {
+          if (cst[a])
+            ev[e * A + a] = alias_draw_cpu(u_a1, u_a2, php[a], pha[a], V);
+          else
+            ev[e * A + a] = alias_draw_cpu(
+                u_a1, u_a2, pwp[a] + (int64_t)(keff - 1) * V,
+                pwa[a] + (int64_t)(keff - 1) * V, V);
+        } else {
+          const double target = u_sel * tot;
+          int64_t j = std::upper_bound(cum.begin(), cum.end(), target) -
+                      cum.begin();
+          if (j > (int64_t)cum.size() - 1) j = (int64_t)cum.size() - 1;
+          ev[e * A + a] = ucol[j];
+        }
+      }
+    }
+  }
+  std::vector<int64_t> fb;
+  for (auto& v : fb_per_thread) fb.insert(fb.end(), v.begin(), v.end());
+  std::sort(fb.begin(), fb.end());
+  auto fbt = torch::empty({(int64_t)fb.size()},
+                          torch::TensorOptions().dtype(torch::kInt64));
+  std::copy(fb.begin(), fb.end(), fbt.data_ptr<int64_t>());
+  return {new_ev, fbt};
 }
 
 }  // namespace dblink
