@@ -1259,31 +1259,32 @@ __device__ void value_update_pair(const ValueArgs& args, int64_t pair, int lane,
       const int64_t row_hi = args.csr_row_ptr[v0 + ux + 1];
       for (int64_t j = row_lo + lane; j < row_hi; j += WAVE) {
         const int32_t v = args.csr_col[j];
-        // dedupe: skip if an earlier unit's row already contains v
+        // one ascending pass fuses the dedupe test (an EARLIER unit's row
+        // containing v means that unit owns the value) with the log-factor
+        // accumulation: one binary search per (entry, unit) instead of two
         bool first = true;
-        for (int u2 = 0; u2 < u && first; ++u2) {
-          int32_t x2, m2, f2;
-          unit_of(u2, &x2, &m2, &f2);
-          const int64_t lo2 = args.csr_row_ptr[v0 + x2];
-          const int64_t hi2 = args.csr_row_ptr[v0 + x2 + 1];
-          if (contains_i32(args.csr_col, lo2, hi2, v)) first = false;
-        }
-        if (!first) continue;
-        // accumulate log factors over ALL units
         float L = 0.0f;
         for (int u2 = 0; u2 < n_units; ++u2) {
           int32_t x2, m2, f2;
           unit_of(u2, &x2, &m2, &f2);
+          float s2;
+          if (u2 == u) {
+            s2 = args.csr_sim[j];
+          } else {
+            const int64_t lo2 = args.csr_row_ptr[v0 + x2];
+            const int64_t hi2 = args.csr_row_ptr[v0 + x2 + 1];
+            const int64_t p2 = lower_bound_i32(args.csr_col, lo2, hi2, v);
+            const bool found = p2 < hi2 && args.csr_col[p2] == v;
+            if (found && u2 < u) { first = false; break; }
+            s2 = found ? args.csr_sim[p2] : 0.0f;
+          }
           const float se2 = se_of(x2, f2);
-          const float s2 = (u2 == u)
-                               ? args.csr_sim[j]
-                               : sim_lookup(args.csr_row_ptr, args.csr_col,
-                                            args.csr_sim, v0 + x2, v);
           if (v == x2 && se2 > 0.0f)
             L += (float)m2 * __logf(__expf(s2) + se2);
           else if (s2 != 0.0f)
             L += (float)m2 * s2;
         }
+        if (!first) continue;
         const float log_expm1 = L + __logf(1.0f - __expf(-L));
         const float logw = log_base_prob(v) + log_expm1;
         W += (logw < 80.0f) ? (double)__expf(logw) : exp((double)logw);
