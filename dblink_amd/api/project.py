@@ -267,8 +267,32 @@ class SampleStep:
         self.sampler = sampler
         self.checkpoint_interval = checkpoint_interval
 
+    def scale_warning(self, num_records=None):
+        """PCG-II / Gibbs-Sequential score every record against every entity
+        of its partition (GibbsUpdates.scala:363-395, 434-466) — quadratic
+        by design. Returns a warning string above a size threshold."""
+        if self.sampler not in ("PCG-II", "Gibbs-Sequential"):
+            return None
+        try:
+            n = num_records if num_records is not None else self.p.table.num_records
+        except Exception:
+            return None
+        parts = max(1, getattr(self.p.partitioner, "num_partitions", 1))
+        per = n // parts
+        if per > 20000:
+            return (
+                f"sampler {self.sampler} is quadratic in partition size "
+                f"(~{per} records/entities per partition here): sweeps will "
+                "be orders of magnitude slower than PCG-I; increase the "
+                "partitioner's numLevels or use PCG-I"
+            )
+        return None
+
     def execute(self):
         log.info(self.mk_string())
+        warn = self.scale_warning()
+        if warn:
+            log.warning(warn)
         state = (self.p.saved_state() if self.resume else None) or self.p.generate_initial_state()
         flags = SamplerFlags.for_sampler(self.sampler)
         sampler_m.sample(
@@ -540,6 +564,10 @@ def check_config(path):
     errs, warns = native_limit_problems(project, table)
     problems.extend(errs)
     warnings.extend(warns)
+    for s in steps:
+        w = s.scale_warning(n) if isinstance(s, SampleStep) else None
+        if w:
+            warnings.append(w)
     for msg in warnings:
         print(f"WARNING: {msg}")
     if problems:
