@@ -895,7 +895,8 @@ constexpr int VAL_STRIDE = 8;  // pairs examined per wave in kobs mode
 // kobs-mode kernels (the monolithic kernel sat at 118 VGPRs = 4 waves/SIMD).
 template <int VMODE>
 __device__ void value_update_pair(const ValueArgs& args, int64_t pair, int lane,
-                                  int32_t* keys, float* vals, int32_t* gbuf);
+                                  int32_t* keys, float* vals, int32_t* gbuf,
+                                  int64_t* glo, int32_t* gn, float* gse);
 
 template <int VMODE>
 __global__ void __launch_bounds__(WAVES_PER_BLOCK_VAL * WAVE)
@@ -907,6 +908,10 @@ value_update_kernel_t(ValueArgs args) {
   __shared__ int32_t h_key[WAVES_PER_BLOCK_VAL][HASH_CAP];
   __shared__ float h_val[WAVES_PER_BLOCK_VAL][HASH_CAP];
   __shared__ int32_t g_buf[WAVES_PER_BLOCK_VAL][3 * VAL_DMAX + 1];
+  // merge-path per-unit hoists: (row_lo, row_len, se) per distinct group
+  __shared__ int64_t g_lo[WAVES_PER_BLOCK_VAL][VAL_DMAX];
+  __shared__ int32_t g_n[WAVES_PER_BLOCK_VAL][VAL_DMAX];
+  __shared__ float g_se[WAVES_PER_BLOCK_VAL][VAL_DMAX];
 
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave = threadIdx.x / WAVE;
@@ -914,7 +919,8 @@ value_update_kernel_t(ValueArgs args) {
   if (args.pair_list != nullptr) {
     if (widx >= args.n_pairs) return;
     value_update_pair<VMODE>(args, args.pair_list[widx], lane, h_key[wave],
-                             h_val[wave], g_buf[wave]);
+                             h_val[wave], g_buf[wave], g_lo[wave], g_n[wave],
+                             g_se[wave]);
   } else {
     // kobs self-selection: each wave examines VAL_STRIDE consecutive pairs
     // and runs the (rare) k >= 2 ones serially
@@ -922,14 +928,15 @@ value_update_kernel_t(ValueArgs args) {
     for (int64_t pair = p0; pair < p0 + VAL_STRIDE && pair < args.n_pairs; ++pair) {
       if (args.kobs[pair] >= 2)
         value_update_pair<VMODE>(args, pair, lane, h_key[wave], h_val[wave],
-                                 g_buf[wave]);
+                                 g_buf[wave], g_lo[wave], g_n[wave], g_se[wave]);
     }
   }
 }
 
 template <int VMODE>
 __device__ void value_update_pair(const ValueArgs& args, int64_t pair, int lane,
-                                  int32_t* keys, float* vals, int32_t* gbuf) {
+                                  int32_t* keys, float* vals, int32_t* gbuf,
+                                  int64_t* glo, int32_t* gn, float* gse) {
   const int64_t e = pair / args.A;
   const int a = (int)(pair % args.A);
   const bool is_const = args.attr_const[a];
@@ -1224,6 +1231,18 @@ __device__ void value_update_pair(const ValueArgs& args, int64_t pair, int lane,
     // union-merge path (hash would overflow): iterate every entry of every
     // DISTINCT row, but process a value only from the FIRST unit containing
     // it; full L_v comes from binary searches in the other units' rows.
+    // Per-unit metadata (row bounds, the theta-dependent self term) is
+    // hoisted into LDS once per pair — the entry loop touches ~10^2-10^3
+    // values per unit and must not recompute it per (entry, unit).
+    const bool hoist = !gover && !is_const;
+    if (hoist && lane == 0) {
+      for (int u = 0; u < n_units; ++u) {
+        const int32_t x2 = gbuf[u];
+        glo[u] = args.csr_row_ptr[v0 + x2];
+        gn[u] = (int32_t)(args.csr_row_ptr[v0 + x2 + 1] - glo[u]);
+        gse[u] = se_of(x2, gbuf[VAL_DMAX + u]);
+      }
+    }
     for (int u = 0; u < n_units; ++u) {
       int32_t ux, um, uf;
       unit_of(u, &ux, &um, &uf);
@@ -1265,20 +1284,38 @@ __device__ void value_update_pair(const ValueArgs& args, int64_t pair, int lane,
         bool first = true;
         float L = 0.0f;
         for (int u2 = 0; u2 < n_units; ++u2) {
-          int32_t x2, m2, f2;
-          unit_of(u2, &x2, &m2, &f2);
+          int32_t x2, m2;
+          float se2;
           float s2;
-          if (u2 == u) {
-            s2 = args.csr_sim[j];
+          if (hoist) {
+            x2 = gbuf[u2];
+            m2 = gbuf[2 * VAL_DMAX + u2];
+            se2 = gse[u2];
+            if (u2 == u) {
+              s2 = args.csr_sim[j];
+            } else {
+              const int64_t lo2 = glo[u2];
+              const int64_t hi2 = lo2 + gn[u2];
+              const int64_t p2 = lower_bound_i32(args.csr_col, lo2, hi2, v);
+              const bool found = p2 < hi2 && args.csr_col[p2] == v;
+              if (found && u2 < u) { first = false; break; }
+              s2 = found ? args.csr_sim[p2] : 0.0f;
+            }
           } else {
-            const int64_t lo2 = args.csr_row_ptr[v0 + x2];
-            const int64_t hi2 = args.csr_row_ptr[v0 + x2 + 1];
-            const int64_t p2 = lower_bound_i32(args.csr_col, lo2, hi2, v);
-            const bool found = p2 < hi2 && args.csr_col[p2] == v;
-            if (found && u2 < u) { first = false; break; }
-            s2 = found ? args.csr_sim[p2] : 0.0f;
+            int32_t f2;
+            unit_of(u2, &x2, &m2, &f2);
+            if (u2 == u) {
+              s2 = args.csr_sim[j];
+            } else {
+              const int64_t lo2 = args.csr_row_ptr[v0 + x2];
+              const int64_t hi2 = args.csr_row_ptr[v0 + x2 + 1];
+              const int64_t p2 = lower_bound_i32(args.csr_col, lo2, hi2, v);
+              const bool found = p2 < hi2 && args.csr_col[p2] == v;
+              if (found && u2 < u) { first = false; break; }
+              s2 = found ? args.csr_sim[p2] : 0.0f;
+            }
+            se2 = se_of(x2, f2);
           }
-          const float se2 = se_of(x2, f2);
           if (v == x2 && se2 > 0.0f)
             L += (float)m2 * __logf(__expf(s2) + se2);
           else if (s2 != 0.0f)
