@@ -238,6 +238,38 @@ class GpuModel:
         self.ktab_rawsum = rawsum
         self.ktab_max = kmax
 
+        # 2-distinct-value tables: raw_{k,m}(x-row entry c) =
+        # phi(c) norm(c)^k (e^{m s} - 1) for 2 <= k <= k2max, 1 <= m < k
+        # (level L = (k-2)(k-1)/2 + m - 1). The d=2 class then draws from
+        # two table CDFs plus an exact residual over the row intersection.
+        self.k2tab_excl = self.k2tab_rawsum = None
+        self.k2tab_max = 0
+        cap2 = int(os.environ.get("DBLINK_KTAB2_MAX_BYTES", str(64 << 30)))
+        k2 = 2
+        while k2 < min(kmax, 5) and \
+                ((k2 + 1) * k2 // 2) * 16 * (nnz + self.Vtot) <= cap2:
+            k2 += 1
+        lv2 = k2 * (k2 - 1) // 2
+        if lv2 * 16 * (nnz + self.Vtot) > cap2 or lv2 < 1:
+            return
+        excl2 = torch.empty((lv2, nnz), dtype=torch.float64, device=device)
+        rawsum2 = torch.zeros((lv2, self.Vtot), dtype=torch.float64, device=device)
+        for k in range(2, k2 + 1):
+            for m in range(1, k):
+                L = (k - 2) * (k - 1) // 2 + m - 1
+                raw = torch.exp(base_log + (k - 1) * ln_g) * torch.expm1(m * sim)
+                c = torch.cumsum(raw, 0)
+                base_c = torch.where(start > 0, c[torch.clamp(start - 1, min=0)],
+                                     torch.zeros((), dtype=torch.float64,
+                                                 device=device))
+                excl2[L] = c - raw - torch.repeat_interleave(base_c, row_counts)
+                rawsum2[L] = torch.where(
+                    row_counts > 0, c[torch.clamp(ends, min=0)],
+                    torch.zeros((), dtype=torch.float64, device=device)) - base_c
+        self.k2tab_excl = excl2
+        self.k2tab_rawsum = rawsum2
+        self.k2tab_max = k2
+
 
 class GpuStateTensors:
     """Device mirror of ChainState's mutable arrays."""
@@ -304,16 +336,21 @@ class GpuEngine(CpuEngine):
         self.device = device
         self.C = _require_native()
         self.model = GpuModel(cache, device, cache_kc(cache, partitioner))
+        zt = torch.empty(0, dtype=torch.float64)
         if self.model.ktab_max >= 2 and os.environ.get("DBLINK_KTAB", "1") != "0":
             self.C.set_value_ktables(
                 self.model.ktab_excl, self.model.ktab_rawsum,
                 self.model.self_expsim, self.model.ktab_max,
                 int(self.model.csr_col.numel()))
         else:
-            self.C.set_value_ktables(
-                torch.empty(0, dtype=torch.float64),
-                torch.empty(0, dtype=torch.float64),
-                torch.empty(0, dtype=torch.float32), 0, 0)
+            self.C.set_value_ktables(zt, zt, torch.empty(0, dtype=torch.float32),
+                                     0, 0)
+        if self.model.k2tab_max >= 2 and os.environ.get("DBLINK_KTAB2", "1") != "0":
+            self.C.set_value_k2tables(self.model.k2tab_excl,
+                                      self.model.k2tab_rawsum,
+                                      self.model.k2tab_max)
+        else:
+            self.C.set_value_k2tables(zt, zt, 0)
         self.flat_tree = None
         self._gs = None
         self._ent_id_base = rank << 40

@@ -138,6 +138,7 @@ void build_keys(torch::Tensor ent_part, torch::Tensor ent_values,
 void set_value_stats(torch::Tensor t);
 void set_value_ktables(torch::Tensor excl, torch::Tensor rawsum,
                        torch::Tensor self_expsim, int64_t kmax, int64_t nnz);
+void set_value_k2tables(torch::Tensor excl, torch::Tensor rawsum, int64_t k2max);
 void mfma_score_bench(torch::Tensor rcode, torch::Tensor rbonus,
                       torch::Tensor ecode, int64_t K, torch::Tensor score);
 void scalar_score_bench(torch::Tensor rcode, torch::Tensor rbonus,
@@ -194,6 +195,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "install the optional value-phase work-counter buffer");
   m.def("set_value_ktables", &dblink::set_value_ktables,
         "install the k>=2 single-value perturbation tables");
+  m.def("set_value_k2tables", &dblink::set_value_k2tables,
+        "install the two-distinct-value (k, m) perturbation tables");
   m.def("mfma_score_bench", &dblink::mfma_score_bench,
         "MFMA one-hot categorical scorer (experiment)");
   m.def("scalar_score_bench", &dblink::scalar_score_bench,

@@ -618,6 +618,10 @@ struct ValueArgs {
   const float* self_expsim;    // [Vtot]
   int ktab_max;                // largest k covered (0 = disabled)
   int64_t nnz;
+  // 2-distinct-value (k, m) tables: level (k-2)(k-1)/2 + m - 1
+  const double* tab2_excl;
+  const double* tab2_rawsum;
+  int k2tab_max;
   const int32_t* rec_values;
   const uint8_t* rec_dist;
   const int32_t* rec_file;
@@ -882,6 +886,163 @@ __global__ void value_update_kd1_kernel(ValueArgs args) {
       if (jidx < row_lo) jidx = row_lo;
       if (jidx >= row_hi) jidx = row_hi - 1;
       v_new = args.csr_col[jidx];
+    }
+  }
+  args.ent_values[e * args.A + a] = (int32_t)v_new;
+}
+
+// Thread-per-pair value update for clusters spanning exactly TWO distinct
+// values of an attribute (the remaining k >= 2 mass after kd1): the
+// perturbation decomposes into two PRECOMPUTED single-row measures
+// phi norm^k (e^{m_i s_i} - 1) (set_value_k2tables, level (k-2)(k-1)/2+m-1)
+// plus an exact non-negative residual supported on the two self points and
+// the row intersection:
+//   w(v) = base_k(v)(F1 F2 - 1)
+//        = plain1(v) + plain2(v) + base_k(v)(F1-1)(F2-1) [+ self-se boosts]
+// Component sampling: two table CDF searches or a replayed residual walk —
+// no per-entry Philox/exp over the row union.
+__global__ void value_update_kd2_kernel(ValueArgs args) {
+  if (args.ctrl != nullptr) {
+    args.seed = (uint64_t)args.ctrl[0];
+    args.iteration = (uint32_t)args.ctrl[1];
+  }
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= args.n_pairs) return;
+  const int64_t pair = i;
+  const int k = args.kobs[pair];
+  if (k < 2 || k > args.k2tab_max) return;
+  const int64_t e = pair / args.A;
+  const int a = (int)(pair % args.A);
+  if (args.attr_const[a]) return;
+  const int64_t v0 = args.voff[a];
+  const int V = (int)(args.voff[a + 1] - v0);
+  const int64_t Vtot = args.voff[args.A];
+
+  // exactly two first-seen (value, file) groups with DISTINCT values
+  const int64_t r_lo = args.ent_rec_ptr[e], r_hi = args.ent_rec_ptr[e + 1];
+  int32_t x1 = -1, f1 = -1, x2 = -1, f2 = -1;
+  int m1 = 0, m2 = 0;
+  for (int64_t j = r_lo; j < r_hi; ++j) {
+    const int64_t r = args.ent_rec_idx[j];
+    const int32_t x = args.rec_values[r * args.A + a];
+    if (x < 0) continue;
+    if (!args.collapsed && !args.rec_dist[r * args.A + a]) return;
+    const int32_t ff = args.collapsed ? args.rec_file[r] : 0;
+    if (x1 < 0 || (x == x1 && ff == f1)) { x1 = x; f1 = ff; ++m1; }
+    else if (x2 < 0 || (x == x2 && ff == f2)) { x2 = x; f2 = ff; ++m2; }
+    else return;  // three or more groups: wave path
+  }
+  if (x2 < 0 || x1 == x2) return;  // d != 2 distinct values
+
+  const int lb = (k - 2) * (k - 1) / 2 - 1;
+  const double T1 = args.tab2_rawsum[(int64_t)(lb + m1) * Vtot + v0 + x1];
+  const double T2 = args.tab2_rawsum[(int64_t)(lb + m2) * Vtot + v0 + x2];
+
+  auto braw = [&](int32_t v) -> double {
+    return exp((double)args.log_phi[v0 + v] +
+               (double)k * (double)args.log_norm[v0 + v]);
+  };
+  auto se_at = [&](int32_t x, int f) -> double {
+    if (!args.collapsed) return 0.0;
+    const double th = (double)args.theta[a * args.F + f];
+    return (1.0 / th - 1.0) /
+           ((double)args.phi[v0 + x] * (double)args.norm_lin[v0 + x]);
+  };
+  const double se1 = se_at(x1, f1), se2 = se_at(x2, f2);
+
+  // residual at a self point v = xs (own group s boosted by se, other group
+  // contributes via its sim to xs, never boosted since x1 != x2)
+  auto resid_self = [&](int32_t xs, double ses, int ms, int32_t xo,
+                        int mo) -> double {
+    const double es = (double)args.self_expsim[v0 + xs];
+    const float so = sim_lookup(args.csr_row_ptr, args.csr_col, args.csr_sim,
+                                v0 + xo, xs);
+    const double Fs = pow(es + ses, (double)ms);
+    const double Fo = exp((double)so * (double)mo);
+    double r = (Fs * Fo - 1.0) - (pow(es, (double)ms) - 1.0);
+    if (so != 0.0f) r -= Fo - 1.0;
+    r *= braw(xs);
+    return r > 0.0 ? r : 0.0;
+  };
+
+  // residual walk: self points first, then the shorter row's entries that
+  // also appear in the other row (skipping the self points)
+  const int64_t lo1 = args.csr_row_ptr[v0 + x1], hi1 = args.csr_row_ptr[v0 + x1 + 1];
+  const int64_t lo2 = args.csr_row_ptr[v0 + x2], hi2 = args.csr_row_ptr[v0 + x2 + 1];
+  const bool first_shorter = (hi1 - lo1) <= (hi2 - lo2);
+  const int64_t wlo = first_shorter ? lo1 : lo2;
+  const int64_t whi = first_shorter ? hi1 : hi2;
+  const int64_t olo = first_shorter ? lo2 : lo1;
+  const int64_t ohi = first_shorter ? hi2 : hi1;
+  const int mw = first_shorter ? m1 : m2;
+  const int mo = first_shorter ? m2 : m1;
+
+  // pass callback: accumulate (and optionally select at a target)
+  auto residual_pass = [&](double target, int32_t* out_v) -> double {
+    double R = 0.0;
+    R += resid_self(x1, se1, m1, x2, m2);
+    if (out_v != nullptr && R >= target && R > 0.0) { *out_v = x1; return R; }
+    R += resid_self(x2, se2, m2, x1, m1);
+    if (out_v != nullptr && R >= target) { *out_v = x2; return R; }
+    for (int64_t j = wlo; j < whi; ++j) {
+      const int32_t v = args.csr_col[j];
+      if (v == x1 || v == x2) continue;
+      const int64_t p = lower_bound_i32(args.csr_col, olo, ohi, v);
+      if (p >= ohi || args.csr_col[p] != v) continue;
+      const double Fw = exp((double)args.csr_sim[j] * (double)mw) - 1.0;
+      const double Fo = exp((double)args.csr_sim[p] * (double)mo) - 1.0;
+      R += braw(v) * Fw * Fo;
+      if (out_v != nullptr && R >= target) { *out_v = v; return R; }
+    }
+    return R;
+  };
+  const double R = residual_pass(0.0, nullptr);
+
+  const double W = T1 + T2 + R;
+  const double Zk = exp((double)args.log_pow_total[a * (args.Kc + 1) + k]);
+  const double Wn = W / Zk;
+
+  const uint64_t elem = (args.ent_id_base + (uint64_t)e) * 32u + (uint64_t)a;
+  u32x4 rnd = philox4x32(args.seed, (uint32_t)elem, (uint32_t)(elem >> 32),
+                         args.iteration ^ (PH_VALM << 24), 0xFFFF0000u);
+  const double u_mix = ((double)rnd.x + 0.5) * 2.3283064365386963e-10;
+  const double u_sel = ((double)rnd.y + 0.5) * 2.3283064365386963e-10;
+  const float u_a1 = u32_to_uniform(rnd.z);
+  const float u_a2 = u32_to_uniform(rnd.w);
+
+  int v_new;
+  if (u_mix < 1.0 / (1.0 + Wn) || W <= 0.0) {
+    const int64_t off = args.pow_off[a] + (int64_t)(k - 1) * V;
+    v_new = alias_draw(args.pow_prob + off, args.pow_alias + off, V, u_a1, u_a2);
+  } else {
+    double t = u_sel * W;
+    if (t < T1) {
+      const double* ex = args.tab2_excl + (int64_t)(lb + m1) * args.nnz;
+      int64_t lo = lo1, hi = hi1;
+      while (lo < hi) {
+        const int64_t mid = (lo + hi) >> 1;
+        if (ex[mid] <= t) lo = mid + 1; else hi = mid;
+      }
+      int64_t jidx = lo - 1;
+      if (jidx < lo1) jidx = lo1;
+      if (jidx >= hi1) jidx = hi1 - 1;
+      v_new = args.csr_col[jidx];
+    } else if (t < T1 + T2) {
+      t -= T1;
+      const double* ex = args.tab2_excl + (int64_t)(lb + m2) * args.nnz;
+      int64_t lo = lo2, hi = hi2;
+      while (lo < hi) {
+        const int64_t mid = (lo + hi) >> 1;
+        if (ex[mid] <= t) lo = mid + 1; else hi = mid;
+      }
+      int64_t jidx = lo - 1;
+      if (jidx < lo2) jidx = lo2;
+      if (jidx >= hi2) jidx = hi2 - 1;
+      v_new = args.csr_col[jidx];
+    } else {
+      int32_t sel = x1;
+      residual_pass(t - T1 - T2, &sel);
+      v_new = sel;
     }
   }
   args.ent_values[e * args.A + a] = (int32_t)v_new;
@@ -1164,6 +1325,10 @@ __device__ void value_update_pair(const ValueArgs& args, int64_t pair, int lane,
   // (kobs mode only; explicit pair lists keep the full path)
   if (args.kobs != nullptr && args.tab_excl != nullptr && !is_const &&
       !gover && d == 1 && k_obs <= args.ktab_max && k_obs <= args.Kc)
+    return;
+  // two distinct VALUES with cached (k, m) tables: the kd2 kernel's
+  if (args.kobs != nullptr && args.tab2_excl != nullptr && !is_const &&
+      !gover && d == 2 && k_obs <= args.k2tab_max && gbuf[0] != gbuf[1])
     return;
 
   double W = 0.0;            // total perturbation weight
@@ -2322,6 +2487,24 @@ static const float* g_self_expsim = nullptr;
 static int g_ktab_max = 0;
 static int64_t g_tab_nnz = 0;
 static std::vector<torch::Tensor> g_ktab_keep;
+static const double* g_tab2_excl = nullptr;
+static const double* g_tab2_rawsum = nullptr;
+static int g_k2tab_max = 0;
+static std::vector<torch::Tensor> g_k2tab_keep;
+
+void set_value_k2tables(torch::Tensor excl, torch::Tensor rawsum, int64_t k2max) {
+  if (k2max >= 2 && excl.numel()) {
+    g_k2tab_keep = {excl, rawsum};
+    g_tab2_excl = excl.data_ptr<double>();
+    g_tab2_rawsum = rawsum.data_ptr<double>();
+    g_k2tab_max = (int)k2max;
+  } else {
+    g_k2tab_keep.clear();
+    g_tab2_excl = nullptr;
+    g_tab2_rawsum = nullptr;
+    g_k2tab_max = 0;
+  }
+}
 
 // Install the k>=2 single-distinct-value perturbation tables (one level per
 // k in [2, kmax]); pairs whose linked records all share one (value, file)
@@ -2376,6 +2559,9 @@ static ValueArgs make_value_args(
   a.self_expsim = g_self_expsim;
   a.ktab_max = g_ktab_max;
   a.nnz = g_tab_nnz;
+  a.tab2_excl = g_tab2_excl;
+  a.tab2_rawsum = g_tab2_rawsum;
+  a.k2tab_max = g_k2tab_max;
   a.csr_excl = csr_excl.numel() ? csr_excl.data_ptr<double>() : nullptr;
   a.csr_rawsum = csr_rawsum.data_ptr<double>();
   a.z1 = z1.data_ptr<double>();
@@ -2456,6 +2642,9 @@ void value_update(
                        at::cuda::getCurrentCUDAStream(), args);
     if (args.tab_excl != nullptr)
       hipLaunchKernelGGL(value_update_kd1_kernel, tgrid, dim3(256), 0,
+                         at::cuda::getCurrentCUDAStream(), args);
+    if (args.tab2_excl != nullptr)
+      hipLaunchKernelGGL(value_update_kd2_kernel, tgrid, dim3(256), 0,
                          at::cuda::getCurrentCUDAStream(), args);
     const int64_t n_waves = (args.n_pairs + VAL_STRIDE - 1) / VAL_STRIDE;
     dim3 wgrid((unsigned)wave_grid(n_waves, WAVES_PER_BLOCK_VAL));

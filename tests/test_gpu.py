@@ -1000,3 +1000,66 @@ def test_value_kd1_kernel_distribution():
     # reset globals so later direct-kernel tests see a clean slate
     z = torch.empty(0, dtype=torch.float64)
     C.set_value_ktables(z, z, torch.empty(0, dtype=torch.float32), 0, 0)
+
+
+@gpu
+def test_value_kd2_kernel_distribution():
+    """k=2 clusters with TWO distinct values (the kd2 table path): empirical
+    frequencies vs the exact mixture with w(v) = base2(v) (F1 F2 - 1),
+    F_i boosted by the collapsed self term at v = x_i."""
+    cache, model = make_model(DEV)
+    assert model.k2tab_max >= 2
+    a = 1
+    idx = cache.indexed_attributes[a].index
+    x1, x2 = 0, 1
+    th = 0.1
+    model.theta.copy_(torch.full((2, 1), th))
+    N = 40000
+    rec_values = np.tile(np.array([[-1, x1], [-1, x2]], dtype=np.int32), (N, 1))
+    rec_dist = np.ones((2 * N, 2), dtype=np.uint8)
+    ent_vals = np.zeros((N, 2), dtype=np.int32)
+    ent_rec_ptr = np.arange(0, 2 * N + 1, 2, dtype=np.int64)
+    ent_rec_idx = np.arange(2 * N, dtype=np.int64)
+    kobs = np.zeros((N, 2), dtype=np.int32)
+    kobs[:, a] = 2
+
+    C.set_value_ktables(model.ktab_excl, model.ktab_rawsum, model.self_expsim,
+                        model.ktab_max, int(model.csr_col.numel()))
+    C.set_value_k2tables(model.k2tab_excl, model.k2tab_rawsum, model.k2tab_max)
+    err = torch.zeros(1, dtype=torch.int32, device=DEV)
+    ev = _dev(ent_vals, torch.int32)
+    empty64 = torch.empty(0, dtype=torch.int64, device=DEV)
+    C.value_update(
+        _dev(rec_values, torch.int32), _dev(rec_dist, torch.uint8),
+        _dev(np.zeros(2 * N, np.int32), torch.int32),
+        _dev(ent_rec_ptr, torch.int64), _dev(ent_rec_idx, torch.int64), ev,
+        model.theta, model.phi, model.log_phi, model.norm_lin, model.log_norm,
+        model.voff, model.csr_row_ptr, model.csr_col, model.csr_sim,
+        model.phi_prob, model.phi_alias, model.pow_prob, model.pow_alias,
+        model.pow_off, model.log_pow_total, model.attr_const, model.Kc,
+        1, 0, 909, 4, 0, err, empty64, empty64, empty64,
+        model.csr_excl, model.csr_rawsum, model.z1, empty64,
+        _dev(kobs.reshape(-1), torch.int32))
+    sel = ev.cpu().numpy()[:, a]
+
+    V = idx.num_values
+    phi = idx.probs
+    norms = idx.sim_norms
+    z2 = idx.sim_norm_total(2)
+    base2 = phi * norms ** 2 / z2
+    F1 = np.ones(V)
+    F2 = np.ones(V)
+    si = idx.sim_index
+    for x, F, se in ((x1, F1, (1/th - 1)/(phi[x1]*norms[x1])),
+                     (x2, F2, (1/th - 1)/(phi[x2]*norms[x2]))):
+        for j in range(si.row_ptr[x], si.row_ptr[x + 1]):
+            c = si.col[j]
+            F[c] = si.expsim[j] + (se if c == x else 0.0)
+    w = base2 * (F1 * F2 - 1.0)
+    W = w.sum()
+    exact = (base2 + w) / (1.0 + W)
+    emp = np.bincount(sel, minlength=V) / N
+    assert tv_distance(emp, exact) < 0.02, tv_distance(emp, exact)
+    z = torch.empty(0, dtype=torch.float64)
+    C.set_value_ktables(z, z, torch.empty(0, dtype=torch.float32), 0, 0)
+    C.set_value_k2tables(z, z, 0)
