@@ -891,7 +891,7 @@ __global__ void value_update_kd1_kernel(ValueArgs args) {
   args.ent_values[e * args.A + a] = (int32_t)v_new;
 }
 
-// Thread-per-pair value update for clusters spanning exactly TWO distinct
+// Wave-per-pair value update for clusters spanning exactly TWO distinct
 // values of an attribute (the remaining k >= 2 mass after kd1): the
 // perturbation decomposes into two PRECOMPUTED single-row measures
 // phi norm^k (e^{m_i s_i} - 1) (set_value_k2tables, level (k-2)(k-1)/2+m-1)
@@ -899,16 +899,13 @@ __global__ void value_update_kd1_kernel(ValueArgs args) {
 // the row intersection:
 //   w(v) = base_k(v)(F1 F2 - 1)
 //        = plain1(v) + plain2(v) + base_k(v)(F1-1)(F2-1) [+ self-se boosts]
-// Component sampling: two table CDF searches or a replayed residual walk —
-// no per-entry Philox/exp over the row union.
-__global__ void value_update_kd2_kernel(ValueArgs args) {
-  if (args.ctrl != nullptr) {
-    args.seed = (uint64_t)args.ctrl[0];
-    args.iteration = (uint32_t)args.ctrl[1];
-  }
-  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= args.n_pairs) return;
-  const int64_t pair = i;
+// Component sampling: two table CDF binary searches, or a residual re-walk.
+// The residual walk strides the SHORTER row across the wave's 64 lanes with
+// an ordered f64 prefix scan (identical op order in the total and selection
+// passes), so it is deterministic and exact.
+
+__device__ void value_update_kd2_pair(const ValueArgs& args, int64_t pair,
+                                      int lane) {
   const int k = args.kobs[pair];
   if (k < 2 || k > args.k2tab_max) return;
   const int64_t e = pair / args.A;
@@ -919,6 +916,7 @@ __global__ void value_update_kd2_kernel(ValueArgs args) {
   const int64_t Vtot = args.voff[args.A];
 
   // exactly two first-seen (value, file) groups with DISTINCT values
+  // (wave-uniform scalar loads; every lane runs the same scan)
   const int64_t r_lo = args.ent_rec_ptr[e], r_hi = args.ent_rec_ptr[e + 1];
   int32_t x1 = -1, f1 = -1, x2 = -1, f2 = -1;
   int m1 = 0, m2 = 0;
@@ -930,7 +928,7 @@ __global__ void value_update_kd2_kernel(ValueArgs args) {
     const int32_t ff = args.collapsed ? args.rec_file[r] : 0;
     if (x1 < 0 || (x == x1 && ff == f1)) { x1 = x; f1 = ff; ++m1; }
     else if (x2 < 0 || (x == x2 && ff == f2)) { x2 = x; f2 = ff; ++m2; }
-    else return;  // three or more groups: wave path
+    else return;  // three or more groups: wave merge path
   }
   if (x2 < 0 || x1 == x2) return;  // d != 2 distinct values
 
@@ -950,8 +948,6 @@ __global__ void value_update_kd2_kernel(ValueArgs args) {
   };
   const double se1 = se_at(x1, f1), se2 = se_at(x2, f2);
 
-  // residual at a self point v = xs (own group s boosted by se, other group
-  // contributes via its sim to xs, never boosted since x1 != x2)
   auto resid_self = [&](int32_t xs, double ses, int ms, int32_t xo,
                         int mo) -> double {
     const double es = (double)args.self_expsim[v0 + xs];
@@ -965,8 +961,6 @@ __global__ void value_update_kd2_kernel(ValueArgs args) {
     return r > 0.0 ? r : 0.0;
   };
 
-  // residual walk: self points first, then the shorter row's entries that
-  // also appear in the other row (skipping the self points)
   const int64_t lo1 = args.csr_row_ptr[v0 + x1], hi1 = args.csr_row_ptr[v0 + x1 + 1];
   const int64_t lo2 = args.csr_row_ptr[v0 + x2], hi2 = args.csr_row_ptr[v0 + x2 + 1];
   const bool first_shorter = (hi1 - lo1) <= (hi2 - lo2);
@@ -977,26 +971,57 @@ __global__ void value_update_kd2_kernel(ValueArgs args) {
   const int mw = first_shorter ? m1 : m2;
   const int mo = first_shorter ? m2 : m1;
 
-  // pass callback: accumulate (and optionally select at a target)
-  auto residual_pass = [&](double target, int32_t* out_v) -> double {
-    double R = 0.0;
-    R += resid_self(x1, se1, m1, x2, m2);
-    if (out_v != nullptr && R >= target && R > 0.0) { *out_v = x1; return R; }
-    R += resid_self(x2, se2, m2, x1, m1);
-    if (out_v != nullptr && R >= target) { *out_v = x2; return R; }
-    for (int64_t j = wlo; j < whi; ++j) {
-      const int32_t v = args.csr_col[j];
-      if (v == x1 || v == x2) continue;
-      const int64_t p = lower_bound_i32(args.csr_col, olo, ohi, v);
-      if (p >= ohi || args.csr_col[p] != v) continue;
-      const double Fw = exp((double)args.csr_sim[j] * (double)mw) - 1.0;
-      const double Fo = exp((double)args.csr_sim[p] * (double)mo) - 1.0;
-      R += braw(v) * Fw * Fo;
-      if (out_v != nullptr && R >= target) { *out_v = v; return R; }
-    }
-    return R;
+  // one lane's residual contribution for walk element j (0 if not in the
+  // other row or a self point)
+  auto walk_elem = [&](int64_t j) -> double {
+    const int32_t v = args.csr_col[j];
+    if (v == x1 || v == x2) return 0.0;
+    const int64_t p = lower_bound_i32(args.csr_col, olo, ohi, v);
+    if (p >= ohi || args.csr_col[p] != v) return 0.0;
+    const double Fw = exp((double)args.csr_sim[j] * (double)mw) - 1.0;
+    const double Fo = exp((double)args.csr_sim[p] * (double)mo) - 1.0;
+    return braw(v) * Fw * Fo;
   };
-  const double R = residual_pass(0.0, nullptr);
+
+  const double r_self1 = resid_self(x1, se1, m1, x2, m2);
+  const double r_self2 = resid_self(x2, se2, m2, x1, m1);
+
+  // residual pass: lanes stride 64-wide chunks with an ordered inclusive
+  // f64 prefix (carry chained across chunks). When target >= 0 the pass
+  // selects the first element whose running total crosses it.
+  auto residual_pass = [&](double target, int32_t* out_v) -> double {
+    double run = r_self1 + r_self2;
+    if (out_v != nullptr) {
+      if (r_self1 >= target && r_self1 > 0.0) { *out_v = x1; return run; }
+      if (run >= target) { *out_v = x2; return run; }
+    }
+    for (int64_t cb = wlo; cb < whi; cb += WAVE) {
+      const int64_t j = cb + lane;
+      double mine = (j < whi) ? walk_elem(j) : 0.0;
+      // inclusive prefix across lanes (deterministic shfl ladder)
+      double pre = mine;
+#pragma unroll
+      for (int off = 1; off < WAVE; off <<= 1) {
+        const double v = __shfl_up(pre, off);
+        if (lane >= off) pre += v;
+      }
+      const double chunk_total = __shfl(pre, WAVE - 1);
+      if (out_v != nullptr && run + chunk_total >= target) {
+        // first lane whose running total crosses the target
+        const bool crossed = (run + pre >= target) && j < whi;
+        const unsigned long long mask = __ballot(crossed);
+        if (mask != 0ull) {
+          const int fl = __ffsll((long long)mask) - 1;
+          const int32_t v = args.csr_col[cb + fl];
+          *out_v = v;
+          return run + chunk_total;
+        }
+      }
+      run += chunk_total;
+    }
+    return run;
+  };
+  const double R = residual_pass(-1.0, nullptr);
 
   const double W = T1 + T2 + R;
   const double Zk = exp((double)args.log_pow_total[a * (args.Kc + 1) + k]);
@@ -1045,7 +1070,22 @@ __global__ void value_update_kd2_kernel(ValueArgs args) {
       v_new = sel;
     }
   }
-  args.ent_values[e * args.A + a] = (int32_t)v_new;
+  if (lane == 0) args.ent_values[e * args.A + a] = (int32_t)v_new;
+}
+
+__global__ void __launch_bounds__(WAVES_PER_BLOCK_VAL * WAVE)
+value_update_kd2_kernel(ValueArgs args) {
+  if (args.ctrl != nullptr) {
+    args.seed = (uint64_t)args.ctrl[0];
+    args.iteration = (uint32_t)args.ctrl[1];
+  }
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = threadIdx.x / WAVE;
+  const int64_t widx = (int64_t)blockIdx.x * WAVES_PER_BLOCK_VAL + wave;
+  const int64_t p0 = widx * 8;
+  for (int64_t pair = p0; pair < p0 + 8 && pair < args.n_pairs; ++pair) {
+    if (args.kobs[pair] >= 2) value_update_kd2_pair(args, pair, lane);
+  }
 }
 
 constexpr int VAL_STRIDE = 8;  // pairs examined per wave in kobs mode
@@ -2643,9 +2683,13 @@ void value_update(
     if (args.tab_excl != nullptr)
       hipLaunchKernelGGL(value_update_kd1_kernel, tgrid, dim3(256), 0,
                          at::cuda::getCurrentCUDAStream(), args);
-    if (args.tab2_excl != nullptr)
-      hipLaunchKernelGGL(value_update_kd2_kernel, tgrid, dim3(256), 0,
+    if (args.tab2_excl != nullptr) {
+      const int64_t nw2 = (args.n_pairs + 7) / 8;
+      hipLaunchKernelGGL(value_update_kd2_kernel,
+                         dim3((unsigned)wave_grid(nw2, WAVES_PER_BLOCK_VAL)),
+                         dim3(WAVES_PER_BLOCK_VAL * WAVE), 0,
                          at::cuda::getCurrentCUDAStream(), args);
+    }
     const int64_t n_waves = (args.n_pairs + VAL_STRIDE - 1) / VAL_STRIDE;
     dim3 wgrid((unsigned)wave_grid(n_waves, WAVES_PER_BLOCK_VAL));
     hipLaunchKernelGGL(value_update_kernel_t<1>, wgrid,
