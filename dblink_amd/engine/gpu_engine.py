@@ -242,6 +242,11 @@ class GpuModel:
         # phi(c) norm(c)^k (e^{m s} - 1) for 2 <= k <= k2max, 1 <= m < k
         # (level L = (k-2)(k-1)/2 + m - 1). The d=2 class then draws from
         # two table CDFs plus an exact residual over the row intersection.
+        # MEASURED NEGATIVE (DBLINK_KTAB2=1 to enable): the residual walk
+        # still pays the per-entry cross-row binary searches that dominate
+        # the merge path, so it saves nothing (10M stationary value phase
+        # 55 -> 60 ms) — kept as a validated experiment (kd2 distribution
+        # test) and for schemas where intersections are tiny.
         self.k2tab_excl = self.k2tab_rawsum = None
         self.k2tab_max = 0
         cap2 = int(os.environ.get("DBLINK_KTAB2_MAX_BYTES", str(64 << 30)))
@@ -345,7 +350,7 @@ class GpuEngine(CpuEngine):
         else:
             self.C.set_value_ktables(zt, zt, torch.empty(0, dtype=torch.float32),
                                      0, 0)
-        if self.model.k2tab_max >= 2 and os.environ.get("DBLINK_KTAB2", "1") != "0":
+        if self.model.k2tab_max >= 2 and os.environ.get("DBLINK_KTAB2", "0") == "1":
             self.C.set_value_k2tables(self.model.k2tab_excl,
                                       self.model.k2tab_rawsum,
                                       self.model.k2tab_max)
