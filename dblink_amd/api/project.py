@@ -588,6 +588,20 @@ def run_config(config_path, rank=None, world_size=None, device=None):
     project = Project(cfg, rank=rank, world_size=world_size, device=device)
     os.makedirs(project.output_path, exist_ok=True)
     if rank == 0:
+        # rolling run log next to the outputs (parity: the reference's
+        # log4j RollingFileAppender writing ./dblink.log)
+        import logging.handlers
+
+        root = logging.getLogger()
+        if not any(isinstance(h, logging.handlers.RotatingFileHandler)
+                   for h in root.handlers):
+            fh = logging.handlers.RotatingFileHandler(
+                os.path.join(project.output_path, "dblink.log"),
+                maxBytes=10_000_000, backupCount=3)
+            fh.setFormatter(logging.Formatter(
+                "%(asctime)s %(levelname)s %(name)s: %(message)s"))
+            root.addHandler(fh)
+    if rank == 0:
         with open(os.path.join(project.output_path, "run.txt"), "w") as f:
             f.write(project.mk_string())
     steps = parse_steps(cfg, project)
