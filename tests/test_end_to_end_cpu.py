@@ -419,3 +419,35 @@ def test_check_config_warns_on_quadratic_sampler(tmp_path, capsys):
     step = prj.parse_steps(cfg, project)[0]
     assert step.scale_warning(num_records=10_000_000) is not None
     assert step.scale_warning(num_records=300) is None
+
+
+def test_strings8_schema_cpu_chain():
+    """The 8-Levenshtein-attribute schema (BASELINE config #4's shape)
+    through the CPU engine: exercises multi-attribute od-masks, the native
+    link/value kernels at A=8, and summary bookkeeping."""
+    import os
+    import sys
+
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    from bench import build_cache_and_records
+
+    cache, rv, rf = build_cache_and_records(400, seed=13, schema="strings8")
+    assert rv.shape[1] == 8
+    partitioner = KDTreePartitioner(1, [0])
+    state = deterministic_init(rv, rf, np.arange(400, dtype=np.int64),
+                               cache, partitioner, seed=13)
+    engine = CpuEngine(cache, partitioner)
+    engine.initial_summary(state)
+    flags = SamplerFlags.for_sampler("PCG-I")
+    lls = []
+    for _ in range(30):
+        engine.step(state, flags)
+        lls.append(state.summary.log_likelihood)
+    assert all(np.isfinite(lls))
+    assert state.summary.agg_distortions.shape == (8, 1)
+    # conservation + value sanity
+    assert state.rec_ent.min() >= 0 and state.rec_ent.max() < state.num_entities
+    for a in range(8):
+        V = cache.indexed_attributes[a].index.num_values
+        assert state.ent_values[:, a].min() >= 0
+        assert state.ent_values[:, a].max() < V
