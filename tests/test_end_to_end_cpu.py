@@ -431,6 +431,10 @@ def test_strings8_schema_cpu_chain():
     sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
     from bench import build_cache_and_records
 
+    from dblink_amd.engine.cpu_engine import CpuEngine, SamplerFlags
+    from dblink_amd.engine.init import deterministic_init
+    from dblink_amd.parallel.partitioning import KDTreePartitioner
+
     cache, rv, rf = build_cache_and_records(400, seed=13, schema="strings8")
     assert rv.shape[1] == 8
     partitioner = KDTreePartitioner(1, [0])
