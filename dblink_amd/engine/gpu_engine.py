@@ -249,6 +249,8 @@ class GpuModel:
         # test) and for schemas where intersections are tiny.
         self.k2tab_excl = self.k2tab_rawsum = None
         self.k2tab_max = 0
+        if os.environ.get("DBLINK_KTAB2", "0") != "1":
+            return  # experiment off by default: skip the multi-GB build too
         cap2 = int(os.environ.get("DBLINK_KTAB2_MAX_BYTES", str(64 << 30)))
         k2 = 2
         while k2 < min(kmax, 5) and \

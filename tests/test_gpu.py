@@ -1007,7 +1007,13 @@ def test_value_kd2_kernel_distribution():
     """k=2 clusters with TWO distinct values (the kd2 table path): empirical
     frequencies vs the exact mixture with w(v) = base2(v) (F1 F2 - 1),
     F_i boosted by the collapsed self term at v = x_i."""
-    cache, model = make_model(DEV)
+    import os
+
+    os.environ["DBLINK_KTAB2"] = "1"  # experiment tables are off by default
+    try:
+        cache, model = make_model(DEV)
+    finally:
+        os.environ.pop("DBLINK_KTAB2", None)
     assert model.k2tab_max >= 2
     a = 1
     idx = cache.indexed_attributes[a].index
