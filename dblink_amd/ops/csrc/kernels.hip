@@ -277,7 +277,6 @@ __global__ void link_update_small_kernel(
 constexpr int HEAVY_EXC_CAP = 512;   // A* exclusions per record before fallback
 constexpr int HEAVY_HASH = 1024;     // LDS exclusion hash slots per wave
 constexpr int HEAVY_WAVES = 4;
-constexpr int HEAVY_ROW_CAP = 320;   // LDS-staged sim-row entries per od attr
 
 __global__ void link_update_heavy_kernel(
     const uint8_t* __restrict__ mode,        // [R] 2 = heavy
@@ -314,52 +313,6 @@ __global__ void link_update_heavy_kernel(
   const int32_t p = rec_part[r];
   const int64_t e0 = ent_ptr[p], e1 = ent_ptr[p + 1];
   const uint64_t gid = (uint64_t)rec_gid[r];
-
-  // Stage the record's observed-distorted sim rows in LDS: the per-visit
-  // score needs sim(x_a, y) for every candidate, and the global binary
-  // search's dependent cache-line chain is what the PMC counters show the
-  // kernel waiting on. Up to two rows of <= HEAVY_ROW_CAP entries (covers
-  // the RLdata-schema od sets; longer rows fall back to the global search).
-  __shared__ int32_t rowc_s[HEAVY_WAVES][2][HEAVY_ROW_CAP];
-  __shared__ float rows_s[HEAVY_WAVES][2][HEAVY_ROW_CAP];
-  int staged_attr0 = -1, staged_attr1 = -1;
-  int staged_len0 = 0, staged_len1 = 0;
-  {
-    int slot = 0;
-    for (uint32_t m = od_mask; m && slot < 2;) {
-      const int a = __ffs(m) - 1;
-      m &= m - 1;
-      const int64_t row = voff[a] + rec_values[r * A + a];
-      const int64_t lo = csr_row_ptr[row];
-      const int n = (int)(csr_row_ptr[row + 1] - lo);
-      if (n > HEAVY_ROW_CAP) continue;
-      for (int i = lane; i < n; i += WAVE) {
-        rowc_s[wid][slot][i] = csr_col[lo + i];
-        rows_s[wid][slot][i] = csr_sim[lo + i];
-      }
-      if (slot == 0) { staged_attr0 = a; staged_len0 = n; }
-      else { staged_attr1 = a; staged_len1 = n; }
-      ++slot;
-    }
-  }
-  // wave-lockstep: LDS writes above are visible to all lanes below
-  auto sim_of = [&](int ao, int32_t y) -> float {
-    const int32_t* rc;
-    const float* rs;
-    int n;
-    if (ao == staged_attr0) { rc = rowc_s[wid][0]; rs = rows_s[wid][0]; n = staged_len0; }
-    else if (ao == staged_attr1) { rc = rowc_s[wid][1]; rs = rows_s[wid][1]; n = staged_len1; }
-    else {
-      return sim_lookup(csr_row_ptr, csr_col, csr_sim,
-                        voff[ao] + rec_values[r * A + ao], y);
-    }
-    int lo = 0, hi = n;
-    while (lo < hi) {
-      const int mid = (lo + hi) >> 1;
-      if (rc[mid] < y) lo = mid + 1; else hi = mid;
-    }
-    return (lo < n && rc[lo] == y) ? rs[lo] : 0.0f;
-  };
 
   // proposal pool: the whole partition (no nd attr) or the base posting
   // segment of the single nd attr (postings are stable-sorted when this
@@ -438,7 +391,10 @@ __global__ void link_update_heavy_kernel(
           const int ao = __ffs(mo) - 1;
           mo &= mo - 1;
           const int32_t y = ent_values[(int64_t)e * A + ao];
-          logw += log_norm[voff[ao] + y] + (ao == a ? dsim : sim_of(ao, y));
+          logw += log_norm[voff[ao] + y] +
+                  (ao == a ? dsim
+                           : sim_lookup(csr_row_ptr, csr_col, csr_sim,
+                                        voff[ao] + rec_values[r * A + ao], y));
         }
         const float g = gumbel_from_uniform(
             philox_uniform(seed, iteration, PH_LINK, gid, (uint32_t)e));
@@ -518,7 +474,8 @@ __global__ void link_update_heavy_kernel(
         const int ao = __ffs(mo) - 1;
         mo &= mo - 1;
         const int32_t y = ent_values[(int64_t)e * A + ao];
-        const float s = sim_of(ao, y);
+        const float s = sim_lookup(csr_row_ptr, csr_col, csr_sim,
+                                   voff[ao] + rec_values[r * A + ao], y);
         if (s >= tau) { in_big = true; break; }  // scored exactly in S_r
         t += (double)(log_norm[voff[ao] + y] + s);
       }
