@@ -417,7 +417,7 @@ class GpuEngine(CpuEngine):
         # hierarchical A* link sampler for huge candidate sets (stationary
         # high-distortion regime); DBLINK_HEAVY=0 reverts to pure scans
         self._heavy_thresh = (
-            int(os.environ.get("DBLINK_HEAVY_THRESH", "512"))
+            int(os.environ.get("DBLINK_HEAVY_THRESH", "2048"))
             if os.environ.get("DBLINK_HEAVY", "1") != "0" else 0
         )
         # the hierarchical sampler pays when candidate scans dominate (big
