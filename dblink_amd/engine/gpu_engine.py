@@ -362,7 +362,6 @@ class GpuEngine(CpuEngine):
         self._gs = None
         self._ent_id_base = rank << 40
         self._err = torch.zeros(1, dtype=torch.int32, device=device)
-        self.debug_classes = os.environ.get("DBLINK_DEBUG_CLASSES", "") == "1"
         # per-phase HIP-event timers (SURVEY.md §5.1 observability): eager mode
         # only; accumulate totals, report via phase_times()
         self.phase_timers = os.environ.get("DBLINK_PHASE_TIMERS", "") == "1"
