@@ -22,7 +22,7 @@ cache, rv, rf = b.build_cache_and_records(n, seed=77)
 print(f"[e2e] data+cache {time.time()-t0:.1f}s", flush=True)
 
 t0 = time.time()
-part = KDTreePartitioner(3, [3, 4])
+part = KDTreePartitioner(6, [3, 4])  # 64 partitions: BASELINE config-3 granularity
 state = deterministic_init(rv, rf, np.arange(n, dtype=np.int64), cache, part, seed=319158)
 engine = GpuEngine(cache, part, device=torch.device("cuda", 0))
 engine.rec_ids_array = cols["rec_id"].astype(object)
