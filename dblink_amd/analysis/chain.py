@@ -144,62 +144,141 @@ def _flatten_chain(table):
     return record_ids, cluster_gidx, iter_per_cluster, cluster_sizes, inner_offsets
 
 
-def most_probable_clusters_fast(table):
-    """Vectorized MPC over large chains: clusters are identified by an
-    order-independent composite hash of their member record ids (two
-    independent 64-bit hash aggregates + size — collisions are negligible).
-    Returns {record_id -> (cluster member tuple, frequency)}."""
+def _mpc_core(table):
+    """Vectorized MPC internals over large chains: clusters are identified
+    by an order-independent composite hash of their member record ids (two
+    independent 64-bit mixes of the records' dictionary codes + size —
+    collisions are negligible). All grouping runs on integer codes with
+    contiguous-segment reductions (clusters are contiguous in the flattened
+    layout), so no per-entry string hashing or string groupbys. Returns
+    (best_codes, best_keys, best_freq, first_idx, inner_offsets, codes_i,
+    uniq_rids, kcode)."""
     import pandas as pd
+    import pyarrow as pa
 
-    record_ids, cluster_gidx, iter_per_cluster, cluster_sizes, inner_offsets = (
-        _flatten_chain(table)
-    )
+    # flatten in Arrow and dictionary-encode the member ids natively: no
+    # 50M-element python-object string arrays, no string factorize
+    col = table["linkageStructure"].combine_chunks()
+    if isinstance(col, pa.ChunkedArray):
+        col = col.combine_chunks()
+    clusters = col.flatten()
+    inner_offsets = np.asarray(clusters.offsets)
+    cluster_sizes = np.diff(inner_offsets)
+    members = clusters.flatten()
+    if isinstance(members, pa.ChunkedArray):
+        members = members.combine_chunks()
+    enc = members.dictionary_encode()
+    codes_i = np.asarray(enc.indices)
+    uniq_rids = np.asarray(enc.dictionary.to_numpy(zero_copy_only=False))
     num_samples = len(np.unique(np.asarray(table["iteration"].to_numpy())))
-    rid_ser = pd.Series(record_ids)
-    h1 = pd.util.hash_pandas_object(rid_ser, index=False).to_numpy()
-    h2 = pd.util.hash_pandas_object(rid_ser + "#2", index=False).to_numpy()
-    df = pd.DataFrame({
-        "c": cluster_gidx,
-        "h1": h1.astype(np.uint64),
-        "h2": h2.astype(np.uint64),
-    })
-    agg = df.groupby("c", sort=False).agg(
-        s1=("h1", "sum"), s2=("h2", "sum"), n=("h1", "size")
-    )
-    # composite cluster-content key
-    key = (agg["s1"].to_numpy() ^ (agg["s2"].to_numpy() * np.uint64(0x9E3779B97F4A7C15))
-           ) + agg["n"].to_numpy().astype(np.uint64)
+    codes = codes_i.astype(np.uint64)
+    off64 = inner_offsets.astype(np.int64)
+
+    def mix(x):  # splitmix64 finalizer
+        x = (x + np.uint64(0x9E3779B97F4A7C15))
+        x ^= x >> np.uint64(30)
+        x *= np.uint64(0xBF58476D1CE4E5B9)
+        x ^= x >> np.uint64(27)
+        x *= np.uint64(0x94D049BB133111EB)
+        x ^= x >> np.uint64(31)
+        return x
+
+    with np.errstate(over="ignore"):
+        h1 = mix(codes)
+        # second independent sum: a nonlinear remix of h1 (h1 is already
+        # well-mixed, so one multiply round suffices for pair independence)
+        h2 = (h1 ^ (h1 >> np.uint64(29))) * np.uint64(0xD6E8FEB86659FD93)
+        h2 ^= h2 >> np.uint64(32)
+        # order-independent per-cluster aggregates: clusters are contiguous,
+        # so segment sums come from one zero-prefixed cumsum (mod-2^64
+        # wraparound is part of the hash; np.add.reduceat pays per-segment
+        # overhead at ~50M segments, and the zero prefix removes all
+        # boundary special-casing)
+        n_flat = len(h1)
+        cc1 = np.empty(n_flat + 1, dtype=np.uint64)
+        cc1[0] = np.uint64(0)
+        np.cumsum(h1, out=cc1[1:])
+        cc2 = np.empty(n_flat + 1, dtype=np.uint64)
+        cc2[0] = np.uint64(0)
+        np.cumsum(h2, out=cc2[1:])
+        s1 = cc1[off64[1:]] - cc1[off64[:-1]]
+        s2 = cc2[off64[1:]] - cc2[off64[:-1]]
+        key = (s1 ^ (s2 * np.uint64(0x9E3779B97F4A7C15))) + cluster_sizes.astype(
+            np.uint64)
     # frequency of each distinct cluster content
-    kdf = pd.DataFrame({"key": key})
-    counts = kdf.groupby("key", sort=False).size()
-    freq_per_cluster = counts.loc[kdf["key"]].to_numpy() / num_samples
-    # one representative instance per distinct key (first occurrence)
-    first_idx = kdf.drop_duplicates("key").index.to_numpy()
-    rep_of_key = dict(zip(kdf["key"].to_numpy()[first_idx], first_idx))
-    # per record: best (max freq) cluster among those containing it
-    rec_df = pd.DataFrame({
-        "rid": record_ids,
-        "key": kdf["key"].to_numpy()[cluster_gidx],
-        "freq": freq_per_cluster[cluster_gidx],
-    })
-    best = rec_df.loc[rec_df.groupby("rid", sort=False)["freq"].idxmax()]
+    kcode, _ = pd.factorize(key)
+    kcounts = np.bincount(kcode)
+    freq_per_cluster = kcounts[kcode] / num_samples
+    # one representative cluster instance per distinct key: reversed fancy
+    # assignment leaves the FIRST occurrence (no sort)
+    first_idx = np.zeros(len(kcounts), dtype=np.int64)
+    first_idx[kcode[::-1]] = np.arange(len(kcode) - 1, -1, -1, dtype=np.int64)
+    # per record-code: best (max count, first entry on ties) cluster entry
+    # via ONE composite int64 sort: code | count | inverted entry index
+    entry_counts = np.repeat(kcounts[kcode].astype(np.int64), cluster_sizes)
+    n_ent = len(codes)
+    idx_bits = max(1, int(n_ent - 1).bit_length())
+    cnt_bits = max(1, int(num_samples).bit_length())
+    code_bits = max(1, int(len(uniq_rids) - 1).bit_length())
+    if code_bits + cnt_bits + idx_bits <= 62:
+        inv_idx = (np.int64(n_ent - 1) - np.arange(n_ent, dtype=np.int64))
+        combo = ((codes.astype(np.int64) << np.int64(cnt_bits + idx_bits))
+                 | (entry_counts << np.int64(idx_bits)) | inv_idx)
+        combo.sort()
+        dec_code = combo >> np.int64(cnt_bits + idx_bits)
+        last = np.flatnonzero(np.r_[dec_code[1:] != dec_code[:-1], True])
+        best_entries = (np.int64(n_ent - 1)
+                        - (combo[last]
+                           & ((np.int64(1) << np.int64(idx_bits)) - 1)))
+        best_codes = dec_code[last]
+    else:  # astronomically large chains: 3-key lexsort instead of packing
+        order = np.lexsort((-np.arange(n_ent, dtype=np.int64), entry_counts,
+                            codes.astype(np.int64)))
+        c_sorted = codes[order].astype(np.int64)
+        last = np.flatnonzero(np.r_[c_sorted[1:] != c_sorted[:-1], True])
+        best_entries = order[last]
+        best_codes = c_sorted[last]
+    # cluster of each winning entry: searchsorted over the offsets (only
+    # ~n_records winners, so no 50M-element repeat of cluster ids needed)
+    best_cluster = np.searchsorted(off64, best_entries, side="right") - 1
+    best_keys = kcode[best_cluster]
+    best_freq = freq_per_cluster[best_cluster]
+    return (best_codes, best_keys, best_freq, first_idx, inner_offsets,
+            codes_i, uniq_rids, kcode)
+
+
+def most_probable_clusters_fast(table):
+    """{record_id -> (most-probable cluster frozenset, frequency)} — the
+    per-record dict view over _mpc_core (LinkageChain.scala:52-64)."""
+    (best_codes, best_keys, best_freq, first_idx, inner_offsets, codes_i,
+     uniq_rids, _kcode) = _mpc_core(table)
+    uniq_list = uniq_rids.tolist()
+    bc = best_codes.tolist()
+    bk = best_keys.tolist()
+    bf = best_freq.tolist()
     out = {}
     fs_cache = {}
-    for rid, k, f in zip(best["rid"], best["key"], best["freq"]):
+    for rc, k, f in zip(bc, bk, bf):
         fs = fs_cache.get(k)
         if fs is None:
-            ci = rep_of_key[k]
+            ci = first_idx[k]  # representative cluster instance for this key
             lo, hi = inner_offsets[ci], inner_offsets[ci + 1]
-            fs = frozenset(record_ids[lo:hi].tolist())
+            fs = frozenset(uniq_rids[codes_i[lo:hi]].tolist())
             fs_cache[k] = fs
-        out[rid] = (fs, float(f))
+        out[uniq_list[rc]] = (fs, float(f))
     return out
 
 
 def shared_most_probable_clusters_fast(table):
-    """Vectorized sMPC (LinkageChain.scala:75-109 semantics)."""
-    mpc = most_probable_clusters_fast(table)
-    agg = defaultdict(set)
-    for rid, (cluster, _) in mpc.items():
-        agg[cluster].add(rid)
-    return [set(v) for v in agg.values()]
+    """Vectorized sMPC (LinkageChain.scala:75-109 semantics): records
+    grouped by the content key of their most-probable cluster — built
+    straight from the integer core, no per-record dict or frozensets."""
+    (best_codes, best_keys, _bf, _fi, _io, _ci, uniq_rids, _kc) = (
+        _mpc_core(table))
+    order = np.argsort(best_keys, kind="stable")
+    sk = best_keys[order]
+    bounds = np.flatnonzero(np.r_[True, sk[1:] != sk[:-1]])
+    bounds = np.r_[bounds, len(sk)]
+    members = uniq_rids[best_codes[order]]
+    return [set(members[bounds[i]:bounds[i + 1]].tolist())
+            for i in range(len(bounds) - 1)]
