@@ -37,6 +37,83 @@ constexpr uint32_t PH_LINK = 1, PH_DIST = 2, PH_VALG = 3, PH_VALM = 4,
                    PH_LINKH = 7, PH_LINKF = 8;
 
 // ---------------------------------------------------------------------------
+// Staged similarity rows. A record's od-attribute CSR sim rows are FIXED for
+// the whole candidate scan, yet sim_lookup pays a log2(row)-deep
+// dependent-load binary search per (candidate, attr). PMC counters
+// (profiles/pmc_r02_1M64.txt) show the link kernels latency-bound
+// (SQ_WAIT_ANY >> SQ_ACTIVE_INST), so that chain is the bottleneck: staging
+// each row once into a per-wave LDS open-addressing hash turns it into ~1
+// LDS probe. Sub-table descriptors are nibble-packed per attr (runtime-
+// indexed local arrays would spill to scratch): logpk nibble = log2(slots)
+// (0 = not staged -> global binary search, 15 = empty row -> sim 0), offpk
+// nibble = offset/32. Load factor <= 1/2; rows that don't fit stay global.
+// ---------------------------------------------------------------------------
+constexpr int SIMH_CAP = 512;  // staged entries per wave (4 KB key+value)
+
+__device__ inline void simh_stage(
+    int32_t* kh, float* vh, int lane, const int64_t* __restrict__ csr_row_ptr,
+    const int32_t* __restrict__ csr_col, const float* __restrict__ csr_sim,
+    const int64_t* __restrict__ voff, const int32_t* __restrict__ rec_values,
+    int64_t r, int A, uint32_t od_mask, uint64_t& logpk, uint64_t& offpk) {
+  logpk = 0;
+  offpk = 0;
+  int cursor = 0;
+  for (uint32_t m = od_mask; m;) {
+    const int a = __ffs(m) - 1;
+    m &= m - 1;
+    const int64_t row = voff[a] + rec_values[r * A + a];
+    const int64_t lo = csr_row_ptr[row];
+    const int len = (int)(csr_row_ptr[row + 1] - lo);
+    if (len == 0) {
+      logpk |= 15ull << (4 * a);
+      continue;
+    }
+    int lg = 5;  // >= 32 slots keeps offsets on /32 nibble boundaries
+    while ((1 << lg) < 2 * len) ++lg;
+    if (lg > 9 || cursor + (1 << lg) > SIMH_CAP) continue;  // stays global
+    const int sz = 1 << lg, base = cursor;
+    for (int i = lane; i < sz; i += WAVE) kh[base + i] = -1;
+    // wave-lockstep: the CAS claims below see the -1 fill of this wave
+    for (int i = lane; i < len; i += WAVE) {
+      const int32_t y = csr_col[lo + i];
+      int h = (int)(((uint32_t)y * 2654435761u) >> (32 - lg));
+      for (;;) {
+        const int32_t prev = atomicCAS(&kh[base + h], -1, y);
+        if (prev == -1) {
+          vh[base + h] = csr_sim[lo + i];
+          break;
+        }
+        h = (h + 1) & (sz - 1);
+      }
+    }
+    logpk |= (uint64_t)lg << (4 * a);
+    offpk |= (uint64_t)(base >> 5) << (4 * a);
+    cursor += sz;
+  }
+}
+
+__device__ inline float simh_lookup(const int32_t* kh, const float* vh,
+                                    uint64_t logpk, uint64_t offpk, int a,
+                                    const int64_t* __restrict__ csr_row_ptr,
+                                    const int32_t* __restrict__ csr_col,
+                                    const float* __restrict__ csr_sim,
+                                    int64_t row, int32_t y) {
+  const int lg = (int)((logpk >> (4 * a)) & 15ull);
+  if (lg == 15) return 0.0f;  // empty sim row
+  if (lg == 0)                // row not staged
+    return sim_lookup(csr_row_ptr, csr_col, csr_sim, row, y);
+  const int base = (int)(((offpk >> (4 * a)) & 15ull) << 5);
+  int h = (int)(((uint32_t)y * 2654435761u) >> (32 - lg));
+  const int msk = (1 << lg) - 1;
+  for (;;) {
+    const int32_t k = kh[base + h];
+    if (k == y) return vh[base + h];
+    if (k == -1) return 0.0f;  // staged table holds the FULL row
+    h = (h + 1) & msk;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // K3+K4+K5: link update (PCG-I / Gibbs indexed path)
 //
 // Posting lists are used ONLY to enumerate the smallest candidate set;
@@ -70,9 +147,12 @@ __global__ void link_update_kernel(
     uint64_t seed, uint32_t iteration, const int64_t* __restrict__ ctrl,
     int64_t* __restrict__ rec_ent_out,       // [R]
     const int64_t* __restrict__ rec_ent_in,  // [R]
-    int* __restrict__ error_count) {
+    int* __restrict__ error_count, int simh_on) {
   const int lane = threadIdx.x & (WAVE - 1);
-  int64_t r = (int64_t)blockIdx.x * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
+  const int wid = threadIdx.x / WAVE;
+  __shared__ int32_t simh_k_s[4][SIMH_CAP];  // launcher uses WPB == 4
+  __shared__ float simh_v_s[4][SIMH_CAP];
+  int64_t r = (int64_t)blockIdx.x * (blockDim.x / WAVE) + wid;
   if (r >= R) return;
   if (small_mask != nullptr && small_mask[r]) return;
   if (ctrl != nullptr) { seed = (uint64_t)ctrl[0]; iteration = (uint32_t)ctrl[1]; }
@@ -119,6 +199,14 @@ __global__ void link_update_kernel(
       base_attr >= 0 ? (nd_mask & ~(1u << base_attr)) : nd_mask;
 
   const uint64_t gid = (uint64_t)rec_gid[r];
+  // stage this record's od-attr sim rows into LDS when the scan is long
+  // enough to amortize the fill (short scans: the binary search is cheaper)
+  uint64_t logpk = 0, offpk = 0;
+  int32_t* simh_k = simh_k_s[wid];
+  float* simh_v = simh_v_s[wid];
+  if (simh_on && od_mask && base_n * __popc(od_mask) >= 32)
+    simh_stage(simh_k, simh_v, lane, csr_row_ptr, csr_col, csr_sim, voff,
+               rec_values, r, A, od_mask, logpk, offpk);
   float best_score = -INFINITY;
   long long best_e = -1;
   for (int64_t i = lane; i < base_n; i += WAVE) {
@@ -139,8 +227,8 @@ __global__ void link_update_kernel(
       m &= m - 1;
       const int32_t y = ent_values[(int64_t)e * A + a];
       logw += log_norm[voff[a] + y] +
-              sim_lookup(csr_row_ptr, csr_col, csr_sim,
-                         voff[a] + rec_values[r * A + a], y);
+              simh_lookup(simh_k, simh_v, logpk, offpk, a, csr_row_ptr,
+                          csr_col, csr_sim, voff[a] + rec_values[r * A + a], y);
     }
     float g = gumbel_from_uniform(
         philox_uniform(seed, iteration, PH_LINK, gid, (uint32_t)e));
@@ -294,10 +382,12 @@ __global__ void link_update_heavy_kernel(
     uint64_t seed, uint32_t iteration, const int64_t* __restrict__ ctrl,
     int64_t* __restrict__ rec_ent_out, const int64_t* __restrict__ rec_ent_in,
     int* __restrict__ error_count,
-    unsigned long long* __restrict__ stats) {
+    unsigned long long* __restrict__ stats, int simh_on) {
   // stats[4] (or null): {A* iterations, fallbacks, S_r visits, heavy records}
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
+  __shared__ int32_t simh_k_s[HEAVY_WAVES][SIMH_CAP];
+  __shared__ float simh_v_s[HEAVY_WAVES][SIMH_CAP];
   const int64_t r = (int64_t)blockIdx.x * HEAVY_WAVES + wid;
   if (r >= R || mode[r] != 2) return;
   if (ctrl != nullptr) { seed = (uint64_t)ctrl[0]; iteration = (uint32_t)ctrl[1]; }
@@ -328,6 +418,14 @@ __global__ void link_update_heavy_kernel(
     plo = e0;
     pn = e1 - e0;
   }
+
+  // heavy scans are always long: stage the record's od-attr sim rows in LDS
+  uint64_t logpk = 0, offpk = 0;
+  int32_t* simh_k = simh_k_s[wid];
+  float* simh_v = simh_v_s[wid];
+  if (simh_on && od_mask)
+    simh_stage(simh_k, simh_v, lane, csr_row_ptr, csr_col, csr_sim, voff,
+               rec_values, r, A, od_mask, logpk, offpk);
 
   // ---- 1. exact scan over the similar set S_r ----------------------------
   // The sim-row's posting segments average only a few entries, so a
@@ -393,8 +491,9 @@ __global__ void link_update_heavy_kernel(
           const int32_t y = ent_values[(int64_t)e * A + ao];
           logw += log_norm[voff[ao] + y] +
                   (ao == a ? dsim
-                           : sim_lookup(csr_row_ptr, csr_col, csr_sim,
-                                        voff[ao] + rec_values[r * A + ao], y));
+                           : simh_lookup(simh_k, simh_v, logpk, offpk, ao,
+                                         csr_row_ptr, csr_col, csr_sim,
+                                         voff[ao] + rec_values[r * A + ao], y));
         }
         const float g = gumbel_from_uniform(
             philox_uniform(seed, iteration, PH_LINK, gid, (uint32_t)e));
@@ -474,8 +573,9 @@ __global__ void link_update_heavy_kernel(
         const int ao = __ffs(mo) - 1;
         mo &= mo - 1;
         const int32_t y = ent_values[(int64_t)e * A + ao];
-        const float s = sim_lookup(csr_row_ptr, csr_col, csr_sim,
-                                   voff[ao] + rec_values[r * A + ao], y);
+        const float s = simh_lookup(simh_k, simh_v, logpk, offpk, ao,
+                                    csr_row_ptr, csr_col, csr_sim,
+                                    voff[ao] + rec_values[r * A + ao], y);
         if (s >= tau) { in_big = true; break; }  // scored exactly in S_r
         t += (double)(log_norm[voff[ao] + y] + s);
       }
@@ -511,8 +611,9 @@ __global__ void link_update_heavy_kernel(
         mo &= mo - 1;
         const int32_t y = ent_values[e * A + ao];
         logw += log_norm[voff[ao] + y] +
-                sim_lookup(csr_row_ptr, csr_col, csr_sim,
-                           voff[ao] + rec_values[r * A + ao], y);
+                simh_lookup(simh_k, simh_v, logpk, offpk, ao, csr_row_ptr,
+                            csr_col, csr_sim,
+                            voff[ao] + rec_values[r * A + ao], y);
       }
       const float g = gumbel_from_uniform(
           philox_uniform(seed, iteration, PH_LINKF, gid, (uint32_t)(e - e0)));
@@ -2256,6 +2357,14 @@ static int64_t wave_grid(int64_t items, int waves_per_block) {
   return (items + waves_per_block - 1) / waves_per_block;
 }
 
+// LDS sim-row staging in the link kernels; DBLINK_SIMH=0 restores the
+// global-memory binary-search path (A/B; results are bitwise identical).
+// Re-read per launch so one process can time both arms.
+static int simh_enabled() {
+  const char* e = std::getenv("DBLINK_SIMH");
+  return e ? std::atoi(e) : 1;
+}
+
 void link_update(
     torch::Tensor rec_values, torch::Tensor rec_dist, torch::Tensor rec_gid,
     torch::Tensor rec_part, torch::Tensor cand_lo, torch::Tensor cand_hi,
@@ -2292,7 +2401,8 @@ void link_update(
                        NP, mask_ptr, R, A,
                        (uint64_t)seed, (uint32_t)iteration, ctrl_ptr,
                        rec_ent_out.data_ptr<int64_t>(),
-                       rec_ent_in.data_ptr<int64_t>(), error_count.data_ptr<int>());
+                       rec_ent_in.data_ptr<int64_t>(), error_count.data_ptr<int>(),
+                       simh_enabled());
   }
   if (split) {
     dim3 grid((unsigned)((R + 255) / 256));
@@ -2466,7 +2576,8 @@ void link_update_heavy(
                      rec_ent_out.data_ptr<int64_t>(), rec_ent_in.data_ptr<int64_t>(),
                      error_count.data_ptr<int>(),
                      stats.numel() ? (unsigned long long*)stats.data_ptr<int64_t>()
-                                   : nullptr);
+                                   : nullptr,
+                     simh_enabled());
 }
 
 void cand_ranges(torch::Tensor rec_part, torch::Tensor rec_values,
