@@ -756,6 +756,7 @@ struct ValueArgs {
   // optional [8]: {k2_pairs, hash_pairs, merge_pairs, sum_entries,
   //                sum_units, sum_kobs, rare_pairs, merge_entries}
   unsigned long long* stats;
+  int vchunk;  // 1 = chunked hash-accumulate dense path (DBLINK_VCHUNK)
 };
 
 // Draw from p(v) ~ phi(v)*norm(v)^k by a dense Gumbel scan (rare path for
@@ -1533,8 +1534,90 @@ __device__ void value_update_pair(const ValueArgs& args, int64_t pair, int lane,
           philox_uniform(args.seed, args.iteration, PH_VALG, elem, (uint32_t)v));
       if (logw + g > best) { best = logw + g; best_v = v; }
     }
+  } else if (VMODE != 1 && !gover && !is_const && args.vchunk) {
+    // chunked hash-accumulate (union too big for one LDS table): sweep the
+    // d rows in ascending value-range chunks sized so each chunk's entries
+    // fit the table, then run the same insert/transform/reduce as the small
+    // path per chunk. Replaces the union-merge path's (d-1) dependent-load
+    // binary searches PER ENTRY with one binary search per unit PER CHUNK.
+    // L_v accumulation order per value (ascending unit) matches the small
+    // path, so replay stays deterministic. Cursor = row_end - remaining:
+    // glo holds the fixed row end, gn the mutable remaining count.
+    if (lane == 0) {
+      for (int u = 0; u < n_units; ++u) {
+        const int32_t x2 = gbuf[u];
+        const int64_t lo2 = args.csr_row_ptr[v0 + x2];
+        glo[u] = args.csr_row_ptr[v0 + x2 + 1];
+        gn[u] = (int32_t)(glo[u] - lo2);
+        gse[u] = se_of(x2, gbuf[VAL_DMAX + u]);
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_wave_barrier();
+    const int ccap = (HASH_CAP * 3) / 4;
+    const int per_u = ccap / n_units > 0 ? ccap / n_units : 1;
+    for (;;) {
+      // chunk bound (exclusive): min over units of the value per_u entries
+      // ahead of the cursor; every per-unit segment is then <= per_u
+      int32_t vhi = INT32_MAX;
+      bool any = false;
+      for (int u = 0; u < n_units; ++u) {
+        const int64_t end2 = glo[u];
+        const int64_t cu = end2 - gn[u];
+        if (cu < end2) any = true;
+        if (cu + per_u < end2) {
+          const int32_t cv = args.csr_col[cu + per_u];
+          if (cv < vhi) vhi = cv;
+        }
+      }
+      if (!any) break;
+      for (int i = lane; i < HASH_CAP; i += WAVE) { keys[i] = -1; vals[i] = 0.0f; }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_wave_barrier();
+      for (int u = 0; u < n_units; ++u) {
+        const int64_t end2 = glo[u];
+        const int64_t cu = end2 - gn[u];
+        const int64_t seg_end =
+            (vhi == INT32_MAX) ? end2
+                               : lower_bound_i32(args.csr_col, cu, end2, vhi);
+        const int32_t ux = gbuf[u];
+        const float self_extra = gse[u];
+        const float fm = (float)gbuf[2 * VAL_DMAX + u];
+        for (int64_t j = cu + lane; j < seg_end; j += WAVE) {
+          const int32_t v = args.csr_col[j];
+          const float s = args.csr_sim[j];
+          const float factor_log = (v == ux && self_extra > 0.0f)
+                                       ? fm * __logf(__expf(s) + self_extra)
+                                       : fm * s;
+          uint32_t h = ((uint32_t)v * 2654435761u) & (HASH_CAP - 1);
+          while (true) {
+            const int32_t prev = atomicCAS(&keys[h], -1, v);
+            if (prev == -1 || prev == v) { atomicAdd(&vals[h], factor_log); break; }
+            h = (h + 1) & (HASH_CAP - 1);
+          }
+        }
+        if (lane == 0) gn[u] = (int32_t)(end2 - seg_end);
+        __builtin_amdgcn_wave_barrier();
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_wave_barrier();
+      for (int i = lane; i < HASH_CAP; i += WAVE) {
+        const int32_t v = keys[i];
+        if (v < 0) continue;
+        const float L = vals[i];
+        const float log_expm1 = L + __logf(1.0f - __expf(-L));
+        const float logw = log_base_prob(v) + log_expm1;
+        W += (logw < 80.0f) ? (double)__expf(logw) : exp((double)logw);
+        const float g = gumbel_from_uniform(
+            philox_uniform(args.seed, args.iteration, PH_VALG, elem, (uint32_t)v));
+        if (logw + g > best) { best = logw + g; best_v = v; }
+      }
+      // reads of this chunk's table must drain before the next chunk's clear
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_wave_barrier();
+    }
   } else if (VMODE != 1) {
-    // union-merge path (hash would overflow): iterate every entry of every
+    // union-merge path (gover / const attrs): iterate every entry of every
     // DISTINCT row, but process a value only from the FIRST unit containing
     // it; full L_v comes from binary searches in the other units' rows.
     // Per-unit metadata (row bounds, the theta-dependent self term) is
@@ -2705,6 +2788,10 @@ static ValueArgs make_value_args(
   a.kobs = nullptr;
   a.pair_list = nullptr;
   a.stats = g_value_stats;
+  {
+    const char* e = std::getenv("DBLINK_VCHUNK");
+    a.vchunk = e ? std::atoi(e) : 1;
+  }
   a.tab_excl = g_tab_excl;
   a.tab_rawsum = g_tab_rawsum;
   a.self_expsim = g_self_expsim;

@@ -28,7 +28,7 @@ def _dev(arr, dtype):
 
 
 def make_model(device, values_weights=None, threshold=5.0, max_sim=10.0, Kc=10,
-               priors=(0.5, 50.0)):
+               priors=(0.5, 50.0), names=None):
     """Small two-attribute model (1 const, 1 Levenshtein) on the GPU."""
     from dblink_amd.engine.gpu_engine import GpuModel
     from dblink_amd.models.attribute_index import _python_sim_pairs
@@ -36,12 +36,17 @@ def make_model(device, values_weights=None, threshold=5.0, max_sim=10.0, Kc=10,
     from dblink_amd.models.similarity import ConstantSimilarityFn, LevenshteinSimilarityFn
 
     rng = np.random.default_rng(0)
-    names = ["ANNA", "ANNE", "ANNAH", "BOB", "BORB", "CLAIRE", "CLAIR", "DAVE"]
+    if names is None:
+        names = ["ANNA", "ANNE", "ANNAH", "BOB", "BORB", "CLAIRE", "CLAIR", "DAVE"]
     years = [str(y) for y in range(1950, 1960)]
     rows = []
-    for i in range(400):
-        rows.append([str(rng.choice(years)), str(rng.choice(names))])
-    table = RecordsTable.from_rows([str(i) for i in range(400)], ["0"] * 400, rows)
+    for i in range(max(400, len(names))):
+        # every supplied name appears at least once so the full vocab builds
+        nm = names[i] if i < len(names) else str(rng.choice(names))
+        rows.append([str(rng.choice(years)), nm])
+    n_rows = len(rows)
+    table = RecordsTable.from_rows([str(i) for i in range(n_rows)],
+                                   ["0"] * n_rows, rows)
     prior = BetaShapeParameters(*priors)
     attrs = [
         Attribute("year", ConstantSimilarityFn(), prior),
@@ -1069,3 +1074,81 @@ def test_value_kd2_kernel_distribution():
     z = torch.empty(0, dtype=torch.float64)
     C.set_value_ktables(z, z, torch.empty(0, dtype=torch.float32), 0, 0)
     C.set_value_k2tables(z, z, 0)
+
+
+@gpu
+def test_value_dense_merge_distribution():
+    """k=2 clusters over TWO distinct values whose sim rows overflow the LDS
+    hash (total entries > 3/4 * HASH_CAP): exercises the chunked
+    hash-accumulate dense path of value_update_kernel_t<2>. Empirical value
+    frequencies vs the exact fp64 mixture (GibbsUpdates.scala:533-570):
+    w(v) = base2(v) * (f_x1(v) * f_x2(v) - 1), f_x(x) boosted by the
+    collapsed self term."""
+    rng = np.random.default_rng(11)
+    # mutation ball: every pair within edit distance 4 -> sim rows ~ V long
+    base = list("ABCABCABCA")
+    pool = set()
+    while len(pool) < 650:
+        s = base.copy()
+        for _ in range(rng.integers(1, 3)):
+            s[rng.integers(0, 10)] = "ABCDE"[rng.integers(0, 5)]
+        pool.add("".join(s))
+    cache, model = make_model(DEV, threshold=7.0, names=sorted(pool))
+    a = 1
+    idx = cache.indexed_attributes[a].index
+    V = idx.num_values
+    si = idx.sim_index
+    row_lens = np.diff(si.row_ptr)
+    x2i, x1i = np.argsort(row_lens)[-2:]
+    x1, x2 = int(x1i), int(x2i)
+    assert x1 != x2
+    # the dense path must actually engage (HASH_CAP = 1024 in kernels.hip)
+    assert row_lens[x1] + row_lens[x2] > (1024 * 3) // 4, (
+        row_lens[x1], row_lens[x2])
+
+    th = 0.1
+    model.theta.copy_(torch.full((2, 1), th))
+    N = 80000
+    # N entities, each with one record of value x1 and one of x2 (file 0)
+    rv = np.full((2 * N, 2), -1, dtype=np.int32)
+    rv[0::2, a] = x1
+    rv[1::2, a] = x2
+    rec_dist = np.ones((2 * N, 2), dtype=np.uint8)
+    ent_vals = np.zeros((N, 2), dtype=np.int32)
+    ent_rec_ptr = np.arange(0, 2 * N + 1, 2, dtype=np.int64)
+    ent_rec_idx = np.arange(2 * N, dtype=np.int64)
+    kobs = np.zeros((N, 2), dtype=np.int32)
+    kobs[:, a] = 2
+
+    err = torch.zeros(1, dtype=torch.int32, device=DEV)
+    ev = _dev(ent_vals, torch.int32)
+    empty64 = torch.empty(0, dtype=torch.int64, device=DEV)
+    C.value_update(
+        _dev(rv, torch.int32), _dev(rec_dist, torch.uint8),
+        _dev(np.zeros(2 * N, np.int32), torch.int32),
+        _dev(ent_rec_ptr, torch.int64), _dev(ent_rec_idx, torch.int64), ev,
+        model.theta, model.phi, model.log_phi, model.norm_lin, model.log_norm,
+        model.voff, model.csr_row_ptr, model.csr_col, model.csr_sim,
+        model.phi_prob, model.phi_alias, model.pow_prob, model.pow_alias,
+        model.pow_off, model.log_pow_total, model.attr_const, model.Kc,
+        1, 0, 4242, 9, 0, err, empty64, empty64, empty64,
+        model.csr_excl, model.csr_rawsum, model.z1, empty64,
+        _dev(kobs.reshape(-1), torch.int32))
+    sel = ev.cpu().numpy()[:, a]
+
+    phi = idx.probs
+    norms = idx.sim_norms
+    z2 = idx.sim_norm_total(2)
+    base2 = phi * norms ** 2 / z2
+    L = np.zeros(V)
+    for x in (x1, x2):
+        se = (1.0 / th - 1.0) / (phi[x] * norms[x])
+        for j in range(si.row_ptr[x], si.row_ptr[x + 1]):
+            c = si.col[j]
+            f = si.expsim[j] + (se if c == x else 0.0)
+            L[c] += np.log(f)
+    w = base2 * np.expm1(L)
+    W = w.sum()
+    exact = (base2 + w) / (1.0 + W)
+    emp = np.bincount(sel, minlength=V) / N
+    assert tv_distance(emp, exact) < 0.05, tv_distance(emp, exact)
