@@ -144,6 +144,11 @@ def _flatten_chain(table):
     return record_ids, cluster_gidx, iter_per_cluster, cluster_sizes, inner_offsets
 
 
+# widest (code | count | index) packing that fits a signed int64 composite
+# sort key; beyond it _mpc_core falls back to a 3-key lexsort
+_PACK_BITS = 62
+
+
 def _mpc_core(table):
     """Vectorized MPC internals over large chains: clusters are identified
     by an order-independent composite hash of their member record ids (two
@@ -220,7 +225,7 @@ def _mpc_core(table):
     idx_bits = max(1, int(n_ent - 1).bit_length())
     cnt_bits = max(1, int(num_samples).bit_length())
     code_bits = max(1, int(len(uniq_rids) - 1).bit_length())
-    if code_bits + cnt_bits + idx_bits <= 62:
+    if code_bits + cnt_bits + idx_bits <= _PACK_BITS:
         inv_idx = (np.int64(n_ent - 1) - np.arange(n_ent, dtype=np.int64))
         combo = ((codes.astype(np.int64) << np.int64(cnt_bits + idx_bits))
                  | (entry_counts << np.int64(idx_bits)) | inv_idx)

@@ -171,3 +171,42 @@ def test_mpc_fast_equals_slow_randomized():
                 == {frozenset(c) for c in shared_most_probable_clusters_fast(t)})
 
     check()
+
+
+def test_fast_mpc_lexsort_fallback(monkeypatch):
+    """Chains too large for the packed int64 composite sort fall back to a
+    3-key lexsort; force that branch and check it agrees with the direct
+    implementation."""
+    import pyarrow as pa
+
+    from dblink_amd.analysis import chain as chain_mod
+    from dblink_amd.analysis.chain import (
+        most_probable_clusters,
+        most_probable_clusters_fast,
+        shared_most_probable_clusters,
+        shared_most_probable_clusters_fast,
+    )
+
+    monkeypatch.setattr(chain_mod, "_PACK_BITS", 1)  # always too narrow
+    rng = np.random.default_rng(9)
+    rows = {"iteration": [], "partitionId": [], "linkageStructure": []}
+    rec = [f"r{i}" for i in range(25)]
+    for it in range(10):
+        labels = rng.integers(0, 10, 25)
+        clusters = [[rec[i] for i in np.flatnonzero(labels == l)]
+                    for l in range(10)]
+        rows["iteration"].append(it)
+        rows["partitionId"].append(0)
+        rows["linkageStructure"].append([c for c in clusters if c])
+    table = pa.table(rows, schema=pa.schema([
+        ("iteration", pa.int64()), ("partitionId", pa.int32()),
+        ("linkageStructure", pa.list_(pa.list_(pa.string()))),
+    ]))
+    ref = most_probable_clusters(table)
+    fast = most_probable_clusters_fast(table)
+    assert set(ref) == set(fast)
+    for rid in ref:
+        assert ref[rid][0] == fast[rid][0], rid
+        assert ref[rid][1] == pytest.approx(fast[rid][1])
+    assert (sorted(map(sorted, shared_most_probable_clusters(table)))
+            == sorted(map(sorted, shared_most_probable_clusters_fast(table))))
