@@ -649,15 +649,32 @@ __global__ void link_update_dense_kernel(
     const uint8_t* __restrict__ attr_const, int64_t R, int A, int F,
     int collapsed,  // 1 = PCG-II weights, 0 = Gibbs-Sequential weights
     uint64_t seed, uint32_t iteration, const int64_t* __restrict__ ctrl,
-    int64_t* __restrict__ rec_ent_out) {
+    int64_t* __restrict__ rec_ent_out, int simh_on) {
   const int lane = threadIdx.x & (WAVE - 1);
-  const int64_t r = (int64_t)blockIdx.x * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
+  const int wid = threadIdx.x / WAVE;
+  __shared__ int32_t simh_k_s[4][SIMH_CAP];  // launcher uses WPB == 4
+  __shared__ float simh_v_s[4][SIMH_CAP];
+  const int64_t r = (int64_t)blockIdx.x * (blockDim.x / WAVE) + wid;
   if (r >= R) return;
   if (ctrl != nullptr) { seed = (uint64_t)ctrl[0]; iteration = (uint32_t)ctrl[1]; }
   const int32_t p = rec_part[r];
   const int64_t e0 = ent_ptr[p], e1 = ent_ptr[p + 1];
   const int f = rec_file[r];
   const uint64_t gid = (uint64_t)rec_gid[r];
+
+  // stage the record's sim rows once per wave: the dense scan pays a
+  // sim_lookup per (entity, attr), all against rows fixed for this record
+  uint32_t st_mask = 0;
+  for (int a = 0; a < A; ++a) {
+    if (rec_values[r * A + a] < 0 || attr_const[a]) continue;
+    if (collapsed || rec_dist[r * A + a]) st_mask |= 1u << a;
+  }
+  uint64_t logpk = 0, offpk = 0;
+  int32_t* simh_k = simh_k_s[wid];
+  float* simh_v = simh_v_s[wid];
+  if (simh_on && st_mask && (e1 - e0) * __popc(st_mask) >= 32)
+    simh_stage(simh_k, simh_v, lane, csr_row_ptr, csr_col, csr_sim, voff,
+               rec_values, r, A, st_mask, logpk, offpk);
 
   float best_score = -INFINITY;
   long long best_e = -1;
@@ -671,7 +688,8 @@ __global__ void link_update_dense_kernel(
         const float th = theta[a * F + f];
         float like = phi[voff[a] + x];
         if (!attr_const[a]) {
-          float s = sim_lookup(csr_row_ptr, csr_col, csr_sim, voff[a] + x, y);
+          float s = simh_lookup(simh_k, simh_v, logpk, offpk, a, csr_row_ptr,
+                                csr_col, csr_sim, voff[a] + x, y);
           like *= norm_lin[voff[a] + y] * __expf(s);
         }
         float w = (y == x ? 1.0f - th : 0.0f) + th * like;
@@ -681,7 +699,8 @@ __global__ void link_update_dense_kernel(
         if (!rec_dist[r * A + a]) {
           if (x != y) { logw = -INFINITY; break; }
         } else if (!attr_const[a]) {
-          float s = sim_lookup(csr_row_ptr, csr_col, csr_sim, voff[a] + x, y);
+          float s = simh_lookup(simh_k, simh_v, logpk, offpk, a, csr_row_ptr,
+                                csr_col, csr_sim, voff[a] + x, y);
           logw += __logf(norm_lin[voff[a] + y]) + s;  // phi(x) constant: cancels
         }
       }
@@ -2708,7 +2727,7 @@ void link_update_dense(
                      csr_col.data_ptr<int32_t>(), csr_sim.data_ptr<float>(),
                      attr_const.data_ptr<uint8_t>(), R, A, F, (int)collapsed,
                      (uint64_t)seed, (uint32_t)iteration, ctrl_ptr,
-                     rec_ent_out.data_ptr<int64_t>());
+                     rec_ent_out.data_ptr<int64_t>(), simh_enabled());
 }
 
 // Opt-in value-phase work counters (DBLink_VALUE_STATS): set once from
