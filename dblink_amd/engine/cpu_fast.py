@@ -41,6 +41,23 @@ _DENSE_WS_THRESHOLD = 8_000_000
 _DENSE_LOGSIM_MAX_V = 2048
 
 
+def _stable_argsort(keys, k=None):
+    """Stable argsort of non-negative int keys. Small key ranges route to the
+    native O(n + k) counting sort (identical permutation to numpy's stable
+    mergesort, which measured ~25% of the real-RLdata10000 stationary
+    sweep); anything else falls back to np.argsort(kind="stable")."""
+    if k is not None and keys.size and k <= 8 * keys.size + 65536:
+        from .. import ops
+
+        if ops.have_native() and hasattr(ops.native(), "counting_argsort_cpu"):
+            import torch
+
+            keys64 = np.ascontiguousarray(keys, dtype=np.int64)
+            return ops.native().counting_argsort_cpu(
+                torch.from_numpy(keys64), int(k)).numpy()
+    return np.argsort(keys, kind="stable")
+
+
 def _philox_uniform4(seed, iteration, phase, ids, draw, rank=0):
     """Philox4x32-10 keyed uniforms in (0, 1): FOUR independent streams per
     id (the four 32-bit output words) for the price of one keyed evaluation.
@@ -312,7 +329,7 @@ def _link_indexed_native(state, fm, num_partitions, seed, it):
     for a, ia in enumerate(fm.attrs):
         Va = ia.index.num_values
         keys = state.ent_part.astype(np.int64) * Va + state.ent_values[:, a]
-        perm = np.argsort(keys, kind="stable").astype(np.int32)
+        perm = _stable_argsort(keys, num_partitions * Va).astype(np.int32)
         cnt = np.bincount(keys, minlength=num_partitions * Va)
         ptr = np.zeros(num_partitions * Va + 1, np.int64)
         ptr[1:] = np.cumsum(cnt)
@@ -362,7 +379,7 @@ def _link_dense_collapsed_native(state, fm, num_partitions, seed, it, rank):
             continue
         Va = ia.index.num_values
         keys = state.ent_part.astype(np.int64) * Va + state.ent_values[:, a]
-        perm = np.argsort(keys, kind="stable").astype(np.int32)
+        perm = _stable_argsort(keys, num_partitions * Va).astype(np.int32)
         cnt = np.bincount(keys, minlength=num_partitions * Va)
         ptr = np.zeros(num_partitions * Va + 1, np.int64)
         ptr[1:] = np.cumsum(cnt)
@@ -708,7 +725,7 @@ def _value_and_rest(state, fm, partitioner, num_partitions, rank, seed, it,
     # ---- phase 2: collapsed entity-value update -----------------------------
     # records in entity order, computed once (native + numpy paths and the
     # oracle fallback all address records through it)
-    rorder_all = np.argsort(state.rec_ent, kind="stable")
+    rorder_all = _stable_argsort(state.rec_ent, state.ent_values.shape[0])
     native = _value_native(state, fm, collapsed, seed, it, rank, rorder_all)
     if native is not None:
         new_ev, fallback = native

@@ -94,13 +94,16 @@ class ChainState:
         preserved (matters for reproducibility of per-partition RNG streams).
         """
         self.cpu_sorted = True
-        order = np.argsort(self.ent_part, kind="stable")
+        from .cpu_fast import _stable_argsort
+
+        order = _stable_argsort(self.ent_part, int(self.ent_part.max()) + 1
+                                if self.ent_part.size else 1)
         inv = np.empty_like(order)
         inv[order] = np.arange(order.size)
         self.ent_values = np.ascontiguousarray(self.ent_values[order])
         self.ent_part = np.ascontiguousarray(self.ent_part[order])
         new_rec_ent = inv[self.rec_ent]
-        rec_order = np.argsort(new_rec_ent, kind="stable")
+        rec_order = _stable_argsort(new_rec_ent, order.size)
         self.rec_ent = np.ascontiguousarray(new_rec_ent[rec_order])
         self.rec_values = np.ascontiguousarray(self.rec_values[rec_order])
         self.rec_file = np.ascontiguousarray(self.rec_file[rec_order])

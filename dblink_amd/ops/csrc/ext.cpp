@@ -26,6 +26,8 @@ std::tuple<torch::Tensor, int64_t> pcg1_link_cpu(
     torch::Tensor attr_const, std::vector<torch::Tensor> post_perm,
     std::vector<torch::Tensor> post_ptr, int64_t seed, int64_t iteration);
 
+torch::Tensor counting_argsort_cpu(torch::Tensor keys, int64_t k);
+
 std::tuple<torch::Tensor, torch::Tensor> value_update_cpu(
     torch::Tensor rec_values, torch::Tensor rec_file, torch::Tensor rec_dist,
     torch::Tensor rec_ent, torch::Tensor rorder, torch::Tensor starts,
@@ -162,6 +164,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "indexed PCG-I/Gibbs link update (OpenMP, keyed Philox gumbels)");
   m.def("pcg2_link_cpu", &dblink::pcg2_link_cpu,
         "PCG-II dense link update (OpenMP, f64 log-space)");
+  m.def("counting_argsort_cpu", &dblink::counting_argsort_cpu,
+        "stable counting argsort for small-range int64 keys");
   m.def("sim_pairs_cpu", &dblink::sim_pairs_cpu,
         "banded Levenshtein sim-pair sweep (CPU/OpenMP)");
   m.def("sim_pairs_gpu", &dblink::sim_pairs_gpu,

@@ -471,3 +471,29 @@ std::tuple<torch::Tensor, torch::Tensor> value_update_cpu(
 }
 
 }  // namespace dblink
+
+namespace dblink {
+
+// Stable counting argsort over small-range non-negative int keys: the CPU
+// fast path re-sorts entities/records by (partition, value)-style keys every
+// sweep, and numpy's stable mergesort was ~25% of the real-RLdata10000
+// stationary sweep. O(n + k), identical permutation to
+// np.argsort(kind="stable").
+torch::Tensor counting_argsort_cpu(torch::Tensor keys, int64_t k) {
+  TORCH_CHECK(keys.dtype() == torch::kInt64, "counting_argsort: int64 keys");
+  TORCH_CHECK(keys.is_contiguous());
+  const int64_t n = keys.numel();
+  auto out = torch::empty({n}, torch::kInt64);
+  const int64_t* kp = keys.data_ptr<int64_t>();
+  int64_t* op = out.data_ptr<int64_t>();
+  std::vector<int64_t> ptr((size_t)k + 1, 0);
+  for (int64_t i = 0; i < n; ++i) {
+    TORCH_CHECK(kp[i] >= 0 && kp[i] < k, "counting_argsort: key out of range");
+    ++ptr[(size_t)kp[i] + 1];
+  }
+  for (int64_t v = 0; v < k; ++v) ptr[(size_t)v + 1] += ptr[(size_t)v];
+  for (int64_t i = 0; i < n; ++i) op[ptr[(size_t)kp[i]]++] = i;
+  return out;
+}
+
+}  // namespace dblink

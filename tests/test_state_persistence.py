@@ -769,3 +769,22 @@ def test_pcg2_native_link_matches_numpy_posterior(monkeypatch):
     ll_p, pairs_p = run(False)
     assert abs(ll_n - ll_p) / abs(ll_p) < 0.02, (ll_n, ll_p)
     assert abs(pairs_n - pairs_p) <= max(2.5, 0.4 * pairs_p), (pairs_n, pairs_p)
+
+
+def test_counting_argsort_matches_numpy_stable():
+    """The native counting argsort must be the IDENTICAL permutation to
+    np.argsort(kind='stable') (chain reproducibility depends on it)."""
+    from dblink_amd import ops
+    from dblink_amd.engine.cpu_fast import _stable_argsort
+
+    rng = np.random.default_rng(3)
+    for n, k in ((0, 1), (1, 1), (1000, 7), (20000, 3000), (5000, 40000)):
+        keys = rng.integers(0, k, n).astype(np.int64)
+        got = _stable_argsort(keys, k)
+        np.testing.assert_array_equal(got, np.argsort(keys, kind="stable"))
+    if ops.have_native():
+        import torch
+
+        with pytest.raises(Exception):
+            ops.native().counting_argsort_cpu(
+                torch.tensor([0, 5], dtype=torch.int64), 3)
