@@ -474,10 +474,11 @@ def compute_summary(state: ChainState, cache, dist_probs) -> SummaryVars:
     s = SummaryVars.zeros(A, F)
 
     loglik = 0.0
-    # entity prior terms, all entities
+    # entity prior terms, all entities (gathering the cached log table is
+    # bitwise-identical to logging the gathered probs)
     for a in range(A):
-        probs = attrs[a].index.probs
-        loglik += float(np.sum(np.log(probs[state.ent_values[:, a]])))
+        log_probs = attrs[a].index.log_probs
+        loglik += float(np.sum(log_probs[state.ent_values[:, a]]))
     # isolates
     linked_counts = np.bincount(state.rec_ent, minlength=state.num_entities)
     s.num_isolates = int(np.sum(linked_counts == 0))
@@ -496,7 +497,7 @@ def compute_summary(state: ChainState, cache, dist_probs) -> SummaryVars:
         if np.any(obs):
             xo = x[obs]
             if ia.is_constant:
-                loglik += float(np.sum(np.log(ia.index.probs[xo])))
+                loglik += float(np.sum(ia.index.log_probs[xo]))
             else:
                 y = state.ent_values[state.rec_ent[d], a][obs]
                 es = _exp_sim_pairs(ia.index, xo, y)

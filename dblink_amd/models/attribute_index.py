@@ -158,6 +158,24 @@ class AttributeIndex:
         return len(self.values)
 
     @property
+    def log_probs(self):
+        """log(phi), cached: the per-sweep summary reduction gathers these
+        for every entity and distorted record."""
+        lp = getattr(self, "_log_probs", None)
+        if lp is None:
+            lp = np.log(self.probs)
+            self._log_probs = lp
+        return lp
+
+    @property
+    def log_sim_norms(self):
+        lsn = getattr(self, "_log_sim_norms", None)
+        if lsn is None:
+            lsn = np.log(self.sim_norms)
+            self._log_sim_norms = lsn
+        return lsn
+
+    @property
     def distribution(self) -> AliasTable:
         return self._distribution
 

@@ -135,7 +135,9 @@ class KDTreePartitioner:
         for level in range(self.num_levels):
             attr_id = self.attribute_ids[it % len(self.attribute_ids)]
             it += 1
-            # node id per row under the current tree
+            # node id per row under the current tree (descent cache rebuilt
+            # each level while the tree grows)
+            self._flat_cache = None
             node_ids = self._leaf_node_ids(values)
             col = values[:, attr_id]
             # group (node, value) weights
@@ -154,33 +156,49 @@ class KDTreePartitioner:
                         "Poor quality split (%.1f%%) at node %d.", splitter.split_quality * 100, nid
                     )
                 self.tree.split_node(int(nid), attr_id, splitter)
+        self._flat_cache = None
         return self
 
     def _leaf_node_ids(self, values):
+        # flat-array descent (the per-node masking variant re-scanned the
+        # whole active set per node and per level); range splits — the
+        # common case for string attributes — are fully vectorized, set
+        # splits use a per-node sorted-slice binary search
+        flat = getattr(self, "_flat_cache", None)
+        if flat is None:
+            flat = self.as_flat()
+            self._flat_cache = flat
+        kind, attr, a, b, rset = (flat["kind"], flat["attr"], flat["a"],
+                                  flat["b"], flat["rset"])
         N = values.shape[0]
-        out = np.zeros(N, dtype=np.int64)
-        # vectorized descent level by level
-        active = np.arange(N)
-        while active.size:
-            nids = out[active]
-            nodes = self.tree.nodes
-            next_active = []
-            # group rows by current node
-            for nid in np.unique(nids):
-                node = nodes[nid] if nid < len(nodes) else None
-                if node is None or node.splitter is None:
-                    continue
-                rows = active[nids == nid]
-                col = values[rows, node.attr_id]
-                if isinstance(node.splitter, RangeSplitter):
-                    right = col > node.splitter.split_value
-                else:
-                    rs = node.splitter.right_set
-                    right = np.fromiter((int(v) in rs for v in col), dtype=bool, count=len(col))
-                out[rows] = np.where(right, 2 * nid + 2, 2 * nid + 1)
-                next_active.append(rows)
-            active = np.concatenate(next_active) if next_active else np.empty(0, dtype=np.int64)
-        return out
+        node = np.zeros(N, dtype=np.int64)
+        for _ in range(self.num_levels):
+            k = kind[node]
+            act = k != 0
+            if not act.any():
+                break
+            right = np.zeros(N, dtype=bool)
+            rsel = k == 1
+            if rsel.any():
+                col = values[rsel, attr[node[rsel]]]
+                right[rsel] = col > a[node[rsel]]
+            ssel = k == 2
+            if ssel.any():
+                idx = np.flatnonzero(ssel)
+                nids = node[idx]
+                col = values[idx, attr[nids]]
+                for nid in np.unique(nids):
+                    m = nids == nid
+                    lo, ln = int(a[nid]), int(b[nid])
+                    if ln == 0:
+                        continue
+                    sl = rset[lo:lo + ln]
+                    cm = col[m]
+                    p = np.searchsorted(sl, cm)
+                    right[idx[m]] = (p < ln) & (sl[np.minimum(p, ln - 1)] == cm)
+            node = np.where(act, np.where(right, 2 * node + 2, 2 * node + 1),
+                            node)
+        return node
 
     def get_partition_id(self, values) -> int:
         return self.tree.leaf_number(values)
@@ -222,7 +240,7 @@ class KDTreePartitioner:
                 continue
             if node.splitter is None:
                 kind[i] = 0
-                a[i] = node.value
+                a[i] = node.value if node.value is not None else 0
             elif isinstance(node.splitter, RangeSplitter):
                 kind[i] = 1
                 attr[i] = node.attr_id
