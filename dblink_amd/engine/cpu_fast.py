@@ -241,6 +241,38 @@ def _pcg2_native_model(fm):
     return got
 
 
+def _distortion_native(state, fm, seed, it, rank):
+    """OpenMP distortion resample (distortion_update_cpu): bitwise-identical
+    to the numpy phase-3 block (same packed Philox stream, same f64
+    expressions). Returns the [R, A] uint8 array or None when unavailable."""
+    import os as _os
+
+    if _os.environ.get("DBLINK_NATIVE_DIST", "1") == "0":
+        return None
+    from .. import ops
+
+    if not ops.have_native() or not hasattr(ops.native(), "distortion_update_cpu"):
+        return None
+    import torch
+
+    sm = getattr(fm, "_self_mass_cat", None)
+    if sm is None:
+        sm = torch.from_numpy(np.ascontiguousarray(np.concatenate(
+            [ia.index.self_mass for ia in fm.attrs])))
+        voff = np.zeros(fm.A + 1, dtype=np.int64)
+        np.cumsum([ia.index.num_values for ia in fm.attrs], out=voff[1:])
+        fm._self_mass_cat = sm
+        fm._voff_cat = torch.from_numpy(voff)
+    return ops.native().distortion_update_cpu(
+        torch.from_numpy(np.ascontiguousarray(state.rec_values)),
+        torch.from_numpy(np.ascontiguousarray(state.rec_file)),
+        torch.from_numpy(np.ascontiguousarray(state.rec_ent)),
+        torch.from_numpy(np.ascontiguousarray(state.ent_values)),
+        torch.from_numpy(np.ascontiguousarray(state.dist_probs.probs)),
+        sm, fm._voff_cat, int(seed), int(it), int(rank),
+    ).numpy()
+
+
 def _value_native(state, fm, collapsed, seed, it, rank, rorder_all):
     """OpenMP entity-value update (value_update_cpu): same Philox streams and
     f64 op order as the numpy block below (draws agree except ulp-boundary
@@ -914,19 +946,23 @@ def _value_and_rest(state, fm, partitioner, num_partitions, rank, seed, it,
     _mark("value")
 
     # ---- phase 3: distortion update -----------------------------------------
-    y_link = state.ent_values[state.rec_ent]  # [R, A]
-    u_d = _philox_dense(seed, it, _PH_DIST, R * A, rank).reshape(R, A)
-    self_mass = np.stack(
-        [ia.index.self_mass[np.maximum(rv[:, a], 0)] for a, ia in enumerate(attrs)], 1
-    )
-    pr1 = theta_ra * self_mass
-    p_agree = pr1 / (pr1 + (1.0 - theta_ra))
-    z = np.where(
-        ~obs,
-        u_d < theta_ra,
-        np.where(rv == y_link, u_d < p_agree, True),
-    )
-    state.rec_dist = z.astype(np.uint8)
+    z8 = _distortion_native(state, fm, seed, it, rank)
+    if z8 is not None:
+        state.rec_dist = z8
+    else:
+        y_link = state.ent_values[state.rec_ent]  # [R, A]
+        u_d = _philox_dense(seed, it, _PH_DIST, R * A, rank).reshape(R, A)
+        self_mass = np.stack(
+            [ia.index.self_mass[np.maximum(rv[:, a], 0)] for a, ia in enumerate(attrs)], 1
+        )
+        pr1 = theta_ra * self_mass
+        p_agree = pr1 / (pr1 + (1.0 - theta_ra))
+        z = np.where(
+            ~obs,
+            u_d < theta_ra,
+            np.where(rv == y_link, u_d < p_agree, True),
+        )
+        state.rec_dist = z.astype(np.uint8)
     _mark("distortion")
 
     # ---- partition reassignment + bookkeeping -------------------------------

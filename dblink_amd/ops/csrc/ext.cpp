@@ -28,6 +28,11 @@ std::tuple<torch::Tensor, int64_t> pcg1_link_cpu(
 
 torch::Tensor counting_argsort_cpu(torch::Tensor keys, int64_t k);
 
+torch::Tensor distortion_update_cpu(
+    torch::Tensor rec_values, torch::Tensor rec_file, torch::Tensor rec_ent,
+    torch::Tensor ent_values, torch::Tensor theta, torch::Tensor self_mass,
+    torch::Tensor voff, int64_t seed, int64_t iteration, int64_t rank);
+
 std::tuple<torch::Tensor, torch::Tensor> value_update_cpu(
     torch::Tensor rec_values, torch::Tensor rec_file, torch::Tensor rec_dist,
     torch::Tensor rec_ent, torch::Tensor rorder, torch::Tensor starts,
@@ -166,6 +171,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "PCG-II dense link update (OpenMP, f64 log-space)");
   m.def("counting_argsort_cpu", &dblink::counting_argsort_cpu,
         "stable counting argsort for small-range int64 keys");
+  m.def("distortion_update_cpu", &dblink::distortion_update_cpu,
+        "distortion resample (OpenMP, bitwise-matches the numpy fast path)");
   m.def("sim_pairs_cpu", &dblink::sim_pairs_cpu,
         "banded Levenshtein sim-pair sweep (CPU/OpenMP)");
   m.def("sim_pairs_gpu", &dblink::sim_pairs_gpu,

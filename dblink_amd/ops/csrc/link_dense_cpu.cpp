@@ -497,3 +497,56 @@ torch::Tensor counting_argsort_cpu(torch::Tensor keys, int64_t k) {
 }
 
 }  // namespace dblink
+
+namespace dblink {
+
+// K7 distortion resample (GibbsUpdates.scala:324-359) for the CPU fast
+// path: bitwise-identical to the numpy phase (same Philox stream — packed
+// four positions per counter like cpu_fast._philox_dense — and the same
+// f64 expressions).
+torch::Tensor distortion_update_cpu(
+    torch::Tensor rec_values, torch::Tensor rec_file, torch::Tensor rec_ent,
+    torch::Tensor ent_values, torch::Tensor theta, torch::Tensor self_mass,
+    torch::Tensor voff, int64_t seed, int64_t iteration, int64_t rank) {
+  const int64_t R = rec_values.size(0);
+  const int A = (int)rec_values.size(1);
+  const int64_t F = theta.size(1);
+  auto out = torch::empty({R, (int64_t)A}, torch::kUInt8);
+  const int32_t* rv = rec_values.data_ptr<int32_t>();
+  const int32_t* rf = rec_file.data_ptr<int32_t>();
+  const int64_t* re = rec_ent.data_ptr<int64_t>();
+  const int32_t* ev = ent_values.data_ptr<int32_t>();
+  const double* th = theta.data_ptr<double>();
+  const double* sm = self_mass.data_ptr<double>();
+  const int64_t* vo = voff.data_ptr<int64_t>();
+  uint8_t* z = out.data_ptr<uint8_t>();
+  const int64_t n = R * (int64_t)A;
+  const int64_t blocks = (n + 3) / 4;
+#pragma omp parallel for schedule(static)
+  for (int64_t blk = 0; blk < blocks; ++blk) {
+    double u[4];
+    philox_cpu4((uint64_t)seed, (uint32_t)iteration, 2u, (uint64_t)blk, 0u,
+                (int)rank, u);
+    const int64_t lim = std::min<int64_t>(4, n - blk * 4);
+    for (int64_t w = 0; w < lim; ++w) {
+      const int64_t p = blk * 4 + w;
+      const int64_t r = p / A;
+      const int a = (int)(p % A);
+      const int32_t x = rv[p];
+      const double ta = th[(int64_t)a * F + rf[r]];
+      bool zd;
+      if (x < 0) {
+        zd = u[w] < ta;
+      } else if (x == ev[re[r] * A + a]) {
+        const double pr1 = ta * sm[vo[a] + x];
+        zd = u[w] < pr1 / (pr1 + (1.0 - ta));
+      } else {
+        zd = true;
+      }
+      z[p] = zd ? 1 : 0;
+    }
+  }
+  return out;
+}
+
+}  // namespace dblink

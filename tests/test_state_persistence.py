@@ -788,3 +788,40 @@ def test_counting_argsort_matches_numpy_stable():
         with pytest.raises(Exception):
             ops.native().counting_argsort_cpu(
                 torch.tensor([0, 5], dtype=torch.int64), 3)
+
+
+def test_native_distortion_matches_numpy_bitwise(monkeypatch):
+    """DBLINK_NATIVE_DIST on/off must produce bitwise-identical chains (the
+    native kernel replicates the packed Philox stream and f64 expressions)."""
+    from dblink_amd import ops
+
+    if not (ops.have_native()
+            and hasattr(ops.native(), "distortion_update_cpu")):
+        pytest.skip("native extension unavailable")
+    import os as _os
+    import sys as _sys
+    _sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
+    from bench import build_cache_and_records
+
+    from dblink_amd.engine.cpu_engine import CpuEngine, SamplerFlags
+    from dblink_amd.engine.init import deterministic_init
+    from dblink_amd.parallel.partitioning import KDTreePartitioner
+
+    def run(flag):
+        monkeypatch.setenv("DBLINK_NATIVE_DIST", flag)
+        cache, rv, rf = build_cache_and_records(1500, seed=5)
+        part = KDTreePartitioner(2, [3, 4])
+        st = deterministic_init(rv, rf, np.arange(1500, dtype=np.int64),
+                                cache, part, seed=11)
+        eng = CpuEngine(cache, part)
+        eng.initial_summary(st)
+        fl = SamplerFlags.for_sampler("PCG-I")
+        for _ in range(40):
+            eng.step(st, fl)
+        return st
+
+    a, c = run("1"), run("0")
+    np.testing.assert_array_equal(a.rec_dist, c.rec_dist)
+    np.testing.assert_array_equal(a.ent_values, c.ent_values)
+    np.testing.assert_array_equal(a.rec_ent, c.rec_ent)
+    assert a.summary.log_likelihood == c.summary.log_likelihood
