@@ -462,12 +462,79 @@ def _exp_sim_pairs(index, xs, ys):
     return np.where(es > 0.0, es, 1.0)
 
 
+def _summary_native(state: ChainState, cache):
+    """OpenMP summary reduction (summary_cpu). Integer outputs are exact;
+    the log-likelihood terms match the numpy path but the float summation
+    ORDER differs (OMP blocks vs numpy pairwise), so the value can differ
+    in ulps — it is a diagnostic series everywhere. Returns None when the
+    native extension is unavailable or DBLINK_NATIVE_SUMMARY=0."""
+    import os as _os
+
+    if _os.environ.get("DBLINK_NATIVE_SUMMARY", "1") == "0":
+        return None
+    from .. import ops
+
+    if not ops.have_native() or not hasattr(ops.native(), "summary_cpu"):
+        return None
+    import torch
+
+    cat = getattr(cache, "_summary_cat", None)
+    if cat is None:
+        attrs = cache.indexed_attributes
+        voff = np.zeros(len(attrs) + 1, dtype=np.int64)
+        np.cumsum([ia.index.num_values for ia in attrs], out=voff[1:])
+        probs = np.concatenate([ia.index.probs for ia in attrs])
+        log_probs = np.concatenate([ia.index.log_probs for ia in attrs])
+        sim_norms = np.concatenate([ia.index.sim_norms for ia in attrs])
+        rp_cat = np.zeros(voff[-1] + 1, dtype=np.int64)
+        col_parts, es_parts = [], []
+        base = 0
+        for a, ia in enumerate(attrs):
+            if ia.is_constant:
+                rp_cat[voff[a] + 1:voff[a + 1] + 1] = base
+            else:
+                si = ia.index.sim_index
+                rp_cat[voff[a] + 1:voff[a + 1] + 1] = base + si.row_ptr[1:]
+                col_parts.append(si.col.astype(np.int32))
+                es_parts.append(si.expsim.astype(np.float64))
+                base += len(si.col)
+        col_cat = (np.concatenate(col_parts) if col_parts
+                   else np.zeros(0, np.int32))
+        es_cat = (np.concatenate(es_parts) if es_parts
+                  else np.zeros(0, np.float64))
+        cst = np.array([1 if ia.is_constant else 0 for ia in attrs],
+                       dtype=np.uint8)
+        cat = tuple(
+            torch.from_numpy(np.ascontiguousarray(x))
+            for x in (probs, log_probs, sim_norms, voff, rp_cat, col_cat,
+                      es_cat, cst))
+        cache._summary_cat = cat
+    probs_t, lp_t, sn_t, voff_t, rp_t, col_t, es_t, cst_t = cat
+    ll, iso, agg, hist = ops.native().summary_cpu(
+        torch.from_numpy(np.ascontiguousarray(state.rec_values)),
+        torch.from_numpy(np.ascontiguousarray(state.rec_dist)),
+        torch.from_numpy(np.ascontiguousarray(state.rec_file)),
+        torch.from_numpy(np.ascontiguousarray(state.rec_ent)),
+        torch.from_numpy(np.ascontiguousarray(state.ent_values)),
+        probs_t, lp_t, sn_t, voff_t, rp_t, col_t, es_t, cst_t,
+        int(cache.num_files))
+    s = SummaryVars.zeros(len(cache.indexed_attributes), cache.num_files)
+    s.log_likelihood = float(ll)
+    s.num_isolates = int(iso)
+    s.agg_distortions = agg.numpy()
+    s.rec_distortions = hist.numpy()
+    return s
+
+
 def compute_summary(state: ChainState, cache, dist_probs) -> SummaryVars:
     """Summary variables (GibbsUpdates.scala:219-301) for the LOCAL shard.
 
     The Beta-prior terms (driver-side in the reference) are added by the
     caller after the cross-rank reduction — see ``add_prior_terms``.
     """
+    native = _summary_native(state, cache)
+    if native is not None:
+        return native
     attrs = cache.indexed_attributes
     A = len(attrs)
     F = cache.num_files

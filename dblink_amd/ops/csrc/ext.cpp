@@ -28,6 +28,13 @@ std::tuple<torch::Tensor, int64_t> pcg1_link_cpu(
 
 torch::Tensor counting_argsort_cpu(torch::Tensor keys, int64_t k);
 
+std::tuple<double, int64_t, torch::Tensor, torch::Tensor> summary_cpu(
+    torch::Tensor rec_values, torch::Tensor rec_dist, torch::Tensor rec_file,
+    torch::Tensor rec_ent, torch::Tensor ent_values, torch::Tensor probs,
+    torch::Tensor log_probs, torch::Tensor sim_norms, torch::Tensor voff,
+    torch::Tensor csr_row_ptr, torch::Tensor csr_col, torch::Tensor csr_expsim,
+    torch::Tensor attr_const, int64_t num_files);
+
 torch::Tensor distortion_update_cpu(
     torch::Tensor rec_values, torch::Tensor rec_file, torch::Tensor rec_ent,
     torch::Tensor ent_values, torch::Tensor theta, torch::Tensor self_mass,
@@ -173,6 +180,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "stable counting argsort for small-range int64 keys");
   m.def("distortion_update_cpu", &dblink::distortion_update_cpu,
         "distortion resample (OpenMP, bitwise-matches the numpy fast path)");
+  m.def("summary_cpu", &dblink::summary_cpu,
+        "summary reduction: loglik + isolates + distortion counts (OpenMP)");
   m.def("sim_pairs_cpu", &dblink::sim_pairs_cpu,
         "banded Levenshtein sim-pair sweep (CPU/OpenMP)");
   m.def("sim_pairs_gpu", &dblink::sim_pairs_gpu,

@@ -550,3 +550,111 @@ torch::Tensor distortion_update_cpu(
 }
 
 }  // namespace dblink
+
+namespace dblink {
+
+// K8 summary reduction (GibbsUpdates.scala:219-301) for the CPU engine:
+// log-likelihood + isolates + per-(attr, file) distortion counts + the
+// per-record distortion histogram in one OpenMP pass. Matches the numpy
+// compute_summary term-for-term; float summation ORDER differs (OMP
+// reduction vs numpy pairwise), so the log-likelihood may differ in ulps —
+// every consumer treats it as a diagnostic series.
+std::tuple<double, int64_t, torch::Tensor, torch::Tensor> summary_cpu(
+    torch::Tensor rec_values, torch::Tensor rec_dist, torch::Tensor rec_file,
+    torch::Tensor rec_ent, torch::Tensor ent_values, torch::Tensor probs,
+    torch::Tensor log_probs, torch::Tensor sim_norms, torch::Tensor voff,
+    torch::Tensor csr_row_ptr, torch::Tensor csr_col, torch::Tensor csr_expsim,
+    torch::Tensor attr_const, int64_t num_files) {
+  const int64_t R = rec_values.size(0);
+  const int A = (int)rec_values.size(1);
+  const int64_t E = ent_values.size(0);
+  const int64_t F = num_files;
+  const int32_t* rv = rec_values.data_ptr<int32_t>();
+  const uint8_t* rd = rec_dist.data_ptr<uint8_t>();
+  const int32_t* rf = rec_file.data_ptr<int32_t>();
+  const int64_t* re = rec_ent.data_ptr<int64_t>();
+  const int32_t* ev = ent_values.data_ptr<int32_t>();
+  const double* ph = probs.data_ptr<double>();
+  const double* lp = log_probs.data_ptr<double>();
+  const double* sn = sim_norms.data_ptr<double>();
+  const int64_t* vo = voff.data_ptr<int64_t>();
+  const int64_t* rp = csr_row_ptr.data_ptr<int64_t>();
+  const int32_t* cc = csr_col.data_ptr<int32_t>();
+  const double* ce = csr_expsim.data_ptr<double>();
+  const uint8_t* cst = attr_const.data_ptr<uint8_t>();
+
+  auto agg = torch::zeros({(int64_t)A, F}, torch::kInt64);
+  auto hist = torch::zeros({(int64_t)A + 1}, torch::kInt64);
+  int64_t* aggp = agg.data_ptr<int64_t>();
+  int64_t* hp = hist.data_ptr<int64_t>();
+
+  // per-thread partials summed in THREAD ORDER afterwards: the reduction
+  // must be deterministic run-to-run (diagnostics.csv reproducibility) —
+  // an omp critical / reduction clause would sum in arrival order
+  const int nthr = omp_get_max_threads();
+  std::vector<double> ll_ent(nthr, 0.0), ll_rec(nthr, 0.0);
+#pragma omp parallel
+  {
+    const int tid = omp_get_thread_num();
+    double ll = 0.0;
+#pragma omp for schedule(static)
+    for (int64_t e = 0; e < E; ++e)
+      for (int a = 0; a < A; ++a) ll += lp[vo[a] + ev[e * A + a]];
+    ll_ent[tid] = ll;
+  }
+
+  std::vector<uint8_t> linked(E, 0);
+  for (int64_t r = 0; r < R; ++r) linked[re[r]] = 1;
+  int64_t iso = 0;
+  for (int64_t e = 0; e < E; ++e) iso += linked[e] == 0;
+
+#pragma omp parallel
+  {
+    const int tid = omp_get_thread_num();
+    std::vector<int64_t> agg_l((size_t)A * F, 0);
+    std::vector<int64_t> hist_l((size_t)A + 1, 0);
+    double ll = 0.0;
+#pragma omp for schedule(static) nowait
+    for (int64_t r = 0; r < R; ++r) {
+      int nd = 0;
+      for (int a = 0; a < A; ++a) {
+        if (!rd[r * A + a]) continue;
+        ++nd;
+        ++agg_l[(size_t)a * F + rf[r]];
+        const int32_t x = rv[r * A + a];
+        if (x < 0) continue;
+        const int64_t row = vo[a] + x;
+        if (cst[a]) {  // constant attr: term = log(phi(x))
+          ll += lp[row];
+          continue;
+        }
+        const int32_t y = ev[re[r] * A + a];
+        // expsim(x, y): binary search x's row; absent pairs have expsim 1
+        double es = 1.0;
+        int64_t lo = rp[row], hi = rp[row + 1];
+        while (lo < hi) {
+          const int64_t mid = (lo + hi) >> 1;
+          if (cc[mid] < y)
+            lo = mid + 1;
+          else
+            hi = mid;
+        }
+        if (lo < rp[row + 1] && cc[lo] == y) es = ce[lo];
+        ll += std::log(ph[row] * sn[vo[a] + y] * es);
+      }
+      ++hist_l[nd];
+    }
+    ll_rec[tid] = ll;
+#pragma omp critical
+    {
+      for (size_t i = 0; i < agg_l.size(); ++i) aggp[i] += agg_l[i];
+      for (size_t i = 0; i < hist_l.size(); ++i) hp[i] += hist_l[i];
+    }
+  }
+  double loglik = 0.0;
+  for (int t = 0; t < nthr; ++t) loglik += ll_ent[t];
+  for (int t = 0; t < nthr; ++t) loglik += ll_rec[t];
+  return {loglik, iso, agg, hist};
+}
+
+}  // namespace dblink

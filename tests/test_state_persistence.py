@@ -825,3 +825,44 @@ def test_native_distortion_matches_numpy_bitwise(monkeypatch):
     np.testing.assert_array_equal(a.ent_values, c.ent_values)
     np.testing.assert_array_equal(a.rec_ent, c.rec_ent)
     assert a.summary.log_likelihood == c.summary.log_likelihood
+
+
+def test_native_summary_matches_numpy():
+    """summary_cpu: integer outputs exact, log-likelihood equal to the numpy
+    reduction up to summation-order ulps."""
+    import os as _os
+    import sys as _sys
+    _sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
+    from bench import build_cache_and_records
+
+    from dblink_amd import ops
+    from dblink_amd.engine import cpu_engine as ce
+    from dblink_amd.engine.init import deterministic_init
+    from dblink_amd.parallel.partitioning import KDTreePartitioner
+
+    if not (ops.have_native() and hasattr(ops.native(), "summary_cpu")):
+        pytest.skip("native extension unavailable")
+    cache, rv, rf = build_cache_and_records(2000, seed=6)
+    part = KDTreePartitioner(2, [3, 4])
+    st = deterministic_init(rv, rf, np.arange(2000, dtype=np.int64), cache,
+                            part, seed=12)
+    eng = ce.CpuEngine(cache, part)
+    eng.initial_summary(st)
+    fl = ce.SamplerFlags.for_sampler("PCG-I")
+    for _ in range(40):
+        eng.step(st, fl)
+    nat = ce._summary_native(st, cache)
+    assert nat is not None
+    old = _os.environ.get("DBLINK_NATIVE_SUMMARY")
+    _os.environ["DBLINK_NATIVE_SUMMARY"] = "0"
+    try:
+        ref = ce.compute_summary(st, cache, st.dist_probs)
+    finally:
+        if old is None:
+            _os.environ.pop("DBLINK_NATIVE_SUMMARY", None)
+        else:
+            _os.environ["DBLINK_NATIVE_SUMMARY"] = old
+    assert nat.num_isolates == ref.num_isolates
+    np.testing.assert_array_equal(nat.agg_distortions, ref.agg_distortions)
+    np.testing.assert_array_equal(nat.rec_distortions, ref.rec_distortions)
+    assert nat.log_likelihood == pytest.approx(ref.log_likelihood, rel=1e-12)
