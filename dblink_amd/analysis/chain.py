@@ -149,6 +149,57 @@ def _flatten_chain(table):
 _PACK_BITS = 62
 
 
+def _native_mpc():
+    """The OpenMP MPC helpers, or None (numpy fallback). Their integer
+    arithmetic is bitwise-identical to the numpy expressions they stand
+    in for (DBLINK_NATIVE_MPC=0 forces the numpy path)."""
+    import os
+
+    if os.environ.get("DBLINK_NATIVE_MPC", "1") == "0":
+        return None
+    try:
+        from .. import ops
+
+        if ops.have_native() and hasattr(ops.native(), "mpc_cluster_keys"):
+            return ops.native()
+    except Exception:
+        pass
+    return None
+
+
+def _mpc_keys_numpy(codes, off64, cluster_sizes):
+    """Order-independent cluster content keys: two splitmix-derived per-code
+    hashes summed per cluster mod 2^64 (zero-prefixed cumsum + segment
+    diff), combined with the cluster size. mpc_cluster_keys computes the
+    exact same integers natively."""
+
+    def mix(x):  # splitmix64 finalizer
+        x = (x + np.uint64(0x9E3779B97F4A7C15))
+        x ^= x >> np.uint64(30)
+        x *= np.uint64(0xBF58476D1CE4E5B9)
+        x ^= x >> np.uint64(27)
+        x *= np.uint64(0x94D049BB133111EB)
+        x ^= x >> np.uint64(31)
+        return x
+
+    h1 = mix(codes)
+    # second independent sum: a nonlinear remix of h1 (h1 is already
+    # well-mixed, so one multiply round suffices for pair independence)
+    h2 = (h1 ^ (h1 >> np.uint64(29))) * np.uint64(0xD6E8FEB86659FD93)
+    h2 ^= h2 >> np.uint64(32)
+    n_flat = len(h1)
+    cc1 = np.empty(n_flat + 1, dtype=np.uint64)
+    cc1[0] = np.uint64(0)
+    np.cumsum(h1, out=cc1[1:])
+    cc2 = np.empty(n_flat + 1, dtype=np.uint64)
+    cc2[0] = np.uint64(0)
+    np.cumsum(h2, out=cc2[1:])
+    s1 = cc1[off64[1:]] - cc1[off64[:-1]]
+    s2 = cc2[off64[1:]] - cc2[off64[:-1]]
+    return (s1 ^ (s2 * np.uint64(0x9E3779B97F4A7C15))) + cluster_sizes.astype(
+        np.uint64)
+
+
 def _mpc_core(table):
     """Vectorized MPC internals over large chains: clusters are identified
     by an order-independent composite hash of their member record ids (two
@@ -178,57 +229,54 @@ def _mpc_core(table):
     num_samples = len(np.unique(np.asarray(table["iteration"].to_numpy())))
     codes = codes_i.astype(np.uint64)
     off64 = inner_offsets.astype(np.int64)
+    nat = _native_mpc()
+    if nat is not None:
+        import torch
 
-    def mix(x):  # splitmix64 finalizer
-        x = (x + np.uint64(0x9E3779B97F4A7C15))
-        x ^= x >> np.uint64(30)
-        x *= np.uint64(0xBF58476D1CE4E5B9)
-        x ^= x >> np.uint64(27)
-        x *= np.uint64(0x94D049BB133111EB)
-        x ^= x >> np.uint64(31)
-        return x
-
-    with np.errstate(over="ignore"):
-        h1 = mix(codes)
-        # second independent sum: a nonlinear remix of h1 (h1 is already
-        # well-mixed, so one multiply round suffices for pair independence)
-        h2 = (h1 ^ (h1 >> np.uint64(29))) * np.uint64(0xD6E8FEB86659FD93)
-        h2 ^= h2 >> np.uint64(32)
-        # order-independent per-cluster aggregates: clusters are contiguous,
-        # so segment sums come from one zero-prefixed cumsum (mod-2^64
-        # wraparound is part of the hash; np.add.reduceat pays per-segment
-        # overhead at ~50M segments, and the zero prefix removes all
-        # boundary special-casing)
-        n_flat = len(h1)
-        cc1 = np.empty(n_flat + 1, dtype=np.uint64)
-        cc1[0] = np.uint64(0)
-        np.cumsum(h1, out=cc1[1:])
-        cc2 = np.empty(n_flat + 1, dtype=np.uint64)
-        cc2[0] = np.uint64(0)
-        np.cumsum(h2, out=cc2[1:])
-        s1 = cc1[off64[1:]] - cc1[off64[:-1]]
-        s2 = cc2[off64[1:]] - cc2[off64[:-1]]
-        key = (s1 ^ (s2 * np.uint64(0x9E3779B97F4A7C15))) + cluster_sizes.astype(
-            np.uint64)
+        c32 = np.ascontiguousarray(codes_i, dtype=np.int32)
+        if not c32.flags.writeable:  # arrow buffers are read-only
+            c32 = c32.copy()
+        codes32 = torch.from_numpy(c32)
+        off_t = torch.from_numpy(np.ascontiguousarray(off64))
+        key = nat.mpc_cluster_keys(codes32, off_t).numpy().view(np.uint64)
+    else:
+        with np.errstate(over="ignore"):
+            key = _mpc_keys_numpy(codes, off64, cluster_sizes)
     # frequency of each distinct cluster content
     kcode, _ = pd.factorize(key)
     kcounts = np.bincount(kcode)
     freq_per_cluster = kcounts[kcode] / num_samples
-    # one representative cluster instance per distinct key: reversed fancy
-    # assignment leaves the FIRST occurrence (no sort)
-    first_idx = np.zeros(len(kcounts), dtype=np.int64)
-    first_idx[kcode[::-1]] = np.arange(len(kcode) - 1, -1, -1, dtype=np.int64)
+    # one representative cluster instance per distinct key: first occurrence
+    if nat is not None:
+        first_idx = nat.first_occurrence(
+            __import__("torch").from_numpy(
+                np.ascontiguousarray(kcode, dtype=np.int64)),
+            int(len(kcounts))).numpy()
+    else:
+        # reversed fancy assignment leaves the FIRST occurrence (no sort)
+        first_idx = np.zeros(len(kcounts), dtype=np.int64)
+        first_idx[kcode[::-1]] = np.arange(len(kcode) - 1, -1, -1,
+                                           dtype=np.int64)
     # per record-code: best (max count, first entry on ties) cluster entry
     # via ONE composite int64 sort: code | count | inverted entry index
-    entry_counts = np.repeat(kcounts[kcode].astype(np.int64), cluster_sizes)
     n_ent = len(codes)
     idx_bits = max(1, int(n_ent - 1).bit_length())
     cnt_bits = max(1, int(num_samples).bit_length())
     code_bits = max(1, int(len(uniq_rids) - 1).bit_length())
     if code_bits + cnt_bits + idx_bits <= _PACK_BITS:
-        inv_idx = (np.int64(n_ent - 1) - np.arange(n_ent, dtype=np.int64))
-        combo = ((codes.astype(np.int64) << np.int64(cnt_bits + idx_bits))
-                 | (entry_counts << np.int64(idx_bits)) | inv_idx)
+        if nat is not None:
+            import torch
+
+            combo = nat.mpc_combo(
+                codes32, off_t,
+                torch.from_numpy(kcounts[kcode].astype(np.int64)),
+                int(cnt_bits), int(idx_bits)).numpy()
+        else:
+            entry_counts = np.repeat(kcounts[kcode].astype(np.int64),
+                                     cluster_sizes)
+            inv_idx = (np.int64(n_ent - 1) - np.arange(n_ent, dtype=np.int64))
+            combo = ((codes.astype(np.int64) << np.int64(cnt_bits + idx_bits))
+                     | (entry_counts << np.int64(idx_bits)) | inv_idx)
         combo.sort()
         dec_code = combo >> np.int64(cnt_bits + idx_bits)
         last = np.flatnonzero(np.r_[dec_code[1:] != dec_code[:-1], True])
@@ -237,6 +285,8 @@ def _mpc_core(table):
                            & ((np.int64(1) << np.int64(idx_bits)) - 1)))
         best_codes = dec_code[last]
     else:  # astronomically large chains: 3-key lexsort instead of packing
+        entry_counts = np.repeat(kcounts[kcode].astype(np.int64),
+                                 cluster_sizes)
         order = np.lexsort((-np.arange(n_ent, dtype=np.int64), entry_counts,
                             codes.astype(np.int64)))
         c_sorted = codes[order].astype(np.int64)

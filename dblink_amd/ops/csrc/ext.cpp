@@ -28,6 +28,14 @@ std::tuple<torch::Tensor, int64_t> pcg1_link_cpu(
 
 torch::Tensor counting_argsort_cpu(torch::Tensor keys, int64_t k);
 
+torch::Tensor mpc_cluster_keys(torch::Tensor codes, torch::Tensor offsets);
+
+torch::Tensor mpc_combo(torch::Tensor codes, torch::Tensor offsets,
+                        torch::Tensor cluster_counts, int64_t cnt_bits,
+                        int64_t idx_bits);
+
+torch::Tensor first_occurrence(torch::Tensor kcode, int64_t k);
+
 std::tuple<double, int64_t, torch::Tensor, torch::Tensor> summary_cpu(
     torch::Tensor rec_values, torch::Tensor rec_dist, torch::Tensor rec_file,
     torch::Tensor rec_ent, torch::Tensor ent_values, torch::Tensor probs,
@@ -178,6 +186,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "PCG-II dense link update (OpenMP, f64 log-space)");
   m.def("counting_argsort_cpu", &dblink::counting_argsort_cpu,
         "stable counting argsort for small-range int64 keys");
+  m.def("mpc_cluster_keys", &dblink::mpc_cluster_keys,
+        "order-independent cluster content hashes (MPC/sMPC core)");
+  m.def("mpc_combo", &dblink::mpc_combo,
+        "composite (code | count | inv index) sort keys (MPC core)");
+  m.def("first_occurrence", &dblink::first_occurrence,
+        "first index of each key value");
   m.def("distortion_update_cpu", &dblink::distortion_update_cpu,
         "distortion resample (OpenMP, bitwise-matches the numpy fast path)");
   m.def("summary_cpu", &dblink::summary_cpu,

@@ -658,3 +658,74 @@ std::tuple<double, int64_t, torch::Tensor, torch::Tensor> summary_cpu(
 }
 
 }  // namespace dblink
+
+namespace dblink {
+
+// MPC/sMPC integer core helpers (analysis/chain.py): order-independent
+// cluster content keys, composite sort keys and first-occurrence — all
+// integer arithmetic, bitwise-identical to the numpy expressions they
+// replace (mod-2^64 sums commute).
+
+static inline uint64_t splitmix64(uint64_t x) {
+  x += 0x9E3779B97F4A7C15ull;
+  x ^= x >> 30;
+  x *= 0xBF58476D1CE4E5B9ull;
+  x ^= x >> 27;
+  x *= 0x94D049BB133111EBull;
+  x ^= x >> 31;
+  return x;
+}
+
+torch::Tensor mpc_cluster_keys(torch::Tensor codes, torch::Tensor offsets) {
+  const int64_t C = offsets.numel() - 1;
+  auto out = torch::empty({C}, torch::kInt64);
+  const int32_t* cp = codes.data_ptr<int32_t>();
+  const int64_t* op = offsets.data_ptr<int64_t>();
+  int64_t* kp = out.data_ptr<int64_t>();
+#pragma omp parallel for schedule(static)
+  for (int64_t c = 0; c < C; ++c) {
+    uint64_t s1 = 0, s2 = 0;
+    for (int64_t j = op[c]; j < op[c + 1]; ++j) {
+      const uint64_t h1 = splitmix64((uint64_t)(uint32_t)cp[j]);
+      uint64_t h2 = (h1 ^ (h1 >> 29)) * 0xD6E8FEB86659FD93ull;
+      h2 ^= h2 >> 32;
+      s1 += h1;
+      s2 += h2;
+    }
+    const uint64_t key =
+        (s1 ^ (s2 * 0x9E3779B97F4A7C15ull)) + (uint64_t)(op[c + 1] - op[c]);
+    kp[c] = (int64_t)key;
+  }
+  return out;
+}
+
+torch::Tensor mpc_combo(torch::Tensor codes, torch::Tensor offsets,
+                        torch::Tensor cluster_counts, int64_t cnt_bits,
+                        int64_t idx_bits) {
+  const int64_t C = offsets.numel() - 1;
+  const int64_t n = codes.numel();
+  auto out = torch::empty({n}, torch::kInt64);
+  const int32_t* cp = codes.data_ptr<int32_t>();
+  const int64_t* op = offsets.data_ptr<int64_t>();
+  const int64_t* cc = cluster_counts.data_ptr<int64_t>();
+  int64_t* ob = out.data_ptr<int64_t>();
+#pragma omp parallel for schedule(static)
+  for (int64_t c = 0; c < C; ++c) {
+    const int64_t cnt_sh = cc[c] << idx_bits;
+    for (int64_t j = op[c]; j < op[c + 1]; ++j)
+      ob[j] = ((int64_t)cp[j] << (cnt_bits + idx_bits)) | cnt_sh | (n - 1 - j);
+  }
+  return out;
+}
+
+torch::Tensor first_occurrence(torch::Tensor kcode, int64_t k) {
+  const int64_t C = kcode.numel();
+  auto out = torch::full({k}, -1, torch::kInt64);
+  const int64_t* kp = kcode.data_ptr<int64_t>();
+  int64_t* op = out.data_ptr<int64_t>();
+  for (int64_t i = 0; i < C; ++i)
+    if (op[kp[i]] < 0) op[kp[i]] = i;
+  return out;
+}
+
+}  // namespace dblink
