@@ -210,3 +210,54 @@ def test_fast_mpc_lexsort_fallback(monkeypatch):
         assert ref[rid][1] == pytest.approx(fast[rid][1])
     assert (sorted(map(sorted, shared_most_probable_clusters(table)))
             == sorted(map(sorted, shared_most_probable_clusters_fast(table))))
+
+
+def test_fast_mpc_matches_reference_fuzz():
+    """Property fuzz: on random linkage chains (random cluster counts,
+    sizes, partition splits, missing records per sample) the vectorized
+    MPC/sMPC must agree with the direct python implementation."""
+    import pyarrow as pa
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    from dblink_amd.analysis.chain import (
+        most_probable_clusters,
+        most_probable_clusters_fast,
+        shared_most_probable_clusters,
+        shared_most_probable_clusters_fast,
+    )
+
+    @settings(max_examples=40, deadline=None)
+    @given(st.integers(0, 2**31 - 1), st.integers(4, 25), st.integers(2, 8))
+    def check(seed, n_rec, n_samples):
+        rng = np.random.default_rng(seed)
+        rec = [f"r{i}" for i in range(n_rec)]
+        rows = {"iteration": [], "partitionId": [], "linkageStructure": []}
+        for it in range(n_samples):
+            # random subset of records present (a record can be absent from
+            # a partition's sample), random clustering, random 2-way split
+            present = np.flatnonzero(rng.random(n_rec) < 0.9)
+            labels = rng.integers(0, max(1, n_rec // 2), len(present))
+            clusters = {}
+            for r, l in zip(present, labels):
+                clusters.setdefault(int(l), []).append(rec[r])
+            cl = list(clusters.values())
+            half = rng.integers(0, len(cl) + 1)
+            for pid, part in ((0, cl[:half]), (1, cl[half:])):
+                rows["iteration"].append(it)
+                rows["partitionId"].append(pid)
+                rows["linkageStructure"].append(part)
+        table = pa.table(rows, schema=pa.schema([
+            ("iteration", pa.int64()), ("partitionId", pa.int32()),
+            ("linkageStructure", pa.list_(pa.list_(pa.string()))),
+        ]))
+        ref = most_probable_clusters(table)
+        fast = most_probable_clusters_fast(table)
+        assert set(ref) == set(fast)
+        for rid in ref:
+            assert ref[rid][0] == fast[rid][0], rid
+            assert abs(ref[rid][1] - fast[rid][1]) < 1e-12
+        assert (sorted(map(sorted, shared_most_probable_clusters(table)))
+                == sorted(map(sorted, shared_most_probable_clusters_fast(table))))
+
+    check()
